@@ -1,0 +1,136 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: GPT-3-6.7B, Fleet sharding stage-3, bf16.
+
+Driver contract (BASELINE.json): tokens/sec whole-node on N GPUs of one
+node, weak scaling (fixed per-GPU batch), synthetic data, random-init
+weights.  Launched directly (N=1) or via torch.distributed.run with one
+rank per GPU over RCCL.
+
+  python bench.py --gpus N --steps K --warmup W
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=6)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--model", type=str, default="gpt3-6.7b")
+    ap.add_argument("--batch", type=int, default=2, help="per-GPU micro batch")
+    ap.add_argument("--seq", type=int, default=2048)
+    ap.add_argument("--sharding-stage", type=int, default=3)
+    ap.add_argument("--recompute", action="store_true")
+    ap.add_argument("--cpu-smoke", action="store_true",
+                    help="tiny CPU run for plumbing checks")
+    args = ap.parse_args()
+
+    import paddle_amd as paddle
+    from paddle_amd.distributed import fleet
+    from paddle_amd.distributed.fleet.sharding import (GroupShardedStage3,
+                                                       ShardedAdamW)
+    from paddle_amd.models import GPTPretrainingCriterion, build_gpt
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    on_gpu = torch.cuda.is_available() and not args.cpu_smoke
+
+    if world > 1:
+        paddle.distributed.init_parallel_env()
+    if on_gpu:
+        local = int(os.environ.get("LOCAL_RANK", "0"))
+        torch.cuda.set_device(local)
+        paddle.set_device(f"gpu:{local}")
+
+    model_name = args.model
+    if args.cpu_smoke:
+        model_name = "gpt3-tiny"
+        args.seq = min(args.seq, 128)
+
+    paddle.seed(1234 + rank)
+    model = build_gpt(model_name, max_seq_len=args.seq,
+                      use_recompute=args.recompute)
+    loss_fn = GPTPretrainingCriterion()
+    dtype = torch.bfloat16 if on_gpu else torch.float32
+    dev = torch.device("cuda", torch.cuda.current_device()) if on_gpu else torch.device("cpu")
+    model = model.to(device=dev, dtype=dtype)
+
+    wrapped = GroupShardedStage3(model, device=dev)
+    opt = ShardedAdamW(wrapped, learning_rate=1e-4, beta1=0.9, beta2=0.95,
+                       epsilon=1e-8, weight_decay=0.1,
+                       grad_clip=paddle.nn.ClipGradByGlobalNorm(1.0))
+
+    vocab = model.cfg.vocab_size
+    ids = torch.randint(0, vocab, (args.batch, args.seq), device=dev)
+    labels = torch.randint(0, vocab, (args.batch, args.seq), device=dev)
+
+    def step():
+        loss = loss_fn(wrapped(ids), labels)
+        loss.backward()
+        opt.step()
+        opt.clear_grad()
+        return loss
+
+    for _ in range(args.warmup):
+        step()
+
+    if world > 1:
+        paddle.distributed.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = step()
+    if on_gpu:
+        torch.cuda.synchronize()
+    if world > 1:
+        paddle.distributed.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=dev if on_gpu else "cpu")
+        paddle.distributed.all_reduce(t, op=paddle.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world if world > 1 else (1 if on_gpu else 0) or 1
+    total_tokens = args.batch * args.seq * args.steps * n_gpus
+    tps = total_tokens / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        out = {
+            "metric": "tokens/sec (whole node) GPT-3-6.7B Fleet sharding-3",
+            "value": round(tps, 1),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if on_gpu else "float32",
+            "data": "synthetic",
+            "config": {
+                "model": model_name,
+                "global_batch": args.batch * n_gpus,
+                "seq_len": args.seq,
+                "parallelism": f"sharding3_dp{n_gpus}",
+                "loss": round(float(loss.detach().float().cpu()), 4),
+                "peak_mem_gb": (round(torch.cuda.max_memory_allocated() / 2**30, 2)
+                                if on_gpu else None),
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

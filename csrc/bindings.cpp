@@ -1,0 +1,351 @@
+// torch binding layer for paddle_amd._C (compiled host-side by g++;
+// kernels live in kernels/*.hip compiled by hipcc for gfx950).
+//
+// Replaces the reference's pybind layer for our op set
+// (paddle/fluid/pybind/eager_functions.cc + generated python_c) -- here
+// the autograd glue lives in Python (torch.autograd.Function); this file
+// only validates tensors and launches.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include "kernels/api.h"
+
+namespace pa {
+void mfma_probe(const void* a, const void* bt, float* c, hipStream_t s);
+}
+
+namespace {
+
+using torch::Tensor;
+
+hipStream_t cur_stream() {
+  return (hipStream_t)c10::hip::getCurrentHIPStream().stream();
+}
+
+int dt_of(const Tensor& t) {
+  switch (t.scalar_type()) {
+    case torch::kBFloat16: return pa::kBF16;
+    case torch::kFloat: return pa::kF32;
+    case torch::kHalf: return pa::kF16;
+    default: TORCH_CHECK(false, "unsupported dtype ", t.scalar_type());
+  }
+  return -1;
+}
+
+#define CHECK_IN(t) \
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous(), #t " must be contiguous GPU tensor")
+
+// ---- norms ----------------------------------------------------------------
+std::vector<Tensor> layer_norm_fwd(const Tensor& x, const Tensor& w,
+                                   const c10::optional<Tensor>& b, double eps) {
+  CHECK_IN(x); CHECK_IN(w);
+  int64_t d = x.size(-1), n = x.numel() / d;
+  TORCH_CHECK(d % 8 == 0, "layer_norm: D must be multiple of 8");
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({n}, x.options().dtype(torch::kFloat));
+  auto rstd = torch::empty({n}, x.options().dtype(torch::kFloat));
+  pa::layer_norm_fwd(x.const_data_ptr(), w.const_data_ptr(),
+                     b.has_value() ? b->const_data_ptr() : nullptr,
+                     y.mutable_data_ptr(), mean.mutable_data_ptr<float>(),
+                     rstd.mutable_data_ptr<float>(), n, d, (float)eps, dt_of(x),
+                     cur_stream());
+  return {y, mean, rstd};
+}
+
+std::vector<Tensor> layer_norm_bwd(const Tensor& dy, const Tensor& x,
+                                   const Tensor& w, const Tensor& mean,
+                                   const Tensor& rstd, bool has_bias) {
+  CHECK_IN(dy); CHECK_IN(x); CHECK_IN(w);
+  int64_t d = x.size(-1), n = x.numel() / d;
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({d}, x.options().dtype(torch::kFloat));
+  auto db = torch::zeros({d}, x.options().dtype(torch::kFloat));
+  pa::layer_norm_bwd_dx(dy.const_data_ptr(), x.const_data_ptr(), w.const_data_ptr(),
+                        mean.const_data_ptr<float>(), rstd.const_data_ptr<float>(),
+                        dx.mutable_data_ptr(), n, d, dt_of(x), cur_stream());
+  pa::layer_norm_bwd_dwdb(dy.const_data_ptr(), x.const_data_ptr(),
+                          mean.const_data_ptr<float>(), rstd.const_data_ptr<float>(),
+                          dw.mutable_data_ptr<float>(), db.mutable_data_ptr<float>(),
+                          n, d, dt_of(x), cur_stream());
+  return {dx, dw.to(x.scalar_type()), db.to(x.scalar_type())};
+}
+
+std::vector<Tensor> rms_norm_fwd(const Tensor& x, const c10::optional<Tensor>& residual,
+                                 const Tensor& w, double eps) {
+  CHECK_IN(x); CHECK_IN(w);
+  int64_t d = x.size(-1), n = x.numel() / d;
+  TORCH_CHECK(d % 8 == 0, "rms_norm: D must be multiple of 8");
+  auto y = torch::empty_like(x);
+  auto rstd = torch::empty({n}, x.options().dtype(torch::kFloat));
+  Tensor res_out;
+  const void* resp = nullptr;
+  void* res_outp = nullptr;
+  if (residual.has_value()) {
+    CHECK_IN((*residual));
+    res_out = torch::empty_like(x);
+    resp = residual->const_data_ptr();
+    res_outp = res_out.mutable_data_ptr();
+  }
+  pa::rms_norm_fwd(x.const_data_ptr(), resp, w.const_data_ptr(),
+                   y.mutable_data_ptr(), res_outp, rstd.mutable_data_ptr<float>(),
+                   n, d, (float)eps, dt_of(x), cur_stream());
+  if (residual.has_value()) return {y, rstd, res_out};
+  return {y, rstd};
+}
+
+std::vector<Tensor> rms_norm_bwd(const Tensor& dy, const Tensor& x, const Tensor& w,
+                                 const Tensor& rstd) {
+  CHECK_IN(dy); CHECK_IN(x); CHECK_IN(w);
+  int64_t d = x.size(-1), n = x.numel() / d;
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({d}, x.options().dtype(torch::kFloat));
+  pa::rms_norm_bwd_dx(dy.const_data_ptr(), x.const_data_ptr(), w.const_data_ptr(),
+                      rstd.const_data_ptr<float>(), dx.mutable_data_ptr(), n, d,
+                      dt_of(x), cur_stream());
+  pa::rms_norm_bwd_dw(dy.const_data_ptr(), x.const_data_ptr(),
+                      rstd.const_data_ptr<float>(), dw.mutable_data_ptr<float>(),
+                      n, d, dt_of(x), cur_stream());
+  return {dx, dw.to(x.scalar_type())};
+}
+
+// ---- cross entropy --------------------------------------------------------
+std::vector<Tensor> softmax_ce_fwd(const Tensor& logits, const Tensor& labels,
+                                   int64_t ignore_index) {
+  CHECK_IN(logits); CHECK_IN(labels);
+  int64_t v = logits.size(-1), n = logits.numel() / v;
+  auto loss = torch::empty({n}, logits.options().dtype(torch::kFloat));
+  auto lse = torch::empty({n}, logits.options().dtype(torch::kFloat));
+  pa::softmax_ce_fwd(logits.const_data_ptr(), labels.const_data_ptr<int64_t>(),
+                     loss.mutable_data_ptr<float>(), lse.mutable_data_ptr<float>(),
+                     n, v, ignore_index, dt_of(logits), cur_stream());
+  return {loss, lse};
+}
+
+Tensor softmax_ce_bwd(const Tensor& dloss, const Tensor& logits,
+                      const Tensor& labels, const Tensor& lse, int64_t ignore_index) {
+  CHECK_IN(logits);
+  int64_t v = logits.size(-1), n = logits.numel() / v;
+  auto dlogits = torch::empty_like(logits);
+  pa::softmax_ce_bwd(dloss.const_data_ptr<float>(), logits.const_data_ptr(),
+                     labels.const_data_ptr<int64_t>(), lse.const_data_ptr<float>(),
+                     dlogits.mutable_data_ptr(), n, v, ignore_index, dt_of(logits),
+                     cur_stream());
+  return dlogits;
+}
+
+// ---- elementwise ----------------------------------------------------------
+Tensor bias_gelu_fwd(const Tensor& x, const c10::optional<Tensor>& bias) {
+  CHECK_IN(x);
+  int64_t d = x.size(-1), n = x.numel() / d;
+  TORCH_CHECK(d % 8 == 0);
+  auto y = torch::empty_like(x);
+  pa::bias_gelu_fwd(x.const_data_ptr(), bias.has_value() ? bias->const_data_ptr() : nullptr,
+                    y.mutable_data_ptr(), n, d, dt_of(x), cur_stream());
+  return y;
+}
+
+Tensor bias_gelu_bwd(const Tensor& dy, const Tensor& x, const c10::optional<Tensor>& bias) {
+  CHECK_IN(dy); CHECK_IN(x);
+  int64_t d = x.size(-1), n = x.numel() / d;
+  auto dx = torch::empty_like(x);
+  pa::bias_gelu_bwd(dy.const_data_ptr(), x.const_data_ptr(),
+                    bias.has_value() ? bias->const_data_ptr() : nullptr,
+                    dx.mutable_data_ptr(), n, d, dt_of(x), cur_stream());
+  return dx;
+}
+
+Tensor swiglu_fwd(const Tensor& x) {
+  CHECK_IN(x);
+  int64_t d2 = x.size(-1), n = x.numel() / d2, d = d2 / 2;
+  TORCH_CHECK(d % 8 == 0);
+  auto sizes = x.sizes().vec();
+  sizes.back() = d;
+  auto y = torch::empty(sizes, x.options());
+  pa::swiglu_fwd(x.const_data_ptr(), y.mutable_data_ptr(), n, d, dt_of(x), cur_stream());
+  return y;
+}
+
+Tensor swiglu_bwd(const Tensor& dy, const Tensor& x) {
+  CHECK_IN(dy); CHECK_IN(x);
+  int64_t d2 = x.size(-1), n = x.numel() / d2, d = d2 / 2;
+  auto dx = torch::empty_like(x);
+  pa::swiglu_bwd(dy.const_data_ptr(), x.const_data_ptr(), dx.mutable_data_ptr(),
+                 n, d, dt_of(x), cur_stream());
+  return dx;
+}
+
+Tensor rope_fwd(const Tensor& x, const Tensor& cos_t, const Tensor& sin_t,
+                int64_t pos_offset, bool conj) {
+  CHECK_IN(x); CHECK_IN(cos_t); CHECK_IN(sin_t);
+  TORCH_CHECK(x.dim() == 4, "rope expects [B,S,H,D]");
+  int64_t b = x.size(0), sl = x.size(1), h = x.size(2), dh = x.size(3);
+  TORCH_CHECK((dh / 2) % 4 == 0, "rope: head_dim/2 must be multiple of 4");
+  auto y = torch::empty_like(x);
+  pa::rope_fwd(x.const_data_ptr(), cos_t.const_data_ptr<float>(),
+               sin_t.const_data_ptr<float>(), y.mutable_data_ptr(), b, sl, h, dh,
+               pos_offset, conj, dt_of(x), cur_stream());
+  return y;
+}
+
+Tensor colsum(const Tensor& x) {
+  CHECK_IN(x);
+  int64_t d = x.size(-1), n = x.numel() / d;
+  auto out = torch::zeros({d}, x.options().dtype(torch::kFloat));
+  pa::colsum(x.const_data_ptr(), out.mutable_data_ptr<float>(), n, d, dt_of(x),
+             cur_stream());
+  return out;
+}
+
+// ---- adamw ----------------------------------------------------------------
+void adamw(Tensor& master, c10::optional<Tensor> param_out, const Tensor& grad,
+           Tensor& m, Tensor& v, double lr, double beta1, double beta2,
+           double eps, double wd, double beta1_pow, double beta2_pow) {
+  CHECK_IN(master); CHECK_IN(grad); CHECK_IN(m); CHECK_IN(v);
+  int64_t numel = master.numel();
+  TORCH_CHECK(numel % 4 == 0, "adamw: flat shard numel must be multiple of 4");
+  bool bf16out = false;
+  void* pout = nullptr;
+  if (param_out.has_value()) {
+    bf16out = param_out->scalar_type() == torch::kBFloat16;
+    pout = param_out->mutable_data_ptr();
+  }
+  pa::adamw(master.mutable_data_ptr<float>(), pout, grad.const_data_ptr(),
+            m.mutable_data_ptr<float>(), v.mutable_data_ptr<float>(), numel,
+            (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd,
+            (float)beta1_pow, (float)beta2_pow, dt_of(grad), bf16out,
+            cur_stream());
+}
+
+Tensor l2norm_sq(const Tensor& x) {
+  CHECK_IN(x);
+  auto out = torch::zeros({1}, x.options().dtype(torch::kFloat));
+  pa::l2norm_sq(x.const_data_ptr(), out.mutable_data_ptr<float>(), x.numel(),
+                dt_of(x), cur_stream());
+  return out;
+}
+
+// ---- flash attention ------------------------------------------------------
+std::vector<Tensor> flash_attn_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
+                                   double scale, bool causal) {
+  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v);
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "flash_attn: bf16 only");
+  TORCH_CHECK(q.dim() == 4, "flash_attn expects [B,H,S,D]");
+  int64_t b = q.size(0), h = q.size(1), sq = q.size(2), d = q.size(3);
+  int64_t hkv = k.size(1), skv = k.size(2);
+  TORCH_CHECK(d == 128 || d == 64, "flash_attn: head_dim must be 64/128");
+  TORCH_CHECK(h % hkv == 0);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({b, h, sq}, q.options().dtype(torch::kFloat));
+  pa::flash_attn_fwd(q.const_data_ptr(), k.const_data_ptr(), v.const_data_ptr(),
+                     o.mutable_data_ptr(), lse.mutable_data_ptr<float>(), b, h,
+                     hkv, sq, skv, d, (float)scale, causal, cur_stream());
+  return {o, lse};
+}
+
+std::vector<Tensor> flash_attn_bwd(const Tensor& dout, const Tensor& q, const Tensor& k,
+                                   const Tensor& v, const Tensor& o, const Tensor& lse,
+                                   double scale, bool causal) {
+  CHECK_IN(dout); CHECK_IN(q); CHECK_IN(k); CHECK_IN(v); CHECK_IN(o);
+  int64_t b = q.size(0), h = q.size(1), sq = q.size(2), d = q.size(3);
+  int64_t hkv = k.size(1), skv = k.size(2);
+  TORCH_CHECK(hkv == h, "flash_attn_bwd kernel requires hkv==h (expand KV upstream)");
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto delta = torch::empty({b, h, sq}, q.options().dtype(torch::kFloat));
+  pa::flash_attn_bwd(dout.const_data_ptr(), q.const_data_ptr(), k.const_data_ptr(),
+                     v.const_data_ptr(), o.const_data_ptr(), lse.const_data_ptr<float>(),
+                     dq.mutable_data_ptr(), dk.mutable_data_ptr(), dv.mutable_data_ptr(),
+                     delta.mutable_data_ptr<float>(), b, h, hkv, sq, skv, d,
+                     (float)scale, causal, cur_stream());
+  return {dq, dk, dv};
+}
+
+// ---- dropout_add ----------------------------------------------------------
+std::vector<Tensor> dropout_add_fwd(const Tensor& x, const c10::optional<Tensor>& residual,
+                                    double p, int64_t seed, int64_t offset) {
+  CHECK_IN(x);
+  auto y = torch::empty_like(x);
+  Tensor mask;
+  uint8_t* maskp = nullptr;
+  if (p > 0) {
+    mask = torch::empty(x.sizes(), x.options().dtype(torch::kUInt8));
+    maskp = mask.mutable_data_ptr<uint8_t>();
+  }
+  pa::dropout_add_fwd(x.const_data_ptr(),
+                      residual.has_value() ? residual->const_data_ptr() : nullptr,
+                      y.mutable_data_ptr(), maskp, x.numel(), (float)p,
+                      (uint64_t)seed, (uint64_t)offset, dt_of(x), cur_stream());
+  if (p > 0) return {y, mask};
+  return {y};
+}
+
+Tensor dropout_add_bwd(const Tensor& dy, const Tensor& mask, double p) {
+  CHECK_IN(dy); CHECK_IN(mask);
+  auto dx = torch::empty_like(dy);
+  pa::dropout_add_bwd(dy.const_data_ptr(), mask.const_data_ptr<uint8_t>(),
+                      dx.mutable_data_ptr(), dy.numel(), (float)p, dt_of(dy),
+                      cur_stream());
+  return dx;
+}
+
+// ---- embedding ------------------------------------------------------------
+Tensor embedding_fwd(const Tensor& table, const Tensor& ids, int64_t padding_idx) {
+  CHECK_IN(table); CHECK_IN(ids);
+  int64_t vocab = table.size(0), d = table.size(1), n = ids.numel();
+  TORCH_CHECK(d % 8 == 0);
+  auto sizes = ids.sizes().vec();
+  sizes.push_back(d);
+  auto out = torch::empty(sizes, table.options());
+  pa::embedding_fwd(table.const_data_ptr(), ids.const_data_ptr<int64_t>(),
+                    out.mutable_data_ptr(), n, d, vocab, padding_idx,
+                    dt_of(table), cur_stream());
+  return out;
+}
+
+Tensor embedding_bwd(const Tensor& dout, const Tensor& ids, int64_t vocab,
+                     int64_t padding_idx) {
+  CHECK_IN(dout); CHECK_IN(ids);
+  int64_t d = dout.size(-1), n = ids.numel();
+  auto dtable = torch::zeros({vocab, d}, dout.options().dtype(torch::kFloat));
+  pa::embedding_bwd(dout.const_data_ptr(), ids.const_data_ptr<int64_t>(),
+                    dtable.mutable_data_ptr<float>(), n, d, vocab, padding_idx,
+                    dt_of(dout), cur_stream());
+  return dtable;
+}
+
+// ---- probe ----------------------------------------------------------------
+Tensor mfma_probe(const Tensor& a, const Tensor& bt) {
+  CHECK_IN(a); CHECK_IN(bt);
+  auto c = torch::empty({16, 16}, a.options().dtype(torch::kFloat));
+  pa::mfma_probe(a.const_data_ptr(), bt.const_data_ptr(),
+                 c.mutable_data_ptr<float>(), cur_stream());
+  return c;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layer_norm_fwd", &layer_norm_fwd);
+  m.def("layer_norm_bwd", &layer_norm_bwd);
+  m.def("rms_norm_fwd", &rms_norm_fwd);
+  m.def("rms_norm_bwd", &rms_norm_bwd);
+  m.def("softmax_ce_fwd", &softmax_ce_fwd);
+  m.def("softmax_ce_bwd", &softmax_ce_bwd);
+  m.def("bias_gelu_fwd", &bias_gelu_fwd);
+  m.def("bias_gelu_bwd", &bias_gelu_bwd);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("rope_fwd", &rope_fwd);
+  m.def("colsum", &colsum);
+  m.def("adamw", &adamw);
+  m.def("l2norm_sq", &l2norm_sq);
+  m.def("flash_attn_fwd", &flash_attn_fwd);
+  m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("dropout_add_fwd", &dropout_add_fwd);
+  m.def("dropout_add_bwd", &dropout_add_bwd);
+  m.def("embedding_fwd", &embedding_fwd);
+  m.def("embedding_bwd", &embedding_bwd);
+  m.def("mfma_probe", &mfma_probe);
+  m.attr("compiled_arch") = "gfx950";
+}

@@ -1,0 +1,108 @@
+// C ABI between the torch binding layer (bindings.cpp, compiled by g++)
+// and the gfx950 kernel library (*.hip, compiled by hipcc).
+// Raw pointers + hipStream_t only -- no torch types cross this line.
+#pragma once
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+namespace pa {
+
+// dtype tags for the few we support natively
+enum DType : int { kBF16 = 0, kF32 = 1, kF16 = 2 };
+
+// ---- norms ----------------------------------------------------------------
+void layer_norm_fwd(const void* x, const void* w, const void* b, void* y,
+                    float* mean, float* rstd, int64_t n, int64_t d, float eps,
+                    int dtype, hipStream_t s);
+void layer_norm_bwd_dx(const void* dy, const void* x, const void* w,
+                       const float* mean, const float* rstd, void* dx,
+                       int64_t n, int64_t d, int dtype, hipStream_t s);
+void layer_norm_bwd_dwdb(const void* dy, const void* x, const float* mean,
+                         const float* rstd, float* dw, float* db, int64_t n,
+                         int64_t d, int dtype, hipStream_t s);
+
+void rms_norm_fwd(const void* x, const void* residual, const void* w, void* y,
+                  void* res_out, float* rstd, int64_t n, int64_t d, float eps,
+                  int dtype, hipStream_t s);
+void rms_norm_bwd_dx(const void* dy, const void* x, const void* w,
+                     const float* rstd, void* dx, int64_t n, int64_t d,
+                     int dtype, hipStream_t s);
+void rms_norm_bwd_dw(const void* dy, const void* x, const float* rstd,
+                     float* dw, int64_t n, int64_t d, int dtype, hipStream_t s);
+
+// ---- fused softmax cross-entropy ------------------------------------------
+// logits [n, v]; labels int64 [n]; loss/lse fp32 [n]
+void softmax_ce_fwd(const void* logits, const int64_t* labels, float* loss,
+                    float* lse, int64_t n, int64_t v, int64_t ignore_index,
+                    int dtype, hipStream_t s);
+// dlogits written in logits dtype; dloss fp32 [n]
+void softmax_ce_bwd(const float* dloss, const void* logits,
+                    const int64_t* labels, const float* lse, void* dlogits,
+                    int64_t n, int64_t v, int64_t ignore_index, int dtype,
+                    hipStream_t s);
+
+// ---- elementwise fusions ---------------------------------------------------
+// y = gelu(x + bias); bias [d] may be null
+void bias_gelu_fwd(const void* x, const void* bias, void* y, int64_t n,
+                   int64_t d, int dtype, hipStream_t s);
+// dx = gelu'(x+bias) * dy  (dbias reduced by caller or dwdb-style kernel)
+void bias_gelu_bwd(const void* dy, const void* x, const void* bias, void* dx,
+                   int64_t n, int64_t d, int dtype, hipStream_t s);
+
+// swiglu: x [n, 2d] = [gate | up]; y [n, d] = silu(gate) * up
+void swiglu_fwd(const void* x, void* y, int64_t n, int64_t d, int dtype,
+                hipStream_t s);
+void swiglu_bwd(const void* dy, const void* x, void* dx, int64_t n, int64_t d,
+                int dtype, hipStream_t s);
+
+// rope (neox rotate-half): qk [b, s, h, dh], cos/sin fp32 [s, dh/2]
+// backward == forward with conj=true (sin negated)
+void rope_fwd(const void* x, const float* cos_t, const float* sin_t, void* y,
+              int64_t b, int64_t sl, int64_t h, int64_t dh, int64_t pos_offset,
+              bool conj, int dtype, hipStream_t s);
+
+// column-sum of [n, d] into fp32 [d] (bias grads)
+void colsum(const void* x, float* out, int64_t n, int64_t d, int dtype,
+            hipStream_t s);
+
+// ---- fused AdamW (flat shards; fp32 master + bf16 model params) -----------
+void adamw(float* master, void* param_bf16, const void* grad, float* m,
+           float* v, int64_t numel, float lr, float beta1, float beta2,
+           float eps, float wd, float beta1_pow, float beta2_pow,
+           int grad_dtype, bool param_out_bf16, hipStream_t s);
+
+// multi-tensor l2-norm^2 of a flat fp32/bf16 buffer -> out[0] (fp32, add)
+void l2norm_sq(const void* x, float* out, int64_t numel, int dtype,
+               hipStream_t s);
+
+// ---- flash attention (bf16, head_dim 128 or 64) ----------------------------
+// q,k,v,o: [b, h, s, dh] contiguous bf16; lse fp32 [b, h, s]
+void flash_attn_fwd(const void* q, const void* k, const void* v, void* o,
+                    float* lse, int64_t b, int64_t h, int64_t hkv, int64_t sq,
+                    int64_t skv, int64_t dh, float scale, bool causal,
+                    hipStream_t s);
+// delta fp32 [b,h,sq] workspace computed internally
+void flash_attn_bwd(const void* dout, const void* q, const void* k,
+                    const void* v, const void* o, const float* lse,
+                    void* dq, void* dk, void* dv, float* delta, int64_t b,
+                    int64_t h, int64_t hkv, int64_t sq, int64_t skv,
+                    int64_t dh, float scale, bool causal, hipStream_t s);
+
+// ---- dropout + residual add ------------------------------------------------
+// y = dropout(x, p) + residual; mask stored as uint8 per element
+void dropout_add_fwd(const void* x, const void* residual, void* y,
+                     uint8_t* mask, int64_t numel, float p, uint64_t seed,
+                     uint64_t offset, int dtype, hipStream_t s);
+void dropout_add_bwd(const void* dy, const uint8_t* mask, void* dx,
+                     int64_t numel, float p, int dtype, hipStream_t s);
+
+// ---- embedding -------------------------------------------------------------
+void embedding_fwd(const void* table, const int64_t* ids, void* out,
+                   int64_t n_ids, int64_t d, int64_t vocab, int64_t padding_idx,
+                   int dtype, hipStream_t s);
+// grad table accumulated fp32 (atomics)
+void embedding_bwd(const void* dout, const int64_t* ids, float* dtable,
+                   int64_t n_ids, int64_t d, int64_t vocab, int64_t padding_idx,
+                   int dtype, hipStream_t s);
+
+}  // namespace pa

@@ -1,0 +1,312 @@
+// LayerNorm / RMSNorm forward+backward for gfx950.
+//
+// Design (MI355X): these are HBM-bandwidth-bound at D=4096..8192; the
+// kernels do one vectorized pass per row (256-thread block per row,
+// 16B/lane loads), fp32 accumulation, wave64 shuffle + 4-slot LDS block
+// reduction.  dgamma/dbeta use a column-parallel kernel with one fp32
+// atomic per (column, row-chunk).
+//
+// Reference behavior parity: paddle/phi/kernels/gpu/layer_norm_kernel.cu
+// (LayerNormFwdWithWelford) and rms_norm_kernel.cu (cuApplyRMSNorm) --
+// re-derived, not ported.
+#include "common.h"
+#include "api.h"
+
+namespace pa {
+
+// -------- type abstraction: load/store 8 elems as fp32 ---------------------
+template <int DT> struct VIO;
+
+template <> struct VIO<kBF16> {
+  using ST = short;
+  static __device__ __forceinline__ void load8(const void* p, int64_t idx, float* f) {
+    shortx8 v = *reinterpret_cast<const shortx8*>((const short*)p + idx);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) f[i] = bf2f(v[i]);
+  }
+  static __device__ __forceinline__ void store8(void* p, int64_t idx, const float* f) {
+    shortx8 v;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) v[i] = f2bf(f[i]);
+    *reinterpret_cast<shortx8*>((short*)p + idx) = v;
+  }
+  static __device__ __forceinline__ float load1(const void* p, int64_t idx) {
+    return bf2f(((const short*)p)[idx]);
+  }
+  static __device__ __forceinline__ void store1(void* p, int64_t idx, float f) {
+    ((short*)p)[idx] = f2bf(f);
+  }
+};
+
+template <> struct VIO<kF32> {
+  using ST = float;
+  static __device__ __forceinline__ void load8(const void* p, int64_t idx, float* f) {
+    const float4* q = reinterpret_cast<const float4*>((const float*)p + idx);
+    float4 a = q[0], b = q[1];
+    f[0] = a.x; f[1] = a.y; f[2] = a.z; f[3] = a.w;
+    f[4] = b.x; f[5] = b.y; f[6] = b.z; f[7] = b.w;
+  }
+  static __device__ __forceinline__ void store8(void* p, int64_t idx, const float* f) {
+    float4* q = reinterpret_cast<float4*>((float*)p + idx);
+    q[0] = make_float4(f[0], f[1], f[2], f[3]);
+    q[1] = make_float4(f[4], f[5], f[6], f[7]);
+  }
+  static __device__ __forceinline__ float load1(const void* p, int64_t idx) {
+    return ((const float*)p)[idx];
+  }
+  static __device__ __forceinline__ void store1(void* p, int64_t idx, float f) {
+    ((float*)p)[idx] = f;
+  }
+};
+
+// ---------------------------------------------------------------------------
+// LayerNorm forward: one 256-thread block per row.
+// ---------------------------------------------------------------------------
+template <int DT>
+__global__ void ln_fwd_kernel(const void* __restrict__ x, const void* __restrict__ w,
+                              const void* __restrict__ b, void* __restrict__ y,
+                              float* __restrict__ mean_out, float* __restrict__ rstd_out,
+                              int64_t n, int64_t d, float eps) {
+  __shared__ float red[8];
+  for (int64_t row = blockIdx.x; row < n; row += gridDim.x) {
+    const int64_t base = row * d;
+    float sum = 0.f, sumsq = 0.f;
+    for (int64_t i = threadIdx.x * 8; i < d; i += blockDim.x * 8) {
+      float f[8];
+      VIO<DT>::load8(x, base + i, f);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) { sum += f[k]; sumsq += f[k] * f[k]; }
+    }
+    float ts = block_reduce_256(sum, SumOp(), red, 0.f);
+    __syncthreads();
+    float tss = block_reduce_256(sumsq, SumOp(), red + 4, 0.f);
+    float mu = ts / d;
+    float var = fmaxf(tss / d - mu * mu, 0.f);
+    float rstd = rsqrtf(var + eps);
+    if (threadIdx.x == 0) { mean_out[row] = mu; rstd_out[row] = rstd; }
+    for (int64_t i = threadIdx.x * 8; i < d; i += blockDim.x * 8) {
+      float f[8], wf[8], bf[8];
+      VIO<DT>::load8(x, base + i, f);
+      VIO<DT>::load8(w, i, wf);
+      if (b) VIO<DT>::load8(b, i, bf);
+#pragma unroll
+      for (int k = 0; k < 8; ++k)
+        f[k] = (f[k] - mu) * rstd * wf[k] + (b ? bf[k] : 0.f);
+      VIO<DT>::store8(y, base + i, f);
+    }
+    __syncthreads();
+  }
+}
+
+// dx = rstd * (g - mean(g) - xhat * mean(g*xhat)),  g = dy*w
+template <int DT>
+__global__ void ln_bwd_dx_kernel(const void* __restrict__ dy, const void* __restrict__ x,
+                                 const void* __restrict__ w, const float* __restrict__ mean,
+                                 const float* __restrict__ rstd, void* __restrict__ dx,
+                                 int64_t n, int64_t d) {
+  __shared__ float red[8];
+  for (int64_t row = blockIdx.x; row < n; row += gridDim.x) {
+    const int64_t base = row * d;
+    const float mu = mean[row], rs = rstd[row];
+    float s1 = 0.f, s2 = 0.f;  // sum(g*xhat), sum(g)
+    for (int64_t i = threadIdx.x * 8; i < d; i += blockDim.x * 8) {
+      float gy[8], xf[8], wf[8];
+      VIO<DT>::load8(dy, base + i, gy);
+      VIO<DT>::load8(x, base + i, xf);
+      VIO<DT>::load8(w, i, wf);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float g = gy[k] * wf[k];
+        float xh = (xf[k] - mu) * rs;
+        s1 += g * xh;
+        s2 += g;
+      }
+    }
+    float t1 = block_reduce_256(s1, SumOp(), red, 0.f);
+    __syncthreads();
+    float t2 = block_reduce_256(s2, SumOp(), red + 4, 0.f);
+    const float inv_d = 1.f / d;
+    for (int64_t i = threadIdx.x * 8; i < d; i += blockDim.x * 8) {
+      float gy[8], xf[8], wf[8];
+      VIO<DT>::load8(dy, base + i, gy);
+      VIO<DT>::load8(x, base + i, xf);
+      VIO<DT>::load8(w, i, wf);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float g = gy[k] * wf[k];
+        float xh = (xf[k] - mu) * rs;
+        gy[k] = rs * (g - t2 * inv_d - xh * t1 * inv_d);
+      }
+      VIO<DT>::store8(dx, base + i, gy);
+    }
+    __syncthreads();
+  }
+}
+
+// dw[d] = sum_n dy*xhat ; db[d] = sum_n dy.  Column-parallel: thread owns a
+// column, walks a row chunk, one atomicAdd per (column, chunk).
+template <int DT, bool RMS>
+__global__ void ln_bwd_dwdb_kernel(const void* __restrict__ dy, const void* __restrict__ x,
+                                   const float* __restrict__ mean, const float* __restrict__ rstd,
+                                   float* __restrict__ dw, float* __restrict__ db,
+                                   int64_t n, int64_t d) {
+  int64_t col = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= d) return;
+  int64_t rows_per = (n + gridDim.y - 1) / gridDim.y;
+  int64_t r0 = (int64_t)blockIdx.y * rows_per;
+  int64_t r1 = min(n, r0 + rows_per);
+  float aw = 0.f, ab = 0.f;
+  for (int64_t r = r0; r < r1; ++r) {
+    float g = VIO<DT>::load1(dy, r * d + col);
+    float xf = VIO<DT>::load1(x, r * d + col);
+    float xh = RMS ? xf * rstd[r] : (xf - mean[r]) * rstd[r];
+    aw += g * xh;
+    ab += g;
+  }
+  atomicAdd(&dw[col], aw);
+  if (db) atomicAdd(&db[col], ab);
+}
+
+// ---------------------------------------------------------------------------
+// RMSNorm (optionally fused residual-add: xr = x + residual; y = rms(xr)*w,
+// res_out = xr) -- matches fused_rms_norm semantics.
+// ---------------------------------------------------------------------------
+template <int DT, bool HAS_RES>
+__global__ void rms_fwd_kernel(const void* __restrict__ x, const void* __restrict__ res,
+                               const void* __restrict__ w, void* __restrict__ y,
+                               void* __restrict__ res_out, float* __restrict__ rstd_out,
+                               int64_t n, int64_t d, float eps) {
+  __shared__ float red[4];
+  for (int64_t row = blockIdx.x; row < n; row += gridDim.x) {
+    const int64_t base = row * d;
+    float sumsq = 0.f;
+    for (int64_t i = threadIdx.x * 8; i < d; i += blockDim.x * 8) {
+      float f[8];
+      VIO<DT>::load8(x, base + i, f);
+      if (HAS_RES) {
+        float r[8];
+        VIO<DT>::load8(res, base + i, r);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) f[k] += r[k];
+        VIO<DT>::store8(res_out, base + i, f);
+      }
+#pragma unroll
+      for (int k = 0; k < 8; ++k) sumsq += f[k] * f[k];
+    }
+    if (HAS_RES) __syncthreads();  // res_out must be visible? (same thread re-reads its own slice only)
+    float tss = block_reduce_256(sumsq, SumOp(), red, 0.f);
+    float rstd = rsqrtf(tss / d + eps);
+    if (threadIdx.x == 0) rstd_out[row] = rstd;
+    const void* src = HAS_RES ? res_out : x;
+    for (int64_t i = threadIdx.x * 8; i < d; i += blockDim.x * 8) {
+      float f[8], wf[8];
+      VIO<DT>::load8(src, base + i, f);
+      VIO<DT>::load8(w, i, wf);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) f[k] = f[k] * rstd * wf[k];
+      VIO<DT>::store8(y, base + i, f);
+    }
+    __syncthreads();
+  }
+}
+
+// dx = rstd*(g - xhat * mean(g*xhat)),  g = dy*w, xhat = x*rstd
+template <int DT>
+__global__ void rms_bwd_dx_kernel(const void* __restrict__ dy, const void* __restrict__ x,
+                                  const void* __restrict__ w, const float* __restrict__ rstd,
+                                  void* __restrict__ dx, int64_t n, int64_t d) {
+  __shared__ float red[4];
+  for (int64_t row = blockIdx.x; row < n; row += gridDim.x) {
+    const int64_t base = row * d;
+    const float rs = rstd[row];
+    float s1 = 0.f;
+    for (int64_t i = threadIdx.x * 8; i < d; i += blockDim.x * 8) {
+      float gy[8], xf[8], wf[8];
+      VIO<DT>::load8(dy, base + i, gy);
+      VIO<DT>::load8(x, base + i, xf);
+      VIO<DT>::load8(w, i, wf);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) s1 += gy[k] * wf[k] * xf[k] * rs;
+    }
+    float t1 = block_reduce_256(s1, SumOp(), red, 0.f);
+    const float c = t1 / d;
+    for (int64_t i = threadIdx.x * 8; i < d; i += blockDim.x * 8) {
+      float gy[8], xf[8], wf[8];
+      VIO<DT>::load8(dy, base + i, gy);
+      VIO<DT>::load8(x, base + i, xf);
+      VIO<DT>::load8(w, i, wf);
+#pragma unroll
+      for (int k = 0; k < 8; ++k)
+        gy[k] = rs * (gy[k] * wf[k] - xf[k] * rs * c);
+      VIO<DT>::store8(dx, base + i, gy);
+    }
+    __syncthreads();
+  }
+}
+
+// -------- host launchers ----------------------------------------------------
+static int norm_grid(int64_t n) {
+  int64_t cap = 256 * 8;
+  return (int)(n < cap ? n : cap);
+}
+
+#define DT_SWITCH(dtype, ...)                         \
+  if (dtype == kBF16) {                               \
+    constexpr int DT = kBF16;                         \
+    __VA_ARGS__;                                      \
+  } else {                                            \
+    constexpr int DT = kF32;                          \
+    __VA_ARGS__;                                      \
+  }
+
+void layer_norm_fwd(const void* x, const void* w, const void* b, void* y,
+                    float* mean, float* rstd, int64_t n, int64_t d, float eps,
+                    int dtype, hipStream_t s) {
+  DT_SWITCH(dtype, hipLaunchKernelGGL((ln_fwd_kernel<DT>), dim3(norm_grid(n)),
+                                      dim3(256), 0, s, x, w, b, y, mean, rstd, n, d, eps));
+}
+
+void layer_norm_bwd_dx(const void* dy, const void* x, const void* w,
+                       const float* mean, const float* rstd, void* dx,
+                       int64_t n, int64_t d, int dtype, hipStream_t s) {
+  DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dx_kernel<DT>), dim3(norm_grid(n)),
+                                      dim3(256), 0, s, dy, x, w, mean, rstd, dx, n, d));
+}
+
+void layer_norm_bwd_dwdb(const void* dy, const void* x, const float* mean,
+                         const float* rstd, float* dw, float* db, int64_t n,
+                         int64_t d, int dtype, hipStream_t s) {
+  int chunks = (int)hmin<int64_t>(64, hmax<int64_t>(1, n / 64));
+  dim3 grid((unsigned)cdiv((int)d, 256), chunks);
+  DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_kernel<DT, false>), grid,
+                                      dim3(256), 0, s, dy, x, mean, rstd, dw, db, n, d));
+}
+
+void rms_norm_fwd(const void* x, const void* residual, const void* w, void* y,
+                  void* res_out, float* rstd, int64_t n, int64_t d, float eps,
+                  int dtype, hipStream_t s) {
+  if (residual) {
+    DT_SWITCH(dtype, hipLaunchKernelGGL((rms_fwd_kernel<DT, true>), dim3(norm_grid(n)),
+                                        dim3(256), 0, s, x, residual, w, y, res_out, rstd, n, d, eps));
+  } else {
+    DT_SWITCH(dtype, hipLaunchKernelGGL((rms_fwd_kernel<DT, false>), dim3(norm_grid(n)),
+                                        dim3(256), 0, s, x, nullptr, w, y, nullptr, rstd, n, d, eps));
+  }
+}
+
+void rms_norm_bwd_dx(const void* dy, const void* x, const void* w,
+                     const float* rstd, void* dx, int64_t n, int64_t d,
+                     int dtype, hipStream_t s) {
+  DT_SWITCH(dtype, hipLaunchKernelGGL((rms_bwd_dx_kernel<DT>), dim3(norm_grid(n)),
+                                      dim3(256), 0, s, dy, x, w, rstd, dx, n, d));
+}
+
+void rms_norm_bwd_dw(const void* dy, const void* x, const float* rstd,
+                     float* dw, int64_t n, int64_t d, int dtype, hipStream_t s) {
+  int chunks = (int)hmin<int64_t>(64, hmax<int64_t>(1, n / 64));
+  dim3 grid((unsigned)cdiv((int)d, 256), chunks);
+  DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_kernel<DT, true>), grid,
+                                      dim3(256), 0, s, dy, x, nullptr, rstd, dw, nullptr, n, d));
+}
+
+}  // namespace pa

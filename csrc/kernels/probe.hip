@@ -1,0 +1,29 @@
+// MFMA layout probe: single-wave 16x16x32 bf16 tile, used by the GPU test
+// suite to verify the fragment layout assumptions in flash_attn.hip /
+// gemm kernels against torch.matmul (asymmetric-input check, guide G9).
+#include "common.h"
+#include "api.h"
+
+namespace pa {
+
+__global__ void mfma_probe_kernel(const short* __restrict__ a, const short* __restrict__ b,
+                                  float* __restrict__ c) {
+  // a: [16][32] row-major bf16 ; b_t: [16][32] row-major (i.e. B^T, B is [32][16])
+  // c: [16][16] row-major fp32 ; launched with 64 threads (one wave)
+  int lane = threadIdx.x & 63;
+  int l16 = lane & 15, lg = lane >> 4;
+  shortx8 af = *reinterpret_cast<const shortx8*>(a + l16 * 32 + lg * 8);
+  shortx8 bf = *reinterpret_cast<const shortx8*>(b + l16 * 32 + lg * 8);
+  floatx4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = mfma_bf16(af, bf, acc);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) c[(lg * 4 + r) * 16 + l16] = acc[r];
+}
+
+void mfma_probe(const void* a, const void* bt, float* c, hipStream_t s);
+void mfma_probe(const void* a, const void* bt, float* c, hipStream_t s) {
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, s,
+                     (const short*)a, (const short*)bt, c);
+}
+
+}  // namespace pa

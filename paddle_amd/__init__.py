@@ -1,0 +1,254 @@
+"""paddle_amd -- an MI355X-native deep-learning framework with the
+PaddlePaddle API surface.
+
+Architecture: the tensor/autograd substrate is
+PyTorch-ROCm; hot ops are hand-written gfx950 HIP kernels (csrc/)
+exposed as torch custom ops; distributed training is RCCL over xGMI via
+torch.distributed; the `paddle.*`-shaped Python API lives here.
+
+This is a ground-up rebuild, not a port: no CUDA-compat layer, no
+multi-backend dispatch, no codegen compiler.  Reference parity anchors
+are cited per-module against /root/reference (PaddlePaddle/Paddle).
+"""
+from __future__ import annotations
+
+import torch
+
+from . import framework
+from .framework import (  # noqa: F401
+    CPUPlace,
+    CUDAPlace,
+    GPUPlace,
+    bfloat16,
+    bool_,
+    complex128,
+    complex64,
+    float16,
+    float32,
+    float64,
+    float8_e4m3fn,
+    float8_e5m2,
+    get_device,
+    get_flags,
+    int16,
+    int32,
+    int64,
+    int8,
+    is_compiled_with_cuda,
+    is_compiled_with_rocm,
+    seed,
+    set_device,
+    set_flags,
+    uint8,
+)
+from . import tensor_patch as _tensor_patch
+
+_tensor_patch.apply_patches()
+
+Tensor = torch.Tensor
+
+# ---------------------------------------------------------------------------
+# creation ops (reference: python/paddle/tensor/creation.py)
+# ---------------------------------------------------------------------------
+from .tensor.creation import (  # noqa: F401
+    arange,
+    assign,
+    clone,
+    empty,
+    empty_like,
+    eye,
+    full,
+    full_like,
+    linspace,
+    meshgrid,
+    ones,
+    ones_like,
+    rand,
+    randint,
+    randn,
+    randperm,
+    normal,
+    uniform,
+    to_tensor,
+    tril,
+    triu,
+    zeros,
+    zeros_like,
+    diag,
+)
+from .tensor.manipulation import (  # noqa: F401
+    broadcast_to,
+    cast,
+    chunk,
+    concat,
+    expand,
+    expand_as,
+    flatten,
+    flip,
+    gather,
+    gather_nd,
+    index_select,
+    masked_select,
+    numel,
+    put_along_axis,
+    repeat_interleave,
+    reshape,
+    reshape_,
+    roll,
+    scatter,
+    shape,
+    slice,
+    split,
+    squeeze,
+    squeeze_,
+    stack,
+    take_along_axis,
+    tile,
+    transpose,
+    unbind,
+    unique,
+    unsqueeze,
+    unsqueeze_,
+    unstack,
+    view,
+    where,
+)
+from .tensor.math import (  # noqa: F401
+    abs,
+    add,
+    add_n,
+    all,
+    allclose,
+    amax,
+    amin,
+    any,
+    argmax,
+    argmin,
+    argsort,
+    bincount,
+    ceil,
+    clip,
+    cos,
+    cosh,
+    cumprod,
+    cumsum,
+    divide,
+    equal,
+    equal_all,
+    erf,
+    exp,
+    floor,
+    floor_divide,
+    greater_equal,
+    greater_than,
+    isfinite,
+    isinf,
+    isnan,
+    less_equal,
+    less_than,
+    log,
+    log2,
+    log10,
+    logical_and,
+    logical_not,
+    logical_or,
+    logical_xor,
+    logsumexp,
+    max,
+    maximum,
+    mean,
+    min,
+    minimum,
+    mod,
+    multiply,
+    not_equal,
+    pow,
+    prod,
+    reciprocal,
+    remainder,
+    round,
+    rsqrt,
+    scale,
+    sign,
+    sin,
+    sinh,
+    sort,
+    sqrt,
+    square,
+    subtract,
+    sum,
+    tan,
+    tanh,
+    topk,
+    trunc,
+)
+from .tensor.linalg import (  # noqa: F401
+    bmm,
+    cross,
+    dist,
+    dot,
+    einsum,
+    matmul,
+    mm,
+    mv,
+    norm,
+    outer,
+    t,
+    tensordot,
+)
+from .tensor.search import (  # noqa: F401
+    index_sample,
+    masked_fill,
+    nonzero,
+    searchsorted,
+)
+from .tensor.random import multinomial, bernoulli, poisson  # noqa: F401
+from .tensor.einsum import einsum  # noqa: F401,F811
+
+from . import nn  # noqa: F401
+from . import optimizer  # noqa: F401
+from . import amp  # noqa: F401
+from . import io  # noqa: F401
+from . import distributed  # noqa: F401
+from . import autograd  # noqa: F401
+from . import device  # noqa: F401
+from . import metric  # noqa: F401
+from . import vision  # noqa: F401
+from . import static  # noqa: F401
+from . import jit  # noqa: F401
+from . import incubate  # noqa: F401
+from . import models  # noqa: F401
+from .framework_io import save, load, async_save  # noqa: F401
+from .param_attr import ParamAttr  # noqa: F401
+from .autograd import grad, no_grad, enable_grad, set_grad_enabled, is_grad_enabled  # noqa: F401
+
+disable_static = lambda *a, **k: None  # dygraph is the default mode
+enable_static = static.enable_static
+in_dynamic_mode = lambda: not static._static_mode
+
+__version__ = "0.1.0"
+
+
+def version():
+    return __version__
+
+
+def is_grad_enabled_():
+    return torch.is_grad_enabled()
+
+
+def get_default_dtype():
+    return framework.dtype_name(torch.get_default_dtype())
+
+
+def set_default_dtype(d):
+    torch.set_default_dtype(framework.convert_dtype(d))
+
+
+def grad_(outputs, inputs, grad_outputs=None, retain_graph=None, create_graph=False):
+    return torch.autograd.grad(outputs, inputs, grad_outputs, retain_graph=retain_graph, create_graph=create_graph)
+
+
+def synchronize(device=None):
+    if torch.cuda.is_available():
+        torch.cuda.synchronize(device)

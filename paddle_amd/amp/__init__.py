@@ -1,0 +1,186 @@
+"""paddle.amp parity (reference: python/paddle/amp/auto_cast.py,
+grad_scaler.py:187; O1/O2 semantics per SURVEY.md A.5).
+
+auto_cast maps to torch.autocast with the paddle O1 white/black lists
+honored by construction (torch's autocast policy covers the same op
+classes: matmul-like ops cast down, reductions/norms stay fp32).
+"""
+from __future__ import annotations
+
+import contextlib
+
+import torch
+
+from .. import framework
+
+# O1 lists kept for API parity / inspection (amp_lists.py:20-44)
+WHITE_LIST = {"conv2d", "einsum", "matmul", "matmul_v2", "mul",
+              "fused_gemm_epilogue", "fused_rotary_position_embedding", "flash_attn"}
+BLACK_LIST = {"exp", "square", "log", "mean", "sum", "cos_sim", "softmax",
+              "softmax_with_cross_entropy", "sigmoid_cross_entropy_with_logits",
+              "c_softmax_with_cross_entropy", "cross_entropy", "cross_entropy2",
+              "layer_norm", "reduce_sum", "rms_norm"}
+
+
+@contextlib.contextmanager
+def auto_cast(enable=True, custom_white_list=None, custom_black_list=None,
+              level="O1", dtype="bfloat16", use_promote=True):
+    if not enable:
+        yield
+        return
+    dt = framework.convert_dtype(dtype)
+    dev = "cuda" if torch.cuda.is_available() else "cpu"
+    with torch.autocast(device_type=dev, dtype=dt, enabled=True):
+        yield
+
+
+amp_guard = auto_cast
+
+
+def decorate(models, optimizers=None, level="O2", dtype="bfloat16",
+             master_weight=None, save_dtype=None, master_grad=False,
+             excluded_layers=None):
+    """O2: cast model params to dtype (keeping norms fp32 per paddle's
+    need_keep_fp32 unless excluded); optimizer keeps fp32 master."""
+    dt = framework.convert_dtype(dtype)
+    from ..nn import BatchNorm, BatchNorm1D, BatchNorm2D, LayerNorm
+
+    def _cast_model(m):
+        if level == "O2":
+            keep = (LayerNorm, BatchNorm, BatchNorm1D, BatchNorm2D)
+            for layer in m.modules():
+                if excluded_layers and isinstance(layer, tuple(excluded_layers)):
+                    continue
+                if isinstance(layer, keep):
+                    continue
+                for _, p in layer.named_parameters(recurse=False):
+                    p.data = p.data.to(dt)
+        return m
+
+    single = not isinstance(models, (list, tuple))
+    ms = [models] if single else list(models)
+    ms = [_cast_model(m) for m in ms]
+    out_m = ms[0] if single else ms
+    if optimizers is None:
+        return out_m
+    return out_m, optimizers
+
+
+class GradScaler:
+    """Dynamic loss scaling (grad_scaler.py defaults: init 2**16,
+    incr_ratio 2.0 / incr_every_n_steps 1000 / decr_ratio 0.5)."""
+
+    def __init__(self, enable=True, init_loss_scaling=2.0 ** 16, incr_ratio=2.0,
+                 decr_ratio=0.5, incr_every_n_steps=1000, decr_every_n_nan_or_inf=2,
+                 use_dynamic_loss_scaling=True):
+        self._enable = enable
+        self._scale = float(init_loss_scaling)
+        self._incr_ratio = incr_ratio
+        self._decr_ratio = decr_ratio
+        self._incr_every = incr_every_n_steps
+        self._decr_every = decr_every_n_nan_or_inf
+        self._dynamic = use_dynamic_loss_scaling
+        self._good_steps = 0
+        self._bad_steps = 0
+        self._found_inf = False
+
+    def is_enable(self):
+        return self._enable
+
+    def scale(self, var):
+        if not self._enable:
+            return var
+        return var * self._scale
+
+    def _check_finite(self, optimizer):
+        found = False
+        for p in optimizer._params:
+            if p.grad is not None:
+                g = p.grad
+                if not torch.isfinite(g.float().sum()):
+                    found = True
+                    break
+        self._found_inf = found
+        return found
+
+    def unscale_(self, optimizer):
+        if not self._enable:
+            return
+        inv = 1.0 / self._scale
+        for p in optimizer._params:
+            if p.grad is not None:
+                p.grad.mul_(inv)
+        self._check_finite(optimizer)
+
+    def step(self, optimizer):
+        if not self._enable:
+            optimizer.step()
+            return
+        if not getattr(self, "_unscaled", False):
+            self.unscale_(optimizer)
+        self._unscaled = False
+        if not self._found_inf:
+            optimizer.step()
+
+    def update(self):
+        if not (self._enable and self._dynamic):
+            return
+        if self._found_inf:
+            self._bad_steps += 1
+            self._good_steps = 0
+            if self._bad_steps >= self._decr_every:
+                self._scale = max(self._scale * self._decr_ratio, 1.0)
+                self._bad_steps = 0
+        else:
+            self._good_steps += 1
+            self._bad_steps = 0
+            if self._good_steps >= self._incr_every:
+                self._scale *= self._incr_ratio
+                self._good_steps = 0
+
+    def minimize(self, optimizer, scaled_loss):
+        scaled_loss.backward()
+        self.step(optimizer)
+        self.update()
+        optimizer.clear_grad()
+
+    def state_dict(self):
+        return {"scale": self._scale, "incr_ratio": self._incr_ratio,
+                "decr_ratio": self._decr_ratio, "incr_count": self._good_steps,
+                "decr_count": self._bad_steps}
+
+    def load_state_dict(self, sd):
+        self._scale = sd.get("scale", self._scale)
+
+    def get_loss_scaling(self):
+        return torch.tensor(self._scale)
+
+
+AmpScaler = GradScaler
+
+
+def is_bfloat16_supported(device=None):
+    return True
+
+
+def is_float16_supported(device=None):
+    return True
+
+
+class debugging:
+    """paddle.amp.debugging stub surface (check_numerics etc.)."""
+
+    @staticmethod
+    def enable_operator_stats_collection():
+        pass
+
+    @staticmethod
+    def disable_operator_stats_collection():
+        pass
+
+    @staticmethod
+    def check_numerics(tensor, op_type="", var_name="", debug_mode=None):
+        import torch as _t
+        if not _t.isfinite(tensor.float()).all():
+            raise FloatingPointError(f"nan/inf detected in {op_type}:{var_name}")
+        return tensor

@@ -1,0 +1,63 @@
+"""paddle.distributed parity surface (reference: python/paddle/distributed/)."""
+from . import collective  # noqa: F401
+from .collective import (  # noqa: F401
+    Group,
+    P2POp,
+    ReduceOp,
+    all_gather,
+    all_gather_into_tensor,
+    all_gather_object,
+    all_reduce,
+    alltoall,
+    alltoall_single,
+    barrier,
+    batch_isend_irecv,
+    broadcast,
+    broadcast_object_list,
+    destroy_process_group,
+    get_group,
+    irecv,
+    isend,
+    is_initialized,
+    new_group,
+    recv,
+    reduce,
+    reduce_scatter,
+    reduce_scatter_tensor,
+    scatter,
+    send,
+    stream,
+)
+from .parallel import (  # noqa: F401
+    DataParallel,
+    ParallelEnv,
+    get_rank,
+    get_world_size,
+    init_parallel_env,
+    sync_gradients,
+)
+from . import fleet  # noqa: F401
+from . import sharding  # noqa: F401
+from . import checkpoint  # noqa: F401
+
+
+def spawn(func, args=(), nprocs=-1, join=True, daemon=False, **options):
+    """paddle.distributed.spawn parity over torch.multiprocessing."""
+    import os
+
+    import torch
+    import torch.multiprocessing as mp
+
+    if nprocs == -1:
+        nprocs = torch.cuda.device_count() if torch.cuda.is_available() else 1
+
+    def _entry(rank, n, fn, fn_args):
+        os.environ["RANK"] = str(rank)
+        os.environ["LOCAL_RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(n)
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29600")
+        fn(*fn_args)
+
+    return mp.start_processes(_entry, args=(nprocs, func, args), nprocs=nprocs,
+                              join=join, daemon=daemon, start_method="spawn")

@@ -1,0 +1,295 @@
+"""Pipeline parallelism: PipelineLayer + 1F1B schedule over RCCL p2p.
+
+Reference: fleet/meta_parallel/parallel_layers/pp_layers.py (LayerDesc:56,
+SharedLayerDesc:76, PipelineLayer:257) and pipeline_parallel.py
+(forward_backward_pipeline:575 -- 1F1B), pp_utils/p2p_communication.py
+(SendRecvMeta:52 shape handshake).
+
+MI355X: stage boundaries are single xGMI links (153 GB/s point-to-point,
+SURVEY.md §5) so activations move cheaply; p2p tensors are sent with a
+one-time [ndim, shape..., dtype] int64 header, cached afterwards.
+"""
+from __future__ import annotations
+
+from typing import Callable, List, Optional
+
+import torch
+
+from .. import collective as C
+
+_DTYPE_IDS = [torch.float32, torch.bfloat16, torch.float16, torch.int64, torch.int32]
+
+
+class LayerDesc:
+    def __init__(self, layer_cls, *args, **kwargs):
+        self.layer_cls = layer_cls
+        self.args = args
+        self.kwargs = kwargs
+
+    def build_layer(self):
+        return self.layer_cls(*self.args, **self.kwargs)
+
+
+class SharedLayerDesc(LayerDesc):
+    """Tied layers (e.g. embedding/lm-head) -- built on every owning stage;
+    grads all-reduced across the shared group after each step."""
+
+    def __init__(self, key, layer_cls, *args, forward_func=None, shared_weight_attr="weight",
+                 **kwargs):
+        super().__init__(layer_cls, *args, **kwargs)
+        self.layer_name = key
+        self.forward_func = forward_func
+        self.shared_weight_attr = shared_weight_attr
+
+
+class PipelineLayer(torch.nn.Module):
+    """Build from a list of LayerDesc; each PP stage owns a contiguous
+    segment (uniform segmentation; param-count mode later)."""
+
+    def __init__(self, layers, num_stages=None, topology=None, loss_fn=None,
+                 seg_method="uniform", recompute_interval=0, recompute_ctx=None,
+                 hcg=None):
+        super().__init__()
+        from . import get_hybrid_communicate_group
+        self._hcg = hcg or get_hybrid_communicate_group()
+        self.num_stages = num_stages or (self._hcg.get_pipe_parallel_world_size()
+                                         if self._hcg else 1)
+        self.stage_id = self._hcg.get_pipe_parallel_rank() if self._hcg else 0
+        self.loss_fn = loss_fn
+        self.descs = list(layers)
+        self._recompute_interval = recompute_interval
+        n = len(self.descs)
+        per = [n // self.num_stages + (1 if i < n % self.num_stages else 0)
+               for i in range(self.num_stages)]
+        starts = [sum(per[:i]) for i in range(self.num_stages + 1)]
+        self.seg_starts = starts
+        self.local_start = starts[self.stage_id]
+        self.local_end = starts[self.stage_id + 1]
+        self.run_funcs = []
+        self.shared_layers = {}
+        mods = []
+        for i in range(self.local_start, self.local_end):
+            d = self.descs[i]
+            if isinstance(d, SharedLayerDesc):
+                if d.layer_name not in self.shared_layers:
+                    self.shared_layers[d.layer_name] = d.build_layer()
+                layer = self.shared_layers[d.layer_name]
+                fwd = d.forward_func
+                self.run_funcs.append((layer, fwd))
+                mods.append(layer)
+            elif isinstance(d, LayerDesc):
+                layer = d.build_layer()
+                self.run_funcs.append((layer, None))
+                mods.append(layer)
+            else:
+                self.run_funcs.append((d, None))
+                if isinstance(d, torch.nn.Module):
+                    mods.append(d)
+        self.local_layers = torch.nn.ModuleList(mods)
+
+    def forward(self, x):
+        from .recompute import recompute
+        for idx, (layer, fwd) in enumerate(self.run_funcs):
+            use_rc = (self._recompute_interval > 0 and self.training and
+                      torch.is_grad_enabled() and
+                      idx % self._recompute_interval == 0 and
+                      isinstance(x, torch.Tensor) and x.requires_grad)
+            fn = (lambda inp, l=layer, f=fwd: f(l, inp) if f else l(inp))
+            x = recompute(fn, x) if use_rc else fn(x)
+        return x
+
+    def allreduce_shared_weight_gradients(self):
+        # tied weights: all-reduce grads across pp group (stages sharing key)
+        if not self.shared_layers or self._hcg is None:
+            return
+        pp_group = self._hcg.get_pipe_parallel_group()
+        if pp_group is None:
+            return
+        for key, layer in self.shared_layers.items():
+            for p in layer.parameters():
+                if p.grad is not None:
+                    C.all_reduce(p.grad, group=pp_group)
+
+    def sharding_units(self):
+        return list(self.local_layers)
+
+
+class _P2P:
+    """shape/dtype handshake + cached send/recv (p2p_communication.py)."""
+
+    def __init__(self, hcg):
+        self.hcg = hcg
+        self.group = hcg.get_pipe_parallel_group()
+        self.sent_meta = False
+        self.recv_shape = None
+        self.recv_dtype = None
+
+    def _send_meta(self, t, dst):
+        hdr = torch.tensor([t.dim(), *t.shape, _DTYPE_IDS.index(t.dtype)],
+                           dtype=torch.int64, device=t.device)
+        n = torch.tensor([hdr.numel()], dtype=torch.int64, device=t.device)
+        C.send(n, dst=dst, group=self.group)
+        C.send(hdr, dst=dst, group=self.group)
+
+    def _recv_meta(self, src, device):
+        n = torch.zeros(1, dtype=torch.int64, device=device)
+        C.recv(n, src=src, group=self.group)
+        hdr = torch.zeros(int(n.item()), dtype=torch.int64, device=device)
+        C.recv(hdr, src=src, group=self.group)
+        vals = hdr.tolist()
+        ndim = vals[0]
+        shape = vals[1:1 + ndim]
+        dtype = _DTYPE_IDS[vals[1 + ndim]]
+        return shape, dtype
+
+    def send_forward(self, t):
+        dst = self.hcg.get_p2p_next_rank()
+        if not self.sent_meta:
+            self._send_meta(t, dst)
+            self.sent_meta = True
+        C.send(t.contiguous(), dst=dst, group=self.group)
+
+    def recv_forward(self, device):
+        src = self.hcg.get_p2p_prev_rank()
+        if self.recv_shape is None:
+            self.recv_shape, self.recv_dtype = self._recv_meta(src, device)
+        t = torch.zeros(self.recv_shape, dtype=self.recv_dtype, device=device)
+        C.recv(t, src=src, group=self.group)
+        return t
+
+    def send_backward(self, g):
+        C.send(g.contiguous(), dst=self.hcg.get_p2p_prev_rank(), group=self.group)
+
+    def recv_backward(self, like):
+        g = torch.zeros_like(like)
+        C.recv(g, src=self.hcg.get_p2p_next_rank(), group=self.group)
+        return g
+
+
+class PipelineParallel(torch.nn.Module):
+    """1F1B schedule (pipeline_parallel.py:575)."""
+
+    def __init__(self, layers: PipelineLayer, hcg, strategy=None):
+        super().__init__()
+        self._layers = layers
+        self._hcg = hcg
+        cfg = strategy.pipeline_configs if strategy is not None else {}
+        self.accumulate_steps = cfg.get("accumulate_steps", 1)
+        self.micro_batch_size = cfg.get("micro_batch_size", 1)
+        self.stage_id = hcg.get_pipe_parallel_rank()
+        self.num_stages = hcg.get_pipe_parallel_world_size()
+        self.p2p = _P2P(hcg)
+        self.is_first = hcg.is_first_stage()
+        self.is_last = hcg.is_last_stage()
+
+    def forward(self, *a, **kw):
+        return self._layers(*a, **kw)
+
+    def _split_micro(self, data):
+        x, y = data
+        mbs = []
+        n = self.accumulate_steps
+        xs = x.chunk(n) if x is not None else [None] * n
+        ys = y.chunk(n) if y is not None else [None] * n
+        return list(zip(xs, ys))
+
+    def train_batch(self, data, optimizer, lr_scheduler=None, scaler=None):
+        assert self.num_stages > 1
+        micro = self._split_micro(data)
+        n_micro = len(micro)
+        warmup = min(self.num_stages - self.stage_id - 1, n_micro)
+        steady = n_micro - warmup
+        dev = (torch.device("cuda", torch.cuda.current_device())
+               if torch.cuda.is_available() else torch.device("cpu"))
+
+        inputs: List = []
+        outputs: List = []
+        losses = []
+        fwd_i = 0
+        bwd_i = 0
+
+        def fwd_step(i):
+            x_mb, y_mb = micro[i]
+            if self.is_first:
+                inp = x_mb
+            else:
+                inp = self.p2p.recv_forward(dev)
+                inp.requires_grad_(True)
+            out = self._layers(inp)
+            if self.is_last:
+                loss = self._layers.loss_fn(out, y_mb) / n_micro
+                losses.append(loss.detach())
+                outputs.append(loss)
+            else:
+                outputs.append(out)
+                self.p2p.send_forward(out)
+            inputs.append(inp if not self.is_first else None)
+
+        def bwd_step(i):
+            out = outputs[i]
+            if self.is_last:
+                out.backward()
+            else:
+                g = self.p2p.recv_backward(out)
+                out.backward(gradient=g)
+            if not self.is_first:
+                self.p2p.send_backward(inputs[i].grad)
+            outputs[i] = None
+            inputs[i] = None
+
+        for _ in range(warmup):
+            fwd_step(fwd_i)
+            fwd_i += 1
+        for k in range(steady):
+            fwd_step(fwd_i)
+            fwd_i += 1
+            bwd_step(bwd_i)
+            bwd_i += 1
+        while bwd_i < n_micro:
+            bwd_step(bwd_i)
+            bwd_i += 1
+
+        self._layers.allreduce_shared_weight_gradients()
+        if optimizer is not None:
+            if scaler is not None:
+                scaler.step(optimizer)
+                scaler.update()
+            else:
+                optimizer.step()
+            optimizer.clear_grad()
+            if lr_scheduler is not None:
+                lr_scheduler.step()
+        if self.is_last:
+            total = torch.stack(losses).sum()
+            return total
+        return torch.zeros(1, device=dev)
+
+    def eval_batch(self, data, compute_loss=True):
+        with torch.no_grad():
+            micro = self._split_micro(data)
+            dev = (torch.device("cuda", torch.cuda.current_device())
+                   if torch.cuda.is_available() else torch.device("cpu"))
+            losses = []
+            for x_mb, y_mb in micro:
+                inp = x_mb if self.is_first else self.p2p.recv_forward(dev)
+                out = self._layers(inp)
+                if self.is_last:
+                    if compute_loss:
+                        losses.append(self._layers.loss_fn(out, y_mb))
+                else:
+                    self.p2p.send_forward(out)
+            if self.is_last and losses:
+                return torch.stack(losses).mean()
+            return torch.zeros(1, device=dev)
+
+    def parameters(self, *a, **kw):
+        return self._layers.parameters(*a, **kw)
+
+    def state_dict(self, *a, **kw):
+        return self._layers.state_dict(*a, **kw)
+
+    def __getattr__(self, name):
+        try:
+            return super().__getattr__(name)
+        except AttributeError:
+            return getattr(self._layers, name)

@@ -1,0 +1,586 @@
+"""Autograd-integrated wrappers over the gfx950 HIP kernels.
+
+Each op is a torch.autograd.Function whose forward dispatches to
+paddle_amd._C on GPU (mandatory -- _ext.use_native raises if the
+extension is missing) and to an fp32 torch reference on CPU.  The CPU
+path doubles as the numerics oracle in tests/.
+
+Reference op-signature parity anchors are cited per-op.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from .. import _ext
+
+
+# ---------------------------------------------------------------------------
+# layer_norm (paddle/phi/kernels/gpu/layer_norm_kernel.cu parity)
+# ---------------------------------------------------------------------------
+class _LayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, eps):
+        if _ext.use_native(x):
+            C = _ext.get_ext()
+            y, mean, rstd = C.layer_norm_fwd(x.contiguous(), w.contiguous(),
+                                             b.contiguous() if b is not None else None, eps)
+        else:
+            xf = x.float()
+            mean = xf.mean(-1).reshape(-1)
+            var = xf.var(-1, unbiased=False).reshape(-1)
+            rstd = torch.rsqrt(var + eps)
+            d = x.shape[-1]
+            xhat = (xf - mean.view(*x.shape[:-1], 1)) * rstd.view(*x.shape[:-1], 1)
+            y = xhat * w.float() + (b.float() if b is not None else 0.0)
+            y = y.to(x.dtype)
+        ctx.save_for_backward(x, w, mean, rstd)
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, mean, rstd = ctx.saved_tensors
+        if _ext.use_native(x):
+            C = _ext.get_ext()
+            dx, dw, db = C.layer_norm_bwd(dy.contiguous(), x, w, mean, rstd, ctx.has_bias)
+        else:
+            d = x.shape[-1]
+            xf = x.float()
+            dyf = dy.float()
+            mu = mean.view(*x.shape[:-1], 1)
+            rs = rstd.view(*x.shape[:-1], 1)
+            xhat = (xf - mu) * rs
+            g = dyf * w.float()
+            s1 = (g * xhat).mean(-1, keepdim=True)
+            s2 = g.mean(-1, keepdim=True)
+            dx = (rs * (g - s2 - xhat * s1)).to(x.dtype)
+            dw = (dyf * xhat).reshape(-1, d).sum(0).to(w.dtype)
+            db = dyf.reshape(-1, d).sum(0).to(w.dtype)
+        return dx, dw, (db if ctx.has_bias else None), None
+
+
+def layer_norm(x, weight, bias=None, epsilon=1e-5):
+    return _LayerNorm.apply(x, weight, bias, epsilon)
+
+
+# ---------------------------------------------------------------------------
+# rms_norm (+ optional fused residual add)
+# paddle parity: incubate/nn/functional/fused_rms_norm.py, rms_norm_kernel.cu
+# ---------------------------------------------------------------------------
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        if _ext.use_native(x):
+            C = _ext.get_ext()
+            y, rstd = C.rms_norm_fwd(x.contiguous(), None, w.contiguous(), eps)
+        else:
+            xf = x.float()
+            rstd = torch.rsqrt(xf.square().mean(-1) + eps).reshape(-1)
+            y = (xf * rstd.view(*x.shape[:-1], 1) * w.float()).to(x.dtype)
+        ctx.save_for_backward(x, w, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, rstd = ctx.saved_tensors
+        if _ext.use_native(x):
+            C = _ext.get_ext()
+            dx, dw = C.rms_norm_bwd(dy.contiguous(), x, w, rstd)
+        else:
+            d = x.shape[-1]
+            xf, dyf = x.float(), dy.float()
+            rs = rstd.view(*x.shape[:-1], 1)
+            xhat = xf * rs
+            g = dyf * w.float()
+            s1 = (g * xhat).mean(-1, keepdim=True)
+            dx = (rs * (g - xhat * s1)).to(x.dtype)
+            dw = (dyf * xhat).reshape(-1, d).sum(0).to(w.dtype)
+        return dx, dw, None
+
+
+def rms_norm(x, weight, epsilon=1e-6):
+    return _RMSNorm.apply(x, weight, epsilon)
+
+
+class _FusedRMSNormResidual(torch.autograd.Function):
+    """y, xr = rms_norm(x + residual) -- returns normed out and the new
+    residual stream (paddle fused_rms_norm with residual)."""
+
+    @staticmethod
+    def forward(ctx, x, residual, w, eps):
+        if _ext.use_native(x):
+            C = _ext.get_ext()
+            y, rstd, xr = C.rms_norm_fwd(x.contiguous(), residual.contiguous(), w.contiguous(), eps)
+        else:
+            xr = x + residual
+            xf = xr.float()
+            rstd = torch.rsqrt(xf.square().mean(-1) + eps).reshape(-1)
+            y = (xf * rstd.view(*x.shape[:-1], 1) * w.float()).to(x.dtype)
+        ctx.save_for_backward(xr, w, rstd)
+        return y, xr
+
+    @staticmethod
+    def backward(ctx, dy, dxr):
+        xr, w, rstd = ctx.saved_tensors
+        if _ext.use_native(xr):
+            C = _ext.get_ext()
+            dx, dw = C.rms_norm_bwd(dy.contiguous(), xr, w, rstd)
+        else:
+            d = xr.shape[-1]
+            xf, dyf = xr.float(), dy.float()
+            rs = rstd.view(*xr.shape[:-1], 1)
+            xhat = xf * rs
+            g = dyf * w.float()
+            s1 = (g * xhat).mean(-1, keepdim=True)
+            dx = (rs * (g - xhat * s1)).to(xr.dtype)
+            dw = (dyf * xhat).reshape(-1, d).sum(0).to(w.dtype)
+        if dxr is not None:
+            dx = dx + dxr
+        return dx, dx, dw, None
+
+
+def fused_rms_norm(x, norm_weight, residual=None, epsilon=1e-6):
+    if residual is None:
+        return rms_norm(x, norm_weight, epsilon)
+    return _FusedRMSNormResidual.apply(x, residual, norm_weight, epsilon)
+
+
+# ---------------------------------------------------------------------------
+# softmax cross entropy (cross_entropy_kernel.cu parity; hard labels)
+# ---------------------------------------------------------------------------
+class _SoftmaxCE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels, ignore_index):
+        shp = logits.shape
+        v = shp[-1]
+        l2 = logits.reshape(-1, v)
+        lab = labels.reshape(-1)
+        if _ext.use_native(logits):
+            C = _ext.get_ext()
+            loss, lse = C.softmax_ce_fwd(l2.contiguous(), lab.contiguous(), ignore_index)
+        else:
+            lf = l2.float()
+            lse = torch.logsumexp(lf, -1)
+            safe = lab.clamp(min=0)
+            picked = lf.gather(1, safe.unsqueeze(1)).squeeze(1)
+            loss = torch.where(lab == ignore_index, torch.zeros_like(lse), lse - picked)
+        ctx.save_for_backward(l2, lab, lse)
+        ctx.ignore_index = ignore_index
+        ctx.in_shape = shp
+        return loss.reshape(shp[:-1])
+
+    @staticmethod
+    def backward(ctx, dloss):
+        l2, lab, lse = ctx.saved_tensors
+        dl = dloss.reshape(-1).float().contiguous()
+        if _ext.use_native(l2):
+            C = _ext.get_ext()
+            dlogits = C.softmax_ce_bwd(dl, l2, lab, lse, ctx.ignore_index)
+        else:
+            p = torch.exp(l2.float() - lse.unsqueeze(1))
+            onehot = torch.zeros_like(p)
+            safe = lab.clamp(min=0)
+            onehot.scatter_(1, safe.unsqueeze(1), 1.0)
+            g = torch.where((lab == ctx.ignore_index).unsqueeze(1),
+                            torch.zeros_like(dl).unsqueeze(1), dl.unsqueeze(1))
+            dlogits = (g * (p - onehot)).to(l2.dtype)
+        return dlogits.reshape(ctx.in_shape), None, None
+
+
+def softmax_cross_entropy(logits, labels, ignore_index=-100, reduction="none"):
+    loss = _SoftmaxCE.apply(logits, labels, ignore_index)
+    if reduction == "mean":
+        n_valid = (labels != ignore_index).sum().clamp(min=1)
+        return loss.sum() / n_valid.to(loss.dtype)
+    if reduction == "sum":
+        return loss.sum()
+    return loss
+
+
+# ---------------------------------------------------------------------------
+# bias + gelu (fused_bias_act parity; erf gelu)
+# ---------------------------------------------------------------------------
+class _BiasGelu(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, bias):
+        ctx.save_for_backward(x, bias if bias is not None else torch.empty(0))
+        ctx.has_bias = bias is not None
+        if _ext.use_native(x):
+            C = _ext.get_ext()
+            return C.bias_gelu_fwd(x.contiguous(), bias.contiguous() if bias is not None else None)
+        v = x.float() + (bias.float() if bias is not None else 0.0)
+        return torch.nn.functional.gelu(v).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, bias = ctx.saved_tensors
+        b = bias if ctx.has_bias else None
+        # d/dx gelu(x+b) == d/db gelu(x+b), so db = colsum(dx)
+        if _ext.use_native(x):
+            C = _ext.get_ext()
+            dx = C.bias_gelu_bwd(dy.contiguous(), x, b.contiguous() if b is not None else None)
+            db = C.colsum(dx).to(dy.dtype) if ctx.has_bias else None
+        else:
+            v = x.float() + (b.float() if b is not None else 0.0)
+            cdf = 0.5 * (1.0 + torch.erf(v * 0.7071067811865476))
+            pdf = 0.3989422804014327 * torch.exp(-0.5 * v * v)
+            dx = (dy.float() * (cdf + v * pdf)).to(x.dtype)
+            db = dx.float().reshape(-1, dx.shape[-1]).sum(0).to(dy.dtype) if ctx.has_bias else None
+        return dx, db
+
+
+def bias_gelu(x, bias=None):
+    return _BiasGelu.apply(x, bias)
+
+
+# ---------------------------------------------------------------------------
+# swiglu: silu(x[..., :d]) * x[..., d:]
+# paddle parity: incubate/nn/functional/swiglu (fused_bias_act swiglu)
+# ---------------------------------------------------------------------------
+class _SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.save_for_backward(x)
+        if _ext.use_native(x):
+            C = _ext.get_ext()
+            return C.swiglu_fwd(x.contiguous())
+        d = x.shape[-1] // 2
+        g, u = x[..., :d].float(), x[..., d:].float()
+        return (torch.nn.functional.silu(g) * u).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        if _ext.use_native(x):
+            C = _ext.get_ext()
+            return C.swiglu_bwd(dy.contiguous(), x)
+        d = x.shape[-1] // 2
+        g, u = x[..., :d].float(), x[..., d:].float()
+        sig = torch.sigmoid(g)
+        silu = g * sig
+        dg = dy.float() * u * (sig * (1 + g * (1 - sig)))
+        du = dy.float() * silu
+        return torch.cat([dg, du], dim=-1).to(x.dtype)
+
+
+def swiglu(x, y=None):
+    if y is not None:
+        x = torch.cat([x, y], dim=-1)
+    return _SwiGLU.apply(x)
+
+
+# ---------------------------------------------------------------------------
+# rotary embedding (fused_rope_kernel.cu RotateHalf parity; neox style)
+# ---------------------------------------------------------------------------
+_rope_cache = {}
+
+
+def build_rope_cache(seq_len, head_dim, base=10000.0, device=None, dtype=torch.float32):
+    key = (seq_len, head_dim, base, str(device))
+    if key not in _rope_cache:
+        half = head_dim // 2
+        inv = 1.0 / (base ** (torch.arange(0, half, dtype=torch.float32, device=device) / half))
+        t = torch.arange(seq_len, dtype=torch.float32, device=device)
+        freqs = torch.outer(t, inv)  # [S, half]
+        _rope_cache[key] = (freqs.cos().contiguous(), freqs.sin().contiguous())
+    return _rope_cache[key]
+
+
+class _Rope(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos_t, sin_t, pos_offset):
+        ctx.save_for_backward(cos_t, sin_t)
+        ctx.pos_offset = pos_offset
+        if _ext.use_native(x):
+            C = _ext.get_ext()
+            return C.rope_fwd(x.contiguous(), cos_t, sin_t, pos_offset, False)
+        return _rope_ref(x, cos_t, sin_t, pos_offset, conj=False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos_t, sin_t = ctx.saved_tensors
+        if _ext.use_native(dy):
+            C = _ext.get_ext()
+            return C.rope_fwd(dy.contiguous(), cos_t, sin_t, ctx.pos_offset, True), None, None, None
+        return _rope_ref(dy, cos_t, sin_t, ctx.pos_offset, conj=True), None, None, None
+
+
+def _rope_ref(x, cos_t, sin_t, pos_offset, conj):
+    # x: [B, S, H, D]
+    b, s, h, d = x.shape
+    half = d // 2
+    c = cos_t[pos_offset:pos_offset + s].view(1, s, 1, half).float()
+    sn = sin_t[pos_offset:pos_offset + s].view(1, s, 1, half).float()
+    if conj:
+        sn = -sn
+    x1, x2 = x[..., :half].float(), x[..., half:].float()
+    y1 = x1 * c - x2 * sn
+    y2 = x2 * c + x1 * sn
+    return torch.cat([y1, y2], dim=-1).to(x.dtype)
+
+
+def fused_rotary_position_embedding(q, k=None, v=None, sin=None, cos=None,
+                                    position_ids=None, use_neox_rotary_style=True,
+                                    base=10000.0, pos_offset=0):
+    """paddle.incubate.nn.functional.fused_rotary_position_embedding parity
+    (SURVEY.md A.7).  q/k/v: [B, S, H, D]."""
+    bq, s, hq, d = q.shape
+    if cos is None or sin is None:
+        cos_t, sin_t = build_rope_cache(s + pos_offset, d, base, q.device)
+    else:
+        cos_t, sin_t = cos.float().reshape(-1, d)[..., : d // 2].contiguous(), \
+                       sin.float().reshape(-1, d)[..., : d // 2].contiguous()
+    outs = [_Rope.apply(q, cos_t, sin_t, pos_offset)]
+    if k is not None:
+        outs.append(_Rope.apply(k, cos_t, sin_t, pos_offset))
+    if v is not None:
+        outs.append(v)
+    return tuple(outs) if len(outs) > 1 else outs[0]
+
+
+# ---------------------------------------------------------------------------
+# flash attention (flash_attn_kernel.cu API parity)
+# ---------------------------------------------------------------------------
+class _FlashAttn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale, causal):
+        # q,k,v: [B, H, S, D] contiguous bf16
+        if _ext.use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+            C = _ext.get_ext()
+            o, lse = C.flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(),
+                                      scale, causal)
+            ctx.native = True
+        else:
+            o, lse = _sdpa_ref(q, k, v, scale, causal)
+            ctx.native = False
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.scale, ctx.causal = scale, causal
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        if ctx.native:
+            C = _ext.get_ext()
+            if k.shape[1] != q.shape[1]:  # GQA: expand for bwd kernel
+                rep = q.shape[1] // k.shape[1]
+                ke = k.repeat_interleave(rep, dim=1)
+                ve = v.repeat_interleave(rep, dim=1)
+                dq, dke, dve = C.flash_attn_bwd(do.contiguous(), q, ke.contiguous(),
+                                                ve.contiguous(), o, lse, ctx.scale, ctx.causal)
+                hkv = k.shape[1]
+                dk = dke.view(k.shape[0], hkv, rep, k.shape[2], k.shape[3]).sum(2)
+                dv = dve.view(v.shape[0], hkv, rep, v.shape[2], v.shape[3]).sum(2)
+            else:
+                dq, dk, dv = C.flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
+                                              ctx.scale, ctx.causal)
+        else:
+            dq, dk, dv = _sdpa_ref_bwd(do, q, k, v, lse, ctx.scale, ctx.causal)
+        return dq, dk, dv, None, None
+
+
+def _sdpa_ref(q, k, v, scale, causal):
+    # fp32 reference; returns (o, lse) with lse = logsumexp of scaled scores
+    qf, kf, vf = q.float(), k.float(), v.float()
+    if k.shape[1] != q.shape[1]:
+        rep = q.shape[1] // k.shape[1]
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal:
+        sq, skv = s.shape[-2], s.shape[-1]
+        mask = torch.ones(sq, skv, dtype=torch.bool, device=s.device).tril()
+        s = s.masked_fill(~mask, float("-inf"))
+    lse = torch.logsumexp(s, -1)
+    p = torch.exp(s - lse.unsqueeze(-1))
+    o = torch.matmul(p, vf)
+    return o.to(q.dtype), lse
+
+
+def _sdpa_ref_bwd(do, q, k, v, lse, scale, causal):
+    qf, kf, vf, dof = q.float(), k.float(), v.float(), do.float()
+    rep = 1
+    if k.shape[1] != q.shape[1]:
+        rep = q.shape[1] // k.shape[1]
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal:
+        sq, skv = s.shape[-2], s.shape[-1]
+        mask = torch.ones(sq, skv, dtype=torch.bool, device=s.device).tril()
+        s = s.masked_fill(~mask, float("-inf"))
+    p = torch.exp(s - lse.unsqueeze(-1).float())
+    dv = torch.matmul(p.transpose(-1, -2), dof)
+    dp = torch.matmul(dof, vf.transpose(-1, -2))
+    delta = (dp * p).sum(-1, keepdim=True)
+    ds = p * (dp - delta) * scale
+    dq = torch.matmul(ds, kf)
+    dk = torch.matmul(ds.transpose(-1, -2), qf)
+    if rep > 1:
+        b, h, skv, d = dk.shape
+        dk = dk.view(b, h // rep, rep, skv, d).sum(2)
+        dv = dv.view(b, h // rep, rep, skv, d).sum(2)
+    return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype)
+
+
+def flash_attention(q, k, v, dropout=0.0, causal=False, scale=None,
+                    return_softmax_lse=False, layout="bshd"):
+    """paddle.nn.functional.flash_attention parity
+    (python/paddle/nn/functional/flash_attention.py:195).
+
+    layout "bshd": q [batch, seq, heads, head_dim] (paddle convention);
+    internally computed as [b, h, s, d].
+    """
+    assert dropout == 0.0, "attention dropout not yet supported in the HIP kernel"
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if layout == "bshd":
+        qt = q.transpose(1, 2).contiguous()
+        kt = k.transpose(1, 2).contiguous()
+        vt = v.transpose(1, 2).contiguous()
+    else:
+        qt, kt, vt = q, k, v
+    o = _FlashAttn.apply(qt, kt, vt, scale, causal)
+    if layout == "bshd":
+        o = o.transpose(1, 2)
+    if return_softmax_lse:
+        return o, None  # lse kept internal for now
+    return o, None
+
+
+def scaled_dot_product_attention(query, key, value, attn_mask=None,
+                                 dropout_p=0.0, is_causal=False, training=True):
+    """paddle.nn.functional.scaled_dot_product_attention parity (bshd)."""
+    assert attn_mask is None, "mask path not yet fused; use causal"
+    out, _ = flash_attention(query, key, value, dropout=dropout_p if training else 0.0,
+                             causal=is_causal)
+    return out
+
+
+# ---------------------------------------------------------------------------
+# dropout + residual add (fused_dropout_add_kernel.cu parity)
+# ---------------------------------------------------------------------------
+class _DropoutAdd(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, residual, p, training):
+        ctx.p = p if training else 0.0
+        if _ext.use_native(x):
+            C = _ext.get_ext()
+            seed = int(torch.cuda.default_generators[x.device.index].initial_seed()) & 0x7FFFFFFF \
+                if x.is_cuda else 0
+            import random
+            offset = random.getrandbits(31)
+            outs = C.dropout_add_fwd(x.contiguous(),
+                                     residual.contiguous() if residual is not None else None,
+                                     ctx.p, seed, offset)
+            if ctx.p > 0:
+                y, mask = outs
+                ctx.save_for_backward(mask)
+            else:
+                y = outs[0]
+                ctx.save_for_backward()
+        else:
+            if ctx.p > 0:
+                mask = (torch.rand_like(x, dtype=torch.float32) >= ctx.p)
+                y = x * mask.to(x.dtype) / (1 - ctx.p)
+                ctx.save_for_backward(mask)
+            else:
+                y = x.clone()
+                ctx.save_for_backward()
+            if residual is not None:
+                y = y + residual
+        ctx.has_res = residual is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        p = ctx.p
+        dres = dy if ctx.has_res else None
+        if p == 0:
+            return dy, dres, None, None
+        (mask,) = ctx.saved_tensors
+        if _ext.use_native(dy):
+            C = _ext.get_ext()
+            dx = C.dropout_add_bwd(dy.contiguous(), mask, p)
+        else:
+            dx = dy * mask.to(dy.dtype) / (1 - p)
+        return dx, dres, None, None
+
+
+def dropout_add(x, residual=None, p=0.0, training=True):
+    return _DropoutAdd.apply(x, residual, p, training)
+
+
+# ---------------------------------------------------------------------------
+# embedding (embedding_grad_kernel.cu parity; atomic scatter-add grad)
+# ---------------------------------------------------------------------------
+class _Embedding(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, ids, table, padding_idx):
+        ctx.save_for_backward(ids)
+        ctx.vocab = table.shape[0]
+        ctx.padding_idx = padding_idx
+        ctx.table_dtype = table.dtype
+        if _ext.use_native(table):
+            C = _ext.get_ext()
+            return C.embedding_fwd(table.contiguous(), ids.contiguous().long(),
+                                   padding_idx if padding_idx is not None else -1)
+        return torch.nn.functional.embedding(ids.long(), table, padding_idx=padding_idx)
+
+    @staticmethod
+    def backward(ctx, dout):
+        (ids,) = ctx.saved_tensors
+        if _ext.use_native(dout):
+            C = _ext.get_ext()
+            dtable = C.embedding_bwd(dout.contiguous(), ids.contiguous().long(), ctx.vocab,
+                                     ctx.padding_idx if ctx.padding_idx is not None else -1)
+            dtable = dtable.to(ctx.table_dtype)
+        else:
+            d = dout.shape[-1]
+            dtable = torch.zeros(ctx.vocab, d, dtype=torch.float32, device=dout.device)
+            flat_ids = ids.reshape(-1).long()
+            ok = torch.ones_like(flat_ids, dtype=torch.bool)
+            if ctx.padding_idx is not None:
+                ok = flat_ids != ctx.padding_idx
+            dtable.index_add_(0, flat_ids[ok], dout.reshape(-1, d)[ok].float())
+            dtable = dtable.to(ctx.table_dtype)
+        return None, dtable, None
+
+
+def embedding(ids, table, padding_idx=None):
+    return _Embedding.apply(ids, table, padding_idx)
+
+
+# ---------------------------------------------------------------------------
+# optimizer primitives
+# ---------------------------------------------------------------------------
+def fused_adamw_step(master, param_out, grad, m, v, lr, beta1, beta2, eps,
+                     weight_decay, step):
+    """In-place fused AdamW on a flat fp32 master shard.  param_out may be
+    a bf16 view of the model weights (written by the kernel) or None.
+    adamw_ parity: paddle/phi/kernels/gpu/adamw_kernel.cu."""
+    if master.is_cuda and _ext.use_native(master):
+        C = _ext.get_ext()
+        C.adamw(master, param_out, grad.contiguous(), m, v, lr, beta1, beta2,
+                eps, weight_decay, beta1 ** step, beta2 ** step)
+        return
+    # torch reference (CPU tests)
+    g = grad.float()
+    master.mul_(1 - lr * weight_decay)
+    m.mul_(beta1).add_(g, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+    mhat = m / (1 - beta1 ** step)
+    vhat = v / (1 - beta2 ** step)
+    master.addcdiv_(mhat, vhat.sqrt().add_(eps), value=-lr)
+    if param_out is not None:
+        param_out.copy_(master.to(param_out.dtype))
+
+
+def l2_norm_squared(x):
+    if x.is_cuda and _ext.use_native(x):
+        C = _ext.get_ext()
+        return C.l2norm_sq(x.contiguous())
+    return x.float().square().sum().reshape(1)

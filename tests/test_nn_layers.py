@@ -1,0 +1,107 @@
+"""Layer semantics: shapes, state_dict naming, hooks (reference Layer
+semantics from python/paddle/nn/layer/layers.py)."""
+import numpy as np
+import pytest
+import torch
+
+import paddle_amd as paddle
+from paddle_amd import nn
+
+
+def test_linear_weight_layout():
+    l = nn.Linear(4, 3)
+    assert tuple(l.weight.shape) == (4, 3)  # paddle layout [in, out]
+    x = paddle.randn([2, 4])
+    y = l(x)
+    assert y.shape == (2, 3)
+    np.testing.assert_allclose(y.numpy(), (x @ l.weight + l.bias).detach().numpy(),
+                               rtol=1e-5)
+
+
+def test_layer_state_dict_roundtrip():
+    l = nn.Sequential(nn.Linear(4, 8), nn.ReLU(), nn.Linear(8, 2))
+    sd = l.state_dict()
+    l2 = nn.Sequential(nn.Linear(4, 8), nn.ReLU(), nn.Linear(8, 2))
+    l2.set_state_dict(sd)
+    x = paddle.randn([3, 4])
+    np.testing.assert_allclose(l(x).detach().numpy(), l2(x).detach().numpy(), rtol=1e-6)
+
+
+def test_sublayers_and_parameters():
+    class M(nn.Layer):
+        def __init__(self):
+            super().__init__()
+            self.fc = nn.Linear(2, 2)
+            self.inner = nn.Sequential(nn.Linear(2, 2))
+
+        def forward(self, x):
+            return self.inner(self.fc(x))
+
+    m = M()
+    subs = m.sublayers()
+    assert len(subs) >= 3
+    assert len(m.parameters()) == 4
+    names = [n for n, _ in m.named_parameters()]
+    assert "fc.weight" in names and "inner.0.bias" in names
+
+
+def test_forward_hooks():
+    l = nn.Linear(2, 2)
+    calls = []
+    h1 = l.register_forward_pre_hook(lambda layer, inp: calls.append("pre"))
+    h2 = l.register_forward_post_hook(lambda layer, inp, out: calls.append("post"))
+    l(paddle.randn([1, 2]))
+    assert calls == ["pre", "post"]
+    h1.remove()
+    h2.remove()
+
+
+def test_layernorm_matches_torch():
+    ln = nn.LayerNorm(16)
+    x = paddle.randn([4, 16])
+    ref = torch.nn.functional.layer_norm(x, (16,), ln.weight, ln.bias, 1e-5)
+    torch.testing.assert_close(ln(x), ref, rtol=1e-5, atol=1e-6)
+
+
+def test_embedding_padding_idx():
+    e = nn.Embedding(10, 4, padding_idx=0)
+    out = e(paddle.to_tensor([[0, 1]]))
+    assert out[0, 0].abs().sum() == 0
+
+
+def test_dropout_modes():
+    d = nn.Dropout(0.5)
+    d.eval()
+    x = paddle.ones([10, 10])
+    torch.testing.assert_close(d(x), x)
+    d.train()
+    y = d(x)
+    assert (y == 0).sum() > 0
+
+
+def test_transformer_encoder_layer():
+    layer = nn.TransformerEncoderLayer(d_model=32, nhead=4, dim_feedforward=64,
+                                       dropout=0.0)
+    enc = nn.TransformerEncoder(layer, 2)
+    x = paddle.randn([2, 8, 32])
+    y = enc(x)
+    assert y.shape == (2, 8, 32)
+
+
+def test_multihead_attention_mask():
+    mha = nn.MultiHeadAttention(32, 4, dropout=0.0)
+    x = paddle.randn([2, 6, 32])
+    mask = torch.zeros(2, 4, 6, 6)
+    y = mha(x, x, x, attn_mask=mask)
+    assert y.shape == (2, 6, 32)
+
+
+def test_clip_grad_by_global_norm():
+    m = nn.Linear(4, 4)
+    opt = paddle.optimizer.AdamW(parameters=m.parameters(),
+                                 grad_clip=nn.ClipGradByGlobalNorm(0.1))
+    loss = m(paddle.randn([8, 4])).square().sum() * 100
+    loss.backward()
+    opt._clip_grads()
+    total = sum(float(p.grad.square().sum()) for p in m.parameters())
+    assert total ** 0.5 <= 0.11

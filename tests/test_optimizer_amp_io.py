@@ -1,0 +1,155 @@
+"""Optimizers vs torch references; GradScaler; .pdparams/.pdopt roundtrip;
+LR schedulers (reference: optimizer.py / lr.py / grad_scaler.py / io.py)."""
+import math
+import os
+
+import numpy as np
+import pytest
+import torch
+
+import paddle_amd as paddle
+from paddle_amd import nn
+
+
+def _train_pair(opt_ours_fn, opt_ref_fn, steps=5):
+    torch.manual_seed(0)
+    m1 = nn.Linear(8, 8)
+    m2 = nn.Linear(8, 8)
+    m2.set_state_dict(m1.state_dict())
+    o1 = opt_ours_fn(m1)
+    o2 = opt_ref_fn(m2)
+    x = paddle.randn([16, 8])
+    for _ in range(steps):
+        l1 = m1(x).square().mean()
+        l1.backward()
+        o1.step()
+        o1.clear_grad()
+        l2 = m2(x).square().mean()
+        l2.backward()
+        o2.step()
+        o2.zero_grad()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        torch.testing.assert_close(p1, p2, rtol=1e-4, atol=1e-5)
+
+
+def test_adamw_vs_torch():
+    _train_pair(
+        lambda m: paddle.optimizer.AdamW(learning_rate=1e-2, parameters=m.parameters(),
+                                         weight_decay=0.01),
+        lambda m: torch.optim.AdamW(m.parameters(), lr=1e-2, weight_decay=0.01))
+
+
+def test_sgd_vs_torch():
+    _train_pair(
+        lambda m: paddle.optimizer.SGD(learning_rate=1e-2, parameters=m.parameters()),
+        lambda m: torch.optim.SGD(m.parameters(), lr=1e-2))
+
+
+def test_momentum_vs_torch():
+    _train_pair(
+        lambda m: paddle.optimizer.Momentum(learning_rate=1e-2, momentum=0.9,
+                                            parameters=m.parameters()),
+        lambda m: torch.optim.SGD(m.parameters(), lr=1e-2, momentum=0.9))
+
+
+def test_lr_schedulers():
+    s = paddle.optimizer.lr.CosineAnnealingDecay(0.1, T_max=10)
+    vals = []
+    for _ in range(11):
+        vals.append(s())
+        s.step()
+    assert abs(vals[0] - 0.1) < 1e-9
+    assert vals[-1] < 0.002
+    w = paddle.optimizer.lr.LinearWarmup(0.1, warmup_steps=5, start_lr=0.0, end_lr=0.1)
+    assert w() == 0.0
+    for _ in range(5):
+        w.step()
+    assert abs(w() - 0.1) < 1e-9
+    n = paddle.optimizer.lr.NoamDecay(d_model=512, warmup_steps=100)
+    n.step()
+    assert n() > 0
+
+
+def test_scheduler_drives_optimizer():
+    m = nn.Linear(2, 2)
+    sched = paddle.optimizer.lr.StepDecay(0.1, step_size=1, gamma=0.5)
+    opt = paddle.optimizer.SGD(learning_rate=sched, parameters=m.parameters())
+    assert opt.get_lr() == 0.1
+    sched.step()
+    assert opt.get_lr() == 0.05
+
+
+def test_grad_scaler_flow():
+    m = nn.Linear(4, 4)
+    opt = paddle.optimizer.SGD(learning_rate=0.1, parameters=m.parameters())
+    scaler = paddle.amp.GradScaler(init_loss_scaling=2.0)
+    loss = m(paddle.randn([4, 4])).mean()
+    scaled = scaler.scale(loss)
+    assert abs(float(scaled) - 2 * float(loss)) < 1e-6
+    scaled.backward()
+    scaler.step(opt)
+    scaler.update()
+    opt.clear_grad()
+
+
+def test_grad_scaler_inf_skips_step():
+    m = nn.Linear(2, 2)
+    w0 = m.weight.detach().clone()
+    opt = paddle.optimizer.SGD(learning_rate=1.0, parameters=m.parameters())
+    scaler = paddle.amp.GradScaler(init_loss_scaling=4.0, decr_every_n_nan_or_inf=1)
+    loss = m(paddle.randn([2, 2])).mean()
+    scaler.scale(loss).backward()
+    m.weight.grad[0, 0] = float("inf")
+    scaler.step(opt)
+    scaler.update()
+    torch.testing.assert_close(m.weight.detach(), w0)
+    assert scaler._scale == 2.0  # halved after inf
+
+
+def test_save_load_pdparams(tmp_path):
+    m = nn.Sequential(nn.Linear(4, 4), nn.Linear(4, 2))
+    p = str(tmp_path / "model.pdparams")
+    paddle.save(m.state_dict(), p)
+    loaded = paddle.load(p)
+    m2 = nn.Sequential(nn.Linear(4, 4), nn.Linear(4, 2))
+    m2.set_state_dict(loaded)
+    x = paddle.randn([2, 4])
+    torch.testing.assert_close(m(x), m2(x))
+
+
+def test_save_load_bf16(tmp_path):
+    t = {"w": torch.randn(4, 4).bfloat16(), "step": 7}
+    p = str(tmp_path / "x.pdparams")
+    paddle.save(t, p)
+    back = paddle.load(p)
+    assert back["w"].dtype == torch.bfloat16
+    torch.testing.assert_close(back["w"].float(), t["w"].float())
+    assert back["step"] == 7
+
+
+def test_optimizer_state_dict_roundtrip(tmp_path):
+    m = nn.Linear(4, 4)
+    opt = paddle.optimizer.AdamW(learning_rate=1e-3, parameters=m.parameters())
+    loss = m(paddle.randn([2, 4])).mean()
+    loss.backward()
+    opt.step()
+    p = str(tmp_path / "o.pdopt")
+    paddle.save(opt.state_dict(), p)
+    opt2 = paddle.optimizer.AdamW(learning_rate=1e-3, parameters=m.parameters())
+    opt2.set_state_dict(paddle.load(p))
+    assert opt2._step_count == 1
+
+
+def test_amp_autocast_cpu_noop():
+    with paddle.amp.auto_cast(dtype="bfloat16"):
+        x = paddle.randn([2, 2])
+        y = paddle.matmul(x, x)
+    assert y is not None
+
+
+def test_async_save(tmp_path):
+    p = str(tmp_path / "a.pdparams")
+    t = paddle.async_save({"x": torch.ones(3)}, p)
+    paddle.framework_io.clear_async_save_task_queue()
+    assert os.path.exists(p)
+    assert paddle.load(p)["x"].sum() == 3
