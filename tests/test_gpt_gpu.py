@@ -1,0 +1,76 @@
+"""End-to-end GPU tests: tiny GPT bf16 on the full native kernel path +
+sharding-3 single-GPU step + smoke parity with CPU reference."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    import paddle_amd as paddle
+    from paddle_amd.distributed.fleet.sharding import GroupShardedStage3, ShardedAdamW
+    from paddle_amd.models import GPTPretrainingCriterion, build_gpt
+
+
+def test_gpt_tiny_bf16_step():
+    paddle.seed(0)
+    m = build_gpt("gpt3-tiny").to(device="cuda:0", dtype=torch.bfloat16)
+    loss_fn = GPTPretrainingCriterion()
+    opt = paddle.optimizer.AdamW(learning_rate=3e-4, parameters=m.parameters())
+    ids = torch.randint(0, 1024, (2, 128), device="cuda:0")
+    losses = []
+    for _ in range(10):
+        loss = loss_fn(m(ids), ids)
+        loss.backward()
+        opt.step()
+        opt.clear_grad()
+        losses.append(float(loss.float()))
+    assert losses[-1] < losses[0] - 0.5, losses
+
+
+def test_gpt_sharding3_gpu_world1():
+    paddle.seed(0)
+    m = build_gpt("gpt3-tiny").to(device="cuda:0", dtype=torch.bfloat16)
+    w = GroupShardedStage3(m)
+    opt = ShardedAdamW(w, learning_rate=3e-4,
+                       grad_clip=paddle.nn.ClipGradByGlobalNorm(1.0))
+    loss_fn = GPTPretrainingCriterion()
+    ids = torch.randint(0, 1024, (2, 128), device="cuda:0")
+    losses = []
+    for _ in range(8):
+        loss = loss_fn(w(ids), ids)
+        loss.backward()
+        opt.step()
+        opt.clear_grad()
+        losses.append(float(loss.float()))
+    assert losses[-1] < losses[0] - 0.3, losses
+
+
+def test_gpt_forward_matches_cpu_fp32():
+    """bf16 GPU forward vs fp32 CPU reference of the same weights."""
+    paddle.seed(0)
+    m_cpu = build_gpt("gpt3-tiny")
+    m_gpu = build_gpt("gpt3-tiny")
+    m_gpu.set_state_dict(m_cpu.state_dict())
+    m_gpu = m_gpu.to(device="cuda:0", dtype=torch.bfloat16)
+    ids = torch.randint(0, 1024, (1, 64))
+    with torch.no_grad():
+        lc = m_cpu(ids)
+        lg = m_gpu(ids.to("cuda:0"))
+    # bf16 stack: loose tolerance, but logits must correlate strongly
+    a = lc.flatten().float()
+    b = lg.float().cpu().flatten()
+    corr = torch.corrcoef(torch.stack([a, b]))[0, 1]
+    assert float(corr) > 0.99, float(corr)
+
+
+def test_native_kernels_actually_used():
+    """The GPU path must run our HIP extension, not a torch fallback."""
+    from paddle_amd import _ext
+    assert _ext.has_ext()
+    x = torch.randn(8, 64, device="cuda:0", dtype=torch.bfloat16)
+    w = torch.ones(64, device="cuda:0", dtype=torch.bfloat16)
+    assert _ext.use_native(x)
+    # sanity: flag off forces fallback
+    paddle.set_flags({"FLAGS_use_native_kernels": False})
+    assert not _ext.use_native(x)
+    paddle.set_flags({"FLAGS_use_native_kernels": True})
