@@ -290,8 +290,10 @@ Tensor dropout_add_bwd(const Tensor& dy, const Tensor& mask, double p) {
 }
 
 // ---- embedding ------------------------------------------------------------
-Tensor embedding_fwd(const Tensor& table, const Tensor& ids, int64_t padding_idx) {
-  CHECK_IN(table); CHECK_IN(ids);
+Tensor embedding_fwd(const Tensor& table, const Tensor& ids_in, int64_t padding_idx) {
+  CHECK_IN(table);
+  TORCH_CHECK(ids_in.is_cuda(), "ids must be on GPU");
+  auto ids = ids_in.contiguous();
   int64_t vocab = table.size(0), d = table.size(1), n = ids.numel();
   TORCH_CHECK(d % 8 == 0);
   auto sizes = ids.sizes().vec();
@@ -303,9 +305,11 @@ Tensor embedding_fwd(const Tensor& table, const Tensor& ids, int64_t padding_idx
   return out;
 }
 
-Tensor embedding_bwd(const Tensor& dout, const Tensor& ids, int64_t vocab,
+Tensor embedding_bwd(const Tensor& dout, const Tensor& ids_in, int64_t vocab,
                      int64_t padding_idx) {
-  CHECK_IN(dout); CHECK_IN(ids);
+  CHECK_IN(dout);
+  TORCH_CHECK(ids_in.is_cuda(), "ids must be on GPU");
+  auto ids = ids_in.contiguous();
   int64_t d = dout.size(-1), n = ids.numel();
   auto dtable = torch::zeros({vocab, d}, dout.options().dtype(torch::kFloat));
   pa::embedding_bwd(dout.const_data_ptr(), ids.const_data_ptr<int64_t>(),
