@@ -225,39 +225,65 @@ Tensor l2norm_sq(const Tensor& x) {
 }
 
 // ---- flash attention ------------------------------------------------------
+// Tensors are logical [B, H, S, D] but may be strided VIEWS (e.g. of a packed
+// [B, S, 3, H, D] qkv) as long as D is contiguous and rows are 16B-aligned.
+static void fa_strides(const Tensor& t, int64_t* out) {
+  TORCH_CHECK(t.stride(3) == 1, "flash_attn: head_dim must be contiguous");
+  out[0] = t.stride(0);  // batch
+  out[1] = t.stride(1);  // head
+  out[2] = t.stride(2);  // seq
+  TORCH_CHECK(out[2] % 8 == 0, "flash_attn: seq stride must keep 16B alignment");
+}
+
 std::vector<Tensor> flash_attn_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
+                                   c10::optional<Tensor> o_out,
                                    double scale, bool causal) {
-  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v);
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "flash_attn: bf16 only");
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16, "flash_attn: bf16 GPU only");
   TORCH_CHECK(q.dim() == 4, "flash_attn expects [B,H,S,D]");
   int64_t b = q.size(0), h = q.size(1), sq = q.size(2), d = q.size(3);
   int64_t hkv = k.size(1), skv = k.size(2);
   TORCH_CHECK(d == 128 || d == 64, "flash_attn: head_dim must be 64/128");
   TORCH_CHECK(h % hkv == 0);
-  auto o = torch::empty_like(q);
+  int64_t qs[3], ks[3], vs[3], os[3];
+  fa_strides(q, qs); fa_strides(k, ks); fa_strides(v, vs);
+  TORCH_CHECK(ks[0] == vs[0] && ks[1] == vs[1] && ks[2] == vs[2],
+              "flash_attn: k/v must share strides");
+  auto o = o_out.has_value() ? *o_out : torch::empty({b, h, sq, d}, q.options());
+  fa_strides(o, os);
   auto lse = torch::empty({b, h, sq}, q.options().dtype(torch::kFloat));
   pa::flash_attn_fwd(q.const_data_ptr(), k.const_data_ptr(), v.const_data_ptr(),
                      o.mutable_data_ptr(), lse.mutable_data_ptr<float>(), b, h,
-                     hkv, sq, skv, d, (float)scale, causal, cur_stream());
+                     hkv, sq, skv, d, (float)scale, causal, qs, ks, os,
+                     cur_stream());
   return {o, lse};
 }
 
 std::vector<Tensor> flash_attn_bwd(const Tensor& dout, const Tensor& q, const Tensor& k,
                                    const Tensor& v, const Tensor& o, const Tensor& lse,
+                                   c10::optional<Tensor> dq_out,
+                                   c10::optional<Tensor> dk_out,
+                                   c10::optional<Tensor> dv_out,
                                    double scale, bool causal) {
-  CHECK_IN(dout); CHECK_IN(q); CHECK_IN(k); CHECK_IN(v); CHECK_IN(o);
   int64_t b = q.size(0), h = q.size(1), sq = q.size(2), d = q.size(3);
   int64_t hkv = k.size(1), skv = k.size(2);
   TORCH_CHECK(hkv == h, "flash_attn_bwd kernel requires hkv==h (expand KV upstream)");
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
+  auto dq = dq_out.has_value() ? *dq_out : torch::empty({b, h, sq, d}, q.options());
+  auto dk = dk_out.has_value() ? *dk_out : torch::empty({b, hkv, skv, d}, k.options());
+  auto dv = dv_out.has_value() ? *dv_out : torch::empty({b, hkv, skv, d}, v.options());
   auto delta = torch::empty({b, h, sq}, q.options().dtype(torch::kFloat));
+  int64_t qs[3], ks[3], vs[3], dos[3], os[3], dqs[3], dks[3], dvs[3];
+  fa_strides(q, qs); fa_strides(k, ks); fa_strides(v, vs);
+  fa_strides(dout, dos); fa_strides(o, os);
+  fa_strides(dq, dqs); fa_strides(dk, dks); fa_strides(dv, dvs);
+  TORCH_CHECK(ks[0] == vs[0] && ks[1] == vs[1] && ks[2] == vs[2]);
+  TORCH_CHECK(dks[0] == dvs[0] && dks[1] == dvs[1] && dks[2] == dvs[2],
+              "dk/dv must share strides");
   pa::flash_attn_bwd(dout.const_data_ptr(), q.const_data_ptr(), k.const_data_ptr(),
                      v.const_data_ptr(), o.const_data_ptr(), lse.const_data_ptr<float>(),
                      dq.mutable_data_ptr(), dk.mutable_data_ptr(), dv.mutable_data_ptr(),
                      delta.mutable_data_ptr<float>(), b, h, hkv, sq, skv, d,
-                     (float)scale, causal, cur_stream());
+                     (float)scale, causal, qs, ks, dos, os, dqs, dks,
+                     cur_stream());
   return {dq, dk, dv};
 }
 

@@ -76,17 +76,23 @@ void l2norm_sq(const void* x, float* out, int64_t numel, int dtype,
                hipStream_t s);
 
 // ---- flash attention (bf16, head_dim 128 or 64) ----------------------------
-// q,k,v,o: [b, h, s, dh] contiguous bf16; lse fp32 [b, h, s]
+// Logical layout [b, h, s, dh]; arbitrary (batch, head, seq) ELEMENT strides
+// per tensor (d must be innermost/contiguous, 16B-aligned rows) so packed
+// [b, s, 3, h, dh] qkv views need no transpose copies.  Strides are int64[3]
+// = {batch, head, seq}.  lse/delta fp32 [b, h, sq] contiguous.
 void flash_attn_fwd(const void* q, const void* k, const void* v, void* o,
                     float* lse, int64_t b, int64_t h, int64_t hkv, int64_t sq,
                     int64_t skv, int64_t dh, float scale, bool causal,
+                    const int64_t* qs, const int64_t* ks, const int64_t* os,
                     hipStream_t s);
-// delta fp32 [b,h,sq] workspace computed internally
 void flash_attn_bwd(const void* dout, const void* q, const void* k,
                     const void* v, const void* o, const float* lse,
                     void* dq, void* dk, void* dv, float* delta, int64_t b,
                     int64_t h, int64_t hkv, int64_t sq, int64_t skv,
-                    int64_t dh, float scale, bool causal, hipStream_t s);
+                    int64_t dh, float scale, bool causal,
+                    const int64_t* qs, const int64_t* ks, const int64_t* dos,
+                    const int64_t* os, const int64_t* dqs, const int64_t* dks,
+                    hipStream_t s);
 
 // ---- dropout + residual add ------------------------------------------------
 // y = dropout(x, p) + residual; mask stored as uint8 per element

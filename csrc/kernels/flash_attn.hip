@@ -41,7 +41,10 @@ __launch_bounds__(256)
 __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restrict__ kg,
                               const short* __restrict__ vg, short* __restrict__ og,
                               float* __restrict__ lseg, int B, int H, int HKV,
-                              int Sq, int Skv, float scale) {
+                              int Sq, int Skv, float scale,
+                              long long q_sb, long long q_sh, long long q_ss,
+                              long long k_sb, long long k_sh, long long k_ss,
+                              long long o_sb, long long o_sh, long long o_ss) {
   constexpr int QB = 64, KB = 64;
   constexpr int NKK = D / 32;   // mfma k-steps over head dim
   constexpr int NDT = D / 16;   // output d tiles
@@ -57,8 +60,9 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
   const int b = bh / H, h = bh % H;
   const int hkv = h / (H / HKV);
   const int q0 = qblk * QB;
-  const long long qbase = ((long long)(b * H + h)) * Sq * D;
-  const long long kbase = ((long long)(b * HKV + hkv)) * Skv * D;
+  const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
+  const long long kbase = (long long)b * k_sb + (long long)hkv * k_sh;
+  const long long obase = (long long)b * o_sb + (long long)h * o_sh;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -74,7 +78,7 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
 #pragma unroll
     for (int kk = 0; kk < NKK; ++kk) {
       if (ok)
-        qf[kk] = *reinterpret_cast<const shortx8*>(qg + qbase + (long long)row * D + kk * 32 + lg * 8);
+        qf[kk] = *reinterpret_cast<const shortx8*>(qg + qbase + (long long)row * q_ss + kk * 32 + lg * 8);
       else
         for (int i = 0; i < 8; ++i) qf[kk][i] = 0;
     }
@@ -97,7 +101,7 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
         int row = flat / D, col = flat % D;
         shortx8 val;
         if (kv0 + row < Skv)
-          val = *reinterpret_cast<const shortx8*>(kg + kbase + (long long)(kv0 + row) * D + col);
+          val = *reinterpret_cast<const shortx8*>(kg + kbase + (long long)(kv0 + row) * k_ss + col);
         else
           for (int i = 0; i < 8; ++i) val[i] = 0;
         *reinterpret_cast<shortx8*>(k_lds + lds_off(row, col * 2, K_RS)) = val;
@@ -107,7 +111,7 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
         int row = flat / D, col = flat % D;  // row=kv, col=d
         shortx8 val;
         if (kv0 + row < Skv)
-          val = *reinterpret_cast<const shortx8*>(vg + kbase + (long long)(kv0 + row) * D + col);
+          val = *reinterpret_cast<const shortx8*>(vg + kbase + (long long)(kv0 + row) * k_ss + col);
         else
           for (int i = 0; i < 8; ++i) val[i] = 0;
 #pragma unroll
@@ -203,7 +207,7 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
     float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
 #pragma unroll
     for (int dt = 0; dt < NDT; ++dt)
-      og[qbase + (long long)row * D + dt * 16 + l16] = f2bf(oacc[dt][r] * inv_l);
+      og[obase + (long long)row * o_ss + dt * 16 + l16] = f2bf(oacc[dt][r] * inv_l);
     if (l16 == 0)
       lseg[(long long)(b * H + h) * Sq + row] =
           (l_run[r] > 0.f) ? m_run[r] + __logf(l_run[r]) : -INFINITY;
@@ -215,17 +219,25 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
 // ---------------------------------------------------------------------------
 template <int D>
 __global__ void fa_bwd_delta_kernel(const short* __restrict__ dog, const short* __restrict__ og,
-                                    float* __restrict__ delta, long long rows) {
+                                    float* __restrict__ delta, int H, int Sq,
+                                    long long rows,
+                                    long long do_sb, long long do_sh, long long do_ss,
+                                    long long o_sb, long long o_sh, long long o_ss) {
   long long wid = ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   int lane = threadIdx.x & 63;
   long long nwaves = ((long long)gridDim.x * blockDim.x) >> 6;
   for (long long r = wid; r < rows; r += nwaves) {
+    int sq = (int)(r % Sq);
+    int h = (int)((r / Sq) % H);
+    int b = (int)(r / ((long long)Sq * H));
+    const long long dob = b * do_sb + h * do_sh + sq * do_ss;
+    const long long ob = b * o_sb + h * o_sh + sq * o_ss;
     float acc = 0.f;
     for (int j = lane * 8; j < D; j += 64 * 8) {
-      shortx8 a = *reinterpret_cast<const shortx8*>(dog + r * D + j);
-      shortx8 b = *reinterpret_cast<const shortx8*>(og + r * D + j);
+      shortx8 a = *reinterpret_cast<const shortx8*>(dog + dob + j);
+      shortx8 bb = *reinterpret_cast<const shortx8*>(og + ob + j);
 #pragma unroll
-      for (int k = 0; k < 8; ++k) acc += bf2f(a[k]) * bf2f(b[k]);
+      for (int k = 0; k < 8; ++k) acc += bf2f(a[k]) * bf2f(bb[k]);
     }
     acc = wave_reduce(acc, SumOp());
     if (lane == 0) delta[r] = acc;
@@ -242,7 +254,11 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
                                   const short* __restrict__ kg, const short* __restrict__ vg,
                                   const float* __restrict__ lseg, const float* __restrict__ deltag,
                                   short* __restrict__ dkg, short* __restrict__ dvg,
-                                  int B, int H, int Sq, int Skv, float scale) {
+                                  int B, int H, int Sq, int Skv, float scale,
+                                  long long q_sb, long long q_sh, long long q_ss,
+                                  long long k_sb, long long k_sh, long long k_ss,
+                                  long long do_sb, long long do_sh, long long do_ss,
+                                  long long dk_sb, long long dk_sh, long long dk_ss) {
   constexpr int KB = 64, QB = 64;
   constexpr int NKK = D / 32;
   constexpr int NDT = D / 16;
@@ -256,10 +272,12 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
 
   const int kvblk = blockIdx.x;
   const int bh = blockIdx.y;
-  const int b = bh / H, h = bh % H;
   const int kv0 = kvblk * KB;
-  const long long base = ((long long)bh) * Sq * D;       // Sq == Skv for causal; general below
-  const long long kvbase = ((long long)bh) * Skv * D;
+  const int b = bh / H, h = bh % H;
+  const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
+  const long long dobase = (long long)b * do_sb + (long long)h * do_sh;
+  const long long kvbase = (long long)b * k_sb + (long long)h * k_sh;
+  const long long dkbase = (long long)b * dk_sb + (long long)h * dk_sh;
   const long long lse_base = ((long long)bh) * Sq;
 
   const int tid = threadIdx.x;
@@ -276,8 +294,8 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
 #pragma unroll
     for (int kk = 0; kk < NKK; ++kk) {
       if (ok) {
-        kf[kk] = *reinterpret_cast<const shortx8*>(kg + kvbase + (long long)row * D + kk * 32 + lg * 8);
-        vf[kk] = *reinterpret_cast<const shortx8*>(vg + kvbase + (long long)row * D + kk * 32 + lg * 8);
+        kf[kk] = *reinterpret_cast<const shortx8*>(kg + kvbase + (long long)row * k_ss + kk * 32 + lg * 8);
+        vf[kk] = *reinterpret_cast<const shortx8*>(vg + kvbase + (long long)row * k_ss + kk * 32 + lg * 8);
       } else {
         for (int i = 0; i < 8; ++i) { kf[kk][i] = 0; vf[kk][i] = 0; }
       }
@@ -301,8 +319,8 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
         int row = flat / D, col = flat % D;
         shortx8 qv, dv;
         if (q0 + row < Sq) {
-          qv = *reinterpret_cast<const shortx8*>(qg + base + (long long)(q0 + row) * D + col);
-          dv = *reinterpret_cast<const shortx8*>(dog + base + (long long)(q0 + row) * D + col);
+          qv = *reinterpret_cast<const shortx8*>(qg + qbase + (long long)(q0 + row) * q_ss + col);
+          dv = *reinterpret_cast<const shortx8*>(dog + dobase + (long long)(q0 + row) * do_ss + col);
         } else {
           for (int i = 0; i < 8; ++i) { qv[i] = 0; dv[i] = 0; }
         }
@@ -411,8 +429,8 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
     if (row >= Skv) continue;
 #pragma unroll
     for (int dt = 0; dt < NDT; ++dt) {
-      dkg[kvbase + (long long)row * D + dt * 16 + l16] = f2bf(dk_acc[dt][r]);
-      dvg[kvbase + (long long)row * D + dt * 16 + l16] = f2bf(dv_acc[dt][r]);
+      dkg[dkbase + (long long)row * dk_ss + dt * 16 + l16] = f2bf(dk_acc[dt][r]);
+      dvg[dkbase + (long long)row * dk_ss + dt * 16 + l16] = f2bf(dv_acc[dt][r]);
     }
   }
 }
@@ -426,7 +444,11 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
                                  const short* __restrict__ kg, const short* __restrict__ vg,
                                  const float* __restrict__ lseg, const float* __restrict__ deltag,
                                  short* __restrict__ dqg, int B, int H, int Sq, int Skv,
-                                 float scale) {
+                                 float scale,
+                                 long long q_sb, long long q_sh, long long q_ss,
+                                 long long k_sb, long long k_sh, long long k_ss,
+                                 long long do_sb, long long do_sh, long long do_ss,
+                                 long long dq_sb, long long dq_sh, long long dq_ss) {
   constexpr int QB = 64, KB = 64;
   constexpr int NKK = D / 32;
   constexpr int NDT = D / 16;
@@ -439,9 +461,12 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
 
   const int qblk = blockIdx.x;
   const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
   const int q0 = qblk * QB;
-  const long long base = ((long long)bh) * Sq * D;
-  const long long kvbase = ((long long)bh) * Skv * D;
+  const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
+  const long long dobase = (long long)b * do_sb + (long long)h * do_sh;
+  const long long dqbase = (long long)b * dq_sb + (long long)h * dq_sh;
+  const long long kvbase = (long long)b * k_sb + (long long)h * k_sh;
   const long long lse_base = ((long long)bh) * Sq;
 
   const int tid = threadIdx.x;
@@ -457,8 +482,8 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
 #pragma unroll
     for (int kk = 0; kk < NKK; ++kk) {
       if (ok) {
-        qf[kk] = *reinterpret_cast<const shortx8*>(qg + base + (long long)row * D + kk * 32 + lg * 8);
-        dof[kk] = *reinterpret_cast<const shortx8*>(dog + base + (long long)row * D + kk * 32 + lg * 8);
+        qf[kk] = *reinterpret_cast<const shortx8*>(qg + qbase + (long long)row * q_ss + kk * 32 + lg * 8);
+        dof[kk] = *reinterpret_cast<const shortx8*>(dog + dobase + (long long)row * do_ss + kk * 32 + lg * 8);
       } else {
         for (int i = 0; i < 8; ++i) { qf[kk][i] = 0; dof[kk][i] = 0; }
       }
@@ -484,8 +509,8 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
         int row = flat / D, col = flat % D;
         shortx8 kv_, vv;
         if (kv0 + row < Skv) {
-          kv_ = *reinterpret_cast<const shortx8*>(kg + kvbase + (long long)(kv0 + row) * D + col);
-          vv = *reinterpret_cast<const shortx8*>(vg + kvbase + (long long)(kv0 + row) * D + col);
+          kv_ = *reinterpret_cast<const shortx8*>(kg + kvbase + (long long)(kv0 + row) * k_ss + col);
+          vv = *reinterpret_cast<const shortx8*>(vg + kvbase + (long long)(kv0 + row) * k_ss + col);
         } else {
           for (int i = 0; i < 8; ++i) { kv_[i] = 0; vv[i] = 0; }
         }
@@ -546,7 +571,7 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
     if (row >= Sq) continue;
 #pragma unroll
     for (int dt = 0; dt < NDT; ++dt)
-      dqg[base + (long long)row * D + dt * 16 + l16] = f2bf(dq_acc[dt][r]);
+      dqg[dqbase + (long long)row * dq_ss + dt * 16 + l16] = f2bf(dq_acc[dt][r]);
   }
 }
 
@@ -556,13 +581,16 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
 void flash_attn_fwd(const void* q, const void* k, const void* v, void* o,
                     float* lse, int64_t b, int64_t h, int64_t hkv, int64_t sq,
                     int64_t skv, int64_t dh, float scale, bool causal,
+                    const int64_t* qs, const int64_t* ks, const int64_t* os,
                     hipStream_t s) {
   dim3 grid((unsigned)cdiv((int)sq, 64), (unsigned)(b * h));
   dim3 blk(256);
-#define FAF(D, C)                                                            \
+#define FAF(D, C)                                                             \
   hipLaunchKernelGGL((fa_fwd_kernel<D, C>), grid, blk, 0, s, (const short*)q, \
                      (const short*)k, (const short*)v, (short*)o, lse,        \
-                     (int)b, (int)h, (int)hkv, (int)sq, (int)skv, scale)
+                     (int)b, (int)h, (int)hkv, (int)sq, (int)skv, scale,      \
+                     qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],                \
+                     os[0], os[1], os[2])
   if (dh == 128) { if (causal) FAF(128, true); else FAF(128, false); }
   else           { if (causal) FAF(64, true);  else FAF(64, false); }
 #undef FAF
@@ -572,15 +600,20 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
                     const void* v, const void* o, const float* lse,
                     void* dq, void* dk, void* dv, float* delta, int64_t b,
                     int64_t h, int64_t hkv, int64_t sq, int64_t skv,
-                    int64_t dh, float scale, bool causal, hipStream_t s) {
+                    int64_t dh, float scale, bool causal,
+                    const int64_t* qs, const int64_t* ks, const int64_t* dos,
+                    const int64_t* os, const int64_t* dqs, const int64_t* dks,
+                    hipStream_t s) {
   long long rows = b * h * sq;
   dim3 dgrid((unsigned)hmin<long long>(2048LL, (rows + 3) / 4));
   if (dh == 128)
     hipLaunchKernelGGL((fa_bwd_delta_kernel<128>), dgrid, dim3(256), 0, s,
-                       (const short*)dout, (const short*)o, delta, rows);
+                       (const short*)dout, (const short*)o, delta, (int)h,
+                       (int)sq, rows, dos[0], dos[1], dos[2], os[0], os[1], os[2]);
   else
     hipLaunchKernelGGL((fa_bwd_delta_kernel<64>), dgrid, dim3(256), 0, s,
-                       (const short*)dout, (const short*)o, delta, rows);
+                       (const short*)dout, (const short*)o, delta, (int)h,
+                       (int)sq, rows, dos[0], dos[1], dos[2], os[0], os[1], os[2]);
 
   dim3 gkv((unsigned)cdiv((int)skv, 64), (unsigned)(b * h));
   dim3 gq((unsigned)cdiv((int)sq, 64), (unsigned)(b * h));
@@ -589,11 +622,15 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
     hipLaunchKernelGGL((fa_bwd_dkv_kernel<D, C>), gkv, dim3(256), 0, s,        \
                        (const short*)dout, (const short*)q, (const short*)k,   \
                        (const short*)v, lse, delta, (short*)dk, (short*)dv,    \
-                       (int)b, (int)h, (int)sq, (int)skv, scale);              \
+                       (int)b, (int)h, (int)sq, (int)skv, scale,               \
+                       qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],               \
+                       dos[0], dos[1], dos[2], dks[0], dks[1], dks[2]);        \
     hipLaunchKernelGGL((fa_bwd_dq_kernel<D, C>), gq, dim3(256), 0, s,          \
                        (const short*)dout, (const short*)q, (const short*)k,   \
                        (const short*)v, lse, delta, (short*)dq, (int)b,        \
-                       (int)h, (int)sq, (int)skv, scale);                      \
+                       (int)h, (int)sq, (int)skv, scale,                       \
+                       qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],               \
+                       dos[0], dos[1], dos[2], dqs[0], dqs[1], dqs[2]);        \
   } while (0)
   if (dh == 128) { if (causal) FAB(128, true); else FAB(128, false); }
   else           { if (causal) FAB(64, true);  else FAB(64, false); }

@@ -89,14 +89,15 @@ class _Unit:
         if self.full is not None:
             return
         with torch.no_grad():
-            self.full = torch.empty(self.total, dtype=self.dtype, device=self.shard.device)
             if self.world == 1:
-                self.full.copy_(self.shard)
+                # degenerate shard == full: alias, no copy (1-GPU fast path)
+                self.full = self.shard
                 self.gather_work = None
-            else:
-                self.gather_work = C.all_gather_into_tensor(self.full, self.shard,
-                                                            group=self.group,
-                                                            sync_op=not async_op)
+                return
+            self.full = torch.empty(self.total, dtype=self.dtype, device=self.shard.device)
+            self.gather_work = C.all_gather_into_tensor(self.full, self.shard,
+                                                        group=self.group,
+                                                        sync_op=not async_op)
 
     def wait_gather_and_bind(self):
         if self.full is None:
@@ -119,25 +120,37 @@ class _Unit:
         self.gathered = False
 
     # -- grads ---------------------------------------------------------------
-    def reduce_grads(self, accumulate):
-        grads = []
-        flat = torch.zeros(self.total, dtype=self.dtype, device=self.shard.device)
+    # autograd accumulates DIRECTLY into grad_flat: bind_grad_views() points
+    # every p.grad at a slice before the unit's backward runs, so there is no
+    # flatten copy at reduce time (the reference's FusedCommBuffer does the
+    # same for stage-1-v2; here it covers stage-3 too).
+    def bind_grad_views(self):
+        if self.grad_flat is None:
+            self.grad_flat = torch.zeros(self.total, dtype=self.dtype,
+                                         device=self.shard.device)
         for p, o, n in zip(self.params, self.offsets, self.numels):
-            if p.grad is not None:
-                flat[o:o + n].copy_(p.grad.reshape(-1))
-                p.grad = None
-        out = torch.empty(self.shard_size, dtype=self.dtype, device=self.shard.device)
+            p.grad = self.grad_flat[o:o + n].view(p._orig_shape)
+
+    def reduce_grads(self, accumulate):
+        flat = self.grad_flat
+        for p in self.params:
+            p.grad = None
         if self.world == 1:
-            out.copy_(flat[0:self.shard_size])
+            if accumulate and self.accum_steps > 0:
+                self.grad_shard_fp32.add_(flat[0:self.shard_size])
+            else:
+                self.grad_shard_fp32.copy_(flat[0:self.shard_size])
         else:
             flat.div_(self.world)
+            out = torch.empty(self.shard_size, dtype=self.dtype,
+                              device=self.shard.device)
             C.reduce_scatter_tensor(out, flat, group=self.group)
-        if accumulate and self.accum_steps > 0:
-            self.grad_shard_fp32.add_(out.float())
-        else:
-            self.grad_shard_fp32.copy_(out.float())
+            if accumulate and self.accum_steps > 0:
+                self.grad_shard_fp32.add_(out)
+            else:
+                self.grad_shard_fp32.copy_(out)
         self.accum_steps += 1
-        del flat
+        flat.zero_()
 
 
 class GroupShardedStage3(torch.nn.Module):
@@ -254,6 +267,7 @@ class GroupShardedStage3(torch.nn.Module):
                 self._order_recorded = True
             u.in_backward = True
             u.wait_gather_and_bind()
+            u.bind_grad_views()  # grads accumulate straight into the flat buffer
             # prefetch previous unit (next to run backward)
             pos = self._order.index(u.idx)
             if pos - 1 >= 0:

@@ -88,11 +88,8 @@ class GPTAttention(nn.Layer):
     def forward(self, x):
         b, s, h = x.shape
         qkv = self.qkv_proj(x).reshape(b, s, 3, self.num_heads, self.head_dim)
-        q, k, v = qkv.unbind(2)  # [b, s, nh, hd]
-        out, _ = hot.flash_attention(q, k, v,
-                                     dropout=self.attn_dropout if self.training else 0.0,
-                                     causal=True)
-        out = out.reshape(b, s, h)
+        # packed zero-copy path: strided q/k/v views + in-place dqkv
+        out = hot.qkv_flash_attention(qkv, causal=True)
         return self.out_proj(out)
 
 

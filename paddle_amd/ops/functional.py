@@ -344,14 +344,18 @@ def fused_rotary_position_embedding(q, k=None, v=None, sin=None, cos=None,
 # ---------------------------------------------------------------------------
 # flash attention (flash_attn_kernel.cu API parity)
 # ---------------------------------------------------------------------------
+def _fa_native_ok(q):
+    return (_ext.use_native(q) and q.dtype == torch.bfloat16
+            and q.shape[-1] in (64, 128) and q.stride(-1) == 1)
+
+
 class _FlashAttn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale, causal):
-        # q,k,v: [B, H, S, D] contiguous bf16
-        if _ext.use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+        # q,k,v: [B, H, S, D] bf16 (views allowed; D contiguous)
+        if _fa_native_ok(q):
             C = _ext.get_ext()
-            o, lse = C.flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(),
-                                      scale, causal)
+            o, lse = C.flash_attn_fwd(q, k, v, None, scale, causal)
             ctx.native = True
         else:
             o, lse = _sdpa_ref(q, k, v, scale, causal)
@@ -365,21 +369,87 @@ class _FlashAttn(torch.autograd.Function):
         q, k, v, o, lse = ctx.saved_tensors
         if ctx.native:
             C = _ext.get_ext()
+            if do.stride(-1) != 1:
+                do = do.contiguous()
             if k.shape[1] != q.shape[1]:  # GQA: expand for bwd kernel
                 rep = q.shape[1] // k.shape[1]
                 ke = k.repeat_interleave(rep, dim=1)
                 ve = v.repeat_interleave(rep, dim=1)
-                dq, dke, dve = C.flash_attn_bwd(do.contiguous(), q, ke.contiguous(),
-                                                ve.contiguous(), o, lse, ctx.scale, ctx.causal)
+                dq, dke, dve = C.flash_attn_bwd(do, q, ke.contiguous(), ve.contiguous(),
+                                                o, lse, None, None, None,
+                                                ctx.scale, ctx.causal)
                 hkv = k.shape[1]
                 dk = dke.view(k.shape[0], hkv, rep, k.shape[2], k.shape[3]).sum(2)
                 dv = dve.view(v.shape[0], hkv, rep, v.shape[2], v.shape[3]).sum(2)
             else:
-                dq, dk, dv = C.flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
+                dq, dk, dv = C.flash_attn_bwd(do, q, k, v, o, lse, None, None, None,
                                               ctx.scale, ctx.causal)
         else:
             dq, dk, dv = _sdpa_ref_bwd(do, q, k, v, lse, ctx.scale, ctx.causal)
         return dq, dk, dv, None, None
+
+
+class _QKVFlashAttn(torch.autograd.Function):
+    """Zero-copy causal attention on a packed [B, S, 3, H, D] qkv tensor
+    (the GPT hot path): q/k/v are strided views, the output is written
+    directly in [B, S, H*D] layout, and backward writes dqkv in place --
+    no transpose/cat kernels at all."""
+
+    @staticmethod
+    def forward(ctx, qkv, scale, causal):
+        b, s, three, h, d = qkv.shape
+        q = qkv[:, :, 0].permute(0, 2, 1, 3)  # [b,h,s,d] strided view
+        k = qkv[:, :, 1].permute(0, 2, 1, 3)
+        v = qkv[:, :, 2].permute(0, 2, 1, 3)
+        if _fa_native_ok(q):
+            C = _ext.get_ext()
+            out = torch.empty(b, s, h, d, dtype=qkv.dtype, device=qkv.device)
+            o_view = out.permute(0, 2, 1, 3)
+            _, lse = C.flash_attn_fwd(q, k, v, o_view, scale, causal)
+            ctx.native = True
+            ctx.save_for_backward(qkv, out, lse)
+        else:
+            o, lse = _sdpa_ref(q.contiguous(), k.contiguous(), v.contiguous(),
+                               scale, causal)
+            out = o.permute(0, 2, 1, 3).reshape(b, s, h, d)
+            ctx.native = False
+            ctx.save_for_backward(qkv, out, lse)
+        ctx.scale, ctx.causal = scale, causal
+        return out.reshape(b, s, h * d)
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, out, lse = ctx.saved_tensors
+        b, s, three, h, d = qkv.shape
+        q = qkv[:, :, 0].permute(0, 2, 1, 3)
+        k = qkv[:, :, 1].permute(0, 2, 1, 3)
+        v = qkv[:, :, 2].permute(0, 2, 1, 3)
+        o_view = out.permute(0, 2, 1, 3)
+        do_view = do.reshape(b, s, h, d).permute(0, 2, 1, 3)
+        if ctx.native:
+            C = _ext.get_ext()
+            if do_view.stride(-1) != 1:
+                do_view = do.contiguous().reshape(b, s, h, d).permute(0, 2, 1, 3)
+            dqkv = torch.empty_like(qkv)
+            dq = dqkv[:, :, 0].permute(0, 2, 1, 3)
+            dk = dqkv[:, :, 1].permute(0, 2, 1, 3)
+            dv = dqkv[:, :, 2].permute(0, 2, 1, 3)
+            C.flash_attn_bwd(do_view, q, k, v, o_view, lse, dq, dk, dv,
+                             ctx.scale, ctx.causal)
+        else:
+            dq, dk, dv = _sdpa_ref_bwd(do_view.contiguous(), q.contiguous(),
+                                       k.contiguous(), v.contiguous(), lse,
+                                       ctx.scale, ctx.causal)
+            # each [b,h,s,d]; stack -> [3,b,h,s,d]; permute -> [b,s,3,h,d]
+            dqkv = torch.stack([dq, dk, dv], dim=0).permute(1, 3, 0, 2, 4).contiguous()
+        return dqkv, None, None
+
+
+def qkv_flash_attention(qkv, scale=None, causal=True):
+    """qkv: [B, S, 3, H, D] packed (straight out of the fused QKV GEMM)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(qkv.shape[-1])
+    return _QKVFlashAttn.apply(qkv, scale, causal)
 
 
 def _sdpa_ref(q, k, v, scale, causal):
@@ -438,9 +508,10 @@ def flash_attention(q, k, v, dropout=0.0, causal=False, scale=None,
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if layout == "bshd":
-        qt = q.transpose(1, 2).contiguous()
-        kt = k.transpose(1, 2).contiguous()
-        vt = v.transpose(1, 2).contiguous()
+        # strided views -- the kernel is stride-aware, no copies
+        qt = q.transpose(1, 2)
+        kt = k.transpose(1, 2)
+        vt = v.transpose(1, 2)
     else:
         qt, kt, vt = q, k, v
     o = _FlashAttn.apply(qt, kt, vt, scale, causal)

@@ -48,7 +48,8 @@ def test_gpt_sharding3_gpu_world1():
 def test_gpt_forward_matches_cpu_fp32():
     """bf16 GPU forward vs fp32 CPU reference of the same weights."""
     paddle.seed(0)
-    m_cpu = build_gpt("gpt3-tiny")
+    # on a GPU box params default to cuda -- pin the reference copy to CPU
+    m_cpu = build_gpt("gpt3-tiny").to(device="cpu")
     m_gpu = build_gpt("gpt3-tiny")
     m_gpu.set_state_dict(m_cpu.state_dict())
     m_gpu = m_gpu.to(device="cuda:0", dtype=torch.bfloat16)
