@@ -6,11 +6,14 @@ attn FLOPs (causal): fwd = 2 * 2 * B*H*S^2*D * 0.5 ; bwd = 2.5x fwd.
 """
 import argparse
 import math
+import os
+import sys
 import time
 
 import torch
 
-import paddle_amd  # noqa: F401  (patches + ext)
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import paddle_amd  # noqa: F401,E402  (patches + ext)
 from paddle_amd import _ext
 
 
