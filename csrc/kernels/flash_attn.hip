@@ -114,8 +114,13 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
           val = *reinterpret_cast<const shortx8*>(vg + kbase + (long long)(kv0 + row) * k_ss + col);
         else
           for (int i = 0; i < 8; ++i) val[i] = 0;
+        // rotate write order by lane: without it all 16 lanes of a row-
+        // group hit ONE bank (16-way conflict, the PMC hotspot); rotated
+        // they spread over 8 banks (2-way).
+        const int rot = threadIdx.x & 7;
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
+        for (int j = 0; j < 8; ++j) {
+          int i = (j + rot) & 7;
           *reinterpret_cast<short*>(vt_lds + lds_off(col + i, row * 2, VT_RS)) = val[i];
         }
       }
@@ -136,17 +141,29 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
 
     // ---- mask + scale + online softmax ------------------------------------
     const int q_abs = q0 + wq * 16 + lg * 4;  // + r
+    const bool boundary = (kv0 + KB > Skv) || (CAUSAL && kv0 + KB > q_abs);
     float mx[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+    if (boundary) {
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) {
-      const int kv_abs = kv0 + nt * 16 + l16;
+      for (int nt = 0; nt < 4; ++nt) {
+        const int kv_abs = kv0 + nt * 16 + l16;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float val = sacc[nt][r] * scale;
-        if (kv_abs >= Skv || (CAUSAL && kv_abs > q_abs + r)) val = -INFINITY;
-        sacc[nt][r] = val;
-        mx[r] = fmaxf(mx[r], val);
+        for (int r = 0; r < 4; ++r) {
+          float val = sacc[nt][r] * scale;
+          if (kv_abs >= Skv || (CAUSAL && kv_abs > q_abs + r)) val = -INFINITY;
+          sacc[nt][r] = val;
+          mx[r] = fmaxf(mx[r], val);
+        }
       }
+    } else {
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float val = sacc[nt][r] * scale;
+          sacc[nt][r] = val;
+          mx[r] = fmaxf(mx[r], val);
+        }
     }
     float corr[4], psum[4];
 #pragma unroll
@@ -326,8 +343,10 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
         }
         *reinterpret_cast<shortx8*>(q_lds + lds_off(row, col * 2, NAT_RS)) = qv;
         *reinterpret_cast<shortx8*>(do_lds + lds_off(row, col * 2, NAT_RS)) = dv;
+        const int rot = threadIdx.x & 7;
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
+        for (int j = 0; j < 8; ++j) {
+          int i = (j + rot) & 7;
           *reinterpret_cast<short*>(qt_lds + lds_off(col + i, row * 2, TR_RS)) = qv[i];
           *reinterpret_cast<short*>(dot_lds + lds_off(col + i, row * 2, TR_RS)) = dv[i];
         }
@@ -356,12 +375,20 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
         st = mfma_bf16(kf[kk], qfr, st);
       }
       const int q_abs = q0 + nt * 16 + l16;
+      const bool bnd = (q0 + QB > Sq) || (kv0 + KB > Skv) ||
+                       (CAUSAL && q0 < kv0 + KB);
+      if (bnd) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int kv_abs = kv_abs0 + r;
-        float p = __expf(st[r] * scale - lse_v[nt]);
-        if (q_abs >= Sq || kv_abs >= Skv || (CAUSAL && kv_abs > q_abs)) p = 0.f;
-        pt[nt][r] = p;
+        for (int r = 0; r < 4; ++r) {
+          int kv_abs = kv_abs0 + r;
+          float p = __expf(st[r] * scale - lse_v[nt]);
+          if (q_abs >= Sq || kv_abs >= Skv || (CAUSAL && kv_abs > q_abs)) p = 0.f;
+          pt[nt][r] = p;
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          pt[nt][r] = __expf(st[r] * scale - lse_v[nt]);
       }
     }
 
@@ -516,9 +543,12 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
         }
         *reinterpret_cast<shortx8*>(k_lds + lds_off(row, col * 2, NAT_RS)) = kv_;
         *reinterpret_cast<shortx8*>(v_lds + lds_off(row, col * 2, NAT_RS)) = vv;
+        const int rot = threadIdx.x & 7;
 #pragma unroll
-        for (int i = 0; i < 8; ++i)
+        for (int j = 0; j < 8; ++j) {
+          int i = (j + rot) & 7;
           *reinterpret_cast<short*>(kt_lds + lds_off(col + i, row * 2, TR_RS)) = kv_[i];
+        }
       }
     }
     __syncthreads();
@@ -536,11 +566,20 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
         dp = mfma_bf16(dof[kk], vfr, dp);
       }
       const int kv_abs = kv0 + nt * 16 + l16;
+      const bool bnd = (kv0 + KB > Skv) || (CAUSAL && kv0 + KB > q_abs0);
+      if (bnd) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float p = __expf(s[r] * scale - lse_r[r]);
-        if (kv_abs >= Skv || (CAUSAL && kv_abs > q_abs0 + r)) p = 0.f;
-        ds[nt][r] = p * (dp[r] - delta_r[r]) * scale;
+        for (int r = 0; r < 4; ++r) {
+          float p = __expf(s[r] * scale - lse_r[r]);
+          if (kv_abs >= Skv || (CAUSAL && kv_abs > q_abs0 + r)) p = 0.f;
+          ds[nt][r] = p * (dp[r] - delta_r[r]) * scale;
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float p = __expf(s[r] * scale - lse_r[r]);
+          ds[nt][r] = p * (dp[r] - delta_r[r]) * scale;
+        }
       }
     }
 
