@@ -40,14 +40,10 @@ __device__ __forceinline__ float bf2f(short u) {
   return c.f;
 }
 
+// RNE via the hardware convert (v_cvt); bit-math emulation costs ~6 VALU ops
 __device__ __forceinline__ short f2bf(float f) {
-  union { float f; unsigned int i; } c;
-  c.f = f;
-  unsigned int lsb = (c.i >> 16) & 1u;
-  unsigned int rounded = c.i + 0x7fffu + lsb;  // round-to-nearest-even
-  // NaN guard
-  if ((c.i & 0x7f800000u) == 0x7f800000u && (c.i & 0x007fffffu)) rounded = c.i;
-  return (short)(rounded >> 16);
+  __hip_bfloat16 h = __float2bfloat16(f);
+  return *reinterpret_cast<short*>(&h);
 }
 
 // float -> bf16 for a vector of 8

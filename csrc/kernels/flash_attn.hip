@@ -93,9 +93,31 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
 
   const int kv_end = CAUSAL ? min(Skv, q0 + QB) : Skv;
 
+  // K staging source offsets (pre-swizzled global address so the async
+  // global_load_lds DMA lands the swizzled layout into linear LDS -- guide
+  // §5 m173 pattern; the read side XORs the same mask back out)
+  int k_row[KB * D / (256 * 8)], k_colp[KB * D / (256 * 8)];
+#pragma unroll
+  for (int it = 0; it < KB * D / (256 * 8); ++it) {
+    int flat = it * 256 * 8 + tid * 8;
+    int row = flat / D, col = flat % D;
+    k_row[it] = row;
+    k_colp[it] = col ^ ((row & 7) << 3);
+  }
+
   for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
     // ---- stage K tile [KB][D] (swizzled rows) -----------------------------
-    {
+    if (kv0 + KB <= Skv) {
+      // fast path: async DMA straight to LDS, no VGPR round-trip
+#pragma unroll
+      for (int it = 0; it < KB * D / (256 * 8); ++it) {
+        const short* src = kg + kbase + (long long)(kv0 + k_row[it]) * k_ss + k_colp[it];
+        char* dst = k_lds + it * 256 * 16 + (tid >> 6) * 64 * 16;
+        __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)src,
+                                         (__attribute__((address_space(3))) unsigned int*)dst,
+                                         16, 0, 0);
+      }
+    } else {
       constexpr int elems = KB * D;
       for (int flat = tid * 8; flat < elems; flat += 256 * 8) {
         int row = flat / D, col = flat % D;
@@ -106,6 +128,9 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
           for (int i = 0; i < 8; ++i) val[i] = 0;
         *reinterpret_cast<shortx8*>(k_lds + lds_off(row, col * 2, K_RS)) = val;
       }
+    }
+    {
+      constexpr int elems = KB * D;
       // ---- stage V^T tile [D][KB] ------------------------------------------
       for (int flat = tid * 8; flat < elems; flat += 256 * 8) {
         int row = flat / D, col = flat % D;  // row=kv, col=d
@@ -165,13 +190,21 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
           mx[r] = fmaxf(mx[r], val);
         }
     }
-    float corr[4], psum[4];
+    float psum[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       mx[r] = group16_reduce(mx[r], MaxOp());
-      float m_new = fmaxf(m_run[r], mx[r]);
-      corr[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_new);
-      m_run[r] = m_new;
+      // defer-max (guide T13): skip the O(NDT) rescale while the running
+      // max hasn't grown by more than THR (P stays bounded by e^THR)
+      const float THR = 8.f;
+      if (mx[r] > m_run[r] + THR || m_run[r] == -INFINITY) {
+        float m_new = fmaxf(m_run[r], mx[r]);
+        float corr = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_new);
+        l_run[r] *= corr;
+#pragma unroll
+        for (int dt = 0; dt < NDT; ++dt) oacc[dt][r] *= corr;
+        m_run[r] = m_new;
+      }
       psum[r] = 0.f;
     }
 #pragma unroll
@@ -186,9 +219,7 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       psum[r] = group16_reduce(psum[r], SumOp());
-      l_run[r] = l_run[r] * corr[r] + psum[r];
-#pragma unroll
-      for (int dt = 0; dt < NDT; ++dt) oacc[dt][r] *= corr[r];
+      l_run[r] += psum[r];
     }
 
     // ---- P -> per-wave LDS (bf16, swizzled 128B rows) ---------------------
