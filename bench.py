@@ -30,7 +30,15 @@ def main():
     ap.add_argument("--recompute", action="store_true")
     ap.add_argument("--cpu-smoke", action="store_true",
                     help="tiny CPU run for plumbing checks")
+    ap.add_argument("--tunableop", action="store_true",
+                    help="autotune hipBLASLt GEMM algo selection during warmup")
     args = ap.parse_args()
+
+    if args.tunableop:
+        import torch.cuda.tunable as tunable
+        tunable.enable(True)
+        tunable.tuning_enable(True)
+        tunable.set_max_tuning_duration(20)
 
     import paddle_amd as paddle
     from paddle_amd.distributed import fleet
@@ -80,6 +88,11 @@ def main():
 
     for _ in range(args.warmup):
         step()
+
+    if args.tunableop:
+        import torch.cuda.tunable as tunable
+        tunable.tuning_enable(False)
+        step()  # one settled step after tuning
 
     if world > 1:
         paddle.distributed.barrier()

@@ -37,7 +37,7 @@ __device__ __forceinline__ shortx8 lds_read8(const char* lds, unsigned row,
 // forward
 // ---------------------------------------------------------------------------
 template <int D, bool CAUSAL>
-__launch_bounds__(256)
+__launch_bounds__(512)
 __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restrict__ kg,
                               const short* __restrict__ vg, short* __restrict__ og,
                               float* __restrict__ lseg, int B, int H, int HKV,
@@ -45,7 +45,9 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
                               long long q_sb, long long q_sh, long long q_ss,
                               long long k_sb, long long k_sh, long long k_ss,
                               long long o_sb, long long o_sh, long long o_ss) {
-  constexpr int QB = 64, KB = 64;
+  constexpr int NW = 8;         // waves per block (16 q rows each)
+  constexpr int QB = 16 * NW, KB = 64;
+  constexpr int NT = NW * 64;   // threads
   constexpr int NKK = D / 32;   // mfma k-steps over head dim
   constexpr int NDT = D / 16;   // output d tiles
   // LDS layout
@@ -53,7 +55,7 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
   constexpr unsigned VT_RS = KB * 2;          // V^T tile row stride
   __shared__ char k_lds[KB * D * 2];
   __shared__ char vt_lds[D * KB * 2];
-  __shared__ char p_lds[4 * 16 * KB * 2];     // per-wave P tiles
+  __shared__ char p_lds[NW * 16 * KB * 2];    // per-wave P tiles
 
   const int qblk = blockIdx.x;
   const int bh = blockIdx.y;
@@ -96,10 +98,10 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
   // K staging source offsets (pre-swizzled global address so the async
   // global_load_lds DMA lands the swizzled layout into linear LDS -- guide
   // §5 m173 pattern; the read side XORs the same mask back out)
-  int k_row[KB * D / (256 * 8)], k_colp[KB * D / (256 * 8)];
+  int k_row[KB * D / (NT * 8)], k_colp[KB * D / (NT * 8)];
 #pragma unroll
-  for (int it = 0; it < KB * D / (256 * 8); ++it) {
-    int flat = it * 256 * 8 + tid * 8;
+  for (int it = 0; it < KB * D / (NT * 8); ++it) {
+    int flat = it * NT * 8 + tid * 8;
     int row = flat / D, col = flat % D;
     k_row[it] = row;
     k_colp[it] = col ^ ((row & 7) << 3);
@@ -110,16 +112,16 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
     if (kv0 + KB <= Skv) {
       // fast path: async DMA straight to LDS, no VGPR round-trip
 #pragma unroll
-      for (int it = 0; it < KB * D / (256 * 8); ++it) {
+      for (int it = 0; it < KB * D / (NT * 8); ++it) {
         const short* src = kg + kbase + (long long)(kv0 + k_row[it]) * k_ss + k_colp[it];
-        char* dst = k_lds + it * 256 * 16 + (tid >> 6) * 64 * 16;
+        char* dst = k_lds + it * NT * 16 + (tid >> 6) * 64 * 16;
         __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)src,
                                          (__attribute__((address_space(3))) unsigned int*)dst,
                                          16, 0, 0);
       }
     } else {
       constexpr int elems = KB * D;
-      for (int flat = tid * 8; flat < elems; flat += 256 * 8) {
+      for (int flat = tid * 8; flat < elems; flat += NT * 8) {
         int row = flat / D, col = flat % D;
         shortx8 val;
         if (kv0 + row < Skv)
@@ -132,7 +134,7 @@ __global__ void fa_fwd_kernel(const short* __restrict__ qg, const short* __restr
     {
       constexpr int elems = KB * D;
       // ---- stage V^T tile [D][KB] ------------------------------------------
-      for (int flat = tid * 8; flat < elems; flat += 256 * 8) {
+      for (int flat = tid * 8; flat < elems; flat += NT * 8) {
         int row = flat / D, col = flat % D;  // row=kv, col=d
         shortx8 val;
         if (kv0 + row < Skv)
@@ -297,7 +299,7 @@ __global__ void fa_bwd_delta_kernel(const short* __restrict__ dog, const short* 
 // iterate q tiles.  S^T computed as K·Q^T so kv is the C-row.
 // ---------------------------------------------------------------------------
 template <int D, bool CAUSAL>
-__launch_bounds__(256)
+__launch_bounds__(512)
 __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __restrict__ qg,
                                   const short* __restrict__ kg, const short* __restrict__ vg,
                                   const float* __restrict__ lseg, const float* __restrict__ deltag,
@@ -307,7 +309,10 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
                                   long long k_sb, long long k_sh, long long k_ss,
                                   long long do_sb, long long do_sh, long long do_ss,
                                   long long dk_sb, long long dk_sh, long long dk_ss) {
-  constexpr int KB = 64, QB = 64;
+  constexpr int NW = 8;               // waves; block owns NW*16 kv rows
+  constexpr int NT = NW * 64;
+  constexpr int KVEXT = NW * 16;
+  constexpr int QB = 64;
   constexpr int NKK = D / 32;
   constexpr int NDT = D / 16;
   constexpr unsigned NAT_RS = D * 2;   // natural [64][D]
@@ -316,11 +321,11 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
   __shared__ char qt_lds[D * QB * 2];
   __shared__ char do_lds[QB * D * 2];
   __shared__ char dot_lds[D * QB * 2];
-  __shared__ char ps_lds[4 * 16 * QB * 2];
+  __shared__ char ps_lds[NW * 16 * QB * 2];
 
   const int kvblk = blockIdx.x;
   const int bh = blockIdx.y;
-  const int kv0 = kvblk * KB;
+  const int kv0 = kvblk * KVEXT;
   const int b = bh / H, h = bh % H;
   const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
   const long long dobase = (long long)b * do_sb + (long long)h * do_sh;
@@ -363,7 +368,7 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
     // stage Q, dO natural + transposed
     {
       constexpr int elems = QB * D;
-      for (int flat = tid * 8; flat < elems; flat += 256 * 8) {
+      for (int flat = tid * 8; flat < elems; flat += NT * 8) {
         int row = flat / D, col = flat % D;
         shortx8 qv, dv;
         if (q0 + row < Sq) {
@@ -406,8 +411,8 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
         st = mfma_bf16(kf[kk], qfr, st);
       }
       const int q_abs = q0 + nt * 16 + l16;
-      const bool bnd = (q0 + QB > Sq) || (kv0 + KB > Skv) ||
-                       (CAUSAL && q0 < kv0 + KB);
+      const bool bnd = (q0 + QB > Sq) || (kv0 + KVEXT > Skv) ||
+                       (CAUSAL && q0 < kv0 + KVEXT);
       if (bnd) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -497,7 +502,7 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
 // backward dQ: block owns a 64-row Q tile; iterate kv tiles.
 // ---------------------------------------------------------------------------
 template <int D, bool CAUSAL>
-__launch_bounds__(256)
+__launch_bounds__(512)
 __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __restrict__ qg,
                                  const short* __restrict__ kg, const short* __restrict__ vg,
                                  const float* __restrict__ lseg, const float* __restrict__ deltag,
@@ -507,7 +512,10 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
                                  long long k_sb, long long k_sh, long long k_ss,
                                  long long do_sb, long long do_sh, long long do_ss,
                                  long long dq_sb, long long dq_sh, long long dq_ss) {
-  constexpr int QB = 64, KB = 64;
+  constexpr int NW = 8;               // waves; block owns NW*16 q rows
+  constexpr int NT = NW * 64;
+  constexpr int QEXT = NW * 16;
+  constexpr int KB = 64;
   constexpr int NKK = D / 32;
   constexpr int NDT = D / 16;
   constexpr unsigned NAT_RS = D * 2;
@@ -515,12 +523,12 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
   __shared__ char k_lds[KB * D * 2];
   __shared__ char v_lds[KB * D * 2];
   __shared__ char kt_lds[D * KB * 2];
-  __shared__ char ps_lds[4 * 16 * KB * 2];
+  __shared__ char ps_lds[NW * 16 * KB * 2];
 
   const int qblk = blockIdx.x;
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
-  const int q0 = qblk * QB;
+  const int q0 = qblk * QEXT;
   const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
   const long long dobase = (long long)b * do_sb + (long long)h * do_sh;
   const long long dqbase = (long long)b * dq_sb + (long long)h * dq_sh;
@@ -559,11 +567,11 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
 #pragma unroll
   for (int dt = 0; dt < NDT; ++dt) dq_acc[dt] = {0.f, 0.f, 0.f, 0.f};
 
-  const int kv_end = CAUSAL ? min(Skv, q0 + QB) : Skv;
+  const int kv_end = CAUSAL ? min(Skv, q0 + QEXT) : Skv;
   for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
     {
       constexpr int elems = KB * D;
-      for (int flat = tid * 8; flat < elems; flat += 256 * 8) {
+      for (int flat = tid * 8; flat < elems; flat += NT * 8) {
         int row = flat / D, col = flat % D;
         shortx8 kv_, vv;
         if (kv0 + row < Skv) {
@@ -653,8 +661,8 @@ void flash_attn_fwd(const void* q, const void* k, const void* v, void* o,
                     int64_t skv, int64_t dh, float scale, bool causal,
                     const int64_t* qs, const int64_t* ks, const int64_t* os,
                     hipStream_t s) {
-  dim3 grid((unsigned)cdiv((int)sq, 64), (unsigned)(b * h));
-  dim3 blk(256);
+  dim3 grid((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
+  dim3 blk(512);
 #define FAF(D, C)                                                             \
   hipLaunchKernelGGL((fa_fwd_kernel<D, C>), grid, blk, 0, s, (const short*)q, \
                      (const short*)k, (const short*)v, (short*)o, lse,        \
@@ -685,17 +693,17 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
                        (const short*)dout, (const short*)o, delta, (int)h,
                        (int)sq, rows, dos[0], dos[1], dos[2], os[0], os[1], os[2]);
 
-  dim3 gkv((unsigned)cdiv((int)skv, 64), (unsigned)(b * h));
-  dim3 gq((unsigned)cdiv((int)sq, 64), (unsigned)(b * h));
+  dim3 gkv((unsigned)cdiv((int)skv, 128), (unsigned)(b * h));
+  dim3 gq((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
 #define FAB(D, C)                                                              \
   do {                                                                         \
-    hipLaunchKernelGGL((fa_bwd_dkv_kernel<D, C>), gkv, dim3(256), 0, s,        \
+    hipLaunchKernelGGL((fa_bwd_dkv_kernel<D, C>), gkv, dim3(512), 0, s,        \
                        (const short*)dout, (const short*)q, (const short*)k,   \
                        (const short*)v, lse, delta, (short*)dk, (short*)dv,    \
                        (int)b, (int)h, (int)sq, (int)skv, scale,               \
                        qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],               \
                        dos[0], dos[1], dos[2], dks[0], dks[1], dks[2]);        \
-    hipLaunchKernelGGL((fa_bwd_dq_kernel<D, C>), gq, dim3(256), 0, s,          \
+    hipLaunchKernelGGL((fa_bwd_dq_kernel<D, C>), gq, dim3(512), 0, s,          \
                        (const short*)dout, (const short*)q, (const short*)k,   \
                        (const short*)v, lse, delta, (short*)dq, (int)b,        \
                        (int)h, (int)sq, (int)skv, scale,                       \
