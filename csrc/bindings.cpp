@@ -12,6 +12,7 @@
 
 namespace pa {
 void mfma_probe(const void* a, const void* bt, float* c, hipStream_t s);
+void mfma_probe32(const void* a, const void* bt, float* c, hipStream_t s);
 }
 
 namespace {
@@ -251,10 +252,10 @@ std::vector<Tensor> flash_attn_fwd(const Tensor& q, const Tensor& k, const Tenso
   auto o = o_out.has_value() ? *o_out : torch::empty({b, h, sq, d}, q.options());
   fa_strides(o, os);
   auto lse = torch::empty({b, h, sq}, q.options().dtype(torch::kFloat));
-  pa::flash_attn_fwd(q.const_data_ptr(), k.const_data_ptr(), v.const_data_ptr(),
-                     o.mutable_data_ptr(), lse.mutable_data_ptr<float>(), b, h,
-                     hkv, sq, skv, d, (float)scale, causal, qs, ks, os,
-                     cur_stream());
+  pa::flash_attn_fwd32(q.const_data_ptr(), k.const_data_ptr(), v.const_data_ptr(),
+                       o.mutable_data_ptr(), lse.mutable_data_ptr<float>(), b, h,
+                       hkv, sq, skv, d, (float)scale, causal, qs, ks, os,
+                       cur_stream());
   return {o, lse};
 }
 
@@ -353,6 +354,14 @@ Tensor mfma_probe(const Tensor& a, const Tensor& bt) {
   return c;
 }
 
+Tensor mfma_probe32(const Tensor& a, const Tensor& bt) {
+  CHECK_IN(a); CHECK_IN(bt);
+  auto c = torch::empty({32, 32}, a.options().dtype(torch::kFloat));
+  pa::mfma_probe32(a.const_data_ptr(), bt.const_data_ptr(),
+                   c.mutable_data_ptr<float>(), cur_stream());
+  return c;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -377,5 +386,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);
   m.def("mfma_probe", &mfma_probe);
+  m.def("mfma_probe32", &mfma_probe32);
   m.attr("compiled_arch") = "gfx950";
 }

@@ -1,0 +1,291 @@
+// FlashAttention-2 forward, 32x32-MFMA variant (gfx950).
+//
+// Design (guide §B "fused attention prefill" 8-warp ladder, re-derived for
+// our wave-per-32-q-rows layout):
+//   * mfma_f32_32x32x16_bf16; each wave owns 32 q rows; 4 waves/block
+//     -> 128-row q blocks; KV tiles of 64.
+//   * SWAPPED QK^T: S^T = mfma(K, Q) puts a full q-COLUMN of scores in
+//     each lane (C layout col = lane&31 = q), so the online softmax is
+//     per-lane scalar + ONE shfl_xor(32) -- no 16-lane reductions.
+//   * P stays in registers: the PV A-fragments are assembled with packed
+//     bf16 pairs + 2x permlane32_swap per k-step (guide T12) -- no P LDS
+//     round-trip, no lgkmcnt(0) stall.
+//   * K staged via async global_load_lds with pre-swizzled source (m173);
+//     V^T staged with bank-rotated scalar writes.
+//   * defer-max rescale (T13), boundary-only masking.
+// C/D layout for 32x32 (measured, guide §3): col=lane&31,
+// row=(reg&3)+8*(reg>>2)+4*(lane>>5).
+#include "common.h"
+#include "api.h"
+
+namespace pa {
+
+typedef __attribute__((ext_vector_type(16))) float floatx16;
+typedef __attribute__((ext_vector_type(2))) int intx2;
+
+__device__ __forceinline__ floatx16 mfma32_bf16(shortx8 a, shortx8 b, floatx16 c) {
+  return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+
+__device__ __forceinline__ unsigned lds_off32(unsigned row, unsigned col_bytes,
+                                              unsigned row_stride) {
+  return row * row_stride + (col_bytes ^ ((row & 7u) << 4));
+}
+
+__device__ __forceinline__ int pack_bf2(float a, float b) {
+  unsigned lo = (unsigned short)f2bf(a);
+  unsigned hi = (unsigned short)f2bf(b);
+  return (int)(lo | (hi << 16));
+}
+
+template <int D, bool CAUSAL>
+__launch_bounds__(256)
+__global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __restrict__ kg,
+                                const short* __restrict__ vg, short* __restrict__ og,
+                                float* __restrict__ lseg, int B, int H, int HKV,
+                                int Sq, int Skv, float scale,
+                                long long q_sb, long long q_sh, long long q_ss,
+                                long long k_sb, long long k_sh, long long k_ss,
+                                long long o_sb, long long o_sh, long long o_ss) {
+  constexpr int NW = 4;            // waves per block, 32 q rows each
+  constexpr int NT = NW * 64;
+  constexpr int QB = NW * 32;      // 128 q rows per block
+  constexpr int KB = 64;           // kv tile
+  constexpr int NKS = D / 16;      // q/k k-steps over head dim (mfma K=16)
+  constexpr int NDT = D / 32;      // output d tiles (32 wide)
+  constexpr unsigned K_RS = D * 2;
+  constexpr unsigned VT_RS = KB * 2;
+  __shared__ char k_lds[KB * D * 2];
+  __shared__ char vt_lds[D * KB * 2];
+
+  const int qblk = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int hkv = h / (H / HKV);
+  const int q0 = qblk * QB;
+  const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
+  const long long kbase = (long long)b * k_sb + (long long)hkv * k_sh;
+  const long long obase = (long long)b * o_sb + (long long)h * o_sh;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wq = tid >> 6;
+  const int l32 = lane & 31;       // q column (C layout)
+  const int hi = lane >> 5;        // half-wave id
+  const int q0w = q0 + wq * 32;
+
+  // Q fragments (B operand of the swapped QK^T): lane holds
+  // Q[q0w + l32][16*ks + 8*hi .. +7]
+  shortx8 qf[NKS];
+  {
+    int row = q0w + l32;
+    bool ok = row < Sq;
+#pragma unroll
+    for (int ks = 0; ks < NKS; ++ks) {
+      if (ok)
+        qf[ks] = *reinterpret_cast<const shortx8*>(
+            qg + qbase + (long long)row * q_ss + ks * 16 + hi * 8);
+      else
+        for (int i = 0; i < 8; ++i) qf[ks][i] = 0;
+    }
+  }
+
+  floatx16 oacc[NDT];
+#pragma unroll
+  for (int dt = 0; dt < NDT; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) oacc[dt][r] = 0.f;
+  float m_run = -INFINITY, l_run = 0.f;   // stats for q column l32
+
+  const int kv_end = CAUSAL ? min(Skv, q0 + QB) : Skv;
+
+  // K-staging source offsets (pre-swizzled so async DMA lands swizzled rows)
+  int k_row[KB * D / (NT * 8)], k_colp[KB * D / (NT * 8)];
+#pragma unroll
+  for (int it = 0; it < KB * D / (NT * 8); ++it) {
+    int flat = it * NT * 8 + tid * 8;
+    int row = flat / D, col = flat % D;
+    k_row[it] = row;
+    k_colp[it] = col ^ ((row & 7) << 3);
+  }
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+    // ---- stage K [KB][D] (async, swizzled) + V^T [D][KB] ------------------
+    if (kv0 + KB <= Skv) {
+#pragma unroll
+      for (int it = 0; it < KB * D / (NT * 8); ++it) {
+        const short* src = kg + kbase + (long long)(kv0 + k_row[it]) * k_ss + k_colp[it];
+        char* dst = k_lds + it * NT * 16 + (tid >> 6) * 64 * 16;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)src,
+            (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
+      }
+    } else {
+      for (int flat = tid * 8; flat < KB * D; flat += NT * 8) {
+        int row = flat / D, col = flat % D;
+        shortx8 val;
+        if (kv0 + row < Skv)
+          val = *reinterpret_cast<const shortx8*>(kg + kbase + (long long)(kv0 + row) * k_ss + col);
+        else
+          for (int i = 0; i < 8; ++i) val[i] = 0;
+        *reinterpret_cast<shortx8*>(k_lds + lds_off32(row, col * 2, K_RS)) = val;
+      }
+    }
+    {
+      const int rot = tid & 7;
+      for (int flat = tid * 8; flat < KB * D; flat += NT * 8) {
+        int row = flat / D, col = flat % D;  // row=kv, col=d
+        shortx8 val;
+        if (kv0 + row < Skv)
+          val = *reinterpret_cast<const shortx8*>(vg + kbase + (long long)(kv0 + row) * k_ss + col);
+        else
+          for (int i = 0; i < 8; ++i) val[i] = 0;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int i = (j + rot) & 7;
+          *reinterpret_cast<short*>(vt_lds + lds_off32(col + i, row * 2, VT_RS)) = val[i];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- S^T = K Q^T : 2 kv tiles of 32 -----------------------------------
+    floatx16 st[2];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) st[t][r] = 0.f;
+#pragma unroll
+      for (int ks = 0; ks < NKS; ++ks) {
+        shortx8 kf = *reinterpret_cast<const shortx8*>(
+            k_lds + lds_off32(t * 32 + l32, (ks * 16 + hi * 8) * 2, K_RS));
+        st[t] = mfma32_bf16(kf, qf[ks], st[t]);
+      }
+    }
+
+    // ---- mask + scale + per-lane online softmax ---------------------------
+    const int q_abs = q0w + l32;
+    float mx = -INFINITY;
+    const bool boundary = (kv0 + KB > Skv) || (CAUSAL && kv0 + KB > q_abs);
+    if (boundary) {
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kv_abs = kv0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float v = st[t][r] * scale;
+          if (kv_abs >= Skv || (CAUSAL && kv_abs > q_abs)) v = -INFINITY;
+          st[t][r] = v;
+          mx = fmaxf(mx, v);
+        }
+    } else {
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float v = st[t][r] * scale;
+          st[t][r] = v;
+          mx = fmaxf(mx, v);
+        }
+    }
+    mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+    const float THR = 8.f;
+    // per-lane stats are divergent; compute corr for every lane so the
+    // broadcast shfl below is wave-uniform-safe, and vote before paying
+    // for the O rescale
+    const bool grow = (mx > m_run + THR) || (m_run == -INFINITY);
+    float m_new = grow ? fmaxf(m_run, mx) : m_run;
+    float corr = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+    if (__any(grow && m_run != -INFINITY)) {
+      l_run *= corr;
+#pragma unroll
+      for (int dt = 0; dt < NDT; ++dt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int rq = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float c = __shfl(corr, rq, 64);
+          oacc[dt][r] *= c;
+        }
+    }
+    m_run = m_new;
+    float psum = 0.f;
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float p = (st[t][r] == -INFINITY) ? 0.f : __expf(st[t][r] - m_run);
+        st[t][r] = p;
+        psum += p;
+      }
+    psum += __shfl_xor(psum, 32, 64);
+    l_run += psum;
+
+    // ---- O += P V : assemble P A-frags in-register ------------------------
+#pragma unroll
+    for (int ks = 0; ks < KB / 16; ++ks) {
+      const int t = ks >> 1;
+      const int kp = ks & 1;
+      const int uo = 2 * kp + hi;        // own reg-group
+      const int us = 2 * kp + 1 - hi;    // supply reg-group (partner's hi)
+      int O1 = pack_bf2(st[t][4 * uo], st[t][4 * uo + 1]);
+      int O2 = pack_bf2(st[t][4 * uo + 2], st[t][4 * uo + 3]);
+      int S1 = pack_bf2(st[t][4 * us], st[t][4 * us + 1]);
+      int S2 = pack_bf2(st[t][4 * us + 2], st[t][4 * us + 3]);
+      intx2 ra = __builtin_amdgcn_permlane32_swap(S1, S2, false, false);
+      intx2 rb = __builtin_amdgcn_permlane32_swap(S2, S1, false, false);
+      // lanes<32: partner supply = (ra[1], rb[1]); lanes>=32: (rb[0], ra[0])
+      int P1 = hi ? rb[0] : ra[1];
+      int P2 = hi ? ra[0] : rb[1];
+      // A-frag words in kv order: hi==0 -> {own, partner}; hi==1 -> {partner, own}
+      intx4 paw;
+      paw[0] = hi ? P1 : O1;
+      paw[1] = hi ? P2 : O2;
+      paw[2] = hi ? O1 : P1;
+      paw[3] = hi ? O2 : P2;
+      shortx8 pa = *reinterpret_cast<shortx8*>(&paw);
+#pragma unroll
+      for (int dt = 0; dt < NDT; ++dt) {
+        shortx8 vf = *reinterpret_cast<const shortx8*>(
+            vt_lds + lds_off32(dt * 32 + l32, (ks * 16 + hi * 8) * 2, VT_RS));
+        oacc[dt] = mfma32_bf16(pa, vf, oacc[dt]);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: O /= l, store O + LSE ------------------------------------
+  float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int rq = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    float il = __shfl(inv_l, rq, 64);
+    int row = q0w + rq;
+    if (row >= Sq) continue;
+#pragma unroll
+    for (int dt = 0; dt < NDT; ++dt)
+      og[obase + (long long)row * o_ss + dt * 32 + l32] = f2bf(oacc[dt][r] * il);
+  }
+  if (hi == 0 && q0w + l32 < Sq)
+    lseg[(long long)(b * H + h) * Sq + q0w + l32] =
+        (l_run > 0.f) ? m_run + __logf(l_run) : -INFINITY;
+}
+
+void flash_attn_fwd32(const void* q, const void* k, const void* v, void* o,
+                      float* lse, int64_t b, int64_t h, int64_t hkv, int64_t sq,
+                      int64_t skv, int64_t dh, float scale, bool causal,
+                      const int64_t* qs, const int64_t* ks, const int64_t* os,
+                      hipStream_t s) {
+  dim3 grid((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
+  dim3 blk(256);
+#define FAF(D, C)                                                               \
+  hipLaunchKernelGGL((fa_fwd32_kernel<D, C>), grid, blk, 0, s, (const short*)q, \
+                     (const short*)k, (const short*)v, (short*)o, lse,          \
+                     (int)b, (int)h, (int)hkv, (int)sq, (int)skv, scale,        \
+                     qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],                  \
+                     os[0], os[1], os[2])
+  if (dh == 128) { if (causal) FAF(128, true); else FAF(128, false); }
+  else           { if (causal) FAF(64, true);  else FAF(64, false); }
+#undef FAF
+}
+
+}  // namespace pa

@@ -27,3 +27,30 @@ void mfma_probe(const void* a, const void* bt, float* c, hipStream_t s) {
 }
 
 }  // namespace pa
+
+namespace pa {
+typedef __attribute__((ext_vector_type(16))) float floatx16;
+
+__global__ void mfma_probe32_kernel(const short* __restrict__ a, const short* __restrict__ b,
+                                    float* __restrict__ c) {
+  // a: [32][16] row-major bf16 ; b_t: [32][16] row-major (B^T, B is [16][32])
+  // c: [32][32] row-major fp32 ; one wave
+  int lane = threadIdx.x & 63;
+  int l32 = lane & 31, hi = lane >> 5;
+  shortx8 af = *reinterpret_cast<const shortx8*>(a + l32 * 16 + hi * 8);
+  shortx8 bf = *reinterpret_cast<const shortx8*>(b + l32 * 16 + hi * 8);
+  floatx16 acc;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) acc[r] = 0.f;
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    c[((r & 3) + 8 * (r >> 2) + 4 * hi) * 32 + l32] = acc[r];
+}
+
+void mfma_probe32(const void* a, const void* bt, float* c, hipStream_t s);
+void mfma_probe32(const void* a, const void* bt, float* c, hipStream_t s) {
+  hipLaunchKernelGGL(mfma_probe32_kernel, dim3(1), dim3(64), 0, s,
+                     (const short*)a, (const short*)bt, c);
+}
+}  // namespace pa
