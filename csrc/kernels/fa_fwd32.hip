@@ -39,7 +39,7 @@ __device__ __forceinline__ int pack_bf2(float a, float b) {
 }
 
 template <int D, bool CAUSAL>
-__launch_bounds__(256)
+__launch_bounds__(256, 2)
 __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __restrict__ kg,
                                 const short* __restrict__ vg, short* __restrict__ og,
                                 float* __restrict__ lseg, int B, int H, int HKV,
