@@ -55,8 +55,9 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
   constexpr int NDT = D / 32;      // output d tiles (32 wide)
   constexpr unsigned K_RS = D * 2;
   constexpr unsigned VT_RS = KB * 2;
-  __shared__ char k_lds[KB * D * 2];
-  __shared__ char vt_lds[D * KB * 2];
+  // double-buffered: stage tile t+1 while the MFMAs chew tile t
+  __shared__ char k_lds[2][KB * D * 2];
+  __shared__ char vt_lds[2][D * KB * 2];
 
   const int qblk = blockIdx.x;
   const int bh = blockIdx.y;
@@ -109,13 +110,13 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
     k_colp[it] = col ^ ((row & 7) << 3);
   }
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+  auto stage = [&](int buf, int kv0) {
     // ---- stage K [KB][D] (async, swizzled) + V^T [D][KB] ------------------
     if (kv0 + KB <= Skv) {
 #pragma unroll
       for (int it = 0; it < KB * D / (NT * 8); ++it) {
         const short* src = kg + kbase + (long long)(kv0 + k_row[it]) * k_ss + k_colp[it];
-        char* dst = k_lds + it * NT * 16 + (tid >> 6) * 64 * 16;
+        char* dst = k_lds[buf] + it * NT * 16 + (tid >> 6) * 64 * 16;
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) unsigned int*)src,
             (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
@@ -128,7 +129,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
           val = *reinterpret_cast<const shortx8*>(kg + kbase + (long long)(kv0 + row) * k_ss + col);
         else
           for (int i = 0; i < 8; ++i) val[i] = 0;
-        *reinterpret_cast<shortx8*>(k_lds + lds_off32(row, col * 2, K_RS)) = val;
+        *reinterpret_cast<shortx8*>(k_lds[buf] + lds_off32(row, col * 2, K_RS)) = val;
       }
     }
     {
@@ -143,11 +144,17 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int i = (j + rot) & 7;
-          *reinterpret_cast<short*>(vt_lds + lds_off32(col + i, row * 2, VT_RS)) = val[i];
+          *reinterpret_cast<short*>(vt_lds[buf] + lds_off32(col + i, row * 2, VT_RS)) = val[i];
         }
       }
     }
-    __syncthreads();
+  };
+
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+    if (kv0 + KB < kv_end) stage(cur ^ 1, kv0 + KB);
 
     // ---- S^T = K Q^T : 2 kv tiles of 32 -----------------------------------
     floatx16 st[2];
@@ -158,7 +165,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
 #pragma unroll
       for (int ks = 0; ks < NKS; ++ks) {
         shortx8 kf = *reinterpret_cast<const shortx8*>(
-            k_lds + lds_off32(t * 32 + l32, (ks * 16 + hi * 8) * 2, K_RS));
+            k_lds[cur] + lds_off32(t * 32 + l32, (ks * 16 + hi * 8) * 2, K_RS));
         st[t] = mfma32_bf16(kf, qf[ks], st[t]);
       }
     }
@@ -246,11 +253,12 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
 #pragma unroll
       for (int dt = 0; dt < NDT; ++dt) {
         shortx8 vf = *reinterpret_cast<const shortx8*>(
-            vt_lds + lds_off32(dt * 32 + l32, (ks * 16 + hi * 8) * 2, VT_RS));
+            vt_lds[cur] + lds_off32(dt * 32 + l32, (ks * 16 + hi * 8) * 2, VT_RS));
         oacc[dt] = mfma32_bf16(pa, vf, oacc[dt]);
       }
     }
     __syncthreads();
+    cur ^= 1;
   }
 
   // ---- epilogue: O /= l, store O + LSE ------------------------------------
