@@ -24,7 +24,7 @@ def main():
     ap.add_argument("--steps", type=int, default=6)
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--model", type=str, default="gpt3-6.7b")
-    ap.add_argument("--batch", type=int, default=2, help="per-GPU micro batch")
+    ap.add_argument("--batch", type=int, default=8, help="per-GPU micro batch")
     ap.add_argument("--seq", type=int, default=2048)
     ap.add_argument("--sharding-stage", type=int, default=3)
     ap.add_argument("--recompute", action="store_true")
