@@ -1,0 +1,195 @@
+"""Mixture-of-Experts GPT with expert parallelism over xGMI all-to-all.
+
+Reference parity: python/paddle/incubate/distributed/models/moe/
+moe_layer.py:263 (MoELayer; MoEScatter/MoEGather PyLayers over
+global_scatter/global_gather) and gate/{gshard,switch}_gate.py.
+
+MI355X design: EP all-to-all is the best case on the 8-GPU xGMI full
+mesh (single hop, all 7 links concurrent -- SURVEY.md §5); dispatch
+uses a single alltoall_single on a packed [capacity-padded] buffer.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+
+from .. import nn
+from ..distributed import collective as C
+from ..nn.initializer import Normal, _apply_initializer
+from ..ops import functional as hot
+from .gpt import GPTAttention, GPTConfig, GPTEmbeddings
+
+
+class _AllToAll(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        if group is None or group.nranks == 1:
+            return x
+        out = torch.empty_like(x)
+        C.alltoall_single(x.contiguous(), out, group=group)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        if ctx.group is None or ctx.group.nranks == 1:
+            return dy, None
+        out = torch.empty_like(dy)
+        C.alltoall_single(dy.contiguous(), out, group=ctx.group)
+        return out, None
+
+
+class TopKGate(nn.Layer):
+    """Switch (k=1) / GShard (k=2) style gate with capacity + aux loss."""
+
+    def __init__(self, hidden_size, num_experts, k=2, capacity_factor=1.25):
+        super().__init__()
+        self.num_experts = num_experts
+        self.k = k
+        self.capacity_factor = capacity_factor
+        self.wg = nn.Linear(hidden_size, num_experts, bias_attr=False)
+        _apply_initializer(Normal(0.0, 0.02), self.wg.weight)
+
+    def forward(self, x):
+        # x: [tokens, h]
+        logits = self.wg(x.float())
+        probs = torch.softmax(logits, -1)
+        topv, topi = probs.topk(self.k, dim=-1)            # [T, k]
+        # aux load-balance loss (gshard): num_experts * sum(me * ce)
+        me = probs.mean(0)
+        ce = torch.zeros_like(me).scatter_add_(
+            0, topi[:, 0], torch.ones_like(topi[:, 0], dtype=me.dtype))
+        ce = ce / x.shape[0]
+        aux = (me * ce).sum() * self.num_experts
+        return topv, topi, aux
+
+
+class ExpertMLP(nn.Layer):
+    def __init__(self, hidden_size, inter_size):
+        super().__init__()
+        self.fc1 = nn.Linear(hidden_size, inter_size)
+        self.fc2 = nn.Linear(inter_size, hidden_size)
+
+    def forward(self, x):
+        return self.fc2(hot.bias_gelu(torch.matmul(x, self.fc1.weight), self.fc1.bias))
+
+
+class MoELayer(nn.Layer):
+    """Token dispatch -> EP all-to-all -> local experts -> combine."""
+
+    def __init__(self, hidden_size, inter_size, num_experts, k=2,
+                 capacity_factor=2.0, ep_group=None):
+        super().__init__()
+        self.ep_group = ep_group
+        self.ep_size = ep_group.nranks if ep_group else 1
+        assert num_experts % self.ep_size == 0
+        self.num_experts = num_experts
+        self.local_experts = num_experts // self.ep_size
+        self.k = k
+        self.capacity_factor = capacity_factor
+        self.gate = TopKGate(hidden_size, num_experts, k, capacity_factor)
+        self.experts = nn.LayerList([ExpertMLP(hidden_size, inter_size)
+                                     for _ in range(self.local_experts)])
+        self.aux_loss = None
+
+    def forward(self, x):
+        b, s, h = x.shape
+        tokens = x.reshape(-1, h)
+        T = tokens.shape[0]
+        topv, topi, aux = self.gate(tokens)
+        self.aux_loss = aux
+        cap = int(self.capacity_factor * T * self.k / self.num_experts) + 1
+        cap = max(cap, 4)
+
+        # build per-expert capacity-padded dispatch buffer [E, cap, h]
+        dispatch = torch.zeros(self.num_experts, cap, h, dtype=x.dtype, device=x.device)
+        combine_w = torch.zeros(T, self.k, dtype=x.dtype, device=x.device)
+        slot_of = torch.full((T, self.k), -1, dtype=torch.long, device=x.device)
+        with torch.no_grad():
+            for kk in range(self.k):
+                e = topi[:, kk]
+                # position of each token within its expert queue
+                pos = torch.zeros_like(e)
+                counts = torch.zeros(self.num_experts, dtype=torch.long, device=x.device)
+                # stable per-expert positions via sort
+                order = torch.argsort(e, stable=True)
+                sorted_e = e[order]
+                ones = torch.ones_like(sorted_e)
+                seg_start = torch.searchsorted(sorted_e, torch.arange(
+                    self.num_experts, device=x.device))
+                idx_in_seg = torch.arange(T, device=x.device) - seg_start[sorted_e]
+                pos[order] = idx_in_seg
+                keep = pos < cap
+                slot = e * cap + pos
+                slot_of[:, kk] = torch.where(keep, slot, torch.full_like(slot, -1))
+        flat_dispatch = dispatch.reshape(-1, h)
+        for kk in range(self.k):
+            valid = slot_of[:, kk] >= 0
+            idx = slot_of[valid, kk]
+            flat_dispatch.index_copy_(0, idx, tokens[valid])
+            combine_w[:, kk] = torch.where(valid, topv[:, kk].to(x.dtype),
+                                           torch.zeros_like(combine_w[:, kk]))
+
+        # EP all-to-all: [E, cap, h] -> experts-local [ep, local_E, cap, h]
+        dd = _AllToAll.apply(dispatch.reshape(self.ep_size, self.local_experts, cap, h)
+                             .contiguous().view(-1, h), self.ep_group)
+        dd = dd.view(self.ep_size, self.local_experts, cap, h)
+
+        outs = []
+        for i, expert in enumerate(self.experts):
+            outs.append(expert(dd[:, i].reshape(-1, h)).view(self.ep_size, cap, h))
+        expert_out = torch.stack(outs, dim=1)  # [ep, local_E, cap, h]
+
+        back = _AllToAll.apply(expert_out.contiguous().view(-1, h), self.ep_group)
+        back = back.view(self.num_experts * cap, h)
+
+        out = torch.zeros_like(tokens)
+        for kk in range(self.k):
+            valid = slot_of[:, kk] >= 0
+            idx = slot_of[valid, kk]
+            out[valid] += back[idx] * combine_w[valid, kk].unsqueeze(-1)
+        return out.view(b, s, h)
+
+
+class MoEDecoderLayer(nn.Layer):
+    def __init__(self, cfg: GPTConfig, num_experts, k=2, ep_group=None):
+        super().__init__()
+        self.ln1 = nn.LayerNorm(cfg.hidden_size)
+        self.attn = GPTAttention(cfg)
+        self.ln2 = nn.LayerNorm(cfg.hidden_size)
+        self.moe = MoELayer(cfg.hidden_size, cfg.intermediate_size, num_experts,
+                            k=k, ep_group=ep_group)
+
+    def forward(self, x):
+        h = x + self.attn(self.ln1(x))
+        return h + self.moe(self.ln2(h))
+
+
+class GPTMoEForPretraining(nn.Layer):
+    def __init__(self, cfg: GPTConfig, num_experts=64, k=2, ep_group=None,
+                 aux_weight=0.01):
+        super().__init__()
+        self.cfg = cfg
+        self.aux_weight = aux_weight
+        self.embeddings = GPTEmbeddings(cfg)
+        self.layers = nn.LayerList([MoEDecoderLayer(cfg, num_experts, k, ep_group)
+                                    for _ in range(cfg.num_layers)])
+        self.final_norm = nn.LayerNorm(cfg.hidden_size)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias_attr=False)
+
+    def forward(self, input_ids):
+        x = self.embeddings(input_ids)
+        for l in self.layers:
+            x = l(x)
+        return self.lm_head(self.final_norm(x))
+
+    def aux_loss(self):
+        total = 0.0
+        for l in self.layers:
+            if l.moe.aux_loss is not None:
+                total = total + l.moe.aux_loss
+        return self.aux_weight * total
+
+    def sharding_units(self):
+        return [self.embeddings, *self.layers, self.final_norm, self.lm_head]

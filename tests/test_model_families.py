@@ -1,0 +1,102 @@
+"""Model-family CPU tests: Llama (rms/rope/swiglu/GQA), BERT (fused
+layers), GPT-MoE (gate/dispatch/combine)."""
+import pytest
+import torch
+
+import paddle_amd as paddle
+from paddle_amd.models import build_bert, build_llama
+from paddle_amd.models.gpt import GPTConfig
+from paddle_amd.models.llama import LlamaPretrainingCriterion
+from paddle_amd.models.moe import GPTMoEForPretraining, MoELayer
+
+
+def test_llama_tiny_trains():
+    paddle.seed(0)
+    m = build_llama("llama-tiny")
+    loss_fn = LlamaPretrainingCriterion()
+    opt = paddle.optimizer.AdamW(learning_rate=3e-4, parameters=m.parameters())
+    ids = paddle.randint(0, 1024, (2, 64))
+    losses = []
+    for _ in range(8):
+        loss = loss_fn(m(ids), ids)
+        loss.backward()
+        opt.step()
+        opt.clear_grad()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] - 0.3, losses
+
+
+def test_llama_gqa_heads():
+    m = build_llama("llama-tiny")
+    ids = paddle.randint(0, 1024, (1, 32))
+    out = m(ids)
+    assert out.shape == (1, 32, 1024)
+
+
+def test_bert_tiny_forward_and_mlm_loss():
+    paddle.seed(0)
+    m = build_bert("bert-tiny")
+    ids = paddle.randint(0, 1024, (2, 32))
+    mask = torch.ones(2, 32)
+    mlm, nsp = m(ids, attention_mask=mask)
+    assert mlm.shape == (2, 32, 1024)
+    assert nsp.shape == (2, 2)
+    loss = paddle.nn.functional.cross_entropy(mlm.reshape(-1, 1024), ids.reshape(-1))
+    loss.backward()
+    assert m.bert.encoder[0].attn.qkv_weight.grad is not None
+
+
+def test_moe_layer_single_process():
+    paddle.seed(0)
+    layer = MoELayer(hidden_size=64, inter_size=128, num_experts=4, k=2,
+                     ep_group=None)
+    x = paddle.randn([2, 16, 64])
+    x.requires_grad_(True)
+    out = layer(x)
+    assert out.shape == x.shape
+    (out.sum() + layer.aux_loss).backward()
+    assert x.grad is not None
+    assert layer.gate.wg.weight.grad is not None
+    for e in layer.experts:
+        assert e.fc1.weight.grad is not None
+
+
+def test_moe_gpt_trains():
+    paddle.seed(0)
+    cfg = GPTConfig(vocab_size=512, hidden_size=64, num_layers=2, num_heads=2,
+                    max_seq_len=64)
+    m = GPTMoEForPretraining(cfg, num_experts=4, k=2)
+    from paddle_amd.models import GPTPretrainingCriterion
+    loss_fn = GPTPretrainingCriterion()
+    opt = paddle.optimizer.AdamW(learning_rate=1e-3, parameters=m.parameters())
+    ids = paddle.randint(0, 512, (2, 32))
+    losses = []
+    for _ in range(6):
+        loss = loss_fn(m(ids), ids) + m.aux_loss()
+        loss.backward()
+        opt.step()
+        opt.clear_grad()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0], losses
+
+
+def test_moe_ep_two_ranks():
+    from dist_util import run_dist
+    run_dist("""
+        import torch
+        import paddle_amd as paddle
+        from paddle_amd.models.moe import MoELayer
+        paddle.distributed.init_parallel_env()
+        r = paddle.distributed.get_rank()
+        g = paddle.distributed.new_group([0, 1])
+        paddle.seed(0)  # same experts both ranks for the parity check
+        layer = MoELayer(hidden_size=32, inter_size=64, num_experts=4, k=1,
+                         capacity_factor=8.0, ep_group=g)
+        paddle.seed(100 + r)
+        x = paddle.randn([1, 8, 32]).requires_grad_(True)
+        out = layer(x)
+        assert out.shape == x.shape
+        out.sum().backward()
+        assert x.grad is not None
+        print("rank", r, "moe ep ok")
+    """)
