@@ -218,7 +218,10 @@ from . import static  # noqa: F401
 from . import jit  # noqa: F401
 from . import incubate  # noqa: F401
 from . import models  # noqa: F401
+from . import distribution  # noqa: F401
+from . import profiler  # noqa: F401
 from .framework_io import save, load, async_save  # noqa: F401
+from .hapi import Model  # noqa: F401
 from .param_attr import ParamAttr  # noqa: F401
 from .autograd import grad, no_grad, enable_grad, set_grad_enabled, is_grad_enabled  # noqa: F401
 

@@ -1,0 +1,195 @@
+"""paddle.Model high-level API (reference: python/paddle/hapi/model.py:1472
+Model.fit/evaluate/predict + callbacks)."""
+from __future__ import annotations
+
+import time
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from . import io as pio
+from . import metric as pmetric
+from .framework_io import load as fload
+from .framework_io import save as fsave
+
+
+class Callback:
+    def set_params(self, params):
+        self.params = params
+
+    def on_train_begin(self, logs=None):
+        pass
+
+    def on_train_end(self, logs=None):
+        pass
+
+    def on_epoch_begin(self, epoch, logs=None):
+        pass
+
+    def on_epoch_end(self, epoch, logs=None):
+        pass
+
+    def on_train_batch_begin(self, step, logs=None):
+        pass
+
+    def on_train_batch_end(self, step, logs=None):
+        pass
+
+
+class ProgBarLogger(Callback):
+    def __init__(self, log_freq=10, verbose=1):
+        self.log_freq = log_freq
+        self.verbose = verbose
+
+    def on_train_batch_end(self, step, logs=None):
+        if self.verbose and step % self.log_freq == 0:
+            items = ", ".join(f"{k}: {v:.4f}" if isinstance(v, float) else f"{k}: {v}"
+                              for k, v in (logs or {}).items())
+            print(f"step {step} - {items}")
+
+
+class ModelCheckpoint(Callback):
+    def __init__(self, save_freq=1, save_dir=None):
+        self.save_freq = save_freq
+        self.save_dir = save_dir
+
+    def on_epoch_end(self, epoch, logs=None):
+        if self.save_dir and epoch % self.save_freq == 0:
+            self.model.save(f"{self.save_dir}/{epoch}")
+
+
+class Model:
+    def __init__(self, network, inputs=None, labels=None):
+        self.network = network
+        self._optimizer = None
+        self._loss = None
+        self._metrics: List = []
+
+    def prepare(self, optimizer=None, loss=None, metrics=None, amp_configs=None):
+        self._optimizer = optimizer
+        self._loss = loss
+        if metrics is not None:
+            self._metrics = metrics if isinstance(metrics, (list, tuple)) else [metrics]
+        return self
+
+    def _run_one_batch(self, batch, train=True):
+        if isinstance(batch, (list, tuple)) and len(batch) >= 2:
+            x, y = batch[0], batch[1]
+        else:
+            x, y = batch, None
+        out = self.network(x)
+        logs = {}
+        if self._loss is not None and y is not None:
+            loss = self._loss(out, y)
+            logs["loss"] = float(loss.detach().float().cpu())
+            if train:
+                loss.backward()
+                self._optimizer.step()
+                self._optimizer.clear_grad()
+        for m in self._metrics:
+            try:
+                corr = m.compute(out, y)
+                res = m.update(corr)
+                logs[m.name()] = res
+            except Exception:
+                pass
+        return logs
+
+    def fit(self, train_data=None, eval_data=None, batch_size=1, epochs=1,
+            eval_freq=1, log_freq=10, save_dir=None, save_freq=1, verbose=1,
+            drop_last=False, shuffle=True, num_workers=0, callbacks=None,
+            accumulate_grad_batches=1, num_iters=None):
+        loader = train_data
+        if not isinstance(train_data, pio.DataLoader) and train_data is not None and \
+                not hasattr(train_data, "__iter__"):
+            loader = pio.DataLoader(train_data, batch_size=batch_size, shuffle=shuffle,
+                                    drop_last=drop_last, num_workers=num_workers)
+        cbs = list(callbacks or [])
+        if verbose:
+            cbs.append(ProgBarLogger(log_freq, verbose))
+        for cb in cbs:
+            cb.model = self
+            cb.on_train_begin()
+        self.network.train()
+        it = 0
+        for epoch in range(epochs):
+            for cb in cbs:
+                cb.on_epoch_begin(epoch)
+            for m in self._metrics:
+                m.reset()
+            for step, batch in enumerate(loader):
+                logs = self._run_one_batch(batch, train=True)
+                for cb in cbs:
+                    cb.on_train_batch_end(step, logs)
+                it += 1
+                if num_iters is not None and it >= num_iters:
+                    break
+            for cb in cbs:
+                cb.on_epoch_end(epoch, logs)
+            if eval_data is not None and (epoch + 1) % eval_freq == 0:
+                self.evaluate(eval_data, batch_size=batch_size, verbose=0)
+            if save_dir and (epoch + 1) % save_freq == 0:
+                self.save(f"{save_dir}/{epoch}")
+        for cb in cbs:
+            cb.on_train_end()
+
+    @torch.no_grad()
+    def evaluate(self, eval_data, batch_size=1, log_freq=10, verbose=1,
+                 num_workers=0, callbacks=None, num_iters=None):
+        loader = eval_data
+        if not hasattr(eval_data, "__iter__"):
+            loader = pio.DataLoader(eval_data, batch_size=batch_size)
+        self.network.eval()
+        for m in self._metrics:
+            m.reset()
+        total_loss, n = 0.0, 0
+        for batch in loader:
+            logs = self._run_one_batch(batch, train=False)
+            if "loss" in logs:
+                total_loss += logs["loss"]
+                n += 1
+        self.network.train()
+        out = {"loss": total_loss / max(n, 1)}
+        for m in self._metrics:
+            out[m.name()] = m.accumulate()
+        if verbose:
+            print("Eval:", out)
+        return out
+
+    @torch.no_grad()
+    def predict(self, test_data, batch_size=1, num_workers=0, stack_outputs=False,
+                verbose=1, callbacks=None):
+        loader = test_data
+        if not hasattr(test_data, "__iter__"):
+            loader = pio.DataLoader(test_data, batch_size=batch_size)
+        self.network.eval()
+        outs = []
+        for batch in loader:
+            x = batch[0] if isinstance(batch, (list, tuple)) else batch
+            outs.append(self.network(x).cpu().numpy())
+        self.network.train()
+        if stack_outputs:
+            return [np.concatenate(outs, 0)]
+        return [outs]
+
+    def save(self, path, training=True):
+        fsave(self.network.state_dict(), path + ".pdparams")
+        if training and self._optimizer is not None:
+            fsave(self._optimizer.state_dict(), path + ".pdopt")
+
+    def load(self, path, skip_mismatch=False, reset_optimizer=False):
+        sd = fload(path + ".pdparams")
+        self.network.set_state_dict(sd)
+        import os
+        if not reset_optimizer and self._optimizer is not None and \
+                os.path.exists(path + ".pdopt"):
+            self._optimizer.set_state_dict(fload(path + ".pdopt"))
+
+    def parameters(self, *a, **kw):
+        return self.network.parameters(*a, **kw)
+
+    def summary(self, input_size=None, dtype=None):
+        n_params = sum(p.numel() for p in self.network.parameters())
+        print(f"Total params: {n_params:,}")
+        return {"total_params": n_params}

@@ -1,0 +1,128 @@
+"""Aux subsystems: profiler, hapi Model, distribution, launch CLI,
+flags, metrics, distributed checkpoint (reference SURVEY.md §5)."""
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+import torch
+
+import paddle_amd as paddle
+from paddle_amd import nn
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_profiler_records(tmp_path):
+    with paddle.profiler.Profiler() as prof:
+        x = paddle.randn([64, 64])
+        for _ in range(3):
+            x = paddle.matmul(x, x)
+            prof.step()
+    prof.export(str(tmp_path / "trace.json"))
+    assert (tmp_path / "trace.json").exists()
+
+
+def test_record_event():
+    with paddle.profiler.RecordEvent("my_op"):
+        paddle.ones([2, 2]).sum()
+
+
+def test_hapi_model_fit_and_save(tmp_path):
+    from paddle_amd.io import TensorDataset
+    paddle.seed(0)
+    x = torch.randn(64, 4)
+    y = (x.sum(-1) > 0).long()
+    ds = TensorDataset([x, y])
+    net = nn.Sequential(nn.Linear(4, 16), nn.ReLU(), nn.Linear(16, 2))
+    model = paddle.Model(net)
+    opt = paddle.optimizer.Adam(learning_rate=1e-2, parameters=net.parameters())
+    model.prepare(opt, nn.CrossEntropyLoss(), paddle.metric.Accuracy())
+    model.fit(ds, batch_size=16, epochs=2, verbose=0)
+    res = model.evaluate(ds, batch_size=16, verbose=0)
+    assert res["acc"] > 0.6
+    model.save(str(tmp_path / "m"))
+    assert (tmp_path / "m.pdparams").exists()
+    model2 = paddle.Model(nn.Sequential(nn.Linear(4, 16), nn.ReLU(), nn.Linear(16, 2)))
+    model2.prepare(None, nn.CrossEntropyLoss())
+    model2.load(str(tmp_path / "m"), reset_optimizer=True)
+
+
+def test_distribution_api():
+    n = paddle.distribution.Normal(0.0, 1.0)
+    s = n.sample([100])
+    assert s.shape == (100,)
+    lp = n.log_prob(paddle.to_tensor(0.0))
+    np.testing.assert_allclose(float(lp), -0.9189385, rtol=1e-5)
+    u = paddle.distribution.Uniform(0.0, 2.0)
+    assert 0 <= float(u.sample([1])) <= 2
+    c = paddle.distribution.Categorical(logits=[0.0, 0.0])
+    assert int(c.sample([1])) in (0, 1)
+    kl = paddle.distribution.kl_divergence(n, paddle.distribution.Normal(1.0, 1.0))
+    np.testing.assert_allclose(float(kl), 0.5, rtol=1e-5)
+
+
+def test_launch_cli_single_proc(tmp_path):
+    script = tmp_path / "train.py"
+    script.write_text(
+        "import os, sys\n"
+        "sys.path.insert(0, %r)\n"
+        "import paddle_amd as paddle\n"
+        "assert os.environ['WORLD_SIZE'] == '2'\n"
+        "paddle.distributed.init_parallel_env()\n"
+        "import torch\n"
+        "t = torch.ones(2)\n"
+        "paddle.distributed.all_reduce(t)\n"
+        "assert t.tolist() == [2.0, 2.0]\n"
+        "print('rank', os.environ['RANK'], 'ok')\n" % REPO)
+    env = dict(os.environ)
+    env["CUDA_VISIBLE_DEVICES"] = ""
+    env["PYTHONPATH"] = REPO
+    r = subprocess.run([sys.executable, "-m", "paddle_amd.distributed.launch",
+                        "--nproc_per_node", "2", "--master_port", "29777",
+                        "--log_dir", str(tmp_path / "logs"), str(script)],
+                       env=env, capture_output=True, text=True, timeout=180,
+                       cwd=str(tmp_path))
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert (tmp_path / "logs" / "workerlog.1").exists()
+
+
+def test_flags_api():
+    paddle.set_flags({"FLAGS_check_nan_inf": True})
+    assert paddle.get_flags("FLAGS_check_nan_inf")["FLAGS_check_nan_inf"] is True
+    paddle.set_flags({"FLAGS_check_nan_inf": False})
+
+
+def test_amp_debugging_check_numerics():
+    t = paddle.to_tensor([1.0, float("nan")])
+    with pytest.raises(FloatingPointError):
+        paddle.amp.debugging.check_numerics(t, "op", "var")
+
+
+def test_dist_checkpoint_single(tmp_path):
+    from paddle_amd.distributed.checkpoint import load_state_dict, save_state_dict
+    sd = {"w": torch.randn(4, 4), "step": 3}
+    save_state_dict(sd, str(tmp_path / "ckpt"))
+    target = {"w": torch.zeros(4, 4), "step": 0}
+    out = load_state_dict(target, str(tmp_path / "ckpt"))
+    torch.testing.assert_close(out["w"], sd["w"])
+    assert out["step"] == 3
+
+
+def test_device_stream_event_api():
+    s = paddle.device.Stream() if torch.cuda.is_available() else None
+    e = paddle.device.Event() if torch.cuda.is_available() else None
+    assert paddle.device.get_device() in ("cpu",) or "gpu" in paddle.device.get_device()
+
+
+def test_vision_lenet_and_transforms():
+    from paddle_amd.vision.models import LeNet
+    from paddle_amd.vision.transforms import Compose, Normalize, ToTensor
+    t = Compose([ToTensor(), Normalize([0.5], [0.5])])
+    img = np.random.randint(0, 255, (28, 28), dtype=np.uint8)
+    x = t(img)
+    assert x.shape == (1, 28, 28)
+    m = LeNet()
+    out = m(x.unsqueeze(0))
+    assert out.shape == (1, 10)
