@@ -165,6 +165,26 @@ class _P2P:
         C.recv(g, src=self.hcg.get_p2p_next_rank(), group=self.group)
         return g
 
+    # combined pairs (deadlock-free: post isend+irecv, then wait both --
+    # the reference's batched _batched_p2p_ops, p2p_communication.py:327)
+    def send_forward_recv_backward(self, out, grad_like):
+        nxt = self.hcg.get_p2p_next_rank()
+        g = torch.zeros_like(grad_like)
+        w1 = C.isend(out.contiguous(), nxt, group=self.group)
+        w2 = C.irecv(g, nxt, group=self.group)
+        w1.wait()
+        w2.wait()
+        return g
+
+    def send_backward_recv_forward(self, grad_in, device):
+        prv = self.hcg.get_p2p_prev_rank()
+        x = torch.zeros(self.recv_shape, dtype=self.recv_dtype, device=device)
+        w1 = C.isend(grad_in.contiguous(), prv, group=self.group)
+        w2 = C.irecv(x, prv, group=self.group)
+        w1.wait()
+        w2.wait()
+        return x
+
 
 class PipelineParallel(torch.nn.Module):
     """1F1B schedule (pipeline_parallel.py:575)."""
@@ -195,6 +215,7 @@ class PipelineParallel(torch.nn.Module):
 
     def train_batch(self, data, optimizer, lr_scheduler=None, scaler=None):
         assert self.num_stages > 1
+        import collections
         micro = self._split_micro(data)
         n_micro = len(micro)
         warmup = min(self.num_stages - self.stage_id - 1, n_micro)
@@ -202,52 +223,67 @@ class PipelineParallel(torch.nn.Module):
         dev = (torch.device("cuda", torch.cuda.current_device())
                if torch.cuda.is_available() else torch.device("cpu"))
 
-        inputs: List = []
-        outputs: List = []
         losses = []
+        pending = collections.deque()  # (inp, out) awaiting backward, FIFO
         fwd_i = 0
-        bwd_i = 0
 
-        def fwd_step(i):
-            x_mb, y_mb = micro[i]
+        def get_input(i):
             if self.is_first:
-                inp = x_mb
-            else:
-                inp = self.p2p.recv_forward(dev)
-                inp.requires_grad_(True)
+                return micro[i][0]
+            inp = self.p2p.recv_forward(dev)
+            inp.requires_grad_(True)
+            return inp
+
+        def run_fwd(inp, i):
             out = self._layers(inp)
             if self.is_last:
-                loss = self._layers.loss_fn(out, y_mb) / n_micro
+                loss = self._layers.loss_fn(out, micro[i][1]) / n_micro
                 losses.append(loss.detach())
-                outputs.append(loss)
-            else:
-                outputs.append(out)
-                self.p2p.send_forward(out)
-            inputs.append(inp if not self.is_first else None)
+                out = loss
+            return out
 
-        def bwd_step(i):
-            out = outputs[i]
+        def run_bwd(binp, bout, grad_out):
             if self.is_last:
-                out.backward()
+                bout.backward()
             else:
-                g = self.p2p.recv_backward(out)
-                out.backward(gradient=g)
-            if not self.is_first:
-                self.p2p.send_backward(inputs[i].grad)
-            outputs[i] = None
-            inputs[i] = None
+                bout.backward(gradient=grad_out)
+            return binp.grad if (binp is not None and not self.is_first) else None
 
+        # warmup forwards (never on the last stage: its warmup == 0)
         for _ in range(warmup):
-            fwd_step(fwd_i)
+            inp = get_input(fwd_i)
+            out = run_fwd(inp, fwd_i)
+            self.p2p.send_forward(out)
+            pending.append((None if self.is_first else inp, out))
             fwd_i += 1
+        # steady 1F1B
+        inp = get_input(fwd_i) if steady > 0 else None
         for k in range(steady):
-            fwd_step(fwd_i)
+            out = run_fwd(inp, fwd_i)
             fwd_i += 1
-            bwd_step(bwd_i)
-            bwd_i += 1
-        while bwd_i < n_micro:
-            bwd_step(bwd_i)
-            bwd_i += 1
+            pending.append((None if self.is_first else inp, out))
+            binp, bout = pending.popleft()
+            if self.is_last:
+                grad_out = None
+            else:
+                grad_out = self.p2p.send_forward_recv_backward(out, bout)
+            grad_in = run_bwd(binp, bout, grad_out)
+            last_k = k == steady - 1
+            if self.is_first:
+                if not last_k:
+                    inp = get_input(fwd_i)
+            elif last_k:
+                self.p2p.send_backward(grad_in)
+            else:
+                inp = self.p2p.send_backward_recv_forward(grad_in, dev)
+                inp.requires_grad_(True)
+        # cooldown: drain deferred backwards
+        while pending:
+            binp, bout = pending.popleft()
+            grad_out = None if self.is_last else self.p2p.recv_backward(bout)
+            grad_in = run_bwd(binp, bout, grad_out)
+            if not self.is_first:
+                self.p2p.send_backward(grad_in)
 
         self._layers.allreduce_shared_weight_gradients()
         if optimizer is not None:
