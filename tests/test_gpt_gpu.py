@@ -75,3 +75,49 @@ def test_native_kernels_actually_used():
     paddle.set_flags({"FLAGS_use_native_kernels": False})
     assert not _ext.use_native(x)
     paddle.set_flags({"FLAGS_use_native_kernels": True})
+
+
+def test_llama_tiny_gpu_bf16():
+    from paddle_amd.models import build_llama
+    from paddle_amd.models.llama import LlamaPretrainingCriterion
+    paddle.seed(0)
+    m = build_llama("llama-tiny").to(device="cuda:0", dtype=torch.bfloat16)
+    loss_fn = LlamaPretrainingCriterion()
+    opt = paddle.optimizer.AdamW(learning_rate=3e-4, parameters=m.parameters())
+    ids = torch.randint(0, 1024, (2, 128), device="cuda:0")
+    losses = []
+    for _ in range(8):
+        loss = loss_fn(m(ids), ids)
+        loss.backward()
+        opt.step()
+        opt.clear_grad()
+        losses.append(float(loss.detach().float()))
+    assert losses[-1] < losses[0] - 0.3, losses
+
+
+def test_bert_tiny_gpu_bf16():
+    from paddle_amd.models import build_bert
+    paddle.seed(0)
+    m = build_bert("bert-tiny").to(device="cuda:0", dtype=torch.bfloat16)
+    ids = torch.randint(0, 1024, (2, 64), device="cuda:0")
+    mask = torch.ones(2, 64, device="cuda:0")
+    mlm, nsp = m(ids, attention_mask=mask)
+    loss = paddle.nn.functional.cross_entropy(mlm.reshape(-1, 1024), ids.reshape(-1))
+    loss.backward()
+    assert torch.isfinite(loss.float())
+
+
+def test_moe_tiny_gpu_bf16():
+    from paddle_amd.models.gpt import GPTConfig
+    from paddle_amd.models.moe import GPTMoEForPretraining
+    from paddle_amd.models import GPTPretrainingCriterion
+    paddle.seed(0)
+    cfg = GPTConfig(vocab_size=512, hidden_size=128, num_layers=2, num_heads=2,
+                    max_seq_len=128)
+    m = GPTMoEForPretraining(cfg, num_experts=4, k=2).to(device="cuda:0",
+                                                         dtype=torch.bfloat16)
+    loss_fn = GPTPretrainingCriterion()
+    ids = torch.randint(0, 512, (2, 64), device="cuda:0")
+    loss = loss_fn(m(ids), ids) + m.aux_loss()
+    loss.backward()
+    assert torch.isfinite(loss.float())
