@@ -52,8 +52,8 @@ class TopKGate(nn.Layer):
         _apply_initializer(Normal(0.0, 0.02), self.wg.weight)
 
     def forward(self, x):
-        # x: [tokens, h]
-        logits = self.wg(x.float())
+        # x: [tokens, h]; gate math in fp32 regardless of model dtype
+        logits = torch.matmul(x.float(), self.wg.weight.float())
         probs = torch.softmax(logits, -1)
         topv, topi = probs.topk(self.k, dim=-1)            # [T, k]
         # aux load-balance loss (gshard): num_experts * sum(me * ce)

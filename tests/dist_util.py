@@ -23,9 +23,15 @@ def free_port():
     return p
 
 
-def run_dist(script_body: str, world_size: int = 2, timeout: int = 240):
+_RETRYABLE = ("Address already in use", "EADDRINUSE", "Connection refused",
+              "Connection reset")
+
+
+def run_dist(script_body: str, world_size: int = 2, timeout: int = 240,
+             _attempt: int = 0):
     """Run `script_body` (python source) in world_size processes over gloo.
-    The body can `import paddle_amd as paddle` and use torch.distributed."""
+    The body can `import paddle_amd as paddle` and use torch.distributed.
+    Retries once on rendezvous races (port TIME_WAIT etc.)."""
     port = free_port()
     script = textwrap.dedent(script_body)
     procs = []
@@ -51,6 +57,11 @@ def run_dist(script_body: str, world_size: int = 2, timeout: int = 240):
                 q.kill()
             raise
         outs.append(out.decode())
-    for rank, (p, out) in enumerate(zip(procs, outs)):
-        assert p.returncode == 0, f"rank {rank} failed:\n{out}"
+    failed = [(rank, out) for rank, (p, out) in enumerate(zip(procs, outs))
+              if p.returncode != 0]
+    if failed and _attempt == 0 and any(m in out for _, out in failed
+                                        for m in _RETRYABLE):
+        return run_dist(script_body, world_size, timeout, _attempt=1)
+    for rank, out in failed:
+        raise AssertionError(f"rank {rank} failed:\n{out}")
     return outs
