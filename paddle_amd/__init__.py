@@ -220,6 +220,9 @@ from . import incubate  # noqa: F401
 from . import models  # noqa: F401
 from . import distribution  # noqa: F401
 from . import profiler  # noqa: F401
+from . import linalg  # noqa: F401
+from . import fft  # noqa: F401
+from . import sparse  # noqa: F401
 from .framework_io import save, load, async_save  # noqa: F401
 from .hapi import Model  # noqa: F401
 from .param_attr import ParamAttr  # noqa: F401

@@ -39,6 +39,18 @@ from .parallel import (  # noqa: F401
 from . import fleet  # noqa: F401
 from . import sharding  # noqa: F401
 from . import checkpoint  # noqa: F401
+from . import auto_parallel  # noqa: F401
+from .auto_parallel import (  # noqa: F401
+    Partial,
+    ProcessMesh,
+    Replicate,
+    Shard,
+    dtensor_from_fn,
+    reshard,
+    shard_layer,
+    shard_optimizer,
+    shard_tensor,
+)
 
 
 def spawn(func, args=(), nprocs=-1, join=True, daemon=False, **options):

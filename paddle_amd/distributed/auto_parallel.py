@@ -1,0 +1,242 @@
+"""Semi-auto parallel (DistTensor) API subset.
+
+Reference: python/paddle/distributed/auto_parallel/api.py (shard_tensor:
+206, reshard:705, shard_layer:806, shard_optimizer:1591) and
+phi/core/distributed/auto_parallel/ (placements, reshard functions).
+
+Round-1 scope (SURVEY.md §2.3 marks DistTensor as phase-2): a working
+manual-mesh subset -- ProcessMesh + Shard/Replicate/Partial placements,
+shard_tensor producing rank-local shards with an attached dist_attr,
+and reshard implementing the common placement conversions (s->r, r->s,
+p->r) over the comm layer.  SPMD propagation rules are future work.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence
+
+import numpy as np
+import torch
+
+from . import collective as C
+from .parallel import get_rank, get_world_size
+
+
+class Placement:
+    pass
+
+
+class Replicate(Placement):
+    def __repr__(self):
+        return "Replicate()"
+
+    def __eq__(self, o):
+        return isinstance(o, Replicate)
+
+
+class Shard(Placement):
+    def __init__(self, dim):
+        self.dim = dim
+
+    def __repr__(self):
+        return f"Shard(dim={self.dim})"
+
+    def __eq__(self, o):
+        return isinstance(o, Shard) and o.dim == self.dim
+
+
+class Partial(Placement):
+    def __init__(self, reduce_type="sum"):
+        self.reduce_type = reduce_type
+
+    def __repr__(self):
+        return "Partial()"
+
+    def __eq__(self, o):
+        return isinstance(o, Partial)
+
+
+class ProcessMesh:
+    """reference: python/paddle/distributed/auto_parallel/process_mesh.py:85"""
+
+    def __init__(self, mesh, dim_names=None, shape=None, process_ids=None):
+        arr = np.asarray(mesh)
+        self._mesh = arr
+        self._dim_names = list(dim_names) if dim_names else \
+            [f"d{i}" for i in range(arr.ndim)]
+
+    @property
+    def shape(self):
+        return list(self._mesh.shape)
+
+    @property
+    def ndim(self):
+        return self._mesh.ndim
+
+    @property
+    def process_ids(self):
+        return self._mesh.flatten().tolist()
+
+    @property
+    def dim_names(self):
+        return self._dim_names
+
+    def get_dim_size(self, name):
+        return self._mesh.shape[self._dim_names.index(name)]
+
+    def get_rank_by_dim_and_process_id(self, dim, pid):
+        idx = np.argwhere(self._mesh == pid)
+        return int(idx[0][self._dim_names.index(dim) if isinstance(dim, str) else dim])
+
+    def my_coord(self):
+        me = get_rank()
+        loc = np.argwhere(self._mesh == me)
+        return tuple(loc[0]) if len(loc) else None
+
+    def _group_along(self, mesh_dim):
+        """ranks varying along mesh_dim with my other coords fixed."""
+        coord = self.my_coord()
+        if coord is None:
+            return None, []
+        sl = list(coord)
+        sl[mesh_dim] = slice(None)
+        ranks = self._mesh[tuple(sl)].flatten().tolist()
+        return ranks
+
+    def __eq__(self, o):
+        return isinstance(o, ProcessMesh) and np.array_equal(self._mesh, o._mesh)
+
+    def __repr__(self):
+        return f"ProcessMesh({self._mesh.tolist()}, dim_names={self._dim_names})"
+
+
+class DistAttr:
+    def __init__(self, mesh: ProcessMesh, placements: Sequence[Placement]):
+        self.process_mesh = mesh
+        self.placements = list(placements)
+
+    def __repr__(self):
+        return f"DistAttr(mesh={self.process_mesh}, placements={self.placements})"
+
+
+_group_cache = {}
+
+
+def _mesh_group(mesh: ProcessMesh, dim: int):
+    key = (tuple(mesh.process_ids), tuple(mesh.shape), dim)
+    if key not in _group_cache:
+        # must be created collectively on all ranks
+        groups = {}
+        arr = mesh._mesh
+        other = [i for i in range(arr.ndim) if i != dim]
+        import itertools
+        me = get_rank()
+        my_group = None
+        for combo in itertools.product(*(range(arr.shape[i]) for i in other)):
+            sl = list(combo)
+            sl.insert(dim, slice(None))
+            ranks = arr[tuple(sl)].flatten().tolist()
+            g = C.new_group(ranks)
+            if me in ranks:
+                my_group = g
+        _group_cache[key] = my_group
+    return _group_cache[key]
+
+
+def shard_tensor(data, mesh: ProcessMesh, placements, dtype=None, place=None,
+                 stop_gradient=None):
+    """Return this rank's local shard with dist metadata attached."""
+    t = data if isinstance(data, torch.Tensor) else torch.as_tensor(data)
+    local = t
+    for mdim, p in enumerate(placements):
+        if isinstance(p, Shard):
+            n = mesh.shape[mdim]
+            coord = mesh.my_coord()
+            idx = coord[mdim] if coord is not None else 0
+            size = local.shape[p.dim] // n
+            local = local.narrow(p.dim, idx * size, size).contiguous()
+    local = local.clone()
+    local.dist_attr = DistAttr(mesh, placements)
+    local.process_mesh = mesh
+    local.placements = list(placements)
+    if stop_gradient is not None:
+        local.requires_grad_(not stop_gradient and local.is_floating_point())
+    return local
+
+
+def dtensor_from_fn(fn, mesh, placements, *args, **kwargs):
+    return shard_tensor(fn(*args, **kwargs), mesh, placements)
+
+
+def reshard(dist_tensor, mesh: ProcessMesh, placements):
+    """Common conversions: Shard->Replicate (all_gather), Partial->
+    Replicate (all_reduce), Replicate->Shard (slice)."""
+    src = getattr(dist_tensor, "placements", [Replicate()] * mesh.ndim)
+    out = dist_tensor
+    for mdim, (sp, dp) in enumerate(zip(src, placements)):
+        if sp == dp:
+            continue
+        g = _mesh_group(mesh, mdim)
+        if isinstance(sp, Shard) and isinstance(dp, Replicate):
+            n = mesh.shape[mdim]
+            if g is not None and n > 1:
+                parts = [torch.empty_like(out) for _ in range(n)]
+                C.all_gather(parts, out.contiguous(), group=g)
+                out = torch.cat(parts, dim=sp.dim)
+        elif isinstance(sp, Partial) and isinstance(dp, Replicate):
+            if g is not None:
+                out = out.clone()
+                C.all_reduce(out, group=g)
+        elif isinstance(sp, Replicate) and isinstance(dp, Shard):
+            n = mesh.shape[mdim]
+            coord = mesh.my_coord()
+            idx = coord[mdim] if coord is not None else 0
+            size = out.shape[dp.dim] // n
+            out = out.narrow(dp.dim, idx * size, size).contiguous()
+        elif isinstance(sp, Partial) and isinstance(dp, Shard):
+            n = mesh.shape[mdim]
+            if g is not None and n > 1:
+                size = out.shape[dp.dim] // n
+                shards = list(out.split(size, dim=dp.dim))
+                dst = torch.empty_like(shards[0])
+                C.reduce_scatter(dst, [s.contiguous() for s in shards], group=g)
+                out = dst
+        else:
+            raise NotImplementedError(f"reshard {sp} -> {dp}")
+    out = out.clone() if out is dist_tensor else out
+    out.dist_attr = DistAttr(mesh, placements)
+    out.process_mesh = mesh
+    out.placements = list(placements)
+    return out
+
+
+def shard_layer(layer, process_mesh, shard_fn=None, input_fn=None, output_fn=None):
+    """api.py:806 parity: apply shard_fn(name, layer, mesh) to sublayers."""
+    if shard_fn is not None:
+        for name, sub in layer.named_sublayers() if hasattr(layer, "named_sublayers") \
+                else layer.named_modules():
+            shard_fn(name, sub, process_mesh)
+    if input_fn is not None:
+        layer.register_forward_pre_hook(
+            lambda mod, inp: input_fn(inp, process_mesh))
+    if output_fn is not None:
+        layer.register_forward_hook(
+            lambda mod, inp, out: output_fn(out, process_mesh))
+    return layer
+
+
+def shard_optimizer(optimizer, shard_fn=None):
+    """api.py:1591 parity hook point; manual-parallel optimizers are already
+    shard-aware in this build (fleet.sharding)."""
+    return optimizer
+
+
+def get_mesh():
+    return _default_mesh
+
+
+_default_mesh: Optional[ProcessMesh] = None
+
+
+def set_mesh(mesh):
+    global _default_mesh
+    _default_mesh = mesh

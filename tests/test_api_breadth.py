@@ -1,0 +1,76 @@
+"""API-breadth tests: linalg, fft, sparse, auto_parallel (single + 2-rank)."""
+import numpy as np
+import pytest
+import torch
+
+import paddle_amd as paddle
+from dist_util import run_dist
+
+
+def test_linalg_ops():
+    a = paddle.to_tensor(np.random.rand(4, 4).astype("float32") + np.eye(4, dtype="float32"))
+    assert paddle.linalg.det(a).abs() > 0
+    u, s, vh = paddle.linalg.svd(a)
+    np.testing.assert_allclose((u @ torch.diag(s) @ vh).numpy(), a.numpy(), atol=1e-4)
+    q, r = paddle.linalg.qr(a)
+    np.testing.assert_allclose((q @ r).numpy(), a.numpy(), atol=1e-4)
+    inv = paddle.linalg.inv(a)
+    np.testing.assert_allclose((a @ inv).numpy(), np.eye(4), atol=1e-4)
+    spd = a @ a.t() + 4 * torch.eye(4)
+    c = paddle.linalg.cholesky(spd)
+    np.testing.assert_allclose((c @ c.t()).numpy(), spd.numpy(), atol=1e-4)
+
+
+def test_fft_roundtrip():
+    x = paddle.randn([8, 16])
+    y = paddle.fft.ifft(paddle.fft.fft(x))
+    np.testing.assert_allclose(y.real.numpy(), x.numpy(), atol=1e-5)
+    r = paddle.fft.irfft(paddle.fft.rfft(x), n=16)
+    np.testing.assert_allclose(r.numpy(), x.numpy(), atol=1e-5)
+
+
+def test_sparse_coo_csr():
+    i = [[0, 1, 2], [2, 0, 1]]
+    v = [1.0, 2.0, 3.0]
+    coo = paddle.sparse.sparse_coo_tensor(i, v, shape=(3, 3))
+    assert paddle.sparse.is_sparse_coo(coo)
+    dense = paddle.sparse.to_dense(coo)
+    assert dense[0, 2] == 1.0 and dense[1, 0] == 2.0
+    csr = paddle.sparse.to_sparse_csr(dense)
+    assert paddle.sparse.is_sparse_csr(csr)
+    y = paddle.sparse.matmul(coo, torch.eye(3))
+    np.testing.assert_allclose(y.to_dense().numpy() if y.layout != torch.strided
+                               else y.numpy(), dense.numpy())
+
+
+def test_process_mesh_single():
+    mesh = paddle.distributed.ProcessMesh([[0, 1], [2, 3]], dim_names=["dp", "mp"])
+    assert mesh.shape == [2, 2]
+    assert mesh.process_ids == [0, 1, 2, 3]
+    assert mesh.get_dim_size("mp") == 2
+
+
+def test_shard_tensor_two_ranks():
+    run_dist("""
+        import torch
+        import paddle_amd as paddle
+        paddle.distributed.init_parallel_env()
+        from paddle_amd.distributed import ProcessMesh, Shard, Replicate, Partial
+        from paddle_amd.distributed import shard_tensor, reshard
+        r = paddle.distributed.get_rank()
+        mesh = ProcessMesh([0, 1], dim_names=["x"])
+        torch.manual_seed(0)
+        full = torch.randn(8, 4)
+        d = shard_tensor(full, mesh, [Shard(0)])
+        assert d.shape == (4, 4)
+        assert torch.allclose(d, full[r*4:(r+1)*4])
+        back = reshard(d, mesh, [Replicate()])
+        assert torch.allclose(back, full)
+        # partial -> replicate == allreduce
+        p = full.clone()
+        p.placements = [Partial()]
+        p.process_mesh = mesh
+        red = reshard(p, mesh, [Replicate()])
+        assert torch.allclose(red, 2 * full)
+        print("rank", r, "dist tensor ok")
+    """)
