@@ -140,17 +140,38 @@ class _Unit:
                 self.grad_shard_fp32.add_(flat[0:self.shard_size])
             else:
                 self.grad_shard_fp32.copy_(flat[0:self.shard_size])
-        else:
+            flat.zero_()
+        elif accumulate:
+            # grad-accumulation path: keep it synchronous (flat is re-bound
+            # next micro-step, must not race the collective)
             flat.div_(self.world)
             out = torch.empty(self.shard_size, dtype=self.dtype,
                               device=self.shard.device)
             C.reduce_scatter_tensor(out, flat, group=self.group)
-            if accumulate and self.accum_steps > 0:
+            if self.accum_steps > 0:
                 self.grad_shard_fp32.add_(out)
             else:
                 self.grad_shard_fp32.copy_(out)
+            flat.zero_()
+        else:
+            # async reduce-scatter overlapped with the remaining backward;
+            # the optimizer's step() calls finish_grad_reduce() first
+            flat.div_(self.world)
+            out = torch.empty(self.shard_size, dtype=self.dtype,
+                              device=self.shard.device)
+            work = C.reduce_scatter_tensor(out, flat, group=self.group,
+                                           sync_op=False)
+            self.grad_work = (work, out)
         self.accum_steps += 1
-        flat.zero_()
+
+    def finish_grad_reduce(self):
+        if self.grad_work is not None:
+            work, out = self.grad_work
+            if work is not None:
+                work.wait()
+            self.grad_shard_fp32.copy_(out)
+            self.grad_flat.zero_()
+            self.grad_work = None
 
 
 class GroupShardedStage3(torch.nn.Module):
@@ -399,6 +420,8 @@ class ShardedAdamW:
     def _global_grad_norm(self):
         total = None
         for u in self.model._units:
+            u.finish_grad_reduce()
+        for u in self.model._units:
             sq = hot.l2_norm_squared(u.grad_shard_fp32)
             total = sq if total is None else total + sq
         if total is None:
@@ -411,6 +434,8 @@ class ShardedAdamW:
     def step(self):
         self.step_count += 1
         lr = self.get_lr()
+        for u in self.model._units:
+            u.finish_grad_reduce()
         clip_coeff = None
         if self.grad_clip is not None:
             clip_norm = getattr(self.grad_clip, "clip_norm", self.grad_clip)
@@ -428,6 +453,7 @@ class ShardedAdamW:
 
     def clear_grad(self, set_to_zero=True):
         for u in self.model._units:
+            u.finish_grad_reduce()  # consume any pending async reduce
             u.grad_shard_fp32.zero_()
             u.accum_steps = 0
 
