@@ -1,0 +1,47 @@
+"""paddle.text parity surface: dataset shells (no network in this
+environment -- datasets construct from local files or synthetic data)."""
+from __future__ import annotations
+
+from ..io import Dataset
+
+
+class Imdb(Dataset):
+    def __init__(self, data_file=None, mode="train", cutoff=150):
+        raise NotImplementedError(
+            "dataset downloads need network; pass local files via data_file "
+            "to a custom paddle.io.Dataset")
+
+
+class Conll05st(Dataset):
+    def __init__(self, *a, **k):
+        raise NotImplementedError("no network for dataset download")
+
+
+class UCIHousing(Dataset):
+    def __init__(self, *a, **k):
+        raise NotImplementedError("no network for dataset download")
+
+
+class ViterbiDecoder:
+    def __init__(self, transitions, include_bos_eos_tag=True):
+        import torch
+        self.trans = torch.as_tensor(transitions)
+        self.with_tag = include_bos_eos_tag
+
+    def __call__(self, potentials, lengths):
+        import torch
+        b, s, n = potentials.shape
+        scores = potentials[:, 0]
+        hist = []
+        for t in range(1, s):
+            prev = scores.unsqueeze(2) + self.trans.unsqueeze(0)
+            best, idx = prev.max(1)
+            scores = best + potentials[:, t]
+            hist.append(idx)
+        best_final, last = scores.max(-1)
+        paths = [last]
+        for idx in reversed(hist):
+            last = idx.gather(1, last.unsqueeze(1)).squeeze(1)
+            paths.append(last)
+        paths.reverse()
+        return best_final, torch.stack(paths, dim=1)
