@@ -345,6 +345,22 @@ Tensor embedding_bwd(const Tensor& dout, const Tensor& ids_in, int64_t vocab,
   return dtable;
 }
 
+// ---- hand-written GEMM ----------------------------------------------------
+Tensor gemm_bf16(const Tensor& a, const Tensor& b, bool b_is_nt) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2);
+  TORCH_CHECK(a.stride(1) == 1 && b.stride(1) == 1, "row-major inputs only");
+  int64_t m = a.size(0), k = a.size(1);
+  int64_t n = b_is_nt ? b.size(0) : b.size(1);
+  TORCH_CHECK(b_is_nt ? b.size(1) == k : b.size(0) == k);
+  TORCH_CHECK(k % 64 == 0 && k >= 128, "gemm_bf16: K must be multiple of 64");
+  auto c = torch::empty({m, n}, a.options());
+  pa::gemm_bf16(a.const_data_ptr(), b.const_data_ptr(), c.mutable_data_ptr(),
+                m, n, k, a.stride(0), b.stride(0), c.stride(0), b_is_nt,
+                cur_stream());
+  return c;
+}
+
 // ---- probe ----------------------------------------------------------------
 Tensor mfma_probe(const Tensor& a, const Tensor& bt) {
   CHECK_IN(a); CHECK_IN(bt);
@@ -385,6 +401,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_add_bwd", &dropout_add_bwd);
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);
+  m.def("gemm_bf16", &gemm_bf16);
   m.def("mfma_probe", &mfma_probe);
   m.def("mfma_probe32", &mfma_probe32);
   m.attr("compiled_arch") = "gfx950";

@@ -117,4 +117,10 @@ void embedding_bwd(const void* dout, const int64_t* ids, float* dtable,
                    int64_t n_ids, int64_t d, int64_t vocab, int64_t padding_idx,
                    int dtype, hipStream_t s);
 
+// ---- hand-written bf16 MFMA GEMM (C[m][n] = A[m][k] x B) ------------------
+// b_is_nt: B passed as Bt[n][k] row-major (fast path); else B[k][n].
+void gemm_bf16(const void* a, const void* b, void* c, int64_t m, int64_t n,
+               int64_t k, int64_t lda, int64_t ldb, int64_t ldc, bool b_is_nt,
+               hipStream_t s);
+
 }  // namespace pa

@@ -39,7 +39,11 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--tunableop", action="store_true")
     ap.add_argument("--duration", type=int, default=100)
+    ap.add_argument("--custom", action="store_true")
     args = ap.parse_args()
+    if args.custom:
+        bench_custom()
+        return
     if args.tunableop:
         import torch.cuda.tunable as tunable
         tunable.enable(True)
@@ -63,3 +67,27 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def bench_custom():
+    from paddle_amd import _ext
+    C = _ext.get_ext()
+    print("== hand-written MFMA gemm_bf16 ==")
+    for name, M, K, N in SHAPES:
+        a = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+        bt = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+        b = torch.randn(K, N, device="cuda", dtype=torch.bfloat16)
+        flops = 2.0 * M * K * N
+        # correctness (small slice check vs torch)
+        out_nt = C.gemm_bf16(a, bt, True)
+        ref_nt = a.float() @ bt.float().t()
+        err = (out_nt.float() - ref_nt).abs().max() / ref_nt.abs().max()
+        out_nn = C.gemm_bf16(a, b, False)
+        ref_nn = a.float() @ b.float()
+        err2 = (out_nn.float() - ref_nn).abs().max() / ref_nn.abs().max()
+        t_nt = bench(lambda: C.gemm_bf16(a, bt, True))
+        t_nn = bench(lambda: C.gemm_bf16(a, b, False))
+        print(f"{name:12s}: NT {flops/t_nt/1e12:7.1f} TF (relerr {err:.2e})  "
+              f"NN {flops/t_nn/1e12:7.1f} TF (relerr {err2:.2e})")
+        del a, bt, b
+        torch.cuda.empty_cache()
