@@ -65,10 +65,6 @@ def main():
         torch.cuda.empty_cache()
 
 
-if __name__ == "__main__":
-    main()
-
-
 def bench_custom():
     from paddle_amd import _ext
     C = _ext.get_ext()
@@ -91,3 +87,7 @@ def bench_custom():
               f"NN {flops/t_nn/1e12:7.1f} TF (relerr {err2:.2e})")
         del a, bt, b
         torch.cuda.empty_cache()
+
+
+if __name__ == "__main__":
+    main()
