@@ -1,46 +1,47 @@
 // Hand-written bf16 MFMA GEMM for gfx950 (C = A x B, fp32 accumulate,
 // bf16 out).
 //
-// Geometry: block tile 256x256, K-step 64 (two 32-wide k-half phases),
-// 8 waves (2M x 4N, per-wave C = 128x64 = acc[8][4] of
-// mfma_f32_16x16x32_bf16).
+// Geometry (guide §5 "256² 8-phase template" -- re-derived, simplified
+// schedule):
+//   * block tile 256x256, K-step 64, 8 waves (2M x 4N), 512 threads
+//   * per-wave output 128x64 = acc[8][4] mfma_f32_16x16x32_bf16 fragments
+//   * LDS: double-buffered A[256][64] + B^T[256][64] bf16 = 128 KiB
+//   * staging for K-tile t+1 issued at the START of tile t's compute via
+//     async global_load_lds (pre-swizzled source) -- ~64 MFMA of cover
+//     for the HBM latency (T14 issue-early at K-tile granularity)
+//   * st_16x32-style XOR swizzle on LDS rows; s_setprio around the MFMA
+//     cluster (T5); XCD-aware bijective workgroup swizzle (T1, m204)
 //
-// Pipeline (the load-bearing part, guide §5.5 T3/T4): LDS is a RING of
-// 10 half-tile slots (16 KiB each = [256 rows][32 k] bf16) giving
-// 2-K-tile lookahead.  Each phase stages one K-tile-half pair for tile
-// kt+2 via async global_load_lds (pre-swizzled source), computes 32
-// MFMAs, then waits a COUNTED `s_waitcnt vmcnt(16)` -- staged loads stay
-// in flight across barriers instead of draining (the m97-ceiling
-// mistake), followed by a raw s_barrier.  s_setprio(1) wraps the MFMA
-// cluster (T5); XCD-aware bijective workgroup swizzle (T1/m204).
-//
-// Layouts: NT: A[M][K] rm + Bt[N][K] rm (both DMA-staged, fast).
-//          NN: A[M][K] rm + B[K][N] rm (B transposed via reg staging).
+// Operand layouts (templates):
+//   NT: A[M][K] rm, Bt[N][K] rm  (fastest: both DMA-staged)
+//   NN: A[M][K] rm, B[K][N] rm   (B transposed into LDS via reg staging)
+// TN (wgrad) stays on hipBLASLt for now (python autotune picks per shape).
 #include "common.h"
 #include "api.h"
 
 namespace pa {
 
-#define RAW_BARRIER() __builtin_amdgcn_s_barrier()
-#define WAITCNT_VM(N) asm volatile("s_waitcnt vmcnt(" #N ")" ::: "memory")
-#define WAITCNT_LGKM0() asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
+typedef __attribute__((ext_vector_type(2))) int intx2;
 
-// half-tile slot: [256][32] bf16; row stride 64B.  XOR the 8-elem granule
-// (bits 3..4) with row&3: a 16-lane column-slice read spreads over
-// (row parity x 4 granules) = 8 bank-slots -> ~2-way.
-__device__ __forceinline__ unsigned h_swz(unsigned row, unsigned col) {
-  return row * 32 + (col ^ ((row & 3u) << 3));
+#define GEMM_SETPRIO(x) __builtin_amdgcn_s_setprio(x)
+
+// LDS element offset with row-XOR swizzle: rows are 64 bf16 = 128 B; XOR
+// the 16B-granule index (elem bits 3..5) with row&7 so a 16-lane column
+// slice spreads over 8 bank-slots (2-way; same fix as flash_attn.hip).
+__device__ __forceinline__ unsigned g_swz(unsigned row, unsigned col) {
+  return row * 64 + (col ^ ((row & 7u) << 3));
 }
 
-template <bool BT>
+template <bool BT>  // BT=true: B supplied as Bt[N][K] (NT); false: B[K][N]
 __launch_bounds__(512, 2)
 __global__ void gemm_bf16_kernel(const short* __restrict__ ag, const short* __restrict__ bg,
                                  short* __restrict__ cg, int M, int N, int K,
                                  long long lda, long long ldb, long long ldc) {
   constexpr int BM = 256, BN = 256, BK = 64;
-  constexpr int NSLOT = 10;
-  __shared__ short ring[NSLOT][256 * 32];
+  __shared__ short a_lds[2][BM * BK];
+  __shared__ short b_lds[2][BN * BK];
 
+  // XCD-aware bijective workgroup swizzle (m204): contiguous chunks per XCD
   const int nwg = gridDim.x * gridDim.y;
   int orig = blockIdx.y * gridDim.x + blockIdx.x;
   {
@@ -49,106 +50,119 @@ __global__ void gemm_bf16_kernel(const short* __restrict__ ag, const short* __re
     int xcd = orig % nx, pos = orig / nx;
     orig = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + pos;
   }
-  const int gm = (M + BM - 1) / BM;
-  const int bm = orig % gm;
-  const int bn = orig / gm;
+  const int bm = orig % ((M + BM - 1) / BM);
+  const int bn = orig / ((M + BM - 1) / BM);
   const int row0 = bm * BM, col0 = bn * BN;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int wr = wid >> 2;
-  const int wc = wid & 3;
+  const int wr = wid >> 2;          // 0..1: which 128 rows
+  const int wc = wid & 3;           // 0..3: which 64 cols
   const int l16 = lane & 15;
   const int lg = lane >> 4;
 
-  // staging map: half-tile 256x32 = 8192 elems; 512thr x 8 = 2 rounds
-  int s_r[2], s_cp[2];
+  // staging thread->element maps (once)
+  // A tile: 256x64 = 16384 elems; 512 thr x 8 = 4096/round -> 4 rounds
+  int a_r[4], a_cp[4];
 #pragma unroll
-  for (int it = 0; it < 2; ++it) {
+  for (int it = 0; it < 4; ++it) {
     int flat = it * 4096 + tid * 8;
-    int r = flat / 32, c = flat % 32;
-    s_r[it] = r;
-    s_cp[it] = c ^ ((r & 3) << 3);
+    int r = flat / BK, c = flat % BK;
+    a_r[it] = r;
+    a_cp[it] = c ^ ((r & 7) << 3);   // pre-swizzled source column
   }
 
-  const bool interior = (row0 + BM <= M) && (col0 + BN <= N) && (K % BK == 0);
+  const bool interior = (row0 + BM <= M) && (col0 + BN <= N);
 
-  // slot index for half h of tile kt: h_global = kt*4 + idx,
-  // idx: 0=A_k0 1=B_k0 2=A_k1 3=B_k1
-  auto slot = [&](int kt, int idx) { return (kt * 4 + idx) % NSLOT; };
-
-  // stage one A half (k-half kh) of tile kt into its slot (2 DMA / thread)
-  auto stage_a = [&](int kt, int kh) {
-    const long long k0 = (long long)kt * BK + kh * 32;
-    short* dst = ring[slot(kt, kh * 2)];
-    if (interior) {
+  auto stage = [&](int buf, int kt) {
+    const long long k0 = (long long)kt * BK;
+    if (interior && K - k0 >= BK) {
+      // async DMA path
 #pragma unroll
-      for (int it = 0; it < 2; ++it) {
-        const short* src = ag + (long long)(row0 + s_r[it]) * lda + k0 + s_cp[it];
+      for (int it = 0; it < 4; ++it) {
+        const short* src = ag + (long long)(row0 + a_r[it]) * lda + k0 + a_cp[it];
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) unsigned int*)src,
-            (__attribute__((address_space(3))) unsigned int*)&dst[it * 4096 + tid * 8],
+            (__attribute__((address_space(3))) unsigned int*)&a_lds[buf][it * 4096 + tid * 8],
             16, 0, 0);
       }
-    } else {
+      if (BT) {
 #pragma unroll
-      for (int it = 0; it < 2; ++it) {
-        int r = s_r[it], c = (it * 4096 + tid * 8) % 32;
-        shortx8 v;
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          v[j] = (row0 + r < M && k0 + c + j < K)
-                     ? ag[(long long)(row0 + r) * lda + k0 + c + j] : (short)0;
-        *reinterpret_cast<shortx8*>(&dst[h_swz(r, c)]) = v;
-      }
-    }
-  };
-
-  auto stage_b = [&](int kt, int kh) {
-    const long long k0 = (long long)kt * BK + kh * 32;
-    short* dst = ring[slot(kt, kh * 2 + 1)];
-    if (BT) {
-      if (interior) {
-#pragma unroll
-        for (int it = 0; it < 2; ++it) {
-          const short* src = bg + (long long)(col0 + s_r[it]) * ldb + k0 + s_cp[it];
+        for (int it = 0; it < 4; ++it) {
+          const short* src = bg + (long long)(col0 + a_r[it]) * ldb + k0 + a_cp[it];
           __builtin_amdgcn_global_load_lds(
               (const __attribute__((address_space(1))) unsigned int*)src,
-              (__attribute__((address_space(3))) unsigned int*)&dst[it * 4096 + tid * 8],
+              (__attribute__((address_space(3))) unsigned int*)&b_lds[buf][it * 4096 + tid * 8],
               16, 0, 0);
         }
       } else {
+        // B [K][N]: load rows of B (contiguous n), write transposed to LDS
+        const int rot = tid & 7;
 #pragma unroll
-        for (int it = 0; it < 2; ++it) {
-          int r = s_r[it], c = (it * 4096 + tid * 8) % 32;
-          shortx8 v;
+        for (int it = 0; it < 4; ++it) {
+          int flat = it * 4096 + tid * 8;
+          int kr = flat / BN;             // 0..63  (k within tile)
+          int nc = flat % BN;             // 0..255 (n within tile)
+          shortx8 v = *reinterpret_cast<const shortx8*>(
+              bg + (k0 + kr) * ldb + col0 + nc);
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            v[j] = (col0 + r < N && k0 + c + j < K)
-                       ? bg[(long long)(col0 + r) * ldb + k0 + c + j] : (short)0;
-          *reinterpret_cast<shortx8*>(&dst[h_swz(r, c)]) = v;
+          for (int j = 0; j < 8; ++j) {
+            int i = (j + rot) & 7;
+            b_lds[buf][g_swz(nc + i, kr)] = v[i];
+          }
         }
       }
     } else {
-      // B [K][N]: rows are 32 k x 256 n; transpose into [n][k] slot
-      const int rot = tid & 7;
+      // boundary tile: guarded reg staging (zero-fill)
 #pragma unroll
-      for (int it = 0; it < 2; ++it) {
+      for (int it = 0; it < 4; ++it) {
         int flat = it * 4096 + tid * 8;
-        int kr = flat / BN, nc = flat % BN;
+        int r = flat / BK, c = flat % BK;
         shortx8 v;
-        if (interior || (k0 + kr < K && col0 + nc + 7 < N))
-          v = *reinterpret_cast<const shortx8*>(bg + (k0 + kr) * ldb + col0 + nc);
+        if (row0 + r < M && k0 + c + 7 < K)
+          v = *reinterpret_cast<const shortx8*>(ag + (long long)(row0 + r) * lda + k0 + c);
         else
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            v[j] = (k0 + kr < K && col0 + nc + j < N)
-                       ? bg[(k0 + kr) * ldb + col0 + nc + j] : (short)0;
+            v[j] = (row0 + r < M && k0 + c + j < K)
+                       ? ag[(long long)(row0 + r) * lda + k0 + c + j] : (short)0;
+        *reinterpret_cast<shortx8*>(&a_lds[buf][g_swz(r, c)]) = v;  // c%8==0 -> 16B aligned
+      }
+      if (BT) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int i = (j + rot) & 7;
-          dst[h_swz(nc + i, kr)] = v[i];
+        for (int it = 0; it < 4; ++it) {
+          int flat = it * 4096 + tid * 8;
+          int r = flat / BK, c = flat % BK;
+          shortx8 v;
+          if (col0 + r < N && k0 + c + 7 < K)
+            v = *reinterpret_cast<const shortx8*>(bg + (long long)(col0 + r) * ldb + k0 + c);
+          else
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              v[j] = (col0 + r < N && k0 + c + j < K)
+                         ? bg[(long long)(col0 + r) * ldb + k0 + c + j] : (short)0;
+          *reinterpret_cast<shortx8*>(&b_lds[buf][g_swz(r, c)]) = v;
+        }
+      } else {
+        const int rot = tid & 7;
+#pragma unroll
+        for (int it = 0; it < 4; ++it) {
+          int flat = it * 4096 + tid * 8;
+          int kr = flat / BN, nc = flat % BN;
+          shortx8 v;
+          if (k0 + kr < K && col0 + nc + 7 < N)
+            v = *reinterpret_cast<const shortx8*>(bg + (k0 + kr) * ldb + col0 + nc);
+          else
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              v[j] = (k0 + kr < K && col0 + nc + j < N)
+                         ? bg[(k0 + kr) * ldb + col0 + nc + j] : (short)0;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            int i = (j + rot) & 7;
+            b_lds[buf][g_swz(nc + i, kr)] = v[i];
+          }
         }
       }
     }
@@ -161,65 +175,53 @@ __global__ void gemm_bf16_kernel(const short* __restrict__ ag, const short* __re
     for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
 
   const int nkt = (K + BK - 1) / BK;
-  // DMA-count discipline only valid when every stage in this block is DMA
-  const bool counted = interior && BT;  // NN's B staging is reg-based
-
-  // prologue: tiles 0 and 1 fully staged
-  stage_a(0, 0); stage_b(0, 0); stage_a(0, 1); stage_b(0, 1);
-  if (nkt > 1) { stage_a(1, 0); stage_b(1, 0); stage_a(1, 1); stage_b(1, 1); }
-  if (counted && nkt > 1) {
-    WAITCNT_VM(12);  // tile0 phase-0 halves done; 3 newer stage-events fly
-  } else {
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  }
-  WAITCNT_LGKM0();
-  RAW_BARRIER();
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
 
   for (int kt = 0; kt < nkt; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < nkt) stage(cur ^ 1, kt + 1);  // issue early: covered by MFMAs
+
+    // ---- compute K-tile kt: 4 sub-phases of 2 mf x 4 nf x 2 ks ----------
+    // B frags (shared across mf): load once
+    shortx8 bf[4][2];
 #pragma unroll
-    for (int kh = 0; kh < 2; ++kh) {
-      // stage tile kt+2's matching halves (slots freed by the barrier
-      // that ended the previous phase -- ring arithmetic in header note)
-      if (kt + 2 < nkt) {
-        stage_a(kt + 2, kh);
-        stage_b(kt + 2, kh);
-      }
-      const short* a_sl = ring[slot(kt, kh * 2)];
-      const short* b_sl = ring[slot(kt, kh * 2 + 1)];
-      shortx8 af[8], bf[4];
+    for (int nf = 0; nf < 4; ++nf)
 #pragma unroll
-      for (int m = 0; m < 8; ++m)
-        af[m] = *reinterpret_cast<const shortx8*>(
-            &a_sl[h_swz(wr * 128 + m * 16 + l16, lg * 8)]);
+      for (int ks = 0; ks < 2; ++ks)
+        bf[nf][ks] = *reinterpret_cast<const shortx8*>(
+            &b_lds[cur][g_swz(wc * 64 + nf * 16 + l16, ks * 32 + lg * 8)]);
 #pragma unroll
-      for (int n = 0; n < 4; ++n)
-        bf[n] = *reinterpret_cast<const shortx8*>(
-            &b_sl[h_swz(wc * 64 + n * 16 + l16, lg * 8)]);
-      WAITCNT_LGKM0();
+    for (int q = 0; q < 4; ++q) {
+      shortx8 af[2][2];
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          af[m][ks] = *reinterpret_cast<const shortx8*>(
+              &a_lds[cur][g_swz(wr * 128 + (q * 2 + m) * 16 + l16, ks * 32 + lg * 8)]);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_sched_barrier(0);
-      __builtin_amdgcn_s_setprio(1);
+      GEMM_SETPRIO(1);
 #pragma unroll
-      for (int m = 0; m < 8; ++m)
+      for (int m = 0; m < 2; ++m)
 #pragma unroll
-        for (int n = 0; n < 4; ++n)
-          acc[m][n] = mfma_bf16(af[m], bf[n], acc[m][n]);
-      __builtin_amdgcn_s_setprio(0);
-      // counted wait: the NEXT phase's halves were staged 2 tiles ago;
-      // 3 stage-events (4 DMA each) were issued after them -> allow 12
-      if (counted) {
-        WAITCNT_VM(12);
-      } else {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        WAITCNT_LGKM0();
-      }
-      RAW_BARRIER();
+        for (int nf = 0; nf < 4; ++nf)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[q * 2 + m][nf] = mfma_bf16(af[m][ks], bf[nf][ks], acc[q * 2 + m][nf]);
+      GEMM_SETPRIO(0);
     }
+    // next tile's loads must have landed before we flip buffers
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
   }
 
-  // epilogue
+  // ---- epilogue: bf16 store -------------------------------------------
 #pragma unroll
   for (int m = 0; m < 8; ++m) {
-    int row = row0 + wr * 128 + m * 16 + lg * 4;
+    int row = row0 + wr * 128 + m * 16 + (lg * 4);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       if (row + r >= M) continue;

@@ -25,7 +25,12 @@ sdp_kernel = None
 
 
 def linear(x, weight, bias=None, name=None):
-    # paddle weight layout: [in_features, out_features]
+    # paddle weight layout: [in_features, out_features].  addmm fuses the
+    # bias into the hipBLASLt epilogue (saves one HBM pass per call).
+    if bias is not None and x.dim() >= 2:
+        x2 = x.reshape(-1, x.shape[-1])
+        out = torch.addmm(bias, x2, weight)
+        return out.reshape(*x.shape[:-1], weight.shape[-1])
     out = torch.matmul(x, weight)
     if bias is not None:
         out = out + bias
