@@ -364,9 +364,46 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
 
   const int q_start = CAUSAL ? (kv0 / QB) * QB : 0;
 
+  // DMA staging maps for the natural [64][D] tiles (pre-swizzled source)
+  int n_row[QB * D / (NT * 8)], n_colp[QB * D / (NT * 8)];
+#pragma unroll
+  for (int it = 0; it < QB * D / (NT * 8); ++it) {
+    int flat = it * NT * 8 + tid * 8;
+    int row = flat / D, col = flat % D;
+    n_row[it] = row;
+    n_colp[it] = col ^ ((row & 7) << 3);
+  }
+
   for (int q0 = q_start; q0 < Sq; q0 += QB) {
-    // stage Q, dO natural + transposed
-    {
+    // stage Q, dO natural (async DMA when in-bounds) + transposed (reg)
+    if (q0 + QB <= Sq) {
+#pragma unroll
+      for (int it = 0; it < QB * D / (NT * 8); ++it) {
+        const short* qsrc = qg + qbase + (long long)(q0 + n_row[it]) * q_ss + n_colp[it];
+        const short* dsrc = dog + dobase + (long long)(q0 + n_row[it]) * do_ss + n_colp[it];
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)qsrc,
+            (__attribute__((address_space(3))) unsigned int*)(q_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
+            16, 0, 0);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)dsrc,
+            (__attribute__((address_space(3))) unsigned int*)(do_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
+            16, 0, 0);
+      }
+      constexpr int elems = QB * D;
+      const int rot = threadIdx.x & 7;
+      for (int flat = tid * 8; flat < elems; flat += NT * 8) {
+        int row = flat / D, col = flat % D;
+        shortx8 qv = *reinterpret_cast<const shortx8*>(qg + qbase + (long long)(q0 + row) * q_ss + col);
+        shortx8 dv = *reinterpret_cast<const shortx8*>(dog + dobase + (long long)(q0 + row) * do_ss + col);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int i = (j + rot) & 7;
+          *reinterpret_cast<short*>(qt_lds + lds_off(col + i, row * 2, TR_RS)) = qv[i];
+          *reinterpret_cast<short*>(dot_lds + lds_off(col + i, row * 2, TR_RS)) = dv[i];
+        }
+      }
+    } else {
       constexpr int elems = QB * D;
       for (int flat = tid * 8; flat < elems; flat += NT * 8) {
         int row = flat / D, col = flat % D;
@@ -568,8 +605,41 @@ __global__ void fa_bwd_dq_kernel(const short* __restrict__ dog, const short* __r
   for (int dt = 0; dt < NDT; ++dt) dq_acc[dt] = {0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = CAUSAL ? min(Skv, q0 + QEXT) : Skv;
+  int n_row[KB * D / (NT * 8)], n_colp[KB * D / (NT * 8)];
+#pragma unroll
+  for (int it = 0; it < KB * D / (NT * 8); ++it) {
+    int flat = it * NT * 8 + tid * 8;
+    int row = flat / D, col = flat % D;
+    n_row[it] = row;
+    n_colp[it] = col ^ ((row & 7) << 3);
+  }
   for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
-    {
+    if (kv0 + KB <= Skv) {
+#pragma unroll
+      for (int it = 0; it < KB * D / (NT * 8); ++it) {
+        const short* ksrc = kg + kvbase + (long long)(kv0 + n_row[it]) * k_ss + n_colp[it];
+        const short* vsrc = vg + kvbase + (long long)(kv0 + n_row[it]) * k_ss + n_colp[it];
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)ksrc,
+            (__attribute__((address_space(3))) unsigned int*)(k_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
+            16, 0, 0);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)vsrc,
+            (__attribute__((address_space(3))) unsigned int*)(v_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
+            16, 0, 0);
+      }
+      constexpr int elems = KB * D;
+      const int rot = threadIdx.x & 7;
+      for (int flat = tid * 8; flat < elems; flat += NT * 8) {
+        int row = flat / D, col = flat % D;
+        shortx8 kv_ = *reinterpret_cast<const shortx8*>(kg + kvbase + (long long)(kv0 + row) * k_ss + col);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int i = (j + rot) & 7;
+          *reinterpret_cast<short*>(kt_lds + lds_off(col + i, row * 2, TR_RS)) = kv_[i];
+        }
+      }
+    } else {
       constexpr int elems = KB * D;
       for (int flat = tid * 8; flat < elems; flat += NT * 8) {
         int row = flat / D, col = flat % D;
