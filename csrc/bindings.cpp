@@ -345,6 +345,25 @@ Tensor embedding_bwd(const Tensor& dout, const Tensor& ids_in, int64_t vocab,
   return dtable;
 }
 
+// ---- decode attention ------------------------------------------------------
+Tensor decode_attention(const Tensor& q, const Tensor& kcache, const Tensor& vcache,
+                        const Tensor& block_table, const Tensor& seq_lens,
+                        double scale) {
+  CHECK_IN(q); CHECK_IN(kcache); CHECK_IN(vcache);
+  TORCH_CHECK(q.dim() == 3, "q: [B, H, D] (one decode step)");
+  int64_t b = q.size(0), h = q.size(1), d = q.size(2);
+  TORCH_CHECK(kcache.dim() == 4, "k_cache: [nblocks, block_size, HKV, D]");
+  int64_t bs = kcache.size(1), hkv = kcache.size(2);
+  int64_t max_blocks = block_table.size(1);
+  auto o = torch::empty_like(q);
+  pa::decode_attention(q.const_data_ptr(), kcache.const_data_ptr(),
+                       vcache.const_data_ptr(),
+                       block_table.const_data_ptr<int>(),
+                       seq_lens.const_data_ptr<int>(), o.mutable_data_ptr(),
+                       b, h, hkv, bs, max_blocks, d, (float)scale, cur_stream());
+  return o;
+}
+
 // ---- hand-written GEMM ----------------------------------------------------
 Tensor gemm_bf16(const Tensor& a, const Tensor& b, bool b_is_nt) {
   TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16);
@@ -402,6 +421,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);
   m.def("gemm_bf16", &gemm_bf16);
+  m.def("decode_attention", &decode_attention);
   m.def("mfma_probe", &mfma_probe);
   m.def("mfma_probe32", &mfma_probe32);
   m.attr("compiled_arch") = "gfx950";

@@ -117,6 +117,13 @@ void embedding_bwd(const void* dout, const int64_t* ids, float* dtable,
                    int64_t n_ids, int64_t d, int64_t vocab, int64_t padding_idx,
                    int dtype, hipStream_t s);
 
+// ---- paged-KV decode attention (serving) ----------------------------------
+void decode_attention(const void* q, const void* kcache, const void* vcache,
+                      const int* block_table, const int* seq_lens, void* o,
+                      int64_t b, int64_t h, int64_t hkv, int64_t bs,
+                      int64_t max_blocks, int64_t dh, float scale,
+                      hipStream_t s);
+
 // ---- hand-written bf16 MFMA GEMM (C[m][n] = A[m][k] x B) ------------------
 // b_is_nt: B passed as Bt[n][k] row-major (fast path); else B[k][n].
 void gemm_bf16(const void* a, const void* b, void* c, int64_t m, int64_t n,

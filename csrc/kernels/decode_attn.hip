@@ -1,0 +1,156 @@
+// Paged-KV single-token decode attention for gfx950 (serving path).
+//
+// Reference role parity: paddle/phi/kernels/fusion/gpu/
+// masked_multihead_attention_kernel.cu + block_multi_head_attention
+// (paged "block" KV cache) -- re-derived for wave64:
+//   * one 256-thread block per (batch, q-head); GQA via head mapping
+//   * KV streamed from a paged cache: k/v_cache [nblocks, bs, hkv, D],
+//     block_table [B, max_blocks] -- HBM-bound, 16B/lane loads
+//   * 16 lanes per position (D<=128: 8 elems/lane), 4 pos/wave/step,
+//     online softmax in registers, cross-wave merge via LDS
+#include "common.h"
+#include "api.h"
+
+namespace pa {
+
+template <int D>
+__launch_bounds__(256)
+__global__ void decode_attn_kernel(const short* __restrict__ qg,
+                                   const short* __restrict__ kcache,
+                                   const short* __restrict__ vcache,
+                                   const int* __restrict__ block_table,
+                                   const int* __restrict__ seq_lens,
+                                   short* __restrict__ og,
+                                   int B, int H, int HKV, int bs,
+                                   int max_blocks, float scale) {
+  const int b = blockIdx.x / H;
+  const int h = blockIdx.x % H;
+  const int hkv = h / (H / HKV);
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l16 = lane & 15;   // element group within D
+  const int lp = lane >> 4;    // 0..3: position slot within wave
+  const int S = seq_lens[b];
+
+  // q for this (b,h): D elems; 16-lane group covers D -> E = D/16 per lane
+  constexpr int E = D / 16;
+  float qv[E];
+  {
+    const short* qp = qg + ((long long)b * H + h) * D;
+#pragma unroll
+    for (int i = 0; i < E; ++i) qv[i] = bf2f(qp[l16 * E + i]);
+  }
+
+  float m = -INFINITY, l = 0.f;
+  float oacc[E];
+#pragma unroll
+  for (int i = 0; i < E; ++i) oacc[i] = 0.f;
+
+  // each wave processes 4 positions per step; 4 waves -> 16 pos/block/step
+  for (int pos0 = wid * 4; pos0 < S; pos0 += 16) {
+    int pos = pos0 + lp;
+    bool ok = pos < S;
+    int phys = 0, off = 0;
+    if (ok) {
+      phys = block_table[(long long)b * max_blocks + pos / bs];
+      off = pos % bs;
+    }
+    const short* kp = kcache + (((long long)phys * bs + off) * HKV + hkv) * D;
+    const short* vp = vcache + (((long long)phys * bs + off) * HKV + hkv) * D;
+    float sc = 0.f;
+    if (ok) {
+      if (E == 8) {
+        shortx8 kv8 = *reinterpret_cast<const shortx8*>(kp + l16 * E);
+#pragma unroll
+        for (int i = 0; i < E; ++i) sc += qv[i] * bf2f(kv8[i]);
+      } else {
+        shortx4 kv4 = *reinterpret_cast<const shortx4*>(kp + l16 * E);
+#pragma unroll
+        for (int i = 0; i < E; ++i) sc += qv[i] * bf2f(kv4[i & 3]);
+      }
+    }
+    // reduce over the 16-lane group
+    sc = group16_reduce(sc, SumOp());
+    sc = ok ? sc * scale : -INFINITY;
+    // online softmax across the 4 position-slots handled by this wave-step
+    float m_new = fmaxf(m, sc);
+    // also fold the other slots' maxima later; here per-lane-slot state:
+    float corr = (m == -INFINITY) ? 0.f : __expf(m - m_new);
+    float p = (sc == -INFINITY) ? 0.f : __expf(sc - m_new);
+    l = l * corr + p;
+    m = m_new;
+    float vvf[E];
+#pragma unroll
+    for (int i = 0; i < E; ++i) vvf[i] = 0.f;
+    if (ok) {
+      if (E == 8) {
+        shortx8 vv8 = *reinterpret_cast<const shortx8*>(vp + l16 * E);
+#pragma unroll
+        for (int i = 0; i < E; ++i) vvf[i] = bf2f(vv8[i]);
+      } else {
+        shortx4 vv4 = *reinterpret_cast<const shortx4*>(vp + l16 * E);
+#pragma unroll
+        for (int i = 0; i < E; ++i) vvf[i] = bf2f(vv4[i & 3]);
+      }
+    }
+#pragma unroll
+    for (int i = 0; i < E; ++i) oacc[i] = oacc[i] * corr + p * vvf[i];
+  }
+
+  // ---- merge the 16 position-slot partials (4 waves x 4 slots) ----------
+  // each (wid, lp) slot has (m, l, oacc[8]) per l16 group
+  __shared__ float sm[16], sl[16];
+  __shared__ float so[16][D];
+  const int slot_id = wid * 4 + lp;
+  if (l16 == 0) {
+    sm[slot_id] = m;
+    sl[slot_id] = l;
+  }
+#pragma unroll
+  for (int i = 0; i < E; ++i) so[slot_id][l16 * E + i] = oacc[i];
+  __syncthreads();
+  if (tid < 64) {
+    // wave 0 merges 16 slots
+    float gm = -INFINITY;
+#pragma unroll
+    for (int s2 = 0; s2 < 16; ++s2) gm = fmaxf(gm, sm[s2]);
+    float gl = 0.f;
+    float out[E];
+#pragma unroll
+    for (int i = 0; i < E; ++i) out[i] = 0.f;
+#pragma unroll
+    for (int s2 = 0; s2 < 16; ++s2) {
+      float w = (sm[s2] == -INFINITY) ? 0.f : __expf(sm[s2] - gm);
+      gl += sl[s2] * w;
+#pragma unroll
+      for (int i = 0; i < E; ++i) out[i] += w * so[s2][l16 * E + i];
+    }
+    if (lane < 16) {
+      float inv = gl > 0.f ? 1.f / gl : 0.f;
+      short* op = og + ((long long)b * H + h) * D;
+#pragma unroll
+      for (int i = 0; i < E; ++i) op[l16 * E + i] = f2bf(out[i] * inv);
+    }
+  }
+}
+
+void decode_attention(const void* q, const void* kcache, const void* vcache,
+                      const int* block_table, const int* seq_lens, void* o,
+                      int64_t b, int64_t h, int64_t hkv, int64_t bs,
+                      int64_t max_blocks, int64_t dh, float scale,
+                      hipStream_t s) {
+  dim3 grid((unsigned)(b * h));
+  if (dh == 128)
+    hipLaunchKernelGGL((decode_attn_kernel<128>), grid, dim3(256), 0, s,
+                       (const short*)q, (const short*)kcache, (const short*)vcache,
+                       block_table, seq_lens, (short*)o, (int)b, (int)h,
+                       (int)hkv, (int)bs, (int)max_blocks, scale);
+  else
+    hipLaunchKernelGGL((decode_attn_kernel<64>), grid, dim3(256), 0, s,
+                       (const short*)q, (const short*)kcache, (const short*)vcache,
+                       block_table, seq_lens, (short*)o, (int)b, (int)h,
+                       (int)hkv, (int)bs, (int)max_blocks, scale);
+}
+
+}  // namespace pa

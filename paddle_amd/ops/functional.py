@@ -655,3 +655,36 @@ def l2_norm_squared(x):
         C = _ext.get_ext()
         return C.l2norm_sq(x.contiguous())
     return x.float().square().sum().reshape(1)
+
+
+# ---------------------------------------------------------------------------
+# paged-KV decode attention (serving; block_multihead_attention parity)
+# ---------------------------------------------------------------------------
+def paged_decode_attention(q, k_cache, v_cache, block_table, seq_lens, scale=None):
+    """One decode step. q: [B, H, D]; k/v_cache: [nblocks, block_size, HKV, D];
+    block_table: int32 [B, max_blocks]; seq_lens: int32 [B]."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _ext.use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+        C = _ext.get_ext()
+        return C.decode_attention(q.contiguous(), k_cache.contiguous(),
+                                  v_cache.contiguous(), block_table.int().contiguous(),
+                                  seq_lens.int().contiguous(), scale)
+    # reference: gather each sequence's KV then plain attention
+    B, H, D = q.shape
+    HKV = k_cache.shape[2]
+    bs = k_cache.shape[1]
+    rep = H // HKV
+    out = torch.empty_like(q)
+    for b in range(B):
+        S = int(seq_lens[b])
+        nb = (S + bs - 1) // bs
+        blocks = block_table[b, :nb].long()
+        k = k_cache[blocks].reshape(-1, HKV, D)[:S]  # [S, HKV, D]
+        v = v_cache[blocks].reshape(-1, HKV, D)[:S]
+        kf = k.float().repeat_interleave(rep, dim=1)   # [S, H, D]
+        vf = v.float().repeat_interleave(rep, dim=1)
+        s = torch.einsum("hd,shd->hs", q[b].float(), kf) * scale
+        p = torch.softmax(s, dim=-1)
+        out[b] = torch.einsum("hs,shd->hd", p, vf).to(q.dtype)
+    return out

@@ -231,3 +231,21 @@ def test_colsum_gpu():
     x = torch.randn(512, 1024, device=DEV, dtype=torch.bfloat16)
     out = C.colsum(x)
     torch.testing.assert_close(out, x.float().sum(0), rtol=1e-2, atol=0.5)
+
+
+def test_paged_decode_attention_gpu():
+    torch.manual_seed(7)
+    B, H, HKV, D, bs, max_blocks = 4, 8, 2, 128, 16, 16
+    nblocks = 64
+    q = torch.randn(B, H, D, device=DEV, dtype=torch.bfloat16)
+    kc = torch.randn(nblocks, bs, HKV, D, device=DEV, dtype=torch.bfloat16)
+    vc = torch.randn_like(kc)
+    bt = torch.randperm(nblocks, device=DEV)[: B * max_blocks].reshape(B, max_blocks).int()
+    seq = torch.tensor([7, 16, 100, 250], device=DEV, dtype=torch.int32)
+    out = hot.paged_decode_attention(q, kc, vc, bt, seq)
+    # reference path (force fallback)
+    import paddle_amd
+    paddle_amd.set_flags({"FLAGS_use_native_kernels": False})
+    ref = hot.paged_decode_attention(q, kc, vc, bt, seq)
+    paddle_amd.set_flags({"FLAGS_use_native_kernels": True})
+    _assert_close_bf16(out, ref.float(), atol=3e-2, rtol=3e-2)
