@@ -42,7 +42,7 @@ __global__ void decode_attn_kernel(const short* __restrict__ qg,
     for (int i = 0; i < E; ++i) qv[i] = bf2f(qp[l16 * E + i]);
   }
 
-  float m = -INFINITY, l = 0.f;
+  float m = -1e30f, l = 0.f;  // finite sentinel: -ffast-math breaks +-inf
   float oacc[E];
 #pragma unroll
   for (int i = 0; i < E; ++i) oacc[i] = 0.f;
@@ -72,12 +72,11 @@ __global__ void decode_attn_kernel(const short* __restrict__ qg,
     }
     // reduce over the 16-lane group
     sc = group16_reduce(sc, SumOp());
-    sc = ok ? sc * scale : -INFINITY;
+    sc = ok ? sc * scale : -1e30f;
     // online softmax across the 4 position-slots handled by this wave-step
     float m_new = fmaxf(m, sc);
-    // also fold the other slots' maxima later; here per-lane-slot state:
-    float corr = (m == -INFINITY) ? 0.f : __expf(m - m_new);
-    float p = (sc == -INFINITY) ? 0.f : __expf(sc - m_new);
+    float corr = __expf(m - m_new);       // underflows to 0 for the sentinel
+    float p = ok ? __expf(sc - m_new) : 0.f;
     l = l * corr + p;
     m = m_new;
     float vvf[E];
@@ -112,7 +111,7 @@ __global__ void decode_attn_kernel(const short* __restrict__ qg,
   __syncthreads();
   if (tid < 64) {
     // wave 0 merges 16 slots
-    float gm = -INFINITY;
+    float gm = -1e30f;
 #pragma unroll
     for (int s2 = 0; s2 < 16; ++s2) gm = fmaxf(gm, sm[s2]);
     float gl = 0.f;
@@ -121,7 +120,7 @@ __global__ void decode_attn_kernel(const short* __restrict__ qg,
     for (int i = 0; i < E; ++i) out[i] = 0.f;
 #pragma unroll
     for (int s2 = 0; s2 < 16; ++s2) {
-      float w = (sm[s2] == -INFINITY) ? 0.f : __expf(sm[s2] - gm);
+      float w = (sl[s2] > 0.f) ? __expf(sm[s2] - gm) : 0.f;
       gl += sl[s2] * w;
 #pragma unroll
       for (int i = 0; i < E; ++i) out[i] += w * so[s2][l16 * E + i];
