@@ -227,6 +227,7 @@ from . import inference  # noqa: F401
 from . import quantization  # noqa: F401
 from . import text  # noqa: F401
 from . import audio  # noqa: F401
+from . import onnx  # noqa: F401
 from .framework_io import save, load, async_save  # noqa: F401
 from .hapi import Model  # noqa: F401
 from .param_attr import ParamAttr  # noqa: F401

@@ -1,0 +1,87 @@
+"""Reference-style OpTest numeric-gradient checks on the CPU oracle
+implementations of the hot ops (SURVEY §4 test-strategy mirror)."""
+import numpy as np
+import torch
+
+from op_test import OpTest
+
+from paddle_amd.ops import functional as hot
+
+
+class TestLayerNormOp(OpTest):
+    rtol, atol = 1e-5, 1e-6
+
+    def __init__(self):
+        torch.manual_seed(0)
+
+    def make_inputs(self):
+        g = torch.Generator().manual_seed(3)
+        return [torch.randn(3, 16, dtype=torch.float64, generator=g),
+                torch.randn(16, dtype=torch.float64, generator=g),
+                torch.randn(16, dtype=torch.float64, generator=g)]
+
+    @staticmethod
+    def fn(x, w, b):
+        return hot.layer_norm(x.float(), w.float(), b.float(), 1e-5).double()
+
+    @staticmethod
+    def oracle(x, w, b):
+        xn = x.numpy()
+        mu = xn.mean(-1, keepdims=True)
+        var = xn.var(-1, keepdims=True)
+        return torch.from_numpy((xn - mu) / np.sqrt(var + 1e-5) * w.numpy() + b.numpy())
+
+
+class TestRMSNormOp(OpTest):
+    rtol, atol = 1e-5, 1e-6
+
+    def make_inputs(self):
+        g = torch.Generator().manual_seed(4)
+        return [torch.randn(3, 16, dtype=torch.float64, generator=g),
+                torch.randn(16, dtype=torch.float64, generator=g)]
+
+    @staticmethod
+    def fn(x, w):
+        return hot.rms_norm(x.float(), w.float(), 1e-6).double()
+
+    @staticmethod
+    def oracle(x, w):
+        xn = x.numpy()
+        return torch.from_numpy(xn / np.sqrt((xn ** 2).mean(-1, keepdims=True) + 1e-6)
+                                * w.numpy())
+
+
+class TestSwigluOp(OpTest):
+    rtol, atol = 1e-5, 1e-6
+
+    def make_inputs(self):
+        g = torch.Generator().manual_seed(5)
+        return [torch.randn(4, 16, dtype=torch.float64, generator=g)]
+
+    @staticmethod
+    def fn(x):
+        return hot.swiglu(x.float()).double()
+
+    @staticmethod
+    def oracle(x):
+        xn = x.numpy()
+        gg, u = xn[:, :8], xn[:, 8:]
+        return torch.from_numpy(gg / (1 + np.exp(-gg)) * u)
+
+
+def test_layer_norm_optest():
+    t = TestLayerNormOp()
+    t.check_output()
+    t.check_grad()
+
+
+def test_rms_norm_optest():
+    t = TestRMSNormOp()
+    t.check_output()
+    t.check_grad()
+
+
+def test_swiglu_optest():
+    t = TestSwigluOp()
+    t.check_output()
+    t.check_grad()
