@@ -25,6 +25,12 @@ def export(layer, path, input_spec=None, opset_version=17, **configs):
     if d:
         os.makedirs(d, exist_ok=True)
     layer.eval()
+    try:
+        import onnx as _onnx  # noqa: F401
+    except ImportError as e:
+        raise RuntimeError(
+            "paddle.onnx.export requires the `onnx` package, which is not "
+            "installed in this offline environment") from e
     torch.onnx.export(layer, tuple(examples), out, opset_version=opset_version,
                       dynamo=False)
     return out
