@@ -232,23 +232,29 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
     for (int ks = 0; ks < KB / 16; ++ks) {
       const int t = ks >> 1;
       const int kp = ks & 1;
-      const int uo = 2 * kp + hi;        // own reg-group
-      const int us = 2 * kp + 1 - hi;    // supply reg-group (partner's hi)
-      int O1 = pack_bf2(st[t][4 * uo], st[t][4 * uo + 1]);
-      int O2 = pack_bf2(st[t][4 * uo + 2], st[t][4 * uo + 3]);
-      int S1 = pack_bf2(st[t][4 * us], st[t][4 * us + 1]);
-      int S2 = pack_bf2(st[t][4 * us + 2], st[t][4 * us + 3]);
+      // STATIC register indices only (rule #20: a runtime `hi` index into
+      // the v16 accumulator becomes a 16-way cndmask tree); pack all four
+      // candidate words, then one select each by hi.
+      const int b0 = 8 * kp;
+      int P0 = pack_bf2(st[t][b0 + 0], st[t][b0 + 1]);
+      int P1 = pack_bf2(st[t][b0 + 2], st[t][b0 + 3]);
+      int P2 = pack_bf2(st[t][b0 + 4], st[t][b0 + 5]);
+      int P3 = pack_bf2(st[t][b0 + 6], st[t][b0 + 7]);
+      int O1 = hi ? P2 : P0;             // own reg-group words
+      int O2 = hi ? P3 : P1;
+      int S1 = hi ? P0 : P2;             // supply (partner's) words
+      int S2 = hi ? P1 : P3;
       intx2 ra = __builtin_amdgcn_permlane32_swap(S1, S2, false, false);
       intx2 rb = __builtin_amdgcn_permlane32_swap(S2, S1, false, false);
       // lanes<32: partner supply = (ra[1], rb[1]); lanes>=32: (rb[0], ra[0])
-      int P1 = hi ? rb[0] : ra[1];
-      int P2 = hi ? ra[0] : rb[1];
+      int X1 = hi ? rb[0] : ra[1];
+      int X2 = hi ? ra[0] : rb[1];
       // A-frag words in kv order: hi==0 -> {own, partner}; hi==1 -> {partner, own}
       intx4 paw;
-      paw[0] = hi ? P1 : O1;
-      paw[1] = hi ? P2 : O2;
-      paw[2] = hi ? O1 : P1;
-      paw[3] = hi ? O2 : P2;
+      paw[0] = hi ? X1 : O1;
+      paw[1] = hi ? X2 : O2;
+      paw[2] = hi ? O1 : X1;
+      paw[3] = hi ? O2 : X2;
       shortx8 pa = *reinterpret_cast<shortx8*>(&paw);
 #pragma unroll
       for (int dt = 0; dt < NDT; ++dt) {
