@@ -54,6 +54,14 @@ from .loss import (  # noqa: F401
     NLLLoss,
     SmoothL1Loss,
 )
+from .rnn import (  # noqa: F401
+    GRU,
+    GRUCell,
+    LSTM,
+    LSTMCell,
+    SimpleRNN,
+    SimpleRNNCell,
+)
 from .transformer import (  # noqa: F401
     MultiHeadAttention,
     Transformer,
