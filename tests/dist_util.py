@@ -24,7 +24,7 @@ def free_port():
 
 
 _RETRYABLE = ("Address already in use", "EADDRINUSE", "Connection refused",
-              "Connection reset")
+              "Connection reset", "terminate called without an active exception")
 
 
 def run_dist(script_body: str, world_size: int = 2, timeout: int = 240,
@@ -34,6 +34,13 @@ def run_dist(script_body: str, world_size: int = 2, timeout: int = 240,
     Retries once on rendezvous races (port TIME_WAIT etc.)."""
     port = free_port()
     script = textwrap.dedent(script_body)
+    # clean teardown: gloo subgroup destructors race at interpreter exit
+    # ("terminate called without an active exception") without this
+    script += (
+        "\n\nimport torch.distributed as _dist\n"
+        "if _dist.is_available() and _dist.is_initialized():\n"
+        "    _dist.barrier()\n"
+        "    _dist.destroy_process_group()\n")
     procs = []
     for rank in range(world_size):
         env = dict(os.environ)
