@@ -91,6 +91,12 @@ void flash_attn_fwd32(const void* q, const void* k, const void* v, void* o,
                       int64_t skv, int64_t dh, float scale, bool causal,
                       const int64_t* qs, const int64_t* ks, const int64_t* os,
                       hipStream_t s);
+void flash_attn_bwd_dq32(const void* dout, const void* q, const void* k,
+                         const void* v, const float* lse, const float* delta,
+                         void* dq, int64_t b, int64_t h, int64_t sq, int64_t skv,
+                         int64_t dh, float scale, bool causal,
+                         const int64_t* qs, const int64_t* ks, const int64_t* dos,
+                         const int64_t* dqs, hipStream_t s);
 void flash_attn_bwd(const void* dout, const void* q, const void* k,
                     const void* v, const void* o, const float* lse,
                     void* dq, void* dk, void* dv, float* delta, int64_t b,

@@ -302,4 +302,240 @@ void flash_attn_fwd32(const void* q, const void* k, const void* v, void* o,
 #undef FAF
 }
 
+
+
+// ---------------------------------------------------------------------------
+// backward dQ, 32x32 variant: same swapped layout as the forward -- S^T and
+// dP^T put the q-COLUMN in each lane (per-lane lse/delta, no broadcasts),
+// and the dS A-fragments for dQ += dS.K are assembled with the identical
+// static-pack + permlane32_swap exchange (no LDS round trip).
+// K/V staged natural via async DMA; K^T staged with rotated writes for the
+// dQ B-fragments.
+// ---------------------------------------------------------------------------
+namespace {
+constexpr int DQ_NW = 4;
+}
+
+template <int D, bool CAUSAL>
+__launch_bounds__(256, 2)
+__global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* __restrict__ qg,
+                                   const short* __restrict__ kg, const short* __restrict__ vg,
+                                   const float* __restrict__ lseg, const float* __restrict__ deltag,
+                                   short* __restrict__ dqg, int B, int H, int Sq, int Skv,
+                                   float scale,
+                                   long long q_sb, long long q_sh, long long q_ss,
+                                   long long k_sb, long long k_sh, long long k_ss,
+                                   long long do_sb, long long do_sh, long long do_ss,
+                                   long long dq_sb, long long dq_sh, long long dq_ss) {
+  constexpr int NW = DQ_NW;
+  constexpr int NT = NW * 64;
+  constexpr int QB = NW * 32;       // 128 q rows / block
+  constexpr int KB = 64;
+  constexpr int NKS = D / 16;
+  constexpr int NDT = D / 32;
+  constexpr unsigned NAT_RS = D * 2;
+  constexpr unsigned KT_RS = KB * 2;
+  __shared__ char k_lds[KB * D * 2];
+  __shared__ char v_lds[KB * D * 2];
+  __shared__ char kt_lds[D * KB * 2];
+
+  const int qblk = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int q0 = qblk * QB;
+  const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
+  const long long dobase = (long long)b * do_sb + (long long)h * do_sh;
+  const long long dqbase = (long long)b * dq_sb + (long long)h * dq_sh;
+  const long long kvbase = (long long)b * k_sb + (long long)h * k_sh;
+  const long long lse_base = ((long long)bh) * Sq;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wq = tid >> 6;
+  const int l32 = lane & 31;
+  const int hi = lane >> 5;
+  const int q0w = q0 + wq * 32;
+
+  // Q and dO as B-fragments (lane holds the q column l32)
+  shortx8 qf[NKS], dof[NKS];
+  {
+    int row = q0w + l32;
+    bool ok = row < Sq;
+#pragma unroll
+    for (int ks = 0; ks < NKS; ++ks) {
+      if (ok) {
+        qf[ks] = *reinterpret_cast<const shortx8*>(
+            qg + qbase + (long long)row * q_ss + ks * 16 + hi * 8);
+        dof[ks] = *reinterpret_cast<const shortx8*>(
+            dog + dobase + (long long)row * do_ss + ks * 16 + hi * 8);
+      } else {
+        for (int i = 0; i < 8; ++i) { qf[ks][i] = 0; dof[ks][i] = 0; }
+      }
+    }
+  }
+  const int q_abs = q0w + l32;
+  const float lse_v = (q_abs < Sq) ? lseg[lse_base + q_abs] : 1e30f;
+  const float delta_v = (q_abs < Sq) ? deltag[lse_base + q_abs] : 0.f;
+
+  floatx16 dq_acc[NDT];
+#pragma unroll
+  for (int dt = 0; dt < NDT; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dq_acc[dt][r] = 0.f;
+
+  // DMA staging maps
+  int n_row[KB * D / (NT * 8)], n_colp[KB * D / (NT * 8)];
+#pragma unroll
+  for (int it = 0; it < KB * D / (NT * 8); ++it) {
+    int flat = it * NT * 8 + tid * 8;
+    int row = flat / D, col = flat % D;
+    n_row[it] = row;
+    n_colp[it] = col ^ ((row & 7) << 3);
+  }
+
+  const int kv_end = CAUSAL ? min(Skv, q0 + QB) : Skv;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+    if (kv0 + KB <= Skv) {
+#pragma unroll
+      for (int it = 0; it < KB * D / (NT * 8); ++it) {
+        const short* ksrc = kg + kvbase + (long long)(kv0 + n_row[it]) * k_ss + n_colp[it];
+        const short* vsrc = vg + kvbase + (long long)(kv0 + n_row[it]) * k_ss + n_colp[it];
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)ksrc,
+            (__attribute__((address_space(3))) unsigned int*)(k_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
+            16, 0, 0);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)vsrc,
+            (__attribute__((address_space(3))) unsigned int*)(v_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
+            16, 0, 0);
+      }
+      const int rot = tid & 7;
+      for (int flat = tid * 8; flat < KB * D; flat += NT * 8) {
+        int row = flat / D, col = flat % D;
+        shortx8 kv_ = *reinterpret_cast<const shortx8*>(
+            kg + kvbase + (long long)(kv0 + row) * k_ss + col);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int i = (j + rot) & 7;
+          *reinterpret_cast<short*>(kt_lds + lds_off32(col + i, row * 2, KT_RS)) = kv_[i];
+        }
+      }
+    } else {
+      for (int flat = tid * 8; flat < KB * D; flat += NT * 8) {
+        int row = flat / D, col = flat % D;
+        shortx8 kv_, vv;
+        if (kv0 + row < Skv) {
+          kv_ = *reinterpret_cast<const shortx8*>(kg + kvbase + (long long)(kv0 + row) * k_ss + col);
+          vv = *reinterpret_cast<const shortx8*>(vg + kvbase + (long long)(kv0 + row) * k_ss + col);
+        } else {
+          for (int i = 0; i < 8; ++i) { kv_[i] = 0; vv[i] = 0; }
+        }
+        *reinterpret_cast<shortx8*>(k_lds + lds_off32(row, col * 2, NAT_RS)) = kv_;
+        *reinterpret_cast<shortx8*>(v_lds + lds_off32(row, col * 2, NAT_RS)) = vv;
+        const int rot = tid & 7;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int i = (j + rot) & 7;
+          *reinterpret_cast<short*>(kt_lds + lds_off32(col + i, row * 2, KT_RS)) = kv_[i];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- S^T = K Q^T and dP^T = V dO^T ------------------------------------
+    floatx16 st[2], dp[2];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) { st[t][r] = 0.f; dp[t][r] = 0.f; }
+#pragma unroll
+      for (int ks = 0; ks < NKS; ++ks) {
+        shortx8 kf = *reinterpret_cast<const shortx8*>(
+            k_lds + lds_off32(t * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
+        st[t] = mfma32_bf16(kf, qf[ks], st[t]);
+        shortx8 vf = *reinterpret_cast<const shortx8*>(
+            v_lds + lds_off32(t * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
+        dp[t] = mfma32_bf16(vf, dof[ks], dp[t]);
+      }
+    }
+
+    // ---- dS^T = P^T (dP^T - delta) * scale --------------------------------
+    const bool bnd = (kv0 + KB > Skv) || (CAUSAL && kv0 + KB > q_abs);
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float p = __expf(st[t][r] * scale - lse_v);
+        if (bnd) {
+          int kv_abs = kv0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          if (kv_abs >= Skv || (CAUSAL && kv_abs > q_abs)) p = 0.f;
+        }
+        st[t][r] = p * (dp[t][r] - delta_v) * scale;
+      }
+
+    // ---- dQ += dS K (A-frags via static packs + permlane exchange) --------
+#pragma unroll
+    for (int ks = 0; ks < KB / 16; ++ks) {
+      const int t = ks >> 1;
+      const int kp = ks & 1;
+      const int b0 = 8 * kp;
+      int P0 = pack_bf2(st[t][b0 + 0], st[t][b0 + 1]);
+      int P1 = pack_bf2(st[t][b0 + 2], st[t][b0 + 3]);
+      int P2 = pack_bf2(st[t][b0 + 4], st[t][b0 + 5]);
+      int P3 = pack_bf2(st[t][b0 + 6], st[t][b0 + 7]);
+      int O1 = hi ? P2 : P0;
+      int O2 = hi ? P3 : P1;
+      int S1 = hi ? P0 : P2;
+      int S2 = hi ? P1 : P3;
+      intx2 ra = __builtin_amdgcn_permlane32_swap(S1, S2, false, false);
+      intx2 rb = __builtin_amdgcn_permlane32_swap(S2, S1, false, false);
+      int X1 = hi ? rb[0] : ra[1];
+      int X2 = hi ? ra[0] : rb[1];
+      intx4 paw;
+      paw[0] = hi ? X1 : O1;
+      paw[1] = hi ? X2 : O2;
+      paw[2] = hi ? O1 : X1;
+      paw[3] = hi ? O2 : X2;
+      shortx8 pa = *reinterpret_cast<shortx8*>(&paw);
+#pragma unroll
+      for (int dt = 0; dt < NDT; ++dt) {
+        shortx8 bf = *reinterpret_cast<const shortx8*>(
+            kt_lds + lds_off32(dt * 32 + l32, (ks * 16 + hi * 8) * 2, KT_RS));
+        dq_acc[dt] = mfma32_bf16(pa, bf, dq_acc[dt]);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue -----------------------------------------------------------
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int rq = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    int row = q0w + rq;
+    if (row >= Sq) continue;
+#pragma unroll
+    for (int dt = 0; dt < NDT; ++dt)
+      dqg[dqbase + (long long)row * dq_ss + dt * 32 + l32] = f2bf(dq_acc[dt][r]);
+  }
+}
+
+void flash_attn_bwd_dq32(const void* dout, const void* q, const void* k,
+                         const void* v, const float* lse, const float* delta,
+                         void* dq, int64_t b, int64_t h, int64_t sq, int64_t skv,
+                         int64_t dh, float scale, bool causal,
+                         const int64_t* qs, const int64_t* ks, const int64_t* dos,
+                         const int64_t* dqs, hipStream_t s) {
+  dim3 grid((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
+  dim3 blk(256);
+#define FDQ(D, C)                                                               \
+  hipLaunchKernelGGL((fa_bwd_dq32_kernel<D, C>), grid, blk, 0, s,               \
+                     (const short*)dout, (const short*)q, (const short*)k,      \
+                     (const short*)v, lse, delta, (short*)dq, (int)b, (int)h,   \
+                     (int)sq, (int)skv, scale, qs[0], qs[1], qs[2],             \
+                     ks[0], ks[1], ks[2], dos[0], dos[1], dos[2],               \
+                     dqs[0], dqs[1], dqs[2])
+  if (dh == 128) { if (causal) FDQ(128, true); else FDQ(128, false); }
+  else           { if (causal) FDQ(64, true);  else FDQ(64, false); }
+#undef FDQ
+}
 }  // namespace pa

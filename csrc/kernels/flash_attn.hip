@@ -773,12 +773,8 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
                        (int)b, (int)h, (int)sq, (int)skv, scale,               \
                        qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],               \
                        dos[0], dos[1], dos[2], dks[0], dks[1], dks[2]);        \
-    hipLaunchKernelGGL((fa_bwd_dq_kernel<D, C>), gq, dim3(512), 0, s,          \
-                       (const short*)dout, (const short*)q, (const short*)k,   \
-                       (const short*)v, lse, delta, (short*)dq, (int)b,        \
-                       (int)h, (int)sq, (int)skv, scale,                       \
-                       qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],               \
-                       dos[0], dos[1], dos[2], dqs[0], dqs[1], dqs[2]);        \
+    flash_attn_bwd_dq32(dout, q, k, v, lse, delta, dq, b, h, sq, skv, dh,     \
+                        scale, causal, qs, ks, dos, dqs, s);                   \
   } while (0)
   if (dh == 128) { if (causal) FAB(128, true); else FAB(128, false); }
   else           { if (causal) FAB(64, true);  else FAB(64, false); }
