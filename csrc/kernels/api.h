@@ -97,6 +97,12 @@ void flash_attn_bwd_dq32(const void* dout, const void* q, const void* k,
                          int64_t dh, float scale, bool causal,
                          const int64_t* qs, const int64_t* ks, const int64_t* dos,
                          const int64_t* dqs, hipStream_t s);
+void flash_attn_bwd_dkv32(const void* dout, const void* q, const void* k,
+                          const void* v, const float* lse, const float* delta,
+                          void* dk, void* dv, int64_t b, int64_t h, int64_t sq,
+                          int64_t skv, int64_t dh, float scale, bool causal,
+                          const int64_t* qs, const int64_t* ks, const int64_t* dos,
+                          const int64_t* dks, hipStream_t s);
 void flash_attn_bwd(const void* dout, const void* q, const void* k,
                     const void* v, const void* o, const float* lse,
                     void* dq, void* dk, void* dv, float* delta, int64_t b,

@@ -538,4 +538,240 @@ void flash_attn_bwd_dq32(const void* dout, const void* q, const void* k,
   else           { if (causal) FDQ(64, true);  else FDQ(64, false); }
 #undef FDQ
 }
+
+// ---------------------------------------------------------------------------
+// backward dV / dK, 32x32 swapped variants.  Split into two kernels so each
+// fits the 2-waves/SIMD register budget (combined dK+dV needs ~290 VGPRs).
+// Wave owns 32 kv rows; q-COLUMN per lane => per-lane lse/delta; the P^T /
+// dS^T A-fragments for the [kv x q] x [q x d] products come straight from
+// the accumulator via the static-pack + permlane exchange (no LDS trip).
+// ---------------------------------------------------------------------------
+template <int D, bool CAUSAL, bool IS_DK>
+__launch_bounds__(256, 2)
+__global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* __restrict__ qg,
+                                    const short* __restrict__ kg, const short* __restrict__ vg,
+                                    const float* __restrict__ lseg, const float* __restrict__ deltag,
+                                    short* __restrict__ outg, int B, int H, int Sq, int Skv,
+                                    float scale,
+                                    long long q_sb, long long q_sh, long long q_ss,
+                                    long long k_sb, long long k_sh, long long k_ss,
+                                    long long do_sb, long long do_sh, long long do_ss,
+                                    long long dk_sb, long long dk_sh, long long dk_ss) {
+  constexpr int NW = 4;
+  constexpr int NT = NW * 64;
+  constexpr int KVB = NW * 32;     // 128 kv rows / block
+  constexpr int QT = 64;           // q tile
+  constexpr int NKS = D / 16;
+  constexpr int NDT = D / 32;
+  constexpr unsigned NAT_RS = D * 2;
+  constexpr unsigned TR_RS = QT * 2;
+  __shared__ char q_lds[QT * D * 2];        // natural (for S^T B-frags)
+  __shared__ char t_lds[D * QT * 2];        // dO^T (dV) or Q^T (dK)
+  __shared__ char d_lds[IS_DK ? QT * D * 2 : 1];  // dO natural (dK only)
+  __shared__ char ps_lds[NW * 32 * 32 * 2]; // per-wave P^T/dS^T round trip
+
+  const int kvblk = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int kv0 = kvblk * KVB;
+  const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
+  const long long dobase = (long long)b * do_sb + (long long)h * do_sh;
+  const long long kvbase = (long long)b * k_sb + (long long)h * k_sh;
+  const long long outbase = (long long)b * dk_sb + (long long)h * dk_sh;
+  const long long lse_base = ((long long)bh) * Sq;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wq = tid >> 6;
+  const int l32 = lane & 31;
+  const int hi = lane >> 5;
+  const int kv0w = kv0 + wq * 32;
+
+  // K (and V for dK) A-fragments: lane = kv row kv0w + l32
+  shortx8 kf[NKS], vf[IS_DK ? NKS : 1];
+  {
+    int row = kv0w + l32;
+    bool ok = row < Skv;
+#pragma unroll
+    for (int ks = 0; ks < NKS; ++ks) {
+      if (ok) {
+        kf[ks] = *reinterpret_cast<const shortx8*>(
+            kg + kvbase + (long long)row * k_ss + ks * 16 + hi * 8);
+        if (IS_DK)
+          vf[ks] = *reinterpret_cast<const shortx8*>(
+              vg + kvbase + (long long)row * k_ss + ks * 16 + hi * 8);
+      } else {
+        for (int i = 0; i < 8; ++i) { kf[ks][i] = 0; if (IS_DK) vf[ks][i] = 0; }
+      }
+    }
+  }
+
+  floatx16 acc[NDT];
+#pragma unroll
+  for (int dt = 0; dt < NDT; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) acc[dt][r] = 0.f;
+
+  int n_row[QT * D / (NT * 8)], n_colp[QT * D / (NT * 8)];
+#pragma unroll
+  for (int it = 0; it < QT * D / (NT * 8); ++it) {
+    int flat = it * NT * 8 + tid * 8;
+    int row = flat / D, col = flat % D;
+    n_row[it] = row;
+    n_colp[it] = col ^ ((row & 7) << 3);
+  }
+
+  const int q_start = CAUSAL ? (kv0 / QT) * QT : 0;
+  for (int q0 = q_start; q0 < Sq; q0 += QT) {
+    // stage Q natural (DMA) + transposed source (dO for dV, Q for dK) +
+    // (dK only) dO natural (DMA)
+    if (q0 + QT <= Sq) {
+#pragma unroll
+      for (int it = 0; it < QT * D / (NT * 8); ++it) {
+        const short* qsrc = qg + qbase + (long long)(q0 + n_row[it]) * q_ss + n_colp[it];
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)qsrc,
+            (__attribute__((address_space(3))) unsigned int*)(q_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
+            16, 0, 0);
+        if (IS_DK) {
+          const short* dsrc = dog + dobase + (long long)(q0 + n_row[it]) * do_ss + n_colp[it];
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) unsigned int*)dsrc,
+              (__attribute__((address_space(3))) unsigned int*)(d_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
+              16, 0, 0);
+        }
+      }
+      const int rot = tid & 7;
+      for (int flat = tid * 8; flat < QT * D; flat += NT * 8) {
+        int row = flat / D, col = flat % D;
+        const short* src = IS_DK
+            ? qg + qbase + (long long)(q0 + row) * q_ss + col
+            : dog + dobase + (long long)(q0 + row) * do_ss + col;
+        shortx8 v = *reinterpret_cast<const shortx8*>(src);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int i = (j + rot) & 7;
+          *reinterpret_cast<short*>(t_lds + lds_off32(col + i, row * 2, TR_RS)) = v[i];
+        }
+      }
+    } else {
+      const int rot = tid & 7;
+      for (int flat = tid * 8; flat < QT * D; flat += NT * 8) {
+        int row = flat / D, col = flat % D;
+        shortx8 qv, dv;
+        if (q0 + row < Sq) {
+          qv = *reinterpret_cast<const shortx8*>(qg + qbase + (long long)(q0 + row) * q_ss + col);
+          dv = *reinterpret_cast<const shortx8*>(dog + dobase + (long long)(q0 + row) * do_ss + col);
+        } else {
+          for (int i = 0; i < 8; ++i) { qv[i] = 0; dv[i] = 0; }
+        }
+        *reinterpret_cast<shortx8*>(q_lds + lds_off32(row, col * 2, NAT_RS)) = qv;
+        if (IS_DK)
+          *reinterpret_cast<shortx8*>(d_lds + lds_off32(row, col * 2, NAT_RS)) = dv;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int i = (j + rot) & 7;
+          *reinterpret_cast<short*>(t_lds + lds_off32(col + i, row * 2, TR_RS)) =
+              IS_DK ? qv[i] : dv[i];
+        }
+      }
+    }
+    __syncthreads();
+
+    // per-lane q-column stats for the two 32-wide q tiles
+#pragma unroll
+    for (int tq = 0; tq < 2; ++tq) {
+      const int q_abs = q0 + tq * 32 + l32;
+      const float lse_v = (q_abs < Sq) ? lseg[lse_base + q_abs] : 1e30f;
+      // S^T = K Q^T for this q tile
+      floatx16 st;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) st[r] = 0.f;
+#pragma unroll
+      for (int ks = 0; ks < NKS; ++ks) {
+        shortx8 qfr = *reinterpret_cast<const shortx8*>(
+            q_lds + lds_off32(tq * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
+        st = mfma32_bf16(kf[ks], qfr, st);
+      }
+      const bool bnd = (q0 + QT > Sq) ||
+                       (CAUSAL && q0 + (tq + 1) * 32 > kv0w) || (kv0w + 32 > Skv);
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float p = __expf(st[r] * scale - lse_v);
+        if (bnd) {
+          int kv_abs = kv0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          if (q_abs >= Sq || kv_abs >= Skv || (CAUSAL && kv_abs > q_abs)) p = 0.f;
+        }
+        st[r] = p;
+      }
+      if (IS_DK) {
+        const float delta_v = (q_abs < Sq) ? deltag[lse_base + q_abs] : 0.f;
+        floatx16 dp;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) dp[r] = 0.f;
+#pragma unroll
+        for (int ks = 0; ks < NKS; ++ks) {
+          shortx8 dfr = *reinterpret_cast<const shortx8*>(
+              d_lds + lds_off32(tq * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
+          dp = mfma32_bf16(vf[ks], dfr, dp);
+        }
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          st[r] = st[r] * (dp[r] - delta_v) * scale;
+      }
+      // acc += (P^T or dS^T)[kv x 32q] @ (dO or Q)[32q x D].  The A-frag
+      // needs a lane<->reg transpose of st (A row = kv = the REG dim) --
+      // permlane cannot do that; use a small per-wave LDS round trip.
+      // ps row stride 64B; swizzle col ^ ((row&3)<<3) elements (4 slots).
+      char* pw = ps_lds + wq * (32 * 32 * 2);
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        unsigned prow = (r & 3) + 8 * (r >> 2) + 4 * hi;  // kv-local
+        unsigned pcol = (unsigned)l32 ^ ((prow & 3u) << 3);
+        *reinterpret_cast<short*>(pw + prow * 64 + pcol * 2) = f2bf(st[r]);
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {   // q 16-steps within the 32-q tile
+        unsigned prow = (unsigned)l32;
+        unsigned pc0 = (unsigned)(ks * 16 + hi * 8) ^ ((prow & 3u) << 3);
+        shortx8 pa = *reinterpret_cast<const shortx8*>(pw + prow * 64 + pc0 * 2);
+#pragma unroll
+        for (int dt = 0; dt < NDT; ++dt) {
+          shortx8 bf = *reinterpret_cast<const shortx8*>(
+              t_lds + lds_off32(dt * 32 + l32, ((tq * 32 + ks * 16) + hi * 8) * 2, TR_RS));
+          acc[dt] = mfma32_bf16(pa, bf, acc[dt]);
+        }
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // reads done before tq+1 overwrites
+    }
+    __syncthreads();
+  }
+
+  // epilogue
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int rkv = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    int row = kv0w + rkv;
+    if (row >= Skv) continue;
+#pragma unroll
+    for (int dt = 0; dt < NDT; ++dt)
+      outg[outbase + (long long)row * dk_ss + dt * 32 + l32] = f2bf(acc[dt][r]);
+  }
+}
+
+void flash_attn_bwd_dkv32(const void* dout, const void* q, const void* k,
+                          const void* v, const float* lse, const float* delta,
+                          void* dk, void* dv, int64_t b, int64_t h, int64_t sq,
+                          int64_t skv, int64_t dh, float scale, bool causal,
+                          const int64_t* qs, const int64_t* ks, const int64_t* dos,
+                          const int64_t* dks, hipStream_t s) {
+  dim3 grid((unsigned)cdiv((int)skv, 128), (unsigned)(b * h));
+  dim3 blk(256);
+#define FDKV(D, C)                                                                do {                                                                              hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C, false>), grid, blk, 0, s,                            (const short*)dout, (const short*)q, (const short*)k,                           (const short*)v, lse, delta, (short*)dv, (int)b, (int)h,                        (int)sq, (int)skv, scale, qs[0], qs[1], qs[2],                                  ks[0], ks[1], ks[2], dos[0], dos[1], dos[2],                                    dks[0], dks[1], dks[2]);                                     hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C, true>), grid, blk, 0, s,                             (const short*)dout, (const short*)q, (const short*)k,                           (const short*)v, lse, delta, (short*)dk, (int)b, (int)h,                        (int)sq, (int)skv, scale, qs[0], qs[1], qs[2],                                  ks[0], ks[1], ks[2], dos[0], dos[1], dos[2],                                    dks[0], dks[1], dks[2]);                                   } while (0)
+  if (dh == 128) { if (causal) FDKV(128, true); else FDKV(128, false); }
+  else           { if (causal) FDKV(64, true);  else FDKV(64, false); }
+#undef FDKV
+}
+
 }  // namespace pa
