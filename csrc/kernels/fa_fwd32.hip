@@ -693,8 +693,9 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
             q_lds + lds_off32(tq * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
         st = mfma32_bf16(kf[ks], qfr, st);
       }
-      const bool bnd = (q0 + QT > Sq) ||
-                       (CAUSAL && q0 + (tq + 1) * 32 > kv0w) || (kv0w + 32 > Skv);
+      // mask needed unless every q in the tile >= every kv of this wave
+      const bool bnd = (q0 + QT > Sq) || (kv0w + 32 > Skv) ||
+                       (CAUSAL && kv0w + 32 > q0 + tq * 32);
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         float p = __expf(st[r] * scale - lse_v);

@@ -767,8 +767,12 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
   dim3 gq((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
 #define FAB(D, C)                                                              \
   do {                                                                         \
-    flash_attn_bwd_dkv32(dout, q, k, v, lse, delta, dk, dv, b, h, sq, skv,    \
-                         dh, scale, causal, qs, ks, dos, dks, s);              \
+    hipLaunchKernelGGL((fa_bwd_dkv_kernel<D, C>), gkv, dim3(512), 0, s,        \
+                       (const short*)dout, (const short*)q, (const short*)k,   \
+                       (const short*)v, lse, delta, (short*)dk, (short*)dv,    \
+                       (int)b, (int)h, (int)sq, (int)skv, scale,               \
+                       qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],               \
+                       dos[0], dos[1], dos[2], dks[0], dks[1], dks[2]);        \
     flash_attn_bwd_dq32(dout, q, k, v, lse, delta, dq, b, h, sq, skv, dh,     \
                         scale, causal, qs, ks, dos, dqs, s);                   \
   } while (0)
