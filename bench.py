@@ -70,10 +70,20 @@ def main():
     dev = torch.device("cuda", torch.cuda.current_device()) if on_gpu else torch.device("cpu")
     model = model.to(device=dev, dtype=dtype)
 
-    wrapped = GroupShardedStage3(model, device=dev)
-    opt = ShardedAdamW(wrapped, learning_rate=1e-4, beta1=0.9, beta2=0.95,
-                       epsilon=1e-8, weight_decay=0.1,
-                       grad_clip=paddle.nn.ClipGradByGlobalNorm(1.0))
+    if args.sharding_stage == 3 or world == 1:
+        # stage-3 flat-shard path (degenerates cleanly at world == 1)
+        wrapped = GroupShardedStage3(model, device=dev)
+        opt = ShardedAdamW(wrapped, learning_rate=1e-4, beta1=0.9, beta2=0.95,
+                           epsilon=1e-8, weight_decay=0.1,
+                           grad_clip=paddle.nn.ClipGradByGlobalNorm(1.0))
+    else:
+        from paddle_amd.distributed.fleet.sharding import group_sharded_parallel
+        inner = paddle.optimizer.AdamW(
+            learning_rate=1e-4, beta1=0.9, beta2=0.95, weight_decay=0.1,
+            parameters=model.parameters(),
+            grad_clip=paddle.nn.ClipGradByGlobalNorm(1.0))
+        level = {1: "os", 2: "os_g"}[args.sharding_stage]
+        wrapped, opt, _ = group_sharded_parallel(model, inner, level)
 
     vocab = model.cfg.vocab_size
     ids = torch.randint(0, vocab, (args.batch, args.seq), device=dev)

@@ -115,6 +115,48 @@ from .tensor.manipulation import (  # noqa: F401
 )
 from .tensor.math import (  # noqa: F401
     abs,
+    acos,
+    acosh,
+    addmm,
+    angle,
+    asin,
+    asinh,
+    atan,
+    atan2,
+    atanh,
+    conj,
+    count_nonzero,
+    deg2rad,
+    diff,
+    digamma,
+    erfinv,
+    expm1,
+    frac,
+    gcd,
+    heaviside,
+    histogram,
+    imag,
+    inner,
+    kron,
+    kthvalue,
+    lcm,
+    lerp,
+    lgamma,
+    log1p,
+    logit,
+    median,
+    mode,
+    nanmean,
+    nanmedian,
+    nansum,
+    nextafter,
+    quantile,
+    rad2deg,
+    real,
+    std,
+    take,
+    trapezoid,
+    var,
     add,
     add_n,
     all,

@@ -345,3 +345,206 @@ def topk(x, k, axis=None, largest=True, sorted=True, name=None):
 
 def bincount(x, weights=None, minlength=0, name=None):
     return torch.bincount(x, weights=weights, minlength=minlength)
+
+
+# -- statistics long tail ----------------------------------------------------
+def std(x, axis=None, unbiased=True, keepdim=False, name=None):
+    axis = _axis(axis)
+    if axis is None:
+        return torch.std(x, correction=1 if unbiased else 0)
+    return torch.std(x, dim=axis, correction=1 if unbiased else 0, keepdim=keepdim)
+
+
+def var(x, axis=None, unbiased=True, keepdim=False, name=None):
+    axis = _axis(axis)
+    if axis is None:
+        return torch.var(x, correction=1 if unbiased else 0)
+    return torch.var(x, dim=axis, correction=1 if unbiased else 0, keepdim=keepdim)
+
+
+def median(x, axis=None, keepdim=False, mode="avg", name=None):
+    axis = _axis(axis)
+    if axis is None:
+        flat = x.flatten().sort().values
+        n = flat.numel()
+        if mode == "avg" and n % 2 == 0:
+            return (flat[n // 2 - 1] + flat[n // 2]) / 2
+        return flat[(n - 1) // 2] if mode == "min" else flat[n // 2]
+    vals, idx = torch.median(x, dim=axis, keepdim=keepdim)
+    return vals
+
+
+def nanmedian(x, axis=None, keepdim=False, name=None):
+    axis = _axis(axis)
+    if axis is None:
+        return torch.nanmedian(x)
+    return torch.nanmedian(x, dim=axis, keepdim=keepdim).values
+
+
+def nanmean(x, axis=None, keepdim=False, name=None):
+    axis = _axis(axis)
+    if axis is None:
+        return torch.nanmean(x)
+    return torch.nanmean(x, dim=axis, keepdim=keepdim)
+
+
+def nansum(x, axis=None, dtype=None, keepdim=False, name=None):
+    from .. import framework
+    dt = framework.convert_dtype(dtype) if dtype else None
+    axis = _axis(axis)
+    if axis is None:
+        return torch.nansum(x, dtype=dt)
+    return torch.nansum(x, dim=axis, keepdim=keepdim, dtype=dt)
+
+
+def kthvalue(x, k, axis=None, keepdim=False, name=None):
+    ax = _axis(axis)
+    if ax is None:
+        ax = x.dim() - 1
+    v, i = torch.kthvalue(x, k, dim=ax, keepdim=keepdim)
+    return v, i
+
+
+def mode(x, axis=-1, keepdim=False, name=None):
+    v, i = torch.mode(x, dim=_axis(axis), keepdim=keepdim)
+    return v, i
+
+
+def quantile(x, q, axis=None, keepdim=False, interpolation="linear", name=None):
+    qq = torch.as_tensor(q, dtype=torch.float64, device=x.device)
+    return torch.quantile(x.double(), qq, dim=_axis(axis), keepdim=keepdim,
+                          interpolation=interpolation).to(x.dtype)
+
+
+def diff(x, n=1, axis=-1, prepend=None, append=None, name=None):
+    return torch.diff(x, n=n, dim=axis, prepend=prepend, append=append)
+
+
+def trapezoid(y, x=None, dx=None, axis=-1, name=None):
+    if dx is not None:
+        return torch.trapezoid(y, dx=dx, dim=axis)
+    return torch.trapezoid(y, x=x, dim=axis)
+
+
+def take(x, index, mode="raise", name=None):
+    return torch.take(x, index.long())
+
+
+def lerp(x, y, weight, name=None):
+    return torch.lerp(x, y, weight)
+
+
+def addmm(input, x, y, beta=1.0, alpha=1.0, name=None):
+    return torch.addmm(input, x, y, beta=beta, alpha=alpha)
+
+
+def inner(x, y, name=None):
+    return torch.inner(x, y)
+
+
+def kron(x, y, name=None):
+    return torch.kron(x, y)
+
+
+def gcd(x, y, name=None):
+    return torch.gcd(x, y)
+
+
+def lcm(x, y, name=None):
+    return torch.lcm(x, y)
+
+
+def heaviside(x, y, name=None):
+    return torch.heaviside(x, y)
+
+
+def frac(x, name=None):
+    return torch.frac(x)
+
+
+def deg2rad(x, name=None):
+    return torch.deg2rad(x)
+
+
+def rad2deg(x, name=None):
+    return torch.rad2deg(x)
+
+
+def angle(x, name=None):
+    return torch.angle(x)
+
+
+def conj(x, name=None):
+    return torch.conj(x)
+
+
+def real(x, name=None):
+    return torch.real(x)
+
+
+def imag(x, name=None):
+    return torch.imag(x)
+
+
+def log1p(x, name=None):
+    return torch.log1p(x)
+
+
+def expm1(x, name=None):
+    return torch.expm1(x)
+
+
+def atan(x, name=None):
+    return torch.atan(x)
+
+
+def atan2(x, y, name=None):
+    return torch.atan2(x, y)
+
+
+def asin(x, name=None):
+    return torch.asin(x)
+
+
+def acos(x, name=None):
+    return torch.acos(x)
+
+
+def asinh(x, name=None):
+    return torch.asinh(x)
+
+
+def acosh(x, name=None):
+    return torch.acosh(x)
+
+
+def atanh(x, name=None):
+    return torch.atanh(x)
+
+
+def erfinv(x, name=None):
+    return torch.erfinv(x)
+
+
+def digamma(x, name=None):
+    return torch.digamma(x)
+
+
+def lgamma(x, name=None):
+    return torch.lgamma(x)
+
+
+def logit(x, eps=None, name=None):
+    return torch.logit(x, eps=eps)
+
+
+def nextafter(x, y, name=None):
+    return torch.nextafter(x, y)
+
+
+def count_nonzero(x, axis=None, keepdim=False, name=None):
+    return torch.count_nonzero(x, dim=_axis(axis))
+
+
+def histogram(input, bins=100, min=0, max=0, name=None):
+    return torch.histc(input.float(), bins=bins, min=min, max=max)
