@@ -146,7 +146,7 @@ def main():
                 "model": model_name,
                 "global_batch": args.batch * n_gpus,
                 "seq_len": args.seq,
-                "parallelism": f"sharding3_dp{n_gpus}",
+                "parallelism": f"sharding{args.sharding_stage}_dp{n_gpus}",
                 "loss": round(float(loss.detach().float().cpu()), 4),
                 "peak_mem_gb": (round(torch.cuda.max_memory_allocated() / 2**30, 2)
                                 if on_gpu else None),

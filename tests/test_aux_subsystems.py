@@ -126,3 +126,31 @@ def test_vision_lenet_and_transforms():
     m = LeNet()
     out = m(x.unsqueeze(0))
     assert out.shape == (1, 10)
+
+
+# -- incubate.asp / autotune -------------------------------------------------
+def test_asp_prune_and_decorate():
+    import torch
+    import paddle_amd as paddle
+    from paddle_amd.incubate import asp
+    m = paddle.nn.Sequential(paddle.nn.Linear(16, 32), paddle.nn.Linear(32, 8))
+    pruned = asp.prune_model(m)
+    assert len(pruned) == 2
+    opt = asp.decorate(paddle.optimizer.AdamW(learning_rate=1e-3,
+                                              parameters=m.parameters()))
+    loss = m(torch.randn(4, 16)).pow(2).mean()
+    loss.backward()
+    opt.step()
+    opt.clear_grad()
+    for p in m.parameters():
+        if p.dim() == 2:
+            assert asp.check_sparsity(p)          # still 2:4 after a step
+            assert 0.3 < asp.calculate_density(p) <= 0.5
+
+
+def test_autotune_config_roundtrip():
+    from paddle_amd.incubate import autotune
+    cfg = autotune.set_config({"kernel": {"enable": False},
+                               "dataloader": {"tuning_steps": 7}})
+    assert cfg["dataloader"]["tuning_steps"] == 7
+    assert autotune.get_config()["kernel"]["enable"] is False
