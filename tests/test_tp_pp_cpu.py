@@ -164,3 +164,70 @@ def test_pipeline_two_stage_matches_single():
                 assert torch.allclose(a, b2, atol=1e-6)
         print("rank", r, "pp ok")
     """, timeout=300)
+
+
+def test_interleaved_vpp_matches_single():
+    """pp=2, v=2 → 4 virtual chunks over 8 blocks; loss and updated params
+    must match the single-process run exactly (relay schedule is numerically
+    identical to interleaved 1F1B)."""
+    run_dist("""
+        import torch
+        import paddle_amd as paddle
+        from paddle_amd import nn
+        from paddle_amd.distributed import fleet
+        strategy = fleet.DistributedStrategy()
+        strategy.hybrid_configs = {"dp_degree": 1, "mp_degree": 1, "pp_degree": 2,
+                                   "sharding_degree": 1}
+        strategy.pipeline_configs = {"accumulate_steps": 4, "micro_batch_size": 1}
+        fleet.init(is_collective=True, strategy=strategy)
+        from paddle_amd.distributed.fleet.pipeline import (
+            LayerDesc, VirtualPipelineLayer, InterleavedPipelineParallel)
+        hcg = fleet.get_hybrid_communicate_group()
+        r = paddle.distributed.get_rank()
+
+        class Block(nn.Layer):
+            def __init__(self, i):
+                super().__init__()
+                torch.manual_seed(142 + i)
+                self.fc = nn.Linear(8, 8)
+            def forward(self, x):
+                return torch.tanh(self.fc(x))
+
+        def loss_fn(out, y):
+            return ((out - y) ** 2).mean()
+
+        descs = [LayerDesc(Block, i) for i in range(8)]
+        pl = VirtualPipelineLayer(descs, loss_fn=loss_fn,
+                                  num_virtual_pipeline_stages=2, hcg=hcg)
+        # rank 0 owns chunks g=0 (blocks 0,1) and g=2 (blocks 4,5)
+        assert pl.my_stages == ([0, 2] if r == 0 else [1, 3])
+        model = InterleavedPipelineParallel(pl, hcg, strategy)
+        opt = paddle.optimizer.SGD(learning_rate=0.1, parameters=pl.parameters())
+        torch.manual_seed(9)
+        x = torch.randn(4, 8)
+        y = torch.randn(4, 8)
+        loss = model.train_batch((x, y), opt)
+
+        blocks = [Block(i) for i in range(8)]
+        opt_ref = paddle.optimizer.SGD(
+            learning_rate=0.1,
+            parameters=[p for b in blocks for p in b.parameters()])
+        total = 0.
+        for mb in range(4):
+            h = x[mb:mb+1]
+            for b in blocks:
+                h = b(h)
+            l = loss_fn(h, y[mb:mb+1]) / 4
+            l.backward()
+            total += float(l)
+        opt_ref.step()
+        if r == 1:
+            assert abs(float(loss) - total) < 1e-5, (float(loss), total)
+            # rank 1 owns blocks 2,3 (chunk g=1) and 6,7 (chunk g=3)
+            mine = list(pl.parameters())
+            theirs = [p for b in (blocks[2:4] + blocks[6:8]) for p in b.parameters()]
+            assert len(mine) == len(theirs)
+            for a, b2 in zip(mine, theirs):
+                assert torch.allclose(a, b2, atol=1e-6)
+        print("rank", r, "vpp ok")
+    """, timeout=300)
