@@ -200,7 +200,8 @@ Tensor colsum(const Tensor& x) {
 // ---- adamw ----------------------------------------------------------------
 void adamw(Tensor& master, c10::optional<Tensor> param_out, const Tensor& grad,
            Tensor& m, Tensor& v, double lr, double beta1, double beta2,
-           double eps, double wd, double beta1_pow, double beta2_pow) {
+           double eps, double wd, double beta1_pow, double beta2_pow,
+           double grad_scale) {
   CHECK_IN(master); CHECK_IN(grad); CHECK_IN(m); CHECK_IN(v);
   int64_t numel = master.numel();
   TORCH_CHECK(numel % 4 == 0, "adamw: flat shard numel must be multiple of 4");
@@ -213,8 +214,8 @@ void adamw(Tensor& master, c10::optional<Tensor> param_out, const Tensor& grad,
   pa::adamw(master.mutable_data_ptr<float>(), pout, grad.const_data_ptr(),
             m.mutable_data_ptr<float>(), v.mutable_data_ptr<float>(), numel,
             (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd,
-            (float)beta1_pow, (float)beta2_pow, dt_of(grad), bf16out,
-            cur_stream());
+            (float)beta1_pow, (float)beta2_pow, (float)grad_scale,
+            dt_of(grad), bf16out, cur_stream());
 }
 
 Tensor l2norm_sq(const Tensor& x) {

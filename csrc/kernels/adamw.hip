@@ -16,7 +16,7 @@ __global__ void adamw_kernel(float* __restrict__ master, void* __restrict__ para
                              const void* __restrict__ grad, float* __restrict__ m,
                              float* __restrict__ v, int64_t numel, float lr,
                              float beta1, float beta2, float eps, float wd,
-                             float bias1, float bias2) {
+                             float bias1, float bias2, float gscale) {
   // bias1 = 1 - beta1^t, bias2 = 1 - beta2^t
   const float inv_b1 = 1.f / bias1;
   const float inv_b2 = 1.f / bias2;
@@ -29,10 +29,10 @@ __global__ void adamw_kernel(float* __restrict__ master, void* __restrict__ para
     if (GDT == kBF16) {
       shortx4 gv = *reinterpret_cast<const shortx4*>((const short*)grad + i);
 #pragma unroll
-      for (int k = 0; k < 4; ++k) g[k] = bf2f(gv[k]);
+      for (int k = 0; k < 4; ++k) g[k] = bf2f(gv[k]) * gscale;
     } else {
       float4 gv = *reinterpret_cast<const float4*>((const float*)grad + i);
-      g[0] = gv.x; g[1] = gv.y; g[2] = gv.z; g[3] = gv.w;
+      g[0] = gv.x * gscale; g[1] = gv.y * gscale; g[2] = gv.z * gscale; g[3] = gv.w * gscale;
     }
     float p[4] = {pm.x, pm.y, pm.z, pm.w};
     float mo[4] = {mm.x, mm.y, mm.z, mm.w};
@@ -64,24 +64,24 @@ __global__ void adamw_kernel(float* __restrict__ master, void* __restrict__ para
 void adamw(float* master, void* param_bf16, const void* grad, float* m,
            float* v, int64_t numel, float lr, float beta1, float beta2,
            float eps, float wd, float beta1_pow, float beta2_pow,
-           int grad_dtype, bool param_out_bf16, hipStream_t s) {
+           float grad_scale, int grad_dtype, bool param_out_bf16, hipStream_t s) {
   float bias1 = 1.f - beta1_pow;
   float bias2 = 1.f - beta2_pow;
   dim3 g((unsigned)elementwise_grid(cdiv((int)hmin<int64_t>(numel / 4 + 1, 1 << 30), 256)));
   if (grad_dtype == kBF16) {
     if (param_out_bf16)
       hipLaunchKernelGGL((adamw_kernel<kBF16, true>), g, dim3(256), 0, s, master,
-                         param_bf16, grad, m, v, numel, lr, beta1, beta2, eps, wd, bias1, bias2);
+                         param_bf16, grad, m, v, numel, lr, beta1, beta2, eps, wd, bias1, bias2, grad_scale);
     else
       hipLaunchKernelGGL((adamw_kernel<kBF16, false>), g, dim3(256), 0, s, master,
-                         param_bf16, grad, m, v, numel, lr, beta1, beta2, eps, wd, bias1, bias2);
+                         param_bf16, grad, m, v, numel, lr, beta1, beta2, eps, wd, bias1, bias2, grad_scale);
   } else {
     if (param_out_bf16)
       hipLaunchKernelGGL((adamw_kernel<kF32, true>), g, dim3(256), 0, s, master,
-                         param_bf16, grad, m, v, numel, lr, beta1, beta2, eps, wd, bias1, bias2);
+                         param_bf16, grad, m, v, numel, lr, beta1, beta2, eps, wd, bias1, bias2, grad_scale);
     else
       hipLaunchKernelGGL((adamw_kernel<kF32, false>), g, dim3(256), 0, s, master,
-                         param_bf16, grad, m, v, numel, lr, beta1, beta2, eps, wd, bias1, bias2);
+                         param_bf16, grad, m, v, numel, lr, beta1, beta2, eps, wd, bias1, bias2, grad_scale);
   }
 }
 

@@ -69,7 +69,7 @@ void colsum(const void* x, float* out, int64_t n, int64_t d, int dtype,
 void adamw(float* master, void* param_bf16, const void* grad, float* m,
            float* v, int64_t numel, float lr, float beta1, float beta2,
            float eps, float wd, float beta1_pow, float beta2_pow,
-           int grad_dtype, bool param_out_bf16, hipStream_t s);
+           float grad_scale, int grad_dtype, bool param_out_bf16, hipStream_t s);
 
 // multi-tensor l2-norm^2 of a flat fp32/bf16 buffer -> out[0] (fp32, add)
 void l2norm_sq(const void* x, float* out, int64_t numel, int dtype,

@@ -74,6 +74,12 @@ class _Unit:
         # fp32 master shard (optimizer state lives here)
         self.master = self.shard.float()
         self.grad_shard_fp32 = torch.zeros_like(self.master)
+        # grad_ready: the tensor step()/grad-norm consume this step -- the
+        # bf16 flat buffer directly on the fast (no-accumulation) paths, or
+        # the fp32 accumulator.  grad_premul folds the 1/world reduce divide.
+        self.grad_ready = None
+        self.grad_premul = 1.0
+        self.defer_flat_zero = False
         del flat
         self.full: Optional[torch.Tensor] = None
         self.gather_work = None
@@ -136,11 +142,20 @@ class _Unit:
         for p in self.params:
             p.grad = None
         if self.world == 1:
-            if accumulate and self.accum_steps > 0:
-                self.grad_shard_fp32.add_(flat[0:self.shard_size])
+            if accumulate:
+                if self.accum_steps > 0:
+                    self.grad_shard_fp32.add_(flat[0:self.shard_size])
+                else:
+                    self.grad_shard_fp32.copy_(flat[0:self.shard_size])
+                flat.zero_()
+                self.grad_ready, self.grad_premul = self.grad_shard_fp32, 1.0
             else:
-                self.grad_shard_fp32.copy_(flat[0:self.shard_size])
-            flat.zero_()
+                # fast path: the optimizer consumes the bf16 flat buffer
+                # directly (fused grad-scale in the AdamW kernel); zeroing
+                # is deferred to clear_grad() after step() has read it.
+                self.grad_ready = flat[0:self.shard_size]
+                self.grad_premul = 1.0
+                self.defer_flat_zero = True
         elif accumulate:
             # grad-accumulation path: keep it synchronous (flat is re-bound
             # next micro-step, must not race the collective)
@@ -153,10 +168,13 @@ class _Unit:
             else:
                 self.grad_shard_fp32.copy_(out)
             flat.zero_()
+            self.grad_ready, self.grad_premul = self.grad_shard_fp32, 1.0
         else:
             # async reduce-scatter overlapped with the remaining backward;
-            # the optimizer's step() calls finish_grad_reduce() first
-            flat.div_(self.world)
+            # the optimizer's step() calls finish_grad_reduce() first.
+            # Grads are SUMMED (no div pass over the 2x-total-bytes flat
+            # buffer); the 1/world lands in grad_premul for the AdamW
+            # kernel's fused grad scale.
             out = torch.empty(self.shard_size, dtype=self.dtype,
                               device=self.shard.device)
             work = C.reduce_scatter_tensor(out, flat, group=self.group,
@@ -169,7 +187,8 @@ class _Unit:
             work, out = self.grad_work
             if work is not None:
                 work.wait()
-            self.grad_shard_fp32.copy_(out)
+            self.grad_ready = out
+            self.grad_premul = 1.0 / self.world
             self.grad_flat.zero_()
             self.grad_work = None
 
@@ -422,7 +441,10 @@ class ShardedAdamW:
         for u in self.model._units:
             u.finish_grad_reduce()
         for u in self.model._units:
-            sq = hot.l2_norm_squared(u.grad_shard_fp32)
+            g = u.grad_ready if u.grad_ready is not None else u.grad_shard_fp32
+            sq = hot.l2_norm_squared(g)
+            if u.grad_premul != 1.0:
+                sq = sq * (u.grad_premul ** 2)
             total = sq if total is None else total + sq
         if total is None:
             return None
@@ -443,18 +465,25 @@ class ShardedAdamW:
             if gnorm is not None:
                 clip_coeff = float(clip_norm) / float(max(float(gnorm), float(clip_norm)))
         for u, m, v in zip(self.model._units, self._m, self._v):
-            g = u.grad_shard_fp32
+            g = u.grad_ready if u.grad_ready is not None else u.grad_shard_fp32
+            scale = u.grad_premul
             if clip_coeff is not None and clip_coeff < 1.0:
-                g.mul_(clip_coeff)
+                scale = scale * clip_coeff
             hot.fused_adamw_step(u.master, u.shard, g, m, v, lr, self.beta1,
                                  self.beta2, self.eps, self.weight_decay,
-                                 self.step_count)
+                                 self.step_count, grad_scale=scale)
             u.accum_steps = 0
+            u.grad_ready = None
 
     def clear_grad(self, set_to_zero=True):
         for u in self.model._units:
             u.finish_grad_reduce()  # consume any pending async reduce
-            u.grad_shard_fp32.zero_()
+            if u.defer_flat_zero and u.grad_flat is not None:
+                u.grad_flat.zero_()
+                u.defer_flat_zero = False
+            # grad_shard_fp32 needs no zeroing: the first micro-step of the
+            # next accumulation window copy_()s rather than add_()s.
+            u.grad_ready = None
             u.accum_steps = 0
 
     clear_gradients = clear_grad

@@ -629,17 +629,22 @@ def embedding(ids, table, padding_idx=None):
 # optimizer primitives
 # ---------------------------------------------------------------------------
 def fused_adamw_step(master, param_out, grad, m, v, lr, beta1, beta2, eps,
-                     weight_decay, step):
+                     weight_decay, step, grad_scale=1.0):
     """In-place fused AdamW on a flat fp32 master shard.  param_out may be
     a bf16 view of the model weights (written by the kernel) or None.
+    grad may be bf16 (read direct from the flat reduce buffer) or fp32;
+    grad_scale folds grad-clip and the 1/world reduce divide into the
+    kernel so no separate mul/cast pass touches HBM.
     adamw_ parity: paddle/phi/kernels/gpu/adamw_kernel.cu."""
     if master.is_cuda and _ext.use_native(master):
         C = _ext.get_ext()
         C.adamw(master, param_out, grad.contiguous(), m, v, lr, beta1, beta2,
-                eps, weight_decay, beta1 ** step, beta2 ** step)
+                eps, weight_decay, beta1 ** step, beta2 ** step, grad_scale)
         return
     # torch reference (CPU tests)
     g = grad.float()
+    if grad_scale != 1.0:
+        g = g * grad_scale
     master.mul_(1 - lr * weight_decay)
     m.mul_(beta1).add_(g, alpha=1 - beta1)
     v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
