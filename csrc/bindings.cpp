@@ -197,6 +197,13 @@ Tensor colsum(const Tensor& x) {
   return out;
 }
 
+namespace pa_lt {
+std::tuple<Tensor, Tensor> fc1_gelu_fwd(const Tensor& x, const Tensor& w,
+                                        const Tensor& bias);
+std::tuple<Tensor, Tensor> fc2_dgrad_dgelu(const Tensor& dy, const Tensor& w2,
+                                           const Tensor& z);
+}  // namespace pa_lt
+
 // ---- adamw ----------------------------------------------------------------
 void adamw(Tensor& master, c10::optional<Tensor> param_out, const Tensor& grad,
            Tensor& m, Tensor& v, double lr, double beta1, double beta2,
@@ -414,6 +421,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_fwd", &rope_fwd);
   m.def("colsum", &colsum);
   m.def("adamw", &adamw);
+  m.def("fc1_gelu_fwd", &pa_lt::fc1_gelu_fwd);
+  m.def("fc2_dgrad_dgelu", &pa_lt::fc2_dgrad_dgelu);
   m.def("l2norm_sq", &l2norm_sq);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);

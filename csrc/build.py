@@ -67,28 +67,29 @@ def build(verbose: bool = False) -> Path:
                    "-ffast-math", f"-I{CSRC}", "-c", str(src), "-o", str(obj)]
             jobs.append(cmd)
 
-    bind_src = CSRC / "bindings.cpp"
-    bind_obj = OUT / "bindings.o"
-    objs.append(bind_obj)
-    if _newer(bind_src, bind_obj, deps):
-        cmd = ["g++", "-O2", "-std=c++17", "-fPIC", "-D__HIP_PLATFORM_AMD__=1",
-               "-DUSE_ROCM=1", "-DTORCH_EXTENSION_NAME=_C",
-               "-D_GLIBCXX_USE_CXX11_ABI=1",
-               f"-I{CSRC}", f"-I{py_inc}", f"-I{ROCM / 'include'}"]
-        cmd += [f"-I{i}" for i in includes]
-        cmd += ["-c", str(bind_src), "-o", str(bind_obj)]
-        jobs.append(cmd)
+    for cpp_name in ("bindings.cpp", "blaslt.cpp"):
+        cpp_src = CSRC / cpp_name
+        cpp_obj = OUT / (cpp_name[:-4] + ".o")
+        objs.append(cpp_obj)
+        if _newer(cpp_src, cpp_obj, deps):
+            cmd = ["g++", "-O2", "-std=c++17", "-fPIC", "-D__HIP_PLATFORM_AMD__=1",
+                   "-DUSE_ROCM=1", "-DTORCH_EXTENSION_NAME=_C",
+                   "-D_GLIBCXX_USE_CXX11_ABI=1",
+                   f"-I{CSRC}", f"-I{py_inc}", f"-I{ROCM / 'include'}"]
+            cmd += [f"-I{i}" for i in includes]
+            cmd += ["-c", str(cpp_src), "-o", str(cpp_obj)]
+            jobs.append(cmd)
 
     if jobs:
         with ThreadPoolExecutor(max_workers=min(8, len(jobs))) as ex:
             list(ex.map(_run, jobs))
 
-    if _newer(bind_obj, TARGET) or any(_newer(o, TARGET) for o in objs):
+    if any(_newer(o, TARGET) for o in objs):
         link = [HIPCC, "-shared", "-fPIC", "-o", str(TARGET)]
         link += [str(o) for o in objs]
         link += [f"-L{torch_lib}", "-ltorch", "-ltorch_python", "-lc10",
                  "-ltorch_hip", "-lc10_hip", f"-Wl,-rpath,{torch_lib}",
-                 f"-L{ROCM / 'lib'}", "-lamdhip64"]
+                 f"-L{ROCM / 'lib'}", "-lamdhip64", "-lhipblaslt"]
         _run(link)
     if verbose:
         print(f"built {TARGET}")

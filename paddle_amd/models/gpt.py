@@ -106,7 +106,12 @@ class GPTMLP(nn.Layer):
         _apply_initializer(out_init, self.fc2.weight)
 
     def forward(self, x):
-        # bias folded into the fused bias+gelu kernel
+        if hot._fused_ffn_available(x):
+            # fc1 bias+GELU in the hipBLASLt epilogue; backward fuses
+            # dGELU + fc1 bias-grad into fc2's dgrad GEMM
+            return hot.fused_ffn(x, self.fc1.weight, self.fc1_bias,
+                                 self.fc2.weight, self.fc2.bias)
+        # CPU / non-bf16 path: fused bias+gelu elementwise kernel
         return self.fc2(hot.bias_gelu(self.fc1(x), self.fc1_bias))
 
 

@@ -693,3 +693,47 @@ def paged_decode_attention(q, k_cache, v_cache, block_table, seq_lens, scale=Non
         p = torch.softmax(s, dim=-1)
         out[b] = torch.einsum("hs,shd->hd", p, vf).to(q.dtype)
     return out
+
+
+# ---------------------------------------------------------------------------
+# fused FFN: fc1(+bias+GELU in the hipBLASLt epilogue) -> fc2.
+# Backward folds dGELU + fc1 bias-grad into fc2's dgrad GEMM (DGELU_BGRAD),
+# so the [tokens, 4h] activation is never touched by a separate elementwise
+# pass.  GELU is the tanh approximation (Tensile epilogue).
+# Parity: paddle incubate FusedFeedForward / fused_gemm_epilogue
+# (paddle/phi/kernels/fusion/gpu/fused_gemm_epilogue_kernel.cu).
+# ---------------------------------------------------------------------------
+class _FusedFFN(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w1, b1, w2, b2):
+        C = _ext.get_ext()
+        xs = x.shape
+        x2 = x.reshape(-1, xs[-1]).contiguous()
+        g, z = C.fc1_gelu_fwd(x2, w1, b1)
+        y = torch.addmm(b2, g, w2)
+        ctx.save_for_backward(x2, w1, w2, z, g)
+        ctx.xshape = xs
+        return y.reshape(*xs[:-1], w2.shape[1])
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = _ext.get_ext()
+        x2, w1, w2, z, g = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dz1, db1 = C.fc2_dgrad_dgelu(dy2, w2, z)   # dGELU + bias-grad fused
+        dw2 = g.t() @ dy2
+        db2 = dy2.sum(0)
+        dx = dz1 @ w1.t()
+        dw1 = x2.t() @ dz1
+        return dx.reshape(ctx.xshape), dw1, db1, dw2, db2
+
+
+def fused_ffn(x, w1, b1, w2, b2):
+    """y = gelu(x @ w1 + b1) @ w2 + b2 with epilogue-fused bias/GELU.
+    Paddle Linear layout: w1 [K, 4K], w2 [4K, K]."""
+    return _FusedFFN.apply(x, w1, b1, w2, b2)
+
+
+def _fused_ffn_available(x):
+    return (x.is_cuda and x.dtype == torch.bfloat16 and _ext.use_native(x)
+            and hasattr(_ext.get_ext(), "fc1_gelu_fwd"))

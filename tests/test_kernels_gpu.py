@@ -268,3 +268,36 @@ def test_paged_decode_attention_gpu():
     ref = hot.paged_decode_attention(q, kc, vc, bt, seq)
     paddle_amd.set_flags({"FLAGS_use_native_kernels": True})
     _assert_close_bf16(out, ref.float(), atol=3e-2, rtol=3e-2)
+
+
+def test_fused_ffn_hipblaslt_gpu():
+    """fc1 bias+GELU epilogue + fc2, backward with DGELU_BGRAD epilogue,
+    vs fp32 torch reference (tanh-approx GELU)."""
+    torch.manual_seed(11)
+    M, K, N, H = 512, 256, 1024, 256
+    x = torch.randn(M, K, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    w1 = torch.randn(K, N, device=DEV, dtype=torch.bfloat16) * 0.05
+    b1 = torch.randn(N, device=DEV, dtype=torch.bfloat16) * 0.1
+    w2 = torch.randn(N, H, device=DEV, dtype=torch.bfloat16) * 0.05
+    b2 = torch.randn(H, device=DEV, dtype=torch.bfloat16) * 0.1
+    for t in (w1, b1, w2, b2):
+        t.requires_grad_(True)
+    y = hot.fused_ffn(x, w1, b1, w2, b2)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xf = x.detach().float().requires_grad_(True)
+    w1f = w1.detach().float().requires_grad_(True)
+    b1f = b1.detach().float().requires_grad_(True)
+    w2f = w2.detach().float().requires_grad_(True)
+    b2f = b2.detach().float().requires_grad_(True)
+    g = torch.nn.functional.gelu(xf @ w1f + b1f, approximate="tanh")
+    yr = g @ w2f + b2f
+    yr.backward(dy.float())
+
+    torch.testing.assert_close(y.float(), yr, atol=0.12, rtol=0.05)
+    torch.testing.assert_close(x.grad.float(), xf.grad, atol=0.15, rtol=0.08)
+    torch.testing.assert_close(b1.grad.float(), b1f.grad, atol=0.8, rtol=0.05)
+    torch.testing.assert_close(b2.grad.float(), b2f.grad, atol=0.8, rtol=0.05)
+    torch.testing.assert_close(w1.grad.float(), w1f.grad, atol=0.8, rtol=0.08)
+    torch.testing.assert_close(w2.grad.float(), w2f.grad, atol=0.8, rtol=0.08)
