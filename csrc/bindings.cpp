@@ -15,6 +15,13 @@ void mfma_probe(const void* a, const void* bt, float* c, hipStream_t s);
 void mfma_probe32(const void* a, const void* bt, float* c, hipStream_t s);
 }
 
+namespace pa_lt {
+std::tuple<torch::Tensor, torch::Tensor> fc1_gelu_fwd(
+    const torch::Tensor& x, const torch::Tensor& w, const torch::Tensor& bias);
+std::tuple<torch::Tensor, torch::Tensor> fc2_dgrad_dgelu(
+    const torch::Tensor& dy, const torch::Tensor& w2, const torch::Tensor& z);
+}  // namespace pa_lt
+
 namespace {
 
 using torch::Tensor;
@@ -196,13 +203,6 @@ Tensor colsum(const Tensor& x) {
              cur_stream());
   return out;
 }
-
-namespace pa_lt {
-std::tuple<Tensor, Tensor> fc1_gelu_fwd(const Tensor& x, const Tensor& w,
-                                        const Tensor& bias);
-std::tuple<Tensor, Tensor> fc2_dgrad_dgelu(const Tensor& dy, const Tensor& w2,
-                                           const Tensor& z);
-}  // namespace pa_lt
 
 // ---- adamw ----------------------------------------------------------------
 void adamw(Tensor& master, c10::optional<Tensor> param_out, const Tensor& grad,

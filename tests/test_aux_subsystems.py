@@ -154,3 +154,13 @@ def test_autotune_config_roundtrip():
                                "dataloader": {"tuning_steps": 7}})
     assert cfg["dataloader"]["tuning_steps"] == 7
     assert autotune.get_config()["kernel"]["enable"] is False
+
+
+def test_extension_imports_when_built():
+    """If the in-tree _C.so exists it must import cleanly -- catches
+    undefined-symbol link errors on the CPU box before any GPU run."""
+    import os
+    import paddle_amd
+    so = os.path.join(os.path.dirname(paddle_amd.__file__), "_C.so")
+    if os.path.exists(so):
+        import paddle_amd._C  # noqa: F401
