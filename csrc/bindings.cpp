@@ -20,6 +20,7 @@ std::tuple<torch::Tensor, torch::Tensor> fc1_gelu_fwd(
     const torch::Tensor& x, const torch::Tensor& w, const torch::Tensor& bias);
 std::tuple<torch::Tensor, torch::Tensor> fc2_dgrad_dgelu(
     const torch::Tensor& dy, const torch::Tensor& w2, const torch::Tensor& z);
+int64_t lt_epilogue_probe(int64_t m, int64_t n, int64_t k, int64_t epi);
 }  // namespace pa_lt
 
 namespace {
@@ -423,6 +424,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw", &adamw);
   m.def("fc1_gelu_fwd", &pa_lt::fc1_gelu_fwd);
   m.def("fc2_dgrad_dgelu", &pa_lt::fc2_dgrad_dgelu);
+  m.def("lt_epilogue_probe", &pa_lt::lt_epilogue_probe);
   m.def("l2norm_sq", &l2norm_sq);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);

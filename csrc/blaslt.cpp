@@ -80,6 +80,10 @@ static void lt_matmul(hipblasOperation_t opA, hipblasOperation_t opB,
         desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_POINTER, &aux, sizeof(aux)));
     LT_CHECK(hipblasLtMatmulDescSetAttribute(
         desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD, &aux_ld, sizeof(aux_ld)));
+    int32_t auxdt = HIP_R_16BF;
+    LT_CHECK(hipblasLtMatmulDescSetAttribute(
+        desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_DATA_TYPE, &auxdt,
+        sizeof(auxdt)));
   }
 
   // layouts: col-major; A is (rows x kk) after opA, B is (kk x cols) after opB
@@ -169,6 +173,51 @@ std::tuple<Tensor, Tensor> fc2_dgrad_dgelu(const Tensor& dy, const Tensor& w2,
             HIPBLASLT_EPILOGUE_DGELU_BGRAD, db.mutable_data_ptr(),
             const_cast<void*>(z.const_data_ptr()), N, dy);
   return {dg, db};
+}
+
+// heuristic-only support probe: how many algos exist for a bf16 GEMM of
+// (m x n x k) with the given epilogue?  Lets Python pick the fused path at
+// runtime instead of failing mid-autograd on builds without aux-epilogue
+// Tensile kernels.
+int64_t lt_epilogue_probe(int64_t m, int64_t n, int64_t k, int64_t epi) {
+  hipblasLtMatmulDesc_t desc;
+  LT_CHECK(hipblasLtMatmulDescCreate(&desc, HIPBLAS_COMPUTE_32F, HIP_R_32F));
+  hipblasOperation_t opN = HIPBLAS_OP_N;
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_TRANSA,
+                                           &opN, sizeof(opN)));
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_TRANSB,
+                                           &opN, sizeof(opN)));
+  hipblasLtEpilogue_t e = (hipblasLtEpilogue_t)epi;
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_EPILOGUE,
+                                           &e, sizeof(e)));
+  if (epi & 128) {  // aux-flavored epilogues
+    int64_t ld = m;
+    LT_CHECK(hipblasLtMatmulDescSetAttribute(
+        desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD, &ld, sizeof(ld)));
+    int32_t auxdt = HIP_R_16BF;
+    LT_CHECK(hipblasLtMatmulDescSetAttribute(
+        desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_DATA_TYPE, &auxdt,
+        sizeof(auxdt)));
+  }
+  hipblasLtMatrixLayout_t la, lb, ld_;
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&la, HIP_R_16BF, m, k, m));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&lb, HIP_R_16BF, k, n, k));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&ld_, HIP_R_16BF, m, n, m));
+  hipblasLtMatmulPreference_t pref;
+  LT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+  size_t wsz = kWorkspace;
+  LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &wsz, sizeof(wsz)));
+  hipblasLtMatmulHeuristicResult_t res[8];
+  int found = 0;
+  hipblasStatus_t st = hipblasLtMatmulAlgoGetHeuristic(
+      handle(), desc, la, lb, ld_, ld_, pref, 8, res, &found);
+  hipblasLtMatmulPreferenceDestroy(pref);
+  hipblasLtMatrixLayoutDestroy(la);
+  hipblasLtMatrixLayoutDestroy(lb);
+  hipblasLtMatrixLayoutDestroy(ld_);
+  hipblasLtMatmulDescDestroy(desc);
+  return st == HIPBLAS_STATUS_SUCCESS ? found : 0;
 }
 
 }  // namespace pa_lt

@@ -734,6 +734,20 @@ def fused_ffn(x, w1, b1, w2, b2):
     return _FusedFFN.apply(x, w1, b1, w2, b2)
 
 
+_lt_ffn_ok = None
+
+
 def _fused_ffn_available(x):
-    return (x.is_cuda and x.dtype == torch.bfloat16 and _ext.use_native(x)
-            and hasattr(_ext.get_ext(), "fc1_gelu_fwd"))
+    global _lt_ffn_ok
+    if not (x.is_cuda and x.dtype == torch.bfloat16 and _ext.use_native(x)):
+        return False
+    if _lt_ffn_ok is None:
+        C = _ext.get_ext()
+        if not hasattr(C, "lt_epilogue_probe"):
+            _lt_ffn_ok = False
+        else:
+            # GELU_AUX_BIAS=164 fwd, DGELU_BGRAD=208 bwd -- both must have
+            # Tensile kernels in this hipBLASLt build
+            _lt_ffn_ok = (C.lt_epilogue_probe(1024, 1024, 1024, 164) > 0 and
+                          C.lt_epilogue_probe(1024, 1024, 1024, 208) > 0)
+    return _lt_ffn_ok
