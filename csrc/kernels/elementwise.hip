@@ -56,24 +56,30 @@ __device__ __forceinline__ float silu_grad_f(float x) {
   return sig * (1.f + x * (1.f - sig));
 }
 
-template <int DT, bool FWD>
+template <int DT, bool FWD, int W>
 __global__ void bias_gelu_kernel(const void* __restrict__ a, const void* __restrict__ x,
                                  const void* __restrict__ bias, void* __restrict__ out,
                                  int64_t n, int64_t d) {
   // FWD: a == x (unused), out = gelu(x+bias).  BWD: a = dy, out = dx.
+  // W=16: two 16B loads in flight per stream -- streaming-friendly MLP.
   int64_t total = n * d;
-  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8; i < total;
-       i += (int64_t)gridDim.x * blockDim.x * 8) {
-    float xf[8], bf[8], af[8];
-    LS8<DT>::load8(x, i, xf);
-    if (bias) LS8<DT>::load8(bias, i % d, bf);
-    if (!FWD) LS8<DT>::load8(a, i, af);
+  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * W; i < total;
+       i += (int64_t)gridDim.x * blockDim.x * W) {
+    float xf[W], bf[W], af[W];
 #pragma unroll
-    for (int k = 0; k < 8; ++k) {
+    for (int v8 = 0; v8 < W / 8; ++v8) {
+      LS8<DT>::load8(x, i + 8 * v8, xf + 8 * v8);
+      if (bias) LS8<DT>::load8(bias, (i + 8 * v8) % d, bf + 8 * v8);
+      if (!FWD) LS8<DT>::load8(a, i + 8 * v8, af + 8 * v8);
+    }
+#pragma unroll
+    for (int k = 0; k < W; ++k) {
       float v = xf[k] + (bias ? bf[k] : 0.f);
       xf[k] = FWD ? gelu_f(v) : af[k] * gelu_grad_f(v);
     }
-    LS8<DT>::store8(out, i, xf);
+#pragma unroll
+    for (int v8 = 0; v8 < W / 8; ++v8)
+      LS8<DT>::store8(out, i + 8 * v8, xf + 8 * v8);
   }
 }
 
@@ -169,20 +175,32 @@ __global__ void rope_kernel(const void* __restrict__ x, const float* __restrict_
   }
 }
 
-// column-sum [n, d] -> fp32 [d]
+// column-sum [n, d] -> fp32 [d].  8-wide column strips per thread (16 B
+// loads) instead of scalar b16 loads; 2-D grid tiles rows.
 template <int DT>
 __global__ void colsum_kernel(const void* __restrict__ x, float* __restrict__ out,
                               int64_t n, int64_t d) {
-  int64_t col = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= d) return;
+  int64_t c0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (c0 >= d) return;
   int64_t rows_per = (n + gridDim.y - 1) / gridDim.y;
   int64_t r0 = (int64_t)blockIdx.y * rows_per, r1 = min(n, r0 + rows_per);
-  float acc = 0.f;
-  for (int64_t r = r0; r < r1; ++r) {
-    if (DT == kBF16) acc += bf2f(((const short*)x)[r * d + col]);
-    else acc += ((const float*)x)[r * d + col];
+  float acc[8] = {0.f};
+  if (c0 + 8 <= d) {
+    for (int64_t r = r0; r < r1; ++r) {
+      float v[8];
+      LS8<DT>::load8(x, r * d + c0, v);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) acc[k] += v[k];
+    }
+  } else {
+    for (int64_t r = r0; r < r1; ++r)
+      for (int64_t c = c0; c < d; ++c)
+        acc[c - c0] += (DT == kBF16) ? bf2f(((const short*)x)[r * d + c])
+                                     : ((const float*)x)[r * d + c];
   }
-  atomicAdd(&out[col], acc);
+#pragma unroll
+  for (int k = 0; k < 8; ++k)
+    if (c0 + k < d) atomicAdd(&out[c0 + k], acc[k]);
 }
 
 // Philox-free dropout: xorshift per-element hash of (seed, offset+idx).
@@ -241,18 +259,32 @@ static dim3 egrid(int64_t numel) {
 
 void bias_gelu_fwd(const void* x, const void* bias, void* y, int64_t n,
                    int64_t d, int dtype, hipStream_t s) {
-  dim3 g = egrid(n * d);
-  EDT(dtype,
-      hipLaunchKernelGGL((bias_gelu_kernel<kBF16, true>), g, dim3(256), 0, s, x, x, bias, y, n, d),
-      hipLaunchKernelGGL((bias_gelu_kernel<kF32, true>), g, dim3(256), 0, s, x, x, bias, y, n, d));
+  if (d % 16 == 0) {
+    dim3 g = egrid(n * d / 2);
+    EDT(dtype,
+        hipLaunchKernelGGL((bias_gelu_kernel<kBF16, true, 16>), g, dim3(256), 0, s, x, x, bias, y, n, d),
+        hipLaunchKernelGGL((bias_gelu_kernel<kF32, true, 16>), g, dim3(256), 0, s, x, x, bias, y, n, d));
+  } else {
+    dim3 g = egrid(n * d);
+    EDT(dtype,
+        hipLaunchKernelGGL((bias_gelu_kernel<kBF16, true, 8>), g, dim3(256), 0, s, x, x, bias, y, n, d),
+        hipLaunchKernelGGL((bias_gelu_kernel<kF32, true, 8>), g, dim3(256), 0, s, x, x, bias, y, n, d));
+  }
 }
 
 void bias_gelu_bwd(const void* dy, const void* x, const void* bias, void* dx,
                    int64_t n, int64_t d, int dtype, hipStream_t s) {
-  dim3 g = egrid(n * d);
-  EDT(dtype,
-      hipLaunchKernelGGL((bias_gelu_kernel<kBF16, false>), g, dim3(256), 0, s, dy, x, bias, dx, n, d),
-      hipLaunchKernelGGL((bias_gelu_kernel<kF32, false>), g, dim3(256), 0, s, dy, x, bias, dx, n, d));
+  if (d % 16 == 0) {
+    dim3 g = egrid(n * d / 2);
+    EDT(dtype,
+        hipLaunchKernelGGL((bias_gelu_kernel<kBF16, false, 16>), g, dim3(256), 0, s, dy, x, bias, dx, n, d),
+        hipLaunchKernelGGL((bias_gelu_kernel<kF32, false, 16>), g, dim3(256), 0, s, dy, x, bias, dx, n, d));
+  } else {
+    dim3 g = egrid(n * d);
+    EDT(dtype,
+        hipLaunchKernelGGL((bias_gelu_kernel<kBF16, false, 8>), g, dim3(256), 0, s, dy, x, bias, dx, n, d),
+        hipLaunchKernelGGL((bias_gelu_kernel<kF32, false, 8>), g, dim3(256), 0, s, dy, x, bias, dx, n, d));
+  }
 }
 
 void swiglu_fwd(const void* x, void* y, int64_t n, int64_t d, int dtype, hipStream_t s) {
@@ -282,8 +314,11 @@ void rope_fwd(const void* x, const float* cos_t, const float* sin_t, void* y,
 }
 
 void colsum(const void* x, float* out, int64_t n, int64_t d, int dtype, hipStream_t s) {
-  int chunks = (int)hmin<int64_t>(64, hmax<int64_t>(1, n / 64));
-  dim3 grid((unsigned)cdiv((int)d, 256), chunks);
+  int xblocks = cdiv((int)d, 256 * 8);
+  // keep >=1024 workgroups in flight (256 CUs / 8 XCDs want oversubscription)
+  int chunks = (int)hmin<int64_t>(hmax<int64_t>(1, n / 32),
+                                  hmax<int64_t>(1, 2048 / xblocks));
+  dim3 grid((unsigned)xblocks, chunks);
   EDT(dtype,
       hipLaunchKernelGGL((colsum_kernel<kBF16>), grid, dim3(256), 0, s, x, out, n, d),
       hipLaunchKernelGGL((colsum_kernel<kF32>), grid, dim3(256), 0, s, x, out, n, d));

@@ -150,21 +150,43 @@ __global__ void ln_bwd_dwdb_kernel(const void* __restrict__ dy, const void* __re
                                    const float* __restrict__ mean, const float* __restrict__ rstd,
                                    float* __restrict__ dw, float* __restrict__ db,
                                    int64_t n, int64_t d) {
-  int64_t col = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= d) return;
+  // 8-wide column strips (16 B loads) per thread; 2-D grid tiles rows;
+  // fp32 atomics once per (strip, row-chunk)
+  int64_t c0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (c0 >= d) return;
   int64_t rows_per = (n + gridDim.y - 1) / gridDim.y;
   int64_t r0 = (int64_t)blockIdx.y * rows_per;
   int64_t r1 = min(n, r0 + rows_per);
-  float aw = 0.f, ab = 0.f;
-  for (int64_t r = r0; r < r1; ++r) {
-    float g = VIO<DT>::load1(dy, r * d + col);
-    float xf = VIO<DT>::load1(x, r * d + col);
-    float xh = RMS ? xf * rstd[r] : (xf - mean[r]) * rstd[r];
-    aw += g * xh;
-    ab += g;
+  float aw[8] = {0.f}, ab[8] = {0.f};
+  if (c0 + 8 <= d) {
+    for (int64_t r = r0; r < r1; ++r) {
+      float g[8], xf[8];
+      VIO<DT>::load8(dy, r * d + c0, g);
+      VIO<DT>::load8(x, r * d + c0, xf);
+      float mu = RMS ? 0.f : mean[r], rs = rstd[r];
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        aw[k] += g[k] * (xf[k] - mu) * rs;
+        ab[k] += g[k];
+      }
+    }
+  } else {
+    for (int64_t r = r0; r < r1; ++r) {
+      float mu = RMS ? 0.f : mean[r], rs = rstd[r];
+      for (int64_t c = c0; c < d; ++c) {
+        float g = VIO<DT>::load1(dy, r * d + c);
+        float xf = VIO<DT>::load1(x, r * d + c);
+        aw[c - c0] += g * (xf - mu) * rs;
+        ab[c - c0] += g;
+      }
+    }
   }
-  atomicAdd(&dw[col], aw);
-  if (db) atomicAdd(&db[col], ab);
+#pragma unroll
+  for (int k = 0; k < 8; ++k)
+    if (c0 + k < d) {
+      atomicAdd(&dw[c0 + k], aw[k]);
+      if (db) atomicAdd(&db[c0 + k], ab[k]);
+    }
 }
 
 // ---------------------------------------------------------------------------
@@ -276,8 +298,10 @@ void layer_norm_bwd_dx(const void* dy, const void* x, const void* w,
 void layer_norm_bwd_dwdb(const void* dy, const void* x, const float* mean,
                          const float* rstd, float* dw, float* db, int64_t n,
                          int64_t d, int dtype, hipStream_t s) {
-  int chunks = (int)hmin<int64_t>(64, hmax<int64_t>(1, n / 64));
-  dim3 grid((unsigned)cdiv((int)d, 256), chunks);
+  int xblocks = cdiv((int)d, 256 * 8);
+  int chunks = (int)hmin<int64_t>(hmax<int64_t>(1, n / 32),
+                                  hmax<int64_t>(1, 2048 / xblocks));
+  dim3 grid((unsigned)xblocks, chunks);
   DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_kernel<DT, false>), grid,
                                       dim3(256), 0, s, dy, x, mean, rstd, dw, db, n, d));
 }
@@ -303,8 +327,10 @@ void rms_norm_bwd_dx(const void* dy, const void* x, const void* w,
 
 void rms_norm_bwd_dw(const void* dy, const void* x, const float* rstd,
                      float* dw, int64_t n, int64_t d, int dtype, hipStream_t s) {
-  int chunks = (int)hmin<int64_t>(64, hmax<int64_t>(1, n / 64));
-  dim3 grid((unsigned)cdiv((int)d, 256), chunks);
+  int xblocks = cdiv((int)d, 256 * 8);
+  int chunks = (int)hmin<int64_t>(hmax<int64_t>(1, n / 32),
+                                  hmax<int64_t>(1, 2048 / xblocks));
+  dim3 grid((unsigned)xblocks, chunks);
   DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_kernel<DT, true>), grid,
                                       dim3(256), 0, s, dy, x, nullptr, rstd, dw, nullptr, n, d));
 }
