@@ -67,7 +67,7 @@ void adamw(float* master, void* param_bf16, const void* grad, float* m,
            float grad_scale, int grad_dtype, bool param_out_bf16, hipStream_t s) {
   float bias1 = 1.f - beta1_pow;
   float bias2 = 1.f - beta2_pow;
-  dim3 g((unsigned)elementwise_grid(cdiv((int)hmin<int64_t>(numel / 4 + 1, 1 << 30), 256)));
+  dim3 g((unsigned)elementwise_grid(cdiv((int)hmin<int64_t>(numel / 8 + 1, 1 << 30), 256)));
   if (grad_dtype == kBF16) {
     if (param_out_bf16)
       hipLaunchKernelGGL((adamw_kernel<kBF16, true>), g, dim3(256), 0, s, master,

@@ -27,8 +27,8 @@ def timeit(fn, iters=30):
     return (time.perf_counter() - t0) / iters
 
 
-def report(name, sec, gbytes):
-    print(f"{name:28s} {sec * 1e6:9.1f} us   {gbytes / sec / 1e12:6.2f} TB/s")
+def report(name, sec, gb):
+    print(f"{name:28s} {sec * 1e6:9.1f} us   {gb / sec / 1e3:6.2f} TB/s")
 
 
 # GPT-6.7B shapes: tokens = 8*2048, hidden 4096, ffn 16384
