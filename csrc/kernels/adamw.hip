@@ -90,8 +90,10 @@ __global__ void l2norm_sq_kernel(const void* __restrict__ x, float* __restrict__
                                  int64_t numel) {
   __shared__ float red[4];
   float acc = 0.f;
-  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4; i < numel;
-       i += (int64_t)gridDim.x * blockDim.x * 4) {
+  int64_t per_blk = ((numel / 4 + gridDim.x - 1) / gridDim.x) * 4;
+  int64_t blk1 = min(numel, (int64_t)(blockIdx.x + 1) * per_blk);
+  for (int64_t i = (int64_t)blockIdx.x * per_blk + (int64_t)threadIdx.x * 4; i < blk1;
+       i += (int64_t)blockDim.x * 4) {
     if (DT == kBF16) {
       shortx4 v = *reinterpret_cast<const shortx4*>((const short*)x + i);
 #pragma unroll
