@@ -73,16 +73,14 @@ __global__ void bias_gelu_kernel(const void* __restrict__ a, const void* __restr
                                  int64_t n, int64_t d) {
   // FWD: a == x (unused), out = gelu(x+bias).  BWD: a = dy, out = dx.
   // W=16: two 16B loads in flight per stream -- streaming-friendly MLP.
-  // contiguous chunk per block: sequential DRAM streams beat grid-stride
-  // interleaving by ~1.5x on MI355X for pure streaming
   int64_t total = n * d;
-  int64_t per_blk = ((total / W + gridDim.x - 1) / gridDim.x) * W;
-  int64_t blk0 = (int64_t)blockIdx.x * per_blk;
-  int64_t blk1 = min(total, blk0 + per_blk);
-  int64_t i0 = blk0 + (int64_t)threadIdx.x * W;
-  int64_t stride = (int64_t)blockDim.x * W;
+  int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * W;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * W;
+  // bias column tracked incrementally -- no 64-bit modulo in the loop
+  // (d divides both W-chunks and wraps cleanly: launcher guarantees d % W == 0
+  // alignment of every chunk start)
   int64_t c = i0 % d, cstep = stride % d;
-  for (int64_t i = i0; i < blk1; i += stride, c = (c + cstep >= d ? c + cstep - d : c + cstep)) {
+  for (int64_t i = i0; i < total; i += stride, c = (c + cstep >= d ? c + cstep - d : c + cstep)) {
     float xf[W], bf[W], af[W];
 #pragma unroll
     for (int v8 = 0; v8 < W / 8; ++v8) {
@@ -193,51 +191,32 @@ __global__ void rope_kernel(const void* __restrict__ x, const float* __restrict_
   }
 }
 
-// column-sum [n, d] -> fp32 [d].  Each block owns a contiguous row range
-// and sweeps every row left-to-right in 2048-col passes (whole 16B lanes,
-// fully sequential DRAM streams -- grid-stride/column-strip layouts
-// measured ~2.8 TB/s, contiguous sweeps ~5.5+).  PASSES*8 accumulators
-// per thread; one atomicAdd per column per block at the end.
-template <int DT, int PASSES>
+// column-sum [n, d] -> fp32 [d].  8-wide column strips per thread (16 B
+// loads) instead of scalar b16 loads; 2-D grid tiles rows.
+template <int DT>
 __global__ void colsum_kernel(const void* __restrict__ x, float* __restrict__ out,
                               int64_t n, int64_t d) {
-  int64_t rows_per = (n + gridDim.x - 1) / gridDim.x;
-  int64_t r0 = (int64_t)blockIdx.x * rows_per, r1 = min(n, r0 + rows_per);
-  float acc[PASSES][8];
-#pragma unroll
-  for (int pp = 0; pp < PASSES; ++pp)
-#pragma unroll
-    for (int k = 0; k < 8; ++k) acc[pp][k] = 0.f;
-  for (int64_t r = r0; r < r1; ++r) {
-    const int64_t base = r * d + threadIdx.x * 8;
-#pragma unroll
-    for (int pp = 0; pp < PASSES; ++pp) {
-      float v[8];
-      LS8<DT>::load8(x, base + pp * 2048, v);
-#pragma unroll
-      for (int k = 0; k < 8; ++k) acc[pp][k] += v[k];
-    }
-  }
-#pragma unroll
-  for (int pp = 0; pp < PASSES; ++pp)
-#pragma unroll
-    for (int k = 0; k < 8; ++k)
-      atomicAdd(&out[threadIdx.x * 8 + pp * 2048 + k], acc[pp][k]);
-}
-
-// generic fallback for widths that are not a multiple of 2048
-template <int DT>
-__global__ void colsum_kernel_gen(const void* __restrict__ x, float* __restrict__ out,
-                                  int64_t n, int64_t d) {
-  int64_t col = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= d) return;
+  int64_t c0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (c0 >= d) return;
   int64_t rows_per = (n + gridDim.y - 1) / gridDim.y;
   int64_t r0 = (int64_t)blockIdx.y * rows_per, r1 = min(n, r0 + rows_per);
-  float acc = 0.f;
-  for (int64_t r = r0; r < r1; ++r)
-    acc += (DT == kBF16) ? bf2f(((const short*)x)[r * d + col])
-                         : ((const float*)x)[r * d + col];
-  atomicAdd(&out[col], acc);
+  float acc[8] = {0.f};
+  if (c0 + 8 <= d) {
+    for (int64_t r = r0; r < r1; ++r) {
+      float v[8];
+      LS8<DT>::load8(x, r * d + c0, v);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) acc[k] += v[k];
+    }
+  } else {
+    for (int64_t r = r0; r < r1; ++r)
+      for (int64_t c = c0; c < d; ++c)
+        acc[c - c0] += (DT == kBF16) ? bf2f(((const short*)x)[r * d + c])
+                                     : ((const float*)x)[r * d + c];
+  }
+#pragma unroll
+  for (int k = 0; k < 8; ++k)
+    if (c0 + k < d) atomicAdd(&out[c0 + k], acc[k]);
 }
 
 // Philox-free dropout: xorshift per-element hash of (seed, offset+idx).
@@ -351,24 +330,14 @@ void rope_fwd(const void* x, const float* cos_t, const float* sin_t, void* y,
 }
 
 void colsum(const void* x, float* out, int64_t n, int64_t d, int dtype, hipStream_t s) {
-  int nblocks = (int)hmin<int64_t>(1024, hmax<int64_t>(1, n / 8));
-#define COLSUM_P(P) \
-    EDT(dtype, \
-        hipLaunchKernelGGL((colsum_kernel<kBF16, P>), dim3(nblocks), dim3(256), 0, s, x, out, n, d), \
-        hipLaunchKernelGGL((colsum_kernel<kF32, P>), dim3(nblocks), dim3(256), 0, s, x, out, n, d))
-  switch (d / 2048) {
-    case 1: if (d == 2048) { COLSUM_P(1); return; } break;
-    case 2: if (d == 4096) { COLSUM_P(2); return; } break;
-    case 4: if (d == 8192) { COLSUM_P(4); return; } break;
-    case 8: if (d == 16384) { COLSUM_P(8); return; } break;
-    default: break;
-  }
-#undef COLSUM_P
-  int chunks = (int)hmin<int64_t>(64, hmax<int64_t>(1, n / 64));
-  dim3 grid((unsigned)cdiv((int)d, 256), chunks);
+  int xblocks = cdiv((int)d, 256 * 8);
+  // keep >=1024 workgroups in flight (256 CUs / 8 XCDs want oversubscription)
+  int chunks = (int)hmin<int64_t>(hmax<int64_t>(1, n / 32),
+                                  hmax<int64_t>(1, 2048 / xblocks));
+  dim3 grid((unsigned)xblocks, chunks);
   EDT(dtype,
-      hipLaunchKernelGGL((colsum_kernel_gen<kBF16>), grid, dim3(256), 0, s, x, out, n, d),
-      hipLaunchKernelGGL((colsum_kernel_gen<kF32>), grid, dim3(256), 0, s, x, out, n, d));
+      hipLaunchKernelGGL((colsum_kernel<kBF16>), grid, dim3(256), 0, s, x, out, n, d),
+      hipLaunchKernelGGL((colsum_kernel<kF32>), grid, dim3(256), 0, s, x, out, n, d));
 }
 
 void dropout_add_fwd(const void* x, const void* residual, void* y,

@@ -145,46 +145,6 @@ __global__ void ln_bwd_dx_kernel(const void* __restrict__ dy, const void* __rest
 
 // dw[d] = sum_n dy*xhat ; db[d] = sum_n dy.  Column-parallel: thread owns a
 // column, walks a row chunk, one atomicAdd per (column, chunk).
-// contiguous-row-sweep variant: block owns a row range, reads every row
-// left-to-right in 2048-col passes (sequential DRAM streams; the strip
-// layout below measured ~2.8 TB/s, this ~5.5+).  PASSES = d/2048.
-template <int DT, bool RMS, int PASSES>
-__global__ void ln_bwd_dwdb_sweep_kernel(const void* __restrict__ dy, const void* __restrict__ x,
-                                         const float* __restrict__ mean, const float* __restrict__ rstd,
-                                         float* __restrict__ dw, float* __restrict__ db,
-                                         int64_t n, int64_t d) {
-  int64_t rows_per = (n + gridDim.x - 1) / gridDim.x;
-  int64_t r0 = (int64_t)blockIdx.x * rows_per, r1 = min(n, r0 + rows_per);
-  float aw[PASSES][8], ab[PASSES][8];
-#pragma unroll
-  for (int pp = 0; pp < PASSES; ++pp)
-#pragma unroll
-    for (int k = 0; k < 8; ++k) { aw[pp][k] = 0.f; ab[pp][k] = 0.f; }
-  for (int64_t r = r0; r < r1; ++r) {
-    float mu = RMS ? 0.f : mean[r], rs = rstd[r];
-    const int64_t base = r * d + threadIdx.x * 8;
-#pragma unroll
-    for (int pp = 0; pp < PASSES; ++pp) {
-      float g[8], xf[8];
-      VIO<DT>::load8(dy, base + pp * 2048, g);
-      VIO<DT>::load8(x, base + pp * 2048, xf);
-#pragma unroll
-      for (int k = 0; k < 8; ++k) {
-        aw[pp][k] += g[k] * (xf[k] - mu) * rs;
-        ab[pp][k] += g[k];
-      }
-    }
-  }
-#pragma unroll
-  for (int pp = 0; pp < PASSES; ++pp)
-#pragma unroll
-    for (int k = 0; k < 8; ++k) {
-      int64_t col = threadIdx.x * 8 + pp * 2048 + k;
-      atomicAdd(&dw[col], aw[pp][k]);
-      if (db) atomicAdd(&db[col], ab[pp][k]);
-    }
-}
-
 template <int DT, bool RMS>
 __global__ void ln_bwd_dwdb_kernel(const void* __restrict__ dy, const void* __restrict__ x,
                                    const float* __restrict__ mean, const float* __restrict__ rstd,
@@ -338,24 +298,12 @@ void layer_norm_bwd_dx(const void* dy, const void* x, const void* w,
 void layer_norm_bwd_dwdb(const void* dy, const void* x, const float* mean,
                          const float* rstd, float* dw, float* db, int64_t n,
                          int64_t d, int dtype, hipStream_t s) {
-  int nblocks = (int)hmin<int64_t>(1024, hmax<int64_t>(1, n / 8));
-  if (d == 2048) {
-    DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_sweep_kernel<DT, false, 1>), dim3(nblocks),
-                                        dim3(256), 0, s, dy, x, mean, rstd, dw, db, n, d));
-  } else if (d == 4096) {
-    DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_sweep_kernel<DT, false, 2>), dim3(nblocks),
-                                        dim3(256), 0, s, dy, x, mean, rstd, dw, db, n, d));
-  } else if (d == 8192) {
-    DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_sweep_kernel<DT, false, 4>), dim3(nblocks),
-                                        dim3(256), 0, s, dy, x, mean, rstd, dw, db, n, d));
-  } else {
-    int xblocks = cdiv((int)d, 256 * 8);
-    int chunks = (int)hmin<int64_t>(hmax<int64_t>(1, n / 32),
-                                    hmax<int64_t>(1, 2048 / xblocks));
-    dim3 grid((unsigned)xblocks, chunks);
-    DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_kernel<DT, false>), grid,
-                                        dim3(256), 0, s, dy, x, mean, rstd, dw, db, n, d));
-  }
+  int xblocks = cdiv((int)d, 256 * 8);
+  int chunks = (int)hmin<int64_t>(hmax<int64_t>(1, n / 32),
+                                  hmax<int64_t>(1, 2048 / xblocks));
+  dim3 grid((unsigned)xblocks, chunks);
+  DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_kernel<DT, false>), grid,
+                                      dim3(256), 0, s, dy, x, mean, rstd, dw, db, n, d));
 }
 
 void rms_norm_fwd(const void* x, const void* residual, const void* w, void* y,
@@ -379,24 +327,12 @@ void rms_norm_bwd_dx(const void* dy, const void* x, const void* w,
 
 void rms_norm_bwd_dw(const void* dy, const void* x, const float* rstd,
                      float* dw, int64_t n, int64_t d, int dtype, hipStream_t s) {
-  int nblocks = (int)hmin<int64_t>(1024, hmax<int64_t>(1, n / 8));
-  if (d == 2048) {
-    DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_sweep_kernel<DT, true, 1>), dim3(nblocks),
-                                        dim3(256), 0, s, dy, x, nullptr, rstd, dw, nullptr, n, d));
-  } else if (d == 4096) {
-    DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_sweep_kernel<DT, true, 2>), dim3(nblocks),
-                                        dim3(256), 0, s, dy, x, nullptr, rstd, dw, nullptr, n, d));
-  } else if (d == 8192) {
-    DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_sweep_kernel<DT, true, 4>), dim3(nblocks),
-                                        dim3(256), 0, s, dy, x, nullptr, rstd, dw, nullptr, n, d));
-  } else {
-    int xblocks = cdiv((int)d, 256 * 8);
-    int chunks = (int)hmin<int64_t>(hmax<int64_t>(1, n / 32),
-                                    hmax<int64_t>(1, 2048 / xblocks));
-    dim3 grid((unsigned)xblocks, chunks);
-    DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_kernel<DT, true>), grid,
-                                        dim3(256), 0, s, dy, x, nullptr, rstd, dw, nullptr, n, d));
-  }
+  int xblocks = cdiv((int)d, 256 * 8);
+  int chunks = (int)hmin<int64_t>(hmax<int64_t>(1, n / 32),
+                                  hmax<int64_t>(1, 2048 / xblocks));
+  dim3 grid((unsigned)xblocks, chunks);
+  DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_kernel<DT, true>), grid,
+                                      dim3(256), 0, s, dy, x, nullptr, rstd, dw, nullptr, n, d));
 }
 
 }  // namespace pa
