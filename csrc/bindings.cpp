@@ -212,7 +212,6 @@ void adamw(Tensor& master, c10::optional<Tensor> param_out, const Tensor& grad,
            double grad_scale) {
   CHECK_IN(master); CHECK_IN(grad); CHECK_IN(m); CHECK_IN(v);
   int64_t numel = master.numel();
-  TORCH_CHECK(numel % 8 == 0, "adamw: flat shard numel must be multiple of 8");
   bool bf16out = false;
   void* pout = nullptr;
   if (param_out.has_value()) {

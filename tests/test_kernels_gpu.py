@@ -219,6 +219,22 @@ def test_fused_adamw_gpu():
     torch.testing.assert_close(param_bf16.float(), master, atol=1e-2, rtol=1e-2)
 
 
+def test_fused_adamw_tail_gpu():
+    # numel not divisible by 8 (per-param path, e.g. a 2-elem bias)
+    for n in (2, 13, 4099):
+        master = torch.randn(n, device=DEV)
+        grad = torch.randn(n, device=DEV, dtype=torch.bfloat16)
+        m = torch.zeros(n, device=DEV)
+        v = torch.zeros(n, device=DEV)
+        ref = master.clone().requires_grad_(True)
+        opt = torch.optim.AdamW([ref], lr=1e-2, betas=(0.9, 0.95), eps=1e-8,
+                                weight_decay=0.1)
+        ref.grad = grad.float()
+        opt.step()
+        hot.fused_adamw_step(master, None, grad, m, v, 1e-2, 0.9, 0.95, 1e-8, 0.1, 1)
+        torch.testing.assert_close(master, ref.detach(), atol=1e-5, rtol=1e-5)
+
+
 def test_fused_adamw_grad_scale_gpu():
     # grad_scale folds clip/1-world into the kernel: scaled bf16 grads must
     # match pre-scaled fp32 grads through torch.optim.AdamW
