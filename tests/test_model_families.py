@@ -57,8 +57,10 @@ def test_moe_layer_single_process():
     (out.sum() + layer.aux_loss).backward()
     assert x.grad is not None
     assert layer.gate.wg.weight.grad is not None
-    for e in layer.experts:
-        assert e.fc1.weight.grad is not None
+    expert_params = (list(layer.experts.parameters()) if layer.grouped
+                     else [p for e in layer.experts for p in e.parameters()])
+    for p_ in expert_params:
+        assert p_.grad is not None
 
 
 def test_moe_gpt_trains():
