@@ -91,7 +91,7 @@ class BertModel(nn.Layer):
         mask = None
         if attention_mask is not None:
             # [b, s] 1/0 -> additive [b, 1, 1, s]
-            mask = (1.0 - attention_mask.to(x.dtype)) * -1e4
+            mask = (1.0 - attention_mask.to(device=x.device, dtype=x.dtype)) * -1e4
             mask = mask.view(mask.shape[0], 1, 1, mask.shape[1])
         for layer in self.encoder:
             x = layer(x, mask)

@@ -52,6 +52,10 @@ def run_dist(script_body: str, world_size: int = 2, timeout: int = 240,
             "MASTER_PORT": str(port),
             "PYTHONPATH": REPO + os.pathsep + env.get("PYTHONPATH", ""),
             "CUDA_VISIBLE_DEVICES": "",
+            # ROCm ignores CUDA_VISIBLE_DEVICES -- hide GPUs properly so the
+            # gloo path is taken even on a GPU box
+            "HIP_VISIBLE_DEVICES": "",
+            "ROCR_VISIBLE_DEVICES": "",
         })
         procs.append(subprocess.Popen([sys.executable, "-c", script], env=env,
                                       stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
