@@ -540,19 +540,28 @@ void flash_attn_bwd_dq32(const void* dout, const void* q, const void* k,
 }
 
 // ---------------------------------------------------------------------------
-// backward dV / dK, 32x32 swapped variants.  Split into two kernels so each
-// fits the 2-waves/SIMD register budget (combined dK+dV needs ~290 VGPRs).
-// Wave owns 32 kv rows; q-COLUMN per lane => per-lane lse/delta; the P^T /
-// dS^T A-fragments for the [kv x q] x [q x d] products come straight from
-// the accumulator via the static-pack + permlane exchange (no LDS trip).
+// backward dV + dK, 32x32 combined variant (v2).
+//
+// Orientation flip vs the retired split version: compute P = mfma(Q, K)
+// and dP = mfma(dO, V) so the OUTPUT COLUMN (kv) sits in the lane dim.
+// P / dS then feed the [d x q][q x kv] products dV^T = dO^T P and
+// dK^T = Q^T dS as B-fragments assembled with the same static-pack +
+// permlane32_swap exchange as the forward's PV step -- the lane<->reg
+// transpose that forced the old per-wave LDS round trip never arises.
+// lse/delta live in the q = REG dim: staged to LDS once per q tile and
+// read back as 16 broadcast (wave-uniform-per-half) ds_reads.
+// K/V B-fragments are loaded once into registers (wave owns 32 kv rows);
+// Q/dO are staged natural (for P/dP A-frags, lane = q) and transposed
+// via bank-rotated writes (for dV^T/dK^T A-frags, lane = d).
+// Outputs are accumulated transposed and written with 8 B packed stores.
 // ---------------------------------------------------------------------------
-template <int D, bool CAUSAL, bool IS_DK>
-__launch_bounds__(256, 2)
+template <int D, bool CAUSAL>
+__launch_bounds__(256, 1)
 __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* __restrict__ qg,
                                     const short* __restrict__ kg, const short* __restrict__ vg,
                                     const float* __restrict__ lseg, const float* __restrict__ deltag,
-                                    short* __restrict__ outg, int B, int H, int Sq, int Skv,
-                                    float scale,
+                                    short* __restrict__ dkg, short* __restrict__ dvg,
+                                    int B, int H, int Sq, int Skv, float scale,
                                     long long q_sb, long long q_sh, long long q_ss,
                                     long long k_sb, long long k_sh, long long k_ss,
                                     long long do_sb, long long do_sh, long long do_ss,
@@ -565,10 +574,11 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
   constexpr int NDT = D / 32;
   constexpr unsigned NAT_RS = D * 2;
   constexpr unsigned TR_RS = QT * 2;
-  __shared__ char q_lds[QT * D * 2];        // natural (for S^T B-frags)
-  __shared__ char t_lds[D * QT * 2];        // dO^T (dV) or Q^T (dK)
-  __shared__ char d_lds[IS_DK ? QT * D * 2 : 1];  // dO natural (dK only)
-  __shared__ char ps_lds[NW * 32 * 32 * 2]; // per-wave P^T/dS^T round trip
+  __shared__ char q_lds[QT * D * 2];        // natural Q  (P A-frags)
+  __shared__ char d_lds[QT * D * 2];        // natural dO (dP A-frags)
+  __shared__ char qt_lds[D * QT * 2];       // Q^T  (dK^T A-frags)
+  __shared__ char dot_lds[D * QT * 2];      // dO^T (dV^T A-frags)
+  __shared__ float lse_s[QT], delta_s[QT];
 
   const int kvblk = blockIdx.x;
   const int bh = blockIdx.y;
@@ -586,31 +596,30 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
   const int l32 = lane & 31;
   const int hi = lane >> 5;
   const int kv0w = kv0 + wq * 32;
+  const int kv_lane = kv0w + l32;            // this lane's kv column
 
-  // K (and V for dK) A-fragments: lane = kv row kv0w + l32
-  shortx8 kf[NKS], vf[IS_DK ? NKS : 1];
+  // K and V as B-fragments (lane = kv row)
+  shortx8 kf[NKS], vf[NKS];
   {
-    int row = kv0w + l32;
-    bool ok = row < Skv;
+    bool ok = kv_lane < Skv;
 #pragma unroll
     for (int ks = 0; ks < NKS; ++ks) {
       if (ok) {
         kf[ks] = *reinterpret_cast<const shortx8*>(
-            kg + kvbase + (long long)row * k_ss + ks * 16 + hi * 8);
-        if (IS_DK)
-          vf[ks] = *reinterpret_cast<const shortx8*>(
-              vg + kvbase + (long long)row * k_ss + ks * 16 + hi * 8);
+            kg + kvbase + (long long)kv_lane * k_ss + ks * 16 + hi * 8);
+        vf[ks] = *reinterpret_cast<const shortx8*>(
+            vg + kvbase + (long long)kv_lane * k_ss + ks * 16 + hi * 8);
       } else {
-        for (int i = 0; i < 8; ++i) { kf[ks][i] = 0; if (IS_DK) vf[ks][i] = 0; }
+        for (int i = 0; i < 8; ++i) { kf[ks][i] = 0; vf[ks][i] = 0; }
       }
     }
   }
 
-  floatx16 acc[NDT];
+  floatx16 dvacc[NDT], dkacc[NDT];
 #pragma unroll
   for (int dt = 0; dt < NDT; ++dt)
 #pragma unroll
-    for (int r = 0; r < 16; ++r) acc[dt][r] = 0.f;
+    for (int r = 0; r < 16; ++r) { dvacc[dt][r] = 0.f; dkacc[dt][r] = 0.f; }
 
   int n_row[QT * D / (NT * 8)], n_colp[QT * D / (NT * 8)];
 #pragma unroll
@@ -623,35 +632,38 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
 
   const int q_start = CAUSAL ? (kv0 / QT) * QT : 0;
   for (int q0 = q_start; q0 < Sq; q0 += QT) {
-    // stage Q natural (DMA) + transposed source (dO for dV, Q for dK) +
-    // (dK only) dO natural (DMA)
+    // ---- stage: Q,dO natural (async DMA) + Q^T,dO^T (rotated writes) ------
+    if (tid < QT) {
+      int qa = q0 + tid;
+      lse_s[tid] = (qa < Sq) ? lseg[lse_base + qa] : 1e30f;
+      delta_s[tid] = (qa < Sq) ? deltag[lse_base + qa] : 0.f;
+    }
     if (q0 + QT <= Sq) {
 #pragma unroll
       for (int it = 0; it < QT * D / (NT * 8); ++it) {
         const short* qsrc = qg + qbase + (long long)(q0 + n_row[it]) * q_ss + n_colp[it];
+        const short* dsrc = dog + dobase + (long long)(q0 + n_row[it]) * do_ss + n_colp[it];
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) unsigned int*)qsrc,
             (__attribute__((address_space(3))) unsigned int*)(q_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
             16, 0, 0);
-        if (IS_DK) {
-          const short* dsrc = dog + dobase + (long long)(q0 + n_row[it]) * do_ss + n_colp[it];
-          __builtin_amdgcn_global_load_lds(
-              (const __attribute__((address_space(1))) unsigned int*)dsrc,
-              (__attribute__((address_space(3))) unsigned int*)(d_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
-              16, 0, 0);
-        }
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)dsrc,
+            (__attribute__((address_space(3))) unsigned int*)(d_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
+            16, 0, 0);
       }
       const int rot = tid & 7;
       for (int flat = tid * 8; flat < QT * D; flat += NT * 8) {
         int row = flat / D, col = flat % D;
-        const short* src = IS_DK
-            ? qg + qbase + (long long)(q0 + row) * q_ss + col
-            : dog + dobase + (long long)(q0 + row) * do_ss + col;
-        shortx8 v = *reinterpret_cast<const shortx8*>(src);
+        shortx8 qv = *reinterpret_cast<const shortx8*>(
+            qg + qbase + (long long)(q0 + row) * q_ss + col);
+        shortx8 dv = *reinterpret_cast<const shortx8*>(
+            dog + dobase + (long long)(q0 + row) * do_ss + col);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int i = (j + rot) & 7;
-          *reinterpret_cast<short*>(t_lds + lds_off32(col + i, row * 2, TR_RS)) = v[i];
+          *reinterpret_cast<short*>(qt_lds + lds_off32(col + i, row * 2, TR_RS)) = qv[i];
+          *reinterpret_cast<short*>(dot_lds + lds_off32(col + i, row * 2, TR_RS)) = dv[i];
         }
       }
     } else {
@@ -666,98 +678,111 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
           for (int i = 0; i < 8; ++i) { qv[i] = 0; dv[i] = 0; }
         }
         *reinterpret_cast<shortx8*>(q_lds + lds_off32(row, col * 2, NAT_RS)) = qv;
-        if (IS_DK)
-          *reinterpret_cast<shortx8*>(d_lds + lds_off32(row, col * 2, NAT_RS)) = dv;
+        *reinterpret_cast<shortx8*>(d_lds + lds_off32(row, col * 2, NAT_RS)) = dv;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int i = (j + rot) & 7;
-          *reinterpret_cast<short*>(t_lds + lds_off32(col + i, row * 2, TR_RS)) =
-              IS_DK ? qv[i] : dv[i];
+          *reinterpret_cast<short*>(qt_lds + lds_off32(col + i, row * 2, TR_RS)) = qv[i];
+          *reinterpret_cast<short*>(dot_lds + lds_off32(col + i, row * 2, TR_RS)) = dv[i];
         }
       }
     }
     __syncthreads();
 
-    // per-lane q-column stats for the two 32-wide q tiles
 #pragma unroll
     for (int tq = 0; tq < 2; ++tq) {
-      const int q_abs = q0 + tq * 32 + l32;
-      const float lse_v = (q_abs < Sq) ? lseg[lse_base + q_abs] : 1e30f;
-      // S^T = K Q^T for this q tile
-      floatx16 st;
+      // causal: whole 32-q tile below this wave's kv rows contributes nothing
+      if (CAUSAL && q0 + tq * 32 + 31 < kv0w) continue;
+      // ---- P = Q K^T, dP = dO V^T (lanes = kv, regs = q) ------------------
+      floatx16 st, dp;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) st[r] = 0.f;
+      for (int r = 0; r < 16; ++r) { st[r] = 0.f; dp[r] = 0.f; }
 #pragma unroll
       for (int ks = 0; ks < NKS; ++ks) {
         shortx8 qfr = *reinterpret_cast<const shortx8*>(
             q_lds + lds_off32(tq * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
-        st = mfma32_bf16(kf[ks], qfr, st);
+        st = mfma32_bf16(qfr, kf[ks], st);
+        shortx8 dfr = *reinterpret_cast<const shortx8*>(
+            d_lds + lds_off32(tq * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
+        dp = mfma32_bf16(dfr, vf[ks], dp);
       }
-      // mask needed unless every q in the tile >= every kv of this wave
+      // ---- P = exp(S*scale - lse[q]); dS = P (dP - delta[q]) --------------
       const bool bnd = (q0 + QT > Sq) || (kv0w + 32 > Skv) ||
                        (CAUSAL && kv0w + 32 > q0 + tq * 32);
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
+        const int rq = (r & 3) + 8 * (r >> 2) + 4 * hi;   // q-local (reg dim)
+        float lse_v = lse_s[tq * 32 + rq];                // broadcast read
         float p = __expf(st[r] * scale - lse_v);
         if (bnd) {
-          int kv_abs = kv0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          if (q_abs >= Sq || kv_abs >= Skv || (CAUSAL && kv_abs > q_abs)) p = 0.f;
+          int q_abs = q0 + tq * 32 + rq;
+          if (q_abs >= Sq || kv_lane >= Skv || (CAUSAL && kv_lane > q_abs)) p = 0.f;
         }
         st[r] = p;
+        dp[r] = p * (dp[r] - delta_s[tq * 32 + rq]) * scale;
       }
-      if (IS_DK) {
-        const float delta_v = (q_abs < Sq) ? deltag[lse_base + q_abs] : 0.f;
-        floatx16 dp;
+      // ---- dV^T += dO^T P ; dK^T += Q^T dS (B-frags via pack+permlane) ----
 #pragma unroll
-        for (int r = 0; r < 16; ++r) dp[r] = 0.f;
-#pragma unroll
-        for (int ks = 0; ks < NKS; ++ks) {
-          shortx8 dfr = *reinterpret_cast<const shortx8*>(
-              d_lds + lds_off32(tq * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
-          dp = mfma32_bf16(vf[ks], dfr, dp);
-        }
-#pragma unroll
-        for (int r = 0; r < 16; ++r)
-          st[r] = st[r] * (dp[r] - delta_v) * scale;
-      }
-      // acc += (P^T or dS^T)[kv x 32q] @ (dO or Q)[32q x D].  The A-frag
-      // needs a lane<->reg transpose of st (A row = kv = the REG dim) --
-      // permlane cannot do that; use a small per-wave LDS round trip.
-      // ps row stride 64B; swizzle col ^ ((row&3)<<3) elements (4 slots).
-      char* pw = ps_lds + wq * (32 * 32 * 2);
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        unsigned prow = (r & 3) + 8 * (r >> 2) + 4 * hi;  // kv-local
-        unsigned pcol = (unsigned)l32 ^ ((prow & 3u) << 3);
-        *reinterpret_cast<short*>(pw + prow * 64 + pcol * 2) = f2bf(st[r]);
-      }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {   // q 16-steps within the 32-q tile
-        unsigned prow = (unsigned)l32;
-        unsigned pc0 = (unsigned)(ks * 16 + hi * 8) ^ ((prow & 3u) << 3);
-        shortx8 pa = *reinterpret_cast<const shortx8*>(pw + prow * 64 + pc0 * 2);
+      for (int ks = 0; ks < 2; ++ks) {
+        const int b0 = 8 * ks;
+        int P0 = pack_bf2(st[b0 + 0], st[b0 + 1]);
+        int P1 = pack_bf2(st[b0 + 2], st[b0 + 3]);
+        int P2 = pack_bf2(st[b0 + 4], st[b0 + 5]);
+        int P3 = pack_bf2(st[b0 + 6], st[b0 + 7]);
+        int D0 = pack_bf2(dp[b0 + 0], dp[b0 + 1]);
+        int D1 = pack_bf2(dp[b0 + 2], dp[b0 + 3]);
+        int D2 = pack_bf2(dp[b0 + 4], dp[b0 + 5]);
+        int D3 = pack_bf2(dp[b0 + 6], dp[b0 + 7]);
+        int O1 = hi ? P2 : P0, O2 = hi ? P3 : P1;
+        int S1 = hi ? P0 : P2, S2 = hi ? P1 : P3;
+        int E1 = hi ? D2 : D0, E2 = hi ? D3 : D1;
+        int T1 = hi ? D0 : D2, T2 = hi ? D1 : D3;
+        intx2 ra = __builtin_amdgcn_permlane32_swap(S1, S2, false, false);
+        intx2 rb = __builtin_amdgcn_permlane32_swap(S2, S1, false, false);
+        intx2 rc = __builtin_amdgcn_permlane32_swap(T1, T2, false, false);
+        intx2 rd = __builtin_amdgcn_permlane32_swap(T2, T1, false, false);
+        int X1 = hi ? rb[0] : ra[1], X2 = hi ? ra[0] : rb[1];
+        int Y1 = hi ? rd[0] : rc[1], Y2 = hi ? rc[0] : rd[1];
+        intx4 paw, daw;
+        paw[0] = hi ? X1 : O1; paw[1] = hi ? X2 : O2;
+        paw[2] = hi ? O1 : X1; paw[3] = hi ? O2 : X2;
+        daw[0] = hi ? Y1 : E1; daw[1] = hi ? Y2 : E2;
+        daw[2] = hi ? E1 : Y1; daw[3] = hi ? E2 : Y2;
+        shortx8 pfrag = *reinterpret_cast<shortx8*>(&paw);
+        shortx8 dsfrag = *reinterpret_cast<shortx8*>(&daw);
 #pragma unroll
         for (int dt = 0; dt < NDT; ++dt) {
-          shortx8 bf = *reinterpret_cast<const shortx8*>(
-              t_lds + lds_off32(dt * 32 + l32, ((tq * 32 + ks * 16) + hi * 8) * 2, TR_RS));
-          acc[dt] = mfma32_bf16(pa, bf, acc[dt]);
+          shortx8 dofr = *reinterpret_cast<const shortx8*>(
+              dot_lds + lds_off32(dt * 32 + l32, ((tq * 32 + ks * 16) + hi * 8) * 2, TR_RS));
+          dvacc[dt] = mfma32_bf16(dofr, pfrag, dvacc[dt]);
+          shortx8 qtfr = *reinterpret_cast<const shortx8*>(
+              qt_lds + lds_off32(dt * 32 + l32, ((tq * 32 + ks * 16) + hi * 8) * 2, TR_RS));
+          dkacc[dt] = mfma32_bf16(qtfr, dsfrag, dkacc[dt]);
         }
       }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // reads done before tq+1 overwrites
     }
     __syncthreads();
   }
 
-  // epilogue
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    int rkv = (r & 3) + 8 * (r >> 2) + 4 * hi;
-    int row = kv0w + rkv;
-    if (row >= Skv) continue;
+  // ---- epilogue: transposed accumulators -> natural dK/dV rows ------------
+  // lane owns kv row kv_lane; d = dt*32 + (r&3) + 8*(r>>2) + 4*hi, so each
+  // (dt, r>>2) group of 4 regs is d-contiguous -> packed 8 B stores.
+  if (kv_lane < Skv) {
+    const long long vrow = outbase + (long long)kv_lane * dk_ss;
 #pragma unroll
     for (int dt = 0; dt < NDT; ++dt)
-      outg[outbase + (long long)row * dk_ss + dt * 32 + l32] = f2bf(acc[dt][r]);
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        shortx4 pv, pk;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          pv[i] = f2bf(dvacc[dt][g * 4 + i]);
+          pk[i] = f2bf(dkacc[dt][g * 4 + i]);
+        }
+        int d = dt * 32 + 8 * g + 4 * hi;
+        *reinterpret_cast<shortx4*>(dvg + vrow + d) = pv;
+        *reinterpret_cast<shortx4*>(dkg + vrow + d) = pk;
+      }
   }
 }
 
@@ -769,7 +794,13 @@ void flash_attn_bwd_dkv32(const void* dout, const void* q, const void* k,
                           const int64_t* dks, hipStream_t s) {
   dim3 grid((unsigned)cdiv((int)skv, 128), (unsigned)(b * h));
   dim3 blk(256);
-#define FDKV(D, C)                                                                do {                                                                              hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C, false>), grid, blk, 0, s,                            (const short*)dout, (const short*)q, (const short*)k,                           (const short*)v, lse, delta, (short*)dv, (int)b, (int)h,                        (int)sq, (int)skv, scale, qs[0], qs[1], qs[2],                                  ks[0], ks[1], ks[2], dos[0], dos[1], dos[2],                                    dks[0], dks[1], dks[2]);                                     hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C, true>), grid, blk, 0, s,                             (const short*)dout, (const short*)q, (const short*)k,                           (const short*)v, lse, delta, (short*)dk, (int)b, (int)h,                        (int)sq, (int)skv, scale, qs[0], qs[1], qs[2],                                  ks[0], ks[1], ks[2], dos[0], dos[1], dos[2],                                    dks[0], dks[1], dks[2]);                                   } while (0)
+#define FDKV(D, C)                                                             \
+  hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C>), grid, blk, 0, s,             \
+                     (const short*)dout, (const short*)q, (const short*)k,     \
+                     (const short*)v, lse, delta, (short*)dk, (short*)dv,      \
+                     (int)b, (int)h, (int)sq, (int)skv, scale,                 \
+                     qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],                 \
+                     dos[0], dos[1], dos[2], dks[0], dks[1], dks[2])
   if (dh == 128) { if (causal) FDKV(128, true); else FDKV(128, false); }
   else           { if (causal) FDKV(64, true);  else FDKV(64, false); }
 #undef FDKV

@@ -765,14 +765,24 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
 
   dim3 gkv((unsigned)cdiv((int)skv, 128), (unsigned)(b * h));
   dim3 gq((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
+  // PA_FA_DKV16=1 selects the 16x16 dKV kernel (A/B probe); default is the
+  // combined 32x32 transposed-output kernel
+  static const bool use_dkv16 = [] {
+    const char* e = getenv("PA_FA_DKV16");
+    return e && e[0] == '1';
+  }();
 #define FAB(D, C)                                                              \
   do {                                                                         \
-    hipLaunchKernelGGL((fa_bwd_dkv_kernel<D, C>), gkv, dim3(512), 0, s,        \
-                       (const short*)dout, (const short*)q, (const short*)k,   \
-                       (const short*)v, lse, delta, (short*)dk, (short*)dv,    \
-                       (int)b, (int)h, (int)sq, (int)skv, scale,               \
-                       qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],               \
-                       dos[0], dos[1], dos[2], dks[0], dks[1], dks[2]);        \
+    if (use_dkv16)                                                             \
+      hipLaunchKernelGGL((fa_bwd_dkv_kernel<D, C>), gkv, dim3(512), 0, s,      \
+                         (const short*)dout, (const short*)q, (const short*)k, \
+                         (const short*)v, lse, delta, (short*)dk, (short*)dv,  \
+                         (int)b, (int)h, (int)sq, (int)skv, scale,             \
+                         qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],             \
+                         dos[0], dos[1], dos[2], dks[0], dks[1], dks[2]);      \
+    else                                                                       \
+      flash_attn_bwd_dkv32(dout, q, k, v, lse, delta, dk, dv, b, h, sq, skv,   \
+                           dh, scale, causal, qs, ks, dos, dks, s);            \
     flash_attn_bwd_dq32(dout, q, k, v, lse, delta, dq, b, h, sq, skv, dh,     \
                         scale, causal, qs, ks, dos, dqs, s);                   \
   } while (0)
