@@ -540,27 +540,25 @@ void flash_attn_bwd_dq32(const void* dout, const void* q, const void* k,
 }
 
 // ---------------------------------------------------------------------------
-// backward dV + dK, 32x32 combined variant (v2).
+// backward dV / dK, 32x32 split variant (v3).
 //
-// Orientation flip vs the retired split version: compute P = mfma(Q, K)
-// and dP = mfma(dO, V) so the OUTPUT COLUMN (kv) sits in the lane dim.
-// P / dS then feed the [d x q][q x kv] products dV^T = dO^T P and
-// dK^T = Q^T dS as B-fragments assembled with the same static-pack +
-// permlane32_swap exchange as the forward's PV step -- the lane<->reg
-// transpose that forced the old per-wave LDS round trip never arises.
-// lse/delta live in the q = REG dim: staged to LDS once per q tile and
-// read back as 16 broadcast (wave-uniform-per-half) ds_reads.
-// K/V B-fragments are loaded once into registers (wave owns 32 kv rows);
-// Q/dO are staged natural (for P/dP A-frags, lane = q) and transposed
-// via bank-rotated writes (for dV^T/dK^T A-frags, lane = d).
-// Outputs are accumulated transposed and written with 8 B packed stores.
+// Orientation: P = mfma(Q, K) and dP = mfma(dO, V) put the kv column in
+// the LANE dim, so P / dS feed dV^T = dO^T P and dK^T = Q^T dS as
+// B-fragments via the forward's static-pack + permlane32_swap exchange --
+// no lane<->reg transpose, no LDS round trip (the v1 split's failure).
+// Split into dV and dK kernels so each fits 2 waves/SIMD (the combined
+// variant needs 128 accumulator VGPRs and runs at occupancy 1, measured
+// slower than the 16x16 kernel).  Q / dO A-fragments are read straight
+// from global (all 4 waves read the same rows -> L2-served); only the
+// transposed operand (dO^T or Q^T) is staged in LDS, double-buffered.
+// Outputs are accumulated transposed and written as packed 8 B stores.
 // ---------------------------------------------------------------------------
-template <int D, bool CAUSAL>
-__launch_bounds__(256, 1)
+template <int D, bool CAUSAL, bool IS_DK>
+__launch_bounds__(256, 2)
 __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* __restrict__ qg,
                                     const short* __restrict__ kg, const short* __restrict__ vg,
                                     const float* __restrict__ lseg, const float* __restrict__ deltag,
-                                    short* __restrict__ dkg, short* __restrict__ dvg,
+                                    short* __restrict__ outg,
                                     int B, int H, int Sq, int Skv, float scale,
                                     long long q_sb, long long q_sh, long long q_ss,
                                     long long k_sb, long long k_sh, long long k_ss,
@@ -569,16 +567,14 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
   constexpr int NW = 4;
   constexpr int NT = NW * 64;
   constexpr int KVB = NW * 32;     // 128 kv rows / block
-  constexpr int QT = 64;           // q tile
+  constexpr int QT = 32;           // q tile (one 32-q mfma tile)
   constexpr int NKS = D / 16;
   constexpr int NDT = D / 32;
-  constexpr unsigned NAT_RS = D * 2;
-  constexpr unsigned TR_RS = QT * 2;
-  __shared__ char q_lds[QT * D * 2];        // natural Q  (P A-frags)
-  __shared__ char d_lds[QT * D * 2];        // natural dO (dP A-frags)
-  __shared__ char qt_lds[D * QT * 2];       // Q^T  (dK^T A-frags)
-  __shared__ char dot_lds[D * QT * 2];      // dO^T (dV^T A-frags)
-  __shared__ float lse_s[QT], delta_s[QT];
+  constexpr unsigned TR_RS = 64 * 2;  // transposed rows padded to 64 cols
+  // transposed operand (dO^T for dV, Q^T for dK): rows = d (128), padded to
+  // 64 bf16 columns so lds_off32's 8-row XOR swizzle stays in-row
+  __shared__ char t_lds[2][D * 64 * 2];
+  __shared__ float stat_s[2][2 * QT];  // [lse | delta]
 
   const int kvblk = blockIdx.x;
   const int bh = blockIdx.y;
@@ -596,10 +592,10 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
   const int l32 = lane & 31;
   const int hi = lane >> 5;
   const int kv0w = kv0 + wq * 32;
-  const int kv_lane = kv0w + l32;            // this lane's kv column
+  const int kv_lane = kv0w + l32;
 
-  // K and V as B-fragments (lane = kv row)
-  shortx8 kf[NKS], vf[NKS];
+  // K (and V for dK) as B-fragments (lane = kv row), loaded once
+  shortx8 kf[NKS], vf[IS_DK ? NKS : 1];
   {
     bool ok = kv_lane < Skv;
 #pragma unroll
@@ -607,181 +603,143 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
       if (ok) {
         kf[ks] = *reinterpret_cast<const shortx8*>(
             kg + kvbase + (long long)kv_lane * k_ss + ks * 16 + hi * 8);
-        vf[ks] = *reinterpret_cast<const shortx8*>(
-            vg + kvbase + (long long)kv_lane * k_ss + ks * 16 + hi * 8);
+        if (IS_DK)
+          vf[ks] = *reinterpret_cast<const shortx8*>(
+              vg + kvbase + (long long)kv_lane * k_ss + ks * 16 + hi * 8);
       } else {
-        for (int i = 0; i < 8; ++i) { kf[ks][i] = 0; vf[ks][i] = 0; }
+        for (int i = 0; i < 8; ++i) { kf[ks][i] = 0; if (IS_DK) vf[ks][i] = 0; }
       }
     }
   }
 
-  floatx16 dvacc[NDT], dkacc[NDT];
+  floatx16 acc[NDT];
 #pragma unroll
   for (int dt = 0; dt < NDT; ++dt)
 #pragma unroll
-    for (int r = 0; r < 16; ++r) { dvacc[dt][r] = 0.f; dkacc[dt][r] = 0.f; }
+    for (int r = 0; r < 16; ++r) acc[dt][r] = 0.f;
 
-  int n_row[QT * D / (NT * 8)], n_colp[QT * D / (NT * 8)];
-#pragma unroll
-  for (int it = 0; it < QT * D / (NT * 8); ++it) {
-    int flat = it * NT * 8 + tid * 8;
-    int row = flat / D, col = flat % D;
-    n_row[it] = row;
-    n_colp[it] = col ^ ((row & 7) << 3);
-  }
-
-  const int q_start = CAUSAL ? (kv0 / QT) * QT : 0;
-  for (int q0 = q_start; q0 < Sq; q0 += QT) {
-    // ---- stage: Q,dO natural (async DMA) + Q^T,dO^T (rotated writes) ------
+  // stage the transposed operand tile (rotated scalar writes) + stats
+  auto stage = [&](int buf, int q0) {
     if (tid < QT) {
       int qa = q0 + tid;
-      lse_s[tid] = (qa < Sq) ? lseg[lse_base + qa] : 1e30f;
-      delta_s[tid] = (qa < Sq) ? deltag[lse_base + qa] : 0.f;
+      stat_s[buf][tid] = (qa < Sq) ? lseg[lse_base + qa] : 1e30f;
+      if (IS_DK)
+        stat_s[buf][QT + tid] = (qa < Sq) ? deltag[lse_base + qa] : 0.f;
     }
-    if (q0 + QT <= Sq) {
-#pragma unroll
-      for (int it = 0; it < QT * D / (NT * 8); ++it) {
-        const short* qsrc = qg + qbase + (long long)(q0 + n_row[it]) * q_ss + n_colp[it];
-        const short* dsrc = dog + dobase + (long long)(q0 + n_row[it]) * do_ss + n_colp[it];
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)qsrc,
-            (__attribute__((address_space(3))) unsigned int*)(q_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
-            16, 0, 0);
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)dsrc,
-            (__attribute__((address_space(3))) unsigned int*)(d_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
-            16, 0, 0);
+    const int rot = tid & 7;
+    for (int flat = tid * 8; flat < QT * D; flat += NT * 8) {
+      int row = flat / D, col = flat % D;   // row = q, col = d
+      shortx8 v;
+      if (q0 + row < Sq) {
+        const short* src = IS_DK
+            ? qg + qbase + (long long)(q0 + row) * q_ss + col
+            : dog + dobase + (long long)(q0 + row) * do_ss + col;
+        v = *reinterpret_cast<const shortx8*>(src);
+      } else {
+        for (int i = 0; i < 8; ++i) v[i] = 0;
       }
-      const int rot = tid & 7;
-      for (int flat = tid * 8; flat < QT * D; flat += NT * 8) {
-        int row = flat / D, col = flat % D;
-        shortx8 qv = *reinterpret_cast<const shortx8*>(
-            qg + qbase + (long long)(q0 + row) * q_ss + col);
-        shortx8 dv = *reinterpret_cast<const shortx8*>(
-            dog + dobase + (long long)(q0 + row) * do_ss + col);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int i = (j + rot) & 7;
-          *reinterpret_cast<short*>(qt_lds + lds_off32(col + i, row * 2, TR_RS)) = qv[i];
-          *reinterpret_cast<short*>(dot_lds + lds_off32(col + i, row * 2, TR_RS)) = dv[i];
-        }
-      }
-    } else {
-      const int rot = tid & 7;
-      for (int flat = tid * 8; flat < QT * D; flat += NT * 8) {
-        int row = flat / D, col = flat % D;
-        shortx8 qv, dv;
-        if (q0 + row < Sq) {
-          qv = *reinterpret_cast<const shortx8*>(qg + qbase + (long long)(q0 + row) * q_ss + col);
-          dv = *reinterpret_cast<const shortx8*>(dog + dobase + (long long)(q0 + row) * do_ss + col);
-        } else {
-          for (int i = 0; i < 8; ++i) { qv[i] = 0; dv[i] = 0; }
-        }
-        *reinterpret_cast<shortx8*>(q_lds + lds_off32(row, col * 2, NAT_RS)) = qv;
-        *reinterpret_cast<shortx8*>(d_lds + lds_off32(row, col * 2, NAT_RS)) = dv;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int i = (j + rot) & 7;
-          *reinterpret_cast<short*>(qt_lds + lds_off32(col + i, row * 2, TR_RS)) = qv[i];
-          *reinterpret_cast<short*>(dot_lds + lds_off32(col + i, row * 2, TR_RS)) = dv[i];
-        }
+      for (int j = 0; j < 8; ++j) {
+        int i = (j + rot) & 7;
+        *reinterpret_cast<short*>(t_lds[buf] + lds_off32(col + i, row * 2, TR_RS)) = v[i];
       }
     }
-    __syncthreads();
+  };
 
+  const int q_start = CAUSAL ? (kv0 / QT) * QT : 0;
+  stage(0, q_start);
+  __syncthreads();
+  int cur = 0;
+  for (int q0 = q_start; q0 < Sq; q0 += QT) {
+    if (q0 + QT < Sq) stage(cur ^ 1, q0 + QT);
+
+    // Q (and dO for dK) A-fragments straight from global: lane = q row
+    const int q_lane = q0 + l32;
+    const bool qok = q_lane < Sq;
+    floatx16 st, dp;
 #pragma unroll
-    for (int tq = 0; tq < 2; ++tq) {
-      // causal: whole 32-q tile below this wave's kv rows contributes nothing
-      if (CAUSAL && q0 + tq * 32 + 31 < kv0w) continue;
-      // ---- P = Q K^T, dP = dO V^T (lanes = kv, regs = q) ------------------
-      floatx16 st, dp;
+    for (int r = 0; r < 16; ++r) { st[r] = 0.f; dp[r] = 0.f; }
 #pragma unroll
-      for (int r = 0; r < 16; ++r) { st[r] = 0.f; dp[r] = 0.f; }
-#pragma unroll
-      for (int ks = 0; ks < NKS; ++ks) {
-        shortx8 qfr = *reinterpret_cast<const shortx8*>(
-            q_lds + lds_off32(tq * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
-        st = mfma32_bf16(qfr, kf[ks], st);
-        shortx8 dfr = *reinterpret_cast<const shortx8*>(
-            d_lds + lds_off32(tq * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
-        dp = mfma32_bf16(dfr, vf[ks], dp);
+    for (int ks = 0; ks < NKS; ++ks) {
+      shortx8 qa;
+      if (qok)
+        qa = *reinterpret_cast<const shortx8*>(
+            qg + qbase + (long long)q_lane * q_ss + ks * 16 + hi * 8);
+      else
+        for (int i = 0; i < 8; ++i) qa[i] = 0;
+      st = mfma32_bf16(qa, kf[ks], st);
+      if (IS_DK) {
+        shortx8 da;
+        if (qok)
+          da = *reinterpret_cast<const shortx8*>(
+              dog + dobase + (long long)q_lane * do_ss + ks * 16 + hi * 8);
+        else
+          for (int i = 0; i < 8; ++i) da[i] = 0;
+        dp = mfma32_bf16(da, vf[ks], dp);
       }
-      // ---- P = exp(S*scale - lse[q]); dS = P (dP - delta[q]) --------------
-      const bool bnd = (q0 + QT > Sq) || (kv0w + 32 > Skv) ||
-                       (CAUSAL && kv0w + 32 > q0 + tq * 32);
+    }
+
+    // per-reg q stats + exp/mask.  For dV, P = exp(S*scale - lse[q]); for
+    // dK additionally dS = P (dP - delta[q]) * scale -- but P needs lse
+    // there too, so dK stages BOTH: lse comes from a second pass of
+    // broadcast loads straight from global (L2-hot after the dV kernel).
+    const bool bnd = (q0 + QT > Sq) || (kv0w + 32 > Skv) ||
+                     (CAUSAL && kv0w + 32 > q0);
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int rq = (r & 3) + 8 * (r >> 2) + 4 * hi;   // q-local (reg dim)
-        float lse_v = lse_s[tq * 32 + rq];                // broadcast read
-        float p = __expf(st[r] * scale - lse_v);
-        if (bnd) {
-          int q_abs = q0 + tq * 32 + rq;
-          if (q_abs >= Sq || kv_lane >= Skv || (CAUSAL && kv_lane > q_abs)) p = 0.f;
-        }
+    for (int r = 0; r < 16; ++r) {
+      const int rq = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const int q_abs = q0 + rq;
+      float p = __expf(st[r] * scale - stat_s[cur][rq]);
+      if (bnd) {
+        if (q_abs >= Sq || kv_lane >= Skv || (CAUSAL && kv_lane > q_abs)) p = 0.f;
+      }
+      if (IS_DK) {
+        st[r] = p * (dp[r] - stat_s[cur][QT + rq]) * scale;
+      } else {
         st[r] = p;
-        dp[r] = p * (dp[r] - delta_s[tq * 32 + rq]) * scale;
       }
-      // ---- dV^T += dO^T P ; dK^T += Q^T dS (B-frags via pack+permlane) ----
+    }
+
+    // ---- acc += (dO^T P) or (Q^T dS): B-frags via pack+permlane ----------
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        const int b0 = 8 * ks;
-        int P0 = pack_bf2(st[b0 + 0], st[b0 + 1]);
-        int P1 = pack_bf2(st[b0 + 2], st[b0 + 3]);
-        int P2 = pack_bf2(st[b0 + 4], st[b0 + 5]);
-        int P3 = pack_bf2(st[b0 + 6], st[b0 + 7]);
-        int D0 = pack_bf2(dp[b0 + 0], dp[b0 + 1]);
-        int D1 = pack_bf2(dp[b0 + 2], dp[b0 + 3]);
-        int D2 = pack_bf2(dp[b0 + 4], dp[b0 + 5]);
-        int D3 = pack_bf2(dp[b0 + 6], dp[b0 + 7]);
-        int O1 = hi ? P2 : P0, O2 = hi ? P3 : P1;
-        int S1 = hi ? P0 : P2, S2 = hi ? P1 : P3;
-        int E1 = hi ? D2 : D0, E2 = hi ? D3 : D1;
-        int T1 = hi ? D0 : D2, T2 = hi ? D1 : D3;
-        intx2 ra = __builtin_amdgcn_permlane32_swap(S1, S2, false, false);
-        intx2 rb = __builtin_amdgcn_permlane32_swap(S2, S1, false, false);
-        intx2 rc = __builtin_amdgcn_permlane32_swap(T1, T2, false, false);
-        intx2 rd = __builtin_amdgcn_permlane32_swap(T2, T1, false, false);
-        int X1 = hi ? rb[0] : ra[1], X2 = hi ? ra[0] : rb[1];
-        int Y1 = hi ? rd[0] : rc[1], Y2 = hi ? rc[0] : rd[1];
-        intx4 paw, daw;
-        paw[0] = hi ? X1 : O1; paw[1] = hi ? X2 : O2;
-        paw[2] = hi ? O1 : X1; paw[3] = hi ? O2 : X2;
-        daw[0] = hi ? Y1 : E1; daw[1] = hi ? Y2 : E2;
-        daw[2] = hi ? E1 : Y1; daw[3] = hi ? E2 : Y2;
-        shortx8 pfrag = *reinterpret_cast<shortx8*>(&paw);
-        shortx8 dsfrag = *reinterpret_cast<shortx8*>(&daw);
+    for (int ks = 0; ks < 2; ++ks) {
+      const int b0 = 8 * ks;
+      int P0 = pack_bf2(st[b0 + 0], st[b0 + 1]);
+      int P1 = pack_bf2(st[b0 + 2], st[b0 + 3]);
+      int P2 = pack_bf2(st[b0 + 4], st[b0 + 5]);
+      int P3 = pack_bf2(st[b0 + 6], st[b0 + 7]);
+      int O1 = hi ? P2 : P0, O2 = hi ? P3 : P1;
+      int S1 = hi ? P0 : P2, S2 = hi ? P1 : P3;
+      intx2 ra = __builtin_amdgcn_permlane32_swap(S1, S2, false, false);
+      intx2 rb = __builtin_amdgcn_permlane32_swap(S2, S1, false, false);
+      int X1 = hi ? rb[0] : ra[1], X2 = hi ? ra[0] : rb[1];
+      intx4 paw;
+      paw[0] = hi ? X1 : O1; paw[1] = hi ? X2 : O2;
+      paw[2] = hi ? O1 : X1; paw[3] = hi ? O2 : X2;
+      shortx8 frag = *reinterpret_cast<shortx8*>(&paw);
 #pragma unroll
-        for (int dt = 0; dt < NDT; ++dt) {
-          shortx8 dofr = *reinterpret_cast<const shortx8*>(
-              dot_lds + lds_off32(dt * 32 + l32, ((tq * 32 + ks * 16) + hi * 8) * 2, TR_RS));
-          dvacc[dt] = mfma32_bf16(dofr, pfrag, dvacc[dt]);
-          shortx8 qtfr = *reinterpret_cast<const shortx8*>(
-              qt_lds + lds_off32(dt * 32 + l32, ((tq * 32 + ks * 16) + hi * 8) * 2, TR_RS));
-          dkacc[dt] = mfma32_bf16(qtfr, dsfrag, dkacc[dt]);
-        }
+      for (int dt = 0; dt < NDT; ++dt) {
+        shortx8 tfr = *reinterpret_cast<const shortx8*>(
+            t_lds[cur] + lds_off32(dt * 32 + l32, (ks * 16 + hi * 8) * 2, TR_RS));
+        acc[dt] = mfma32_bf16(tfr, frag, acc[dt]);
       }
     }
     __syncthreads();
+    cur ^= 1;
   }
 
-  // ---- epilogue: transposed accumulators -> natural dK/dV rows ------------
-  // lane owns kv row kv_lane; d = dt*32 + (r&3) + 8*(r>>2) + 4*hi, so each
-  // (dt, r>>2) group of 4 regs is d-contiguous -> packed 8 B stores.
+  // ---- epilogue: transposed accumulators -> natural rows ------------------
   if (kv_lane < Skv) {
-    const long long vrow = outbase + (long long)kv_lane * dk_ss;
+    const long long orow = outbase + (long long)kv_lane * dk_ss;
 #pragma unroll
     for (int dt = 0; dt < NDT; ++dt)
 #pragma unroll
       for (int g = 0; g < 4; ++g) {
-        shortx4 pv, pk;
+        shortx4 pv;
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          pv[i] = f2bf(dvacc[dt][g * 4 + i]);
-          pk[i] = f2bf(dkacc[dt][g * 4 + i]);
-        }
+        for (int i = 0; i < 4; ++i) pv[i] = f2bf(acc[dt][g * 4 + i]);
         int d = dt * 32 + 8 * g + 4 * hi;
-        *reinterpret_cast<shortx4*>(dvg + vrow + d) = pv;
-        *reinterpret_cast<shortx4*>(dkg + vrow + d) = pk;
+        *reinterpret_cast<shortx4*>(outg + orow + d) = pv;
       }
   }
 }
@@ -795,12 +753,20 @@ void flash_attn_bwd_dkv32(const void* dout, const void* q, const void* k,
   dim3 grid((unsigned)cdiv((int)skv, 128), (unsigned)(b * h));
   dim3 blk(256);
 #define FDKV(D, C)                                                             \
-  hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C>), grid, blk, 0, s,             \
-                     (const short*)dout, (const short*)q, (const short*)k,     \
-                     (const short*)v, lse, delta, (short*)dk, (short*)dv,      \
-                     (int)b, (int)h, (int)sq, (int)skv, scale,                 \
-                     qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],                 \
-                     dos[0], dos[1], dos[2], dks[0], dks[1], dks[2])
+  do {                                                                         \
+    hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C, false>), grid, blk, 0, s,    \
+                       (const short*)dout, (const short*)q, (const short*)k,   \
+                       (const short*)v, lse, delta, (short*)dv,                \
+                       (int)b, (int)h, (int)sq, (int)skv, scale,               \
+                       qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],               \
+                       dos[0], dos[1], dos[2], dks[0], dks[1], dks[2]);        \
+    hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C, true>), grid, blk, 0, s,     \
+                       (const short*)dout, (const short*)q, (const short*)k,   \
+                       (const short*)v, lse, delta, (short*)dk,                \
+                       (int)b, (int)h, (int)sq, (int)skv, scale,               \
+                       qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],               \
+                       dos[0], dos[1], dos[2], dks[0], dks[1], dks[2]);        \
+  } while (0)
   if (dh == 128) { if (causal) FDKV(128, true); else FDKV(128, false); }
   else           { if (causal) FDKV(64, true);  else FDKV(64, false); }
 #undef FDKV
