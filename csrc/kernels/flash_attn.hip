@@ -765,11 +765,13 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
 
   dim3 gkv((unsigned)cdiv((int)skv, 128), (unsigned)(b * h));
   dim3 gq((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
-  // PA_FA_DKV16=1 selects the 16x16 dKV kernel (A/B probe); default is the
-  // combined 32x32 transposed-output kernel
+  // The 16x16 dKV kernel is the default: both 32x32 rewrites measured
+  // slower (combined/transposed-output: 146 TF bwd at occupancy 1;
+  // occupancy-2 split with global A-frags: 95 TF; 16x16 baseline: 168 TF).
+  // PA_FA_DKV32=1 opts into the 32x32 path for future experiments.
   static const bool use_dkv16 = [] {
-    const char* e = getenv("PA_FA_DKV16");
-    return e && e[0] == '1';
+    const char* e = getenv("PA_FA_DKV32");
+    return !(e && e[0] == '1');
   }();
 #define FAB(D, C)                                                              \
   do {                                                                         \
