@@ -275,7 +275,7 @@ from .hapi import Model  # noqa: F401
 from .param_attr import ParamAttr  # noqa: F401
 from .autograd import grad, no_grad, enable_grad, set_grad_enabled, is_grad_enabled  # noqa: F401
 
-disable_static = lambda *a, **k: None  # dygraph is the default mode
+disable_static = static.disable_static
 enable_static = static.enable_static
 in_dynamic_mode = lambda: not static._static_mode
 

@@ -26,7 +26,11 @@ class Optimizer:
                  grad_clip=None, name=None, multi_precision=None):
         self._lr = learning_rate
         if parameters is None:
-            raise ValueError("parameters must be given in dygraph mode")
+            # static-graph style: parameters bound later by minimize()
+            from .. import static as _static
+            if not _static.in_static_mode():
+                raise ValueError("parameters must be given in dygraph mode")
+            parameters = []
         self._param_groups = list(parameters)
         if self._param_groups and isinstance(self._param_groups[0], dict):
             self._params = [p for g in self._param_groups for p in g["params"]]
@@ -126,6 +130,25 @@ class Optimizer:
 
     def _apply_one(self, p):  # pragma: no cover - abstract
         raise NotImplementedError
+
+    # -- static-graph entry points (reference: optimizer.minimize) -----------
+    def minimize(self, loss, startup_program=None, parameters=None,
+                 no_grad_set=None):
+        from .. import static as _static
+        if isinstance(loss, _static.Var):
+            _static.default_main_program().train_ops.append((loss, self))
+            return None, None
+        loss.backward()
+        self.step()
+        self.clear_grad()
+        return None, None
+
+    def _static_step(self, prog):
+        """One optimizer application over the program's parameters."""
+        if not self._params:
+            self._params = [v.tensor for v in prog.params]
+        self.step()
+        self.clear_grad()
 
     # -- state dict (feeds .pdopt format) -------------------------------------
     def state_dict(self):

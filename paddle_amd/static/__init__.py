@@ -1,12 +1,24 @@
-"""paddle.static facade (reference: python/paddle/static/, base/executor.py).
+"""paddle.static: deferred-execution graph over the eager substrate.
 
-Round-1 scope: an executor that runs captured dygraph programs (the
-jit.to_static capture path) plus the data/feed API shape.  The full
-PIR-style instruction scheduler is tracked in SURVEY.md §2 as phase-7
-work; training uses the dygraph path.
+Reference surface: python/paddle/static/ (Program, Executor, data,
+default_main_program, nn.fc) and base/executor.py:1234 (Executor.run
+with feed/fetch).
+
+MI355X-first design: instead of rebuilding the reference's PIR program
+translator + instruction scheduler, the static API records a DEFERRED
+GRAPH of torch ops.  `paddle.static.data` returns a symbolic Var;
+every torch function touching a Var (all paddle_amd ops dispatch to
+torch functions) is intercepted via `__torch_function__` and recorded
+as a graph node.  `Executor.run` binds feeds, evaluates the graph
+(memoized per run) through the SAME HIP kernels as dygraph, and runs
+the recorded optimizer step.  Semantics match the reference's
+build-once / run-many static workflow; execution is eager per run,
+which on MI355X is the right call: kernels are already fused and
+hipGraph capture can wrap the run when launch-bound.
 """
 from __future__ import annotations
 
+import numpy as np
 import torch
 
 from .. import framework
@@ -28,17 +40,119 @@ def in_static_mode():
     return _static_mode
 
 
+class Var:
+    """Symbolic node in a static Program.
+
+    kind: "data" (bound from feed), "param" (persistent tensor),
+    "op" (fn over input Vars/constants).
+    """
+
+    def __init__(self, kind, name=None, shape=None, dtype=None, fn=None,
+                 args=None, kwargs=None, tensor=None):
+        self.kind = kind
+        self.name = name
+        self.shape = shape
+        self.dtype = dtype
+        self.fn = fn
+        self.args = args or ()
+        self.kwargs = kwargs or {}
+        self.tensor = tensor          # params: the persistent torch tensor
+        self.stop_gradient = kind != "param"
+
+    # -- evaluation ---------------------------------------------------------
+    def _eval(self, env, memo):
+        key = id(self)
+        if key in memo:
+            return memo[key]
+        if self.kind == "data":
+            if self.name not in env:
+                raise KeyError(f"feed missing for data var '{self.name}'")
+            v = env[self.name]
+        elif self.kind == "param":
+            v = self.tensor
+        else:
+            args = [a._eval(env, memo) if isinstance(a, Var) else a
+                    for a in self.args]
+            kwargs = {k: (v._eval(env, memo) if isinstance(v, Var) else v)
+                      for k, v in self.kwargs.items()}
+            v = self.fn(*args, **kwargs)
+        memo[key] = v
+        return v
+
+    # -- interception: any torch fn over a Var records an op node -----------
+    @classmethod
+    def __torch_function__(cls, func, types, args=(), kwargs=None):
+        return Var("op", fn=func, args=args, kwargs=kwargs or {})
+
+    # python operators route through torch so they are captured too
+    def _binop(self, other, fn, swap=False):
+        a, b = (other, self) if swap else (self, other)
+        return Var("op", fn=fn, args=(a, b))
+
+    def __add__(self, o):
+        return self._binop(o, torch.add)
+
+    __radd__ = __add__
+
+    def __sub__(self, o):
+        return self._binop(o, torch.subtract)
+
+    def __rsub__(self, o):
+        return self._binop(o, torch.subtract, swap=True)
+
+    def __mul__(self, o):
+        return self._binop(o, torch.multiply)
+
+    __rmul__ = __mul__
+
+    def __truediv__(self, o):
+        return self._binop(o, torch.divide)
+
+    def __rtruediv__(self, o):
+        return self._binop(o, torch.divide, swap=True)
+
+    def __matmul__(self, o):
+        return self._binop(o, torch.matmul)
+
+    def __neg__(self):
+        return Var("op", fn=torch.neg, args=(self,))
+
+    def __getitem__(self, idx):
+        return Var("op", fn=lambda t, i: t[i], args=(self, idx))
+
+    def reshape(self, shape):
+        return Var("op", fn=torch.reshape, args=(self, shape))
+
+    def astype(self, dtype):
+        dt = framework.convert_dtype(dtype)
+        return Var("op", fn=lambda t, d: t.to(d), args=(self, dt))
+
+    def __repr__(self):
+        return f"Var(kind={self.kind}, name={self.name}, shape={self.shape})"
+
+
 class Program:
-    """Minimal Program facade: records a traced callable + params."""
+    """Recorded graph + optimizer step (reference: base/framework.py Program)."""
 
     def __init__(self):
-        self.fn = None
-        self.feed_names = []
-        self.fetch_names = []
+        self.params = []              # persistent param Vars
+        self.train_ops = []           # [(loss_var, optimizer)]
+        self.initializers = []        # callables run by the startup program
 
     def clone(self, for_test=False):
         import copy
-        return copy.copy(self)
+        p = copy.copy(self)
+        if for_test:
+            p = copy.copy(self)
+            p.train_ops = []
+        return p
+
+    # reference API parity
+    def global_block(self):
+        return self
+
+    def all_parameters(self):
+        return list(self.params)
 
 
 _default_main = Program()
@@ -53,36 +167,110 @@ def default_startup_program():
     return _default_startup
 
 
-def data(name, shape, dtype="float32", lod_level=0):
-    class _Var:
-        def __init__(self, name, shape, dtype):
-            self.name, self.shape, self.dtype = name, shape, dtype
+class program_guard:
+    def __init__(self, main_program, startup_program=None):
+        self.main = main_program
+        self.startup = startup_program
 
-    return _Var(name, shape, dtype)
+    def __enter__(self):
+        global _default_main, _default_startup
+        self._saved = (_default_main, _default_startup)
+        _default_main = self.main
+        if self.startup is not None:
+            _default_startup = self.startup
+        return self
+
+    def __exit__(self, *a):
+        global _default_main, _default_startup
+        _default_main, _default_startup = self._saved
+
+
+def data(name, shape, dtype="float32", lod_level=0):
+    return Var("data", name=name, shape=shape,
+               dtype=framework.convert_dtype(dtype))
+
+
+def create_parameter(shape, dtype="float32", initializer=None, program=None):
+    prog = program or _default_main
+    dt = framework.convert_dtype(dtype)
+    t = torch.empty(shape, dtype=dt)
+    v = Var("param", shape=shape, dtype=dt, tensor=t)
+    t.requires_grad_(True)
+
+    def init():
+        with torch.no_grad():
+            if initializer is not None:
+                initializer(t)
+            else:
+                torch.nn.init.xavier_normal_(t) if t.dim() >= 2 else t.zero_()
+    prog.initializers.append(init)
+    prog.params.append(v)
+    return v
+
+
+class nn:
+    """paddle.static.nn subset (reference: python/paddle/static/nn/common.py)."""
+
+    @staticmethod
+    def fc(x, size, activation=None, name=None, num_flatten_dims=1,
+           weight_attr=None, bias_attr=None):
+        in_dim = x.shape[-1]
+        if in_dim is None or in_dim < 0:
+            raise ValueError("static.nn.fc needs a known last-dim size")
+        w = create_parameter([in_dim, size])
+        b = create_parameter([size], initializer=lambda t: t.zero_())
+        out = Var("op", fn=torch.addmm, args=(b, x, w))
+        if activation == "relu":
+            out = Var("op", fn=torch.relu, args=(out,))
+        elif activation == "softmax":
+            out = Var("op", fn=lambda t: torch.softmax(t, -1), args=(out,))
+        elif activation == "tanh":
+            out = Var("op", fn=torch.tanh, args=(out,))
+        out.shape = (list(x.shape[:-1]) + [size]) if x.shape else [None, size]
+        return out
 
 
 class Executor:
-    """reference: python/paddle/base/executor.py:1234.  Runs a captured
-    TracedProgram (from paddle.jit) or a plain callable with feeds."""
+    """reference: python/paddle/base/executor.py:1234."""
 
     def __init__(self, place=None):
         self.place = place
+        self._dev = (torch.device("cuda") if (place is not None and
+                     "GPU" in type(place).__name__ and torch.cuda.is_available())
+                     else torch.device("cpu"))
 
     def run(self, program=None, feed=None, fetch_list=None, return_numpy=True):
+        prog = program if isinstance(program, Program) else _default_main
+        # startup program: run pending initializers
+        if prog is _default_startup or (isinstance(program, Program) and
+                                        program.initializers and not program.train_ops
+                                        and fetch_list is None and not feed):
+            for init in (prog.initializers or _default_main.initializers):
+                init()
+            prog.initializers = []
+            return []
         feed = feed or {}
-        prog = program or _default_main
+        env = {}
+        for k, v in feed.items():
+            t = v if isinstance(v, torch.Tensor) else torch.as_tensor(np.asarray(v))
+            env[k] = t.to(self._dev)
+        memo = {}
+        # legacy program captured via paddle.jit
         fn = getattr(prog, "fn", None)
-        if fn is None:
-            raise RuntimeError(
-                "static Executor.run needs a program captured via paddle.jit "
-                "(dygraph-first build; see SURVEY.md §7 step 7)")
-        tensors = {k: (torch.as_tensor(v) if not isinstance(v, torch.Tensor) else v)
-                   for k, v in feed.items()}
-        outs = fn(**tensors)
-        if not isinstance(outs, (list, tuple)):
-            outs = [outs]
+        if fn is not None:
+            outs = fn(**env)
+            outs = outs if isinstance(outs, (list, tuple)) else [outs]
+        else:
+            for loss_var, opt in prog.train_ops:
+                loss = loss_var._eval(env, memo)
+                loss.backward()
+                opt._static_step(prog)
+            outs = []
+            for f in (fetch_list or []):
+                outs.append(f._eval(env, memo) if isinstance(f, Var) else f)
         if return_numpy:
-            outs = [o.detach().cpu().numpy() if isinstance(o, torch.Tensor) else o for o in outs]
+            outs = [o.detach().cpu().numpy() if isinstance(o, torch.Tensor) else o
+                    for o in outs]
         return list(outs)
 
 
@@ -98,10 +286,8 @@ class InputSpec:
         return cls(list(tensor.shape), str(tensor.dtype), name)
 
 
-def save_inference_model(path_prefix, feed_vars, fetch_vars, executor, program=None,
-                         **kwargs):
-    """Minimal .pdmodel save: stores the jit-scripted program if available."""
-    from .. import jit
+def save_inference_model(path_prefix, feed_vars, fetch_vars, executor,
+                         program=None, **kwargs):
     raise NotImplementedError(
         "save_inference_model requires the jit capture path; use paddle.jit.save")
 
