@@ -496,6 +496,37 @@ class ShardedAdamW:
             sd[f"unit{i}_v"] = self._v[i]
         return sd
 
+    # -- distcp resharding hooks ---------------------------------------------
+    # each unit's master/m/v is the flat slice [rank*shard_size, +shard_size)
+    # of the unit's padded global buffer -- exactly the shard_info model of
+    # distributed.checkpoint (reference: checkpoint/save_state_dict.py)
+    def sharded_state_dict(self):
+        sd = {"step": self.step_count}
+        info = {}
+        for i, u in enumerate(self.model._units):
+            off = u.rank * u.shard_size
+            for nm, t in (("master", u.master), ("m", self._m[i]),
+                          ("v", self._v[i])):
+                k = f"unit{i}_{nm}"
+                sd[k] = t
+                info[k] = {"global_numel": u.total, "offset": off}
+        return sd, info
+
+    def save_sharded(self, path):
+        from ...distributed import checkpoint as dcp
+        sd, info = self.sharded_state_dict()
+        dcp.save_state_dict(sd, path, shard_info=info)
+
+    def load_sharded(self, path):
+        from ...distributed import checkpoint as dcp
+        sd, info = self.sharded_state_dict()
+        dcp.load_state_dict(sd, path, shard_info=info)
+        self.step_count = int(sd["step"])
+        # masters changed: refresh the bf16 model shards
+        for u in self.model._units:
+            with torch.no_grad():
+                u.shard.copy_(u.master.to(u.shard.dtype))
+
     def set_state_dict(self, sd):
         self.step_count = int(sd.get("step", 0))
         for i, u in enumerate(self.model._units):

@@ -178,3 +178,61 @@ def test_vocab_parallel_ce_gloo():
         torch.nn.functional.cross_entropy(full, labels).backward()
         assert torch.allclose(local.grad, full.grad[:, r*10:(r+1)*10], atol=1e-5)
     """)
+
+
+def test_sharded_checkpoint_reshard_w2_to_w1(tmp_path):
+    """Save the ZeRO-3 optimizer state on 2 ranks, restore on 1 rank:
+    load_state_dict reassembles each unit's flat buffers from the
+    overlapping saved slices (reference: distributed/checkpoint/
+    load_state_dict.py resharding)."""
+    ckpt = str(tmp_path / "distcp")
+    run_dist(f"""
+        import torch
+        import paddle_amd as paddle
+        paddle.distributed.init_parallel_env()
+        from paddle_amd.distributed.fleet.sharding import (GroupShardedStage3,
+                                                           ShardedAdamW)
+
+        torch.manual_seed(0)   # same init on both ranks (sync_init also covers it)
+        m = paddle.nn.Sequential(paddle.nn.Linear(16, 32), paddle.nn.Linear(32, 16))
+        wrapped = GroupShardedStage3(m, device=torch.device("cpu"))
+        opt = ShardedAdamW(wrapped, learning_rate=1e-2)
+        x = torch.randn(4, 16)
+        loss = wrapped(x).pow(2).mean()
+        loss.backward()
+        opt.step()
+        opt.clear_grad()
+        opt.save_sharded({ckpt!r})
+    """, world_size=2)
+
+    # restore single-process: world 1 shard == the full flat buffer
+    import torch
+    import paddle_amd as paddle
+    from paddle_amd.distributed.fleet.sharding import (GroupShardedStage3,
+                                                       ShardedAdamW)
+    torch.manual_seed(0)
+    m = paddle.nn.Sequential(paddle.nn.Linear(16, 32), paddle.nn.Linear(32, 16))
+    wrapped = GroupShardedStage3(m, device=torch.device("cpu"))
+    opt = ShardedAdamW(wrapped, learning_rate=1e-2)
+    before = [u.master.clone() for u in wrapped._units]
+    opt.load_sharded(ckpt)
+    assert opt.step_count == 1
+    changed = any(not torch.equal(b, u.master)
+                  for b, u in zip(before, wrapped._units))
+    assert changed, "restore did not modify masters"
+    # the restored single-rank state must equal re-running the same step
+    # locally (identical init + same grads => identical AdamW result)
+    torch.manual_seed(0)
+    m2 = paddle.nn.Sequential(paddle.nn.Linear(16, 32), paddle.nn.Linear(32, 16))
+    w2 = GroupShardedStage3(m2, device=torch.device("cpu"))
+    o2 = ShardedAdamW(w2, learning_rate=1e-2)
+    x = torch.randn(4, 16)   # NOTE: ranks saw different x; see below
+    # grads differ across ranks in the saved run (different torch.randn
+    # per-rank after seed reset? both ranks seeded 0 -> same x), so the
+    # single-process rerun with the same seed reproduces the same step.
+    loss = w2(x).pow(2).mean()
+    loss.backward()
+    o2.step()
+    for u_loaded, u_ref in zip(wrapped._units, w2._units):
+        torch.testing.assert_close(u_loaded.master, u_ref.master,
+                                   atol=1e-5, rtol=1e-5)
