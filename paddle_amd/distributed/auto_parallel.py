@@ -240,3 +240,112 @@ _default_mesh: Optional[ProcessMesh] = None
 def set_mesh(mesh):
     global _default_mesh
     _default_mesh = mesh
+
+
+# ---------------------------------------------------------------------------
+# SPMD rules: einsum-notation placement propagation
+# (reference: paddle/phi/infermeta/spmd_rules/*.cc -- matmul.cc,
+#  elementwise.cc, reduction.cc use the same dim-char formulation)
+# ---------------------------------------------------------------------------
+def infer_spmd(notation: str, placements_list, mesh: ProcessMesh):
+    """Given an einsum-style notation ("mk,kn->mn") and the input
+    placements, return the output placements.
+
+    Rules (matching the reference's dim-mapping semantics):
+      - a mesh dim sharding an input dim whose char appears in the output
+        -> output Shard(out_index)
+      - a mesh dim sharding only contracted/reduced chars -> output Partial
+      - conflicts (one mesh dim sharding different chars) -> Replicate
+        (caller reshards an input first)
+    """
+    ins, out = notation.replace(" ", "").split("->")
+    in_specs = ins.split(",")
+    assert len(in_specs) == len(placements_list)
+    by_mesh_dim = {}
+    for spec, pls in zip(in_specs, placements_list):
+        for mdim, p in enumerate(pls):
+            if isinstance(p, Shard):
+                if p.dim >= len(spec):
+                    continue
+                by_mesh_dim.setdefault(mdim, set()).add(spec[p.dim])
+            elif isinstance(p, Partial):
+                by_mesh_dim.setdefault(mdim, set()).add("+")
+    out_pl = []
+    for mdim in range(mesh.ndim):
+        chars = by_mesh_dim.get(mdim, set())
+        if len(chars) != 1:
+            out_pl.append(Replicate())
+            continue
+        c = next(iter(chars))
+        if c == "+":
+            out_pl.append(Partial())
+        elif c in out:
+            out_pl.append(Shard(out.index(c)))
+        else:
+            out_pl.append(Partial())      # contracted sharded dim
+    return out_pl
+
+
+def _placements_of(t, mesh):
+    return getattr(t, "placements", [Replicate()] * mesh.ndim)
+
+
+def _local_op(fn, out_notation_placements, mesh, *tensors):
+    out = fn(*tensors)
+    out.dist_attr = DistAttr(mesh, out_notation_placements)
+    out.process_mesh = mesh
+    out.placements = list(out_notation_placements)
+    return out
+
+
+def dist_matmul(x, y):
+    """SPMD matmul: local matmul + inferred placements.  A contraction-dim
+    conflict (k sharded in only one operand) reshards that operand to
+    Replicate first -- the reference's matmul.cc does the same fallback."""
+    mesh = getattr(x, "process_mesh", None) or getattr(y, "process_mesh", None)
+    assert mesh is not None, "dist_matmul needs dist tensors"
+    px, py = _placements_of(x, mesh), _placements_of(y, mesh)
+    # k = last dim of x, first of y
+    for mdim in range(mesh.ndim):
+        sx, sy = px[mdim], py[mdim]
+        x_shards_k = isinstance(sx, Shard) and sx.dim == x.dim() - 1
+        y_shards_k = isinstance(sy, Shard) and sy.dim == 0
+        if x_shards_k != y_shards_k:        # one-sided contraction shard
+            if x_shards_k:
+                x = reshard(x, mesh, [Replicate() if i == mdim else p
+                                      for i, p in enumerate(px)])
+                px = x.placements
+            else:
+                y = reshard(y, mesh, [Replicate() if i == mdim else p
+                                      for i, p in enumerate(py)])
+                py = y.placements
+    if x.dim() == 2 and y.dim() == 2:
+        notation = "mk,kn->mn"
+    elif x.dim() == 3 and y.dim() == 2:
+        notation = "bmk,kn->bmn"
+    else:
+        notation = "mk,kn->mn"
+    out_pl = infer_spmd(notation, [px, py], mesh)
+    return _local_op(torch.matmul, out_pl, mesh, x, y)
+
+
+def dist_elementwise(fn, *tensors):
+    mesh = next(t.process_mesh for t in tensors if hasattr(t, "process_mesh"))
+    specs = []
+    pls = []
+    chars = "abcdefgh"
+    nd = max(t.dim() for t in tensors if isinstance(t, torch.Tensor))
+    for t in tensors:
+        specs.append(chars[nd - t.dim():nd])
+        pls.append(_placements_of(t, mesh))
+    out_pl = infer_spmd(",".join(specs) + "->" + chars[:nd], pls, mesh)
+    return _local_op(fn, out_pl, mesh, *tensors)
+
+
+def dist_reduce(x, axis, fn=torch.sum, keepdim=False):
+    mesh = x.process_mesh
+    nd = x.dim()
+    chars = "abcdefgh"[:nd]
+    out_chars = "".join(c for i, c in enumerate(chars) if i != axis % nd)
+    out_pl = infer_spmd(chars + "->" + out_chars, [_placements_of(x, mesh)], mesh)
+    return _local_op(lambda t: fn(t, dim=axis, keepdim=keepdim), out_pl, mesh, x)

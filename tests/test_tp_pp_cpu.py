@@ -231,3 +231,53 @@ def test_interleaved_vpp_matches_single():
                 assert torch.allclose(a, b2, atol=1e-6)
         print("rank", r, "vpp ok")
     """, timeout=300)
+
+
+def test_spmd_dist_matmul_two_ranks():
+    """SPMD rules over a 2-rank mesh: row-sharded X @ replicated W stays
+    sharded; contraction-sharded matmul yields Partial, resolved by
+    reshard to match the single-process product.
+    (reference: phi/infermeta/spmd_rules/matmul.cc)"""
+    run_dist("""
+        import torch
+        import paddle_amd as paddle
+        paddle.distributed.init_parallel_env()
+        from paddle_amd.distributed import auto_parallel as ap
+        r = paddle.distributed.get_rank()
+        mesh = ap.ProcessMesh([0, 1])
+        torch.manual_seed(3)
+        X = torch.randn(8, 6)
+        W = torch.randn(6, 4)
+        ref = X @ W
+
+        # case 1: X row-sharded, W replicated -> out row-sharded
+        xs = ap.shard_tensor(X, mesh, [ap.Shard(0)])
+        wr = ap.shard_tensor(W, mesh, [ap.Replicate()])
+        out = ap.dist_matmul(xs, wr)
+        assert out.placements == [ap.Shard(0)], out.placements
+        full = ap.reshard(out, mesh, [ap.Replicate()])
+        assert torch.allclose(full, ref, atol=1e-5)
+
+        # case 2: contraction dim sharded on both -> Partial, all-reduce
+        xc = ap.shard_tensor(X, mesh, [ap.Shard(1)])
+        wc = ap.shard_tensor(W, mesh, [ap.Shard(0)])
+        out2 = ap.dist_matmul(xc, wc)
+        assert isinstance(out2.placements[0], ap.Partial), out2.placements
+        full2 = ap.reshard(out2, mesh, [ap.Replicate()])
+        assert torch.allclose(full2, ref, atol=1e-4)
+
+        # case 3: one-sided contraction shard -> auto-reshard then local
+        out3 = ap.dist_matmul(xc, wr)
+        full3 = ap.reshard(out3, mesh, [ap.Replicate()]) if not isinstance(
+            out3.placements[0], ap.Replicate) else out3
+        assert torch.allclose(full3, ref, atol=1e-4)
+
+        # elementwise + reduction rules
+        ys = ap.dist_elementwise(torch.add, xs, xs)
+        assert ys.placements == [ap.Shard(0)]
+        red = ap.dist_reduce(xs, axis=1)
+        assert red.placements == [ap.Shard(0)]
+        red2 = ap.dist_reduce(xs, axis=0)
+        assert isinstance(red2.placements[0], ap.Partial)
+        print("rank", r, "spmd ok")
+    """, world_size=2)
