@@ -205,6 +205,22 @@ Tensor colsum(const Tensor& x) {
   return out;
 }
 
+Tensor weight_only_gemv(const Tensor& x, const Tensor& qweight,
+                        const Tensor& scale, const c10::optional<Tensor>& bias) {
+  CHECK_IN(x); CHECK_IN(qweight); CHECK_IN(scale);
+  TORCH_CHECK(qweight.scalar_type() == torch::kChar, "qweight must be int8");
+  int64_t k = x.size(-1), m = x.numel() / k, n = qweight.size(0);
+  TORCH_CHECK(qweight.size(1) == k && k % 8 == 0);
+  auto sizes = x.sizes().vec();
+  sizes.back() = n;
+  auto out = torch::empty(sizes, x.options());
+  const void* bp = bias.has_value() ? bias->const_data_ptr() : nullptr;
+  pa::weight_only_gemv(x.const_data_ptr(), qweight.const_data_ptr(),
+                       scale.const_data_ptr<float>(), bp,
+                       out.mutable_data_ptr(), m, n, k, dt_of(x), cur_stream());
+  return out;
+}
+
 // ---- adamw ----------------------------------------------------------------
 void adamw(Tensor& master, c10::optional<Tensor> param_out, const Tensor& grad,
            Tensor& m, Tensor& v, double lr, double beta1, double beta2,
@@ -424,6 +440,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fc1_gelu_fwd", &pa_lt::fc1_gelu_fwd);
   m.def("fc2_dgrad_dgelu", &pa_lt::fc2_dgrad_dgelu);
   m.def("lt_epilogue_probe", &pa_lt::lt_epilogue_probe);
+  m.def("weight_only_gemv", &weight_only_gemv);
   m.def("l2norm_sq", &l2norm_sq);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);

@@ -142,4 +142,9 @@ void gemm_bf16(const void* a, const void* b, void* c, int64_t m, int64_t n,
                int64_t k, int64_t lda, int64_t ldb, int64_t ldc, bool b_is_nt,
                hipStream_t s);
 
+// weight-only int8 GEMV (decode): qweight [N,K] int8 rows, scale [N] fp32
+void weight_only_gemv(const void* x, const void* wq, const float* scale,
+                      const void* bias, void* out, int64_t m, int64_t n,
+                      int64_t k, int dtype, hipStream_t s);
+
 }  // namespace pa

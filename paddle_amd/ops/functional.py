@@ -751,3 +751,27 @@ def _fused_ffn_available(x):
             _lt_ffn_ok = (C.lt_epilogue_probe(1024, 1024, 1024, 164) > 0 and
                           C.lt_epilogue_probe(1024, 1024, 1024, 208) > 0)
     return _lt_ffn_ok
+
+
+def fused_linear_param_grad_add(x, dy, dweight=None, dbias=None,
+                                multi_precision=False, has_bias=True):
+    """dW += X^T dY (and db += colsum(dY)) in single accumulating GEMM /
+    reduction launches -- no intermediate dW allocation.
+    Parity: paddle/phi/kernels/fusion/gpu/fused_linear_param_grad_add_kernel.cu
+    (used by the sequence-parallel backward overlap)."""
+    x2 = x.reshape(-1, x.shape[-1])
+    dy2 = dy.reshape(-1, dy.shape[-1])
+    if dweight is None:
+        dweight = torch.zeros(x2.shape[1], dy2.shape[1], dtype=x.dtype,
+                              device=x.device)
+    dweight.addmm_(x2.t(), dy2)   # beta=1 accumulate inside the GEMM epilogue
+    if has_bias:
+        if dbias is None:
+            dbias = torch.zeros(dy2.shape[1], dtype=dy.dtype, device=dy.device)
+        if dy.is_cuda and _ext.use_native(dy):
+            C = _ext.get_ext()
+            dbias.add_(C.colsum(dy2.contiguous()).to(dbias.dtype))
+        else:
+            dbias.add_(dy2.sum(0).to(dbias.dtype))
+        return dweight, dbias
+    return dweight, None

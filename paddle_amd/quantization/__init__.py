@@ -57,3 +57,32 @@ class PTQ:
 
     def convert(self, model, inplace=False):
         return model
+
+
+# -- weight-only int8 (serving decode path) ----------------------------------
+# parity: paddle/phi/kernels/weight_quantize_kernel.cu + funcs/weight_only_gemv.cu
+def weight_quantize(w, algo="weight_only_int8"):
+    """w [K, N] (paddle Linear layout) -> (qweight [N, K] int8 row-major,
+    scale [N] fp32).  Per-output-channel absmax; w ~ qw.T * scale / 127."""
+    import torch
+    assert algo in ("weight_only_int8",)
+    wt = w.detach().float().t().contiguous()        # [N, K]
+    scale = wt.abs().amax(dim=1).clamp(min=1e-8)    # [N]
+    q = torch.round(wt / scale.unsqueeze(1) * 127.0).clamp(-127, 127).to(torch.int8)
+    return q, scale
+
+
+def weight_only_linear(x, qweight, scale, bias=None, weight_dtype="int8"):
+    """x [..., K] @ dequant(qweight) + bias.  GPU decode path runs the
+    wave-per-channel int8 GEMV HIP kernel; fallback dequantizes."""
+    import torch
+    from .. import _ext
+    if x.is_cuda and _ext.use_native(x):
+        C = _ext.get_ext()
+        b = bias.to(x.dtype) if bias is not None else None
+        return C.weight_only_gemv(x.contiguous(), qweight, scale, b)
+    w = (qweight.float() * scale.unsqueeze(1) / 127.0).t().to(x.dtype)  # [K,N]
+    out = x @ w
+    if bias is not None:
+        out = out + bias
+    return out

@@ -41,3 +41,15 @@ def test_generate_matches_full_recompute_cpu():
             cur = torch.cat([cur, nxt], dim=1)
     ref = torch.cat(ref_tokens, dim=1)
     assert torch.equal(gen, ref), (gen, ref)
+
+
+def test_weight_only_quant_roundtrip_cpu():
+    import torch
+    from paddle_amd import quantization as Q
+    torch.manual_seed(5)
+    w = torch.randn(128, 64)
+    qw, sc = Q.weight_quantize(w)
+    x = torch.randn(2, 128)
+    out = Q.weight_only_linear(x, qw, sc)
+    rel = (out - x @ w).abs().max() / (x @ w).abs().max()
+    assert rel < 0.02, float(rel)

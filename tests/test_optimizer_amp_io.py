@@ -153,3 +153,17 @@ def test_async_save(tmp_path):
     paddle.framework_io.clear_async_save_task_queue()
     assert os.path.exists(p)
     assert paddle.load(p)["x"].sum() == 3
+
+
+def test_fused_linear_param_grad_add_accumulates():
+    import torch
+    from paddle_amd.ops import functional as hot
+    torch.manual_seed(0)
+    x = torch.randn(6, 4)
+    dy = torch.randn(6, 3)
+    dw = torch.ones(4, 3)
+    db = torch.ones(3)
+    dw2, db2 = hot.fused_linear_param_grad_add(x, dy, dw, db)
+    assert dw2 is dw and db2 is db
+    torch.testing.assert_close(dw, 1.0 + x.t() @ dy)
+    torch.testing.assert_close(db, 1.0 + dy.sum(0))
