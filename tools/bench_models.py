@@ -92,6 +92,49 @@ def main():
         return crit(m(ids), labels)
 
     run("gpt-moe-64e 350M-trunk b8 s1024", moe, moe_batch, moe_loss)
+    del moe
+    torch.cuda.empty_cache()
+    torch.cuda.reset_peak_memory_stats()
+    bench_llama7b()
+
+
+def bench_llama7b():
+    """Llama-2-7B sharding-3 on one GPU (b4 s2048 bf16) -- the dense-model
+    secondary config measured standalone (70B needs the 8-GPU hybrid run)."""
+    import paddle_amd as paddle
+    from paddle_amd.models import build_llama
+    from paddle_amd.distributed.fleet.sharding import (GroupShardedStage3,
+                                                       ShardedAdamW)
+    dev = torch.device("cuda")
+    model = build_llama("llama2-7b", max_seq_len=2048).to(dev, torch.bfloat16)
+    wrapped = GroupShardedStage3(model, device=dev)
+    opt = ShardedAdamW(wrapped, learning_rate=1e-4,
+                       grad_clip=paddle.nn.ClipGradByGlobalNorm(1.0))
+    B, S, V = 4, 2048, 32000
+    ids = torch.randint(0, V, (B, S), device=dev)
+    labels = torch.randint(0, V, (B, S), device=dev)
+
+    import torch.nn.functional as F
+
+    def step():
+        logits = wrapped(ids)
+        loss = F.cross_entropy(logits.float().reshape(-1, V), labels.reshape(-1))
+        loss.backward()
+        opt.step()
+        opt.clear_grad()
+        return loss
+
+    for _ in range(3):
+        step()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(6):
+        loss = step()
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / 6
+    print(f"{'llama2-7b shard3 b4 s2048':34s} {dt * 1e3:8.2f} ms/step  "
+          f"{B * S / dt:12.0f} tokens/s  loss={float(loss):.3f}  "
+          f"peak={torch.cuda.max_memory_allocated() / 2**30:.1f} GB")
 
 
 if __name__ == "__main__":
