@@ -118,7 +118,10 @@ def bench_llama7b():
 
     def step():
         logits = wrapped(ids)
-        loss = F.cross_entropy(logits.float().reshape(-1, V), labels.reshape(-1))
+        # causal shift: predict position t+1 (unshifted CE is trivially
+        # memorizable from the visible input token)
+        loss = F.cross_entropy(logits[:, :-1].float().reshape(-1, V),
+                               labels[:, 1:].reshape(-1))
         loss.backward()
         opt.step()
         opt.clear_grad()
