@@ -271,6 +271,78 @@ from . import text  # noqa: F401
 from . import audio  # noqa: F401
 from . import onnx  # noqa: F401
 from .framework_io import save, load, async_save  # noqa: F401
+
+# remaining top-level parity names
+import torch as _torch  # noqa: E402
+dtype = _torch.dtype
+bool = _torch.bool  # noqa: A001 -- paddle exports `bool` as a dtype
+
+
+def addmm_(input, x, y, beta=1.0, alpha=1.0, name=None):
+    input.addmm_(x, y, beta=beta, alpha=alpha)
+    return input
+
+
+def floor_mod(x, y, name=None):
+    return _torch.remainder(x, y)
+
+
+def floor_mod_(x, y, name=None):
+    x.remainder_(y)
+    return x
+
+
+def batch(reader, batch_size, drop_last=False):
+    def _gen():
+        buf = []
+        for item in reader():
+            buf.append(item)
+            if len(buf) == batch_size:
+                yield buf
+                buf = []
+        if buf and not drop_last:
+            yield buf
+    return _gen
+
+
+def summary(net, input_size=None, dtypes=None, input=None):
+    from .hapi_summary import summary as _s
+    return _s(net, input_size, dtypes, input)
+
+
+def flops(net, input_size, custom_ops=None, print_detail=False):
+    from .hapi_summary import flops as _f
+    return _f(net, input_size, custom_ops, print_detail)
+
+
+def create_parameter(shape, dtype="float32", name=None, attr=None,
+                     is_bias=False, default_initializer=None):
+    from . import framework as _fw
+    t = _torch.empty(shape, dtype=_fw.convert_dtype(dtype))
+    if default_initializer is not None:
+        default_initializer(t)
+    elif is_bias or t.dim() < 2:
+        _torch.nn.init.zeros_(t)
+    else:
+        _torch.nn.init.xavier_normal_(t)
+    t.requires_grad_(True)
+    return t
+
+
+class CUDAPinnedPlace:
+    def __repr__(self):
+        return "Place(gpu_pinned)"
+
+
+DataParallel = distributed.DataParallel
+
+# long-tail export parity: everything public in tensor/extras.py
+from .tensor import extras as _extras  # noqa: E402
+for _n in dir(_extras):
+    if not _n.startswith("_") and _n not in globals():
+        globals()[_n] = getattr(_extras, _n)
+del _extras, _n
+
 from .hapi import Model  # noqa: F401
 from .param_attr import ParamAttr  # noqa: F401
 from .autograd import grad, no_grad, enable_grad, set_grad_enabled, is_grad_enabled  # noqa: F401

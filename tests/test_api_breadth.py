@@ -74,3 +74,20 @@ def test_shard_tensor_two_ranks():
         assert torch.allclose(red, 2 * full)
         print("rank", r, "dist tensor ok")
     """)
+
+
+def test_top_level_export_parity_complete():
+    """Every name in the reference's paddle.__all__ resolves here.
+    (reference: python/paddle/__init__.py __all__, 418 names)"""
+    import os
+    import re
+    ref = "/root/reference/python/paddle/__init__.py"
+    if not os.path.exists(ref):
+        import pytest
+        pytest.skip("reference tree not mounted")
+    src = open(ref).read()
+    names = re.findall(r"'([^']+)'",
+                       re.search(r"__all__ = \[(.*?)\]", src, re.S).group(1))
+    import paddle_amd
+    missing = [n for n in names if not hasattr(paddle_amd, n)]
+    assert not missing, f"missing {len(missing)}: {missing[:20]}"
