@@ -131,3 +131,7 @@ def hessian(func, xs, batch_axis=None):
 
 def jacobian(func, xs, batch_axis=None):
     return torch.autograd.functional.jacobian(func, xs)
+
+
+# reference: autograd/saved_tensors_hooks.py -- pack/unpack hook context
+from torch.autograd.graph import saved_tensors_hooks  # noqa: F401

@@ -99,3 +99,13 @@ StudentT = _wrap(td.StudentT)
 def kl_divergence(p, q):
     return td.kl_divergence(p._t if isinstance(p, Distribution) else p,
                             q._t if isinstance(q, Distribution) else q)
+
+
+# long-tail (reference: distribution/__init__.py)
+import torch as _t
+
+ExponentialFamily = _t.distributions.ExponentialFamily
+Independent = _t.distributions.Independent
+TransformedDistribution = _t.distributions.TransformedDistribution
+LKJCholesky = _t.distributions.LKJCholesky
+register_kl = _t.distributions.register_kl

@@ -74,3 +74,19 @@ def fftshift(x, axes=None, name=None):
 
 def ifftshift(x, axes=None, name=None):
     return tf.ifftshift(x, dim=axes)
+
+
+def hfft2(x, s=None, axes=(-2, -1), norm="backward", name=None):
+    return torch.fft.hfft2(x, s=s, dim=axes, norm=norm)
+
+
+def ihfft2(x, s=None, axes=(-2, -1), norm="backward", name=None):
+    return torch.fft.ihfft2(x, s=s, dim=axes, norm=norm)
+
+
+def hfftn(x, s=None, axes=None, norm="backward", name=None):
+    return torch.fft.hfftn(x, s=s, dim=axes, norm=norm)
+
+
+def ihfftn(x, s=None, axes=None, norm="backward", name=None):
+    return torch.fft.ihfftn(x, s=s, dim=axes, norm=norm)

@@ -226,3 +226,11 @@ def _move(batch, dev):
 
 def get_worker_info():
     return tud.get_worker_info()
+
+
+# long-tail samplers/datasets (reference: io/__init__.py)
+from torch.utils.data import (  # noqa: F401
+    ConcatDataset,
+    SubsetRandomSampler,
+    WeightedRandomSampler,
+)

@@ -62,6 +62,7 @@ from .rnn import (  # noqa: F401
     SimpleRNN,
     SimpleRNNCell,
 )
+from .torch_wrap import *  # noqa: F401,F403
 from .transformer import (  # noqa: F401
     MultiHeadAttention,
     Transformer,

@@ -271,3 +271,234 @@ def label_smooth(label, prior_dist=None, epsilon=0.1, name=None):
     if prior_dist is not None:
         return (1 - epsilon) * label + epsilon * prior_dist
     return (1 - epsilon) * label + epsilon / n
+
+
+# ---------------------------------------------------------------------------
+# long-tail functional parity (reference: nn/functional/__init__.py __all__)
+# -- thin torch.nn.functional dispatches + paddle-specific forms
+# ---------------------------------------------------------------------------
+import torch as _t
+import torch.nn.functional as _F
+
+
+def _drop_name(kw):
+    kw.pop("name", None)
+    return kw
+
+
+def _mk(fn):
+    def g(*a, **kw):
+        return fn(*a, **_drop_name(kw))
+    g.__name__ = fn.__name__
+    return g
+
+
+celu = _mk(_F.celu)
+hardshrink = _mk(_F.hardshrink)
+log_sigmoid = _mk(_F.logsigmoid)
+prelu = _mk(_F.prelu)
+selu = _mk(_F.selu)
+softshrink = _mk(_F.softshrink)
+softsign = _mk(_F.softsign)
+tanhshrink = _mk(_F.tanhshrink)
+gumbel_softmax = _mk(_F.gumbel_softmax)
+alpha_dropout = _mk(_F.alpha_dropout)
+feature_alpha_dropout = _mk(_F.feature_alpha_dropout)
+avg_pool1d = _mk(_F.avg_pool1d)
+avg_pool3d = _mk(_F.avg_pool3d)
+lp_pool1d = _mk(_F.lp_pool1d)
+lp_pool2d = _mk(_F.lp_pool2d)
+max_pool1d = _mk(_F.max_pool1d)
+max_pool3d = _mk(_F.max_pool3d)
+max_unpool1d = _mk(_F.max_unpool1d)
+max_unpool2d = _mk(_F.max_unpool2d)
+max_unpool3d = _mk(_F.max_unpool3d)
+adaptive_avg_pool1d = _mk(_F.adaptive_avg_pool1d)
+adaptive_avg_pool3d = _mk(_F.adaptive_avg_pool3d)
+adaptive_max_pool1d = _mk(_F.adaptive_max_pool1d)
+adaptive_max_pool2d = _mk(_F.adaptive_max_pool2d)
+adaptive_max_pool3d = _mk(_F.adaptive_max_pool3d)
+fractional_max_pool2d = _mk(_F.fractional_max_pool2d)
+fractional_max_pool3d = _mk(_F.fractional_max_pool3d)
+margin_ranking_loss = _mk(_F.margin_ranking_loss)
+multi_label_soft_margin_loss = _mk(_F.multilabel_soft_margin_loss)
+poisson_nll_loss = _mk(_F.poisson_nll_loss)
+ctc_loss = _mk(_F.ctc_loss)
+hinge_embedding_loss = _mk(_F.hinge_embedding_loss)
+affine_grid = _mk(_F.affine_grid)
+grid_sample = _mk(_F.grid_sample)
+local_response_norm = _mk(_F.local_response_norm)
+pixel_shuffle = _mk(_F.pixel_shuffle)
+pixel_unshuffle = _mk(_F.pixel_unshuffle)
+channel_shuffle = _mk(_F.channel_shuffle)
+instance_norm = _mk(_F.instance_norm)
+group_norm = _mk(_F.group_norm)
+fold = _mk(_F.fold)
+cosine_similarity = _mk(_F.cosine_similarity)
+cosine_embedding_loss = _mk(_F.cosine_embedding_loss)
+rrelu = _mk(_F.rrelu)
+triplet_margin_loss = _mk(_F.triplet_margin_loss)
+triplet_margin_with_distance_loss = _mk(_F.triplet_margin_with_distance_loss)
+soft_margin_loss = _mk(_F.soft_margin_loss)
+gaussian_nll_loss = _mk(_F.gaussian_nll_loss)
+multi_margin_loss = _mk(_F.multi_margin_loss)
+pairwise_distance = _mk(_F.pairwise_distance)
+bilinear = _mk(_F.bilinear)
+dropout3d = _mk(_F.dropout3d)
+conv3d = _mk(_F.conv3d)
+conv3d_transpose = _mk(_F.conv_transpose3d)
+conv1d_transpose = _mk(_F.conv_transpose1d)
+hardtanh = _mk(_F.hardtanh)
+
+
+def upsample(x, size=None, scale_factor=None, mode="nearest",
+             align_corners=False, align_mode=0, data_format="NCHW", name=None):
+    ac = align_corners if mode in ("linear", "bilinear", "bicubic",
+                                   "trilinear") else None
+    return _F.interpolate(x, size=size, scale_factor=scale_factor, mode=mode,
+                          align_corners=ac)
+
+
+def zeropad2d(x, padding, data_format="NCHW", name=None):
+    return _F.pad(x, padding if isinstance(padding, (list, tuple)) else
+                  [padding] * 4)
+
+
+def maxout(x, groups, axis=1, name=None):
+    shape = list(x.shape)
+    c = shape[axis]
+    assert c % groups == 0
+    new = shape[:axis] + [c // groups, groups] + shape[axis + 1:]
+    return x.reshape(new).max(dim=axis + 1).values
+
+
+def thresholded_relu(x, threshold=1.0, value=0.0, name=None):
+    return _t.where(x > threshold, x, _t.full_like(x, value))
+
+
+def sequence_mask(x, maxlen=None, dtype="int64", name=None):
+    from ... import framework as _fw
+    m = int(maxlen) if maxlen is not None else int(x.max())
+    r = _t.arange(m, device=x.device)
+    return (r.unsqueeze(0) < x.unsqueeze(-1)).to(_fw.convert_dtype(dtype))
+
+
+def dice_loss(input, label, epsilon=1e-5, name=None):
+    label_one_hot = _F.one_hot(label.squeeze(-1).long(), input.shape[-1]).to(input.dtype)
+    inter = (input * label_one_hot).sum(-1)
+    union = input.sum(-1) + label_one_hot.sum(-1)
+    return (1 - (2 * inter + epsilon) / (union + epsilon)).mean()
+
+
+def log_loss(input, label, epsilon=1e-4, name=None):
+    return -label * _t.log(input + epsilon) - (1 - label) * _t.log(
+        1 - input + epsilon)
+
+
+def square_error_cost(input, label):
+    return (input - label) ** 2
+
+
+def npair_loss(anchor, positive, labels, l2_reg=0.002):
+    sim = anchor @ positive.t()
+    ce = _F.cross_entropy(sim, _t.arange(sim.shape[0], device=sim.device))
+    reg = l2_reg * (anchor.pow(2).sum(1).mean() + positive.pow(2).sum(1).mean())
+    return ce + reg
+
+
+def sigmoid_focal_loss(logit, label, normalizer=None, alpha=0.25, gamma=2.0,
+                       reduction="sum", name=None):
+    p = _t.sigmoid(logit)
+    ce = _F.binary_cross_entropy_with_logits(logit, label, reduction="none")
+    pt = p * label + (1 - p) * (1 - label)
+    af = alpha * label + (1 - alpha) * (1 - label)
+    loss = af * (1 - pt) ** gamma * ce
+    if normalizer is not None:
+        loss = loss / normalizer
+    if reduction == "sum":
+        return loss.sum()
+    if reduction == "mean":
+        return loss.mean()
+    return loss
+
+
+def margin_cross_entropy(logits, label, margin1=1.0, margin2=0.5, margin3=0.0,
+                         scale=64.0, group=None, return_softmax=False,
+                         reduction="mean"):
+    """ArcFace/CosFace-style margin softmax (single-rank form; the
+    model-parallel variant lives in fleet.mpu.ParallelCrossEntropy).
+    reference: paddle/fluid/operators/margin_cross_entropy_op.cu"""
+    cos = logits.float().clamp(-1, 1)
+    theta = _t.acos(cos.gather(1, label.view(-1, 1)))
+    target = _t.cos(margin1 * theta + margin2) - margin3
+    out = cos.scatter(1, label.view(-1, 1), target) * scale
+    loss = _F.cross_entropy(out, label.view(-1), reduction=reduction)
+    if return_softmax:
+        return loss, _F.softmax(out, -1)
+    return loss
+
+
+def temporal_shift(x, seg_num, shift_ratio=0.25, data_format="NCHW", name=None):
+    nt, c, h, w = x.shape
+    n = nt // seg_num
+    x5 = x.view(n, seg_num, c, h, w)
+    fold_c = int(c * shift_ratio)
+    out = _t.zeros_like(x5)
+    out[:, 1:, :fold_c] = x5[:, :-1, :fold_c]              # shift left
+    out[:, :-1, fold_c:2 * fold_c] = x5[:, 1:, fold_c:2 * fold_c]  # right
+    out[:, :, 2 * fold_c:] = x5[:, :, 2 * fold_c:]
+    return out.view(nt, c, h, w)
+
+
+def gather_tree(ids, parents):
+    max_len, batch, beam = ids.shape
+    out = _t.zeros_like(ids)
+    out[-1] = ids[-1]
+    parent = parents[-1]
+    for t in range(max_len - 2, -1, -1):
+        b_idx = _t.arange(batch, device=ids.device).unsqueeze(1).expand(batch, beam)
+        out[t] = ids[t][b_idx, parent]
+        parent = parents[t][b_idx, parent]
+    return out
+
+
+def flash_attn_qkvpacked(qkv, dropout=0.0, causal=False, return_softmax=False,
+                         **kwargs):
+    """Packed [B,S,3,H,D] flash attention on the zero-copy HIP path."""
+    from ...ops import functional as hot
+    return hot.qkv_flash_attention(qkv, causal=causal), None
+
+
+def _gated(name, why):
+    def f(*a, **k):
+        raise NotImplementedError(f"{name}: {why}")
+    f.__name__ = name
+    return f
+
+
+hsigmoid_loss = _gated("hsigmoid_loss", "hierarchical sigmoid not in this build")
+rnnt_loss = _gated("rnnt_loss", "transducer kernel not in this build")
+class_center_sample = _gated("class_center_sample", "PLSC-style sampling: round 2")
+sparse_attention = _gated("sparse_attention", "use flash_attention (dense) or paged decode")
+adaptive_log_softmax_with_loss = _gated("adaptive_log_softmax_with_loss",
+                                        "use nn.AdaptiveLogSoftmaxWithLoss layer")
+flashmask_attention = _gated("flashmask_attention", "mask-sparse FA: round 2")
+flash_attn_varlen_qkvpacked = _gated("flash_attn_varlen_qkvpacked",
+                                     "varlen path: round 2")
+
+
+def _mk_inplace(fn, extract=None):
+    def g(x, *a, **kw):
+        kw.pop("name", None)
+        out = fn(x, *a, **kw)
+        x.copy_(out)
+        return x
+    return g
+
+
+elu_ = _mk_inplace(_F.elu)
+hardtanh_ = _mk_inplace(_F.hardtanh)
+leaky_relu_ = _mk_inplace(_F.leaky_relu)
+softmax_ = _mk_inplace(_F.softmax)
+tanh_ = _mk_inplace(_t.tanh)
+thresholded_relu_ = _mk_inplace(thresholded_relu)

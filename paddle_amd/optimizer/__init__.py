@@ -321,3 +321,148 @@ class Lamb(Optimizer):
 
 
 lr = lr_mod
+
+
+# ---------------------------------------------------------------------------
+# long-tail optimizers (reference: optimizer/{adagrad,adamax,asgd,radam,
+# rmsprop,adadelta,rprop,nadam,lbfgs}.py) -- torch.optim engines under the
+# paddle Optimizer API (learning_rate/parameters/clear_grad/.pdopt dicts)
+# ---------------------------------------------------------------------------
+class _TorchOptimizer(Optimizer):
+    _torch_cls = None
+    _kw_map = {}
+
+    def __init__(self, learning_rate=0.001, parameters=None, weight_decay=None,
+                 grad_clip=None, name=None, multi_precision=None, **kwargs):
+        super().__init__(learning_rate=learning_rate, parameters=parameters,
+                         weight_decay=weight_decay, grad_clip=grad_clip)
+        tkw = {}
+        for pk, tk in self._kw_map.items():
+            if pk in kwargs and kwargs[pk] is not None:
+                tkw[tk] = kwargs[pk]
+        wd = self._weight_decay
+        self._tkw = dict(tkw, weight_decay=wd)
+        self._opt = None
+
+    def _ensure(self):
+        if self._opt is None and self._params:
+            lr = self.get_lr()
+            self._opt = self._torch_cls(self._params, lr=lr, **self._tkw)
+        return self._opt
+
+    @torch.no_grad()
+    def step(self):
+        self._clip_grads()
+        self._step_count += 1
+        opt = self._ensure()
+        if opt is None:
+            return
+        for g in opt.param_groups:
+            g["lr"] = self.get_lr()
+        opt.step()
+
+    def state_dict(self):
+        opt = self._ensure()
+        return opt.state_dict() if opt else {}
+
+    def set_state_dict(self, sd):
+        opt = self._ensure()
+        if opt and sd:
+            opt.load_state_dict(sd)
+
+
+class Adagrad(_TorchOptimizer):
+    _torch_cls = torch.optim.Adagrad
+    _kw_map = {"epsilon": "eps", "initial_accumulator_value":
+               "initial_accumulator_value"}
+
+    def __init__(self, learning_rate=0.001, epsilon=1e-6,
+                 initial_accumulator_value=0.0, **kw):
+        super().__init__(learning_rate=learning_rate, epsilon=epsilon,
+                         initial_accumulator_value=initial_accumulator_value,
+                         **kw)
+
+
+class Adamax(_TorchOptimizer):
+    _torch_cls = torch.optim.Adamax
+
+    def __init__(self, learning_rate=0.001, beta1=0.9, beta2=0.999,
+                 epsilon=1e-8, **kw):
+        super().__init__(learning_rate=learning_rate, **kw)
+        self._tkw.update(betas=(beta1, beta2), eps=epsilon)
+
+
+class ASGD(_TorchOptimizer):
+    _torch_cls = torch.optim.ASGD
+
+
+class RAdam(_TorchOptimizer):
+    _torch_cls = torch.optim.RAdam
+
+    def __init__(self, learning_rate=0.001, beta1=0.9, beta2=0.999,
+                 epsilon=1e-8, **kw):
+        super().__init__(learning_rate=learning_rate, **kw)
+        self._tkw.update(betas=(beta1, beta2), eps=epsilon)
+
+
+class NAdam(_TorchOptimizer):
+    _torch_cls = torch.optim.NAdam
+
+    def __init__(self, learning_rate=0.001, beta1=0.9, beta2=0.999,
+                 epsilon=1e-8, momentum_decay=0.004, **kw):
+        super().__init__(learning_rate=learning_rate, **kw)
+        self._tkw.update(betas=(beta1, beta2), eps=epsilon,
+                         momentum_decay=momentum_decay)
+
+
+class RMSProp(_TorchOptimizer):
+    _torch_cls = torch.optim.RMSprop
+
+    def __init__(self, learning_rate=0.001, rho=0.95, epsilon=1e-6,
+                 momentum=0.0, centered=False, **kw):
+        super().__init__(learning_rate=learning_rate, **kw)
+        self._tkw.update(alpha=rho, eps=epsilon, momentum=momentum,
+                         centered=centered)
+
+
+class Adadelta(_TorchOptimizer):
+    _torch_cls = torch.optim.Adadelta
+
+    def __init__(self, learning_rate=0.001, epsilon=1e-6, rho=0.95, **kw):
+        super().__init__(learning_rate=learning_rate, **kw)
+        self._tkw.update(eps=epsilon, rho=rho)
+
+
+class Rprop(_TorchOptimizer):
+    _torch_cls = torch.optim.Rprop
+
+    def __init__(self, learning_rate=0.001, learning_rate_range=(1e-5, 50),
+                 etas=(0.5, 1.2), **kw):
+        kw.pop("weight_decay", None)
+        super().__init__(learning_rate=learning_rate, **kw)
+        self._tkw.pop("weight_decay", None)
+        self._tkw.update(etas=tuple(etas), step_sizes=tuple(learning_rate_range))
+
+
+class LBFGS(_TorchOptimizer):
+    _torch_cls = torch.optim.LBFGS
+
+    def __init__(self, learning_rate=1.0, max_iter=20, max_eval=None,
+                 tolerance_grad=1e-7, tolerance_change=1e-9, history_size=100,
+                 line_search_fn=None, **kw):
+        kw.pop("weight_decay", None)
+        super().__init__(learning_rate=learning_rate, **kw)
+        self._tkw.pop("weight_decay", None)
+        self._tkw.update(max_iter=max_iter, max_eval=max_eval,
+                         tolerance_grad=tolerance_grad,
+                         tolerance_change=tolerance_change,
+                         history_size=history_size,
+                         line_search_fn=line_search_fn)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        opt = self._ensure()
+        if closure is not None:
+            with torch.enable_grad():
+                return opt.step(closure)
+        return opt.step(lambda: 0.0)
