@@ -337,6 +337,57 @@ class CUDAPinnedPlace:
 
 DataParallel = distributed.DataParallel
 
+
+
+# tensor-method long tail at top level
+def put_along_axis_(arr, indices, values, axis, reduce="assign"):
+    import torch as _tt
+    if reduce == "assign":
+        arr.scatter_(axis, indices.long(),
+                     values if isinstance(values, _tt.Tensor) else
+                     _tt.full_like(indices, values, dtype=arr.dtype))
+    else:
+        arr.scatter_reduce_(axis, indices.long(), values,
+                            {"add": "sum", "mul": "prod"}.get(reduce, reduce))
+    return arr
+
+
+def top_p_sampling(x, ps, threshold=None, seed=None, name=None):
+    """Nucleus sampling over the last dim (reference:
+    paddle/phi/kernels/gpu/top_p_sampling_kernel.cu)."""
+    import torch as _tt
+    probs = _tt.softmax(x.float(), dim=-1)
+    sp, si = probs.sort(dim=-1, descending=True)
+    cum = sp.cumsum(-1)
+    keep = cum - sp < ps.unsqueeze(-1)
+    sp = sp * keep
+    sp = sp / sp.sum(-1, keepdim=True)
+    choice = _tt.multinomial(sp.reshape(-1, sp.shape[-1]), 1)
+    ids = si.reshape(-1, si.shape[-1]).gather(1, choice).reshape(*x.shape[:-1], 1)
+    scores = probs.reshape(-1, probs.shape[-1]).gather(1, choice).reshape(*x.shape[:-1], 1)
+    return scores, ids
+
+
+def scale_(x, scale=1.0, bias=0.0, bias_after_scale=True, act=None, name=None):
+    import torch as _tt
+    with _tt.no_grad():
+        if bias_after_scale:
+            x.mul_(scale).add_(bias)
+        else:
+            x.add_(bias).mul_(scale)
+    return x
+
+
+def create_tensor(dtype, name=None, persistable=False):
+    from . import framework as _fw
+    return _torch.empty(0, dtype=_fw.convert_dtype(dtype))
+
+
+from .linalg import (  # noqa: E402,F401
+    cond, eigvals, eigvalsh, householder_product, lu_unpack, multi_dot,
+    pca_lowrank, pinv, svd_lowrank,
+)
+
 # long-tail export parity: everything public in tensor/extras.py
 from .tensor import extras as _extras  # noqa: E402
 for _n in dir(_extras):
@@ -378,3 +429,8 @@ def grad_(outputs, inputs, grad_outputs=None, retain_graph=None, create_graph=Fa
 def synchronize(device=None):
     if torch.cuda.is_available():
         torch.cuda.synchronize(device)
+
+# bind the tensor-method long tail now that every top-level name exists
+from . import tensor_patch as _tp  # noqa: E402
+_tp._patch_method_long_tail()
+del _tp

@@ -123,3 +123,16 @@ def vector_norm(x, p=2, axis=None, keepdim=False, name=None):
 
 def householder_product(x, tau, name=None):
     return torch.linalg.householder_product(x, tau)
+
+
+def lu_unpack(x, y, unpack_ludata=True, unpack_pivots=True, name=None):
+    return torch.lu_unpack(x, y, unpack_data=unpack_ludata,
+                           unpack_pivots=unpack_pivots)
+
+
+def pca_lowrank(x, q=None, center=True, niter=2, name=None):
+    return torch.pca_lowrank(x, q=q, center=center, niter=niter)
+
+
+def svd_lowrank(x, q=6, niter=2, M=None, name=None):
+    return torch.svd_lowrank(x, q=q, niter=niter, M=M)

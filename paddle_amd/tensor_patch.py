@@ -125,3 +125,39 @@ def apply_patches():
             _names[id(self)] = v
 
         T.name = property(_name_get, _name_set)
+
+
+def _patch_method_long_tail():
+    """Bind the reference's tensor_method_func long tail as torch.Tensor
+    methods (reference: python/paddle/tensor/__init__.py) -- each method
+    is the same paddle function with the tensor as first argument."""
+    import torch
+    from . import __init__ as _root  # noqa: F401  (late-bound lookups below)
+    import paddle_amd as P
+
+    names = [
+        "add_n", "as_complex", "as_real", "atleast_1d", "atleast_2d",
+        "atleast_3d", "block_diag", "broadcast_shape", "broadcast_tensors",
+        "bucketize", "cast_", "cdist", "concat", "cond", "create_parameter",
+        "create_tensor", "cumulative_trapezoid", "eigvals", "eigvalsh",
+        "equal_", "equal_all", "flatten_", "floor_mod", "floor_mod_",
+        "gammainc", "gammainc_", "gammaincc", "gammaincc_", "gammaln",
+        "gammaln_", "gather_nd", "greater_than", "greater_than_",
+        "histogram_bin_edges", "histogramdd", "householder_product", "i0e",
+        "i1", "i1e", "increment", "index_sample", "is_empty", "is_integer",
+        "is_tensor", "isin", "less_than", "less_than_", "lu_unpack", "mod",
+        "mod_", "multi_dot", "multigammaln", "multigammaln_", "multiplex",
+        "pca_lowrank", "pinv", "polar", "put_along_axis", "put_along_axis_",
+        "rank", "reduce_as", "reshape_", "reverse", "scale_", "scatter_nd",
+        "scatter_nd_add", "shard_index", "slice", "stack", "stanh",
+        "strided_slice", "svd_lowrank", "take_along_axis", "tensordot",
+        "top_p_sampling", "trapezoid", "unstack", "vander", "where_",
+    ]
+    T = torch.Tensor
+    for n in names:
+        fn = getattr(P, n, None)
+        if fn is None or hasattr(T, n):
+            continue
+        setattr(T, n, fn)
+
+# called from paddle_amd/__init__.py AFTER all top-level names exist
