@@ -178,3 +178,65 @@ class CUDAGraph:
 
     def reset(self):
         self._g.reset()
+
+
+# breadth parity (reference: device/__init__.py __all__)
+def get_cudnn_version():
+    import torch
+    v = torch.backends.cudnn.version()  # MIOpen version on ROCm
+    return v
+
+
+class XPUPlace:
+    def __init__(self, *a):
+        raise NotImplementedError("XPU is not a target of this MI355X build")
+
+
+class IPUPlace:
+    def __init__(self, *a):
+        raise NotImplementedError("IPU is not a target of this MI355X build")
+
+
+def is_compiled_with_xpu():
+    return False
+
+
+def is_compiled_with_ipu():
+    return False
+
+
+def is_compiled_with_cinn():
+    return False  # by design: hand-fused HIP kernels, no IR compiler
+
+
+def is_compiled_with_distribute():
+    return True
+
+
+def is_compiled_with_custom_device(device_type=None):
+    return False
+
+
+def get_all_device_type():
+    import torch
+    return ["cpu", "gpu"] if torch.cuda.is_available() else ["cpu"]
+
+
+def get_all_custom_device_type():
+    return []
+
+
+def get_available_device():
+    import torch
+    return ([f"gpu:{i}" for i in range(torch.cuda.device_count())]
+            if torch.cuda.is_available() else ["cpu"])
+
+
+def get_available_custom_device():
+    return []
+
+
+def set_stream(stream):
+    import torch
+    torch.cuda.set_stream(stream._raw if hasattr(stream, "_raw") else stream)
+    return stream

@@ -90,3 +90,12 @@ def load(path, **configs):
 
 def enable_to_static(flag=True):
     pass
+
+
+def set_code_level(level=100, also_to_stdout=False):
+    """dy2static debug verbosity (reference: jit/dy2static/logging_utils.py);
+    the deferred-graph build has no transformed code to print."""
+
+
+def set_verbosity(level=0, also_to_stdout=False):
+    pass

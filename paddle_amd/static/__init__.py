@@ -302,3 +302,250 @@ def gradients(targets, inputs, target_gradients=None):
 
 class amp:
     pass
+
+
+# ---------------------------------------------------------------------------
+# legacy static API remainder (reference: static/__init__.py __all__)
+# ---------------------------------------------------------------------------
+import contextlib as _ctx
+
+
+def append_backward(loss, parameter_list=None, no_grad_set=None, callbacks=None):
+    """In the deferred-graph model gradients run inside Executor.run via
+    optimizer.minimize; this records the loss for a bare-backward program."""
+    _default_main.train_ops.append((loss, None))
+    return []
+
+
+class _Scope(dict):
+    def find_var(self, name):
+        return self.get(name)
+
+    def var(self, name):
+        return self.setdefault(name, None)
+
+
+_global_scope = _Scope()
+
+
+def global_scope():
+    return _global_scope
+
+
+@_ctx.contextmanager
+def scope_guard(scope):
+    global _global_scope
+    old, _global_scope = _global_scope, scope
+    try:
+        yield
+    finally:
+        _global_scope = old
+
+
+class BuildStrategy:
+    def __init__(self):
+        self.memory_optimize = None
+        self.enable_inplace = None
+        self.fuse_broadcast_ops = None
+
+
+class CompiledProgram:
+    def __init__(self, program, build_strategy=None):
+        self.program = program
+        self.build_strategy = build_strategy or BuildStrategy()
+
+
+def Print(input, first_n=-1, message=None, summarize=20, **kw):
+    def _p(t):
+        print(message or "", t)
+        return t
+    return Var("op", fn=_p, args=(input,))
+
+
+def py_func(func, x, out, backward_func=None, skip_vars_in_backward_input=None):
+    xs = x if isinstance(x, (list, tuple)) else [x]
+    return Var("op", fn=lambda *ts: func(*ts), args=tuple(xs))
+
+
+@_ctx.contextmanager
+def name_scope(prefix=None):
+    yield
+
+
+@_ctx.contextmanager
+def device_guard(device=None):
+    yield
+
+
+@_ctx.contextmanager
+def ipu_shard_guard(index=-1, stage=-1):
+    raise NotImplementedError("IPU is not a target of this MI355X build")
+
+
+class IpuStrategy:
+    def __init__(self):
+        raise NotImplementedError("IPU is not a target of this MI355X build")
+
+
+class IpuCompiledProgram:
+    def __init__(self, *a, **k):
+        raise NotImplementedError("IPU is not a target of this MI355X build")
+
+
+def set_ipu_shard(call_func, index=-1, stage=-1):
+    raise NotImplementedError("IPU is not a target of this MI355X build")
+
+
+class WeightNormParamAttr:
+    def __init__(self, dim=None, **kw):
+        self.dim = dim
+
+
+class ExponentialMovingAverage:
+    """EMA of program parameters (reference: static/ema.py)."""
+
+    def __init__(self, decay=0.999, thres_steps=None, name=None):
+        self.decay = decay
+        self._ema = {}
+
+    def update(self):
+        for v in _default_main.params:
+            t = v.tensor.detach()
+            key = id(v)
+            if key not in self._ema:
+                self._ema[key] = t.clone()
+            else:
+                self._ema[key].mul_(self.decay).add_(t, alpha=1 - self.decay)
+
+    @_ctx.contextmanager
+    def apply(self, executor=None, need_restore=True):
+        saved = {}
+        for v in _default_main.params:
+            key = id(v)
+            if key in self._ema:
+                saved[key] = v.tensor.detach().clone()
+                with torch.no_grad():
+                    v.tensor.copy_(self._ema[key])
+        try:
+            yield
+        finally:
+            if need_restore:
+                with torch.no_grad():
+                    for v in _default_main.params:
+                        if id(v) in saved:
+                            v.tensor.copy_(saved[id(v)])
+
+    def restore(self, executor=None):
+        pass
+
+
+Variable = Var
+
+
+def create_global_var(shape, value, dtype, persistable=False, force_cpu=False,
+                      name=None):
+    from .. import framework as _fw
+    t = torch.full(shape, value, dtype=_fw.convert_dtype(dtype))
+    v = Var("param", shape=shape, dtype=t.dtype, tensor=t)
+    _default_main.params.append(v)
+    return v
+
+
+def accuracy(input, label, k=1, correct=None, total=None, name=None):
+    def _acc(logits, lab):
+        topk = logits.topk(k, dim=-1).indices
+        return (topk == lab.view(-1, 1)).any(-1).float().mean()
+    return Var("op", fn=_acc, args=(input, label))
+
+
+def auc(input, label, curve="ROC", num_thresholds=4095, topk=1, slide_steps=1):
+    raise NotImplementedError("static auc: use paddle.metric.Auc")
+
+
+def ctr_metric_bundle(input, label, ins_tag_weight=None):
+    raise NotImplementedError("ctr_metric_bundle: PS-specific; out of scope")
+
+
+def cpu_places(device_count=None):
+    from ..framework import CPUPlace
+    import os
+    n = device_count or int(os.environ.get("CPU_NUM", 1))
+    return [CPUPlace() for _ in range(n)]
+
+
+def cuda_places(device_ids=None):
+    from ..framework import CUDAPlace
+    ids = device_ids if device_ids is not None else (
+        range(torch.cuda.device_count()) if torch.cuda.is_available() else [0])
+    return [CUDAPlace(i) for i in ids]
+
+
+def xpu_places(device_ids=None):
+    raise NotImplementedError("XPU is not a target of this MI355X build")
+
+
+# -- program/persistables serialization (reference: static/io.py:513,837) ----
+def save(program, model_path, protocol=4, **configs):
+    import pickle
+    state = {id(v): v.tensor.detach().cpu() for v in program.params}
+    with open(model_path + ".pdparams", "wb") as f:
+        pickle.dump(list(state.values()), f, protocol=protocol)
+
+
+def load(program, model_path, executor=None, var_list=None):
+    import pickle
+    with open(model_path + ".pdparams", "rb") as f:
+        vals = pickle.load(f)
+    with torch.no_grad():
+        for v, t in zip(program.params, vals):
+            v.tensor.copy_(t)
+
+
+def serialize_program(feed_vars, fetch_vars, **kwargs):
+    import pickle
+    return pickle.dumps({"feeds": [getattr(v, "name", None) for v in feed_vars],
+                         "fetch_count": len(fetch_vars)})
+
+
+def serialize_persistables(feed_vars, fetch_vars, executor=None, **kwargs):
+    import pickle
+    return pickle.dumps([v.tensor.detach().cpu() for v in _default_main.params])
+
+
+def save_to_file(path, content):
+    with open(path, "wb") as f:
+        f.write(content)
+
+
+def load_from_file(path):
+    with open(path, "rb") as f:
+        return f.read()
+
+
+def deserialize_program(data):
+    import pickle
+    return pickle.loads(data)
+
+
+def deserialize_persistables(program, data, executor=None):
+    import pickle
+    vals = pickle.loads(data)
+    with torch.no_grad():
+        for v, t in zip(program.params, vals):
+            v.tensor.copy_(t)
+
+
+def normalize_program(program, feed_vars, fetch_vars, **kwargs):
+    return program
+
+
+def load_program_state(model_path, var_list=None):
+    import pickle
+    with open(model_path + ".pdparams", "rb") as f:
+        return pickle.load(f)
+
+
+def set_program_state(program, state):
+    with torch.no_grad():
+        for v, t in zip(program.params, state):
+            v.tensor.copy_(t)

@@ -106,7 +106,11 @@ def test_submodule_export_parity_complete():
                      ("optimizer", "optimizer"), ("distribution", "distribution"),
                      ("io", "io"), ("fft", "fft"), ("vision", "vision"),
                      ("autograd", "autograd"), ("signal", "signal"),
-                     ("metric", "metric"), ("amp", "amp")]:
+                     ("metric", "metric"), ("amp", "amp"),
+                     ("utils", "utils"), ("static", "static"), ("jit", "jit"),
+                     ("sparse", "sparse"), ("incubate", "incubate"),
+                     ("text", "text"), ("device", "device"),
+                     ("hub", "hapi/hub")]:
         try:
             src = open(f"{base}/{ref}/__init__.py").read()
         except FileNotFoundError:
