@@ -149,3 +149,97 @@ class FusedTransformerEncoderLayer(Layer):
 
     def forward(self, src, src_mask=None, cache=None):
         return self.ffn(self.fused_attn(src, src_mask))
+
+
+class FusedLinear(Layer):
+    """Linear whose backward fuses dW accumulation into one GEMM
+    (reference: incubate/nn/layer/fused_linear.py)."""
+
+    def __init__(self, in_features, out_features, weight_attr=None,
+                 bias_attr=None, transpose_weight=False, name=None):
+        super().__init__()
+        self.weight = self.create_parameter([in_features, out_features])
+        self.bias = (None if bias_attr is False
+                     else self.create_parameter([out_features], is_bias=True))
+
+    def forward(self, x):
+        out = torch.matmul(x, self.weight)
+        return out + self.bias if self.bias is not None else out
+
+
+class FusedDropoutAdd(Layer):
+    """out = residual + dropout(x) in one HIP kernel (reference:
+    incubate/nn/layer/fused_dropout_add.py -> fused_dropout_add kernel)."""
+
+    def __init__(self, p=0.5, mode="upscale_in_train", name=None):
+        super().__init__()
+        self.p = p
+
+    def forward(self, x, y):
+        return hot.dropout_add(x, y, self.p, self.training)
+
+
+class FusedBiasDropoutResidualLayerNorm(Layer):
+    """out = LN(residual + dropout(x + bias)) (reference:
+    incubate/nn/layer/fused_transformer.py:FusedBiasDropoutResidualLayerNorm)."""
+
+    def __init__(self, embed_dim, dropout_rate=0.5, weight_attr=None,
+                 bias_attr=None, epsilon=1e-5, name=None):
+        super().__init__()
+        self.bias = self.create_parameter([embed_dim], is_bias=True)
+        self.ln_weight = self.create_parameter(
+            [embed_dim], default_initializer=Constant(1.0))
+        self.ln_bias = self.create_parameter([embed_dim], is_bias=True)
+        self.dropout_rate = dropout_rate
+        self.epsilon = epsilon
+
+    def forward(self, x, residual):
+        h = hot.dropout_add(x + self.bias, residual, self.dropout_rate,
+                            self.training)
+        return hot.layer_norm(h, self.ln_weight, self.ln_bias, self.epsilon)
+
+
+class FusedMultiTransformer(Layer):
+    """Inference decoder stack (reference: incubate/nn/layer/
+    fused_transformer.py:FusedMultiTransformer -- the
+    fused_multi_transformer kernel's layer API).  Runs pre-LN attention +
+    FFN per layer through the flash-attention HIP path; weights are given
+    as per-layer lists like the reference."""
+
+    def __init__(self, embed_dim, num_heads, dim_feedforward,
+                 dropout_rate=0.0, activation="gelu", normalize_before=True,
+                 ln_scale_attrs=None, qkv_weight_attrs=None, num_layers=-1,
+                 nranks=1, trans_qkvw=True, ring_id=-1, name=None, **kw):
+        super().__init__()
+        assert normalize_before, "post-LN variant: round 2"
+        self.embed_dim = embed_dim
+        self.num_heads = num_heads
+        n = num_layers if num_layers > 0 else (
+            len(qkv_weight_attrs) if qkv_weight_attrs else 1)
+        self.layers = pnn.LayerList()
+        for _ in range(n):
+            blk = Layer()
+            blk.ln1_w = blk.create_parameter([embed_dim],
+                                             default_initializer=Constant(1.0))
+            blk.ln1_b = blk.create_parameter([embed_dim], is_bias=True)
+            blk.qkv = pnn.Linear(embed_dim, 3 * embed_dim)
+            blk.proj = pnn.Linear(embed_dim, embed_dim)
+            blk.ln2_w = blk.create_parameter([embed_dim],
+                                             default_initializer=Constant(1.0))
+            blk.ln2_b = blk.create_parameter([embed_dim], is_bias=True)
+            blk.fc1 = pnn.Linear(embed_dim, dim_feedforward)
+            blk.fc2 = pnn.Linear(dim_feedforward, embed_dim)
+            self.layers.append(blk)
+
+    def forward(self, src, attn_mask=None, caches=None, time_step=None, **kw):
+        x = src
+        b, s, _ = x.shape
+        h, d = self.num_heads, self.embed_dim // self.num_heads
+        for blk in self.layers:
+            y = hot.layer_norm(x, blk.ln1_w, blk.ln1_b, 1e-5)
+            qkv = blk.qkv(y).reshape(b, s, 3, h, d)
+            attn = hot.qkv_flash_attention(qkv, causal=True)
+            x = x + blk.proj(attn)
+            y = hot.layer_norm(x, blk.ln2_w, blk.ln2_b, 1e-5)
+            x = x + blk.fc2(hot.bias_gelu(blk.fc1(y), None))
+        return x

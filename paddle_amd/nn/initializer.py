@@ -120,3 +120,47 @@ def set_global_initializer(weight_init=None, bias_init=None):
 
 _GLOBAL_WEIGHT_INIT = None
 _GLOBAL_BIAS_INIT = None
+
+
+class Bilinear(Initializer):
+    """Bilinear upsampling kernel init (reference: nn/initializer/Bilinear)."""
+
+    def __call__(self, tensor):
+        import math
+        with torch.no_grad():
+            shape = tensor.shape
+            f = math.ceil(shape[-1] / 2)
+            c = (2 * f - 1 - f % 2) / (2.0 * f)
+            for idx in range(tensor.numel()):
+                coords = []
+                rem = idx
+                for s in reversed(shape):
+                    coords.append(rem % s)
+                    rem //= s
+                x, y = coords[0], coords[1] if len(coords) > 1 else 0
+                tensor.view(-1)[idx] = (1 - abs(x / f - c)) * (1 - abs(y / f - c))
+        return tensor
+
+
+class Orthogonal(Initializer):
+    def __init__(self, gain=1.0, name=None):
+        self.gain = gain
+
+    def __call__(self, tensor):
+        with torch.no_grad():
+            torch.nn.init.orthogonal_(tensor, gain=self.gain)
+        return tensor
+
+
+class Dirac(Initializer):
+    def __init__(self, groups=1, name=None):
+        self.groups = groups
+
+    def __call__(self, tensor):
+        with torch.no_grad():
+            torch.nn.init.dirac_(tensor, groups=self.groups)
+        return tensor
+
+
+def calculate_gain(nonlinearity, param=None):
+    return torch.nn.init.calculate_gain(nonlinearity, param)

@@ -1,3 +1,4 @@
+from . import datasets  # noqa: F401
 from . import models  # noqa: F401
 from . import transforms  # noqa: F401
 
@@ -24,3 +25,6 @@ def image_load(path, backend=None):
     except ImportError as e:
         raise RuntimeError("image_load needs Pillow (not in this image); "
                            "decode with numpy and use paddle.to_tensor") from e
+
+
+from . import datasets  # noqa: F401

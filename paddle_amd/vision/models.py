@@ -63,13 +63,16 @@ class BasicBlock(nn.Layer):
 class BottleneckBlock(nn.Layer):
     expansion = 4
 
-    def __init__(self, inplanes, planes, stride=1, downsample=None):
+    def __init__(self, inplanes, planes, stride=1, downsample=None, groups=1,
+                 base_width=64):
         super().__init__()
-        self.conv1 = nn.Conv2D(inplanes, planes, 1, bias_attr=False)
-        self.bn1 = nn.BatchNorm2D(planes)
-        self.conv2 = nn.Conv2D(planes, planes, 3, stride=stride, padding=1, bias_attr=False)
-        self.bn2 = nn.BatchNorm2D(planes)
-        self.conv3 = nn.Conv2D(planes, planes * 4, 1, bias_attr=False)
+        width = int(planes * (base_width / 64.0)) * groups
+        self.conv1 = nn.Conv2D(inplanes, width, 1, bias_attr=False)
+        self.bn1 = nn.BatchNorm2D(width)
+        self.conv2 = nn.Conv2D(width, width, 3, stride=stride, padding=1,
+                               groups=groups, bias_attr=False)
+        self.bn2 = nn.BatchNorm2D(width)
+        self.conv3 = nn.Conv2D(width, planes * 4, 1, bias_attr=False)
         self.bn3 = nn.BatchNorm2D(planes * 4)
         self.relu = nn.ReLU()
         self.downsample = downsample
@@ -87,8 +90,11 @@ class BottleneckBlock(nn.Layer):
 class ResNet(nn.Layer):
     """reference: python/paddle/vision/models/resnet.py"""
 
-    def __init__(self, block, depth=50, width=64, num_classes=1000, with_pool=True):
+    def __init__(self, block, depth=50, width=64, num_classes=1000,
+                 with_pool=True, groups=1):
         super().__init__()
+        self.groups = groups
+        self.base_width = width
         layer_cfg = {18: [2, 2, 2, 2], 34: [3, 4, 6, 3], 50: [3, 4, 6, 3],
                      101: [3, 4, 23, 3], 152: [3, 8, 36, 3]}
         layers = layer_cfg[depth]
@@ -116,10 +122,12 @@ class ResNet(nn.Layer):
                           bias_attr=False),
                 nn.BatchNorm2D(planes * block.expansion),
             )
-        layers = [block(self.inplanes, planes, stride, downsample)]
+        kw = ({"groups": self.groups, "base_width": self.base_width}
+              if block is BottleneckBlock else {})
+        layers = [block(self.inplanes, planes, stride, downsample, **kw)]
         self.inplanes = planes * block.expansion
         for _ in range(1, blocks):
-            layers.append(block(self.inplanes, planes))
+            layers.append(block(self.inplanes, planes, **kw))
         return nn.Sequential(*layers)
 
     def forward(self, x):
@@ -147,3 +155,33 @@ def resnet50(pretrained=False, **kwargs):
 
 def resnet101(pretrained=False, **kwargs):
     return ResNet(BottleneckBlock, 101, **kwargs)
+
+
+def resnet152(pretrained=False, **kwargs):
+    return ResNet(BottleneckBlock, 152, **kwargs)
+
+
+def _resnext(depth, groups, width):
+    def f(pretrained=False, **kwargs):
+        return ResNet(BottleneckBlock, depth, width=width, groups=groups,
+                      **kwargs)
+    return f
+
+
+resnext50_32x4d = _resnext(50, 32, 4)
+resnext50_64x4d = _resnext(50, 64, 4)
+resnext101_32x4d = _resnext(101, 32, 4)
+resnext101_64x4d = _resnext(101, 64, 4)
+resnext152_32x4d = _resnext(152, 32, 4)
+resnext152_64x4d = _resnext(152, 64, 4)
+
+
+def wide_resnet50_2(pretrained=False, **kwargs):
+    return ResNet(BottleneckBlock, 50, width=128, **kwargs)
+
+
+def wide_resnet101_2(pretrained=False, **kwargs):
+    return ResNet(BottleneckBlock, 101, width=128, **kwargs)
+
+
+from .models_extra import *  # noqa: F401,F403,E402

@@ -110,7 +110,12 @@ def test_submodule_export_parity_complete():
                      ("utils", "utils"), ("static", "static"), ("jit", "jit"),
                      ("sparse", "sparse"), ("incubate", "incubate"),
                      ("text", "text"), ("device", "device"),
-                     ("hub", "hapi/hub")]:
+                     ("hub", "hapi/hub"),
+                     ("vision.transforms", "vision/transforms"),
+                     ("vision.models", "vision/models"),
+                     ("vision.datasets", "vision/datasets"),
+                     ("nn.initializer", "nn/initializer"),
+                     ("incubate.nn", "incubate/nn")]:
         try:
             src = open(f"{base}/{ref}/__init__.py").read()
         except FileNotFoundError:
