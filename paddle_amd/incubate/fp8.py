@@ -73,9 +73,25 @@ class Fp8Linear(torch.nn.Module):
 
 
 def convert_experts_to_fp8(moe_model):
-    """Swap every MoE ExpertMLP Linear to the fp8 path in-place."""
-    from ..models.moe import ExpertMLP, MoELayer
+    """Swap MoE expert compute to the fp8 GEMM path in-place -- handles
+    both the grouped (stacked-weight bmm) and per-expert layouts."""
+    from ..models.moe import ExpertMLP, GroupedExperts
     for mod in moe_model.modules():
+        if isinstance(mod, GroupedExperts):
+            mod.fp8 = True
+
+            def gfwd(self, x):
+                # per-expert fp8 GEMMs over the stacked weights
+                outs = []
+                for e in range(self.num_local):
+                    h = fp8_matmul(x[e], self.w1[e]) + self.b1[e]
+                    from ..ops import functional as hot
+                    h = (torch.nn.functional.gelu(h.float()).to(x.dtype)
+                         if not x.is_cuda else hot.bias_gelu(h, None))
+                    outs.append(fp8_matmul(h, self.w2[e]) + self.b2[e])
+                return torch.stack(outs)
+
+            mod.forward = gfwd.__get__(mod)
         if isinstance(mod, ExpertMLP):
             mod.fp8 = True
 
