@@ -92,6 +92,14 @@ def main():
         return crit(m(ids), labels)
 
     run("gpt-moe-64e 350M-trunk b8 s1024", moe, moe_batch, moe_loss)
+    # fp8 expert GEMMs (BASELINE config 5: expert-parallel + fp8 MFMA)
+    from paddle_amd.incubate.fp8 import convert_experts_to_fp8
+    torch.cuda.empty_cache()
+    torch.cuda.reset_peak_memory_stats()
+    moe8 = GPTMoEForPretraining(cfg, num_experts=64, k=2)
+    convert_experts_to_fp8(moe8)
+    run("gpt-moe-64e fp8-experts", moe8, moe_batch, moe_loss)
+    del moe8
     del moe
     torch.cuda.empty_cache()
     torch.cuda.reset_peak_memory_stats()
