@@ -22,6 +22,64 @@ def _quant(x: torch.Tensor):
     return q, (1.0 / scale).reshape(1)
 
 
+class _Fp8GroupedFFN(torch.autograd.Function):
+    """Stacked-expert fp8 FFN: ONE quantize pass over the whole stacked
+    activation / weight tensors (per-tensor scales), then per-expert
+    _scaled_mm calls -- drops the 4-kernels-per-expert quantize overhead
+    that made the naive loop launch-bound (measured 8x slower than bf16).
+    Backward stays bf16 batched bmm."""
+
+    @staticmethod
+    def forward(ctx, x, w1, b1, w2, b2):
+        # x [E, N, h]; w1 [E, h, I]; w2 [E, I, h]
+        from ..ops import functional as hot
+        if not x.is_cuda:
+            h = torch.nn.functional.gelu((x.float() @ w1.float() + b1.float().unsqueeze(1)))
+            out = (h @ w2.float() + b2.float().unsqueeze(1)).to(x.dtype)
+            ctx.save_for_backward(x, w1, b1, w2, b2)
+            return out
+        qx, sx = _quant(x)
+        qw1, sw1 = _quant(w1)
+        E = x.shape[0]
+        hs = []
+        for e in range(E):
+            wt = qw1[e].t().contiguous().t()
+            hs.append(torch._scaled_mm(qx[e], wt, scale_a=sx, scale_b=sw1,
+                                       out_dtype=torch.bfloat16))
+        h = torch.stack(hs) + b1.unsqueeze(1)
+        g = hot.bias_gelu(h, None)
+        qg, sg = _quant(g)
+        qw2, sw2 = _quant(w2)
+        outs = []
+        for e in range(E):
+            wt = qw2[e].t().contiguous().t()
+            outs.append(torch._scaled_mm(qg[e], wt, scale_a=sg, scale_b=sw2,
+                                         out_dtype=torch.bfloat16))
+        out = torch.stack(outs) + b2.unsqueeze(1)
+        ctx.save_for_backward(x, w1, b1, w2, b2)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w1, b1, w2, b2 = ctx.saved_tensors
+        xf = x.float()
+        z = torch.baddbmm(b1.float().unsqueeze(1), xf, w1.float())
+        g = torch.nn.functional.gelu(z)
+        dg = torch.bmm(dy.float(), w2.float().transpose(1, 2))
+        dw2 = torch.bmm(g.transpose(1, 2), dy.float())
+        db2 = dy.float().sum(1)
+        # dgelu (erf form)
+        cdf = 0.5 * (1 + torch.erf(z * 0.7071067811865476))
+        pdf = 0.3989422804014327 * torch.exp(-0.5 * z * z)
+        dz = dg * (cdf + z * pdf)
+        dx = torch.bmm(dz, w1.float().transpose(1, 2))
+        dw1 = torch.bmm(xf.transpose(1, 2), dz)
+        db1 = dz.sum(1)
+        dt = x.dtype
+        return (dx.to(dt), dw1.to(w1.dtype), db1.to(b1.dtype),
+                dw2.to(w2.dtype), db2.to(b2.dtype))
+
+
 class _Fp8Matmul(torch.autograd.Function):
     """out = x @ w  with x,w quantized to e4m3 for the MFMA fp8 path."""
 
@@ -81,15 +139,8 @@ def convert_experts_to_fp8(moe_model):
             mod.fp8 = True
 
             def gfwd(self, x):
-                # per-expert fp8 GEMMs over the stacked weights
-                outs = []
-                for e in range(self.num_local):
-                    h = fp8_matmul(x[e], self.w1[e]) + self.b1[e]
-                    from ..ops import functional as hot
-                    h = (torch.nn.functional.gelu(h.float()).to(x.dtype)
-                         if not x.is_cuda else hot.bias_gelu(h, None))
-                    outs.append(fp8_matmul(h, self.w2[e]) + self.b2[e])
-                return torch.stack(outs)
+                return _Fp8GroupedFFN.apply(x, self.w1, self.b1, self.w2,
+                                            self.b2)
 
             mod.forward = gfwd.__get__(mod)
         if isinstance(mod, ExpertMLP):
