@@ -193,3 +193,144 @@ class Model:
         n_params = sum(p.numel() for p in self.network.parameters())
         print(f"Total params: {n_params:,}")
         return {"total_params": n_params}
+
+
+class EarlyStopping(Callback):
+    """Stop fit() when a monitored metric stops improving (reference:
+    hapi/callbacks.py:EarlyStopping)."""
+
+    def __init__(self, monitor="loss", mode="auto", patience=0, verbose=1,
+                 min_delta=0, baseline=None, save_best_model=True):
+        self.monitor = monitor
+        self.patience = patience
+        self.min_delta = abs(min_delta)
+        self.baseline = baseline
+        self.save_best_model = save_best_model
+        if mode == "max" or (mode == "auto" and "acc" in monitor):
+            self.better = lambda cur, best: cur > best + self.min_delta
+            self.best = -float("inf")
+        else:
+            self.better = lambda cur, best: cur < best - self.min_delta
+            self.best = float("inf")
+        if baseline is not None:
+            self.best = baseline
+        self.wait = 0
+        self.stopped_epoch = 0
+
+    def on_eval_end(self, logs=None):
+        logs = logs or {}
+        cur = logs.get(self.monitor)
+        if cur is None:
+            return
+        cur = float(cur[0] if isinstance(cur, (list, tuple)) else cur)
+        if self.better(cur, self.best):
+            self.best = cur
+            self.wait = 0
+        else:
+            self.wait += 1
+            if self.wait >= self.patience:
+                self.model.stop_training = True
+
+
+class LRScheduler(Callback):
+    """Step the optimizer's LRScheduler per epoch/batch (reference:
+    hapi/callbacks.py:LRScheduler)."""
+
+    def __init__(self, by_step=False, by_epoch=True):
+        self.by_step = by_step
+        self.by_epoch = by_epoch
+
+    def _sched(self):
+        opt = getattr(self.model, "_optimizer", None)
+        lr = getattr(opt, "_lr", None) if opt else None
+        return lr if hasattr(lr, "step") else None
+
+    def on_epoch_end(self, epoch, logs=None):
+        s = self._sched()
+        if self.by_epoch and s:
+            s.step()
+
+    def on_train_batch_end(self, step, logs=None):
+        s = self._sched()
+        if self.by_step and s:
+            s.step()
+
+
+class ReduceLROnPlateau(Callback):
+    """Scale LR down when the monitored metric plateaus (reference:
+    hapi/callbacks.py:ReduceLROnPlateau)."""
+
+    def __init__(self, monitor="loss", factor=0.1, patience=10, verbose=1,
+                 mode="auto", min_delta=1e-4, cooldown=0, min_lr=0):
+        self.monitor = monitor
+        self.factor = factor
+        self.patience = patience
+        self.cooldown = cooldown
+        self.min_lr = min_lr
+        self.min_delta = min_delta
+        if mode == "max" or (mode == "auto" and "acc" in monitor):
+            self.better = lambda cur, best: cur > best + min_delta
+            self.best = -float("inf")
+        else:
+            self.better = lambda cur, best: cur < best - min_delta
+            self.best = float("inf")
+        self.wait = 0
+        self.cooldown_counter = 0
+
+    def on_eval_end(self, logs=None):
+        logs = logs or {}
+        cur = logs.get(self.monitor)
+        if cur is None:
+            return
+        cur = float(cur[0] if isinstance(cur, (list, tuple)) else cur)
+        if self.cooldown_counter > 0:
+            self.cooldown_counter -= 1
+            return
+        if self.better(cur, self.best):
+            self.best = cur
+            self.wait = 0
+            return
+        self.wait += 1
+        if self.wait >= self.patience:
+            opt = getattr(self.model, "_optimizer", None)
+            if opt is not None:
+                new_lr = max(opt.get_lr() * self.factor, self.min_lr)
+                opt.set_lr(new_lr)
+            self.cooldown_counter = self.cooldown
+            self.wait = 0
+
+
+class VisualDL(Callback):
+    """Scalar logging; VisualDL itself is not installed, so this writes a
+    plain JSONL the visualdl UI can be pointed at later."""
+
+    def __init__(self, log_dir="./log"):
+        self.log_dir = log_dir
+        self._f = None
+
+    def _write(self, tag, step, logs):
+        import json
+        import os
+        if self._f is None:
+            os.makedirs(self.log_dir, exist_ok=True)
+            self._f = open(f"{self.log_dir}/scalars.jsonl", "a")
+        rec = {"tag": tag, "step": step}
+        for k, v in (logs or {}).items():
+            try:
+                rec[k] = float(v[0] if isinstance(v, (list, tuple)) else v)
+            except (TypeError, ValueError):
+                pass
+        self._f.write(json.dumps(rec) + "\n")
+        self._f.flush()
+
+    def on_train_batch_end(self, step, logs=None):
+        self._write("train", step, logs)
+
+    def on_eval_end(self, logs=None):
+        self._write("eval", 0, logs)
+
+
+class WandbCallback(Callback):
+    def __init__(self, *a, **kw):
+        raise RuntimeError("wandb is not installed in this image (no network "
+                           "egress); use VisualDL for local JSONL logging")

@@ -100,3 +100,81 @@ class Predictor:
 
 def create_predictor(config: Config):
     return Predictor(config)
+
+
+# breadth parity (reference: python/paddle/inference/__init__.py __all__)
+import enum as _enum
+
+import torch as _torch
+
+
+class DataType(_enum.Enum):
+    FLOAT32 = 0
+    INT64 = 1
+    INT32 = 2
+    UINT8 = 3
+    INT8 = 4
+    FLOAT16 = 5
+    BFLOAT16 = 6
+    FLOAT64 = 7
+    BOOL = 8
+
+
+class PlaceType(_enum.Enum):
+    UNK = -1
+    CPU = 0
+    GPU = 1
+
+
+class PrecisionType(_enum.Enum):
+    Float32 = 0
+    Half = 1
+    Int8 = 2
+    Bfloat16 = 3
+
+
+Tensor = _torch.Tensor
+
+
+def get_version():
+    from .. import __version__
+    return f"paddle_amd inference {__version__}"
+
+
+def get_num_bytes_of_data_type(dtype):
+    return {DataType.FLOAT32: 4, DataType.INT64: 8, DataType.INT32: 4,
+            DataType.UINT8: 1, DataType.INT8: 1, DataType.FLOAT16: 2,
+            DataType.BFLOAT16: 2, DataType.FLOAT64: 8, DataType.BOOL: 1}[dtype]
+
+
+def get_trt_compile_version():
+    return (0, 0, 0)  # TensorRT does not exist on ROCm; MIGraphX is round 2
+
+
+def get_trt_runtime_version():
+    return (0, 0, 0)
+
+
+def _get_phi_kernel_name(op_name):
+    return op_name
+
+
+def convert_to_mixed_precision(model_file, params_file, mixed_model_file,
+                               mixed_params_file, mixed_precision, backend,
+                               **kwargs):
+    raise NotImplementedError(
+        "convert_to_mixed_precision operates on .pdmodel graphs; use "
+        "paddle.amp at runtime instead")
+
+
+class PredictorPool:
+    def __init__(self, config, size=1):
+        self._preds = [create_predictor(config) for _ in range(size)]
+
+    def retrieve(self, idx):
+        return self._preds[idx]
+
+
+class XpuConfig:
+    def __init__(self, *a, **kw):
+        raise NotImplementedError("XPU is not a target of this MI355X build")

@@ -115,7 +115,9 @@ def test_submodule_export_parity_complete():
                      ("vision.models", "vision/models"),
                      ("vision.datasets", "vision/datasets"),
                      ("nn.initializer", "nn/initializer"),
-                     ("incubate.nn", "incubate/nn")]:
+                     ("incubate.nn", "incubate/nn"),
+                     ("callbacks", "callbacks"),
+                     ("inference", "inference")]:
         try:
             src = open(f"{base}/{ref}/__init__.py").read()
         except FileNotFoundError:
