@@ -164,3 +164,38 @@ def test_extension_imports_when_built():
     so = os.path.join(os.path.dirname(paddle_amd.__file__), "_C.so")
     if os.path.exists(so):
         import paddle_amd._C  # noqa: F401
+
+
+def test_early_stopping_and_reduce_lr():
+    import paddle_amd as paddle
+    from paddle_amd.callbacks import EarlyStopping, ReduceLROnPlateau
+
+    class _FakeModel:
+        stop_training = False
+
+    es = EarlyStopping(monitor="loss", patience=2, mode="min")
+    es.model = _FakeModel()
+    for loss in (1.0, 0.5, 0.6, 0.7, 0.8):
+        es.on_eval_end({"loss": loss})
+    assert es.model.stop_training
+
+    m = paddle.nn.Linear(2, 2)
+    opt = paddle.optimizer.SGD(learning_rate=1.0, parameters=m.parameters())
+
+    class _M2:
+        _optimizer = opt
+    rl = ReduceLROnPlateau(monitor="loss", factor=0.5, patience=1, min_delta=0)
+    rl.model = _M2()
+    for loss in (1.0, 1.0, 1.0):
+        rl.on_eval_end({"loss": loss})
+    assert abs(opt.get_lr() - 0.5) < 1e-9
+
+
+def test_visualdl_writes_jsonl(tmp_path):
+    import json
+    from paddle_amd.callbacks import VisualDL
+    v = VisualDL(log_dir=str(tmp_path))
+    v.on_train_batch_end(1, {"loss": 0.5})
+    v.on_eval_end({"acc": [0.9]})
+    lines = [json.loads(l) for l in open(tmp_path / "scalars.jsonl")]
+    assert lines[0]["loss"] == 0.5 and lines[1]["acc"] == 0.9
