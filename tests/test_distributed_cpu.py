@@ -236,3 +236,46 @@ def test_sharded_checkpoint_reshard_w2_to_w1(tmp_path):
     for u_loaded, u_ref in zip(wrapped._units, w2._units):
         torch.testing.assert_close(u_loaded.master, u_ref.master,
                                    atol=1e-5, rtol=1e-5)
+
+
+def test_sharding_stage1_stage2_match_single():
+    """group_sharded_parallel levels os (stage-1) and os_g (stage-2) on two
+    ranks produce the same updated params as single-process training
+    (reference: sharding stage semantics, dygraph_sharding_optimizer.py /
+    group_sharded stage2)."""
+    run_dist("""
+        import torch
+        import paddle_amd as paddle
+        paddle.distributed.init_parallel_env()
+        from paddle_amd.distributed.fleet.sharding import group_sharded_parallel
+        r = paddle.distributed.get_rank()
+
+        for level in ("os", "os_g"):
+            torch.manual_seed(7)
+            m = paddle.nn.Sequential(paddle.nn.Linear(8, 16),
+                                     paddle.nn.Linear(16, 8))
+            inner = paddle.optimizer.AdamW(learning_rate=1e-2,
+                                           parameters=m.parameters())
+            wrapped, opt, _ = group_sharded_parallel(m, inner, level)
+            torch.manual_seed(11)
+            x = torch.randn(4, 8)
+            loss = wrapped(x).pow(2).mean()
+            loss.backward()
+            opt.step()
+            opt.clear_grad()
+
+            # single-process reference
+            torch.manual_seed(7)
+            m2 = paddle.nn.Sequential(paddle.nn.Linear(8, 16),
+                                      paddle.nn.Linear(16, 8))
+            o2 = paddle.optimizer.AdamW(learning_rate=1e-2,
+                                        parameters=m2.parameters())
+            torch.manual_seed(11)
+            x2 = torch.randn(4, 8)
+            l2 = m2(x2).pow(2).mean()
+            l2.backward()
+            o2.step()
+            for a, b in zip(m.parameters(), m2.parameters()):
+                torch.testing.assert_close(a, b, atol=2e-5, rtol=2e-5), level
+        print("rank", r, "stage1/2 ok")
+    """, world_size=2)
