@@ -32,15 +32,15 @@ class PagedKVCache:
         self.seq_lens = torch.zeros(batch, device=device, dtype=torch.int32)
 
     def append(self, layer, k_new, v_new, start_pos):
-        """k_new/v_new: [B, S_new, HKV, D]; positions start_pos..start_pos+S_new"""
+        """k_new/v_new: [B, S_new, HKV, D]; positions start_pos..start_pos+S_new
+        (vectorized scatter: one advanced-indexing write per layer)"""
         B, S_new, HKV, D = k_new.shape
-        for b in range(B):
-            for s in range(S_new):
-                pos = start_pos + s
-                blk = int(self.block_table[b, pos // self.bs])
-                off = pos % self.bs
-                self.k[layer][blk, off] = k_new[b, s]
-                self.v[layer][blk, off] = v_new[b, s]
+        pos = torch.arange(start_pos, start_pos + S_new,
+                           device=k_new.device)               # [S]
+        blks = self.block_table.long()[:, pos // self.bs]     # [B, S]
+        offs = (pos % self.bs).unsqueeze(0).expand(B, S_new)  # [B, S]
+        self.k[layer][blks, offs] = k_new
+        self.v[layer][blks, offs] = v_new
 
     def advance(self, n):
         self.seq_lens += n
