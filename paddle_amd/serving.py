@@ -215,7 +215,7 @@ class GPTModelRunner:
     decode runs every active request through `paged_decode_attention`."""
 
     def __init__(self, model, num_blocks=1024, block_size=16,
-                 device=None, dtype=None):
+                 device=None, dtype=None, max_seq=2048, use_graphs=True):
         import math
         from .ops import functional as hot
         self.hot = hot
@@ -228,12 +228,18 @@ class GPTModelRunner:
         dev = device or next(iter(model.state_dict().values())).device
         dt = dtype or model.lm_head.weight.dtype
         self.dev = dev
-        self.k = [torch.zeros(num_blocks, block_size, self.H, self.D,
+        # +1 scratch block: hipGraph-padded rows write their (discarded) KV
+        # into it so every captured bucket replays with fixed shapes
+        self.scratch_blk = num_blocks
+        self.k = [torch.zeros(num_blocks + 1, block_size, self.H, self.D,
                               device=dev, dtype=dt)
                   for _ in range(cfg.num_layers)]
         self.v = [torch.zeros_like(self.k[0]) for _ in range(cfg.num_layers)]
         self.seq_len = {}            # rid -> tokens currently cached
         self.last_token = {}         # rid -> next input token
+        self.max_blk = (max_seq + block_size - 1) // block_size
+        self.use_graphs = use_graphs and dev.type == "cuda"
+        self._graphs = {}            # bucket B -> (graph, static tensors)
 
     def _qkv(self, layer, x):
         b, s, _ = x.shape
@@ -275,20 +281,44 @@ class GPTModelRunner:
         if not rest:
             return out
         B = len(rest)
-        max_blk = max(len(blocks[r.rid]) for r in rest)
-        table = torch.zeros(B, max_blk, device=self.dev, dtype=torch.int32)
+        Bp = B if not self.use_graphs else self._bucket(B)
+        table = torch.zeros(Bp, self.max_blk, device=self.dev,
+                            dtype=torch.int32)
+        lens_l = []
         for i, r in enumerate(rest):
             bl = blocks[r.rid]
             table[i, :len(bl)] = torch.tensor(bl, device=self.dev,
                                               dtype=torch.int32)
-        lens = torch.tensor([self.seq_len[r.rid] for r in rest],
-                            device=self.dev, dtype=torch.int32)
-        toks = torch.tensor([[self.last_token[r.rid]] for r in rest],
-                            device=self.dev, dtype=torch.long)
+            lens_l.append(self.seq_len[r.rid])
+        # padded rows decode a single scratch slot
+        table[B:, 0] = self.scratch_blk
+        lens = torch.tensor(lens_l + [0] * (Bp - B), device=self.dev,
+                            dtype=torch.int32)
+        toks = torch.tensor([[self.last_token[r.rid]] for r in rest] +
+                            [[0]] * (Bp - B), device=self.dev, dtype=torch.long)
         pos_ids = lens.long().unsqueeze(1)
-        write_blk = table.long().gather(
-            1, (pos_ids // self.bs)).squeeze(1)
+        write_blk = table.long().gather(1, (pos_ids // self.bs)).squeeze(1)
         write_off = (pos_ids % self.bs).squeeze(1)
+        write_blk[B:] = self.scratch_blk
+        logits = self._decode(Bp, toks, pos_ids, table, lens, write_blk,
+                              write_off)
+        for i, r in enumerate(rest):
+            t = sample_token(logits[i], r.temperature, r.top_p)
+            self.seq_len[r.rid] += 1
+            self.last_token[r.rid] = t
+            out[r.rid] = t
+        return out
+
+    @staticmethod
+    def _bucket(b):
+        n = 1
+        while n < b:
+            n *= 2
+        return n
+
+    def _decode_forward(self, B, toks, pos_ids, table, lens, write_blk,
+                        write_off):
+        m = self.model
         x = m.gpt.embeddings(toks, pos_ids)
         for li, layer in enumerate(m.gpt.layers):
             h = layer.ln1(x)
@@ -300,10 +330,39 @@ class GPTModelRunner:
                 table, lens + 1, self.scale)
             x = x + layer.attn.out_proj(att.reshape(B, 1, -1))
             x = x + layer.mlp(layer.ln2(x))
-        logits = m.lm_head(m.gpt.final_norm(x[:, 0]))
-        for i, r in enumerate(rest):
-            t = sample_token(logits[i], r.temperature, r.top_p)
-            self.seq_len[r.rid] += 1
-            self.last_token[r.rid] = t
-            out[r.rid] = t
-        return out
+        return m.lm_head(m.gpt.final_norm(x[:, 0]))
+
+    def _decode(self, B, toks, pos_ids, table, lens, write_blk, write_off):
+        if not self.use_graphs:
+            return self._decode_forward(B, toks, pos_ids, table, lens,
+                                        write_blk, write_off)
+        # hipGraph per batch bucket: decode is launch-bound (~200 small
+        # kernels/step); one graph replay replaces them all
+        entry = self._graphs.get(B)
+        if entry is None:
+            st = {"toks": toks.clone(), "pos": pos_ids.clone(),
+                  "table": table.clone(), "lens": lens.clone(),
+                  "wblk": write_blk.clone(), "woff": write_off.clone()}
+            stream = torch.cuda.Stream()
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                for _ in range(2):   # warmup outside capture
+                    self._decode_forward(B, st["toks"], st["pos"], st["table"],
+                                         st["lens"], st["wblk"], st["woff"])
+            torch.cuda.current_stream().wait_stream(stream)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                st["logits"] = self._decode_forward(
+                    B, st["toks"], st["pos"], st["table"], st["lens"],
+                    st["wblk"], st["woff"])
+            entry = (g, st)
+            self._graphs[B] = entry
+        g, st = entry
+        st["toks"].copy_(toks)
+        st["pos"].copy_(pos_ids)
+        st["table"].copy_(table)
+        st["lens"].copy_(lens)
+        st["wblk"].copy_(write_blk)
+        st["woff"].copy_(write_off)
+        g.replay()
+        return st["logits"]
