@@ -238,7 +238,12 @@ class GPTModelRunner:
         self.seq_len = {}            # rid -> tokens currently cached
         self.last_token = {}         # rid -> next input token
         self.max_blk = (max_seq + block_size - 1) // block_size
-        self.use_graphs = use_graphs and dev.type == "cuda"
+        # graphs need the native decode kernel (the python fallback syncs
+        # on seq_lens, which is illegal during stream capture)
+        from . import _ext
+        native_ok = (dev.type == "cuda" and dt == torch.bfloat16 and
+                     self.D in (64, 128) and _ext.has_ext())
+        self.use_graphs = use_graphs and native_ok
         self._graphs = {}            # bucket B -> (graph, static tensors)
 
     def _qkv(self, layer, x):
@@ -282,20 +287,24 @@ class GPTModelRunner:
             return out
         B = len(rest)
         Bp = B if not self.use_graphs else self._bucket(B)
-        table = torch.zeros(Bp, self.max_blk, device=self.dev,
-                            dtype=torch.int32)
+        # build once on host, ship with ONE copy (per-row H2D was per-step
+        # launch overhead)
+        rows = []
         lens_l = []
-        for i, r in enumerate(rest):
+        for r in rest:
             bl = blocks[r.rid]
-            table[i, :len(bl)] = torch.tensor(bl, device=self.dev,
-                                              dtype=torch.int32)
+            rows.append(bl + [0] * (self.max_blk - len(bl)))
             lens_l.append(self.seq_len[r.rid])
-        # padded rows decode a single scratch slot
-        table[B:, 0] = self.scratch_blk
-        lens = torch.tensor(lens_l + [0] * (Bp - B), device=self.dev,
-                            dtype=torch.int32)
+        for _ in range(Bp - B):
+            rows.append([self.scratch_blk] + [0] * (self.max_blk - 1))
+            lens_l.append(0)
+        table = torch.tensor(rows, dtype=torch.int32).to(self.dev,
+                                                         non_blocking=True)
+        lens = torch.tensor(lens_l, dtype=torch.int32).to(self.dev,
+                                                          non_blocking=True)
         toks = torch.tensor([[self.last_token[r.rid]] for r in rest] +
-                            [[0]] * (Bp - B), device=self.dev, dtype=torch.long)
+                            [[0]] * (Bp - B), dtype=torch.long).to(
+                                self.dev, non_blocking=True)
         pos_ids = lens.long().unsqueeze(1)
         write_blk = table.long().gather(1, (pos_ids // self.bs)).squeeze(1)
         write_off = (pos_ids % self.bs).squeeze(1)
