@@ -252,36 +252,47 @@ class GPTModelRunner:
                 .reshape(b, s, 3, self.H, self.D).unbind(2))
 
     @torch.no_grad()
-    def _prefill(self, req, blocks):
+    def _prefill(self, reqs, blocks):
+        """Batched prefill for same-prompt-length requests (one flash-
+        attention pass for the whole group)."""
         m = self.model
-        ids = torch.tensor([req.prompt_ids], device=self.dev, dtype=torch.long)
-        S0 = ids.shape[1]
+        N = len(reqs)
+        S0 = len(reqs[0].prompt_ids)
+        ids = torch.tensor([r.prompt_ids for r in reqs], dtype=torch.long).to(
+            self.dev, non_blocking=True)
         pos = torch.arange(S0, device=self.dev)
-        blk_t = torch.tensor(blocks, device=self.dev, dtype=torch.long)
-        blks = blk_t[pos // self.bs]
-        offs = pos % self.bs
+        blk_t = torch.tensor([blocks[r.rid] for r in reqs],
+                             dtype=torch.long).to(self.dev, non_blocking=True)
+        blks = blk_t[:, pos // self.bs]          # [N, S0]
+        offs = (pos % self.bs).unsqueeze(0).expand(N, S0)
         x = m.gpt.embeddings(ids)
         for li, layer in enumerate(m.gpt.layers):
             h = layer.ln1(x)
             q, k, v = self._qkv(layer, h)
-            self.k[li][blks, offs] = k[0]
-            self.v[li][blks, offs] = v[0]
+            self.k[li][blks, offs] = k
+            self.v[li][blks, offs] = v
             att, _ = self.hot.flash_attention(q, k, v, causal=True)
-            x = x + layer.attn.out_proj(att.reshape(1, S0, -1))
+            x = x + layer.attn.out_proj(att.reshape(N, S0, -1))
             x = x + layer.mlp(layer.ln2(x))
         logits = m.lm_head(m.gpt.final_norm(x[:, -1]))
-        self.seq_len[req.rid] = S0
-        self.last_token[req.rid] = sample_token(logits[0], req.temperature,
-                                                req.top_p)
+        for i, r in enumerate(reqs):
+            self.seq_len[r.rid] = S0
+            self.last_token[r.rid] = sample_token(logits[i], r.temperature,
+                                                  r.top_p)
 
     @torch.no_grad()
     def __call__(self, active, blocks):
         m = self.model
         out = {}
         new = [r for r in active if r.rid not in self.seq_len]
-        for r in new:
-            self._prefill(r, blocks[r.rid])
-            out[r.rid] = self.last_token[r.rid]   # first generated token
+        if new:
+            by_len = {}
+            for r in new:
+                by_len.setdefault(len(r.prompt_ids), []).append(r)
+            for group in by_len.values():
+                self._prefill(group, blocks)
+            for r in new:
+                out[r.rid] = self.last_token[r.rid]  # first generated token
         rest = [r for r in active if r not in new]
         if not rest:
             return out
