@@ -322,8 +322,14 @@ class GPTModelRunner:
         write_blk[B:] = self.scratch_blk
         logits = self._decode(Bp, toks, pos_ids, table, lens, write_blk,
                               write_off)
+        if all(r.temperature <= 0.0 for r in rest):
+            # greedy for the whole batch: ONE argmax + ONE device sync
+            picked = logits[:B].argmax(-1).tolist()
+        else:
+            picked = [sample_token(logits[i], r.temperature, r.top_p)
+                      for i, r in enumerate(rest)]
         for i, r in enumerate(rest):
-            t = sample_token(logits[i], r.temperature, r.top_p)
+            t = int(picked[i])
             self.seq_len[r.rid] += 1
             self.last_token[r.rid] = t
             out[r.rid] = t
