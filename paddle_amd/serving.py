@@ -298,6 +298,32 @@ class GPTModelRunner:
             return out
         B = len(rest)
         Bp = B if not self.use_graphs else self._bucket(B)
+        comp = tuple(r.rid for r in rest)
+        all_greedy = all(r.temperature <= 0.0 for r in rest)
+        # steady-state fast path: same composition as last step, graph
+        # captured, greedy -- advance lens / rotate tokens entirely on
+        # device (zero H2D per step)
+        if (self.use_graphs and all_greedy and
+                comp == getattr(self, "_last_comp", None) and
+                Bp in self._graphs and
+                getattr(self, "_picked_dev", None) is not None):
+            g, st = self._graphs[Bp]
+            st["toks"][:B, 0] = self._picked_dev[:B]
+            st["lens"][:B] += 1
+            st["pos"].copy_(st["lens"].long().unsqueeze(1))
+            st["wblk"].copy_(st["table"].long().gather(
+                1, st["pos"] // self.bs).squeeze(1))
+            st["wblk"][B:] = self.scratch_blk
+            st["woff"].copy_((st["pos"] % self.bs).squeeze(1))
+            g.replay()
+            self._picked_dev = st["logits"][:B].argmax(-1)
+            picked = self._picked_dev.tolist()
+            for i, r in enumerate(rest):
+                t = int(picked[i])
+                self.seq_len[r.rid] += 1
+                self.last_token[r.rid] = t
+                out[r.rid] = t
+            return out
         # build once on host, ship with ONE copy (per-row H2D was per-step
         # launch overhead)
         rows = []
@@ -322,10 +348,13 @@ class GPTModelRunner:
         write_blk[B:] = self.scratch_blk
         logits = self._decode(Bp, toks, pos_ids, table, lens, write_blk,
                               write_off)
-        if all(r.temperature <= 0.0 for r in rest):
+        if all_greedy:
             # greedy for the whole batch: ONE argmax + ONE device sync
-            picked = logits[:B].argmax(-1).tolist()
+            self._picked_dev = logits[:B].argmax(-1)
+            self._last_comp = comp
+            picked = self._picked_dev.tolist()
         else:
+            self._last_comp = None
             picked = [sample_token(logits[i], r.temperature, r.top_p)
                       for i, r in enumerate(rest)]
         for i, r in enumerate(rest):
