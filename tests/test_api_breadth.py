@@ -185,3 +185,20 @@ def test_tensor_method_parity_complete():
         if not ok:
             missing.append(n)
     assert not missing, missing
+
+
+def test_every_module_imports():
+    """Every paddle_amd submodule imports cleanly (catches latent errors in
+    rarely-exercised files)."""
+    import importlib
+    import pkgutil
+    import paddle_amd
+    bad = []
+    for m in pkgutil.walk_packages(paddle_amd.__path__, "paddle_amd."):
+        if m.name.endswith("__main__"):
+            continue
+        try:
+            importlib.import_module(m.name)
+        except Exception as e:  # pragma: no cover
+            bad.append((m.name, repr(e)))
+    assert not bad, bad
