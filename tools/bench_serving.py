@@ -17,29 +17,44 @@ from paddle_amd.models.gpt import GPTConfig, GPTForPretraining  # noqa: E402
 from paddle_amd.serving import Engine, GPTModelRunner, Request  # noqa: E402
 
 
-def main():
-    paddle.seed(0)
-    cfg = GPTConfig(vocab_size=50304, hidden_size=1024, num_layers=24,
-                    num_heads=16, intermediate_size=4096, max_seq_len=2048)
-    m = GPTForPretraining(cfg).to("cuda", torch.bfloat16)
-    runner = GPTModelRunner(m, num_blocks=4096, block_size=16)
-    eng = Engine(runner, num_blocks=4096, block_size=16, max_batch=32)
+def run_one(tag, model, n_req=64, prompt_len=128, gen_len=128, max_batch=32,
+            num_blocks=4096):
+    runner = GPTModelRunner(model, num_blocks=num_blocks, block_size=16)
+    eng = Engine(runner, num_blocks=num_blocks, block_size=16,
+                 max_batch=max_batch)
     import random
     random.seed(0)
-    n_req, prompt_len, gen_len = 64, 128, 128
+    V = model.cfg.vocab_size
     for _ in range(n_req):
         eng.add_request(Request(
-            prompt_ids=[random.randrange(cfg.vocab_size) for _ in range(prompt_len)],
+            prompt_ids=[random.randrange(V) for _ in range(prompt_len)],
             max_new_tokens=gen_len))
     t0 = time.perf_counter()
     eng.run_until_done()
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     s = eng.stats()
-    print(f"serving gpt-350M: {n_req} reqs x (p{prompt_len}+g{gen_len}) in {dt:.1f}s"
+    print(f"serving {tag}: {n_req} reqs x (p{prompt_len}+g{gen_len}) in {dt:.1f}s"
           f"  decode {s['output_tokens'] / dt:.0f} tok/s"
           f"  ttft {s['mean_ttft_s'] * 1e3:.0f} ms"
           f"  peak {torch.cuda.max_memory_allocated() / 2**30:.1f} GB")
+
+
+def main():
+    import argparse
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="gpt-350m",
+                    choices=["gpt-350m", "gpt3-6.7b"])
+    args = ap.parse_args()
+    paddle.seed(0)
+    if args.model == "gpt-350m":
+        cfg = GPTConfig(vocab_size=50304, hidden_size=1024, num_layers=24,
+                        num_heads=16, intermediate_size=4096, max_seq_len=2048)
+        m = GPTForPretraining(cfg).to("cuda", torch.bfloat16)
+        run_one("gpt-350M", m)
+    else:
+        m = build_gpt("gpt3-6.7b", max_seq_len=2048).to("cuda", torch.bfloat16)
+        run_one("gpt3-6.7B", m, n_req=32, max_batch=16, num_blocks=8192)
 
 
 if __name__ == "__main__":
