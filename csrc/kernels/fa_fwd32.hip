@@ -76,18 +76,23 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
   const int q0w = q0 + wq * 32;
 
   // Q fragments (B operand of the swapped QK^T): lane holds
-  // Q[q0w + l32][16*ks + 8*hi .. +7]
+  // Q[q0w + l32][16*ks + 8*hi .. +7].  Pre-scaled by softmax scale ONCE
+  // here -- saves one VALU mul per score element in the kv loop (the
+  // softmax block is what keeps this kernel VALU-issue-bound).
   shortx8 qf[NKS];
   {
     int row = q0w + l32;
     bool ok = row < Sq;
 #pragma unroll
     for (int ks = 0; ks < NKS; ++ks) {
-      if (ok)
-        qf[ks] = *reinterpret_cast<const shortx8*>(
+      if (ok) {
+        shortx8 raw = *reinterpret_cast<const shortx8*>(
             qg + qbase + (long long)row * q_ss + ks * 16 + hi * 8);
-      else
+#pragma unroll
+        for (int i = 0; i < 8; ++i) qf[ks][i] = f2bf(bf2f(raw[i]) * scale);
+      } else {
         for (int i = 0; i < 8; ++i) qf[ks][i] = 0;
+      }
     }
   }
 
@@ -180,7 +185,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           int kv_abs = kv0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          float v = st[t][r] * scale;
+          float v = st[t][r];                 // Q pre-scaled at load
           if (kv_abs >= Skv || (CAUSAL && kv_abs > q_abs)) v = -INFINITY;
           st[t][r] = v;
           mx = fmaxf(mx, v);
@@ -190,9 +195,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
       for (int t = 0; t < 2; ++t)
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          float v = st[t][r] * scale;
-          st[t][r] = v;
-          mx = fmaxf(mx, v);
+          mx = fmaxf(mx, st[t][r]);           // Q pre-scaled at load
         }
     }
     mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
@@ -356,7 +359,8 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
   const int hi = lane >> 5;
   const int q0w = q0 + wq * 32;
 
-  // Q and dO as B-fragments (lane holds the q column l32)
+  // Q and dO as B-fragments (lane holds the q column l32); Q pre-scaled
+  // by the softmax scale so the exp loop skips a per-element mul
   shortx8 qf[NKS], dof[NKS];
   {
     int row = q0w + l32;
@@ -364,8 +368,10 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
 #pragma unroll
     for (int ks = 0; ks < NKS; ++ks) {
       if (ok) {
-        qf[ks] = *reinterpret_cast<const shortx8*>(
+        shortx8 raw = *reinterpret_cast<const shortx8*>(
             qg + qbase + (long long)row * q_ss + ks * 16 + hi * 8);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) qf[ks][i] = f2bf(bf2f(raw[i]) * scale);
         dof[ks] = *reinterpret_cast<const shortx8*>(
             dog + dobase + (long long)row * do_ss + ks * 16 + hi * 8);
       } else {
@@ -465,7 +471,7 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
     for (int t = 0; t < 2; ++t)
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        float p = __expf(st[t][r] * scale - lse_v);
+        float p = __expf(st[t][r] - lse_v);   // Q pre-scaled at load
         if (bnd) {
           int kv_abs = kv0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
           if (kv_abs >= Skv || (CAUSAL && kv_abs > q_abs)) p = 0.f;
