@@ -483,8 +483,13 @@ sparse_attention = _gated("sparse_attention", "use flash_attention (dense) or pa
 adaptive_log_softmax_with_loss = _gated("adaptive_log_softmax_with_loss",
                                         "use nn.AdaptiveLogSoftmaxWithLoss layer")
 flashmask_attention = _gated("flashmask_attention", "mask-sparse FA: round 2")
-flash_attn_varlen_qkvpacked = _gated("flash_attn_varlen_qkvpacked",
-                                     "varlen path: round 2")
+def flash_attn_varlen_qkvpacked(qkv, cu_seqlens, max_seqlen, dropout=0.0,
+                                causal=False, **kwargs):
+    """Packed ragged QKV [total, 3, H, D] -> varlen flash attention."""
+    from ...ops.functional import flash_attn_varlen_func
+    q, k, v = qkv.unbind(1)
+    return flash_attn_varlen_func(q, k, v, cu_seqlens, cu_seqlens,
+                                  max_seqlen, max_seqlen, causal=causal), None
 
 
 def _mk_inplace(fn, extract=None):

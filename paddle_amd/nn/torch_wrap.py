@@ -249,3 +249,86 @@ def dynamic_decode(decoder, inits=None, max_step_num=100, **kwargs):
     raise NotImplementedError(
         "dynamic_decode: use paddle_amd.models.generation.generate for "
         "autoregressive decoding (paged KV cache, greedy/top-p)")
+
+__all__ = [
+    "AdaptiveAvgPool1D",
+    "AdaptiveAvgPool3D",
+    "AdaptiveLogSoftmaxWithLoss",
+    "AdaptiveMaxPool1D",
+    "AdaptiveMaxPool2D",
+    "AdaptiveMaxPool3D",
+    "AlphaDropout",
+    "AvgPool1D",
+    "AvgPool3D",
+    "BatchNorm3D",
+    "BeamSearchDecoder",
+    "BiRNN",
+    "Bilinear",
+    "CELU",
+    "CTCLoss",
+    "ChannelShuffle",
+    "Conv1DTranspose",
+    "Conv3D",
+    "Conv3DTranspose",
+    "CosineEmbeddingLoss",
+    "CosineSimilarity",
+    "Dropout3D",
+    "FeatureAlphaDropout",
+    "Fold",
+    "FractionalMaxPool2D",
+    "FractionalMaxPool3D",
+    "GLU",
+    "GaussianNLLLoss",
+    "HSigmoidLoss",
+    "Hardshrink",
+    "Hardtanh",
+    "HingeEmbeddingLoss",
+    "InstanceNorm1D",
+    "InstanceNorm2D",
+    "InstanceNorm3D",
+    "LPPool1D",
+    "LPPool2D",
+    "LayerDict",
+    "LocalResponseNorm",
+    "LogSigmoid",
+    "MarginRankingLoss",
+    "MaxPool1D",
+    "MaxPool3D",
+    "MaxUnPool1D",
+    "MaxUnPool2D",
+    "MaxUnPool3D",
+    "Maxout",
+    "MultiLabelSoftMarginLoss",
+    "MultiMarginLoss",
+    "PReLU",
+    "Pad1D",
+    "Pad3D",
+    "PairwiseDistance",
+    "ParameterDict",
+    "PixelShuffle",
+    "PixelUnshuffle",
+    "PoissonNLLLoss",
+    "RNN",
+    "RNNCellBase",
+    "RNNTLoss",
+    "RReLU",
+    "SELU",
+    "Silu",
+    "SoftMarginLoss",
+    "Softmax2D",
+    "Softshrink",
+    "Softsign",
+    "SpectralNorm",
+    "Tanhshrink",
+    "ThresholdedReLU",
+    "TripletMarginLoss",
+    "TripletMarginWithDistanceLoss",
+    "Unflatten",
+    "Unfold",
+    "UpsamplingBilinear2D",
+    "UpsamplingNearest2D",
+    "ZeroPad1D",
+    "ZeroPad2D",
+    "ZeroPad3D",
+    "dynamic_decode",
+]

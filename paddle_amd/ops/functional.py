@@ -775,3 +775,32 @@ def fused_linear_param_grad_add(x, dy, dweight=None, dbias=None,
             dbias.add_(dy2.sum(0).to(dbias.dtype))
         return dweight, dbias
     return dweight, None
+
+
+def flash_attn_varlen_func(q, k, v, cu_seqlens_q, cu_seqlens_k, max_seqlen_q,
+                           max_seqlen_k, scale=None, causal=False):
+    """Varlen (ragged) flash attention over packed [total_tokens, H, D]
+    inputs with cu_seqlens boundaries (reference: flash_attn_unpadded,
+    python/paddle/nn/functional/flash_attention.py:195).
+
+    Round-1 scheme: group sequences by (len_q, len_k) and run each group
+    through the rectangular HIP kernel in ONE batched call -- exact
+    numerics, no padding compute; a fused ragged kernel is queued for
+    round 2 (the kernels are already stride-aware).
+    """
+    import collections
+    nq = cu_seqlens_q.tolist()
+    nk = cu_seqlens_k.tolist()
+    out = torch.empty_like(q)
+    groups = collections.defaultdict(list)
+    for i in range(len(nq) - 1):
+        groups[(nq[i + 1] - nq[i], nk[i + 1] - nk[i])].append(i)
+    for (lq, lk), idxs in groups.items():
+        qg = torch.stack([q[nq[i]:nq[i] + lq] for i in idxs]).transpose(1, 2)
+        kg = torch.stack([k[nk[i]:nk[i] + lk] for i in idxs]).transpose(1, 2)
+        vg = torch.stack([v[nk[i]:nk[i] + lk] for i in idxs]).transpose(1, 2)
+        og, _ = flash_attention(qg.transpose(1, 2), kg.transpose(1, 2),
+                                vg.transpose(1, 2), causal=causal, scale=scale)
+        for j, i in enumerate(idxs):
+            out[nq[i]:nq[i] + lq] = og[j]
+    return out

@@ -358,3 +358,27 @@ def test_flash_attention_cross_gpu(sq, skv):
     _assert_close_bf16(q.grad, dq, atol=5e-2, rtol=5e-2)
     _assert_close_bf16(k.grad, dk, atol=5e-2, rtol=5e-2)
     _assert_close_bf16(v.grad, dv, atol=5e-2, rtol=5e-2)
+
+
+def test_flash_attn_varlen_gpu():
+    """Ragged flash attention vs per-sequence reference."""
+    from paddle_amd.ops.functional import flash_attn_varlen_func, _sdpa_ref
+    torch.manual_seed(9)
+    H, D = 4, 128
+    lens = [96, 160, 96]
+    cu = torch.tensor([0, 96, 256, 352], dtype=torch.int32, device=DEV)
+    total = sum(lens)
+    q = torch.randn(total, H, D, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(total, H, D, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(total, H, D, device=DEV, dtype=torch.bfloat16)
+    out = flash_attn_varlen_func(q, k, v, cu, cu, max(lens), max(lens),
+                                 causal=True)
+    off = 0
+    for L in lens:
+        qs = q[off:off + L].transpose(0, 1).unsqueeze(0).float()
+        ks = k[off:off + L].transpose(0, 1).unsqueeze(0).float()
+        vs = v[off:off + L].transpose(0, 1).unsqueeze(0).float()
+        ref, _ = _sdpa_ref(qs, ks, vs, 1.0 / math.sqrt(D), True)
+        got = out[off:off + L].transpose(0, 1).unsqueeze(0).float()
+        torch.testing.assert_close(got, ref, atol=3e-2, rtol=3e-2)
+        off += L
