@@ -368,7 +368,10 @@ class _TorchOptimizer(Optimizer):
     def set_state_dict(self, sd):
         opt = self._ensure()
         if opt and sd:
-            opt.load_state_dict(sd)
+            import copy
+            # torch keeps (not copies) state tensors when dtype/device match;
+            # without the deepcopy two optimizers would share mutable state
+            opt.load_state_dict(copy.deepcopy(sd))
 
 
 class Adagrad(_TorchOptimizer):
