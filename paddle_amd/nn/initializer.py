@@ -128,17 +128,13 @@ class Bilinear(Initializer):
     def __call__(self, tensor):
         import math
         with torch.no_grad():
-            shape = tensor.shape
-            f = math.ceil(shape[-1] / 2)
+            h, w = tensor.shape[-2], tensor.shape[-1]
+            f = math.ceil(w / 2)
             c = (2 * f - 1 - f % 2) / (2.0 * f)
-            for idx in range(tensor.numel()):
-                coords = []
-                rem = idx
-                for s in reversed(shape):
-                    coords.append(rem % s)
-                    rem //= s
-                x, y = coords[0], coords[1] if len(coords) > 1 else 0
-                tensor.view(-1)[idx] = (1 - abs(x / f - c)) * (1 - abs(y / f - c))
+            og_y = torch.arange(h, dtype=torch.float32).reshape(-1, 1)
+            og_x = torch.arange(w, dtype=torch.float32).reshape(1, -1)
+            filt = (1 - (og_x / f - c).abs()) * (1 - (og_y / f - c).abs())
+            tensor.copy_(filt.expand_as(tensor))
         return tensor
 
 

@@ -188,10 +188,12 @@ def bitwise_right_shift(x, y, is_arithmetic=True, out=None, name=None):
 
 def scatter_nd_add(x, index, updates, name=None):
     out = x.clone()
-    flat_idx = index.reshape(-1, index.shape[-1])
-    flat_upd = updates.reshape(flat_idx.shape[0], *updates.shape[index.dim() - 1:])
-    for i in range(flat_idx.shape[0]):
-        out[tuple(flat_idx[i].tolist())] += flat_upd[i]
+    k = index.shape[-1]
+    flat_idx = index.reshape(-1, k).long()
+    flat_upd = updates.reshape(flat_idx.shape[0],
+                               *updates.shape[index.dim() - 1:])
+    out.index_put_(tuple(flat_idx[:, d] for d in range(k)), flat_upd,
+                   accumulate=True)
     return out
 
 
