@@ -408,9 +408,7 @@ in_dynamic_mode = lambda: not static._static_mode
 
 __version__ = "0.1.0"
 
-
-def version():
-    return __version__
+from . import version_mod as version  # noqa: E402  (paddle.version module)
 
 
 def is_grad_enabled_():
