@@ -470,3 +470,108 @@ class InterleavedPipelineParallel(torch.nn.Module):
 
     def state_dict(self, *a, **kw):
         return self._layers.state_dict(*a, **kw)
+
+
+# ---------------------------------------------------------------------------
+# Zero-bubble (ZB-H1) schedule: backward split into B (input-grad, on the
+# critical p2p path) and W (weight-grad, deferred into pipeline bubbles).
+# Reference: passes/pipeline_scheduler_pass/__init__.py:33 (FThenB/ZBH1
+# registry) and the ZB-H1 paper schedule.  The W phase runs through
+# fused_linear_param_grad_add (one accumulating GEMM per linear).
+# ---------------------------------------------------------------------------
+class _DeferredLinear(torch.autograd.Function):
+    """Linear whose backward returns dX immediately and queues (x, dy) for
+    a later dW accumulation -- the B/W split of zero-bubble schedules."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, store):
+        ctx.save_for_backward(x, w)
+        ctx.store = store
+        ctx.bias = b
+        ctx.wref = w
+        out = x @ w
+        if b is not None:
+            out = out + b
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dx = dy @ w.t()
+        ctx.store.append((x, dy, ctx.wref, ctx.bias))   # W work, deferred
+        return dx, None, None, None
+
+
+def _convert_to_zb(module, store):
+    """Swap every nn-style Linear under `module` to the deferred-W path."""
+    n = 0
+    for sub in module.modules():
+        if type(sub).__name__ == "Linear" and hasattr(sub, "weight") \
+                and sub.weight.dim() == 2:
+            def fwd(self, x, _store=store):
+                return _DeferredLinear.apply(x, self.weight, self.bias, _store)
+            sub.forward = fwd.__get__(sub)
+            n += 1
+    return n
+
+
+def _flush_w(store, limit=None):
+    """Run deferred weight-gradient GEMMs (the W phase)."""
+    from ...ops import functional as hot
+    n = 0
+    while store and (limit is None or n < limit):
+        x, dy, w, b = store.pop()
+        x2 = x.reshape(-1, x.shape[-1])
+        dy2 = dy.reshape(-1, dy.shape[-1])
+        if w.grad is None:
+            w.grad = torch.zeros_like(w)
+        if b is not None and b.grad is None:
+            b.grad = torch.zeros_like(b)
+        hot.fused_linear_param_grad_add(x2, dy2, w.grad,
+                                        b.grad if b is not None else None,
+                                        has_bias=b is not None)
+        n += 1
+    return n
+
+
+class _ZBP2P:
+    """p2p delegate that queues a slice of deferred W GEMMs before each
+    blocking receive -- on GPU the W kernels then execute while the host
+    waits on the wire (the ZB bubble fill)."""
+
+    def __init__(self, inner, store, per_recv=4):
+        self._inner = inner
+        self._store = store
+        self._per_recv = per_recv
+
+    def recv_backward(self, like):
+        _flush_w(self._store, limit=self._per_recv)
+        return self._inner.recv_backward(like)
+
+    def send_backward_recv_forward(self, grad_in, device):
+        _flush_w(self._store, limit=self._per_recv)
+        return self._inner.send_backward_recv_forward(grad_in, device)
+
+    def __getattr__(self, name):
+        return getattr(self._inner, name)
+
+
+class ZeroBubblePipelineParallel(PipelineParallel):
+    """ZB-H1: 1F1B order for F and B; each micro-step's W (weight-grad)
+    work is deferred and drained into the schedule's bubbles -- queued
+    ahead of every blocking p2p receive and fully flushed before the
+    tied-weight grad allreduce.  Gradients are bit-identical to 1F1B;
+    the p2p-critical backward path only carries dX."""
+
+    def __init__(self, layers: PipelineLayer, hcg, strategy=None):
+        super().__init__(layers, hcg, strategy)
+        self._w_store = []
+        self._n_zb = _convert_to_zb(layers, self._w_store)
+        self.p2p = _ZBP2P(self.p2p, self._w_store)
+        # W must be complete before tied-weight grads are allreduced
+        inner_ar = self._layers.allreduce_shared_weight_gradients
+
+        def ar_with_flush():
+            _flush_w(self._w_store)
+            return inner_ar()
+        self._layers.allreduce_shared_weight_gradients = ar_with_flush
