@@ -73,3 +73,19 @@ def spawn(func, args=(), nprocs=-1, join=True, daemon=False, **options):
 
     return mp.start_processes(_entry, args=(nprocs, func, args), nprocs=nprocs,
                               join=join, daemon=daemon, start_method="spawn")
+
+# remaining comm API names
+def wait(tensor, group=None, use_calc_stream=True):
+    """Block until async work on `tensor` is visible (stream-sync model)."""
+    import torch
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    return tensor
+
+
+def gather(tensor, gather_list=None, dst=0, group=None, sync_op=True):
+    from . import collective as C
+    return C.gather(tensor, gather_list, dst=dst, group=group)
+
+
+from . import launch  # noqa: E402,F401

@@ -331,3 +331,16 @@ class stream:
     @staticmethod
     def recv(tensor, src, group=None, sync_op=True, use_calc_stream=False):
         return recv(tensor, src, group, sync_op)
+
+
+def gather(tensor, gather_list=None, dst=0, group=None, sync_op=True):
+    """Gather tensors to dst (reference: communication/gather.py).
+    gloo/rccl portable: implemented over all_gather (rccl has no native
+    gather; the extra traffic is one ring pass)."""
+    world = get_world_size(group)
+    parts = [torch.empty_like(tensor) for _ in range(world)]
+    all_gather(parts, tensor, group=group)
+    if get_rank(group) == (dst if group is None else dst) and gather_list is not None:
+        for g, p in zip(gather_list, parts):
+            g.copy_(p)
+    return gather_list

@@ -189,6 +189,7 @@ util = UtilBase()
 
 # meta_parallel namespace parity
 from . import mpu  # noqa: E402,F401
+from . import utils  # noqa: F401
 from .pipeline import PipelineLayer, PipelineParallel, LayerDesc, SharedLayerDesc  # noqa: E402,F401
 from .recompute import recompute  # noqa: E402,F401
 
