@@ -111,6 +111,7 @@ class Model:
         for cb in cbs:
             cb.model = self
             cb.on_train_begin()
+        self.stop_training = False
         self.network.train()
         it = 0
         for epoch in range(epochs):
@@ -128,9 +129,14 @@ class Model:
             for cb in cbs:
                 cb.on_epoch_end(epoch, logs)
             if eval_data is not None and (epoch + 1) % eval_freq == 0:
-                self.evaluate(eval_data, batch_size=batch_size, verbose=0)
+                eval_logs = self.evaluate(eval_data, batch_size=batch_size,
+                                          verbose=0)
+                for cb in cbs:
+                    cb.on_eval_end(eval_logs)
             if save_dir and (epoch + 1) % save_freq == 0:
                 self.save(f"{save_dir}/{epoch}")
+            if getattr(self, "stop_training", False):
+                break                        # EarlyStopping & friends
         for cb in cbs:
             cb.on_train_end()
 

@@ -200,3 +200,28 @@ def test_visualdl_writes_jsonl(tmp_path):
     v.on_eval_end({"acc": [0.9]})
     lines = [json.loads(l) for l in open(tmp_path / "scalars.jsonl")]
     assert lines[0]["loss"] == 0.5 and lines[1]["acc"] == 0.9
+
+
+def test_model_fit_with_early_stopping():
+    import torch
+    import paddle_amd as paddle
+    from paddle_amd.callbacks import EarlyStopping
+
+    class DS(paddle.io.Dataset):
+        def __getitem__(self, i):
+            torch.manual_seed(i)
+            x = torch.randn(4)
+            return x, x.sum(keepdim=True)
+
+        def __len__(self):
+            return 16
+
+    net = paddle.nn.Linear(4, 1)
+    model = paddle.Model(net)
+    opt = paddle.optimizer.SGD(learning_rate=0.0, parameters=net.parameters())
+    model.prepare(opt, paddle.nn.MSELoss())
+    es = EarlyStopping(monitor="loss", patience=1, mode="min")
+    model.fit(DS(), eval_data=DS(), epochs=10, batch_size=4, verbose=0,
+              callbacks=[es])
+    # lr=0 -> eval loss constant -> early stop long before 10 epochs
+    assert model.stop_training
