@@ -211,7 +211,7 @@ def test_model_fit_with_early_stopping():
         def __getitem__(self, i):
             torch.manual_seed(i)
             x = torch.randn(4)
-            return x, x.sum(keepdim=True)
+            return x, x.sum().reshape(1)
 
         def __len__(self):
             return 16
