@@ -1,0 +1,16 @@
+"""paddle.nn.quant (reference: python/paddle/nn/quant/__init__.py --
+weight_only linear + quant stubs)."""
+from ...quantization import (  # noqa: F401
+    weight_only_linear,
+    weight_quantize,
+)
+
+
+def weight_dequantize(qweight, scale, algo="weight_only_int8", out_dtype=None):
+    import torch
+    w = (qweight.float() * scale.unsqueeze(1) / 127.0).t()
+    return w.to(out_dtype or torch.float16)
+
+
+def llm_int8_linear(x, qweight, scale, bias=None, threshold=6.0):
+    return weight_only_linear(x, qweight, scale, bias)
