@@ -141,34 +141,20 @@ class _Unit:
         flat = self.grad_flat
         for p in self.params:
             p.grad = None
+        self.accum_steps += 1
+        if accumulate:
+            # DDP-style no_sync: micro-step grads keep accumulating into the
+            # flat buffer locally (bind_grad_views re-binds the same views on
+            # the next backward, autograd adds in place); the collective runs
+            # only on the final, synchronizing micro-step
+            return
         if self.world == 1:
-            if accumulate:
-                if self.accum_steps > 0:
-                    self.grad_shard_fp32.add_(flat[0:self.shard_size])
-                else:
-                    self.grad_shard_fp32.copy_(flat[0:self.shard_size])
-                flat.zero_()
-                self.grad_ready, self.grad_premul = self.grad_shard_fp32, 1.0
-            else:
-                # fast path: the optimizer consumes the bf16 flat buffer
-                # directly (fused grad-scale in the AdamW kernel); zeroing
-                # is deferred to clear_grad() after step() has read it.
-                self.grad_ready = flat[0:self.shard_size]
-                self.grad_premul = 1.0
-                self.defer_flat_zero = True
-        elif accumulate:
-            # grad-accumulation path: keep it synchronous (flat is re-bound
-            # next micro-step, must not race the collective)
-            flat.div_(self.world)
-            out = torch.empty(self.shard_size, dtype=self.dtype,
-                              device=self.shard.device)
-            C.reduce_scatter_tensor(out, flat, group=self.group)
-            if self.accum_steps > 0:
-                self.grad_shard_fp32.add_(out)
-            else:
-                self.grad_shard_fp32.copy_(out)
-            flat.zero_()
-            self.grad_ready, self.grad_premul = self.grad_shard_fp32, 1.0
+            # fast path: the optimizer consumes the bf16 flat buffer
+            # directly (fused grad-scale in the AdamW kernel); zeroing
+            # is deferred to clear_grad() after step() has read it.
+            self.grad_ready = flat[0:self.shard_size]
+            self.grad_premul = 1.0
+            self.defer_flat_zero = True
         else:
             # async reduce-scatter overlapped with the remaining backward;
             # the optimizer's step() calls finish_grad_reduce() first.
@@ -180,7 +166,6 @@ class _Unit:
             work = C.reduce_scatter_tensor(out, flat, group=self.group,
                                            sync_op=False)
             self.grad_work = (work, out)
-        self.accum_steps += 1
 
     def finish_grad_reduce(self):
         if self.grad_work is not None:

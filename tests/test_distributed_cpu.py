@@ -279,3 +279,48 @@ def test_sharding_stage1_stage2_match_single():
                 torch.testing.assert_close(a, b, atol=2e-5, rtol=2e-5), level
         print("rank", r, "stage1/2 ok")
     """, world_size=2)
+
+
+def test_sharding3_grad_accumulation_matches_single():
+    """no_sync() micro-batch accumulation on 2 ranks equals single-process
+    accumulation over the same 4 micro-batches (exercises the fp32
+    grad-accumulate branch of reduce_grads)."""
+    run_dist("""
+        import torch
+        import paddle_amd as paddle
+        paddle.distributed.init_parallel_env()
+        from paddle_amd.distributed.fleet.sharding import (GroupShardedStage3,
+                                                           ShardedAdamW)
+        r = paddle.distributed.get_rank()
+        torch.manual_seed(0)
+        m = paddle.nn.Sequential(paddle.nn.Linear(8, 16), paddle.nn.Linear(16, 8))
+        wrapped = GroupShardedStage3(m, device=torch.device("cpu"))
+        opt = ShardedAdamW(wrapped, learning_rate=1e-2)
+        torch.manual_seed(17)
+        xs = [torch.randn(2, 8) for _ in range(4)]
+        mine = xs[r * 2:(r + 1) * 2]          # 2 micro-batches per rank
+        with wrapped.no_sync():
+            loss = wrapped(mine[0]).pow(2).mean()
+            loss.backward()
+        # final micro-batch syncs (accumulate branch then reduce)
+        loss = wrapped(mine[1]).pow(2).mean()
+        loss.backward()
+        opt.step()
+        opt.clear_grad()
+
+        # single-process reference: mean over ranks of mean over micro-steps
+        torch.manual_seed(0)
+        m2 = paddle.nn.Sequential(paddle.nn.Linear(8, 16), paddle.nn.Linear(16, 8))
+        w2 = GroupShardedStage3(m2, device=torch.device("cpu"), group=None)
+        o2 = ShardedAdamW(w2, learning_rate=1e-2)
+        # emulate: rank grads = sum of its 2 micro losses' grads; then mean
+        # over ranks -> equivalent single-process grad = mean over ranks of
+        # (sum over micro); build it by scaling
+        for i, x in enumerate(xs):
+            loss = w2(x).pow(2).mean() / 2.0   # 2 ranks averaged
+            loss.backward()
+        o2.step()
+        for a, b in zip(wrapped._units, w2._units):
+            torch.testing.assert_close(a.master, b.master, atol=2e-5, rtol=2e-5)
+        print("rank", r, "accum ok")
+    """, world_size=2)
