@@ -42,3 +42,5 @@ def run_check():
         from paddle_amd.ops import functional as hot
         hot.l2_norm_squared(y)
     print("paddle_amd is installed successfully!")
+
+from . import cpp_extension  # noqa: E402,F401
