@@ -44,7 +44,7 @@ class _Unit:
     flat buffer."""
 
     def __init__(self, idx, module, params, world, rank, device, dtype,
-                 group=None, sync_init=True):
+                 group=None, sync_init=True, use_main_grad=False):
         self.idx = idx
         self.module = module
         self.params = params
@@ -80,6 +80,10 @@ class _Unit:
         self.grad_ready = None
         self.grad_premul = 1.0
         self.defer_flat_zero = False
+        # fp32 main-grad accumulation (reference: main_grad hooks in
+        # mixed_precision_utils) -- full-unit fp32 buffer, lazily allocated
+        self.use_main_grad = use_main_grad
+        self.main_grad = None
         del flat
         self.full: Optional[torch.Tensor] = None
         self.gather_work = None
@@ -142,6 +146,27 @@ class _Unit:
         for p in self.params:
             p.grad = None
         self.accum_steps += 1
+        if self.use_main_grad:
+            # fold this micro-step's bf16 grads into the fp32 accumulator
+            # (one add pass; bf16 rounding only within a single micro-step)
+            if self.main_grad is None:
+                self.main_grad = torch.zeros(self.total, dtype=torch.float32,
+                                             device=self.shard.device)
+            self.main_grad.add_(flat)
+            flat.zero_()
+            if accumulate:
+                return
+            if self.world == 1:
+                self.grad_ready = self.main_grad[0:self.shard_size]
+                self.grad_premul = 1.0
+                self.defer_flat_zero = False
+            else:
+                out = torch.empty(self.shard_size, dtype=torch.float32,
+                                  device=self.shard.device)
+                work = C.reduce_scatter_tensor(out, self.main_grad,
+                                               group=self.group, sync_op=False)
+                self.grad_work = (work, out)
+            return
         if accumulate:
             # DDP-style no_sync: micro-step grads keep accumulating into the
             # flat buffer locally (bind_grad_views re-binds the same views on
@@ -184,8 +209,9 @@ class GroupShardedStage3(torch.nn.Module):
     def __init__(self, layer, optimizer=None, group=None, sync_buffers=False,
                  device=None, segment_size=2 ** 20, pertrain_sync_models=True,
                  offload=False, sync_comm=False, dp_group=None,
-                 exclude_layer=None, param_dtype=None):
+                 exclude_layer=None, param_dtype=None, use_main_grad=False):
         super().__init__()
+        self._use_main_grad = use_main_grad
         self._layer = layer
         self.group = group
         self.world = get_world_size(group)
@@ -248,7 +274,8 @@ class GroupShardedStage3(torch.nn.Module):
                 p._placeholder = torch.empty(0, dtype=p.dtype, device=self.device)
             u = _Unit(len(self._units), mod, params, self.world, self.rank,
                       self.device, dtype, group=self.group,
-                      sync_init=self._sync_init)
+                      sync_init=self._sync_init,
+                      use_main_grad=self._use_main_grad)
             u.owner = self
             self._units.append(u)
             for p in params:
@@ -466,8 +493,8 @@ class ShardedAdamW:
             if u.defer_flat_zero and u.grad_flat is not None:
                 u.grad_flat.zero_()
                 u.defer_flat_zero = False
-            # grad_shard_fp32 needs no zeroing: the first micro-step of the
-            # next accumulation window copy_()s rather than add_()s.
+            if u.use_main_grad and u.main_grad is not None:
+                u.main_grad.zero_()
             u.grad_ready = None
             u.accum_steps = 0
 

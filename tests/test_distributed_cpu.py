@@ -324,3 +324,40 @@ def test_sharding3_grad_accumulation_matches_single():
             torch.testing.assert_close(a.master, b.master, atol=2e-5, rtol=2e-5)
         print("rank", r, "accum ok")
     """, world_size=2)
+
+
+def test_sharding3_main_grad_fp32_accumulation():
+    """use_main_grad=True accumulates micro-grads in fp32 and matches the
+    single-process result (reference: mixed_precision_utils main_grad)."""
+    run_dist("""
+        import torch
+        import paddle_amd as paddle
+        paddle.distributed.init_parallel_env()
+        from paddle_amd.distributed.fleet.sharding import (GroupShardedStage3,
+                                                           ShardedAdamW)
+        r = paddle.distributed.get_rank()
+        torch.manual_seed(0)
+        m = paddle.nn.Sequential(paddle.nn.Linear(8, 16), paddle.nn.Linear(16, 8))
+        wrapped = GroupShardedStage3(m, device=torch.device("cpu"),
+                                     use_main_grad=True)
+        opt = ShardedAdamW(wrapped, learning_rate=1e-2)
+        torch.manual_seed(23)
+        xs = [torch.randn(2, 8) for _ in range(4)]
+        mine = xs[r * 2:(r + 1) * 2]
+        with wrapped.no_sync():
+            wrapped(mine[0]).pow(2).mean().backward()
+        wrapped(mine[1]).pow(2).mean().backward()
+        opt.step()
+        opt.clear_grad()
+
+        torch.manual_seed(0)
+        m2 = paddle.nn.Sequential(paddle.nn.Linear(8, 16), paddle.nn.Linear(16, 8))
+        w2 = GroupShardedStage3(m2, device=torch.device("cpu"))
+        o2 = ShardedAdamW(w2, learning_rate=1e-2)
+        for x in xs:
+            (w2(x).pow(2).mean() / 2.0).backward()
+        o2.step()
+        for a, b in zip(wrapped._units, w2._units):
+            torch.testing.assert_close(a.master, b.master, atol=2e-5, rtol=2e-5)
+        print("rank", r, "main_grad ok")
+    """, world_size=2)
