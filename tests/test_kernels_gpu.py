@@ -382,3 +382,33 @@ def test_flash_attn_varlen_gpu():
         got = out[off:off + L].transpose(0, 1).unsqueeze(0).float()
         torch.testing.assert_close(got, ref, atol=3e-2, rtol=3e-2)
         off += L
+
+
+def test_flash_attention_gqa_bwd_gpu():
+    """GQA (H > HKV) backward: expanded-KV kernel + head-group reduction of
+    dK/dV vs fp32 reference."""
+    torch.manual_seed(21)
+    b, hq, hkv, s, d = 2, 8, 2, 192, 128
+    q = torch.randn(b, hq, s, d, device=DEV, dtype=torch.bfloat16).requires_grad_(True)
+    k = torch.randn(b, hkv, s, d, device=DEV, dtype=torch.bfloat16).requires_grad_(True)
+    v = torch.randn(b, hkv, s, d, device=DEV, dtype=torch.bfloat16).requires_grad_(True)
+    scale = 1.0 / math.sqrt(d)
+    from paddle_amd.ops.functional import _FlashAttn
+    o = _FlashAttn.apply(q, k, v, scale, True)
+    g = torch.randn_like(o)
+    o.backward(g)
+    rep = hq // hkv
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    ke = kf.repeat_interleave(rep, dim=1)
+    ve = vf.repeat_interleave(rep, dim=1)
+    sc = (qf @ ke.transpose(-1, -2)) * scale
+    mask = torch.triu(torch.ones(s, s, device=DEV, dtype=torch.bool), 1)
+    sc = sc.masked_fill(mask, float("-inf"))
+    ref = torch.softmax(sc, -1) @ ve
+    ref.backward(g.float())
+    _assert_close_bf16(o, ref, atol=3e-2, rtol=3e-2)
+    _assert_close_bf16(q.grad, qf.grad, atol=5e-2, rtol=5e-2)
+    _assert_close_bf16(k.grad, kf.grad, atol=8e-2, rtol=8e-2)
+    _assert_close_bf16(v.grad, vf.grad, atol=8e-2, rtol=8e-2)
