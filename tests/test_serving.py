@@ -98,3 +98,20 @@ def test_gpt_runner_end_to_end_cpu():
     eng2.add_request(r2)
     eng2.run_until_done()
     assert r2.out_ids == reqs[0].out_ids
+
+
+def test_engine_fifo_admission():
+    """Requests admit in arrival order when blocks free up."""
+    order = []
+
+    def step(active, blocks):
+        for r in active:
+            if r.rid not in order:
+                order.append(r.rid)
+        return {r.rid: 1 for r in active}
+
+    eng = Engine(step, num_blocks=3, block_size=16, max_batch=8)
+    rids = [eng.add_request(Request(prompt_ids=[1] * 8, max_new_tokens=2))
+            for _ in range(3)]          # each needs 1 block; pool of 3
+    eng.run_until_done()
+    assert order == rids
