@@ -85,3 +85,78 @@ def test_swiglu_optest():
     t = TestSwigluOp()
     t.check_output()
     t.check_grad()
+
+
+class TestBiasGeluOp(OpTest):
+    rtol, atol = 1e-4, 1e-5
+
+    def make_inputs(self):
+        g = torch.Generator().manual_seed(6)
+        return [torch.randn(4, 16, dtype=torch.float64, generator=g),
+                torch.randn(16, dtype=torch.float64, generator=g)]
+
+    @staticmethod
+    def fn(x, b):
+        return hot.bias_gelu(x.float(), b.float()).double()
+
+    @staticmethod
+    def oracle(x, b):
+        from scipy.special import erf
+        v = x.numpy() + b.numpy()
+        return torch.from_numpy(0.5 * v * (1 + erf(v / np.sqrt(2.0))))
+
+
+class TestSoftmaxCrossEntropyOp(OpTest):
+    rtol, atol = 1e-4, 1e-5
+    grad_inputs = (0,)          # labels are integer; only logits get grads
+
+    def make_inputs(self):
+        g = torch.Generator().manual_seed(7)
+        self.labels = torch.randint(0, 8, (5,), generator=g)
+        return [torch.randn(5, 8, dtype=torch.float64, generator=g)]
+
+    def fn(self, logits):
+        return hot.softmax_cross_entropy(logits.float(), self.labels).double()
+
+    def oracle(self, logits):
+        ln = logits.numpy()
+        lse = np.log(np.exp(ln - ln.max(-1, keepdims=True)).sum(-1)) + \
+            ln.max(-1)
+        picked = ln[np.arange(5), self.labels.numpy()]
+        return torch.from_numpy(lse - picked)
+
+
+class TestMaxoutOp(OpTest):
+    rtol, atol = 1e-5, 1e-6
+
+    def make_inputs(self):
+        g = torch.Generator().manual_seed(8)
+        return [torch.randn(2, 8, 3, dtype=torch.float64, generator=g)]
+
+    @staticmethod
+    def fn(x):
+        import paddle_amd as paddle
+        return paddle.nn.functional.maxout(x.float(), groups=2).double()
+
+    @staticmethod
+    def oracle(x):
+        xn = x.numpy().reshape(2, 4, 2, 3)
+        return torch.from_numpy(xn.max(axis=2))
+
+
+def test_bias_gelu_optest():
+    t = TestBiasGeluOp()
+    t.check_output()
+    t.check_grad()
+
+
+def test_softmax_ce_optest():
+    t = TestSoftmaxCrossEntropyOp()
+    t.check_output()
+    t.check_grad()
+
+
+def test_maxout_optest():
+    t = TestMaxoutOp()
+    t.check_output()
+    t.check_grad()
