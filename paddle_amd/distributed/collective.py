@@ -131,12 +131,16 @@ class _Task:
 
 
 def all_reduce(tensor, op=ReduceOp.SUM, group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("all_reduce", tensor)
     g = _get_group(group)
     work = dist.all_reduce(tensor, op=op, group=g.pg, async_op=not sync_op)
     return _Task(work) if not sync_op else None
 
 
 def all_gather(tensor_list, tensor, group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("all_gather", tensor_list)
     g = _get_group(group)
     if isinstance(tensor_list, list) and len(tensor_list) == 0:
         tensor_list.extend(torch.empty_like(tensor) for _ in range(g.nranks))
@@ -146,6 +150,8 @@ def all_gather(tensor_list, tensor, group=None, sync_op=True):
 
 
 def all_gather_into_tensor(out, tensor, group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("all_gather_into_tensor", out)
     g = _get_group(group)
     work = dist.all_gather_into_tensor(out, tensor.contiguous(), group=g.pg,
                                        async_op=not sync_op)
@@ -160,6 +166,8 @@ def all_gather_object(object_list, obj, group=None):
 
 
 def reduce_scatter(tensor, tensor_list, op=ReduceOp.SUM, group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("reduce_scatter", tensor)
     g = _get_group(group)
     if g.pg is not None and dist.get_backend(g.pg) == "gloo":
         # gloo has no reduce_scatter: emulate (CPU test path only)
@@ -173,6 +181,8 @@ def reduce_scatter(tensor, tensor_list, op=ReduceOp.SUM, group=None, sync_op=Tru
 
 
 def reduce_scatter_tensor(out, tensor, op=ReduceOp.SUM, group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("reduce_scatter_tensor", out)
     g = _get_group(group)
     if g.pg is not None and dist.get_backend(g.pg) == "gloo":
         t = tensor.clone()
@@ -186,6 +196,8 @@ def reduce_scatter_tensor(out, tensor, op=ReduceOp.SUM, group=None, sync_op=True
 
 
 def broadcast(tensor, src, group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("broadcast", tensor)
     g = _get_group(group)
     work = dist.broadcast(tensor, src=src, group=g.pg, async_op=not sync_op)
     return _Task(work) if not sync_op else None
@@ -197,12 +209,16 @@ def broadcast_object_list(object_list, src, group=None):
 
 
 def reduce(tensor, dst, op=ReduceOp.SUM, group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("reduce", tensor)
     g = _get_group(group)
     work = dist.reduce(tensor, dst=dst, op=op, group=g.pg, async_op=not sync_op)
     return _Task(work) if not sync_op else None
 
 
 def scatter(tensor, tensor_list=None, src=0, group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("scatter", tensor)
     g = _get_group(group)
     work = dist.scatter(tensor, scatter_list=tensor_list, src=src, group=g.pg,
                         async_op=not sync_op)
@@ -214,6 +230,8 @@ def _is_gloo(g: Group) -> bool:
 
 
 def alltoall(in_tensor_list, out_tensor_list, group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("alltoall", in_tensor_list)
     g = _get_group(group)
     if isinstance(out_tensor_list, list) and len(out_tensor_list) == 0:
         out_tensor_list.extend(torch.empty_like(t) for t in in_tensor_list)
@@ -233,6 +251,8 @@ def alltoall(in_tensor_list, out_tensor_list, group=None, sync_op=True):
 
 def alltoall_single(in_tensor, out_tensor, in_split_sizes=None, out_split_sizes=None,
                     group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("alltoall_single", in_tensor)
     g = _get_group(group)
     if _is_gloo(g):
         w = g.nranks
@@ -251,6 +271,8 @@ def alltoall_single(in_tensor, out_tensor, in_split_sizes=None, out_split_sizes=
 
 
 def send(tensor, dst=0, group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("send", tensor)
     g = _get_group(group)
     if sync_op:
         dist.send(tensor, dst=dst, group=g.pg)
@@ -259,6 +281,8 @@ def send(tensor, dst=0, group=None, sync_op=True):
 
 
 def recv(tensor, src=0, group=None, sync_op=True):
+    from . import comm_check as _cc
+    _cc.record("recv", tensor)
     g = _get_group(group)
     if sync_op:
         dist.recv(tensor, src=src, group=g.pg)

@@ -225,3 +225,33 @@ def test_model_fit_with_early_stopping():
               callbacks=[es])
     # lr=0 -> eval loss constant -> early stop long before 10 epochs
     assert model.stop_training
+
+
+def test_comm_desync_checker():
+    """Desync checker catches ranks issuing different collectives."""
+    import sys
+    sys.path.insert(0, "tests")
+    from dist_util import run_dist
+    run_dist("""
+        import torch
+        import paddle_amd as paddle
+        from paddle_amd.distributed import comm_check
+        paddle.distributed.init_parallel_env()
+        r = paddle.distributed.get_rank()
+        comm_check.enable()
+        t = torch.ones(4)
+        paddle.distributed.all_reduce(t)
+        assert comm_check.verify()           # in sync
+        # now diverge: rank 0 broadcasts a different shape
+        if r == 0:
+            paddle.distributed.broadcast(torch.ones(8), src=0)
+        else:
+            paddle.distributed.broadcast(torch.ones(4, 2), src=0)
+        try:
+            comm_check.verify()
+            raise SystemExit("desync not detected")
+        except RuntimeError as e:
+            assert "desync" in str(e)
+        comm_check.disable()
+        print("rank", r, "desync checker ok")
+    """, world_size=2)
