@@ -1,5 +1,9 @@
 // FlashAttention-2 forward, 32x32-MFMA variant (gfx950).
 //
+// Reference role: paddle/phi/kernels/gpu/flash_attn_kernel.cu:41 (thin
+// dispatch into dynloaded libflashattn) -- here the FA2 algorithm is
+// implemented natively for CDNA4 instead of vendoring a library.
+//
 // Design (guide §B "fused attention prefill" 8-warp ladder, re-derived for
 // our wave-per-32-q-rows layout):
 //   * mfma_f32_32x32x16_bf16; each wave owns 32 q rows; 4 waves/block

@@ -1,6 +1,10 @@
 // Hand-written bf16 MFMA GEMM for gfx950 (C = A x B, fp32 accumulate,
 // bf16 out).
 //
+// Reference role: paddle/phi/kernels/funcs/blas/blaslt_impl.cu.h +
+// matmul_kernel_impl.h:108 (the library-GEMM path) -- here re-derived as
+// a native CDNA4 kernel; hipBLASLt remains the dispatch where it wins.
+//
 // Geometry (guide §5 "256² 8-phase template" -- re-derived, simplified
 // schedule):
 //   * block tile 256x256, K-step 64, 8 waves (2M x 4N), 512 threads

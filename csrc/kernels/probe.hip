@@ -1,6 +1,8 @@
 // MFMA layout probe: single-wave 16x16x32 bf16 tile, used by the GPU test
 // suite to verify the fragment layout assumptions in flash_attn.hip /
 // gemm kernels against torch.matmul (asymmetric-input check, guide G9).
+// No reference counterpart: this exists because CDNA4 fragment layouts
+// are hardware facts the kernels above depend on (SURVEY §2.2 notes).
 #include "common.h"
 #include "api.h"
 
