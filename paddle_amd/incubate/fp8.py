@@ -155,3 +155,31 @@ def convert_experts_to_fp8(moe_model):
 
             mod.forward = fwd.__get__(mod)
     return moe_model
+
+
+def convert_linears_to_fp8(model, min_features=256, skip=("lm_head",)):
+    """Swap paddle Linear forwards to fp8 GEMMs in place (full-model fp8
+    training recipe; small layers and skipped names stay bf16 --
+    reference: the fp8 quant variants of fused_bias_act / cublaslt fp8
+    path in paddle/phi/kernels/fusion/).  Returns the converted count."""
+    n = 0
+    for name, mod in model.named_modules():
+        if type(mod).__name__ != "Linear" or not hasattr(mod, "weight"):
+            continue
+        if mod.weight.dim() != 2:
+            continue
+        if any(s in name for s in skip):
+            continue
+        in_f, out_f = mod.weight.shape
+        if min(in_f, out_f) < min_features:
+            continue
+
+        def fwd(self, x):
+            out = fp8_matmul(x, self.weight)
+            if getattr(self, "bias", None) is not None:
+                out = out + self.bias
+            return out
+
+        mod.forward = fwd.__get__(mod)
+        n += 1
+    return n

@@ -53,3 +53,26 @@ def test_weight_only_quant_roundtrip_cpu():
     out = Q.weight_only_linear(x, qw, sc)
     rel = (out - x @ w).abs().max() / (x @ w).abs().max()
     assert rel < 0.02, float(rel)
+
+
+def test_convert_linears_to_fp8_trains():
+    import torch
+    import paddle_amd as paddle
+    from paddle_amd.incubate.fp8 import convert_linears_to_fp8
+    from paddle_amd.models import GPTPretrainingCriterion, build_gpt
+    paddle.seed(0)
+    m = build_gpt("gpt3-tiny", max_seq_len=64)
+    n = convert_linears_to_fp8(m, min_features=64)
+    assert n > 0
+    opt = paddle.optimizer.AdamW(learning_rate=1e-3, parameters=m.parameters())
+    crit = GPTPretrainingCriterion()
+    ids = torch.randint(0, 1024, (2, 32))
+    first = last = None
+    for _ in range(5):
+        loss = crit(m(ids), ids)
+        loss.backward()
+        opt.step()
+        opt.clear_grad()
+        first = float(loss) if first is None else first
+        last = float(loss)
+    assert last < first
