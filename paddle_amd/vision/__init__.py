@@ -28,3 +28,5 @@ def image_load(path, backend=None):
 
 
 from . import datasets  # noqa: F401
+
+from . import ops  # noqa: F401

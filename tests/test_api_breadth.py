@@ -114,6 +114,7 @@ def test_submodule_export_parity_complete():
                      ("vision.transforms", "vision/transforms"),
                      ("vision.models", "vision/models"),
                      ("vision.datasets", "vision/datasets"),
+                     ("vision.ops", "vision/ops"),
                      ("nn.initializer", "nn/initializer"),
                      ("incubate.nn", "incubate/nn"),
                      ("callbacks", "callbacks"),
@@ -202,3 +203,23 @@ def test_every_module_imports():
         except Exception as e:  # pragma: no cover
             bad.append((m.name, repr(e)))
     assert not bad, bad
+
+
+def test_vision_ops_behavior():
+    import torch
+    import paddle_amd.vision.ops as O
+    boxes = torch.tensor([[0, 0, 10, 10], [1, 1, 11, 11], [50, 50, 60, 60.]])
+    keep = O.nms(boxes, 0.5, torch.tensor([0.9, 0.8, 0.7]))
+    assert keep.tolist() == [0, 2]
+    x = torch.arange(64.).reshape(1, 1, 8, 8)
+    assert O.roi_pool(x, torch.tensor([[0, 0, 4, 4.]]),
+                      torch.tensor([1]), 2).shape == (1, 1, 2, 2)
+    assert O.roi_align(x, torch.tensor([[0, 0, 4, 4.]]),
+                       torch.tensor([1]), 2).shape == (1, 1, 2, 2)
+    pb = torch.tensor([[0, 0, 10, 10.]])
+    dec = O.box_coder(pb, [1., 1, 1, 1], torch.zeros(1, 4),
+                      code_type="decode_center_size")
+    assert torch.allclose(dec, pb, atol=1e-5)
+    priors, var = O.prior_box(torch.zeros(1, 8, 4, 4), torch.zeros(1, 3, 32, 32),
+                              min_sizes=[8.0])
+    assert priors.shape[-1] == 4 and var.shape == priors.shape
