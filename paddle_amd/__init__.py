@@ -265,6 +265,7 @@ from . import profiler  # noqa: F401
 from . import linalg  # noqa: F401
 from . import fft  # noqa: F401
 from . import callbacks  # noqa: F401
+from . import dataset  # noqa: F401
 from . import hub  # noqa: F401
 from . import utils  # noqa: F401
 from . import signal  # noqa: F401

@@ -193,4 +193,4 @@ from . import utils  # noqa: F401
 from .pipeline import PipelineLayer, PipelineParallel, LayerDesc, SharedLayerDesc  # noqa: E402,F401
 from .recompute import recompute  # noqa: E402,F401
 
-meta_parallel = sharding_mod
+from . import meta_parallel  # noqa: E402,F401
