@@ -115,6 +115,9 @@ def test_submodule_export_parity_complete():
                      ("vision.models", "vision/models"),
                      ("vision.datasets", "vision/datasets"),
                      ("vision.ops", "vision/ops"),
+                     ("audio.functional", "audio/functional"),
+                     ("audio.features", "audio/features"),
+                     ("audio.backends", "audio/backends"),
                      ("nn.initializer", "nn/initializer"),
                      ("incubate.nn", "incubate/nn"),
                      ("callbacks", "callbacks"),
