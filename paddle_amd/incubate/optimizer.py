@@ -83,3 +83,6 @@ class ModelAverage:
 
     def minimize(self, loss, **kw):
         raise NotImplementedError("ModelAverage wraps evaluation, not training")
+
+
+from ..optimizer import LBFGS  # noqa: E402,F401

@@ -1,6 +1,7 @@
 """paddle.nn parity surface (reference: python/paddle/nn/__init__.py)."""
 from . import functional  # noqa: F401
 from . import initializer  # noqa: F401
+from . import utils  # noqa: F401
 from .layer import Layer, LayerList, ParameterList, Sequential  # noqa: F401
 from .common import (  # noqa: F401
     Dropout,

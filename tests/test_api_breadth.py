@@ -118,6 +118,9 @@ def test_submodule_export_parity_complete():
                      ("audio.functional", "audio/functional"),
                      ("audio.features", "audio/features"),
                      ("audio.backends", "audio/backends"),
+                     ("incubate.autograd", "incubate/autograd"),
+                     ("incubate.optimizer", "incubate/optimizer"),
+                     ("nn.utils", "nn/utils"),
                      ("nn.initializer", "nn/initializer"),
                      ("incubate.nn", "incubate/nn"),
                      ("callbacks", "callbacks"),
