@@ -264,6 +264,7 @@ from . import distribution  # noqa: F401
 from . import profiler  # noqa: F401
 from . import linalg  # noqa: F401
 from . import fft  # noqa: F401
+from . import base  # noqa: F401
 from . import callbacks  # noqa: F401
 from . import dataset  # noqa: F401
 from . import hub  # noqa: F401
