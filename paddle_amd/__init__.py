@@ -266,6 +266,7 @@ from . import linalg  # noqa: F401
 from . import fft  # noqa: F401
 from . import base  # noqa: F401
 from . import callbacks  # noqa: F401
+from . import cost_model  # noqa: F401
 from . import dataset  # noqa: F401
 from . import hub  # noqa: F401
 from . import utils  # noqa: F401

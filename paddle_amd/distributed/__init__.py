@@ -89,3 +89,4 @@ def gather(tensor, gather_list=None, dst=0, group=None, sync_op=True):
 
 
 from . import launch  # noqa: E402,F401
+from . import passes  # noqa: E402,F401

@@ -194,3 +194,4 @@ from .pipeline import PipelineLayer, PipelineParallel, LayerDesc, SharedLayerDes
 from .recompute import recompute  # noqa: E402,F401
 
 from . import meta_parallel  # noqa: E402,F401
+from . import metrics  # noqa: E402,F401

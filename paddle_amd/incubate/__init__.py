@@ -1,5 +1,6 @@
 from . import asp  # noqa: F401
 from . import autograd  # noqa: F401
+from . import multiprocessing  # noqa: F401
 from . import autotune  # noqa: F401
 from . import fp8  # noqa: F401
 from . import nn  # noqa: F401
