@@ -120,3 +120,36 @@ def load_state_dict(state_dict, path, process_group=None, shard_info=None):
         else:
             state_dict[k] = lv
     return state_dict
+
+
+def merge_sharded(path):
+    """Offline utility: reassemble a sharded (flat-slice) checkpoint into
+    full per-key tensors without a process group (reference:
+    checkpoint/utils.py merge behavior).  Returns {key: tensor} with every
+    shard_info key rebuilt at its global size."""
+    meta = _read_meta(path)
+    ranks_meta = meta["ranks"]
+    files = {}
+
+    def rf(r):
+        if r not in files:
+            with open(os.path.join(path, f"rank_{r}.distcp"), "rb") as f:
+                files[r] = pickle.load(f)
+        return files[r]
+
+    out = {}
+    for r, rmeta in ranks_meta.items():
+        for k, km in rmeta.items():
+            if km.get("shape") is None:      # non-tensor
+                out.setdefault(k, rf(r)[k])
+                continue
+            if "shard" in km:
+                g = km["shard"]["global_numel"]
+                if k not in out:
+                    out[k] = torch.zeros(g, dtype=rf(r)[k].dtype)
+                o = km["shard"]["offset"]
+                n = km["shard"]["numel"]
+                out[k][o:o + n] = rf(r)[k].view(-1)
+            else:
+                out.setdefault(k, rf(r)[k])
+    return out
