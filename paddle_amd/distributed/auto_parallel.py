@@ -349,3 +349,100 @@ def dist_reduce(x, axis, fn=torch.sum, keepdim=False):
     out_chars = "".join(c for i, c in enumerate(chars) if i != axis % nd)
     out_pl = infer_spmd(chars + "->" + out_chars, [_placements_of(x, mesh)], mesh)
     return _local_op(lambda t: fn(t, dim=axis, keepdim=keepdim), out_pl, mesh, x)
+
+
+class Strategy:
+    """auto-parallel strategy knobs (reference: auto_parallel/strategy.py).
+    Dataclass-lite: attribute bags per feature."""
+
+    def __init__(self):
+        self.auto_mode = "semi"
+        self.sharding = type("S", (), {"enable": False, "degree": 1,
+                                       "stage": 1})()
+        self.recompute = type("R", (), {"enable": False})()
+        self.pipeline = type("P", (), {"enable": False,
+                                       "schedule_mode": "1F1B"})()
+        self.amp = type("A", (), {"enable": False, "dtype": "bfloat16"})()
+
+
+class Engine:
+    """Semi-auto training engine (reference: distributed/auto_parallel/
+    static/engine.py:Engine -- fit/evaluate/predict over a DistTensor
+    program).  This build executes the same API on the dygraph path:
+    shard_fn placements are applied via shard_layer, sharding/recompute
+    strategy knobs map onto the Fleet implementations."""
+
+    def __init__(self, model, loss=None, optimizer=None, metrics=None,
+                 strategy=None):
+        self.model = model
+        self.loss = loss
+        self.optimizer = optimizer
+        self.metrics = metrics or []
+        self.strategy = strategy or Strategy()
+        if self.strategy.sharding.enable and self.strategy.sharding.stage == 3:
+            from .fleet.sharding import GroupShardedStage3
+            self.model = GroupShardedStage3(model)
+        if self.strategy.recompute.enable and hasattr(model, "enable_recompute"):
+            model.enable_recompute()
+
+    def _step(self, batch, train=True):
+        import contextlib
+
+        import torch
+        x, y = batch if isinstance(batch, (list, tuple)) else (batch, None)
+        amp = self.strategy.amp
+        ctx = (torch.autocast("cuda", dtype=torch.bfloat16)
+               if amp.enable and torch.cuda.is_available()
+               else contextlib.nullcontext())
+        with ctx:
+            out = self.model(x)
+            loss = self.loss(out, y) if self.loss is not None else out
+        if train:
+            loss.backward()
+            if self.optimizer is not None:
+                self.optimizer.step()
+                self.optimizer.clear_grad()
+        return loss, out
+
+    def fit(self, train_data, epochs=1, batch_size=1, steps_per_epoch=None,
+            log_freq=10, verbose=0):
+        from .. import io as pio
+        loader = train_data if hasattr(train_data, "__iter__") else \
+            pio.DataLoader(train_data, batch_size=batch_size, shuffle=True)
+        history = []
+        for ep in range(epochs):
+            for step, batch in enumerate(loader):
+                loss, _ = self._step(batch, train=True)
+                if steps_per_epoch and step + 1 >= steps_per_epoch:
+                    break
+            history.append(float(loss.detach().float()))
+        return history
+
+    def evaluate(self, eval_data, batch_size=1, steps=None):
+        import torch
+        from .. import io as pio
+        loader = eval_data if hasattr(eval_data, "__iter__") else \
+            pio.DataLoader(eval_data, batch_size=batch_size)
+        tot, n = 0.0, 0
+        with torch.no_grad():
+            for i, batch in enumerate(loader):
+                loss, _ = self._step(batch, train=False)
+                tot += float(loss.detach().float())
+                n += 1
+                if steps and i + 1 >= steps:
+                    break
+        return {"loss": tot / max(n, 1)}
+
+    def predict(self, test_data, batch_size=1, steps=None):
+        import torch
+        from .. import io as pio
+        loader = test_data if hasattr(test_data, "__iter__") else \
+            pio.DataLoader(test_data, batch_size=batch_size)
+        outs = []
+        with torch.no_grad():
+            for i, batch in enumerate(loader):
+                x = batch[0] if isinstance(batch, (list, tuple)) else batch
+                outs.append(self.model(x))
+                if steps and i + 1 >= steps:
+                    break
+        return outs

@@ -345,3 +345,25 @@ def test_zero_bubble_pp_matches_1f1b():
                 assert torch.allclose(a, b2, atol=1e-6)
         print("rank", r, "zb ok")
     """, timeout=300)
+
+
+def test_auto_parallel_engine_fit():
+    import torch
+    import paddle_amd as paddle
+    from paddle_amd.distributed import auto_parallel as ap
+
+    class DS(paddle.io.Dataset):
+        def __getitem__(self, i):
+            torch.manual_seed(i)
+            x = torch.randn(4)
+            return x, (x * 2).sum().reshape(1)
+
+        def __len__(self):
+            return 32
+
+    net = paddle.nn.Linear(4, 1)
+    opt = paddle.optimizer.SGD(learning_rate=0.05, parameters=net.parameters())
+    eng = ap.Engine(net, paddle.nn.MSELoss(), opt, strategy=ap.Strategy())
+    hist = eng.fit(DS(), epochs=6, batch_size=8)
+    assert hist[-1] < hist[0]
+    assert eng.evaluate(DS(), batch_size=8)["loss"] < 1.5
