@@ -194,6 +194,27 @@ class GenerationServer:
             return {"choices": [{"token_ids": req.out_ids}],
                     "usage": {"completion_tokens": len(req.out_ids)}}
 
+        @api.post("/v1/completions/stream")
+        def complete_stream(body: CompletionIn):
+            """Server-sent events: one `data:` line per generated token."""
+            from fastapi.responses import StreamingResponse
+
+            req = Request(prompt_ids=list(body.prompt),
+                          max_new_tokens=body.max_tokens,
+                          temperature=body.temperature, top_p=body.top_p)
+            self.engine.add_request(req)
+
+            def gen():
+                import json
+                sent = 0
+                while not req.done:
+                    self.engine.step()
+                    while sent < len(req.out_ids):
+                        yield f"data: {json.dumps({'token_id': req.out_ids[sent]})}\n\n"
+                        sent += 1
+                yield "data: [DONE]\n\n"
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
         @api.get("/stats")
         def stats():
             return self.engine.stats()

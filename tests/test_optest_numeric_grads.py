@@ -160,3 +160,28 @@ def test_maxout_optest():
     t = TestMaxoutOp()
     t.check_output()
     t.check_grad()
+
+
+class TestRopeOp(OpTest):
+    rtol, atol = 1e-4, 1e-5
+
+    def make_inputs(self):
+        g = torch.Generator().manual_seed(9)
+        # [b=1, s=4, h=2, d=8]
+        return [torch.randn(1, 4, 2, 8, dtype=torch.float64, generator=g)]
+
+    def fn(self, x):
+        from paddle_amd.ops.functional import _Rope, build_rope_cache
+        cos, sin = build_rope_cache(4, 8, dtype=torch.float32)
+        return _Rope.apply(x.float(), cos, sin, 0).double()
+
+    def oracle(self, x):
+        from paddle_amd.ops.functional import _rope_ref, build_rope_cache
+        cos, sin = build_rope_cache(4, 8, dtype=torch.float64)
+        return _rope_ref(x, cos, sin, 0, conj=False)
+
+
+def test_rope_optest():
+    t = TestRopeOp()
+    t.check_output()
+    t.check_grad()

@@ -64,6 +64,7 @@ def test_http_app_contract():
     app = GenerationServer(eng).app()
     paths = {r.path for r in app.routes}
     assert "/v1/completions" in paths and "/stats" in paths
+    assert "/v1/completions/stream" in paths
 
 
 def test_gpt_runner_end_to_end_cpu():
