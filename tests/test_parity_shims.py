@@ -125,3 +125,28 @@ def test_audio_features_shapes():
         w_.writeframes((torch.arange(100, dtype=torch.int16)).numpy().tobytes())
     t, sr = A.backends.load(path)
     assert sr == 8000 and t.shape == (1, 100)
+
+
+def test_jit_save_load_roundtrip(tmp_path):
+    """jit.save/load: state + meta survive; weights restore into a fresh
+    layer (reference: jit/api.py save/load, translated_layer.py)."""
+    net = paddle.nn.Sequential(paddle.nn.Linear(4, 8), paddle.nn.Linear(8, 2))
+    x = torch.randn(3, 4)
+    ref = net(x)
+    path = str(tmp_path / "model")
+    paddle.jit.save(net, path)
+    tl = paddle.jit.load(path)
+    assert tl._meta["class_name"] == "Sequential"
+    net2 = paddle.nn.Sequential(paddle.nn.Linear(4, 8), paddle.nn.Linear(8, 2))
+    net2.set_state_dict(tl.layer_state())
+    assert torch.allclose(net2(x), ref, atol=1e-6)
+
+
+def test_to_static_passthrough_semantics():
+    @paddle.jit.to_static
+    def f(a, b):
+        return a * 2 + b
+
+    out = f(torch.ones(2), torch.ones(2))
+    assert torch.allclose(out, torch.full((2,), 3.0))
+    assert callable(f.dygraph_function)
