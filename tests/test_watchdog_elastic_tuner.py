@@ -41,19 +41,24 @@ def test_watchdog_completed_task_no_fire():
 
 
 def test_elastic_membership(tmp_path):
+    # generous margins: a loaded CI box can stall a heartbeat thread for
+    # hundreds of ms, which must not look like a lost lease
     m1 = ElasticManager(job_id="t", host="a:1", store_dir=str(tmp_path),
-                        heartbeat_interval=0.1, lease_ttl=0.5)
+                        heartbeat_interval=0.2, lease_ttl=3.0)
     m2 = ElasticManager(job_id="t", host="b:2", store_dir=str(tmp_path),
-                        heartbeat_interval=0.1, lease_ttl=0.5)
+                        heartbeat_interval=0.2, lease_ttl=3.0)
     m1.register()
     m2.register()
-    time.sleep(0.3)
+    time.sleep(0.6)
     assert m1.hosts() == ["a:1", "b:2"]
     changes = []
     m1.watch(lambda hosts: changes.append(hosts))
-    time.sleep(0.3)  # let the watch loop capture the 2-node baseline
+    time.sleep(0.6)  # let the watch loop capture the 2-node baseline
     m2.exit()  # node b leaves
-    time.sleep(1.5)
+    for _ in range(40):  # poll up to 8 s instead of a fixed sleep
+        if any(h == ["a:1"] for h in changes):
+            break
+        time.sleep(0.2)
     m1.exit()
     assert any(h == ["a:1"] for h in changes), changes
 
