@@ -277,7 +277,13 @@ from . import quantization  # noqa: F401
 from . import text  # noqa: F401
 from . import audio  # noqa: F401
 from . import onnx  # noqa: F401
-from .framework_io import save, load, async_save  # noqa: F401
+from .framework_io import (  # noqa: F401
+    async_save,
+    load,
+    load_safetensors,
+    save,
+    save_safetensors,
+)
 
 # remaining top-level parity names
 import torch as _torch  # noqa: E402

@@ -95,3 +95,22 @@ def clear_async_save_task_queue():
     for t in _async_threads:
         t.join()
     _async_threads.clear()
+
+
+def save_safetensors(state_dict, path, metadata=None):
+    """Save a flat {name: tensor} dict as .safetensors (ecosystem interchange;
+    HF loaders read these directly).  Non-tensor entries are rejected --
+    use paddle.save's pickle format for optimizer state."""
+    from safetensors.torch import save_file
+    flat = {}
+    for k, v in state_dict.items():
+        if not isinstance(v, torch.Tensor):
+            raise ValueError(f"save_safetensors: '{k}' is not a tensor; "
+                             "use paddle.save for mixed state")
+        flat[k] = v.detach().cpu().contiguous()
+    save_file(flat, path, metadata=metadata or {"format": "paddle_amd"})
+
+
+def load_safetensors(path, device="cpu"):
+    from safetensors.torch import load_file
+    return load_file(path, device=device)

@@ -223,3 +223,17 @@ def test_static_ema_apply_restore():
             assert torch.allclose(w.tensor, torch.full((4,), 3.0))  # restored
     finally:
         paddle.disable_static()
+
+
+def test_safetensors_roundtrip(tmp_path):
+    import torch
+    import paddle_amd as paddle
+    m = paddle.nn.Linear(4, 3)
+    path = str(tmp_path / "w.safetensors")
+    paddle.save_safetensors(m.state_dict(), path)
+    back = paddle.load_safetensors(path)
+    for k, v in m.state_dict().items():
+        torch.testing.assert_close(back[k], v.detach())
+    import pytest
+    with pytest.raises(ValueError):
+        paddle.save_safetensors({"step": 3}, str(tmp_path / "bad.safetensors"))
