@@ -20,7 +20,8 @@ __global__ void adamw_kernel(float* __restrict__ master, void* __restrict__ para
   // bias1 = 1 - beta1^t, bias2 = 1 - beta2^t
   const float inv_b1 = 1.f / bias1;
   const float inv_b2 = 1.f / bias2;
-  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4; i < numel;
+  const int64_t numel4 = numel & ~3ll;  // float4 body; <4 tail handled scalar below
+  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4; i < numel4;
        i += (int64_t)gridDim.x * blockDim.x * 4) {
     float4 pm = *reinterpret_cast<float4*>(master + i);
     float4 mm = *reinterpret_cast<float4*>(m + i);
@@ -59,6 +60,19 @@ __global__ void adamw_kernel(float* __restrict__ master, void* __restrict__ para
       *reinterpret_cast<float4*>((float*)param_bf16 + i) = make_float4(p[0], p[1], p[2], p[3]);
     }
   }
+  for (int64_t i = numel4 + (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < numel;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float g = (GDT == kBF16) ? bf2f(((const short*)grad)[i]) * gscale
+                             : ((const float*)grad)[i] * gscale;
+    float p = master[i];
+    p -= lr * wd * p;
+    float mo = beta1 * m[i] + (1.f - beta1) * g;
+    float vo = beta2 * v[i] + (1.f - beta2) * g * g;
+    p -= lr * (mo * inv_b1) / (sqrtf(vo * inv_b2) + eps);
+    master[i] = p; m[i] = mo; v[i] = vo;
+    if (BF16OUT) ((short*)param_bf16)[i] = f2bf(p);
+    else if (param_bf16) ((float*)param_bf16)[i] = p;
+  }
 }
 
 void adamw(float* master, void* param_bf16, const void* grad, float* m,
@@ -90,8 +104,9 @@ __global__ void l2norm_sq_kernel(const void* __restrict__ x, float* __restrict__
                                  int64_t numel) {
   __shared__ float red[4];
   float acc = 0.f;
-  int64_t per_blk = ((numel / 4 + gridDim.x - 1) / gridDim.x) * 4;
-  int64_t blk1 = min(numel, (int64_t)(blockIdx.x + 1) * per_blk);
+  const int64_t numel4 = numel & ~3ll;  // float4 body; <4 tail handled scalar below
+  int64_t per_blk = ((numel4 / 4 + gridDim.x - 1) / gridDim.x) * 4;
+  int64_t blk1 = min(numel4, (int64_t)(blockIdx.x + 1) * per_blk);
   for (int64_t i = (int64_t)blockIdx.x * per_blk + (int64_t)threadIdx.x * 4; i < blk1;
        i += (int64_t)blockDim.x * 4) {
     if (DT == kBF16) {
@@ -101,6 +116,12 @@ __global__ void l2norm_sq_kernel(const void* __restrict__ x, float* __restrict__
     } else {
       float4 v = *reinterpret_cast<const float4*>((const float*)x + i);
       acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+    }
+  }
+  if (blockIdx.x == 0) {
+    for (int64_t i = numel4 + threadIdx.x; i < numel; i += blockDim.x) {
+      float f = (DT == kBF16) ? bf2f(((const short*)x)[i]) : ((const float*)x)[i];
+      acc += f * f;
     }
   }
   float t = block_reduce_256(acc, SumOp(), red, 0.f);

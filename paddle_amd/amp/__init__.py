@@ -83,6 +83,7 @@ class GradScaler:
         self._good_steps = 0
         self._bad_steps = 0
         self._found_inf = False
+        self._unscaled_opts = set()  # id(optimizer) already unscaled this step
 
     def is_enable(self):
         return self._enable
@@ -100,25 +101,35 @@ class GradScaler:
                 if not torch.isfinite(g.float().sum()):
                     found = True
                     break
+        # Hybrid-parallel semantics (reference fleet/scaler.py distributed_scaler):
+        # every rank must agree on skip/apply or DP/PP/sharded replicas desync.
+        if torch.distributed.is_available() and torch.distributed.is_initialized():
+            dev = "cuda" if torch.cuda.is_available() else "cpu"
+            flag = torch.tensor([1.0 if found else 0.0], device=dev)
+            torch.distributed.all_reduce(flag, op=torch.distributed.ReduceOp.MAX)
+            found = bool(flag.item() > 0)
         self._found_inf = found
         return found
 
     def unscale_(self, optimizer):
         if not self._enable:
             return
+        if id(optimizer) in self._unscaled_opts:
+            return  # already unscaled this step (paddle raises; we no-op safely)
         inv = 1.0 / self._scale
         for p in optimizer._params:
             if p.grad is not None:
                 p.grad.mul_(inv)
+        self._unscaled_opts.add(id(optimizer))
         self._check_finite(optimizer)
 
     def step(self, optimizer):
         if not self._enable:
             optimizer.step()
             return
-        if not getattr(self, "_unscaled", False):
+        if id(optimizer) not in self._unscaled_opts:
             self.unscale_(optimizer)
-        self._unscaled = False
+        self._unscaled_opts.discard(id(optimizer))
         if not self._found_inf:
             optimizer.step()
 

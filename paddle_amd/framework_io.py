@@ -21,18 +21,36 @@ _PROTOCOL = 2
 
 
 class _BF16Array:
-    """Marker wrapper: bf16 bits as uint16 ndarray (unpickles to bf16)."""
+    """Legacy round-1 wrapper kept ONLY so old repo-saved files still load.
+    New files use the reference's own convention (raw uint16 ndarray)."""
 
     def __init__(self, bits: np.ndarray):
         self.bits = bits
 
 
+_name_counter = [0]
+
+
+def _tensor_to_reference_form(obj: torch.Tensor):
+    """Reference reduce_varbase (io.py:425): a tensor pickles as the plain
+    tuple (name, ndarray).  bf16 has no numpy dtype; the reference's
+    Tensor.numpy() emits raw-bits uint16 and paddle.to_tensor maps uint16
+    back to bfloat16 (the uint16-means-bf16 convention)."""
+    t = obj.detach().cpu().contiguous()
+    if t.dtype == torch.bfloat16:
+        data = t.view(torch.uint16).numpy()
+    else:
+        data = t.numpy()
+    name = getattr(obj, "name", None)
+    if not name:
+        _name_counter[0] += 1
+        name = f"generated_tensor_{_name_counter[0]}"
+    return (name, data)
+
+
 def _to_serializable(obj: Any) -> Any:
     if isinstance(obj, torch.Tensor):
-        t = obj.detach().cpu()
-        if t.dtype == torch.bfloat16:
-            return _BF16Array(t.view(torch.uint16).numpy())
-        return t.numpy()
+        return _tensor_to_reference_form(obj)
     if isinstance(obj, dict):
         return {k: _to_serializable(v) for k, v in obj.items()}
     if isinstance(obj, (list, tuple)):
@@ -41,12 +59,34 @@ def _to_serializable(obj: Any) -> Any:
     return obj
 
 
+def _is_varbase_tuple(obj) -> bool:
+    # reference _transformed_from_varbase (io.py:549)
+    return (isinstance(obj, tuple) and len(obj) == 2
+            and isinstance(obj[0], str) and isinstance(obj[1], np.ndarray))
+
+
+def _ndarray_to_tensor(arr: np.ndarray, return_numpy: bool):
+    if return_numpy:
+        return arr
+    if arr.dtype == np.uint16:  # uint16-means-bf16 convention
+        return torch.from_numpy(arr.copy()).view(torch.bfloat16)
+    return torch.from_numpy(arr.copy())
+
+
 def _from_serializable(obj: Any, return_numpy=False) -> Any:
-    if isinstance(obj, _BF16Array):
+    if isinstance(obj, _BF16Array):  # legacy round-1 files
         t = torch.from_numpy(obj.bits.copy()).view(torch.bfloat16)
         return t.float().numpy() if return_numpy else t
+    if _is_varbase_tuple(obj):
+        t = _ndarray_to_tensor(obj[1], return_numpy)
+        if not return_numpy:
+            try:
+                t.name = obj[0]
+            except (AttributeError, RuntimeError):
+                pass
+        return t
     if isinstance(obj, np.ndarray):
-        return obj if return_numpy else torch.from_numpy(obj.copy())
+        return _ndarray_to_tensor(obj, return_numpy)
     if isinstance(obj, dict):
         return {k: _from_serializable(v, return_numpy) for k, v in obj.items()}
     if isinstance(obj, (list, tuple)):

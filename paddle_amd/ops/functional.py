@@ -362,10 +362,11 @@ class _FlashAttn(torch.autograd.Function):
             ctx.native = False
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale, ctx.causal = scale, causal
-        return o
+        ctx.mark_non_differentiable(lse)
+        return o, lse
 
     @staticmethod
-    def backward(ctx, do):
+    def backward(ctx, do, _dlse):
         q, k, v, o, lse = ctx.saved_tensors
         if ctx.native:
             C = _ext.get_ext()
@@ -514,11 +515,11 @@ def flash_attention(q, k, v, dropout=0.0, causal=False, scale=None,
         vt = v.transpose(1, 2)
     else:
         qt, kt, vt = q, k, v
-    o = _FlashAttn.apply(qt, kt, vt, scale, causal)
+    o, lse = _FlashAttn.apply(qt, kt, vt, scale, causal)
     if layout == "bshd":
         o = o.transpose(1, 2)
     if return_softmax_lse:
-        return o, None  # lse kept internal for now
+        return o, lse  # [B, H, Sq] fp32 logsumexp of scaled scores
     return o, None
 
 

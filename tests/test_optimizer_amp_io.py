@@ -1,6 +1,7 @@
 """Optimizers vs torch references; GradScaler; .pdparams/.pdopt roundtrip;
 LR schedulers (reference: optimizer.py / lr.py / grad_scaler.py / io.py)."""
 import math
+import pickle
 import os
 
 import numpy as np
@@ -237,3 +238,97 @@ def test_safetensors_roundtrip(tmp_path):
     import pytest
     with pytest.raises(ValueError):
         paddle.save_safetensors({"step": 3}, str(tmp_path / "bad.safetensors"))
+
+
+def test_gradscaler_unscale_then_step_single_unscale():
+    # unscale_ followed by step() must not divide by scale twice.
+    m = nn.Linear(4, 4)
+    opt = paddle.optimizer.SGD(learning_rate=1.0, parameters=m.parameters())
+    scaler = paddle.amp.GradScaler(init_loss_scaling=2.0 ** 8)
+    x = paddle.randn([2, 4])
+    loss = m(x).sum()
+    scaled = scaler.scale(loss)
+    scaled.backward()
+    scaler.unscale_(opt)  # user-side unscale for grad clipping
+    g_after_unscale = [p.grad.clone() for p in m.parameters()]
+    scaler.step(opt)      # must NOT unscale again
+    # recompute reference grads at scale 1
+    m2 = nn.Linear(4, 4)
+    with torch.no_grad():
+        for p2, p in zip(m2.parameters(), m.parameters()):
+            pass  # shapes match; we only compare grad scale consistency
+    # grads stored on m were unscaled exactly once: loss grad of sum() is all-ones-ish
+    # check: sum-of-abs of grads is within 2x band of the re-derived unscaled grads
+    m.zero_grad()
+    loss2 = m(x).sum()
+    loss2.backward()
+    for g_once, p in zip(g_after_unscale, m.parameters()):
+        torch.testing.assert_close(g_once, p.grad, atol=1e-5, rtol=1e-5)
+
+
+def test_gradscaler_double_unscale_is_noop():
+    m = nn.Linear(3, 3)
+    opt = paddle.optimizer.SGD(learning_rate=0.1, parameters=m.parameters())
+    scaler = paddle.amp.GradScaler(init_loss_scaling=16.0)
+    loss = m(paddle.randn([2, 3])).sum()
+    scaler.scale(loss).backward()
+    scaler.unscale_(opt)
+    g1 = [p.grad.clone() for p in m.parameters()]
+    scaler.unscale_(opt)  # second call: no-op, not another /16
+    for a, p in zip(g1, m.parameters()):
+        torch.testing.assert_close(a, p.grad)
+
+
+def test_bf16_pdparams_reference_interop(tmp_path):
+    """Byte-compat: bf16 tensors must pickle as the reference's own form --
+    a plain (name, uint16 ndarray) tuple with NO repo-private classes
+    (reference python/paddle/framework/io.py:425 reduce_varbase)."""
+    import pickletools, io as _io
+    w = torch.randn(4, 5).bfloat16()
+    p = str(tmp_path / "m.pdparams")
+    paddle.save({"w": w, "b": torch.ones(3)}, p)
+    raw = open(p, "rb").read()
+    # scan pickle opcodes: no GLOBAL/STACK_GLOBAL may reference paddle_amd
+    for op, arg, _ in pickletools.genops(raw):
+        if op.name in ("GLOBAL", "STACK_GLOBAL") and arg and "paddle_amd" in str(arg):
+            raise AssertionError(f"repo-private class in pickle: {arg}")
+    # unpickle with plain pickle (what the reference's loader does first)
+    obj = pickle.loads(raw)
+    assert isinstance(obj["w"], tuple) and len(obj["w"]) == 2
+    name, arr = obj["w"]
+    assert isinstance(name, str) and isinstance(arr, np.ndarray)
+    assert arr.dtype == np.uint16  # uint16-means-bf16 convention
+    # and our loader restores bf16 values exactly
+    back = paddle.load(p)
+    assert back["w"].dtype == torch.bfloat16
+    torch.testing.assert_close(back["w"].float(), w.float())
+    assert back["b"].dtype == torch.float32
+
+
+def test_load_reference_produced_bf16_fixture(tmp_path):
+    """Simulate a file written by the reference: state values are
+    (name, ndarray) tuples, bf16 as raw uint16 bits."""
+    w = torch.randn(6).bfloat16()
+    fixture = {"w": ("linear_0.w_0", w.view(torch.uint16).numpy()),
+               "s": ("scalar_0", np.float64(3.5))}
+    p = str(tmp_path / "ref.pdparams")
+    with open(p, "wb") as f:
+        pickle.dump(fixture, f, protocol=2)
+    back = paddle.load(p)
+    assert back["w"].dtype == torch.bfloat16
+    torch.testing.assert_close(back["w"].float(), w.float())
+
+
+def test_flash_attention_returns_lse_cpu():
+    # CPU fallback path: return_softmax_lse must hand back the real LSE
+    from paddle_amd.ops import functional as F
+    q = torch.randn(2, 8, 4, 16)  # [b, s, h, d] bshd layout
+    k = torch.randn(2, 8, 4, 16)
+    v = torch.randn(2, 8, 4, 16)
+    o, lse = F.flash_attention(q, k, v, causal=True, return_softmax_lse=True)
+    assert lse is not None and lse.shape == (2, 4, 8)  # [b, h, sq]
+    # verify against direct logsumexp of scaled causal scores
+    s = torch.einsum("bshd,bthd->bhst", q, k) / math.sqrt(16)
+    mask = torch.ones(8, 8, dtype=torch.bool).tril()
+    s = s.masked_fill(~mask, float("-inf"))
+    torch.testing.assert_close(lse, torch.logsumexp(s, -1), atol=1e-4, rtol=1e-4)
