@@ -404,6 +404,65 @@ Tensor gemm_bf16(const Tensor& a, const Tensor& b, bool b_is_nt) {
   return c;
 }
 
+// Full GEMM with layout + fused epilogue.  layout: 0=NT 1=NN 2=TN.
+// epilogue: 0=none 1=bias 2=bias_gelu(aux out) 3=dgelu(aux in).
+// Returns {C} or {C, aux} for bias_gelu.  c_in (optional): accumulate into
+// it in-place (epilogue must be 0) -- the fused_linear_param_grad_add path.
+std::vector<Tensor> gemm_bf16_ex(const Tensor& a, const Tensor& b,
+                                 int64_t layout, int64_t epilogue,
+                                 const c10::optional<Tensor>& bias,
+                                 const c10::optional<Tensor>& aux_in,
+                                 const c10::optional<Tensor>& c_in) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(b.is_cuda() && b.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2);
+  TORCH_CHECK(a.stride(1) == 1 && b.stride(1) == 1, "row-major inputs only");
+  int64_t m, n, k;
+  if (layout == 0) {        // A[m][k] x Bt[n][k]
+    m = a.size(0); k = a.size(1); n = b.size(0);
+    TORCH_CHECK(b.size(1) == k, "gemm NT: inner dims mismatch");
+  } else if (layout == 1) { // A[m][k] x B[k][n]
+    m = a.size(0); k = a.size(1); n = b.size(1);
+    TORCH_CHECK(b.size(0) == k, "gemm NN: inner dims mismatch");
+  } else {                  // At[k][m] x B[k][n]
+    k = a.size(0); m = a.size(1); n = b.size(1);
+    TORCH_CHECK(b.size(0) == k, "gemm TN: inner dims mismatch");
+  }
+  TORCH_CHECK(k >= 64, "gemm_bf16_ex: K must be >= 64");
+  bool accumulate = c_in.has_value();
+  Tensor c;
+  if (accumulate) {
+    c = c_in.value();
+    TORCH_CHECK(c.size(0) == m && c.size(1) == n && c.stride(1) == 1 &&
+                c.scalar_type() == torch::kBFloat16);
+    TORCH_CHECK(epilogue == 0, "accumulate requires epilogue=none");
+  } else {
+    c = torch::empty({m, n}, a.options());
+  }
+  const void* bias_p = nullptr;
+  if (epilogue == 1 || epilogue == 2) {
+    TORCH_CHECK(bias.has_value() && bias->numel() == n &&
+                bias->scalar_type() == torch::kBFloat16 && bias->is_contiguous());
+    bias_p = bias->const_data_ptr();
+  }
+  Tensor aux;
+  void* aux_p = nullptr;
+  if (epilogue == 2) {
+    aux = torch::empty({m, n}, a.options());
+    aux_p = aux.mutable_data_ptr();
+  } else if (epilogue == 3) {
+    TORCH_CHECK(aux_in.has_value() && aux_in->size(0) == m &&
+                aux_in->size(1) == n && aux_in->stride(1) == 1 &&
+                aux_in->scalar_type() == torch::kBFloat16);
+    aux_p = const_cast<void*>(aux_in->const_data_ptr());
+  }
+  pa::gemm_bf16_ex(a.const_data_ptr(), b.const_data_ptr(), c.mutable_data_ptr(),
+                   bias_p, aux_p, m, n, k, a.stride(0), b.stride(0), c.stride(0),
+                   (int)layout, (int)epilogue, accumulate, cur_stream());
+  if (epilogue == 2) return {c, aux};
+  return {c};
+}
+
 // ---- probe ----------------------------------------------------------------
 Tensor mfma_probe(const Tensor& a, const Tensor& bt) {
   CHECK_IN(a); CHECK_IN(bt);
@@ -449,6 +508,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);
   m.def("gemm_bf16", &gemm_bf16);
+  m.def("gemm_bf16_ex", &gemm_bf16_ex, py::arg("a"), py::arg("b"),
+        py::arg("layout"), py::arg("epilogue") = 0,
+        py::arg("bias") = c10::nullopt, py::arg("aux") = c10::nullopt,
+        py::arg("c_in") = c10::nullopt);
   m.def("decode_attention", &decode_attention);
   m.def("mfma_probe", &mfma_probe);
   m.def("mfma_probe32", &mfma_probe32);

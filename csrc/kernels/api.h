@@ -136,11 +136,22 @@ void decode_attention(const void* q, const void* kcache, const void* vcache,
                       int64_t max_blocks, int64_t dh, float scale,
                       hipStream_t s);
 
-// ---- hand-written bf16 MFMA GEMM (C[m][n] = A[m][k] x B) ------------------
+// ---- hand-written bf16 MFMA GEMM (C[m][n] = op(A) x op(B)) ----------------
 // b_is_nt: B passed as Bt[n][k] row-major (fast path); else B[k][n].
 void gemm_bf16(const void* a, const void* b, void* c, int64_t m, int64_t n,
                int64_t k, int64_t lda, int64_t ldb, int64_t ldc, bool b_is_nt,
                hipStream_t s);
+
+// full variant with operand layout + fused epilogue:
+//   layout: 0=NT (A[m][k], Bt[n][k])  1=NN (A[m][k], B[k][n])
+//           2=TN (At[k][m], B[k][n])  -- wgrad
+//   epilogue: 0=none 1=+bias[n] 2=gelu(x+bias) writing pre-act to aux
+//             3=dgelu (out = C * gelu'(aux))
+//   accumulate: C += result (epilogue forced to none)
+void gemm_bf16_ex(const void* a, const void* b, void* c, const void* bias,
+                  void* aux, int64_t m, int64_t n, int64_t k, int64_t lda,
+                  int64_t ldb, int64_t ldc, int layout, int epilogue,
+                  bool accumulate, hipStream_t s);
 
 // weight-only int8 GEMV (decode): qweight [N,K] int8 rows, scale [N] fp32
 void weight_only_gemv(const void* x, const void* wq, const float* scale,

@@ -106,6 +106,11 @@ class GPTMLP(nn.Layer):
         _apply_initializer(out_init, self.fc2.weight)
 
     def forward(self, x):
+        if hot._own_linear_ok(x, self.fc1.weight):
+            # own MFMA NT GEMM: fc1 bias+GELU in the kernel epilogue (aux
+            # saved), fc2 dgrad carries dGELU in its epilogue
+            return hot.fused_ffn_own(x, self.fc1.weight, self.fc1_bias,
+                                     self.fc2.weight, self.fc2.bias)
         if hot._fused_ffn_available(x):
             # fc1 bias+GELU in the hipBLASLt epilogue; backward fuses
             # dGELU + fc1 bias-grad into fc2's dgrad GEMM

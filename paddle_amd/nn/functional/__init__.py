@@ -25,8 +25,12 @@ sdp_kernel = None
 
 
 def linear(x, weight, bias=None, name=None):
-    # paddle weight layout: [in_features, out_features].  addmm fuses the
-    # bias into the hipBLASLt epilogue (saves one HBM pass per call).
+    # paddle weight layout: [in_features, out_features].  Shapes the
+    # autotune table assigns to the own MFMA kernel run gemm.hip's 8-phase
+    # NT path (fwd via cached W^T, dgrad direct); everything else goes to
+    # hipBLASLt via addmm (bias fused in its epilogue).
+    if hot._own_linear_ok(x, weight):
+        return hot.fused_linear_own(x, weight, bias)
     if bias is not None and x.dim() >= 2:
         x2 = x.reshape(-1, x.shape[-1])
         out = torch.addmm(bias, x2, weight)

@@ -735,6 +735,104 @@ def fused_ffn(x, w1, b1, w2, b2):
     return _FusedFFN.apply(x, w1, b1, w2, b2)
 
 
+# ---------------------------------------------------------------------------
+# own-MFMA fused linear / FFN (gemm.hip 8-phase kernel + epilogues).
+# fwd runs NT with a per-step cached W^T; dgrad is NT directly because
+# paddle's W[in,out] IS the NT B-operand; wgrad picks own-TN vs hipBLASLt
+# from the autotune table.  Reference: fused_gemm_epilogue_kernel.cu +
+# matmul_kernel_impl.h:914 dispatch pattern.
+# ---------------------------------------------------------------------------
+from . import gemm_dispatch as _gd  # noqa: E402
+
+
+def _wgrad(x2, dy2):
+    if _gd.use_own("tn", x2.shape[1], dy2.shape[1], x2.shape[0]):
+        return _gd.gemm_tn(x2, dy2)
+    return torch.matmul(x2.t(), dy2)
+
+
+def _colsum(dy2, dtype):
+    C = _ext.get_ext()
+    return C.colsum(dy2).to(dtype)
+
+
+class _FusedLinearOwn(torch.autograd.Function):
+    """y = x @ W (+bias) on the own NT kernel (W paddle-layout [K, N])."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        xs = x.shape
+        x2 = x.reshape(-1, xs[-1]).contiguous()
+        wt = _gd.weight_t(w)
+        if bias is not None:
+            y2 = _gd.gemm_nt(x2, wt, epilogue=1, bias=bias.contiguous())
+        else:
+            y2 = _gd.gemm_nt(x2, wt)
+        ctx.save_for_backward(x2, w)
+        ctx.xshape, ctx.has_bias = xs, bias is not None
+        return y2.reshape(*xs[:-1], w.shape[1])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, w = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx = _gd.gemm_nt(dy2, w) if _gd.use_own("nt", dy2.shape[0], w.shape[0], w.shape[1]) \
+            else torch.matmul(dy2, w.t())
+        dw = _wgrad(x2, dy2)
+        db = _colsum(dy2, w.dtype) if ctx.has_bias else None
+        return dx.reshape(ctx.xshape), dw, db
+
+
+class _FusedFFNOwn(torch.autograd.Function):
+    """FFN pair on the own NT kernel: fc1 carries bias+GELU in the GEMM
+    epilogue (pre-activation saved as aux), fc2's dgrad carries dGELU in
+    its epilogue -- the [tokens, 4h] tensor never sees a separate
+    elementwise pass in either direction."""
+
+    @staticmethod
+    def forward(ctx, x, w1, b1, w2, b2):
+        xs = x.shape
+        x2 = x.reshape(-1, xs[-1]).contiguous()
+        g, z = _gd.gemm_nt(x2, _gd.weight_t(w1), epilogue=2, bias=b1.contiguous())
+        if b2 is not None:
+            y = _gd.gemm_nt(g, _gd.weight_t(w2), epilogue=1, bias=b2.contiguous())
+        else:
+            y = _gd.gemm_nt(g, _gd.weight_t(w2))
+        ctx.save_for_backward(x2, w1, w2, z, g)
+        ctx.xshape, ctx.has_b2 = xs, b2 is not None
+        return y.reshape(*xs[:-1], w2.shape[1])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, w1, w2, z, g = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+        # dz = (dy @ W2^T) * gelu'(z): DGELU epilogue on fc2's dgrad
+        dz = _gd.gemm_nt(dy2, w2, epilogue=3, aux=z)
+        dw2 = _wgrad(g, dy2)
+        db2 = _colsum(dy2, w2.dtype) if ctx.has_b2 else None
+        dx = _gd.gemm_nt(dz, w1) if _gd.use_own("nt", dz.shape[0], w1.shape[0], w1.shape[1]) \
+            else torch.matmul(dz, w1.t())
+        dw1 = _wgrad(x2, dz)
+        db1 = _colsum(dz, w1.dtype)
+        return dx.reshape(ctx.xshape), dw1, db1, dw2, db2
+
+
+def _own_linear_ok(x, w, layout="nt"):
+    if not (x.is_cuda and x.dtype == torch.bfloat16 and w.dtype == torch.bfloat16
+            and _ext.use_native(x)):
+        return False
+    m = x.numel() // x.shape[-1]
+    return _gd.use_own(layout, m, w.shape[1], w.shape[0])
+
+
+def fused_linear_own(x, w, bias=None):
+    return _FusedLinearOwn.apply(x, w, bias)
+
+
+def fused_ffn_own(x, w1, b1, w2, b2):
+    return _FusedFFNOwn.apply(x, w1, b1, w2, b2)
+
+
 _lt_ffn_ok = None
 
 

@@ -412,3 +412,169 @@ def test_flash_attention_gqa_bwd_gpu():
     _assert_close_bf16(q.grad, qf.grad, atol=5e-2, rtol=5e-2)
     _assert_close_bf16(k.grad, kf.grad, atol=8e-2, rtol=8e-2)
     _assert_close_bf16(v.grad, vf.grad, atol=8e-2, rtol=8e-2)
+
+
+# ---------------------------------------------------------------------------
+# gemm.hip: 8-phase MFMA GEMM, all layouts + epilogues (vs fp32 torch)
+# ---------------------------------------------------------------------------
+def _gemm_rel_ok(out, ref, tol=3e-2):
+    err = (out.float() - ref).abs().max().item() / max(ref.abs().max().item(), 1e-6)
+    assert err < tol, f"relerr {err:.3e}"
+
+
+@pytest.mark.parametrize("m,n,k", [(512, 512, 512), (768, 1024, 256),
+                                   (2048, 2048, 4096)])
+def test_gemm_nt_numerics(m, n, k):
+    torch.manual_seed(1)
+    C = _ext.get_ext()
+    a = _bf(torch.randn(m, k, device=DEV))
+    bt = _bf(torch.randn(n, k, device=DEV))
+    out = C.gemm_bf16_ex(a, bt, 0)[0]
+    _gemm_rel_ok(out, a.float() @ bt.float().t())
+
+
+@pytest.mark.parametrize("m,n,k", [(512, 512, 512), (2048, 1024, 2048)])
+def test_gemm_nn_numerics(m, n, k):
+    torch.manual_seed(2)
+    C = _ext.get_ext()
+    a = _bf(torch.randn(m, k, device=DEV))
+    b = _bf(torch.randn(k, n, device=DEV))
+    out = C.gemm_bf16_ex(a, b, 1)[0]
+    _gemm_rel_ok(out, a.float() @ b.float())
+
+
+@pytest.mark.parametrize("m,n,k", [(512, 512, 512), (1024, 2048, 2048)])
+def test_gemm_tn_numerics(m, n, k):
+    # wgrad: C[m,n] = At[k,m]^T @ B[k,n], reduction over k (the token dim)
+    torch.manual_seed(3)
+    C = _ext.get_ext()
+    at = _bf(torch.randn(k, m, device=DEV))
+    b = _bf(torch.randn(k, n, device=DEV))
+    out = C.gemm_bf16_ex(at, b, 2)[0]
+    _gemm_rel_ok(out, at.float().t() @ b.float(), tol=5e-2)
+
+
+def test_gemm_boundary_tiles():
+    # M/N not multiples of 256 exercise the guarded kernel + skip-interior
+    torch.manual_seed(4)
+    C = _ext.get_ext()
+    for (m, n, k) in [(300, 520, 512), (512, 777, 256), (130, 200, 128),
+                      (1000, 50304 % 2048 + 304, 512)]:
+        a = _bf(torch.randn(m, k, device=DEV))
+        bt = _bf(torch.randn(n, k, device=DEV))
+        out = C.gemm_bf16_ex(a, bt, 0)[0]
+        _gemm_rel_ok(out, a.float() @ bt.float().t())
+
+
+def test_gemm_k_tail():
+    # K not a multiple of 64: whole grid takes the guarded path
+    torch.manual_seed(5)
+    C = _ext.get_ext()
+    a = _bf(torch.randn(512, 200, device=DEV))
+    bt = _bf(torch.randn(512, 200, device=DEV))
+    out = C.gemm_bf16_ex(a, bt, 0)[0]
+    _gemm_rel_ok(out, a.float() @ bt.float().t())
+
+
+def test_gemm_bias_epilogue():
+    torch.manual_seed(6)
+    C = _ext.get_ext()
+    a = _bf(torch.randn(512, 512, device=DEV))
+    bt = _bf(torch.randn(768, 512, device=DEV))
+    bias = _bf(torch.randn(768, device=DEV))
+    out = C.gemm_bf16_ex(a, bt, 0, 1, bias)[0]
+    _gemm_rel_ok(out, a.float() @ bt.float().t() + bias.float())
+
+
+def test_gemm_bias_gelu_epilogue():
+    torch.manual_seed(7)
+    C = _ext.get_ext()
+    a = _bf(torch.randn(512, 512, device=DEV) * 0.5)
+    bt = _bf(torch.randn(768, 512, device=DEV) * 0.05)
+    bias = _bf(torch.randn(768, device=DEV) * 0.1)
+    out, aux = C.gemm_bf16_ex(a, bt, 0, 2, bias)
+    pre = a.float() @ bt.float().t() + bias.float()
+    _gemm_rel_ok(aux, pre)
+    _gemm_rel_ok(out, torch.nn.functional.gelu(pre), tol=4e-2)
+
+
+def test_gemm_dgelu_epilogue():
+    torch.manual_seed(8)
+    C = _ext.get_ext()
+    dy = _bf(torch.randn(512, 512, device=DEV))
+    w = _bf(torch.randn(768, 512, device=DEV) * 0.05)   # dgrad B-operand
+    z = _bf(torch.randn(512, 768, device=DEV))          # saved pre-act
+    out = C.gemm_bf16_ex(dy, w.t().contiguous(), 0, 3, None, z)[0]
+    zf = z.float().requires_grad_(True)
+    torch.nn.functional.gelu(zf).backward(dy.float() @ w.float())
+    # out = (dy @ w) * gelu'(z)
+    _gemm_rel_ok(out, zf.grad, tol=5e-2)
+
+
+def test_gemm_accumulate():
+    torch.manual_seed(9)
+    C = _ext.get_ext()
+    at = _bf(torch.randn(512, 512, device=DEV))
+    b = _bf(torch.randn(512, 768, device=DEV))
+    c0 = _bf(torch.randn(512, 768, device=DEV))
+    acc = c0.clone()
+    C.gemm_bf16_ex(at, b, 2, 0, None, None, acc)
+    _gemm_rel_ok(acc, c0.float() + at.float().t() @ b.float(), tol=5e-2)
+
+
+def test_fused_linear_own_grads():
+    """fused_linear_own fwd+bwd vs fp32 autograd reference."""
+    from paddle_amd.ops import gemm_dispatch
+    torch.manual_seed(10)
+    m, k, n = 1024, 512, 768
+    x = _bf(torch.randn(2, m // 2, k, device=DEV)).requires_grad_(True)
+    w = (_bf(torch.randn(k, n, device=DEV)) * 0.05).requires_grad_(True)
+    bias = (_bf(torch.randn(n, device=DEV)) * 0.1).requires_grad_(True)
+    y = hot.fused_linear_own(x, w, bias)
+    loss = (y.float() ** 2).mean()
+    loss.backward()
+    xf = x.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    bf_ = bias.detach().float().requires_grad_(True)
+    yf = xf @ wf + bf_
+    ((yf ** 2).mean()).backward()
+    _gemm_rel_ok(y, yf.detach(), tol=4e-2)
+    _gemm_rel_ok(x.grad, xf.grad, tol=5e-2)
+    _gemm_rel_ok(w.grad, wf.grad, tol=5e-2)
+    _gemm_rel_ok(bias.grad, bf_.grad, tol=5e-2)
+
+
+def test_fused_ffn_own_grads():
+    torch.manual_seed(11)
+    m, h, ffn = 512, 512, 1024
+    x = _bf(torch.randn(m, h, device=DEV)).requires_grad_(True)
+    w1 = (_bf(torch.randn(h, ffn, device=DEV)) * 0.05).requires_grad_(True)
+    b1 = (_bf(torch.randn(ffn, device=DEV)) * 0.1).requires_grad_(True)
+    w2 = (_bf(torch.randn(ffn, h, device=DEV)) * 0.05).requires_grad_(True)
+    b2 = (_bf(torch.randn(h, device=DEV)) * 0.1).requires_grad_(True)
+    y = hot.fused_ffn_own(x, w1, b1, w2, b2)
+    (y.float() ** 2).mean().backward()
+    xf = x.detach().float().requires_grad_(True)
+    w1f = w1.detach().float().requires_grad_(True)
+    b1f = b1.detach().float().requires_grad_(True)
+    w2f = w2.detach().float().requires_grad_(True)
+    b2f = b2.detach().float().requires_grad_(True)
+    yf = torch.nn.functional.gelu(xf @ w1f + b1f) @ w2f + b2f
+    (yf ** 2).mean().backward()
+    _gemm_rel_ok(y, yf.detach(), tol=4e-2)
+    _gemm_rel_ok(x.grad, xf.grad, tol=6e-2)
+    _gemm_rel_ok(w1.grad, w1f.grad, tol=6e-2)
+    _gemm_rel_ok(b1.grad, b1f.grad, tol=6e-2)
+    _gemm_rel_ok(w2.grad, w2f.grad, tol=6e-2)
+    _gemm_rel_ok(b2.grad, b2f.grad, tol=6e-2)
+
+
+def test_weight_t_cache_invalidation():
+    from paddle_amd.ops import gemm_dispatch as gd
+    w = _bf(torch.randn(64, 32, device=DEV))
+    t1 = gd.weight_t(w)
+    assert gd.weight_t(w) is t1          # cached
+    w.add_(1.0)                          # version bump
+    t2 = gd.weight_t(w)
+    assert t2 is not t1
+    torch.testing.assert_close(t2, w.t().contiguous())
