@@ -16,12 +16,15 @@
 //     (1 block/CU -- latency hidden by the phase interleave, not TLP)
 //   * 4 phases per K-tile, one C-quadrant (16 MFMA) each, with
 //     s_setprio(1) around the MFMA cluster (T5) and two barriers per
-//     phase; all ds_reads for the tile issued in phases 0-1 and held in
-//     VGPRs (A 64 + B 32 + acc 128 regs)
-//   * staging for tile t+1 issued during tile t (A at phase 0, B at
-//     phase 1) into the other LDS buffer; the ONLY vmcnt wait is a
-//     counted s_waitcnt vmcnt(4) at phase 0 -- prefetch loads stay in
-//     flight across barriers (T3+T4; never vmcnt(0) in the main loop)
+//     phase; fragment ds_reads pipelined one phase ahead of their MFMA
+//     cluster with counted lgkm waits (A 64 + B 32 + acc 128 regs)
+//   * staging for tile t+1 issued during tile t (B at phase 0; A of t+2
+//     split into round-pairs at phases 1/3) into the freed LDS slots;
+//     the ONLY vmcnt wait is a counted vmcnt(2) at phase 2's end --
+//     prefetch loads stay in flight across barriers (T3+T4; never a
+//     mid-loop vmcnt(0))
+//   * CUTLASS-style tile-group rasterization (GM=8) + XCD chunking so
+//     concurrently-resident blocks share A/B panels in L2
 //   * st_16x32-style XOR swizzle on LDS rows (T2); XCD-aware bijective
 //     workgroup swizzle (T1)
 //
@@ -103,11 +106,23 @@ __global__ void gemm_bf16_kernel(const short* __restrict__ ag, const short* __re
   }
   // FAST grid spans only the interior tile rectangle
   const int mt = FAST ? (M / BM) : ((M + BM - 1) / BM);
-  const int bm = orig % mt;
-  const int bn = orig / mt;
+  const int nt = FAST ? (N / BN) : ((N + BN - 1) / BN);
+  // Tile-group rasterization: GM consecutive bm-rows walk bn fastest, so
+  // the ~32 blocks concurrently resident on one XCD touch few A/B panels
+  // and their K-tile streams coincide in L2 (vs column-major order, which
+  // re-reads the B panel from HBM for every bm).
+  constexpr int GM = 8;
+  int bm, bn;
+  {
+    int band = orig / (GM * nt);
+    int rem = orig - band * (GM * nt);
+    int gm_band = mt - band * GM < GM ? mt - band * GM : GM;
+    bm = band * GM + rem % gm_band;
+    bn = rem / gm_band;
+  }
   const int row0 = bm * BM, col0 = bn * BN;
   if (!FAST && skip_interior &&
-      row0 + BM <= M && col0 + BN <= N && (K % BK) == 0)
+      row0 + BM <= M && col0 + BN <= N && (K % 64) == 0)
     return;  // covered by the fast launch
 
   const int tid = threadIdx.x;
@@ -132,7 +147,25 @@ __global__ void gemm_bf16_kernel(const short* __restrict__ ag, const short* __re
   const int nkt = (K + BK - 1) / BK;
 
   // ---- staging lambdas ----------------------------------------------------
-  // A operand, DMA path (A[M][K] row-major; NT/NN)
+  // A operand, DMA path (A[M][K] row-major; NT/NN).  Rounds r0,r1 only:
+  // round r covers tile rows [64r, 64r+64); the fast pipeline stages
+  // rounds {0,2} and {1,3} separately (they free after different phases).
+  auto stage_a_round = [&](int buf, int kt, int it) {
+    const long long k0 = (long long)kt * BK;
+    const short* src = ag + (long long)(row0 + s_r[it]) * lda + k0 + s_cp[it];
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)&a_lds[buf][it * 4096 + tid * 8],
+        16, 0, 0);
+  };
+  auto stage_b_round = [&](int buf, int kt, int it) {
+    const long long k0 = (long long)kt * BK;
+    const short* src = bg + (long long)(col0 + s_r[it]) * ldb + k0 + s_cp[it];
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)&b_lds[buf][it * 4096 + tid * 8],
+        16, 0, 0);
+  };
   auto stage_a_dma = [&](int buf, int kt) {
     const long long k0 = (long long)kt * BK;
 #pragma unroll
@@ -156,33 +189,7 @@ __global__ void gemm_bf16_kernel(const short* __restrict__ ag, const short* __re
           16, 0, 0);
     }
   };
-  // reg-transpose staging, load half: source S[K][X] row-major, 64 k-rows x
-  // 256 x-cols slab -> regs (4 x shortx8); used for B (NN/TN) and A (TN)
-  const int rot = tid & 7;
-  auto load_trans = [&](const short* sg, long long lds_, int x0, int kt, shortx8* v) {
-    const long long k0 = (long long)kt * BK;
-#pragma unroll
-    for (int it = 0; it < 4; ++it) {
-      int flat = it * 4096 + tid * 8;
-      int kr = flat / BN;             // 0..63  (k within tile)
-      int nc = flat % BN;             // 0..255 (x within tile)
-      v[it] = *reinterpret_cast<const shortx8*>(sg + (k0 + kr) * lds_ + x0 + nc);
-    }
-  };
-  // write the transposed slab into LDS[x][k] (rotated to spread banks)
-  auto write_trans = [&](short* lds, const shortx8* v) {
-#pragma unroll
-    for (int it = 0; it < 4; ++it) {
-      int flat = it * 4096 + tid * 8;
-      int kr = flat / BN;
-      int nc = flat % BN;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int i = (j + rot) & 7;
-        lds[g_swz(nc + i, kr)] = v[it][i];
-      }
-    }
-  };
+  const int rot = tid & 7;  // staggers transposed LDS writes across banks
 
   // ---- guarded (boundary / K-tail) staging: zero-fill out of range --------
   auto stage_guarded = [&](int buf, int kt) {
@@ -269,29 +276,29 @@ __global__ void gemm_bf16_kernel(const short* __restrict__ ag, const short* __re
 #pragma unroll
     for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
 
-  // fragment registers, reused across phases (guide's 12/4/8/4 read split):
-  //   af2 holds the current m-half of A (m0 for phases 0-1, m1 for 2-3)
-  //   bf2 holds the current n-half of B (n0, n1, n1, n0 re-read)
-  shortx8 af2[4][2];  // [mfrag within half][ks]
-  shortx8 bf2[2][2];  // [nfrag within half][ks]
+  // fragment registers.  afA/bfA hold the (m0)/(n0) halves, afB/bfB the
+  // (m1)/(n1) halves; the fast path software-pipelines reads one phase
+  // ahead of their MFMA cluster (counted lgkm waits).
+  shortx8 afA[4][2], afB[4][2];  // [mfrag within half][ks]
+  shortx8 bfA[2][2], bfB[2][2];  // [nfrag within half][ks]
 
-  auto read_a_half = [&](int buf, int mh) {
+  auto read_a_half = [&](int buf, int mh, shortx8 (*dst)[2]) {
 #pragma unroll
     for (int m = 0; m < 4; ++m)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
-        af2[m][ks] = *reinterpret_cast<const shortx8*>(
+        dst[m][ks] = *reinterpret_cast<const shortx8*>(
             &a_lds[buf][g_swz(wr * 128 + (mh * 4 + m) * 16 + l16, ks * 32 + lg * 8)]);
   };
-  auto read_b_half = [&](int buf, int nh) {
+  auto read_b_half = [&](int buf, int nh, shortx8 (*dst)[2]) {
 #pragma unroll
     for (int n = 0; n < 2; ++n)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
-        bf2[n][ks] = *reinterpret_cast<const shortx8*>(
+        dst[n][ks] = *reinterpret_cast<const shortx8*>(
             &b_lds[buf][g_swz(wc * 64 + (nh * 2 + n) * 16 + l16, ks * 32 + lg * 8)]);
   };
-  auto mfma_quadrant = [&](int mh, int nh) {
+  auto mfma_quadrant = [&](shortx8 (*af)[2], shortx8 (*bf)[2], int mh, int nh) {
     GEMM_SETPRIO(1);
 #pragma unroll
     for (int m = 0; m < 4; ++m)
@@ -300,91 +307,73 @@ __global__ void gemm_bf16_kernel(const short* __restrict__ ag, const short* __re
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
           acc[mh * 4 + m][nh * 2 + n] =
-              mfma_bf16(af2[m][ks], bf2[n][ks], acc[mh * 4 + m][nh * 2 + n]);
+              mfma_bf16(af[m][ks], bf[n][ks], acc[mh * 4 + m][nh * 2 + n]);
     GEMM_SETPRIO(0);
   };
 
-  if constexpr (FAST) {
+  if constexpr (FAST && LAYOUT == LAY_NT) {
     // =======================================================================
-    // fast path: 4-phase counted-vmcnt pipeline (T3+T4), T5 setprio
+    // fast path (NT only): 4-phase counted-vmcnt pipeline (T3+T4), T5.
+    //
+    // Staging runs ahead of consumption so every mid-loop wait is counted:
+    //   B(t+1): phase 0 of tile t        (into buf^1; free since t-1 ph3)
+    //   A(t+2) rounds {0,2}: phase 1 of t (into buf cur rows freed at ph0)
+    //   A(t+2) rounds {1,3}: phase 3 of t (rows freed at ph2)
+    //   wait for tile t+1's data: END of tile t phase 3 -- vmcnt(4), the
+    //   4 newest (A(t+2)) stay in flight across the barrier.
     // =======================================================================
-    shortx8 breg[4];  // reg-staged B slab (NN/TN)
-    shortx8 areg[4];  // reg-staged A slab (TN)
-
-    // prologue: stage tile 0 into buf 0
-    if (!A_DMA) load_trans(ag, lda, row0, 0, areg);
-    if (!B_DMA) load_trans(bg, ldb, col0, 0, breg);
-    if (A_DMA) stage_a_dma(0, 0);
-    if (B_DMA) stage_b_dma(0, 0);
-    if (!B_DMA || !A_DMA) {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      if (!A_DMA) write_trans(&a_lds[0][0], areg);
-      if (!B_DMA) write_trans(&b_lds[0][0], breg);
-    }
-    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    // 4-phase counted-vmcnt pipeline, fragment reads one phase ahead.
+    // (An 8-phase 2-K-tile variant with fully static banks measured SLOWER
+    // -- register-allocator spills in the hot loop; see profiles/ notes.)
+    stage_a_dma(0, 0);
+    stage_b_dma(0, 0);
+    if (nkt > 1) stage_a_dma(1, 1);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
+    // pre-read tile 0's (m0)/(n0) fragments (the pipeline's "prev ph3")
+    read_a_half(0, 0, afA);
+    read_b_half(0, 0, bfA);
 
     for (int kt = 0; kt < nkt; ++kt) {
       const int cur = kt & 1;
-      const bool pf = (kt + 1 < nkt);
+      const bool pf1 = (kt + 1 < nkt);
+      const bool pf2 = (kt + 2 < nkt);
 
-      // ---- phase 0: quadrant (m0,n0); issue next-tile loads ----
-      read_a_half(cur, 0);
-      read_b_half(cur, 0);
-      if (pf) {
-        if (!A_DMA) load_trans(ag, lda, row0, kt + 1, areg);         // TN A
-        if (!B_DMA && A_DMA) load_trans(bg, ldb, col0, kt + 1, breg);  // NN B
-        if (A_DMA) stage_a_dma(cur ^ 1, kt + 1);
-      }
-      // counted wait for THIS tile's data: prefetch loads stay in flight.
-      // On the last tile nothing newer was issued, so drain fully.
-      if (A_DMA && B_DMA) {
-        if (pf) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-        else    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      } else if (A_DMA) {
-        if (pf) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-        else    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      }
-      // TN: LDS was filled by ds_writes, drained at phase 2's lgkmcnt(0)
+      // ---- phase 0: MFMA (m0,n0); read B(n1); stage B(t+1) ----
+      read_b_half(cur, 1, bfB);
+      if (pf1) stage_b_dma(cur ^ 1, kt + 1);
       __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma_quadrant(0, 0);
+      // afA/bfA (12 reads from prev ph3) must land; bfB's 4 stay in flight
+      asm volatile("s_waitcnt lgkmcnt(4)" ::: "memory");
+      mfma_quadrant(afA, bfA, 0, 0);
       __builtin_amdgcn_s_barrier();
 
-      // ---- phase 1: quadrant (m0,n1); write-transpose A (TN) / B (NN) ----
-      // Writing buf^1 here is safe: tile kt-1's last reads of it (its n0
-      // re-read) were lgkm-drained before kt-1's closing barrier.
-      read_b_half(cur, 1);
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      if (pf && B_DMA) stage_b_dma(cur ^ 1, kt + 1);
-      if (pf && !A_DMA) {
-        if (!B_DMA) load_trans(bg, ldb, col0, kt + 1, breg);  // TN B issue
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");      // areg landed
-        write_trans(&a_lds[cur ^ 1][0], areg);
-      } else if (pf && !B_DMA) {
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");      // breg landed
-        write_trans(&b_lds[cur ^ 1][0], breg);                // NN
-      }
+      // ---- phase 1: MFMA (m0,n1); read A(m1); stage A(t+2) r0,r2 ----
+      read_a_half(cur, 1, afB);
+      if (pf2) { stage_a_round(cur, kt + 2, 0); stage_a_round(cur, kt + 2, 2); }
       __builtin_amdgcn_s_barrier();
-      mfma_quadrant(0, 1);
+      asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");  // bfB landed
+      mfma_quadrant(afA, bfB, 0, 1);
       __builtin_amdgcn_s_barrier();
 
-      // ---- phase 2: quadrant (m1,n1); write-transpose B (TN) ----
-      read_a_half(cur, 1);
-      if (pf && !B_DMA && !A_DMA) {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        write_trans(&b_lds[cur ^ 1][0], breg);
+      // ---- phase 2: MFMA (m1,n1); counted vmcnt for tile t+1 at end ----
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // afB landed
+      mfma_quadrant(afB, bfB, 1, 1);
+      if (pf1) {
+        // tile t+1's A (staged t-1 ph1/ph3) + B (ph0) must have landed so
+        // phase 3 can read them; A(t+2) r0/r2 (2 loads) stay outstanding
+        if (pf2) asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
       __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // A(m1) + our writes
-      mfma_quadrant(1, 1);
-      __builtin_amdgcn_s_barrier();
 
-      // ---- phase 3: quadrant (m1,n0); re-read B(n0) ----
-      read_b_half(cur, 0);
+      // ---- phase 3: MFMA (m1,n0); pre-read tile t+1's (m0)/(n0) ----
+      if (pf1) read_a_half(cur ^ 1, 0, afA);   // afA free since phase 1
+      if (pf2) { stage_a_round(cur, kt + 2, 1); stage_a_round(cur, kt + 2, 3); }
       __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma_quadrant(1, 0);
+      mfma_quadrant(afB, bfA, 1, 0);           // inputs already drained
+      if (pf1) read_b_half(cur ^ 1, 0, bfA);   // after bfA's last use
       __builtin_amdgcn_s_barrier();
     }
   } else {
@@ -397,19 +386,17 @@ __global__ void gemm_bf16_kernel(const short* __restrict__ ag, const short* __re
     for (int kt = 0; kt < nkt; ++kt) {
       const int cur = kt & 1;
       if (kt + 1 < nkt) stage_guarded(cur ^ 1, kt + 1);
-      read_a_half(cur, 0);
-      read_b_half(cur, 0);
+      read_a_half(cur, 0, afA);
+      read_b_half(cur, 0, bfA);
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma_quadrant(0, 0);
-      read_b_half(cur, 1);
+      mfma_quadrant(afA, bfA, 0, 0);
+      read_b_half(cur, 1, bfB);
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma_quadrant(0, 1);
-      read_a_half(cur, 1);
+      mfma_quadrant(afA, bfB, 0, 1);
+      read_a_half(cur, 1, afB);
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma_quadrant(1, 1);
-      read_b_half(cur, 0);
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma_quadrant(1, 0);
+      mfma_quadrant(afB, bfB, 1, 1);
+      mfma_quadrant(afB, bfA, 1, 0);
       asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
       __syncthreads();
     }
@@ -463,8 +450,10 @@ void gemm_bf16_ex(const void* a, const void* b, void* c, const void* bias,
   const int mi = (int)(m / 256), ni = (int)(n / 256);      // interior tiles
   const int gm = (int)((m + 255) / 256), gn = (int)((n + 255) / 256);
   const bool k_ok = (k % 64 == 0);
-  const bool has_fast = k_ok && mi > 0 && ni > 0;
-  const bool has_edge = !k_ok || mi < gm || ni < gn;
+  // the 8-phase pipeline is NT-only; NN/TN run the guarded kernel (the
+  // autotune table keeps them off the hot path where hipBLASLt wins)
+  const bool has_fast = k_ok && mi > 0 && ni > 0 && layout == LAY_NT;
+  const bool has_edge = !has_fast || mi < gm || ni < gn;
   dim3 blk(512);
 #define LAUNCH(L, E, AC)                                                       \
   do {                                                                         \

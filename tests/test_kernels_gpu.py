@@ -502,12 +502,12 @@ def test_gemm_dgelu_epilogue():
     torch.manual_seed(8)
     C = _ext.get_ext()
     dy = _bf(torch.randn(512, 512, device=DEV))
-    w = _bf(torch.randn(768, 512, device=DEV) * 0.05)   # dgrad B-operand
+    w = _bf(torch.randn(768, 512, device=DEV) * 0.05)   # NT B-operand [n,k]
     z = _bf(torch.randn(512, 768, device=DEV))          # saved pre-act
-    out = C.gemm_bf16_ex(dy, w.t().contiguous(), 0, 3, None, z)[0]
+    out = C.gemm_bf16_ex(dy, w, 0, 3, None, z)[0]       # [512, 768]
     zf = z.float().requires_grad_(True)
-    torch.nn.functional.gelu(zf).backward(dy.float() @ w.float())
-    # out = (dy @ w) * gelu'(z)
+    torch.nn.functional.gelu(zf).backward(dy.float() @ w.float().t())
+    # out = (dy @ w^T) * gelu'(z)
     _gemm_rel_ok(out, zf.grad, tol=5e-2)
 
 
