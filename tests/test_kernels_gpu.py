@@ -118,7 +118,7 @@ def test_flash_attention_gpu(causal, shape):
     v = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16).requires_grad_(True)
     scale = 1.0 / math.sqrt(d)
     from paddle_amd.ops.functional import _FlashAttn, _sdpa_ref, _sdpa_ref_bwd
-    o = _FlashAttn.apply(q, k, v, scale, causal)
+    o, _lse = _FlashAttn.apply(q, k, v, scale, causal)
     ref_o, ref_lse = _sdpa_ref(q.detach().float(), k.detach().float(),
                                v.detach().float(), scale, causal)
     _assert_close_bf16(o, ref_o, atol=3e-2, rtol=3e-2)
@@ -139,7 +139,7 @@ def test_flash_attention_gqa_fwd():
     v = torch.randn(b, hkv, s, d, device=DEV, dtype=torch.bfloat16)
     from paddle_amd.ops.functional import _FlashAttn, _sdpa_ref
     with torch.no_grad():
-        o = _FlashAttn.apply(q, k, v, 1.0 / math.sqrt(d), True)
+        o, _lse = _FlashAttn.apply(q, k, v, 1.0 / math.sqrt(d), True)
     ref_o, _ = _sdpa_ref(q.float(), k.float(), v.float(), 1.0 / math.sqrt(d), True)
     _assert_close_bf16(o, ref_o, atol=3e-2, rtol=3e-2)
 
@@ -347,7 +347,7 @@ def test_flash_attention_cross_gpu(sq, skv):
     v = torch.randn(b, h, skv, d, device=DEV, dtype=torch.bfloat16).requires_grad_(True)
     scale = 1.0 / math.sqrt(d)
     from paddle_amd.ops.functional import _FlashAttn, _sdpa_ref, _sdpa_ref_bwd
-    o = _FlashAttn.apply(q, k, v, scale, False)
+    o, _lse = _FlashAttn.apply(q, k, v, scale, False)
     ref_o, ref_lse = _sdpa_ref(q.detach().float(), k.detach().float(),
                                v.detach().float(), scale, False)
     _assert_close_bf16(o, ref_o, atol=3e-2, rtol=3e-2)
@@ -394,7 +394,7 @@ def test_flash_attention_gqa_bwd_gpu():
     v = torch.randn(b, hkv, s, d, device=DEV, dtype=torch.bfloat16).requires_grad_(True)
     scale = 1.0 / math.sqrt(d)
     from paddle_amd.ops.functional import _FlashAttn
-    o = _FlashAttn.apply(q, k, v, scale, True)
+    o, _lse = _FlashAttn.apply(q, k, v, scale, True)
     g = torch.randn_like(o)
     o.backward(g)
     rep = hq // hkv
