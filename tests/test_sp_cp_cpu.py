@@ -83,3 +83,53 @@ def test_ulysses_attention_matches_single():
         assert q2.grad is not None
         print("rank", r, "ulysses ok")
     """)
+
+
+def test_ring_attention_matches_single():
+    run_dist("""
+        import math
+        import torch
+        import paddle_amd as paddle
+        from paddle_amd.distributed import fleet
+        strategy = fleet.DistributedStrategy()
+        strategy.hybrid_configs = {"dp_degree": 1, "mp_degree": 2, "pp_degree": 1,
+                                   "sharding_degree": 1}
+        fleet.init(is_collective=True, strategy=strategy)
+        from paddle_amd.distributed.fleet.context_parallel import (
+            ring_attention, zigzag_split, zigzag_merge)
+        from paddle_amd.ops.functional import flash_attention
+        g = fleet.get_hybrid_communicate_group().get_model_parallel_group()
+        r, w = g.rank, g.nranks
+        torch.manual_seed(7)
+        b, s, h, d = 2, 32, 3, 16
+        q = torch.randn(b, s, h, d)
+        k = torch.randn(b, s, h, d)
+        v = torch.randn(b, s, h, d)
+        for causal in (True, False):
+            # single-process reference fwd+bwd
+            qr = q.clone().requires_grad_(True)
+            kr = k.clone().requires_grad_(True)
+            vr = v.clone().requires_grad_(True)
+            ref, _ = flash_attention(qr, kr, vr, causal=causal)
+            gout = torch.randn_like(ref)
+            ref.backward(gout)
+            # ring on zigzag shards
+            ql = zigzag_split(q, w, r).requires_grad_(True)
+            kl = zigzag_split(k, w, r).requires_grad_(True)
+            vl = zigzag_split(v, w, r).requires_grad_(True)
+            out, lse = ring_attention(ql, kl, vl, cp_group=g, causal=causal)
+            out.backward(zigzag_split(gout, w, r))
+            ref_l = zigzag_split(ref.detach(), w, r)
+            assert torch.allclose(out, ref_l, atol=2e-3, rtol=2e-3), \\
+                (causal, (out - ref_l).abs().max())
+            for got, refg in ((ql.grad, qr.grad), (kl.grad, kr.grad),
+                              (vl.grad, vr.grad)):
+                want = zigzag_split(refg, w, r)
+                assert torch.allclose(got, want, atol=5e-3, rtol=5e-3), \\
+                    (causal, (got - want).abs().max())
+        # zigzag_merge reassembles
+        full = torch.arange(2 * w * 3.0).reshape(1, 2 * w * 3, 1, 1)
+        shards = [zigzag_split(full, w, i) for i in range(w)]
+        assert torch.allclose(zigzag_merge(shards, w), full)
+        print("rank", r, "ring ok")
+    """)
