@@ -260,9 +260,24 @@ static void fa_strides(const Tensor& t, int64_t* out) {
   TORCH_CHECK(out[2] % 8 == 0, "flash_attn: seq stride must keep 16B alignment");
 }
 
+static void fa_mask_strides(const Tensor& m, int64_t b, int64_t h, int64_t sq,
+                            int64_t skv, int64_t* ms) {
+  TORCH_CHECK(m.dim() == 4 && m.size(3) == skv && m.size(2) == sq &&
+              (m.size(1) == 1 || m.size(1) == h) &&
+              (m.size(0) == 1 || m.size(0) == b),
+              "flash_attn mask must be [B|1, H|1, Sq, Skv]");
+  TORCH_CHECK(m.stride(3) == 1 && m.scalar_type() == torch::kBFloat16,
+              "flash_attn mask: bf16 with contiguous last dim");
+  ms[0] = m.size(0) == 1 ? 0 : m.stride(0);
+  ms[1] = m.size(1) == 1 ? 0 : m.stride(1);
+  ms[2] = m.stride(2);
+}
+
 std::vector<Tensor> flash_attn_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
                                    c10::optional<Tensor> o_out,
-                                   double scale, bool causal) {
+                                   double scale, bool causal,
+                                   c10::optional<Tensor> mask,
+                                   double pdrop, int64_t seed, int64_t offset) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16, "flash_attn: bf16 GPU only");
   TORCH_CHECK(q.dim() == 4, "flash_attn expects [B,H,S,D]");
   int64_t b = q.size(0), h = q.size(1), sq = q.size(2), d = q.size(3);
@@ -276,9 +291,16 @@ std::vector<Tensor> flash_attn_fwd(const Tensor& q, const Tensor& k, const Tenso
   auto o = o_out.has_value() ? *o_out : torch::empty({b, h, sq, d}, q.options());
   fa_strides(o, os);
   auto lse = torch::empty({b, h, sq}, q.options().dtype(torch::kFloat));
+  int64_t ms[3] = {0, 0, 0};
+  const void* mp = nullptr;
+  if (mask.has_value()) {
+    fa_mask_strides(*mask, b, h, sq, skv, ms);
+    mp = mask->const_data_ptr();
+  }
   pa::flash_attn_fwd32(q.const_data_ptr(), k.const_data_ptr(), v.const_data_ptr(),
                        o.mutable_data_ptr(), lse.mutable_data_ptr<float>(), b, h,
                        hkv, sq, skv, d, (float)scale, causal, qs, ks, os,
+                       mp, ms, (float)pdrop, (uint64_t)seed, (uint64_t)offset,
                        cur_stream());
   return {o, lse};
 }
@@ -288,7 +310,9 @@ std::vector<Tensor> flash_attn_bwd(const Tensor& dout, const Tensor& q, const Te
                                    c10::optional<Tensor> dq_out,
                                    c10::optional<Tensor> dk_out,
                                    c10::optional<Tensor> dv_out,
-                                   double scale, bool causal) {
+                                   double scale, bool causal,
+                                   c10::optional<Tensor> mask,
+                                   double pdrop, int64_t seed, int64_t offset) {
   int64_t b = q.size(0), h = q.size(1), sq = q.size(2), d = q.size(3);
   int64_t hkv = k.size(1), skv = k.size(2);
   TORCH_CHECK(hkv == h, "flash_attn_bwd kernel requires hkv==h (expand KV upstream)");
@@ -303,13 +327,29 @@ std::vector<Tensor> flash_attn_bwd(const Tensor& dout, const Tensor& q, const Te
   TORCH_CHECK(ks[0] == vs[0] && ks[1] == vs[1] && ks[2] == vs[2]);
   TORCH_CHECK(dks[0] == dvs[0] && dks[1] == dvs[1] && dks[2] == dvs[2],
               "dk/dv must share strides");
+  int64_t ms[3] = {0, 0, 0};
+  const void* mp = nullptr;
+  if (mask.has_value()) {
+    fa_mask_strides(*mask, b, h, sq, skv, ms);
+    mp = mask->const_data_ptr();
+  }
   pa::flash_attn_bwd(dout.const_data_ptr(), q.const_data_ptr(), k.const_data_ptr(),
                      v.const_data_ptr(), o.const_data_ptr(), lse.const_data_ptr<float>(),
                      dq.mutable_data_ptr(), dk.mutable_data_ptr(), dv.mutable_data_ptr(),
                      delta.mutable_data_ptr<float>(), b, h, hkv, sq, skv, d,
                      (float)scale, causal, qs, ks, dos, os, dqs, dks,
+                     mp, ms, (float)pdrop, (uint64_t)seed, (uint64_t)offset,
                      cur_stream());
   return {dq, dk, dv};
+}
+
+Tensor fa_dropout_mask(int64_t b, int64_t h, int64_t sq, int64_t skv,
+                       double p, int64_t seed, int64_t offset) {
+  auto out = torch::empty({b, h, sq, skv},
+                          torch::dtype(torch::kUInt8).device(torch::kCUDA));
+  pa::fa_dropout_mask(out.mutable_data_ptr(), out.numel(), (float)p,
+                      (uint64_t)seed, (uint64_t)offset, cur_stream());
+  return out;
 }
 
 // ---- dropout_add ----------------------------------------------------------
@@ -501,8 +541,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lt_epilogue_probe", &pa_lt::lt_epilogue_probe);
   m.def("weight_only_gemv", &weight_only_gemv);
   m.def("l2norm_sq", &l2norm_sq);
-  m.def("flash_attn_fwd", &flash_attn_fwd);
-  m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("o") = c10::nullopt, py::arg("scale"),
+        py::arg("causal"), py::arg("mask") = c10::nullopt,
+        py::arg("pdrop") = 0.0, py::arg("seed") = 0, py::arg("offset") = 0);
+  m.def("fa_dropout_mask", &fa_dropout_mask);
+  m.def("flash_attn_bwd", &flash_attn_bwd, py::arg("dout"), py::arg("q"),
+        py::arg("k"), py::arg("v"), py::arg("o"), py::arg("lse"),
+        py::arg("dq") = c10::nullopt, py::arg("dk") = c10::nullopt,
+        py::arg("dv") = c10::nullopt, py::arg("scale"), py::arg("causal"),
+        py::arg("mask") = c10::nullopt, py::arg("pdrop") = 0.0,
+        py::arg("seed") = 0, py::arg("offset") = 0);
   m.def("dropout_add_fwd", &dropout_add_fwd);
   m.def("dropout_add_bwd", &dropout_add_bwd);
   m.def("embedding_fwd", &embedding_fwd);

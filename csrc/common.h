@@ -119,3 +119,40 @@ inline int elementwise_grid(long long n_blocks_needed) {
   long long cap = 256LL * 8;  // 256 CUs x 8 blocks
   return (int)(n_blocks_needed < cap ? n_blocks_needed : cap);
 }
+
+// ---------------------------------------------------------------------------
+// Philox4x32-10 counter RNG (reference funcs/dropout_impl.cu.h:129 keeps
+// (seed, offset) pairs so recompute replays the same mask).  One call
+// yields 4 uint32 lanes for 4 consecutive elements.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(4))) unsigned int uintx4;
+
+__device__ __forceinline__ uintx4 philox10(unsigned long long seed,
+                                           unsigned long long ctr) {
+  unsigned key0 = (unsigned)seed, key1 = (unsigned)(seed >> 32);
+  unsigned c0 = (unsigned)ctr, c1 = (unsigned)(ctr >> 32), c2 = 0, c3 = 0;
+#pragma unroll
+  for (int i = 0; i < 10; ++i) {
+    unsigned hi0 = __umulhi(0xD2511F53u, c0), lo0 = 0xD2511F53u * c0;
+    unsigned hi1 = __umulhi(0xCD9E8D57u, c2), lo1 = 0xCD9E8D57u * c2;
+    c0 = hi1 ^ c1 ^ key0; c1 = lo1;
+    c2 = hi0 ^ c3 ^ key1; c3 = lo0;
+    key0 += 0x9E3779B9u; key1 += 0xBB67AE85u;
+  }
+  uintx4 r = {c0, c1, c2, c3};
+  return r;
+}
+
+// keep-decision for attention dropout at element (q_abs, kv_abs) of head
+// slot bh: the counter is the global element index / 4 (4 consecutive kv
+// share one philox call); threshold compare in 24-bit space.
+__device__ __forceinline__ bool fa_keep(unsigned long long seed,
+                                        unsigned long long offset,
+                                        long long bh, int sq_, long long skv_,
+                                        int q_abs, long long kv_abs,
+                                        unsigned thr24) {
+  long long elem = ((bh * sq_ + q_abs) * skv_ + kv_abs);
+  uintx4 r = philox10(seed, offset + (unsigned long long)(elem >> 2));
+  unsigned v = r[elem & 3];
+  return (v >> 8) >= thr24;
+}

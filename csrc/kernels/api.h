@@ -86,23 +86,31 @@ void flash_attn_fwd(const void* q, const void* k, const void* v, void* o,
                     const int64_t* qs, const int64_t* ks, const int64_t* os,
                     hipStream_t s);
 // 32x32-MFMA forward variant (same contract)
+// mask: additive bf16 [B, 1|H, Sq, Skv] (ms = {m_sb, m_sh, m_sq} strides,
+// null = none); pdrop + (seed, offset): Philox attention dropout
+// (dropout_impl.cu.h:129 seed/offset semantics, recompute-deterministic)
 void flash_attn_fwd32(const void* q, const void* k, const void* v, void* o,
                       float* lse, int64_t b, int64_t h, int64_t hkv, int64_t sq,
                       int64_t skv, int64_t dh, float scale, bool causal,
                       const int64_t* qs, const int64_t* ks, const int64_t* os,
-                      hipStream_t s);
+                      const void* mask, const int64_t* ms, float pdrop,
+                      uint64_t seed, uint64_t offset, hipStream_t s);
 void flash_attn_bwd_dq32(const void* dout, const void* q, const void* k,
                          const void* v, const float* lse, const float* delta,
                          void* dq, int64_t b, int64_t h, int64_t sq, int64_t skv,
                          int64_t dh, float scale, bool causal,
                          const int64_t* qs, const int64_t* ks, const int64_t* dos,
-                         const int64_t* dqs, hipStream_t s);
+                         const int64_t* dqs, const void* mask, const int64_t* ms,
+                         float pdrop, uint64_t seed, uint64_t offset,
+                         hipStream_t s);
 void flash_attn_bwd_dkv32(const void* dout, const void* q, const void* k,
                           const void* v, const float* lse, const float* delta,
                           void* dk, void* dv, int64_t b, int64_t h, int64_t sq,
                           int64_t skv, int64_t dh, float scale, bool causal,
                           const int64_t* qs, const int64_t* ks, const int64_t* dos,
-                          const int64_t* dks, hipStream_t s);
+                          const int64_t* dks, const void* mask, const int64_t* ms,
+                          float pdrop, uint64_t seed, uint64_t offset,
+                          hipStream_t s);
 void flash_attn_bwd(const void* dout, const void* q, const void* k,
                     const void* v, const void* o, const float* lse,
                     void* dq, void* dk, void* dv, float* delta, int64_t b,
@@ -110,7 +118,8 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
                     int64_t dh, float scale, bool causal,
                     const int64_t* qs, const int64_t* ks, const int64_t* dos,
                     const int64_t* os, const int64_t* dqs, const int64_t* dks,
-                    hipStream_t s);
+                    const void* mask, const int64_t* ms, float pdrop,
+                    uint64_t seed, uint64_t offset, hipStream_t s);
 
 // ---- dropout + residual add ------------------------------------------------
 // y = dropout(x, p) + residual; mask stored as uint8 per element
@@ -157,5 +166,9 @@ void gemm_bf16_ex(const void* a, const void* b, void* c, const void* bias,
 void weight_only_gemv(const void* x, const void* wq, const float* scale,
                       const void* bias, void* out, int64_t m, int64_t n,
                       int64_t k, int dtype, hipStream_t s);
+
+// materialize the dropout keep-mask (debug/tests): out uint8 [total]
+void fa_dropout_mask(void* out, int64_t total, float p, uint64_t seed,
+                     uint64_t offset, hipStream_t s);
 
 }  // namespace pa

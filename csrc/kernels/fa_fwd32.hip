@@ -42,7 +42,7 @@ __device__ __forceinline__ int pack_bf2(float a, float b) {
   return (int)(lo | (hi << 16));
 }
 
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool MASKED, bool DROPOUT>
 __launch_bounds__(256, 2)
 __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __restrict__ kg,
                                 const short* __restrict__ vg, short* __restrict__ og,
@@ -50,7 +50,11 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
                                 int Sq, int Skv, float scale,
                                 long long q_sb, long long q_sh, long long q_ss,
                                 long long k_sb, long long k_sh, long long k_ss,
-                                long long o_sb, long long o_sh, long long o_ss) {
+                                long long o_sb, long long o_sh, long long o_ss,
+                                const short* __restrict__ maskg, long long m_sb,
+                                long long m_sh, long long m_sq,
+                                float pdrop, unsigned long long rseed,
+                                unsigned long long roffset) {
   constexpr int NW = 4;            // waves per block, 32 q rows each
   constexpr int NT = NW * 64;
   constexpr int QB = NW * 32;      // 128 q rows per block
@@ -181,6 +185,26 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
 
     // ---- mask + scale + per-lane online softmax ---------------------------
     const int q_abs = q0w + l32;
+    if (MASKED && q_abs < Sq) {
+      // additive mask [B, 1|H, Sq, Skv]; reg groups are 4 consecutive kv
+      const short* mrow = maskg + (long long)b * m_sb + (long long)h * m_sh
+                        + (long long)q_abs * m_sq;
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+          long long kvb = kv0 + 32 * t + 8 * g + 4 * hi;
+          if (kvb + 3 < Skv) {
+            shortx4 mv = *reinterpret_cast<const shortx4*>(mrow + kvb);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) st[t][4 * g + j] += bf2f(mv[j]);
+          } else {
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+              if (kvb + j < Skv) st[t][4 * g + j] += bf2f(mrow[kvb + j]);
+          }
+        }
+    }
     float mx = -INFINITY;
     const bool boundary = (kv0 + KB > Skv) || (CAUSAL && kv0 + KB > q_abs);
     if (boundary) {
@@ -233,6 +257,24 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
       }
     psum += __shfl_xor(psum, 32, 64);
     l_run += psum;
+    if (DROPOUT) {
+      // Philox keep-mask on P (normalizer above uses undropped P)
+      const float inv_keep = 1.f / (1.f - pdrop);
+      const unsigned thr24 = (unsigned)(pdrop * 16777216.f);
+      const long long bh_ll = (long long)b * H + h;
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+          long long kvb = kv0 + 32 * t + 8 * g + 4 * hi;
+          long long elem = (bh_ll * Sq + q_abs) * (long long)Skv + kvb;
+          uintx4 rv = philox10(rseed, roffset + (unsigned long long)(elem >> 2));
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            st[t][4 * g + j] = ((rv[j] >> 8) >= thr24)
+                                   ? st[t][4 * g + j] * inv_keep : 0.f;
+        }
+    }
 
     // ---- O += P V : assemble P A-frags in-register ------------------------
 #pragma unroll
@@ -295,18 +337,32 @@ void flash_attn_fwd32(const void* q, const void* k, const void* v, void* o,
                       float* lse, int64_t b, int64_t h, int64_t hkv, int64_t sq,
                       int64_t skv, int64_t dh, float scale, bool causal,
                       const int64_t* qs, const int64_t* ks, const int64_t* os,
-                      hipStream_t s) {
+                      const void* mask, const int64_t* ms, float pdrop,
+                      uint64_t seed, uint64_t offset, hipStream_t s) {
   dim3 grid((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
   dim3 blk(256);
+  const bool masked = mask != nullptr;
+  const bool dropped = pdrop > 0.f;
+  static const int64_t zs[3] = {0, 0, 0};
+  if (!ms) ms = zs;
+#define FAF1(D, C, M, P)                                                        \
+  hipLaunchKernelGGL((fa_fwd32_kernel<D, C, M, P>), grid, blk, 0, s,            \
+                     (const short*)q, (const short*)k, (const short*)v,         \
+                     (short*)o, lse, (int)b, (int)h, (int)hkv, (int)sq,         \
+                     (int)skv, scale, qs[0], qs[1], qs[2],                      \
+                     ks[0], ks[1], ks[2], os[0], os[1], os[2],                  \
+                     (const short*)mask, ms[0], ms[1], ms[2], pdrop, seed, offset)
 #define FAF(D, C)                                                               \
-  hipLaunchKernelGGL((fa_fwd32_kernel<D, C>), grid, blk, 0, s, (const short*)q, \
-                     (const short*)k, (const short*)v, (short*)o, lse,          \
-                     (int)b, (int)h, (int)hkv, (int)sq, (int)skv, scale,        \
-                     qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],                  \
-                     os[0], os[1], os[2])
+  do {                                                                          \
+    if (masked && dropped) FAF1(D, C, true, true);                              \
+    else if (masked)       FAF1(D, C, true, false);                             \
+    else if (dropped)      FAF1(D, C, false, true);                             \
+    else                   FAF1(D, C, false, false);                            \
+  } while (0)
   if (dh == 128) { if (causal) FAF(128, true); else FAF(128, false); }
   else           { if (causal) FAF(64, true);  else FAF(64, false); }
 #undef FAF
+#undef FAF1
 }
 
 
@@ -323,7 +379,7 @@ namespace {
 constexpr int DQ_NW = 4;
 }
 
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool MASKED, bool DROPOUT>
 __launch_bounds__(256, 2)
 __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* __restrict__ qg,
                                    const short* __restrict__ kg, const short* __restrict__ vg,
@@ -333,7 +389,11 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
                                    long long q_sb, long long q_sh, long long q_ss,
                                    long long k_sb, long long k_sh, long long k_ss,
                                    long long do_sb, long long do_sh, long long do_ss,
-                                   long long dq_sb, long long dq_sh, long long dq_ss) {
+                                   long long dq_sb, long long dq_sh, long long dq_ss,
+                                   const short* __restrict__ maskg, long long m_sb,
+                                   long long m_sh, long long m_sq,
+                                   float pdrop, unsigned long long rseed,
+                                   unsigned long long roffset) {
   constexpr int NW = DQ_NW;
   constexpr int NT = NW * 64;
   constexpr int QB = NW * 32;       // 128 q rows / block
@@ -469,10 +529,44 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
       }
     }
 
-    // ---- dS^T = P^T (dP^T - delta) * scale --------------------------------
-    const bool bnd = (kv0 + KB > Skv) || (CAUSAL && kv0 + KB > q_abs);
+    // ---- dS^T = P^T (dP^T o dropout - delta) * scale ----------------------
+    if (MASKED && q_abs < Sq) {
+      const short* mrow = maskg + (long long)b * m_sb + (long long)h * m_sh
+                        + (long long)q_abs * m_sq;
 #pragma unroll
-    for (int t = 0; t < 2; ++t)
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+          long long kvb = kv0 + 32 * t + 8 * g + 4 * hi;
+          if (kvb + 3 < Skv) {
+            shortx4 mv = *reinterpret_cast<const shortx4*>(mrow + kvb);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) st[t][4 * g + j] += bf2f(mv[j]);
+          } else {
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+              if (kvb + j < Skv) st[t][4 * g + j] += bf2f(mrow[kvb + j]);
+          }
+        }
+    }
+    const bool bnd = (kv0 + KB > Skv) || (CAUSAL && kv0 + KB > q_abs);
+    const float inv_keep = DROPOUT ? 1.f / (1.f - pdrop) : 1.f;
+    const unsigned thr24 = DROPOUT ? (unsigned)(pdrop * 16777216.f) : 0u;
+    const long long bh_ll = (long long)b * H + h;
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      if (DROPOUT) {
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+          long long kvb = kv0 + 32 * t + 8 * g + 4 * hi;
+          long long elem = (bh_ll * Sq + q_abs) * (long long)Skv + kvb;
+          uintx4 rv = philox10(rseed, roffset + (unsigned long long)(elem >> 2));
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            dp[t][4 * g + j] = ((rv[j] >> 8) >= thr24)
+                                   ? dp[t][4 * g + j] * inv_keep : 0.f;
+        }
+      }
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         float p = __expf(st[t][r] - lse_v);   // Q pre-scaled at load
@@ -482,6 +576,7 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
         }
         st[t][r] = p * (dp[t][r] - delta_v) * scale;
       }
+    }
 
     // ---- dQ += dS K (A-frags via static packs + permlane exchange) --------
 #pragma unroll
@@ -534,19 +629,34 @@ void flash_attn_bwd_dq32(const void* dout, const void* q, const void* k,
                          void* dq, int64_t b, int64_t h, int64_t sq, int64_t skv,
                          int64_t dh, float scale, bool causal,
                          const int64_t* qs, const int64_t* ks, const int64_t* dos,
-                         const int64_t* dqs, hipStream_t s) {
+                         const int64_t* dqs, const void* mask, const int64_t* ms,
+                         float pdrop, uint64_t seed, uint64_t offset,
+                         hipStream_t s) {
   dim3 grid((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
   dim3 blk(256);
-#define FDQ(D, C)                                                               \
-  hipLaunchKernelGGL((fa_bwd_dq32_kernel<D, C>), grid, blk, 0, s,               \
+  const bool masked = mask != nullptr;
+  const bool dropped = pdrop > 0.f;
+  static const int64_t zs[3] = {0, 0, 0};
+  if (!ms) ms = zs;
+#define FDQ1(D, C, M, P)                                                        \
+  hipLaunchKernelGGL((fa_bwd_dq32_kernel<D, C, M, P>), grid, blk, 0, s,         \
                      (const short*)dout, (const short*)q, (const short*)k,      \
                      (const short*)v, lse, delta, (short*)dq, (int)b, (int)h,   \
                      (int)sq, (int)skv, scale, qs[0], qs[1], qs[2],             \
                      ks[0], ks[1], ks[2], dos[0], dos[1], dos[2],               \
-                     dqs[0], dqs[1], dqs[2])
+                     dqs[0], dqs[1], dqs[2], (const short*)mask,                \
+                     ms[0], ms[1], ms[2], pdrop, seed, offset)
+#define FDQ(D, C)                                                               \
+  do {                                                                          \
+    if (masked && dropped) FDQ1(D, C, true, true);                              \
+    else if (masked)       FDQ1(D, C, true, false);                             \
+    else if (dropped)      FDQ1(D, C, false, true);                             \
+    else                   FDQ1(D, C, false, false);                            \
+  } while (0)
   if (dh == 128) { if (causal) FDQ(128, true); else FDQ(128, false); }
   else           { if (causal) FDQ(64, true);  else FDQ(64, false); }
 #undef FDQ
+#undef FDQ1
 }
 
 // ---------------------------------------------------------------------------
@@ -563,7 +673,7 @@ void flash_attn_bwd_dq32(const void* dout, const void* q, const void* k,
 // transposed operand (dO^T or Q^T) is staged in LDS, double-buffered.
 // Outputs are accumulated transposed and written as packed 8 B stores.
 // ---------------------------------------------------------------------------
-template <int D, bool CAUSAL, bool IS_DK>
+template <int D, bool CAUSAL, bool IS_DK, bool MASKED, bool DROPOUT>
 __launch_bounds__(256, 2)
 __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* __restrict__ qg,
                                     const short* __restrict__ kg, const short* __restrict__ vg,
@@ -573,7 +683,11 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
                                     long long q_sb, long long q_sh, long long q_ss,
                                     long long k_sb, long long k_sh, long long k_ss,
                                     long long do_sb, long long do_sh, long long do_ss,
-                                    long long dk_sb, long long dk_sh, long long dk_ss) {
+                                    long long dk_sb, long long dk_sh, long long dk_ss,
+                                    const short* __restrict__ maskg, long long m_sb,
+                                    long long m_sh, long long m_sq,
+                                    float pdrop, unsigned long long rseed,
+                                    unsigned long long roffset) {
   constexpr int NW = 4;
   constexpr int NT = NW * 64;
   constexpr int KVB = NW * 32;     // 128 kv rows / block
@@ -695,17 +809,30 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
     // broadcast loads straight from global (L2-hot after the dV kernel).
     const bool bnd = (q0 + QT > Sq) || (kv0w + 32 > Skv) ||
                      (CAUSAL && kv0w + 32 > q0);
+    const float inv_keep = DROPOUT ? 1.f / (1.f - pdrop) : 1.f;
+    const unsigned thr24 = DROPOUT ? (unsigned)(pdrop * 16777216.f) : 0u;
+    const long long bh_ll = (long long)b * H + h;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int rq = (r & 3) + 8 * (r >> 2) + 4 * hi;
       const int q_abs = q0 + rq;
-      float p = __expf(st[r] * scale - stat_s[cur][rq]);
+      float sc = st[r] * scale;
+      if (MASKED && q_abs < Sq && kv_lane < Skv)
+        sc += bf2f(maskg[(long long)b * m_sb + (long long)h * m_sh
+                         + (long long)q_abs * m_sq + kv_lane]);
+      float p = __expf(sc - stat_s[cur][rq]);
       if (bnd) {
         if (q_abs >= Sq || kv_lane >= Skv || (CAUSAL && kv_lane > q_abs)) p = 0.f;
       }
+      bool keep = true;
+      if (DROPOUT)
+        keep = fa_keep(rseed, roffset, bh_ll, Sq, Skv, q_abs, kv_lane, thr24);
       if (IS_DK) {
-        st[r] = p * (dp[r] - stat_s[cur][QT + rq]) * scale;
+        float dpe = dp[r];
+        if (DROPOUT) dpe = keep ? dpe * inv_keep : 0.f;
+        st[r] = p * (dpe - stat_s[cur][QT + rq]) * scale;
       } else {
+        if (DROPOUT) p = keep ? p * inv_keep : 0.f;
         st[r] = p;
       }
     }
@@ -759,27 +886,62 @@ void flash_attn_bwd_dkv32(const void* dout, const void* q, const void* k,
                           void* dk, void* dv, int64_t b, int64_t h, int64_t sq,
                           int64_t skv, int64_t dh, float scale, bool causal,
                           const int64_t* qs, const int64_t* ks, const int64_t* dos,
-                          const int64_t* dks, hipStream_t s) {
+                          const int64_t* dks, const void* mask, const int64_t* ms,
+                          float pdrop, uint64_t seed, uint64_t offset,
+                          hipStream_t s) {
   dim3 grid((unsigned)cdiv((int)skv, 128), (unsigned)(b * h));
   dim3 blk(256);
+  const bool masked = mask != nullptr;
+  const bool dropped = pdrop > 0.f;
+  static const int64_t zs[3] = {0, 0, 0};
+  if (!ms) ms = zs;
+#define FDKV2(D, C, DK, M, P, OUT)                                             \
+  hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C, DK, M, P>), grid, blk, 0, s,   \
+                     (const short*)dout, (const short*)q, (const short*)k,     \
+                     (const short*)v, lse, delta, (short*)(OUT),               \
+                     (int)b, (int)h, (int)sq, (int)skv, scale,                 \
+                     qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],                 \
+                     dos[0], dos[1], dos[2], dks[0], dks[1], dks[2],           \
+                     (const short*)mask, ms[0], ms[1], ms[2], pdrop, seed, offset)
+#define FDKV1(D, C, M, P)                                                      \
+  do { FDKV2(D, C, false, M, P, dv); FDKV2(D, C, true, M, P, dk); } while (0)
 #define FDKV(D, C)                                                             \
   do {                                                                         \
-    hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C, false>), grid, blk, 0, s,    \
-                       (const short*)dout, (const short*)q, (const short*)k,   \
-                       (const short*)v, lse, delta, (short*)dv,                \
-                       (int)b, (int)h, (int)sq, (int)skv, scale,               \
-                       qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],               \
-                       dos[0], dos[1], dos[2], dks[0], dks[1], dks[2]);        \
-    hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C, true>), grid, blk, 0, s,     \
-                       (const short*)dout, (const short*)q, (const short*)k,   \
-                       (const short*)v, lse, delta, (short*)dk,                \
-                       (int)b, (int)h, (int)sq, (int)skv, scale,               \
-                       qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],               \
-                       dos[0], dos[1], dos[2], dks[0], dks[1], dks[2]);        \
+    if (masked && dropped) FDKV1(D, C, true, true);                            \
+    else if (masked)       FDKV1(D, C, true, false);                           \
+    else if (dropped)      FDKV1(D, C, false, true);                           \
+    else                   FDKV1(D, C, false, false);                          \
   } while (0)
   if (dh == 128) { if (causal) FDKV(128, true); else FDKV(128, false); }
   else           { if (causal) FDKV(64, true);  else FDKV(64, false); }
 #undef FDKV
+#undef FDKV1
+#undef FDKV2
+}
+
+// debug/test utility: materialize the attention-dropout keep mask the FA
+// kernels derive from (seed, offset) -- lets tests compare fwd/bwd against
+// an exact CPU oracle using the same mask.
+__global__ void fa_dropout_mask_kernel(unsigned char* __restrict__ out,
+                                       float p, unsigned long long seed,
+                                       unsigned long long offset,
+                                       long long total) {
+  long long i4 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long base = i4 * 4;
+  if (base >= total) return;
+  unsigned thr24 = (unsigned)(p * 16777216.f);
+  uintx4 rv = philox10(seed, offset + (unsigned long long)i4);
+#pragma unroll
+  for (int j = 0; j < 4; ++j)
+    if (base + j < total) out[base + j] = ((rv[j] >> 8) >= thr24) ? 1 : 0;
+}
+
+void fa_dropout_mask(void* out, int64_t total, float p, uint64_t seed,
+                     uint64_t offset, hipStream_t s) {
+  long long n4 = (total + 3) / 4;
+  dim3 g((unsigned)hmin<long long>((n4 + 255) / 256, 1 << 30));
+  hipLaunchKernelGGL(fa_dropout_mask_kernel, g, dim3(256), 0, s,
+                     (unsigned char*)out, p, seed, offset, total);
 }
 
 }  // namespace pa

@@ -751,7 +751,8 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
                     int64_t dh, float scale, bool causal,
                     const int64_t* qs, const int64_t* ks, const int64_t* dos,
                     const int64_t* os, const int64_t* dqs, const int64_t* dks,
-                    hipStream_t s) {
+                    const void* mask, const int64_t* ms, float pdrop,
+                    uint64_t seed, uint64_t offset, hipStream_t s) {
   long long rows = b * h * sq;
   dim3 dgrid((unsigned)hmin<long long>(2048LL, (rows + 3) / 4));
   if (dh == 128)
@@ -769,10 +770,12 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
   // slower (combined/transposed-output: 146 TF bwd at occupancy 1;
   // occupancy-2 split with global A-frags: 95 TF; 16x16 baseline: 168 TF).
   // PA_FA_DKV32=1 opts into the 32x32 path for future experiments.
-  static const bool use_dkv16 = [] {
+  static const bool dkv16_env = [] {
     const char* e = getenv("PA_FA_DKV32");
     return !(e && e[0] == '1');
   }();
+  // the 16x16 dKV kernel has no mask/dropout path; route those to dkv32
+  const bool use_dkv16 = dkv16_env && mask == nullptr && pdrop <= 0.f;
 #define FAB(D, C)                                                              \
   do {                                                                         \
     if (use_dkv16)                                                             \
@@ -784,9 +787,11 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
                          dos[0], dos[1], dos[2], dks[0], dks[1], dks[2]);      \
     else                                                                       \
       flash_attn_bwd_dkv32(dout, q, k, v, lse, delta, dk, dv, b, h, sq, skv,   \
-                           dh, scale, causal, qs, ks, dos, dks, s);            \
+                           dh, scale, causal, qs, ks, dos, dks, mask, ms,      \
+                           pdrop, seed, offset, s);                            \
     flash_attn_bwd_dq32(dout, q, k, v, lse, delta, dq, b, h, sq, skv, dh,     \
-                        scale, causal, qs, ks, dos, dqs, s);                   \
+                        scale, causal, qs, ks, dos, dqs, mask, ms, pdrop,      \
+                        seed, offset, s);                                      \
   } while (0)
   if (dh == 128) { if (causal) FAB(128, true); else FAB(128, false); }
   else           { if (causal) FAB(64, true);  else FAB(64, false); }

@@ -89,7 +89,8 @@ class GPTAttention(nn.Layer):
         b, s, h = x.shape
         qkv = self.qkv_proj(x).reshape(b, s, 3, self.num_heads, self.head_dim)
         # packed zero-copy path: strided q/k/v views + in-place dqkv
-        out = hot.qkv_flash_attention(qkv, causal=True)
+        out = hot.qkv_flash_attention(qkv, causal=True, dropout=self.attn_dropout,
+                                      training=self.training)
         return self.out_proj(out)
 
 

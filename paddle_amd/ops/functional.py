@@ -349,19 +349,44 @@ def _fa_native_ok(q):
             and q.shape[-1] in (64, 128) and q.stride(-1) == 1)
 
 
+def _fa_seed_offset(fixed_seed_offset=None):
+    """Reference dropout RNG contract (funcs/dropout_impl.cu.h:129): a
+    (seed, offset) pair; fixed_seed_offset pins it (recompute replays the
+    torch CPU generator state, so the default path is replay-deterministic
+    under fleet.recompute as well)."""
+    if fixed_seed_offset is not None:
+        return int(fixed_seed_offset[0]), int(fixed_seed_offset[1])
+    seed = int(torch.initial_seed()) & 0x7FFFFFFFFFFFFFFF
+    offset = int(torch.randint(0, 2 ** 31, (1,)).item())
+    return seed, offset
+
+
 class _FlashAttn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, scale, causal):
+    def forward(ctx, q, k, v, scale, causal, mask=None, dropout=0.0,
+                seed=0, offset=0):
         # q,k,v: [B, H, S, D] bf16 (views allowed; D contiguous)
         if _fa_native_ok(q):
             C = _ext.get_ext()
-            o, lse = C.flash_attn_fwd(q, k, v, None, scale, causal)
+            mk = mask.to(torch.bfloat16) if mask is not None else None
+            o, lse = C.flash_attn_fwd(q, k, v, None, scale, causal, mk,
+                                      dropout, seed, offset)
             ctx.native = True
+            ctx.mask = mk
         else:
-            o, lse = _sdpa_ref(q, k, v, scale, causal)
+            dm = None
+            if dropout > 0.0:
+                g = torch.Generator(device="cpu").manual_seed(
+                    (seed * 1000003 + offset) & 0x7FFFFFFFFFFFFFFF)
+                dm = (torch.rand(q.shape[0], q.shape[1], q.shape[2], k.shape[2],
+                                 generator=g) >= dropout).to(q.device)
+            o, lse = _sdpa_ref(q, k, v, scale, causal, mask, dm, dropout)
             ctx.native = False
+            ctx.mask = mask
+            ctx.drop_mask = dm
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale, ctx.causal = scale, causal
+        ctx.dropout, ctx.seed, ctx.offset = dropout, seed, offset
         ctx.mark_non_differentiable(lse)
         return o, lse
 
@@ -372,22 +397,24 @@ class _FlashAttn(torch.autograd.Function):
             C = _ext.get_ext()
             if do.stride(-1) != 1:
                 do = do.contiguous()
+            kw = dict(scale=ctx.scale, causal=ctx.causal, mask=ctx.mask,
+                      pdrop=ctx.dropout, seed=ctx.seed, offset=ctx.offset)
             if k.shape[1] != q.shape[1]:  # GQA: expand for bwd kernel
                 rep = q.shape[1] // k.shape[1]
                 ke = k.repeat_interleave(rep, dim=1)
                 ve = v.repeat_interleave(rep, dim=1)
                 dq, dke, dve = C.flash_attn_bwd(do, q, ke.contiguous(), ve.contiguous(),
-                                                o, lse, None, None, None,
-                                                ctx.scale, ctx.causal)
+                                                o, lse, **kw)
                 hkv = k.shape[1]
                 dk = dke.view(k.shape[0], hkv, rep, k.shape[2], k.shape[3]).sum(2)
                 dv = dve.view(v.shape[0], hkv, rep, v.shape[2], v.shape[3]).sum(2)
             else:
-                dq, dk, dv = C.flash_attn_bwd(do, q, k, v, o, lse, None, None, None,
-                                              ctx.scale, ctx.causal)
+                dq, dk, dv = C.flash_attn_bwd(do, q, k, v, o, lse, **kw)
         else:
-            dq, dk, dv = _sdpa_ref_bwd(do, q, k, v, lse, ctx.scale, ctx.causal)
-        return dq, dk, dv, None, None
+            dq, dk, dv = _sdpa_ref_bwd(do, q, k, v, lse, ctx.scale, ctx.causal,
+                                       ctx.mask, getattr(ctx, "drop_mask", None),
+                                       ctx.dropout)
+        return dq, dk, dv, None, None, None, None, None, None
 
 
 class _QKVFlashAttn(torch.autograd.Function):
@@ -397,7 +424,7 @@ class _QKVFlashAttn(torch.autograd.Function):
     no transpose/cat kernels at all."""
 
     @staticmethod
-    def forward(ctx, qkv, scale, causal):
+    def forward(ctx, qkv, scale, causal, dropout=0.0, seed=0, offset=0):
         b, s, three, h, d = qkv.shape
         q = qkv[:, :, 0].permute(0, 2, 1, 3)  # [b,h,s,d] strided view
         k = qkv[:, :, 1].permute(0, 2, 1, 3)
@@ -406,16 +433,25 @@ class _QKVFlashAttn(torch.autograd.Function):
             C = _ext.get_ext()
             out = torch.empty(b, s, h, d, dtype=qkv.dtype, device=qkv.device)
             o_view = out.permute(0, 2, 1, 3)
-            _, lse = C.flash_attn_fwd(q, k, v, o_view, scale, causal)
+            _, lse = C.flash_attn_fwd(q, k, v, o_view, scale, causal, None,
+                                      dropout, seed, offset)
             ctx.native = True
             ctx.save_for_backward(qkv, out, lse)
+            ctx.drop_mask = None
         else:
+            dm = None
+            if dropout > 0.0:
+                g = torch.Generator(device="cpu").manual_seed(
+                    (seed * 1000003 + offset) & 0x7FFFFFFFFFFFFFFF)
+                dm = (torch.rand(b, h, s, s, generator=g) >= dropout).to(qkv.device)
             o, lse = _sdpa_ref(q.contiguous(), k.contiguous(), v.contiguous(),
-                               scale, causal)
+                               scale, causal, None, dm, dropout)
             out = o.permute(0, 2, 1, 3).reshape(b, s, h, d)
             ctx.native = False
             ctx.save_for_backward(qkv, out, lse)
+            ctx.drop_mask = dm
         ctx.scale, ctx.causal = scale, causal
+        ctx.dropout, ctx.seed, ctx.offset = dropout, seed, offset
         return out.reshape(b, s, h * d)
 
     @staticmethod
@@ -436,42 +472,56 @@ class _QKVFlashAttn(torch.autograd.Function):
             dk = dqkv[:, :, 1].permute(0, 2, 1, 3)
             dv = dqkv[:, :, 2].permute(0, 2, 1, 3)
             C.flash_attn_bwd(do_view, q, k, v, o_view, lse, dq, dk, dv,
-                             ctx.scale, ctx.causal)
+                             ctx.scale, ctx.causal, None, ctx.dropout,
+                             ctx.seed, ctx.offset)
         else:
             dq, dk, dv = _sdpa_ref_bwd(do_view.contiguous(), q.contiguous(),
                                        k.contiguous(), v.contiguous(), lse,
-                                       ctx.scale, ctx.causal)
+                                       ctx.scale, ctx.causal, None,
+                                       ctx.drop_mask, ctx.dropout)
             # each [b,h,s,d]; stack -> [3,b,h,s,d]; permute -> [b,s,3,h,d]
             dqkv = torch.stack([dq, dk, dv], dim=0).permute(1, 3, 0, 2, 4).contiguous()
-        return dqkv, None, None
+        return dqkv, None, None, None, None, None
 
 
-def qkv_flash_attention(qkv, scale=None, causal=True):
+def qkv_flash_attention(qkv, scale=None, causal=True, dropout=0.0,
+                        training=True):
     """qkv: [B, S, 3, H, D] packed (straight out of the fused QKV GEMM)."""
     if scale is None:
         scale = 1.0 / math.sqrt(qkv.shape[-1])
-    return _QKVFlashAttn.apply(qkv, scale, causal)
+    p = dropout if training else 0.0
+    seed, offset = _fa_seed_offset() if p > 0 else (0, 0)
+    return _QKVFlashAttn.apply(qkv, scale, causal, p, seed, offset)
 
 
-def _sdpa_ref(q, k, v, scale, causal):
-    # fp32 reference; returns (o, lse) with lse = logsumexp of scaled scores
+def _sdpa_ref(q, k, v, scale, causal, attn_mask=None, drop_mask=None,
+              dropout=0.0):
+    # fp32 reference; returns (o, lse) with lse = logsumexp of scaled scores.
+    # attn_mask: additive [B|1, H|1, Sq, Skv]; drop_mask: uint8 keep mask
+    # (same shape as scores) with 1/(1-p) rescale -- matches the HIP
+    # kernel's semantics (normalizer from undropped P).
     qf, kf, vf = q.float(), k.float(), v.float()
     if k.shape[1] != q.shape[1]:
         rep = q.shape[1] // k.shape[1]
         kf = kf.repeat_interleave(rep, dim=1)
         vf = vf.repeat_interleave(rep, dim=1)
     s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if attn_mask is not None:
+        s = s + attn_mask.float()
     if causal:
         sq, skv = s.shape[-2], s.shape[-1]
         mask = torch.ones(sq, skv, dtype=torch.bool, device=s.device).tril()
         s = s.masked_fill(~mask, float("-inf"))
     lse = torch.logsumexp(s, -1)
     p = torch.exp(s - lse.unsqueeze(-1))
+    if drop_mask is not None:
+        p = p * drop_mask.float() / (1.0 - dropout)
     o = torch.matmul(p, vf)
     return o.to(q.dtype), lse
 
 
-def _sdpa_ref_bwd(do, q, k, v, lse, scale, causal):
+def _sdpa_ref_bwd(do, q, k, v, lse, scale, causal, attn_mask=None,
+                  drop_mask=None, dropout=0.0):
     qf, kf, vf, dof = q.float(), k.float(), v.float(), do.float()
     rep = 1
     if k.shape[1] != q.shape[1]:
@@ -479,11 +529,29 @@ def _sdpa_ref_bwd(do, q, k, v, lse, scale, causal):
         kf = kf.repeat_interleave(rep, dim=1)
         vf = vf.repeat_interleave(rep, dim=1)
     s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if attn_mask is not None:
+        s = s + attn_mask.float()
     if causal:
         sq, skv = s.shape[-2], s.shape[-1]
         mask = torch.ones(sq, skv, dtype=torch.bool, device=s.device).tril()
         s = s.masked_fill(~mask, float("-inf"))
     p = torch.exp(s - lse.unsqueeze(-1).float())
+    if drop_mask is not None:
+        keep = drop_mask.float() / (1.0 - dropout)
+        pd = p * keep
+        dv = torch.matmul(pd.transpose(-1, -2), dof)
+        dp_raw = torch.matmul(dof, vf.transpose(-1, -2))
+        dp = dp_raw * keep
+        # delta = rowsum(dO*O) = rowsum(dPd * Pd)
+        delta = (dp_raw * pd).sum(-1, keepdim=True)
+        ds = p * (dp - delta) * scale
+        dq = torch.matmul(ds, kf)
+        dk = torch.matmul(ds.transpose(-1, -2), qf)
+        if rep > 1:
+            b, h, skv_, d = dk.shape
+            dk = dk.view(b, h // rep, rep, skv_, d).sum(2)
+            dv = dv.view(b, h // rep, rep, skv_, d).sum(2)
+        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype)
     dv = torch.matmul(p.transpose(-1, -2), dof)
     dp = torch.matmul(dof, vf.transpose(-1, -2))
     delta = (dp * p).sum(-1, keepdim=True)
@@ -498,16 +566,20 @@ def _sdpa_ref_bwd(do, q, k, v, lse, scale, causal):
 
 
 def flash_attention(q, k, v, dropout=0.0, causal=False, scale=None,
-                    return_softmax_lse=False, layout="bshd"):
+                    return_softmax_lse=False, layout="bshd", attn_mask=None,
+                    training=True, fixed_seed_offset=None, rng_name="",
+                    name=None):
     """paddle.nn.functional.flash_attention parity
-    (python/paddle/nn/functional/flash_attention.py:195).
+    (python/paddle/nn/functional/flash_attention.py:195; mask + Philox
+    dropout per flash_attn_kernel.cu:41 / dropout_impl.cu.h:129).
 
     layout "bshd": q [batch, seq, heads, head_dim] (paddle convention);
     internally computed as [b, h, s, d].
     """
-    assert dropout == 0.0, "attention dropout not yet supported in the HIP kernel"
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
+    p = dropout if training else 0.0
+    seed, offset = _fa_seed_offset(fixed_seed_offset) if p > 0 else (0, 0)
     if layout == "bshd":
         # strided views -- the kernel is stride-aware, no copies
         qt = q.transpose(1, 2)
@@ -515,7 +587,8 @@ def flash_attention(q, k, v, dropout=0.0, causal=False, scale=None,
         vt = v.transpose(1, 2)
     else:
         qt, kt, vt = q, k, v
-    o, lse = _FlashAttn.apply(qt, kt, vt, scale, causal)
+    o, lse = _FlashAttn.apply(qt, kt, vt, scale, causal, attn_mask, p,
+                              seed, offset)
     if layout == "bshd":
         o = o.transpose(1, 2)
     if return_softmax_lse:
@@ -525,10 +598,18 @@ def flash_attention(q, k, v, dropout=0.0, causal=False, scale=None,
 
 def scaled_dot_product_attention(query, key, value, attn_mask=None,
                                  dropout_p=0.0, is_causal=False, training=True):
-    """paddle.nn.functional.scaled_dot_product_attention parity (bshd)."""
-    assert attn_mask is None, "mask path not yet fused; use causal"
-    out, _ = flash_attention(query, key, value, dropout=dropout_p if training else 0.0,
-                             causal=is_causal)
+    """paddle.nn.functional.scaled_dot_product_attention parity (bshd);
+    boolean masks become additive -inf masks (kernel takes additive bf16)."""
+    if attn_mask is not None and attn_mask.dtype == torch.bool:
+        attn_mask = torch.zeros_like(attn_mask, dtype=query.dtype).masked_fill(
+            ~attn_mask, float("-inf"))
+    if attn_mask is not None and attn_mask.dim() == 3:
+        attn_mask = attn_mask.unsqueeze(1)
+    if attn_mask is not None and attn_mask.dim() == 2:
+        attn_mask = attn_mask.unsqueeze(0).unsqueeze(0)
+    out, _ = flash_attention(query, key, value, dropout=dropout_p,
+                             causal=is_causal, attn_mask=attn_mask,
+                             training=training)
     return out
 
 

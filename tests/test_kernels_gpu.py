@@ -578,3 +578,80 @@ def test_weight_t_cache_invalidation():
     t2 = gd.weight_t(w)
     assert t2 is not t1
     torch.testing.assert_close(t2, w.t().contiguous())
+
+
+# ---------------------------------------------------------------------------
+# flash attention: mask + Philox dropout
+# ---------------------------------------------------------------------------
+def test_flash_attention_mask_gpu():
+    torch.manual_seed(11)
+    b, h, s, d = 2, 4, 256, 128
+    q = _bf(torch.randn(b, h, s, d, device=DEV)).requires_grad_(True)
+    k = _bf(torch.randn(b, h, s, d, device=DEV)).requires_grad_(True)
+    v = _bf(torch.randn(b, h, s, d, device=DEV)).requires_grad_(True)
+    # additive mask with random -inf blocks + finite biases
+    m = torch.zeros(b, 1, s, s, device=DEV)
+    m[:, :, :, ::7] = -10000.0
+    m[:, :, ::5, :] += 0.25
+    m = _bf(m)
+    scale = 1.0 / math.sqrt(d)
+    from paddle_amd.ops.functional import _FlashAttn, _sdpa_ref, _sdpa_ref_bwd
+    o, lse = _FlashAttn.apply(q, k, v, scale, False, m, 0.0, 0, 0)
+    ref_o, ref_lse = _sdpa_ref(q.detach().float(), k.detach().float(),
+                               v.detach().float(), scale, False, m.float())
+    _assert_close_bf16(o, ref_o, atol=3e-2, rtol=3e-2)
+    torch.testing.assert_close(lse, ref_lse, atol=2e-2, rtol=2e-2)
+    g = torch.randn_like(o)
+    o.backward(g)
+    dq, dk, dv = _sdpa_ref_bwd(g.float(), q.detach().float(), k.detach().float(),
+                               v.detach().float(), ref_lse, scale, False, m.float())
+    _assert_close_bf16(q.grad, dq, atol=5e-2, rtol=5e-2)
+    _assert_close_bf16(k.grad, dk, atol=5e-2, rtol=5e-2)
+    _assert_close_bf16(v.grad, dv, atol=5e-2, rtol=5e-2)
+
+
+def test_flash_attention_dropout_gpu():
+    """Exact check: extract the kernel's Philox keep-mask and feed it to the
+    fp32 oracle -- fwd and all three grads must match; plus determinism."""
+    torch.manual_seed(12)
+    C = _ext.get_ext()
+    b, h, s, d = 1, 2, 256, 128
+    p, seed, offset = 0.3, 1234, 999
+    q = _bf(torch.randn(b, h, s, d, device=DEV)).requires_grad_(True)
+    k = _bf(torch.randn(b, h, s, d, device=DEV)).requires_grad_(True)
+    v = _bf(torch.randn(b, h, s, d, device=DEV)).requires_grad_(True)
+    scale = 1.0 / math.sqrt(d)
+    from paddle_amd.ops.functional import _FlashAttn, _sdpa_ref, _sdpa_ref_bwd
+    o, lse = _FlashAttn.apply(q, k, v, scale, False, None, p, seed, offset)
+    o2, _ = _FlashAttn.apply(q, k, v, scale, False, None, p, seed, offset)
+    torch.testing.assert_close(o, o2)           # recompute-deterministic
+    keep = C.fa_dropout_mask(b, h, s, s, p, seed, offset)
+    frac = keep.float().mean().item()
+    assert abs(frac - (1 - p)) < 0.02, frac     # keep rate sane
+    ref_o, ref_lse = _sdpa_ref(q.detach().float(), k.detach().float(),
+                               v.detach().float(), scale, False, None,
+                               keep, p)
+    _assert_close_bf16(o, ref_o, atol=4e-2, rtol=4e-2)
+    g = torch.randn_like(o)
+    o.backward(g)
+    dq, dk, dv = _sdpa_ref_bwd(g.float(), q.detach().float(), k.detach().float(),
+                               v.detach().float(), ref_lse, scale, False, None,
+                               keep, p)
+    _assert_close_bf16(q.grad, dq, atol=6e-2, rtol=6e-2)
+    _assert_close_bf16(k.grad, dk, atol=6e-2, rtol=6e-2)
+    _assert_close_bf16(v.grad, dv, atol=6e-2, rtol=6e-2)
+
+
+def test_flash_attention_causal_mask_combo_gpu():
+    torch.manual_seed(13)
+    b, h, s, d = 1, 2, 200, 64
+    q = _bf(torch.randn(b, h, s, d, device=DEV))
+    k = _bf(torch.randn(b, h, s, d, device=DEV))
+    v = _bf(torch.randn(b, h, s, d, device=DEV))
+    m = _bf(torch.randn(b, h, s, s, device=DEV) * 0.5)
+    scale = 1.0 / math.sqrt(d)
+    from paddle_amd.ops.functional import _FlashAttn, _sdpa_ref
+    with torch.no_grad():
+        o, _ = _FlashAttn.apply(q, k, v, scale, True, m, 0.0, 0, 0)
+    ref_o, _ = _sdpa_ref(q.float(), k.float(), v.float(), scale, True, m.float())
+    _assert_close_bf16(o, ref_o, atol=3e-2, rtol=3e-2)
