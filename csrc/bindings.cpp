@@ -343,6 +343,85 @@ std::vector<Tensor> flash_attn_bwd(const Tensor& dout, const Tensor& q, const Te
   return {dq, dk, dv};
 }
 
+// host-side block map for the varlen kernels: (seq, offset) per block
+static std::pair<Tensor, int64_t> make_bmap(const Tensor& cu_cpu, int64_t blk) {
+  auto acc = cu_cpu.accessor<int, 1>();
+  std::vector<int> map;
+  for (int64_t i = 0; i + 1 < cu_cpu.size(0); ++i) {
+    int len = acc[i + 1] - acc[i];
+    for (int off = 0; off < len; off += (int)blk) {
+      map.push_back((int)i);
+      map.push_back(off);
+    }
+  }
+  auto t = torch::from_blob(map.data(), {(int64_t)map.size()},
+                            torch::dtype(torch::kInt32)).clone().cuda();
+  return {t, (int64_t)map.size() / 2};
+}
+
+std::vector<Tensor> flash_attn_varlen_fwd(const Tensor& q, const Tensor& k,
+                                          const Tensor& v, const Tensor& cu_q,
+                                          const Tensor& cu_k, double scale,
+                                          bool causal, double pdrop,
+                                          int64_t seed, int64_t offset) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16 &&
+              q.dim() == 3 && q.is_contiguous(), "varlen q: [total, H, D] bf16");
+  TORCH_CHECK(cu_q.scalar_type() == torch::kInt32 && cu_k.scalar_type() == torch::kInt32);
+  int64_t tq = q.size(0), h = q.size(1), d = q.size(2);
+  int64_t tk = k.size(0), hkv = k.size(1);
+  TORCH_CHECK(d == 128 || d == 64);
+  auto cu_q_cpu = cu_q.cpu().contiguous();
+  auto cu_k_cpu = cu_k.cpu().contiguous();
+  auto cu_q_gpu = cu_q.is_cuda() ? cu_q.contiguous() : cu_q_cpu.cuda();
+  auto cu_k_gpu = cu_k.is_cuda() ? cu_k.contiguous() : cu_k_cpu.cuda();
+  auto [bmap, nblk] = make_bmap(cu_q_cpu, 128);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({h, tq}, q.options().dtype(torch::kFloat));
+  pa::flash_attn_varlen_fwd32(q.const_data_ptr(), k.const_data_ptr(),
+                              v.const_data_ptr(), o.mutable_data_ptr(),
+                              lse.mutable_data_ptr<float>(), h, hkv, tq, tk, d,
+                              (float)scale, causal, nblk,
+                              cu_q_gpu.const_data_ptr<int>(),
+                              cu_k_gpu.const_data_ptr<int>(),
+                              bmap.const_data_ptr<int>(), (float)pdrop,
+                              (uint64_t)seed, (uint64_t)offset, cur_stream());
+  return {o, lse};
+}
+
+std::vector<Tensor> flash_attn_varlen_bwd(const Tensor& dout, const Tensor& q,
+                                          const Tensor& k, const Tensor& v,
+                                          const Tensor& o, const Tensor& lse,
+                                          const Tensor& cu_q, const Tensor& cu_k,
+                                          double scale, bool causal,
+                                          double pdrop, int64_t seed,
+                                          int64_t offset) {
+  int64_t tq = q.size(0), h = q.size(1), d = q.size(2);
+  int64_t tk = k.size(0);
+  TORCH_CHECK(k.size(1) == h, "varlen bwd requires hkv == h");
+  auto cu_q_cpu = cu_q.cpu().contiguous();
+  auto cu_k_cpu = cu_k.cpu().contiguous();
+  auto cu_q_gpu = cu_q.is_cuda() ? cu_q.contiguous() : cu_q_cpu.cuda();
+  auto cu_k_gpu = cu_k.is_cuda() ? cu_k.contiguous() : cu_k_cpu.cuda();
+  auto [qbmap, nqb] = make_bmap(cu_q_cpu, 128);
+  auto [kvbmap, nkb] = make_bmap(cu_k_cpu, 128);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto delta = torch::empty({h, tq}, q.options().dtype(torch::kFloat));
+  pa::flash_attn_varlen_bwd(dout.contiguous().const_data_ptr(), q.const_data_ptr(),
+                            k.const_data_ptr(), v.const_data_ptr(),
+                            o.const_data_ptr(), lse.const_data_ptr<float>(),
+                            dq.mutable_data_ptr(), dk.mutable_data_ptr(),
+                            dv.mutable_data_ptr(), delta.mutable_data_ptr<float>(),
+                            h, tq, tk, d, (float)scale, causal, nqb, nkb,
+                            cu_q_gpu.const_data_ptr<int>(),
+                            cu_k_gpu.const_data_ptr<int>(),
+                            qbmap.const_data_ptr<int>(),
+                            kvbmap.const_data_ptr<int>(), (float)pdrop,
+                            (uint64_t)seed, (uint64_t)offset, cur_stream());
+  return {dq, dk, dv};
+}
+
 Tensor fa_dropout_mask(int64_t b, int64_t h, int64_t sq, int64_t skv,
                        double p, int64_t seed, int64_t offset) {
   auto out = torch::empty({b, h, sq, skv},
@@ -546,6 +625,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("causal"), py::arg("mask") = c10::nullopt,
         py::arg("pdrop") = 0.0, py::arg("seed") = 0, py::arg("offset") = 0);
   m.def("fa_dropout_mask", &fa_dropout_mask);
+  m.def("flash_attn_varlen_fwd", &flash_attn_varlen_fwd, py::arg("q"),
+        py::arg("k"), py::arg("v"), py::arg("cu_q"), py::arg("cu_k"),
+        py::arg("scale"), py::arg("causal"), py::arg("pdrop") = 0.0,
+        py::arg("seed") = 0, py::arg("offset") = 0);
+  m.def("flash_attn_varlen_bwd", &flash_attn_varlen_bwd);
   m.def("flash_attn_bwd", &flash_attn_bwd, py::arg("dout"), py::arg("q"),
         py::arg("k"), py::arg("v"), py::arg("o"), py::arg("lse"),
         py::arg("dq") = c10::nullopt, py::arg("dk") = c10::nullopt,

@@ -167,6 +167,33 @@ void weight_only_gemv(const void* x, const void* wq, const float* scale,
                       const void* bias, void* out, int64_t m, int64_t n,
                       int64_t k, int dtype, hipStream_t s);
 
+// varlen (flash_attn_unpadded): packed [total, H, D], ONE launch; bmap =
+// int32 (seq, offset-within-seq) pairs per block; lse/delta [H, total_q]
+void flash_attn_varlen_fwd32(const void* q, const void* k, const void* v,
+                             void* o, float* lse, int64_t h, int64_t hkv,
+                             int64_t total_q, int64_t total_k, int64_t dh,
+                             float scale, bool causal, int64_t nblocks,
+                             const int* cu_q, const int* cu_k, const int* bmap,
+                             float pdrop, uint64_t seed, uint64_t offset,
+                             hipStream_t s);
+void flash_attn_varlen_bwd32(const void* dout, const void* q, const void* k,
+                             const void* v, const float* lse, const float* delta,
+                             void* dq, void* dk, void* dv, int64_t h,
+                             int64_t total_q, int64_t total_k, int64_t dh,
+                             float scale, bool causal, int64_t nqblocks,
+                             int64_t nkvblocks, const int* cu_q, const int* cu_k,
+                             const int* qbmap, const int* kvbmap, float pdrop,
+                             uint64_t seed, uint64_t offset, hipStream_t s);
+void flash_attn_varlen_bwd(const void* dout, const void* q, const void* k,
+                           const void* v, const void* o, const float* lse,
+                           void* dq, void* dk, void* dv, float* delta,
+                           int64_t h, int64_t total_q, int64_t total_k,
+                           int64_t dh, float scale, bool causal,
+                           int64_t nqblocks, int64_t nkvblocks,
+                           const int* cu_q, const int* cu_k, const int* qbmap,
+                           const int* kvbmap, float pdrop, uint64_t seed,
+                           uint64_t offset, hipStream_t s);
+
 // materialize the dropout keep-mask (debug/tests): out uint8 [total]
 void fa_dropout_mask(void* out, int64_t total, float p, uint64_t seed,
                      uint64_t offset, hipStream_t s);

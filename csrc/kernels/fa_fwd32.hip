@@ -42,7 +42,11 @@ __device__ __forceinline__ int pack_bf2(float a, float b) {
   return (int)(lo | (hi << 16));
 }
 
-template <int D, bool CAUSAL, bool MASKED, bool DROPOUT>
+// VARLEN: packed ragged batch (flash_attn_unpadded).  Sq/Skv carry the
+// TOTAL q/k token counts; cu_q/cu_k are [nseq+1] prefix sums and bmap
+// holds (seq, q0-within-seq) per grid.x block.  Causal is bottom-right
+// aligned (kv <= q + Skv_seq - Sq_seq).  lse layout [H, total_q].
+template <int D, bool CAUSAL, bool MASKED, bool DROPOUT, bool VARLEN = false>
 __launch_bounds__(256, 2)
 __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __restrict__ kg,
                                 const short* __restrict__ vg, short* __restrict__ og,
@@ -54,7 +58,10 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
                                 const short* __restrict__ maskg, long long m_sb,
                                 long long m_sh, long long m_sq,
                                 float pdrop, unsigned long long rseed,
-                                unsigned long long roffset) {
+                                unsigned long long roffset,
+                                const int* __restrict__ cu_q = nullptr,
+                                const int* __restrict__ cu_k = nullptr,
+                                const int* __restrict__ bmap = nullptr) {
   constexpr int NW = 4;            // waves per block, 32 q rows each
   constexpr int NT = NW * 64;
   constexpr int QB = NW * 32;      // 128 q rows per block
@@ -69,12 +76,29 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
 
   const int qblk = blockIdx.x;
   const int bh = blockIdx.y;
-  const int b = bh / H, h = bh % H;
-  const int hkv = h / (H / HKV);
-  const int q0 = qblk * QB;
-  const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
-  const long long kbase = (long long)b * k_sb + (long long)hkv * k_sh;
-  const long long obase = (long long)b * o_sb + (long long)h * o_sh;
+  int b = bh / H, h = bh % H;
+  int q0, Sq_e = Sq, Skv_e = Skv, cdelta = 0, qglob0 = 0;
+  long long qbase, kbase, obase;
+  if (VARLEN) {
+    b = 0; h = bh;
+    const int seq = bmap[2 * qblk];
+    q0 = bmap[2 * qblk + 1];
+    const int qs0 = cu_q[seq], ks0 = cu_k[seq];
+    Sq_e = cu_q[seq + 1] - qs0;
+    Skv_e = cu_k[seq + 1] - ks0;
+    cdelta = Skv_e - Sq_e;
+    qglob0 = qs0;
+    const int hkv_ = h / (H / HKV);
+    qbase = (long long)qs0 * q_ss + (long long)h * q_sh;
+    kbase = (long long)ks0 * k_ss + (long long)hkv_ * k_sh;
+    obase = (long long)qs0 * o_ss + (long long)h * o_sh;
+  } else {
+    const int hkv_ = h / (H / HKV);
+    q0 = qblk * QB;
+    qbase = (long long)b * q_sb + (long long)h * q_sh;
+    kbase = (long long)b * k_sb + (long long)hkv_ * k_sh;
+    obase = (long long)b * o_sb + (long long)h * o_sh;
+  }
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -90,7 +114,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
   shortx8 qf[NKS];
   {
     int row = q0w + l32;
-    bool ok = row < Sq;
+    bool ok = row < Sq_e;
 #pragma unroll
     for (int ks = 0; ks < NKS; ++ks) {
       if (ok) {
@@ -111,7 +135,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
     for (int r = 0; r < 16; ++r) oacc[dt][r] = 0.f;
   float m_run = -INFINITY, l_run = 0.f;   // stats for q column l32
 
-  const int kv_end = CAUSAL ? min(Skv, q0 + QB) : Skv;
+  const int kv_end = CAUSAL ? min(Skv_e, q0 + QB + cdelta) : Skv_e;
 
   // K-staging source offsets (pre-swizzled so async DMA lands swizzled rows)
   int k_row[KB * D / (NT * 8)], k_colp[KB * D / (NT * 8)];
@@ -125,7 +149,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
 
   auto stage = [&](int buf, int kv0) {
     // ---- stage K [KB][D] (async, swizzled) + V^T [D][KB] ------------------
-    if (kv0 + KB <= Skv) {
+    if (kv0 + KB <= Skv_e) {
 #pragma unroll
       for (int it = 0; it < KB * D / (NT * 8); ++it) {
         const short* src = kg + kbase + (long long)(kv0 + k_row[it]) * k_ss + k_colp[it];
@@ -138,7 +162,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
       for (int flat = tid * 8; flat < KB * D; flat += NT * 8) {
         int row = flat / D, col = flat % D;
         shortx8 val;
-        if (kv0 + row < Skv)
+        if (kv0 + row < Skv_e)
           val = *reinterpret_cast<const shortx8*>(kg + kbase + (long long)(kv0 + row) * k_ss + col);
         else
           for (int i = 0; i < 8; ++i) val[i] = 0;
@@ -150,7 +174,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
       for (int flat = tid * 8; flat < KB * D; flat += NT * 8) {
         int row = flat / D, col = flat % D;  // row=kv, col=d
         shortx8 val;
-        if (kv0 + row < Skv)
+        if (kv0 + row < Skv_e)
           val = *reinterpret_cast<const shortx8*>(vg + kbase + (long long)(kv0 + row) * k_ss + col);
         else
           for (int i = 0; i < 8; ++i) val[i] = 0;
@@ -185,7 +209,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
 
     // ---- mask + scale + per-lane online softmax ---------------------------
     const int q_abs = q0w + l32;
-    if (MASKED && q_abs < Sq) {
+    if (MASKED && q_abs < Sq_e) {
       // additive mask [B, 1|H, Sq, Skv]; reg groups are 4 consecutive kv
       const short* mrow = maskg + (long long)b * m_sb + (long long)h * m_sh
                         + (long long)q_abs * m_sq;
@@ -194,19 +218,19 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
           long long kvb = kv0 + 32 * t + 8 * g + 4 * hi;
-          if (kvb + 3 < Skv) {
+          if (kvb + 3 < Skv_e) {
             shortx4 mv = *reinterpret_cast<const shortx4*>(mrow + kvb);
 #pragma unroll
             for (int j = 0; j < 4; ++j) st[t][4 * g + j] += bf2f(mv[j]);
           } else {
 #pragma unroll
             for (int j = 0; j < 4; ++j)
-              if (kvb + j < Skv) st[t][4 * g + j] += bf2f(mrow[kvb + j]);
+              if (kvb + j < Skv_e) st[t][4 * g + j] += bf2f(mrow[kvb + j]);
           }
         }
     }
     float mx = -INFINITY;
-    const bool boundary = (kv0 + KB > Skv) || (CAUSAL && kv0 + KB > q_abs);
+    const bool boundary = (kv0 + KB > Skv_e) || (CAUSAL && kv0 + KB > q_abs + cdelta);
     if (boundary) {
 #pragma unroll
       for (int t = 0; t < 2; ++t)
@@ -214,7 +238,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
         for (int r = 0; r < 16; ++r) {
           int kv_abs = kv0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
           float v = st[t][r];                 // Q pre-scaled at load
-          if (kv_abs >= Skv || (CAUSAL && kv_abs > q_abs)) v = -INFINITY;
+          if (kv_abs >= Skv_e || (CAUSAL && kv_abs > q_abs + cdelta)) v = -INFINITY;
           st[t][r] = v;
           mx = fmaxf(mx, v);
         }
@@ -267,7 +291,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
           long long kvb = kv0 + 32 * t + 8 * g + 4 * hi;
-          long long elem = (bh_ll * Sq + q_abs) * (long long)Skv + kvb;
+          long long elem = (bh_ll * Sq + (q_abs + qglob0)) * (long long)Skv + kvb;
           uintx4 rv = philox10(rseed, roffset + (unsigned long long)(elem >> 2));
 #pragma unroll
           for (int j = 0; j < 4; ++j)
@@ -323,13 +347,13 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
     int rq = (r & 3) + 8 * (r >> 2) + 4 * hi;
     float il = __shfl(inv_l, rq, 64);
     int row = q0w + rq;
-    if (row >= Sq) continue;
+    if (row >= Sq_e) continue;
 #pragma unroll
     for (int dt = 0; dt < NDT; ++dt)
       og[obase + (long long)row * o_ss + dt * 32 + l32] = f2bf(oacc[dt][r] * il);
   }
-  if (hi == 0 && q0w + l32 < Sq)
-    lseg[(long long)(b * H + h) * Sq + q0w + l32] =
+  if (hi == 0 && q0w + l32 < Sq_e)
+    lseg[(long long)(b * H + h) * Sq + qglob0 + q0w + l32] =
         (l_run > 0.f) ? m_run + __logf(l_run) : -INFINITY;
 }
 
@@ -379,7 +403,7 @@ namespace {
 constexpr int DQ_NW = 4;
 }
 
-template <int D, bool CAUSAL, bool MASKED, bool DROPOUT>
+template <int D, bool CAUSAL, bool MASKED, bool DROPOUT, bool VARLEN = false>
 __launch_bounds__(256, 2)
 __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* __restrict__ qg,
                                    const short* __restrict__ kg, const short* __restrict__ vg,
@@ -393,7 +417,10 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
                                    const short* __restrict__ maskg, long long m_sb,
                                    long long m_sh, long long m_sq,
                                    float pdrop, unsigned long long rseed,
-                                   unsigned long long roffset) {
+                                   unsigned long long roffset,
+                                   const int* __restrict__ cu_q = nullptr,
+                                   const int* __restrict__ cu_k = nullptr,
+                                   const int* __restrict__ bmap = nullptr) {
   constexpr int NW = DQ_NW;
   constexpr int NT = NW * 64;
   constexpr int QB = NW * 32;       // 128 q rows / block
@@ -408,13 +435,30 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
 
   const int qblk = blockIdx.x;
   const int bh = blockIdx.y;
-  const int b = bh / H, h = bh % H;
-  const int q0 = qblk * QB;
-  const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
-  const long long dobase = (long long)b * do_sb + (long long)h * do_sh;
-  const long long dqbase = (long long)b * dq_sb + (long long)h * dq_sh;
-  const long long kvbase = (long long)b * k_sb + (long long)h * k_sh;
-  const long long lse_base = ((long long)bh) * Sq;
+  int b = bh / H, h = bh % H;
+  int q0, Sq_e = Sq, Skv_e = Skv, cdelta = 0, qglob0 = 0;
+  long long qbase, dobase, dqbase, kvbase;
+  if (VARLEN) {
+    b = 0; h = bh;
+    const int seq = bmap[2 * qblk];
+    q0 = bmap[2 * qblk + 1];
+    const int qs0 = cu_q[seq], ks0 = cu_k[seq];
+    Sq_e = cu_q[seq + 1] - qs0;
+    Skv_e = cu_k[seq + 1] - ks0;
+    cdelta = Skv_e - Sq_e;
+    qglob0 = qs0;
+    qbase = (long long)qs0 * q_ss + (long long)h * q_sh;
+    dobase = (long long)qs0 * do_ss + (long long)h * do_sh;
+    dqbase = (long long)qs0 * dq_ss + (long long)h * dq_sh;
+    kvbase = (long long)ks0 * k_ss + (long long)h * k_sh;
+  } else {
+    q0 = qblk * QB;
+    qbase = (long long)b * q_sb + (long long)h * q_sh;
+    dobase = (long long)b * do_sb + (long long)h * do_sh;
+    dqbase = (long long)b * dq_sb + (long long)h * dq_sh;
+    kvbase = (long long)b * k_sb + (long long)h * k_sh;
+  }
+  const long long lse_base = ((long long)bh) * Sq + qglob0;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -428,7 +472,7 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
   shortx8 qf[NKS], dof[NKS];
   {
     int row = q0w + l32;
-    bool ok = row < Sq;
+    bool ok = row < Sq_e;
 #pragma unroll
     for (int ks = 0; ks < NKS; ++ks) {
       if (ok) {
@@ -444,8 +488,8 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
     }
   }
   const int q_abs = q0w + l32;
-  const float lse_v = (q_abs < Sq) ? lseg[lse_base + q_abs] : 1e30f;
-  const float delta_v = (q_abs < Sq) ? deltag[lse_base + q_abs] : 0.f;
+  const float lse_v = (q_abs < Sq_e) ? lseg[lse_base + q_abs] : 1e30f;
+  const float delta_v = (q_abs < Sq_e) ? deltag[lse_base + q_abs] : 0.f;
 
   floatx16 dq_acc[NDT];
 #pragma unroll
@@ -463,9 +507,9 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
     n_colp[it] = col ^ ((row & 7) << 3);
   }
 
-  const int kv_end = CAUSAL ? min(Skv, q0 + QB) : Skv;
+  const int kv_end = CAUSAL ? min(Skv_e, q0 + QB + cdelta) : Skv_e;
   for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
-    if (kv0 + KB <= Skv) {
+    if (kv0 + KB <= Skv_e) {
 #pragma unroll
       for (int it = 0; it < KB * D / (NT * 8); ++it) {
         const short* ksrc = kg + kvbase + (long long)(kv0 + n_row[it]) * k_ss + n_colp[it];
@@ -494,7 +538,7 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
       for (int flat = tid * 8; flat < KB * D; flat += NT * 8) {
         int row = flat / D, col = flat % D;
         shortx8 kv_, vv;
-        if (kv0 + row < Skv) {
+        if (kv0 + row < Skv_e) {
           kv_ = *reinterpret_cast<const shortx8*>(kg + kvbase + (long long)(kv0 + row) * k_ss + col);
           vv = *reinterpret_cast<const shortx8*>(vg + kvbase + (long long)(kv0 + row) * k_ss + col);
         } else {
@@ -530,7 +574,7 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
     }
 
     // ---- dS^T = P^T (dP^T o dropout - delta) * scale ----------------------
-    if (MASKED && q_abs < Sq) {
+    if (MASKED && q_abs < Sq_e) {
       const short* mrow = maskg + (long long)b * m_sb + (long long)h * m_sh
                         + (long long)q_abs * m_sq;
 #pragma unroll
@@ -538,18 +582,18 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
           long long kvb = kv0 + 32 * t + 8 * g + 4 * hi;
-          if (kvb + 3 < Skv) {
+          if (kvb + 3 < Skv_e) {
             shortx4 mv = *reinterpret_cast<const shortx4*>(mrow + kvb);
 #pragma unroll
             for (int j = 0; j < 4; ++j) st[t][4 * g + j] += bf2f(mv[j]);
           } else {
 #pragma unroll
             for (int j = 0; j < 4; ++j)
-              if (kvb + j < Skv) st[t][4 * g + j] += bf2f(mrow[kvb + j]);
+              if (kvb + j < Skv_e) st[t][4 * g + j] += bf2f(mrow[kvb + j]);
           }
         }
     }
-    const bool bnd = (kv0 + KB > Skv) || (CAUSAL && kv0 + KB > q_abs);
+    const bool bnd = (kv0 + KB > Skv_e) || (CAUSAL && kv0 + KB > q_abs + cdelta);
     const float inv_keep = DROPOUT ? 1.f / (1.f - pdrop) : 1.f;
     const unsigned thr24 = DROPOUT ? (unsigned)(pdrop * 16777216.f) : 0u;
     const long long bh_ll = (long long)b * H + h;
@@ -559,7 +603,7 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
           long long kvb = kv0 + 32 * t + 8 * g + 4 * hi;
-          long long elem = (bh_ll * Sq + q_abs) * (long long)Skv + kvb;
+          long long elem = (bh_ll * Sq + (q_abs + qglob0)) * (long long)Skv + kvb;
           uintx4 rv = philox10(rseed, roffset + (unsigned long long)(elem >> 2));
 #pragma unroll
           for (int j = 0; j < 4; ++j)
@@ -572,7 +616,7 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
         float p = __expf(st[t][r] - lse_v);   // Q pre-scaled at load
         if (bnd) {
           int kv_abs = kv0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          if (kv_abs >= Skv || (CAUSAL && kv_abs > q_abs)) p = 0.f;
+          if (kv_abs >= Skv_e || (CAUSAL && kv_abs > q_abs + cdelta)) p = 0.f;
         }
         st[t][r] = p * (dp[t][r] - delta_v) * scale;
       }
@@ -617,7 +661,7 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
   for (int r = 0; r < 16; ++r) {
     int rq = (r & 3) + 8 * (r >> 2) + 4 * hi;
     int row = q0w + rq;
-    if (row >= Sq) continue;
+    if (row >= Sq_e) continue;
 #pragma unroll
     for (int dt = 0; dt < NDT; ++dt)
       dqg[dqbase + (long long)row * dq_ss + dt * 32 + l32] = f2bf(dq_acc[dt][r]);
@@ -673,7 +717,8 @@ void flash_attn_bwd_dq32(const void* dout, const void* q, const void* k,
 // transposed operand (dO^T or Q^T) is staged in LDS, double-buffered.
 // Outputs are accumulated transposed and written as packed 8 B stores.
 // ---------------------------------------------------------------------------
-template <int D, bool CAUSAL, bool IS_DK, bool MASKED, bool DROPOUT>
+template <int D, bool CAUSAL, bool IS_DK, bool MASKED, bool DROPOUT,
+          bool VARLEN = false>
 __launch_bounds__(256, 2)
 __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* __restrict__ qg,
                                     const short* __restrict__ kg, const short* __restrict__ vg,
@@ -687,7 +732,10 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
                                     const short* __restrict__ maskg, long long m_sb,
                                     long long m_sh, long long m_sq,
                                     float pdrop, unsigned long long rseed,
-                                    unsigned long long roffset) {
+                                    unsigned long long roffset,
+                                    const int* __restrict__ cu_q = nullptr,
+                                    const int* __restrict__ cu_k = nullptr,
+                                    const int* __restrict__ bmap = nullptr) {
   constexpr int NW = 4;
   constexpr int NT = NW * 64;
   constexpr int KVB = NW * 32;     // 128 kv rows / block
@@ -702,13 +750,30 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
 
   const int kvblk = blockIdx.x;
   const int bh = blockIdx.y;
-  const int b = bh / H, h = bh % H;
-  const int kv0 = kvblk * KVB;
-  const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
-  const long long dobase = (long long)b * do_sb + (long long)h * do_sh;
-  const long long kvbase = (long long)b * k_sb + (long long)h * k_sh;
-  const long long outbase = (long long)b * dk_sb + (long long)h * dk_sh;
-  const long long lse_base = ((long long)bh) * Sq;
+  int b = bh / H, h = bh % H;
+  int kv0, Sq_e = Sq, Skv_e = Skv, cdelta = 0, qglob0 = 0;
+  long long qbase, dobase, kvbase, outbase;
+  if (VARLEN) {
+    b = 0; h = bh;
+    const int seq = bmap[2 * kvblk];
+    kv0 = bmap[2 * kvblk + 1];
+    const int qs0 = cu_q[seq], ks0 = cu_k[seq];
+    Sq_e = cu_q[seq + 1] - qs0;
+    Skv_e = cu_k[seq + 1] - ks0;
+    cdelta = Skv_e - Sq_e;
+    qglob0 = qs0;
+    qbase = (long long)qs0 * q_ss + (long long)h * q_sh;
+    dobase = (long long)qs0 * do_ss + (long long)h * do_sh;
+    kvbase = (long long)ks0 * k_ss + (long long)h * k_sh;
+    outbase = (long long)ks0 * dk_ss + (long long)h * dk_sh;
+  } else {
+    kv0 = kvblk * KVB;
+    qbase = (long long)b * q_sb + (long long)h * q_sh;
+    dobase = (long long)b * do_sb + (long long)h * do_sh;
+    kvbase = (long long)b * k_sb + (long long)h * k_sh;
+    outbase = (long long)b * dk_sb + (long long)h * dk_sh;
+  }
+  const long long lse_base = ((long long)bh) * Sq + qglob0;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -721,7 +786,7 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
   // K (and V for dK) as B-fragments (lane = kv row), loaded once
   shortx8 kf[NKS], vf[IS_DK ? NKS : 1];
   {
-    bool ok = kv_lane < Skv;
+    bool ok = kv_lane < Skv_e;
 #pragma unroll
     for (int ks = 0; ks < NKS; ++ks) {
       if (ok) {
@@ -746,15 +811,15 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
   auto stage = [&](int buf, int q0) {
     if (tid < QT) {
       int qa = q0 + tid;
-      stat_s[buf][tid] = (qa < Sq) ? lseg[lse_base + qa] : 1e30f;
+      stat_s[buf][tid] = (qa < Sq_e) ? lseg[lse_base + qa] : 1e30f;
       if (IS_DK)
-        stat_s[buf][QT + tid] = (qa < Sq) ? deltag[lse_base + qa] : 0.f;
+        stat_s[buf][QT + tid] = (qa < Sq_e) ? deltag[lse_base + qa] : 0.f;
     }
     const int rot = tid & 7;
     for (int flat = tid * 8; flat < QT * D; flat += NT * 8) {
       int row = flat / D, col = flat % D;   // row = q, col = d
       shortx8 v;
-      if (q0 + row < Sq) {
+      if (q0 + row < Sq_e) {
         const short* src = IS_DK
             ? qg + qbase + (long long)(q0 + row) * q_ss + col
             : dog + dobase + (long long)(q0 + row) * do_ss + col;
@@ -770,16 +835,16 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
     }
   };
 
-  const int q_start = CAUSAL ? (kv0 / QT) * QT : 0;
+  const int q_start = CAUSAL ? (max(0, kv0 - cdelta) / QT) * QT : 0;
   stage(0, q_start);
   __syncthreads();
   int cur = 0;
-  for (int q0 = q_start; q0 < Sq; q0 += QT) {
-    if (q0 + QT < Sq) stage(cur ^ 1, q0 + QT);
+  for (int q0 = q_start; q0 < Sq_e; q0 += QT) {
+    if (q0 + QT < Sq_e) stage(cur ^ 1, q0 + QT);
 
     // Q (and dO for dK) A-fragments straight from global: lane = q row
     const int q_lane = q0 + l32;
-    const bool qok = q_lane < Sq;
+    const bool qok = q_lane < Sq_e;
     floatx16 st, dp;
 #pragma unroll
     for (int r = 0; r < 16; ++r) { st[r] = 0.f; dp[r] = 0.f; }
@@ -807,8 +872,8 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
     // dK additionally dS = P (dP - delta[q]) * scale -- but P needs lse
     // there too, so dK stages BOTH: lse comes from a second pass of
     // broadcast loads straight from global (L2-hot after the dV kernel).
-    const bool bnd = (q0 + QT > Sq) || (kv0w + 32 > Skv) ||
-                     (CAUSAL && kv0w + 32 > q0);
+    const bool bnd = (q0 + QT > Sq_e) || (kv0w + 32 > Skv_e) ||
+                     (CAUSAL && kv0w + 32 > q0 + cdelta);
     const float inv_keep = DROPOUT ? 1.f / (1.f - pdrop) : 1.f;
     const unsigned thr24 = DROPOUT ? (unsigned)(pdrop * 16777216.f) : 0u;
     const long long bh_ll = (long long)b * H + h;
@@ -817,16 +882,18 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
       const int rq = (r & 3) + 8 * (r >> 2) + 4 * hi;
       const int q_abs = q0 + rq;
       float sc = st[r] * scale;
-      if (MASKED && q_abs < Sq && kv_lane < Skv)
+      if (MASKED && q_abs < Sq_e && kv_lane < Skv_e)
         sc += bf2f(maskg[(long long)b * m_sb + (long long)h * m_sh
                          + (long long)q_abs * m_sq + kv_lane]);
       float p = __expf(sc - stat_s[cur][rq]);
       if (bnd) {
-        if (q_abs >= Sq || kv_lane >= Skv || (CAUSAL && kv_lane > q_abs)) p = 0.f;
+        if (q_abs >= Sq_e || kv_lane >= Skv_e ||
+            (CAUSAL && kv_lane > q_abs + cdelta)) p = 0.f;
       }
       bool keep = true;
       if (DROPOUT)
-        keep = fa_keep(rseed, roffset, bh_ll, Sq, Skv, q_abs, kv_lane, thr24);
+        keep = fa_keep(rseed, roffset, bh_ll, Sq, Skv, q_abs + qglob0,
+                       kv_lane, thr24);
       if (IS_DK) {
         float dpe = dp[r];
         if (DROPOUT) dpe = keep ? dpe * inv_keep : 0.f;
@@ -866,7 +933,7 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
   }
 
   // ---- epilogue: transposed accumulators -> natural rows ------------------
-  if (kv_lane < Skv) {
+  if (kv_lane < Skv_e) {
     const long long orow = outbase + (long long)kv_lane * dk_ss;
 #pragma unroll
     for (int dt = 0; dt < NDT; ++dt)
@@ -917,6 +984,74 @@ void flash_attn_bwd_dkv32(const void* dout, const void* q, const void* k,
 #undef FDKV
 #undef FDKV1
 #undef FDKV2
+}
+
+// ---------------------------------------------------------------------------
+// varlen (ragged, flash_attn_unpadded) launchers: ONE launch for the whole
+// packed batch; per-block sequence bounds from bmap/cu_seqlens.
+// Packed layout [total, H, D] contiguous; lse/delta [H, total_q].
+// ---------------------------------------------------------------------------
+void flash_attn_varlen_fwd32(const void* q, const void* k, const void* v,
+                             void* o, float* lse, int64_t h, int64_t hkv,
+                             int64_t total_q, int64_t total_k, int64_t dh,
+                             float scale, bool causal, int64_t nblocks,
+                             const int* cu_q, const int* cu_k, const int* bmap,
+                             float pdrop, uint64_t seed, uint64_t offset,
+                             hipStream_t s) {
+  dim3 grid((unsigned)nblocks, (unsigned)h);
+  dim3 blk(256);
+  const bool dropped = pdrop > 0.f;
+  const long long q_ss = h * dh, k_ss = hkv * dh;
+#define VFF(D, C, P)                                                           \
+  hipLaunchKernelGGL((fa_fwd32_kernel<D, C, false, P, true>), grid, blk, 0, s, \
+                     (const short*)q, (const short*)k, (const short*)v,        \
+                     (short*)o, lse, 1, (int)h, (int)hkv, (int)total_q,        \
+                     (int)total_k, scale, 0, dh, q_ss, 0, dh, k_ss,            \
+                     0, dh, q_ss, nullptr, 0, 0, 0, pdrop, seed, offset,       \
+                     cu_q, cu_k, bmap)
+#define VF(D, C) do { if (dropped) VFF(D, C, true); else VFF(D, C, false); } while (0)
+  if (dh == 128) { if (causal) VF(128, true); else VF(128, false); }
+  else           { if (causal) VF(64, true);  else VF(64, false); }
+#undef VF
+#undef VFF
+}
+
+void flash_attn_varlen_bwd32(const void* dout, const void* q, const void* k,
+                             const void* v, const float* lse, const float* delta,
+                             void* dq, void* dk, void* dv, int64_t h,
+                             int64_t total_q, int64_t total_k, int64_t dh,
+                             float scale, bool causal, int64_t nqblocks,
+                             int64_t nkvblocks, const int* cu_q, const int* cu_k,
+                             const int* qbmap, const int* kvbmap, float pdrop,
+                             uint64_t seed, uint64_t offset, hipStream_t s) {
+  dim3 blk(256);
+  const bool dropped = pdrop > 0.f;
+  const long long ss = h * dh;
+#define VDQ(D, C, P)                                                           \
+  hipLaunchKernelGGL((fa_bwd_dq32_kernel<D, C, false, P, true>),               \
+                     dim3((unsigned)nqblocks, (unsigned)h), blk, 0, s,         \
+                     (const short*)dout, (const short*)q, (const short*)k,     \
+                     (const short*)v, lse, delta, (short*)dq, 1, (int)h,       \
+                     (int)total_q, (int)total_k, scale, 0, dh, ss, 0, dh, ss,  \
+                     0, dh, ss, 0, dh, ss, nullptr, 0, 0, 0, pdrop, seed,      \
+                     offset, cu_q, cu_k, qbmap)
+#define VDKV(D, C, DK, P, OUT)                                                 \
+  hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C, DK, false, P, true>),          \
+                     dim3((unsigned)nkvblocks, (unsigned)h), blk, 0, s,        \
+                     (const short*)dout, (const short*)q, (const short*)k,     \
+                     (const short*)v, lse, delta, (short*)(OUT), 1, (int)h,    \
+                     (int)total_q, (int)total_k, scale, 0, dh, ss, 0, dh, ss,  \
+                     0, dh, ss, 0, dh, ss, nullptr, 0, 0, 0, pdrop, seed,      \
+                     offset, cu_q, cu_k, kvbmap)
+#define VB(D, C, P)                                                            \
+  do { VDQ(D, C, P); VDKV(D, C, false, P, dv); VDKV(D, C, true, P, dk); } while (0)
+#define VBD(D, C) do { if (dropped) VB(D, C, true); else VB(D, C, false); } while (0)
+  if (dh == 128) { if (causal) VBD(128, true); else VBD(128, false); }
+  else           { if (causal) VBD(64, true);  else VBD(64, false); }
+#undef VBD
+#undef VB
+#undef VDKV
+#undef VDQ
 }
 
 // debug/test utility: materialize the attention-dropout keep mask the FA

@@ -798,4 +798,30 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
 #undef FAB
 }
 
+void flash_attn_varlen_bwd(const void* dout, const void* q, const void* k,
+                           const void* v, const void* o, const float* lse,
+                           void* dq, void* dk, void* dv, float* delta,
+                           int64_t h, int64_t total_q, int64_t total_k,
+                           int64_t dh, float scale, bool causal,
+                           int64_t nqblocks, int64_t nkvblocks,
+                           const int* cu_q, const int* cu_k, const int* qbmap,
+                           const int* kvbmap, float pdrop, uint64_t seed,
+                           uint64_t offset, hipStream_t s) {
+  // delta[h, total_q] = rowsum(do * o) over the packed [total, H, D] layout
+  long long rows = h * total_q;
+  dim3 dgrid((unsigned)hmin<long long>(2048LL, (rows + 3) / 4));
+  const long long ss = h * dh;
+  if (dh == 128)
+    hipLaunchKernelGGL((fa_bwd_delta_kernel<128>), dgrid, dim3(256), 0, s,
+                       (const short*)dout, (const short*)o, delta, (int)h,
+                       (int)total_q, rows, 0, dh, ss, 0, dh, ss);
+  else
+    hipLaunchKernelGGL((fa_bwd_delta_kernel<64>), dgrid, dim3(256), 0, s,
+                       (const short*)dout, (const short*)o, delta, (int)h,
+                       (int)total_q, rows, 0, dh, ss, 0, dh, ss);
+  flash_attn_varlen_bwd32(dout, q, k, v, lse, delta, dq, dk, dv, h, total_q,
+                          total_k, dh, scale, causal, nqblocks, nkvblocks,
+                          cu_q, cu_k, qbmap, kvbmap, pdrop, seed, offset, s);
+}
+
 }  // namespace pa

@@ -957,17 +957,53 @@ def fused_linear_param_grad_add(x, dy, dweight=None, dbias=None,
     return dweight, None
 
 
+class _FlashAttnVarlen(torch.autograd.Function):
+    """Single-launch ragged attention over packed [total, H, D] tensors
+    (reference flash_attn_unpadded / flash_attn_kernel.cu:41 varlen via
+    cu_seqlens; ours maps each 128-row block to its sequence in-kernel)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, cu_q, cu_k, scale, causal, dropout, seed, offset):
+        C = _ext.get_ext()
+        o, lse = C.flash_attn_varlen_fwd(q, k, v, cu_q, cu_k, scale, causal,
+                                         dropout, seed, offset)
+        ctx.save_for_backward(q, k, v, o, lse, cu_q, cu_k)
+        ctx.scale, ctx.causal = scale, causal
+        ctx.dropout, ctx.seed, ctx.offset = dropout, seed, offset
+        ctx.mark_non_differentiable(lse)
+        return o, lse
+
+    @staticmethod
+    def backward(ctx, do, _dlse):
+        q, k, v, o, lse, cu_q, cu_k = ctx.saved_tensors
+        C = _ext.get_ext()
+        dq, dk, dv = C.flash_attn_varlen_bwd(do, q, k, v, o, lse, cu_q, cu_k,
+                                             ctx.scale, ctx.causal, ctx.dropout,
+                                             ctx.seed, ctx.offset)
+        return dq, dk, dv, None, None, None, None, None, None, None
+
+
 def flash_attn_varlen_func(q, k, v, cu_seqlens_q, cu_seqlens_k, max_seqlen_q,
-                           max_seqlen_k, scale=None, causal=False):
+                           max_seqlen_k, scale=None, causal=False,
+                           dropout=0.0, training=True, return_softmax_lse=False):
     """Varlen (ragged) flash attention over packed [total_tokens, H, D]
     inputs with cu_seqlens boundaries (reference: flash_attn_unpadded,
-    python/paddle/nn/functional/flash_attention.py:195).
+    python/paddle/nn/functional/flash_attention.py:593).
 
-    Round-1 scheme: group sequences by (len_q, len_k) and run each group
-    through the rectangular HIP kernel in ONE batched call -- exact
-    numerics, no padding compute; a fused ragged kernel is queued for
-    round 2 (the kernels are already stride-aware).
-    """
+    Native path: ONE kernel launch for the whole mixed-length batch
+    (per-block sequence bounds resolved in-kernel); CPU fallback groups
+    by shape."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    p = dropout if training else 0.0
+    if (_ext.use_native(q) and q.dtype == torch.bfloat16
+            and q.shape[-1] in (64, 128) and k.shape[1] == q.shape[1]):
+        cu_q = cu_seqlens_q.to(torch.int32)
+        cu_k = cu_seqlens_k.to(torch.int32)
+        seed, offset = _fa_seed_offset() if p > 0 else (0, 0)
+        out, lse = _FlashAttnVarlen.apply(q, k, v, cu_q, cu_k, scale, causal,
+                                          p, seed, offset)
+        return (out, lse) if return_softmax_lse else out
     import collections
     nq = cu_seqlens_q.tolist()
     nk = cu_seqlens_k.tolist()
@@ -983,4 +1019,6 @@ def flash_attn_varlen_func(q, k, v, cu_seqlens_q, cu_seqlens_k, max_seqlen_q,
                                 vg.transpose(1, 2), causal=causal, scale=scale)
         for j, i in enumerate(idxs):
             out[nq[i]:nq[i] + lq] = og[j]
+    if return_softmax_lse:
+        return out, None
     return out

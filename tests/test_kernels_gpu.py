@@ -655,3 +655,48 @@ def test_flash_attention_causal_mask_combo_gpu():
         o, _ = _FlashAttn.apply(q, k, v, scale, True, m, 0.0, 0, 0)
     ref_o, _ = _sdpa_ref(q.float(), k.float(), v.float(), scale, True, m.float())
     _assert_close_bf16(o, ref_o, atol=3e-2, rtol=3e-2)
+
+
+def test_flash_attn_varlen_single_launch_gpu():
+    """Mixed lengths, ONE kernel launch: fwd+bwd vs the
+    per-sequence fp32 oracle."""
+    torch.manual_seed(21)
+    from paddle_amd.ops.functional import flash_attn_varlen_func, _sdpa_ref, _sdpa_ref_bwd
+    h, d = 4, 128
+    lens = [128, 37, 256, 200]
+    cu = torch.tensor([0] + list(torch.cumsum(torch.tensor(lens), 0)),
+                      dtype=torch.int32, device=DEV)
+    total = sum(lens)
+    q = _bf(torch.randn(total, h, d, device=DEV)).requires_grad_(True)
+    k = _bf(torch.randn(total, h, d, device=DEV)).requires_grad_(True)
+    v = _bf(torch.randn(total, h, d, device=DEV)).requires_grad_(True)
+    scale = 1.0 / math.sqrt(d)
+    for causal in (False, True):
+        if q.grad is not None:
+            q.grad = None; k.grad = None; v.grad = None
+        out = flash_attn_varlen_func(q, k, v, cu, cu, max(lens), max(lens),
+                                     scale=scale, causal=causal)
+        g = torch.randn_like(out)
+        out.backward(g)
+        # per-sequence reference
+        o_ref = torch.empty(total, h, d, device=DEV)
+        dq_ref = torch.zeros(total, h, d, device=DEV)
+        dk_ref = torch.zeros_like(dq_ref)
+        dv_ref = torch.zeros_like(dq_ref)
+        for i, L in enumerate(lens):
+            s0, s1 = int(cu[i]), int(cu[i + 1])
+            qs = q.detach()[s0:s1].transpose(0, 1).unsqueeze(0)  # [1,h,L,d]
+            ks = k.detach()[s0:s1].transpose(0, 1).unsqueeze(0)
+            vs = v.detach()[s0:s1].transpose(0, 1).unsqueeze(0)
+            oo, lse = _sdpa_ref(qs.float(), ks.float(), vs.float(), scale, causal)
+            o_ref[s0:s1] = oo.squeeze(0).transpose(0, 1)
+            gs = g[s0:s1].transpose(0, 1).unsqueeze(0).float()
+            dqs, dks, dvs = _sdpa_ref_bwd(gs, qs.float(), ks.float(), vs.float(),
+                                          lse, scale, causal)
+            dq_ref[s0:s1] = dqs.squeeze(0).transpose(0, 1)
+            dk_ref[s0:s1] = dks.squeeze(0).transpose(0, 1)
+            dv_ref[s0:s1] = dvs.squeeze(0).transpose(0, 1)
+        _assert_close_bf16(out, o_ref, atol=3e-2, rtol=3e-2)
+        _assert_close_bf16(q.grad, dq_ref, atol=5e-2, rtol=5e-2)
+        _assert_close_bf16(k.grad, dk_ref, atol=5e-2, rtol=5e-2)
+        _assert_close_bf16(v.grad, dv_ref, atol=5e-2, rtol=5e-2)
