@@ -13,6 +13,8 @@
 namespace pa {
 void mfma_probe(const void* a, const void* bt, float* c, hipStream_t s);
 void mfma_probe32(const void* a, const void* bt, float* c, hipStream_t s);
+void mfma_probe_fp8mx(const void* a, const void* bt, float* c, int sa, int sb,
+                      hipStream_t s);
 }
 
 namespace pa_lt {
@@ -582,7 +584,37 @@ std::vector<Tensor> gemm_bf16_ex(const Tensor& a, const Tensor& b,
   return {c};
 }
 
+Tensor gemm_fp8_nt(const Tensor& a, const Tensor& bt, double scale_ab,
+                   const c10::optional<Tensor>& bias) {
+  TORCH_CHECK(a.is_cuda() && a.dim() == 2 && bt.dim() == 2);
+  TORCH_CHECK(a.scalar_type() == torch::kFloat8_e4m3fn ||
+              a.scalar_type() == torch::kUInt8, "gemm_fp8: e4m3/uint8 input");
+  TORCH_CHECK(a.stride(1) == 1 && bt.stride(1) == 1, "row-major inputs");
+  int64_t m = a.size(0), k = a.size(1), n = bt.size(0);
+  TORCH_CHECK(bt.size(1) == k && k >= 128);
+  auto c = torch::empty({m, n}, a.options().dtype(torch::kBFloat16));
+  const void* bp = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->scalar_type() == torch::kBFloat16 && bias->numel() == n);
+    bp = bias->const_data_ptr();
+  }
+  pa::gemm_fp8_nt(a.const_data_ptr(), bt.const_data_ptr(), c.mutable_data_ptr(),
+                  bp, (float)scale_ab, m, n, k, a.stride(0), bt.stride(0),
+                  c.stride(0), cur_stream());
+  return c;
+}
+
 // ---- probe ----------------------------------------------------------------
+Tensor mfma_probe_fp8mx(const Tensor& a, const Tensor& bt, int64_t sa, int64_t sb) {
+  CHECK_IN(a); CHECK_IN(bt);
+  TORCH_CHECK(a.scalar_type() == torch::kUInt8 && a.numel() == 16 * 128);
+  auto c = torch::empty({16, 16}, a.options().dtype(torch::kFloat));
+  pa::mfma_probe_fp8mx(a.const_data_ptr(), bt.const_data_ptr(),
+                       c.mutable_data_ptr<float>(), (int)sa, (int)sb,
+                       cur_stream());
+  return c;
+}
+
 Tensor mfma_probe(const Tensor& a, const Tensor& bt) {
   CHECK_IN(a); CHECK_IN(bt);
   auto c = torch::empty({16, 16}, a.options().dtype(torch::kFloat));
@@ -625,6 +657,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("causal"), py::arg("mask") = c10::nullopt,
         py::arg("pdrop") = 0.0, py::arg("seed") = 0, py::arg("offset") = 0);
   m.def("fa_dropout_mask", &fa_dropout_mask);
+  m.def("mfma_probe_fp8mx", &mfma_probe_fp8mx);
+  m.def("gemm_fp8_nt", &gemm_fp8_nt, py::arg("a"), py::arg("bt"),
+        py::arg("scale_ab") = 1.0, py::arg("bias") = c10::nullopt);
   m.def("flash_attn_varlen_fwd", &flash_attn_varlen_fwd, py::arg("q"),
         py::arg("k"), py::arg("v"), py::arg("cu_q"), py::arg("cu_k"),
         py::arg("scale"), py::arg("causal"), py::arg("pdrop") = 0.0,

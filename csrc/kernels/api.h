@@ -198,4 +198,9 @@ void flash_attn_varlen_bwd(const void* dout, const void* q, const void* k,
 void fa_dropout_mask(void* out, int64_t total, float p, uint64_t seed,
                      uint64_t offset, hipStream_t s);
 
+// fp8 e4m3 MX GEMM: C[m,n] (bf16) = scale_ab * (A_fp8[m,k] x Bt_fp8[n,k]^T)
+void gemm_fp8_nt(const void* a, const void* bt, void* c, const void* bias,
+                 float scale_ab, int64_t m, int64_t n, int64_t k, int64_t lda,
+                 int64_t ldb, int64_t ldc, hipStream_t s);
+
 }  // namespace pa

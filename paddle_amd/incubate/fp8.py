@@ -80,6 +80,19 @@ class _Fp8GroupedFFN(torch.autograd.Function):
                 dw2.to(w2.dtype), db2.to(b2.dtype))
 
 
+def _own_fp8_ok(x):
+    from .. import _ext
+    return x.is_cuda and _ext.has_ext()
+
+
+def fp8_gemm_nt(qx, qwt, scale_ab, bias=None):
+    """C = scale_ab * qx @ qwt^T on the own MX MFMA kernel
+    (csrc/kernels/gemm_fp8.hip; ~2x the bf16 MFMA rate)."""
+    from .. import _ext
+    C = _ext.get_ext()
+    return C.gemm_fp8_nt(qx, qwt, float(scale_ab), bias)
+
+
 class _Fp8Matmul(torch.autograd.Function):
     """out = x @ w  with x,w quantized to e4m3 for the MFMA fp8 path."""
 
@@ -87,7 +100,16 @@ class _Fp8Matmul(torch.autograd.Function):
     def forward(ctx, x, w):
         ctx.save_for_backward(x, w)
         x2 = x.reshape(-1, x.shape[-1])
-        if x.is_cuda:
+        # dispatch note (measured, profiles/gemm_fp8_r02.txt): own MX kernel
+        # 1.6-2.0 PF > bf16 hipBLASLt 1.2-1.3 PF; hipBLASLt fp8 via
+        # _scaled_mm 2.4-3.3 PF stays the default (autotune-between-impls)
+        if x.is_cuda and not hasattr(torch, "_scaled_mm") and _own_fp8_ok(x) \
+                and x2.shape[1] % 128 == 0:
+            qx, sx = _quant(x2)
+            qw, sw = _quant(w)
+            out = fp8_gemm_nt(qx.contiguous(), qw.t().contiguous(),
+                              float(sx) * float(sw))
+        elif x.is_cuda:
             qx, sx = _quant(x2)
             qw, sw = _quant(w)
             # _scaled_mm requires column-major B: pass w^T's transpose view

@@ -700,3 +700,44 @@ def test_flash_attn_varlen_single_launch_gpu():
         _assert_close_bf16(q.grad, dq_ref, atol=5e-2, rtol=5e-2)
         _assert_close_bf16(k.grad, dk_ref, atol=5e-2, rtol=5e-2)
         _assert_close_bf16(v.grad, dv_ref, atol=5e-2, rtol=5e-2)
+
+
+# ---------------------------------------------------------------------------
+# fp8 e4m3 MX GEMM (gemm_fp8.hip)
+# ---------------------------------------------------------------------------
+def test_fp8mx_probe_layout():
+    torch.manual_seed(30)
+    a = (torch.randn(16, 128) * 0.5).to(torch.float8_e4m3fn)
+    bt = (torch.randn(16, 128) * 0.5).to(torch.float8_e4m3fn)
+    one = 0x7F7F7F7F
+    C = _ext.get_ext()
+    out = C.mfma_probe_fp8mx(a.view(torch.uint8).cuda(),
+                             bt.view(torch.uint8).cuda(), one, one).cpu()
+    ref = a.float() @ bt.float().t()
+    _gemm_rel_ok(out, ref, tol=1e-3)
+
+
+@pytest.mark.parametrize("m,n,k", [(512, 512, 512), (777, 300, 256),
+                                   (2048, 1024, 4096)])
+def test_gemm_fp8_numerics(m, n, k):
+    torch.manual_seed(31)
+    C = _ext.get_ext()
+    a = (torch.randn(m, k, device=DEV) * 0.3).to(torch.float8_e4m3fn)
+    bt = (torch.randn(n, k, device=DEV) * 0.3).to(torch.float8_e4m3fn)
+    out = C.gemm_fp8_nt(a, bt, 0.731)
+    ref = 0.731 * (a.float() @ bt.float().t())
+    _gemm_rel_ok(out, ref, tol=2e-2)
+    # bias epilogue
+    bias = _bf(torch.randn(n, device=DEV))
+    outb = C.gemm_fp8_nt(a, bt, 1.0, bias)
+    _gemm_rel_ok(outb, a.float() @ bt.float().t() + bias.float(), tol=2e-2)
+
+
+def test_fp8_matmul_roundtrip():
+    from paddle_amd.incubate import fp8
+    torch.manual_seed(32)
+    x = _bf(torch.randn(512, 512, device=DEV))
+    w = _bf(torch.randn(512, 384, device=DEV) * 0.05)
+    y = fp8.fp8_matmul(x, w)
+    ref = x.float() @ w.float()
+    _gemm_rel_ok(y, ref, tol=6e-2)  # fp8 quantization error budget

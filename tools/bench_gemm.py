@@ -65,10 +65,14 @@ def main():
     ap.add_argument("--duration", type=int, default=100)
     ap.add_argument("--custom", action="store_true")
     ap.add_argument("--autotune", action="store_true")
+    ap.add_argument("--fp8", action="store_true")
     ap.add_argument("--iters", type=int, default=10)
     args = ap.parse_args()
     if args.autotune:
         autotune(args.iters)
+        return
+    if args.fp8:
+        bench_fp8()
         return
     if args.custom:
         bench_custom()
@@ -188,6 +192,34 @@ def autotune(iters):
     os.makedirs("gpurun_out", exist_ok=True)
     with open(out, "w") as f:
         json.dump({"version": 1, "entries": entries}, f, indent=1)
+
+
+def bench_fp8():
+    """fp8 MX kernel vs bf16 hipBLASLt vs torch._scaled_mm at bench shapes."""
+    from paddle_amd import _ext
+    C = _ext.get_ext()
+    shapes = [(16384, 4096, 12288), (16384, 16384, 4096), (8192, 8192, 8192),
+              (4096, 4096, 4096), (2048, 14336, 4096)]
+    for M, K, N in shapes:
+        x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(K, N, device="cuda", dtype=torch.bfloat16) * 0.02
+        qx = x.to(torch.float8_e4m3fn)
+        qwt = w.t().contiguous().to(torch.float8_e4m3fn)
+        flops = 2.0 * M * K * N
+        out = C.gemm_fp8_nt(qx, qwt, 1.0)
+        ref = qx.float() @ qwt.float().t()
+        err = (out.float() - ref).abs().max().item() / ref.abs().max().item()
+        t8 = bench(lambda: C.gemm_fp8_nt(qx, qwt, 1.0))
+        tb = bench(lambda: torch.matmul(x, w))
+        wt_col = w.to(torch.float8_e4m3fn).t().contiguous().t()  # [K,N] col-major
+        sc = torch.ones(1, device="cuda")
+        tsm = bench(lambda: torch._scaled_mm(qx, wt_col, scale_a=sc, scale_b=sc,
+                                             out_dtype=torch.bfloat16))
+        print(f"M{M} K{K} N{N}: own-fp8 {flops/t8/1e12:7.1f} TF  "
+              f"bf16-lt {flops/tb/1e12:7.1f}  scaled_mm {flops/tsm/1e12:7.1f}  "
+              f"(relerr {err:.1e})")
+        del x, w, qx, qwt
+        torch.cuda.empty_cache()
 
 
 if __name__ == "__main__":
