@@ -493,19 +493,27 @@ Tensor embedding_bwd(const Tensor& dout, const Tensor& ids_in, int64_t vocab,
 // ---- decode attention ------------------------------------------------------
 Tensor decode_attention(const Tensor& q, const Tensor& kcache, const Tensor& vcache,
                         const Tensor& block_table, const Tensor& seq_lens,
-                        double scale) {
+                        double scale, int64_t blk_str, int64_t pos_str,
+                        int64_t head_str, int64_t bs_override) {
   CHECK_IN(q); CHECK_IN(kcache); CHECK_IN(vcache);
   TORCH_CHECK(q.dim() == 3, "q: [B, H, D] (one decode step)");
   int64_t b = q.size(0), h = q.size(1), d = q.size(2);
-  TORCH_CHECK(kcache.dim() == 4, "k_cache: [nblocks, block_size, HKV, D]");
-  int64_t bs = kcache.size(1), hkv = kcache.size(2);
+  int64_t bs, hkv;
+  if (blk_str == 0) {
+    TORCH_CHECK(kcache.dim() == 4, "k_cache: [nblocks, block_size, HKV, D]");
+    bs = kcache.size(1); hkv = kcache.size(2);
+  } else {
+    bs = bs_override; hkv = bs_override >= 0 ? h : h;  // dense: hkv == h
+    hkv = kcache.size(1) == h ? h : kcache.size(1);
+  }
   int64_t max_blocks = block_table.size(1);
   auto o = torch::empty_like(q);
   pa::decode_attention(q.const_data_ptr(), kcache.const_data_ptr(),
                        vcache.const_data_ptr(),
                        block_table.const_data_ptr<int>(),
                        seq_lens.const_data_ptr<int>(), o.mutable_data_ptr(),
-                       b, h, hkv, bs, max_blocks, d, (float)scale, cur_stream());
+                       b, h, hkv, bs, max_blocks, d, (float)scale,
+                       blk_str, pos_str, head_str, cur_stream());
   return o;
 }
 
@@ -680,7 +688,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("layout"), py::arg("epilogue") = 0,
         py::arg("bias") = c10::nullopt, py::arg("aux") = c10::nullopt,
         py::arg("c_in") = c10::nullopt);
-  m.def("decode_attention", &decode_attention);
+  m.def("decode_attention", &decode_attention, py::arg("q"), py::arg("kcache"),
+        py::arg("vcache"), py::arg("block_table"), py::arg("seq_lens"),
+        py::arg("scale"), py::arg("blk_str") = 0, py::arg("pos_str") = 0,
+        py::arg("head_str") = 0, py::arg("bs_override") = 0);
   m.def("mfma_probe", &mfma_probe);
   m.def("mfma_probe32", &mfma_probe32);
   m.attr("compiled_arch") = "gfx950";

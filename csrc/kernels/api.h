@@ -139,10 +139,13 @@ void embedding_bwd(const void* dout, const int64_t* ids, float* dtable,
                    int dtype, hipStream_t s);
 
 // ---- paged-KV decode attention (serving) ----------------------------------
+// blk/pos/head strides in elements (0,0,0 = default paged layout
+// [nblocks, bs, HKV, D]); dense [B, H, S, D] caches pass their own
 void decode_attention(const void* q, const void* kcache, const void* vcache,
                       const int* block_table, const int* seq_lens, void* o,
                       int64_t b, int64_t h, int64_t hkv, int64_t bs,
                       int64_t max_blocks, int64_t dh, float scale,
+                      int64_t blk_str, int64_t pos_str, int64_t head_str,
                       hipStream_t s);
 
 // ---- hand-written bf16 MFMA GEMM (C[m][n] = op(A) x op(B)) ----------------

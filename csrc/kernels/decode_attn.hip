@@ -22,7 +22,9 @@ __global__ void decode_attn_kernel(const short* __restrict__ qg,
                                    const int* __restrict__ seq_lens,
                                    short* __restrict__ og,
                                    int B, int H, int HKV, int bs,
-                                   int max_blocks, float scale) {
+                                   int max_blocks, float scale,
+                                   long long blk_str, long long pos_str,
+                                   long long head_str) {
   const int b = blockIdx.x / H;
   const int h = blockIdx.x % H;
   const int hkv = h / (H / HKV);
@@ -56,8 +58,10 @@ __global__ void decode_attn_kernel(const short* __restrict__ qg,
       phys = block_table[(long long)b * max_blocks + pos / bs];
       off = pos % bs;
     }
-    const short* kp = kcache + (((long long)phys * bs + off) * HKV + hkv) * D;
-    const short* vp = vcache + (((long long)phys * bs + off) * HKV + hkv) * D;
+    const short* kp = kcache + (long long)phys * blk_str + (long long)off * pos_str
+                     + (long long)hkv * head_str;
+    const short* vp = vcache + (long long)phys * blk_str + (long long)off * pos_str
+                     + (long long)hkv * head_str;
     float sc = 0.f;
     if (ok) {
       if (E == 8) {
@@ -138,18 +142,26 @@ void decode_attention(const void* q, const void* kcache, const void* vcache,
                       const int* block_table, const int* seq_lens, void* o,
                       int64_t b, int64_t h, int64_t hkv, int64_t bs,
                       int64_t max_blocks, int64_t dh, float scale,
+                      int64_t blk_str, int64_t pos_str, int64_t head_str,
                       hipStream_t s) {
   dim3 grid((unsigned)(b * h));
+  if (blk_str == 0) {   // default paged layout [nblocks, bs, HKV, D]
+    blk_str = bs * hkv * dh;
+    pos_str = hkv * dh;
+    head_str = dh;
+  }
   if (dh == 128)
     hipLaunchKernelGGL((decode_attn_kernel<128>), grid, dim3(256), 0, s,
                        (const short*)q, (const short*)kcache, (const short*)vcache,
                        block_table, seq_lens, (short*)o, (int)b, (int)h,
-                       (int)hkv, (int)bs, (int)max_blocks, scale);
+                       (int)hkv, (int)bs, (int)max_blocks, scale,
+                       blk_str, pos_str, head_str);
   else
     hipLaunchKernelGGL((decode_attn_kernel<64>), grid, dim3(256), 0, s,
                        (const short*)q, (const short*)kcache, (const short*)vcache,
                        block_table, seq_lens, (short*)o, (int)b, (int)h,
-                       (int)hkv, (int)bs, (int)max_blocks, scale);
+                       (int)hkv, (int)bs, (int)max_blocks, scale,
+                       blk_str, pos_str, head_str);
 }
 
 }  // namespace pa
