@@ -612,6 +612,35 @@ Tensor gemm_fp8_nt(const Tensor& a, const Tensor& bt, double scale_ab,
   return c;
 }
 
+// ---- MoE routing ----------------------------------------------------------
+std::vector<Tensor> moe_gate_topk(const Tensor& logits, int64_t k) {
+  CHECK_IN(logits);
+  TORCH_CHECK(logits.scalar_type() == torch::kFloat && logits.dim() == 2);
+  int64_t t = logits.size(0), e = logits.size(1);
+  TORCH_CHECK(e <= 64 && k <= 4);
+  auto topv = torch::empty({t, k}, logits.options());
+  auto topi = torch::empty({t, k}, logits.options().dtype(torch::kInt32));
+  auto me = torch::zeros({e}, logits.options());
+  auto ce = torch::zeros({e}, logits.options());
+  pa::moe_gate_topk(logits.const_data_ptr<float>(), topv.mutable_data_ptr<float>(),
+                    topi.mutable_data_ptr<int>(), me.mutable_data_ptr<float>(),
+                    ce.mutable_data_ptr<float>(), t, e, k, cur_stream());
+  return {topv, topi, me, ce};
+}
+
+std::vector<Tensor> moe_assign_slots(const Tensor& topi, int64_t num_experts,
+                                     int64_t cap) {
+  CHECK_IN(topi);
+  TORCH_CHECK(topi.scalar_type() == torch::kInt32 && topi.dim() == 2);
+  int64_t t = topi.size(0), k = topi.size(1);
+  auto slot = torch::full({t, k}, -1, topi.options());
+  auto counts = torch::zeros({num_experts}, topi.options());
+  pa::moe_assign_slots(topi.const_data_ptr<int>(), slot.mutable_data_ptr<int>(),
+                       counts.mutable_data_ptr<int>(), t, num_experts, k, cap,
+                       cur_stream());
+  return {slot, counts};
+}
+
 // ---- probe ----------------------------------------------------------------
 Tensor mfma_probe_fp8mx(const Tensor& a, const Tensor& bt, int64_t sa, int64_t sb) {
   CHECK_IN(a); CHECK_IN(bt);
@@ -666,6 +695,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("pdrop") = 0.0, py::arg("seed") = 0, py::arg("offset") = 0);
   m.def("fa_dropout_mask", &fa_dropout_mask);
   m.def("mfma_probe_fp8mx", &mfma_probe_fp8mx);
+  m.def("moe_gate_topk", &moe_gate_topk);
+  m.def("moe_assign_slots", &moe_assign_slots);
   m.def("gemm_fp8_nt", &gemm_fp8_nt, py::arg("a"), py::arg("bt"),
         py::arg("scale_ab") = 1.0, py::arg("bias") = c10::nullopt);
   m.def("flash_attn_varlen_fwd", &flash_attn_varlen_fwd, py::arg("q"),

@@ -206,4 +206,10 @@ void gemm_fp8_nt(const void* a, const void* bt, void* c, const void* bias,
                  float scale_ab, int64_t m, int64_t n, int64_t k, int64_t lda,
                  int64_t ldb, int64_t ldc, hipStream_t s);
 
+// ---- MoE routing (assign_pos/number_count/gate parity) --------------------
+void moe_gate_topk(const float* logits, float* topv, int* topi, float* me,
+                   float* ce, int64_t t, int64_t e, int64_t k, hipStream_t s);
+void moe_assign_slots(const int* topi, int* slot_of, int* counts, int64_t t,
+                      int64_t e, int64_t k, int64_t cap, hipStream_t s);
+
 }  // namespace pa

@@ -40,6 +40,35 @@ class _AllToAll(torch.autograd.Function):
         return out, None
 
 
+class _FusedGate(torch.autograd.Function):
+    """Fused softmax+top-k+stats on the gfx950 kernel (csrc/kernels/moe.hip,
+    reference gshard_gate.py math); backward re-derives the softmax
+    jacobian for the selected probs and the aux-loss me-term."""
+
+    @staticmethod
+    def forward(ctx, logits, k, num_experts):
+        from .. import _ext
+        C = _ext.get_ext()
+        topv, topi, me, ce = C.moe_gate_topk(logits.contiguous(), k)
+        ctx.save_for_backward(logits, topi, ce)
+        ctx.k, ctx.E = k, num_experts
+        ctx.mark_non_differentiable(topi)
+        aux = (me * ce).sum() * num_experts
+        return topv, topi, aux
+
+    @staticmethod
+    def backward(ctx, dtopv, _dtopi, daux):
+        logits, topi, ce = ctx.saved_tensors
+        T = logits.shape[0]
+        p = torch.softmax(logits, -1)
+        # dL/dp = scatter(dtopv at topi) + daux * E/T * ce  (me = mean p)
+        g = torch.zeros_like(p).scatter_(1, topi.long(), dtopv)
+        if daux is not None:
+            g = g + daux * (ctx.E / T) * ce.unsqueeze(0)
+        dlogits = p * (g - (g * p).sum(-1, keepdim=True))
+        return dlogits, None, None
+
+
 class TopKGate(nn.Layer):
     """Switch (k=1) / GShard (k=2) style gate with capacity + aux loss."""
 
@@ -52,8 +81,12 @@ class TopKGate(nn.Layer):
         _apply_initializer(Normal(0.0, 0.02), self.wg.weight)
 
     def forward(self, x):
+        from .. import _ext
         # x: [tokens, h]; gate math in fp32 regardless of model dtype
         logits = torch.matmul(x.float(), self.wg.weight.float())
+        if (_ext.use_native(x) and self.num_experts <= 64 and self.k <= 4):
+            topv, topi, aux = _FusedGate.apply(logits, self.k, self.num_experts)
+            return topv, topi.long(), aux
         probs = torch.softmax(logits, -1)
         topv, topi = probs.topk(self.k, dim=-1)            # [T, k]
         # aux load-balance loss (gshard): num_experts * sum(me * ce)
@@ -136,23 +169,29 @@ class MoELayer(nn.Layer):
         dispatch = torch.zeros(self.num_experts, cap, h, dtype=x.dtype, device=x.device)
         combine_w = torch.zeros(T, self.k, dtype=x.dtype, device=x.device)
         slot_of = torch.full((T, self.k), -1, dtype=torch.long, device=x.device)
+        from .. import _ext
         with torch.no_grad():
-            for kk in range(self.k):
-                e = topi[:, kk]
-                # position of each token within its expert queue
-                pos = torch.zeros_like(e)
-                counts = torch.zeros(self.num_experts, dtype=torch.long, device=x.device)
-                # stable per-expert positions via sort
-                order = torch.argsort(e, stable=True)
-                sorted_e = e[order]
-                ones = torch.ones_like(sorted_e)
-                seg_start = torch.searchsorted(sorted_e, torch.arange(
-                    self.num_experts, device=x.device))
-                idx_in_seg = torch.arange(T, device=x.device) - seg_start[sorted_e]
-                pos[order] = idx_in_seg
-                keep = pos < cap
-                slot = e * cap + pos
-                slot_of[:, kk] = torch.where(keep, slot, torch.full_like(slot, -1))
+            if _ext.use_native(x) and self.num_experts <= 64:
+                # one ballot-prefix kernel (assign_pos/number_count parity)
+                C = _ext.get_ext()
+                slots, _counts = C.moe_assign_slots(
+                    topi.to(torch.int32).contiguous(), self.num_experts, cap)
+                slot_of = slots.long()
+            else:
+                for kk in range(self.k):
+                    e = topi[:, kk]
+                    # position of each token within its expert queue
+                    pos = torch.zeros_like(e)
+                    order = torch.argsort(e, stable=True)
+                    sorted_e = e[order]
+                    seg_start = torch.searchsorted(sorted_e, torch.arange(
+                        self.num_experts, device=x.device))
+                    idx_in_seg = torch.arange(T, device=x.device) - seg_start[sorted_e]
+                    pos[order] = idx_in_seg
+                    keep = pos < cap
+                    slot = e * cap + pos
+                    slot_of[:, kk] = torch.where(keep, slot,
+                                                 torch.full_like(slot, -1))
         flat_dispatch = dispatch.reshape(-1, h)
         for kk in range(self.k):
             valid = slot_of[:, kk] >= 0
