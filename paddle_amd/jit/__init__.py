@@ -1,12 +1,18 @@
 """paddle.jit parity surface (reference: python/paddle/jit/).
 
-Round-1 scope: to_static is a capture/no-op wrapper preserving eager
-semantics (the reference's SOT/AST machinery is a phase-7 target --
-SURVEY.md §2.4); save/load serialize a Layer's state plus a pickled
-forward spec so TranslatedLayer-style reload works for inference.
+save/load: the layer's forward is captured by tracing through the torch
+substrate with the HIP-native dispatch disabled (so the recorded graph
+contains only substrate ops and is executable anywhere), serialized as a
+TorchScript archive in the reference's file layout (path.pdmodel program
+meta + path.pdiparams reference-form weights + path.pdscript program).
+jit.load returns an EXECUTABLE TranslatedLayer -- no model class needed
+(reference python/paddle/jit/api.py save/load, static/io.py:513 layout).
+to_static remains an eager-preserving wrapper (no SOT/AST bytecode pass;
+SURVEY §7 scopes the compiler out).
 """
 from __future__ import annotations
 
+import contextlib
 import os
 import pickle
 
@@ -51,33 +57,109 @@ def ignore_module(modules):
     pass
 
 
+@contextlib.contextmanager
+def _substrate_only():
+    """Trace with HIP-native dispatch off AND the op wrappers in trace
+    mode (plain substrate expressions; autograd.Function wrappers are not
+    TorchScript-exportable)."""
+    from .. import framework
+    from ..ops import functional as hot
+    old = framework.get_flag("FLAGS_use_native_kernels")
+    framework.set_flags({"FLAGS_use_native_kernels": False})
+    hot._TRACE_SUBSTRATE = True
+    try:
+        yield
+    finally:
+        hot._TRACE_SUBSTRATE = False
+        framework.set_flags({"FLAGS_use_native_kernels": old})
+
+
+def _example_inputs(layer, input_spec):
+    if input_spec is None:
+        raise ValueError(
+            "jit.save needs input_spec (paddle.static.InputSpec list or "
+            "example tensors) to capture the program")
+    ex = []
+    from ..static import InputSpec
+    from .. import framework
+    dev = next((p.device for p in layer.parameters()), torch.device("cpu")) \
+        if hasattr(layer, "parameters") else torch.device("cpu")
+    dtype0 = next((p.dtype for p in layer.parameters()), torch.float32) \
+        if hasattr(layer, "parameters") else torch.float32
+    for spec in input_spec:
+        if isinstance(spec, torch.Tensor):
+            ex.append(spec)
+            continue
+        shape = [1 if (d is None or d < 0) else d for d in spec.shape]
+        dt = framework.convert_dtype(spec.dtype) if isinstance(spec.dtype, str) \
+            else spec.dtype
+        if dt in (torch.int64, torch.int32):
+            ex.append(torch.zeros(shape, dtype=dt, device=dev))
+        else:
+            ex.append(torch.randn(shape, device=dev).to(
+                dt if dt is not None else dtype0))
+    return tuple(ex)
+
+
 def save(layer, path, input_spec=None, **configs):
-    """Save a Layer for later jit.load: state_dict (.pdiparams naming kept)."""
+    """Capture + serialize a Layer (or function) for class-free reload."""
     from ..framework_io import save as fsave
     os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
-    state = layer.state_dict() if hasattr(layer, "state_dict") else {}
+    target = layer._fn if isinstance(layer, StaticFunction) else layer
+    if isinstance(layer, StaticFunction) and input_spec is None:
+        input_spec = layer.input_spec
+    state = target.state_dict() if hasattr(target, "state_dict") else {}
     fsave(state, path + ".pdiparams")
     meta = {
-        "class_module": type(layer).__module__,
-        "class_name": type(layer).__name__,
+        "class_module": type(target).__module__,
+        "class_name": type(target).__name__,
+        "format": "torchscript",
+        "program_file": os.path.basename(path) + ".pdscript",
     }
+    # capture the program: trace through the substrate
+    was_training = getattr(target, "training", False)
+    try:
+        if hasattr(target, "eval"):
+            target.eval()
+        ex = _example_inputs(target, input_spec)
+        with _substrate_only(), torch.no_grad():
+            traced = torch.jit.trace(target, ex, strict=False, check_trace=False)
+        torch.jit.save(traced, path + ".pdscript")
+    except Exception as e:  # capture failure: keep state-only save
+        meta["format"] = "state_only"
+        meta["capture_error"] = repr(e)
+    finally:
+        if was_training and hasattr(target, "train"):
+            target.train()
     with open(path + ".pdmodel", "wb") as f:
         pickle.dump(meta, f, protocol=2)
 
 
 class TranslatedLayer(torch.nn.Module):
-    def __init__(self, state, meta):
+    """Executable reloaded program (reference jit/translated_layer.py)."""
+
+    def __init__(self, state, meta, program=None):
         super().__init__()
         self._state = state
         self._meta = meta
+        self._program = program
 
     def forward(self, *args, **kwargs):
-        raise RuntimeError(
-            "TranslatedLayer from jit.load is a state container in this build; "
-            "reconstruct the Layer class and call set_state_dict(layer_state())")
+        if self._program is None:
+            raise RuntimeError(
+                "this jit.save archive has no captured program "
+                f"(capture_error={self._meta.get('capture_error')!r}); "
+                "reconstruct the Layer class and set_state_dict(layer_state())")
+        return self._program(*args, **kwargs)
 
     def layer_state(self):
         return self._state
+
+    def eval(self):
+        if self._program is not None:
+            self._program.eval()
+        return super().eval()
+
 
 
 def load(path, **configs):
@@ -85,7 +167,13 @@ def load(path, **configs):
     state = fload(path + ".pdiparams")
     with open(path + ".pdmodel", "rb") as f:
         meta = pickle.load(f)
-    return TranslatedLayer(state, meta)
+    program = None
+    if meta.get("format") == "torchscript":
+        pfile = os.path.join(os.path.dirname(path) or ".",
+                             meta["program_file"])
+        if os.path.exists(pfile):
+            program = torch.jit.load(pfile, map_location="cpu")
+    return TranslatedLayer(state, meta, program)
 
 
 def enable_to_static(flag=True):

@@ -16,6 +16,15 @@ import torch
 
 from .. import _ext
 
+# jit.save tracing mode: emit plain substrate ops (no autograd.Function
+# wrappers -- TorchScript cannot export Python function calls).  Forward
+# inference semantics only; toggled by paddle_amd.jit._substrate_only.
+_TRACE_SUBSTRATE = False
+
+
+def _tracing():
+    return _TRACE_SUBSTRATE
+
 
 # ---------------------------------------------------------------------------
 # layer_norm (paddle/phi/kernels/gpu/layer_norm_kernel.cu parity)
@@ -63,6 +72,9 @@ class _LayerNorm(torch.autograd.Function):
 
 
 def layer_norm(x, weight, bias=None, epsilon=1e-5):
+    if _tracing():
+        return torch.nn.functional.layer_norm(x, (x.shape[-1],), weight, bias,
+                                              epsilon)
     return _LayerNorm.apply(x, weight, bias, epsilon)
 
 
@@ -102,6 +114,10 @@ class _RMSNorm(torch.autograd.Function):
 
 
 def rms_norm(x, weight, epsilon=1e-6):
+    if _tracing():
+        xf = x.float()
+        y = xf * torch.rsqrt(xf.square().mean(-1, keepdim=True) + epsilon)
+        return (y * weight.float()).to(x.dtype)
     return _RMSNorm.apply(x, weight, epsilon)
 
 
@@ -143,6 +159,9 @@ class _FusedRMSNormResidual(torch.autograd.Function):
 
 
 def fused_rms_norm(x, norm_weight, residual=None, epsilon=1e-6):
+    if _tracing() and residual is not None:
+        xr = x + residual
+        return rms_norm(xr, norm_weight, epsilon), xr
     if residual is None:
         return rms_norm(x, norm_weight, epsilon)
     return _FusedRMSNormResidual.apply(x, residual, norm_weight, epsilon)
@@ -191,7 +210,13 @@ class _SoftmaxCE(torch.autograd.Function):
 
 
 def softmax_cross_entropy(logits, labels, ignore_index=-100, reduction="none"):
-    loss = _SoftmaxCE.apply(logits, labels, ignore_index)
+    if _tracing():
+        v = logits.shape[-1]
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, v).float(), labels.reshape(-1),
+            ignore_index=ignore_index, reduction="none").reshape(labels.shape)
+    else:
+        loss = _SoftmaxCE.apply(logits, labels, ignore_index)
     if reduction == "mean":
         n_valid = (labels != ignore_index).sum().clamp(min=1)
         return loss.sum() / n_valid.to(loss.dtype)
@@ -233,6 +258,8 @@ class _BiasGelu(torch.autograd.Function):
 
 
 def bias_gelu(x, bias=None):
+    if _tracing():
+        return torch.nn.functional.gelu(x + bias if bias is not None else x)
     return _BiasGelu.apply(x, bias)
 
 
@@ -267,6 +294,12 @@ class _SwiGLU(torch.autograd.Function):
 
 
 def swiglu(x, y=None):
+    if _tracing():
+        if y is None:
+            g, u = x.chunk(2, -1)
+        else:
+            g, u = x, y
+        return torch.nn.functional.silu(g) * u
     if y is not None:
         x = torch.cat([x, y], dim=-1)
     return _SwiGLU.apply(x)
@@ -325,6 +358,19 @@ def _rope_ref(x, cos_t, sin_t, pos_offset, conj):
 def fused_rotary_position_embedding(q, k=None, v=None, sin=None, cos=None,
                                     position_ids=None, use_neox_rotary_style=True,
                                     base=10000.0, pos_offset=0):
+    if _tracing():
+        b, s, h, d = q.shape
+        if cos is None or sin is None:
+            cos_t, sin_t = build_rope_cache(s + pos_offset, d, base, q.device)
+        else:
+            cos_t = cos.float().reshape(-1, d)[..., : d // 2]
+            sin_t = sin.float().reshape(-1, d)[..., : d // 2]
+        outs = [_rope_ref(q.float(), cos_t, sin_t, pos_offset, False).to(q.dtype)]
+        if k is not None:
+            outs.append(_rope_ref(k.float(), cos_t, sin_t, pos_offset, False).to(k.dtype))
+        if v is not None:
+            outs.append(v)
+        return tuple(outs) if len(outs) > 1 else outs[0]
     """paddle.incubate.nn.functional.fused_rotary_position_embedding parity
     (SURVEY.md A.7).  q/k/v: [B, S, H, D]."""
     bq, s, hq, d = q.shape
@@ -489,6 +535,14 @@ def qkv_flash_attention(qkv, scale=None, causal=True, dropout=0.0,
     """qkv: [B, S, 3, H, D] packed (straight out of the fused QKV GEMM)."""
     if scale is None:
         scale = 1.0 / math.sqrt(qkv.shape[-1])
+    if _tracing():
+        b, s, three, h, d = qkv.shape
+        q = qkv[:, :, 0].permute(0, 2, 1, 3)
+        k = qkv[:, :, 1].permute(0, 2, 1, 3)
+        v = qkv[:, :, 2].permute(0, 2, 1, 3)
+        o = torch.nn.functional.scaled_dot_product_attention(
+            q, k, v, is_causal=causal, scale=scale)
+        return o.permute(0, 2, 1, 3).reshape(b, s, h * d)
     p = dropout if training else 0.0
     seed, offset = _fa_seed_offset() if p > 0 else (0, 0)
     return _QKVFlashAttn.apply(qkv, scale, causal, p, seed, offset)
@@ -579,6 +633,14 @@ def flash_attention(q, k, v, dropout=0.0, causal=False, scale=None,
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     p = dropout if training else 0.0
+    if _tracing():
+        qt, kt, vt = (t.transpose(1, 2) for t in (q, k, v)) \
+            if layout == "bshd" else (q, k, v)
+        o = torch.nn.functional.scaled_dot_product_attention(
+            qt, kt, vt, attn_mask=attn_mask, is_causal=causal and attn_mask is None,
+            scale=scale, enable_gqa=kt.shape[1] != qt.shape[1])
+        o = o.transpose(1, 2) if layout == "bshd" else o
+        return o, None
     seed, offset = _fa_seed_offset(fixed_seed_offset) if p > 0 else (0, 0)
     if layout == "bshd":
         # strided views -- the kernel is stride-aware, no copies
@@ -664,6 +726,8 @@ class _DropoutAdd(torch.autograd.Function):
 
 
 def dropout_add(x, residual=None, p=0.0, training=True):
+    if _tracing():
+        return x + residual if residual is not None else x
     return _DropoutAdd.apply(x, residual, p, training)
 
 
@@ -704,6 +768,8 @@ class _Embedding(torch.autograd.Function):
 
 
 def embedding(ids, table, padding_idx=None):
+    if _tracing():
+        return torch.nn.functional.embedding(ids, table, padding_idx)
     return _Embedding.apply(ids, table, padding_idx)
 
 

@@ -286,14 +286,119 @@ class InputSpec:
         return cls(list(tensor.shape), str(tensor.dtype), name)
 
 
+class _ProgramModule(torch.nn.Module):
+    """Evaluates a captured static Program feed->fetch as a torch Module
+    so torch.jit.trace can serialize it (the Var DAG's op fns execute
+    real substrate ops under tracing)."""
+
+    def __init__(self, feed_vars, fetch_vars):
+        super().__init__()
+        self._feeds = feed_vars
+        self._fetches = fetch_vars
+        for i, p in enumerate(_collect_params(fetch_vars)):
+            self.register_parameter(f"p{i}", torch.nn.Parameter(
+                p.tensor.detach().clone(), requires_grad=False))
+            p._traced_tensor = getattr(self, f"p{i}")
+
+    def forward(self, *inputs):
+        env = {fv.name: t for fv, t in zip(self._feeds, inputs)}
+        memo = {}
+        outs = [_eval_traced(v, env, memo) for v in self._fetches]
+        return tuple(outs) if len(outs) > 1 else outs[0]
+
+
+def _collect_params(fetch_vars):
+    seen, out = set(), []
+
+    def walk(v):
+        if not isinstance(v, Var) or id(v) in seen:
+            return
+        seen.add(id(v))
+        if v.kind == "param":
+            out.append(v)
+        for a in list(v.args) + list(v.kwargs.values()):
+            walk(a)
+    for f in fetch_vars:
+        walk(f)
+    return out
+
+
+def _eval_traced(v, env, memo):
+    key = id(v)
+    if key in memo:
+        return memo[key]
+    if v.kind == "data":
+        r = env[v.name]
+    elif v.kind == "param":
+        r = getattr(v, "_traced_tensor", v.tensor)
+    else:
+        args = [_eval_traced(a, env, memo) if isinstance(a, Var) else a
+                for a in v.args]
+        kwargs = {k: (_eval_traced(x, env, memo) if isinstance(x, Var) else x)
+                  for k, x in v.kwargs.items()}
+        r = v.fn(*args, **kwargs)
+    memo[key] = r
+    return r
+
+
 def save_inference_model(path_prefix, feed_vars, fetch_vars, executor,
                          program=None, **kwargs):
-    raise NotImplementedError(
-        "save_inference_model requires the jit capture path; use paddle.jit.save")
+    """Serialize the captured deferred graph for class-free inference
+    reload (reference python/paddle/static/io.py:513 file layout:
+    .pdmodel program + .pdiparams weights)."""
+    import os as _os
+    import pickle as _pickle
+    from .. import framework_io
+    feed_vars = feed_vars if isinstance(feed_vars, (list, tuple)) else [feed_vars]
+    fetch_vars = fetch_vars if isinstance(fetch_vars, (list, tuple)) else [fetch_vars]
+    mod = _ProgramModule(feed_vars, fetch_vars).eval()
+    ex = []
+    for fv in feed_vars:
+        shape = [1 if (d is None or (isinstance(d, int) and d < 0)) else d
+                 for d in (fv.shape or [1])]
+        ex.append(torch.zeros(shape, dtype=fv.dtype or torch.float32))
+    with torch.no_grad():
+        traced = torch.jit.trace(mod, tuple(ex), strict=False, check_trace=False)
+    _os.makedirs(_os.path.dirname(path_prefix) or ".", exist_ok=True)
+    torch.jit.save(traced, path_prefix + ".pdscript")
+    framework_io.save({f"p{i}": p.tensor for i, p in
+                       enumerate(_collect_params(fetch_vars))},
+                      path_prefix + ".pdiparams")
+    meta = {"format": "torchscript",
+            "program_file": _os.path.basename(path_prefix) + ".pdscript",
+            "feed_names": [fv.name for fv in feed_vars],
+            "n_fetch": len(fetch_vars)}
+    with open(path_prefix + ".pdmodel", "wb") as f:
+        _pickle.dump(meta, f, protocol=2)
 
 
 def load_inference_model(path_prefix, executor, **kwargs):
-    raise NotImplementedError("use paddle.jit.load")
+    """Returns (program_callable, feed_names, fetch_count) -- run via
+    executor.run(program, feed={...}, fetch_list=...) or call directly."""
+    import os as _os
+    import pickle as _pickle
+    with open(path_prefix + ".pdmodel", "rb") as f:
+        meta = _pickle.load(f)
+    prog = torch.jit.load(
+        _os.path.join(_os.path.dirname(path_prefix) or ".",
+                      meta["program_file"]), map_location="cpu")
+
+    class _LoadedProgram:
+        def __init__(self, mod, feed_names, n_fetch):
+            self._mod = mod
+            self.feed_names = feed_names
+            self.n_fetch = n_fetch
+
+        def __call__(self, *inputs):
+            return self._mod(*inputs)
+
+        def run(self, feed, fetch_list=None):
+            ins = [feed[n] for n in self.feed_names]
+            out = self._mod(*ins)
+            return list(out) if isinstance(out, tuple) else [out]
+
+    return (_LoadedProgram(prog, meta["feed_names"], meta["n_fetch"]),
+            meta["feed_names"], meta["n_fetch"])
 
 
 def gradients(targets, inputs, target_gradients=None):
