@@ -90,3 +90,128 @@ class ElasticManager:
             os.remove(self._my_key())
         except OSError:
             pass
+
+
+# ---------------------------------------------------------------------------
+# scale in/out (reference fleet/elastic/manager.py:484 _update_elastic_scale_out
+# / :507 _update_elastic_scale_in): membership changes within [min_np, max_np]
+# produce a new world plan; the agent relaunches workers with the new size.
+# ---------------------------------------------------------------------------
+def parse_np_range(np_spec):
+    """'2' -> (2, 2); '2:8' -> (2, 8) (reference --np min:max form)."""
+    if isinstance(np_spec, int):
+        return np_spec, np_spec
+    s = str(np_spec)
+    if ":" in s:
+        lo, hi = s.split(":")
+        return int(lo), int(hi)
+    return int(s), int(s)
+
+
+class ScalePlan:
+    def __init__(self, action, world, hosts):
+        self.action = action      # "none" | "scale_out" | "scale_in" | "wait"
+        self.world = world
+        self.hosts = hosts
+
+    def __repr__(self):
+        return f"ScalePlan({self.action}, world={self.world}, hosts={self.hosts})"
+
+
+class ElasticScaler:
+    """Turns membership snapshots into scale decisions with a debounce
+    window so transient heartbeat blips don't thrash restarts."""
+
+    def __init__(self, manager: "ElasticManager", np_spec, debounce=2.0):
+        self.mgr = manager
+        self.min_np, self.max_np = parse_np_range(np_spec)
+        self.debounce = debounce
+        self.current_world = None
+        self._pending = None
+        self._pending_since = 0.0
+
+    def plan(self, hosts=None, now=None) -> ScalePlan:
+        hosts = self.mgr.hosts() if hosts is None else sorted(hosts)
+        now = time.time() if now is None else now
+        n = len(hosts)
+        if self.current_world is None:
+            world = max(self.min_np, min(n, self.max_np))
+            return ScalePlan("none", world, hosts[:world])
+        if n == self.current_world:
+            self._pending = None
+            return ScalePlan("none", self.current_world, hosts)
+        # debounce: a change must persist for `debounce` seconds
+        if self._pending != n:
+            self._pending = n
+            self._pending_since = now
+            return ScalePlan("wait", self.current_world, hosts)
+        if now - self._pending_since < self.debounce:
+            return ScalePlan("wait", self.current_world, hosts)
+        self._pending = None
+        if n > self.current_world and self.current_world < self.max_np:
+            world = min(n, self.max_np)
+            return ScalePlan("scale_out", world, hosts[:world])
+        if n < self.current_world:
+            if n >= self.min_np:
+                return ScalePlan("scale_in", n, hosts)
+            return ScalePlan("wait", self.current_world, hosts)  # below min: hold
+        return ScalePlan("none", self.current_world, hosts)
+
+    def commit(self, plan: ScalePlan):
+        self.current_world = plan.world
+
+
+class LocalElasticAgent:
+    """Launcher-tier elastic agent: runs `nproc` local workers, watches
+    membership, and on a committed scale decision relaunches the worker
+    set at the new world size (restart-with-checkpoint-resume semantics,
+    the reference's recovery tier 2)."""
+
+    def __init__(self, manager, np_spec, entry_args, debounce=2.0,
+                 python=None):
+        import sys as _sys
+        self.mgr = manager
+        self.scaler = ElasticScaler(manager, np_spec, debounce)
+        self.entry_args = entry_args        # argv after `python`
+        self.python = python or _sys.executable
+        self.procs = []
+        self.restarts = 0
+
+    def _launch(self, world):
+        import subprocess
+        self._stop_workers()
+        env_base = dict(os.environ)
+        env_base["PADDLE_ELASTIC_RESTART"] = str(self.restarts)
+        for r in range(world):
+            env = dict(env_base)
+            env.update({"RANK": str(r), "WORLD_SIZE": str(world),
+                        "MASTER_ADDR": "127.0.0.1",
+                        "PADDLE_TRAINER_ID": str(r),
+                        "PADDLE_TRAINERS_NUM": str(world)})
+            self.procs.append(__import__("subprocess").Popen(
+                [self.python] + self.entry_args, env=env))
+
+    def _stop_workers(self):
+        for p in self.procs:
+            if p.poll() is None:
+                p.terminate()
+        for p in self.procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                p.kill()
+        self.procs = []
+
+    def step(self, hosts=None, now=None):
+        """One watch iteration; returns the plan taken."""
+        plan = self.scaler.plan(hosts, now)
+        if plan.action in ("scale_out", "scale_in") or \
+                (plan.action == "none" and self.scaler.current_world is None):
+            self.scaler.commit(plan)
+            self.restarts += (self.restarts >= 0 and plan.action != "none")
+            self._launch(plan.world)
+        return plan
+
+    def shutdown(self):
+        self._stop_workers()
+        self.mgr.exit()

@@ -85,3 +85,61 @@ def test_auto_tuner_records_best(tmp_path):
     best = tuner.tune()
     assert best["trial"]["micro_batch"] == 2
     assert best["tokens_per_s"] == 200.0
+
+
+def test_elastic_scale_in_out_plan(tmp_path):
+    """membership scale in/out decisions (reference
+    manager.py:484/:507), with debounce and min/max bounds."""
+    from paddle_amd.distributed.elastic import (ElasticManager, ElasticScaler,
+                                                parse_np_range)
+    assert parse_np_range("2:8") == (2, 8)
+    assert parse_np_range(4) == (4, 4)
+    mgr = ElasticManager(job_id="t1", store_dir=str(tmp_path))
+    sc = ElasticScaler(mgr, "2:4", debounce=1.0)
+    # initial formation at 2 hosts
+    p = sc.plan(hosts=["a", "b"], now=100.0)
+    assert p.action == "none" and p.world == 2
+    sc.commit(p)
+    # a third host appears: debounced, then scale_out
+    assert sc.plan(hosts=["a", "b", "c"], now=101.0).action == "wait"
+    p = sc.plan(hosts=["a", "b", "c"], now=102.5)
+    assert p.action == "scale_out" and p.world == 3
+    sc.commit(p)
+    # blip: host drops for < debounce then returns -> no restart
+    assert sc.plan(hosts=["a", "b"], now=103.0).action == "wait"
+    assert sc.plan(hosts=["a", "b", "c"], now=103.5).action == "none"
+    # real loss, persists: scale_in
+    assert sc.plan(hosts=["a", "c"], now=104.0).action == "wait"
+    p = sc.plan(hosts=["a", "c"], now=105.5)
+    assert p.action == "scale_in" and p.world == 2
+    sc.commit(p)
+    # below min: hold (fault-tolerance tier takes over, no shrink below min)
+    sc.plan(hosts=["a"], now=106.0)
+    assert sc.plan(hosts=["a"], now=108.0).action == "wait"
+    # beyond max: capped
+    sc.plan(hosts=list("abcdef"), now=109.0)
+    p = sc.plan(hosts=list("abcdef"), now=111.0)
+    assert p.action == "scale_out" and p.world == 4
+
+
+def test_local_elastic_agent_relaunches(tmp_path):
+    import time
+    from paddle_amd.distributed.elastic import ElasticManager, LocalElasticAgent
+    marker = tmp_path / "runs.log"
+    prog = (f"import os; open(r'{marker}', 'a').write("
+            "os.environ['WORLD_SIZE'] + ':' + os.environ['RANK'] + ':' "
+            "+ os.environ['PADDLE_ELASTIC_RESTART'] + '\\n')")
+    mgr = ElasticManager(job_id="t2", store_dir=str(tmp_path / "s"))
+    agent = LocalElasticAgent(mgr, "1:3", ["-c", prog], debounce=0.5)
+    agent.step(hosts=["a", "b"], now=1.0)          # initial world 2
+    for p in agent.procs:                          # let world-2 run finish
+        p.wait(timeout=20)
+    agent.step(hosts=["a", "b", "c"], now=2.0)     # wait
+    agent.step(hosts=["a", "b", "c"], now=3.0)     # scale_out -> world 3
+    for p in agent.procs:
+        p.wait(timeout=20)
+    agent.shutdown()
+    lines = marker.read_text().strip().splitlines()
+    w2 = [l for l in lines if l.startswith("2:")]
+    w3 = [l for l in lines if l.startswith("3:")]
+    assert len(w2) >= 1 and len(w3) == 3, lines
