@@ -15,6 +15,7 @@ from .pipeline import (  # noqa: F401
     SharedLayerDesc,
     VirtualPipelineLayer,
     ZeroBubblePipelineParallel,
+    ZeroBubbleInterleavedPipelineParallel,
 )
 from .random import RNGStatesTracker, get_rng_state_tracker  # noqa: F401
 from .sharding import (  # noqa: F401

@@ -575,3 +575,38 @@ class ZeroBubblePipelineParallel(PipelineParallel):
             _flush_w(self._w_store)
             return inner_ar()
         self._layers.allreduce_shared_weight_gradients = ar_with_flush
+
+
+class ZeroBubbleInterleavedPipelineParallel(InterleavedPipelineParallel):
+    """ZB-VPP (reference passes/pipeline_scheduler_pass/__init__.py:33-38
+    registry): the interleaved virtual-pipeline schedule with backward
+    split into B (input-grad, stays on the p2p relay) and W (weight-grad
+    GEMMs deferred into bubbles and drained before optimizer.step).
+    Gradients are bit-identical to plain interleaved VPP."""
+
+    def __init__(self, layers: VirtualPipelineLayer, hcg, strategy=None):
+        super().__init__(layers, hcg, strategy)
+        self._w_store = []
+        self._n_zb = _convert_to_zb(layers, self._w_store)
+        self.p2p = _ZBP2P(self.p2p, self._w_store)
+
+    def train_batch(self, data, optimizer, lr_scheduler=None, scaler=None):
+        store = self._w_store
+
+        class _FlushingOpt:
+            """Drains the deferred W phase before the real step."""
+
+            def __init__(self, opt):
+                self._opt = opt
+
+            def step(self):
+                _flush_w(store)
+                self._opt.step()
+
+            def __getattr__(self, name):
+                return getattr(self._opt, name)
+
+        wrapped = _FlushingOpt(optimizer) if optimizer is not None else None
+        out = super().train_batch(data, wrapped, lr_scheduler, scaler)
+        _flush_w(store)   # eval/no-opt paths: leave nothing queued
+        return out
