@@ -200,6 +200,21 @@ def reshard(dist_tensor, mesh: ProcessMesh, placements):
                 dst = torch.empty_like(shards[0])
                 C.reduce_scatter(dst, [s.contiguous() for s in shards], group=g)
                 out = dst
+        elif isinstance(sp, Shard) and isinstance(dp, Shard) and sp.dim != dp.dim:
+            # s_to_s (reference reshard/s_to_s_reshard_function.cc): re-shard
+            # from dim a to dim b with one all-to-all -- no gather round-trip
+            n = mesh.shape[mdim]
+            if g is not None and n > 1:
+                pieces = [p.contiguous() for p in out.chunk(n, dim=dp.dim)]
+                recv = [torch.empty_like(pieces[0]) for _ in range(n)]
+                C.alltoall(pieces, recv, group=g)
+                out = torch.cat(recv, dim=sp.dim)
+        elif isinstance(sp, Replicate) and isinstance(dp, Partial):
+            # r_to_p (reference r_to_p_reshard_function.cc): rank 0 keeps the
+            # value, others zero -- the sum over the mesh dim reproduces it
+            coord = mesh.my_coord()
+            idx = coord[mdim] if coord is not None else 0
+            out = out if idx == 0 else torch.zeros_like(out)
         else:
             raise NotImplementedError(f"reshard {sp} -> {dp}")
     out = out.clone() if out is dist_tensor else out
@@ -446,3 +461,101 @@ class Engine:
                 if steps and i + 1 >= steps:
                     break
         return outs
+
+
+def dist_embedding(ids, weight):
+    """Vocab-sharded embedding (reference c_embedding_op + spmd rule
+    phi/infermeta/spmd_rules/c_embedding.cc): each rank looks up its vocab
+    slice (OOV rows -> 0) and the output is Partial over the sharding
+    mesh dim (reshard/allreduce completes it)."""
+    mesh = getattr(weight, "process_mesh", None)
+    assert mesh is not None
+    pw = _placements_of(weight, mesh)
+    out_pl = []
+    local = None
+    for mdim, p in enumerate(pw):
+        if isinstance(p, Shard) and p.dim == 0:
+            n = mesh.shape[mdim]
+            coord = mesh.my_coord()
+            idx = coord[mdim] if coord is not None else 0
+            per = weight.shape[0]
+            start = idx * per
+            shifted = ids - start
+            ok = (shifted >= 0) & (shifted < per)
+            local = torch.nn.functional.embedding(shifted.clamp(0, per - 1),
+                                                  weight)
+            local = local * ok.unsqueeze(-1).to(local.dtype)
+            out_pl.append(Partial())
+        elif isinstance(p, Shard):          # hidden-dim sharded
+            out_pl.append(Shard(ids.dim()))
+        else:
+            out_pl.append(Replicate())
+    if local is None:
+        local = torch.nn.functional.embedding(ids, weight)
+    local.dist_attr = DistAttr(mesh, out_pl)
+    local.process_mesh = mesh
+    local.placements = out_pl
+    return local
+
+
+def dist_cross_entropy(logits, labels):
+    """Vocab-parallel softmax cross entropy (reference
+    c_softmax_with_cross_entropy_op.cu + its spmd rule): logits sharded on
+    the class dim stay sharded; the loss comes out Replicated via the
+    max/sum/logit allreduce triplet."""
+    mesh = getattr(logits, "process_mesh", None)
+    assert mesh is not None
+    pl = _placements_of(logits, mesh)
+    vdim = logits.dim() - 1
+    shard_mdim = next((m for m, p in enumerate(pl)
+                       if isinstance(p, Shard) and p.dim == vdim), None)
+    if shard_mdim is None:
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, logits.shape[-1]).float(), labels.reshape(-1),
+            reduction="none").reshape(labels.shape)
+        loss.dist_attr = DistAttr(mesh, [Replicate()] * mesh.ndim)
+        loss.process_mesh = mesh
+        loss.placements = [Replicate()] * mesh.ndim
+        return loss
+    g = _mesh_group(mesh, shard_mdim)
+    n = mesh.shape[shard_mdim]
+    coord = mesh.my_coord()
+    idx = coord[shard_mdim] if coord is not None else 0
+    per = logits.shape[-1]
+    lf = logits.float().reshape(-1, per)
+    lab = labels.reshape(-1)
+    mx = lf.max(-1).values
+    if g is not None:
+        C.all_reduce(mx, op=C.ReduceOp.MAX, group=g)
+    ex = torch.exp(lf - mx.unsqueeze(1))
+    den = ex.sum(-1)
+    if g is not None:
+        C.all_reduce(den, group=g)
+    shifted = lab - idx * per
+    ok = (shifted >= 0) & (shifted < per)
+    picked = lf.gather(1, shifted.clamp(0, per - 1).unsqueeze(1)).squeeze(1)
+    picked = torch.where(ok, picked, torch.zeros_like(picked))
+    if g is not None:
+        C.all_reduce(picked, group=g)
+    loss = (torch.log(den) + mx - picked).reshape(labels.shape)
+    out_pl = [Replicate()] * mesh.ndim
+    loss.dist_attr = DistAttr(mesh, out_pl)
+    loss.process_mesh = mesh
+    loss.placements = out_pl
+    return loss
+
+
+def dist_flash_attention(q, k, v, causal=True, scale=None):
+    """Head-sharded flash attention SPMD rule (reference
+    spmd_rules/flash_attention.cc): Shard on the head dim of q/k/v passes
+    straight through (attention is head-wise independent); any seq/batch
+    Partial or contraction conflicts fall back to Replicate inputs."""
+    from ..ops import functional as hot
+    mesh = getattr(q, "process_mesh", None)
+    assert mesh is not None
+    pq = _placements_of(q, mesh)
+    o, _ = hot.flash_attention(q, k, v, causal=causal, scale=scale)
+    o.dist_attr = DistAttr(mesh, list(pq))
+    o.process_mesh = mesh
+    o.placements = list(pq)
+    return o

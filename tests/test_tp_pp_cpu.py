@@ -432,3 +432,92 @@ def test_zero_bubble_vpp_matches_single():
                 assert torch.allclose(a, b2, atol=1e-6)
         print("rank", r, "zbvpp ok")
     """, timeout=300)
+
+
+def test_spmd_broadened_rules_two_ranks():
+    """embedding / cross-entropy / flash-attn SPMD rules
+    + s_to_s and r_to_p reshard; a shard_layer'd GPT-ish stack (vocab-
+    parallel embedding -> head-sharded attention -> TP mlp -> vocab-parallel
+    CE) reproduces the single-process loss."""
+    run_dist("""
+        import math
+        import torch
+        import paddle_amd as paddle
+        from paddle_amd.distributed import fleet
+        strategy = fleet.DistributedStrategy()
+        strategy.hybrid_configs = {"dp_degree": 1, "mp_degree": 2, "pp_degree": 1,
+                                   "sharding_degree": 1}
+        fleet.init(is_collective=True, strategy=strategy)
+        from paddle_amd.distributed import auto_parallel as ap
+        from paddle_amd.distributed.auto_parallel import (
+            ProcessMesh, Shard, Replicate, Partial, shard_tensor, reshard,
+            dist_embedding, dist_cross_entropy, dist_flash_attention,
+            dist_matmul)
+        r = paddle.distributed.get_rank()
+        mesh = ProcessMesh([0, 1], dim_names=["mp"])
+        torch.manual_seed(0)
+
+        # ---- s_to_s reshard: Shard(0) -> Shard(1) via one all-to-all ----
+        full = torch.arange(16.0).reshape(4, 4)
+        x = shard_tensor(full.clone(), mesh, [Shard(0)])
+        y = reshard(x, mesh, [Shard(1)])
+        assert y.shape == (4, 2)
+        assert torch.allclose(y, full[:, r*2:(r+1)*2]), y
+
+        # ---- r_to_p: sum over mesh dim reproduces the value ----
+        z = shard_tensor(full.clone(), mesh, [Replicate()])
+        p = reshard(z, mesh, [Partial()])
+        back = reshard(p, mesh, [Replicate()])
+        assert torch.allclose(back, full)
+
+        # ---- GPT-ish mini stack under SPMD rules vs single process ----
+        V, Hd, Hh, Dh, S = 32, 16, 2, 8, 6
+        torch.manual_seed(1)
+        emb_w = torch.randn(V, Hd) * 0.1
+        wqkv = torch.randn(Hd, 3 * Hd) * 0.1
+        wout = torch.randn(Hd, V) * 0.1
+        ids = torch.randint(0, V, (2, S))
+        labels = torch.randint(0, V, (2, S))
+
+        # single-process reference
+        from paddle_amd.ops.functional import flash_attention
+        h = torch.nn.functional.embedding(ids, emb_w)
+        qkv = h @ wqkv
+        q, k, v = qkv.reshape(2, S, 3, Hh, Dh).unbind(2)
+        att, _ = flash_attention(q, k, v, causal=True)
+        logits = att.reshape(2, S, Hd) @ wout
+        ref_loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, V).float(), labels.reshape(-1), reduction="none")
+
+        # distributed: vocab-sharded embedding -> partial -> replicate
+        emb_l = shard_tensor(emb_w.clone(), mesh, [Shard(0)])
+        assert emb_l.shape == (V // 2, Hd)
+        hd = dist_embedding(ids, emb_l)
+        hd = reshard(hd, mesh, [Replicate()])
+        # column-parallel qkv: W sharded on out dim -> out Shard(last)
+        # reorder so Shard(1) splits by head: [Hd, Hh, 3*Dh]
+        wq_headmajor = wqkv.reshape(Hd, 3, Hh, Dh).permute(0, 2, 1, 3) \
+                            .reshape(Hd, -1).contiguous()
+        wq_l = shard_tensor(wq_headmajor, mesh, [Shard(1)])
+        qkv_d = dist_matmul(hd, wq_l)
+        q_d, k_d, v_d = (t.contiguous() for t in
+                         qkv_d.reshape(2, S, Hh // 2, 3, Dh).unbind(3))
+        for t in (q_d, k_d, v_d):
+            t.process_mesh = mesh
+            t.placements = [Shard(2)]
+        att_d = dist_flash_attention(q_d, k_d, v_d, causal=True)
+        assert att_d.placements == [Shard(2)]
+        # out proj: vocab-sharded -> logits Shard(last); gather heads first
+        att_f = att_d.reshape(2, S, Hd // 2).contiguous()
+        att_f.process_mesh = mesh
+        att_f.placements = [Shard(2)]
+        att_f = reshard(att_f, mesh, [Replicate()])
+        assert att_f.shape == (2, S, Hd)
+        wo_l = shard_tensor(wout.clone(), mesh, [Shard(1)])
+        logits_d = dist_matmul(att_f, wo_l)
+        assert logits_d.placements == [Shard(2)]
+        loss_d = dist_cross_entropy(logits_d, labels)
+        assert torch.allclose(loss_d.reshape(-1), ref_loss, atol=1e-4), \\
+            (loss_d.reshape(-1) - ref_loss).abs().max()
+        print("rank", r, "spmd broadened ok")
+    """, timeout=300)
