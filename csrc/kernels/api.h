@@ -206,6 +206,12 @@ void gemm_fp8_nt(const void* a, const void* bt, void* c, const void* bias,
                  float scale_ab, int64_t m, int64_t n, int64_t k, int64_t lda,
                  int64_t ldb, int64_t ldc, hipStream_t s);
 
+// skinny decode GEMM (M<=32): y = x @ W[K,N] + bias, split-K partials
+// in `workspace` ([ksplit, padded_m, N] fp32)
+void decode_gemm(const void* x, const void* w, const void* bias, void* y,
+                 float* workspace, int64_t m, int64_t n, int64_t k,
+                 int64_t ldw, int64_t ksplit, hipStream_t s);
+
 // ---- MoE routing (assign_pos/number_count/gate parity) --------------------
 void moe_gate_topk(const float* logits, float* topv, int* topi, float* me,
                    float* ce, int64_t t, int64_t e, int64_t k, hipStream_t s);

@@ -31,6 +31,9 @@ def linear(x, weight, bias=None, name=None):
     # hipBLASLt via addmm (bias fused in its epilogue).
     if hot._own_linear_ok(x, weight):
         return hot.fused_linear_own(x, weight, bias)
+    # NOTE: an M<=32 split-K weight-streaming kernel (decode_gemm.hip) was
+    # measured SLOWER in-situ than hipBLASLt's skinny tiles (VALU-bound at
+    # M=32); it stays available via _C.decode_gemm but is not dispatched.
     if bias is not None and x.dim() >= 2:
         x2 = x.reshape(-1, x.shape[-1])
         out = torch.addmm(bias, x2, weight)

@@ -800,3 +800,22 @@ def test_moe_gate_gradients_gpu():
     ce = torch.bincount(ri[:, 0], minlength=E).float() / T
     (rv.square().sum() + 3.0 * (me * ce).sum() * E).backward()
     torch.testing.assert_close(logits.grad, lr.grad, atol=1e-4, rtol=1e-4)
+
+
+# ---------------------------------------------------------------------------
+# skinny decode GEMM (decode_gemm.hip)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("m,k,n", [(1, 4096, 4096), (16, 4096, 16384),
+                                   (32, 16384, 4096), (16, 4096, 50304),
+                                   (7, 2048, 1100)])
+def test_decode_gemm_numerics(m, k, n):
+    torch.manual_seed(50)
+    C = _ext.get_ext()
+    x = _bf(torch.randn(m, k, device=DEV))
+    w = _bf(torch.randn(k, n, device=DEV) * 0.02)
+    bias = _bf(torch.randn(n, device=DEV))
+    y = C.decode_gemm(x, w, bias)
+    ref = x.float() @ w.float() + bias.float()
+    _gemm_rel_ok(y, ref, tol=2e-2)
+    y2 = C.decode_gemm(x, w, None)
+    _gemm_rel_ok(y2, x.float() @ w.float(), tol=2e-2)

@@ -30,12 +30,25 @@ def run_one(tag, model, n_req=64, prompt_len=128, gen_len=128, max_batch=32,
             prompt_ids=[random.randrange(V) for _ in range(prompt_len)],
             max_new_tokens=gen_len))
     t0 = time.perf_counter()
-    eng.run_until_done()
+    eng.step()                      # prefill wave (all same-length prompts)
+    torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    tok0 = sum(len(r.out_ids) for r in (eng.active + eng.completed + eng.waiting))
+    n_steps = 0
+    while eng.active or eng.waiting:
+        eng.step()
+        n_steps += 1
+        if n_steps > 200000:
+            break
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
+    dt_dec = time.perf_counter() - t1
     s = eng.stats()
+    dec_toks = s["output_tokens"] - tok0
     print(f"serving {tag}: {n_req} reqs x (p{prompt_len}+g{gen_len}) in {dt:.1f}s"
-          f"  decode {s['output_tokens'] / dt:.0f} tok/s"
+          f"  decode {s['output_tokens'] / dt:.0f} tok/s (incl prefill)"
+          f"  steady-decode {dec_toks / dt_dec:.0f} tok/s"
+          f" ({dt_dec / max(n_steps,1) * 1e3:.2f} ms/step)"
           f"  ttft {s['mean_ttft_s'] * 1e3:.0f} ms"
           f"  peak {torch.cuda.max_memory_allocated() / 2**30:.1f} GB")
 
