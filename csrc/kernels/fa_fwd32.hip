@@ -36,6 +36,15 @@ __device__ __forceinline__ unsigned lds_off32(unsigned row, unsigned col_bytes,
   return row * row_stride + (col_bytes ^ ((row & 7u) << 4));
 }
 
+// 16-granule variant for 256 B rows (D=128 tiles): fragment reads pull 32
+// DIFFERENT rows at one column, so an 8-way spread leaves 4-way bank
+// conflicts (measured ~1 conflict-cycle/busy-cycle, gpurun fa_pmc.csv);
+// 16 granules reach the b128 2-way floor.
+__device__ __forceinline__ unsigned lds_off32w(unsigned row, unsigned col_bytes,
+                                               unsigned row_stride, unsigned mask) {
+  return row * row_stride + (col_bytes ^ ((row & mask) << 4));
+}
+
 __device__ __forceinline__ int pack_bf2(float a, float b) {
   unsigned lo = (unsigned short)f2bf(a);
   unsigned hi = (unsigned short)f2bf(b);
@@ -166,7 +175,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
           val = *reinterpret_cast<const shortx8*>(kg + kbase + (long long)(kv0 + row) * k_ss + col);
         else
           for (int i = 0; i < 8; ++i) val[i] = 0;
-        *reinterpret_cast<shortx8*>(k_lds[buf] + lds_off32(row, col * 2, K_RS)) = val;
+        *reinterpret_cast<shortx8*>(k_lds[buf] + lds_off32w(row, col * 2, K_RS, KSWZ)) = val;
       }
     }
     {
@@ -202,7 +211,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
 #pragma unroll
       for (int ks = 0; ks < NKS; ++ks) {
         shortx8 kf = *reinterpret_cast<const shortx8*>(
-            k_lds[cur] + lds_off32(t * 32 + l32, (ks * 16 + hi * 8) * 2, K_RS));
+            k_lds[cur] + lds_off32w(t * 32 + l32, (ks * 16 + hi * 8) * 2, K_RS, KSWZ));
         st[t] = mfma32_bf16(kf, qf[ks], st[t]);
       }
     }
@@ -498,13 +507,14 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
     for (int r = 0; r < 16; ++r) dq_acc[dt][r] = 0.f;
 
   // DMA staging maps
+  constexpr unsigned KSWZ = (D >= 128) ? 15u : 7u;
   int n_row[KB * D / (NT * 8)], n_colp[KB * D / (NT * 8)];
 #pragma unroll
   for (int it = 0; it < KB * D / (NT * 8); ++it) {
     int flat = it * NT * 8 + tid * 8;
     int row = flat / D, col = flat % D;
     n_row[it] = row;
-    n_colp[it] = col ^ ((row & 7) << 3);
+    n_colp[it] = col ^ ((row & (int)KSWZ) << 3);
   }
 
   const int kv_end = CAUSAL ? min(Skv_e, q0 + QB + cdelta) : Skv_e;
@@ -544,8 +554,8 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
         } else {
           for (int i = 0; i < 8; ++i) { kv_[i] = 0; vv[i] = 0; }
         }
-        *reinterpret_cast<shortx8*>(k_lds + lds_off32(row, col * 2, NAT_RS)) = kv_;
-        *reinterpret_cast<shortx8*>(v_lds + lds_off32(row, col * 2, NAT_RS)) = vv;
+        *reinterpret_cast<shortx8*>(k_lds + lds_off32w(row, col * 2, NAT_RS, KSWZ)) = kv_;
+        *reinterpret_cast<shortx8*>(v_lds + lds_off32w(row, col * 2, NAT_RS, KSWZ)) = vv;
         const int rot = tid & 7;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -565,10 +575,10 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
 #pragma unroll
       for (int ks = 0; ks < NKS; ++ks) {
         shortx8 kf = *reinterpret_cast<const shortx8*>(
-            k_lds + lds_off32(t * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
+            k_lds + lds_off32w(t * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS, KSWZ));
         st[t] = mfma32_bf16(kf, qf[ks], st[t]);
         shortx8 vf = *reinterpret_cast<const shortx8*>(
-            v_lds + lds_off32(t * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS));
+            v_lds + lds_off32w(t * 32 + l32, (ks * 16 + hi * 8) * 2, NAT_RS, KSWZ));
         dp[t] = mfma32_bf16(vf, dof[ks], dp[t]);
       }
     }
