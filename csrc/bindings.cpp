@@ -661,6 +661,27 @@ std::vector<Tensor> moe_assign_slots(const Tensor& topi, int64_t num_experts,
   return {slot, counts};
 }
 
+Tensor gemm_fp8_nt_batched(const Tensor& a, const Tensor& bt, double scale_ab) {
+  TORCH_CHECK(a.is_cuda() && a.dim() == 3 && bt.dim() == 3);
+  TORCH_CHECK(a.is_contiguous() && bt.is_contiguous());
+  int64_t e = a.size(0), m = a.size(1), k = a.size(2), n = bt.size(1);
+  TORCH_CHECK(bt.size(0) == e && bt.size(2) == k && k >= 128);
+  auto c = torch::empty({e, m, n}, a.options().dtype(torch::kBFloat16));
+  pa::gemm_fp8_nt_batched(a.const_data_ptr(), bt.const_data_ptr(),
+                          c.mutable_data_ptr(), nullptr, (float)scale_ab, e,
+                          m, n, k, k, k, n, m * k, n * k, m * n, cur_stream());
+  return c;
+}
+
+Tensor quant_fp8(const Tensor& x, double scale) {
+  CHECK_IN(x);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.numel() % 8 == 0);
+  auto out = torch::empty(x.sizes(), x.options().dtype(torch::kFloat8_e4m3fn));
+  pa::quant_fp8(x.const_data_ptr(), out.mutable_data_ptr(), (float)scale,
+                x.numel(), cur_stream());
+  return out;
+}
+
 // ---- probe ----------------------------------------------------------------
 Tensor mfma_probe_fp8mx(const Tensor& a, const Tensor& bt, int64_t sa, int64_t sb) {
   CHECK_IN(a); CHECK_IN(bt);
@@ -721,6 +742,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_assign_slots", &moe_assign_slots);
   m.def("gemm_fp8_nt", &gemm_fp8_nt, py::arg("a"), py::arg("bt"),
         py::arg("scale_ab") = 1.0, py::arg("bias") = c10::nullopt);
+  m.def("gemm_fp8_nt_batched", &gemm_fp8_nt_batched);
+  m.def("quant_fp8", &quant_fp8);
   m.def("flash_attn_varlen_fwd", &flash_attn_varlen_fwd, py::arg("q"),
         py::arg("k"), py::arg("v"), py::arg("cu_q"), py::arg("cu_k"),
         py::arg("scale"), py::arg("causal"), py::arg("pdrop") = 0.0,

@@ -205,6 +205,15 @@ void fa_dropout_mask(void* out, int64_t total, float p, uint64_t seed,
 void gemm_fp8_nt(const void* a, const void* bt, void* c, const void* bias,
                  float scale_ab, int64_t m, int64_t n, int64_t k, int64_t lda,
                  int64_t ldb, int64_t ldc, hipStream_t s);
+// grouped/batched variant (one launch over all experts; *_bs = batch strides)
+void gemm_fp8_nt_batched(const void* a, const void* bt, void* c,
+                         const void* bias, float scale_ab, int64_t batch,
+                         int64_t m, int64_t n, int64_t k, int64_t lda,
+                         int64_t ldb, int64_t ldc, int64_t a_bs, int64_t b_bs,
+                         int64_t c_bs, hipStream_t s);
+// bf16 -> e4m3 cast with uniform scale (fused, no fp32 round-trip)
+void quant_fp8(const void* x, void* out, float scale, int64_t numel,
+               hipStream_t s);
 
 // skinny decode GEMM (M<=32): y = x @ W[K,N] + bias, split-K partials
 // in `workspace` ([ksplit, padded_m, N] fp32)

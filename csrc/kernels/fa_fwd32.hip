@@ -147,13 +147,14 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
   const int kv_end = CAUSAL ? min(Skv_e, q0 + QB + cdelta) : Skv_e;
 
   // K-staging source offsets (pre-swizzled so async DMA lands swizzled rows)
+  constexpr unsigned KSWZ = (D >= 128) ? 15u : 7u;
   int k_row[KB * D / (NT * 8)], k_colp[KB * D / (NT * 8)];
 #pragma unroll
   for (int it = 0; it < KB * D / (NT * 8); ++it) {
     int flat = it * NT * 8 + tid * 8;
     int row = flat / D, col = flat % D;
     k_row[it] = row;
-    k_colp[it] = col ^ ((row & 7) << 3);
+    k_colp[it] = col ^ ((row & (int)KSWZ) << 3);
   }
 
   auto stage = [&](int buf, int kv0) {

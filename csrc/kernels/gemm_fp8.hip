@@ -38,7 +38,13 @@ __global__ void gemm_fp8_kernel(const unsigned char* __restrict__ ag,
                                 const short* __restrict__ biasg,
                                 float sab, int M, int N, int K,
                                 long long lda, long long ldb, long long ldc,
-                                int skip_interior, int has_bias) {
+                                int skip_interior, int has_bias,
+                                long long a_bs = 0, long long b_bs = 0,
+                                long long c_bs = 0) {
+  // batched (grouped-expert) mode: blockIdx.z selects the problem
+  ag += (long long)blockIdx.z * a_bs;
+  bg += (long long)blockIdx.z * b_bs;
+  cg += (long long)blockIdx.z * c_bs;
   constexpr int BM = 256, BN = 256, BK = 128;
   __shared__ unsigned char a_lds[2][BM * BK];
   __shared__ unsigned char b_lds[2][BN * BK];
@@ -249,9 +255,11 @@ __global__ void gemm_fp8_kernel(const unsigned char* __restrict__ ag,
   }
 }
 
-void gemm_fp8_nt(const void* a, const void* bt, void* c, const void* bias,
-                 float scale_ab, int64_t m, int64_t n, int64_t k, int64_t lda,
-                 int64_t ldb, int64_t ldc, hipStream_t s) {
+void gemm_fp8_nt_batched(const void* a, const void* bt, void* c,
+                         const void* bias, float scale_ab, int64_t batch,
+                         int64_t m, int64_t n, int64_t k, int64_t lda,
+                         int64_t ldb, int64_t ldc, int64_t a_bs, int64_t b_bs,
+                         int64_t c_bs, hipStream_t s) {
   const int mi = (int)(m / 256), ni = (int)(n / 256);
   const int gm = (int)((m + 255) / 256), gn = (int)((n + 255) / 256);
   const bool k_ok = (k % 128 == 0);
@@ -259,15 +267,55 @@ void gemm_fp8_nt(const void* a, const void* bt, void* c, const void* bias,
   const bool has_edge = !has_fast || mi < gm || ni < gn;
   dim3 blk(512);
   if (has_fast)
-    hipLaunchKernelGGL((gemm_fp8_kernel<true>), dim3((unsigned)mi, (unsigned)ni),
+    hipLaunchKernelGGL((gemm_fp8_kernel<true>),
+                       dim3((unsigned)mi, (unsigned)ni, (unsigned)batch),
                        blk, 0, s, (const unsigned char*)a, (const unsigned char*)bt,
                        (short*)c, (const short*)bias, scale_ab, (int)m, (int)n,
-                       (int)k, lda, ldb, ldc, 0, bias != nullptr);
+                       (int)k, lda, ldb, ldc, 0, bias != nullptr,
+                       a_bs, b_bs, c_bs);
   if (has_edge)
-    hipLaunchKernelGGL((gemm_fp8_kernel<false>), dim3((unsigned)gm, (unsigned)gn),
+    hipLaunchKernelGGL((gemm_fp8_kernel<false>),
+                       dim3((unsigned)gm, (unsigned)gn, (unsigned)batch),
                        blk, 0, s, (const unsigned char*)a, (const unsigned char*)bt,
                        (short*)c, (const short*)bias, scale_ab, (int)m, (int)n,
-                       (int)k, lda, ldb, ldc, has_fast ? 1 : 0, bias != nullptr);
+                       (int)k, lda, ldb, ldc, has_fast ? 1 : 0, bias != nullptr,
+                       a_bs, b_bs, c_bs);
+}
+
+void gemm_fp8_nt(const void* a, const void* bt, void* c, const void* bias,
+                 float scale_ab, int64_t m, int64_t n, int64_t k, int64_t lda,
+                 int64_t ldb, int64_t ldc, hipStream_t s) {
+  gemm_fp8_nt_batched(a, bt, c, bias, scale_ab, 1, m, n, k, lda, ldb, ldc,
+                      0, 0, 0, s);
+}
+
+// bf16 -> e4m3 cast with a uniform scale, packed x8 (v_cvt_pk_fp8_f32);
+// one read + one half-size write, no fp32 materialization
+__global__ void quant_fp8_kernel(const short* __restrict__ x,
+                                 unsigned char* __restrict__ out,
+                                 float scale, long long n8) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < n8; i += stride) {
+    shortx8 v = *reinterpret_cast<const shortx8*>(x + i * 8);
+    int2 packed;
+    int p0 = 0, p1 = 0;
+    p0 = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f(v[0]) * scale, bf2f(v[1]) * scale, p0, false);
+    p0 = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f(v[2]) * scale, bf2f(v[3]) * scale, p0, true);
+    p1 = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f(v[4]) * scale, bf2f(v[5]) * scale, p1, false);
+    p1 = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f(v[6]) * scale, bf2f(v[7]) * scale, p1, true);
+    packed.x = p0;
+    packed.y = p1;
+    *reinterpret_cast<int2*>(out + i * 8) = packed;
+  }
+}
+
+void quant_fp8(const void* x, void* out, float scale, int64_t numel,
+               hipStream_t s) {
+  long long n8 = numel / 8;
+  dim3 g((unsigned)hmin<long long>((n8 + 255) / 256, 2048));
+  hipLaunchKernelGGL(quant_fp8_kernel, g, dim3(256), 0, s, (const short*)x,
+                     (unsigned char*)out, scale, n8);
 }
 
 }  // namespace pa

@@ -128,10 +128,44 @@ class GroupedExperts(nn.Layer):
                 w.normal_(0.0, (2.0 / (fan_in + w.shape[-1])) ** 0.5)
 
     def forward(self, x):
-        # x [E, N, h] -> [E, N, h]
-        z = torch.baddbmm(self.b1.unsqueeze(1), x, self.w1)
+        # x [E, N, h] -> [E, N, h].  Explicit backward: the autograd-
+        # generated baddbmm backward issues strided-batched bf16 GEMMs
+        # with a TRANSPOSED B operand, which memory-faults in this
+        # ROCm/hipBLASLt build at power-of-two token counts (bisect:
+        # tools/moe_prof.py); _GroupedFFN materializes the transposes.
+        return _GroupedFFN.apply(x, self.w1, self.b1, self.w2, self.b2)
+
+
+class _GroupedFFN(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w1, b1, w2, b2):
+        z = torch.baddbmm(b1.unsqueeze(1), x, w1)
         g = hot.bias_gelu(z, None)
-        return torch.baddbmm(self.b2.unsqueeze(1), g, self.w2)
+        out = torch.baddbmm(b2.unsqueeze(1), g, w2)
+        ctx.save_for_backward(x, z, g, w1, w2)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        from .. import _ext
+        x, z, g, w1, w2 = ctx.saved_tensors
+        dyc = dy.contiguous()
+        w2t = w2.transpose(1, 2).contiguous()   # substrate transB-bmm bug
+        w1t = w1.transpose(1, 2).contiguous()
+        dg = torch.bmm(dyc, w2t)
+        dw2 = torch.bmm(g.transpose(1, 2), dyc)
+        db2 = dyc.sum(1, dtype=torch.float32).to(w2.dtype)
+        if _ext.use_native(z):
+            dz = _ext.get_ext().bias_gelu_bwd(dg.contiguous(), z, None)
+        else:
+            zf = z.float()
+            cdf = 0.5 * (1 + torch.erf(zf * 0.7071067811865476))
+            pdf = 0.3989422804014327 * torch.exp(-0.5 * zf * zf)
+            dz = (dg.float() * (cdf + zf * pdf)).to(z.dtype)
+        dx = torch.bmm(dz, w1t)
+        dw1 = torch.bmm(x.transpose(1, 2), dz)
+        db1 = dz.sum(1, dtype=torch.float32).to(w1.dtype)
+        return dx, dw1, db1, dw2, db2
 
 
 class MoELayer(nn.Layer):
