@@ -83,8 +83,11 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
   __shared__ char k_lds[2][KB * D * 2];
   __shared__ char vt_lds[2][D * KB * 2];
 
-  const int qblk = blockIdx.x;
-  const int bh = blockIdx.y;
+  // block index on the SLOW grid dim (y) so causal longest-first ordering
+  // is global: the dispatcher walks x fastest, so a length-ordered y makes
+  // the drain tail all-short blocks instead of one of each length.
+  const int qblk = blockIdx.y;
+  const int bh = blockIdx.x;
   int b = bh / H, h = bh % H;
   int q0, Sq_e = Sq, Skv_e = Skv, cdelta = 0, qglob0 = 0;
   long long qbase, kbase, obase;
@@ -103,7 +106,7 @@ __global__ void fa_fwd32_kernel(const short* __restrict__ qg, const short* __res
     obase = (long long)qs0 * o_ss + (long long)h * o_sh;
   } else {
     const int hkv_ = h / (H / HKV);
-    q0 = qblk * QB;
+    q0 = (CAUSAL ? ((int)gridDim.y - 1 - qblk) : qblk) * QB;
     qbase = (long long)b * q_sb + (long long)h * q_sh;
     kbase = (long long)b * k_sb + (long long)hkv_ * k_sh;
     obase = (long long)b * o_sb + (long long)h * o_sh;
@@ -373,7 +376,7 @@ void flash_attn_fwd32(const void* q, const void* k, const void* v, void* o,
                       const int64_t* qs, const int64_t* ks, const int64_t* os,
                       const void* mask, const int64_t* ms, float pdrop,
                       uint64_t seed, uint64_t offset, hipStream_t s) {
-  dim3 grid((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
+  dim3 grid((unsigned)(b * h), (unsigned)cdiv((int)sq, 128));
   dim3 blk(256);
   const bool masked = mask != nullptr;
   const bool dropped = pdrop > 0.f;
@@ -443,8 +446,8 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
   __shared__ char v_lds[KB * D * 2];
   __shared__ char kt_lds[D * KB * 2];
 
-  const int qblk = blockIdx.x;
-  const int bh = blockIdx.y;
+  const int qblk = blockIdx.y;      // slow dim: global longest-first causal order
+  const int bh = blockIdx.x;
   int b = bh / H, h = bh % H;
   int q0, Sq_e = Sq, Skv_e = Skv, cdelta = 0, qglob0 = 0;
   long long qbase, dobase, dqbase, kvbase;
@@ -462,7 +465,7 @@ __global__ void fa_bwd_dq32_kernel(const short* __restrict__ dog, const short* _
     dqbase = (long long)qs0 * dq_ss + (long long)h * dq_sh;
     kvbase = (long long)ks0 * k_ss + (long long)h * k_sh;
   } else {
-    q0 = qblk * QB;
+    q0 = (CAUSAL ? ((int)gridDim.y - 1 - qblk) : qblk) * QB;   // longest-first
     qbase = (long long)b * q_sb + (long long)h * q_sh;
     dobase = (long long)b * do_sb + (long long)h * do_sh;
     dqbase = (long long)b * dq_sb + (long long)h * dq_sh;
@@ -687,7 +690,7 @@ void flash_attn_bwd_dq32(const void* dout, const void* q, const void* k,
                          const int64_t* dqs, const void* mask, const int64_t* ms,
                          float pdrop, uint64_t seed, uint64_t offset,
                          hipStream_t s) {
-  dim3 grid((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
+  dim3 grid((unsigned)(b * h), (unsigned)cdiv((int)sq, 128));
   dim3 blk(256);
   const bool masked = mask != nullptr;
   const bool dropped = pdrop > 0.f;
@@ -759,8 +762,8 @@ __global__ void fa_bwd_dkv32_kernel(const short* __restrict__ dog, const short* 
   __shared__ char t_lds[2][D * 64 * 2];
   __shared__ float stat_s[2][2 * QT];  // [lse | delta]
 
-  const int kvblk = blockIdx.x;
-  const int bh = blockIdx.y;
+  const int kvblk = blockIdx.y;     // slow dim; kv0=0 (longest) dispatches first
+  const int bh = blockIdx.x;
   int b = bh / H, h = bh % H;
   int kv0, Sq_e = Sq, Skv_e = Skv, cdelta = 0, qglob0 = 0;
   long long qbase, dobase, kvbase, outbase;
@@ -967,7 +970,7 @@ void flash_attn_bwd_dkv32(const void* dout, const void* q, const void* k,
                           const int64_t* dks, const void* mask, const int64_t* ms,
                           float pdrop, uint64_t seed, uint64_t offset,
                           hipStream_t s) {
-  dim3 grid((unsigned)cdiv((int)skv, 128), (unsigned)(b * h));
+  dim3 grid((unsigned)(b * h), (unsigned)cdiv((int)skv, 128));
   dim3 blk(256);
   const bool masked = mask != nullptr;
   const bool dropped = pdrop > 0.f;
@@ -1009,7 +1012,7 @@ void flash_attn_varlen_fwd32(const void* q, const void* k, const void* v,
                              const int* cu_q, const int* cu_k, const int* bmap,
                              float pdrop, uint64_t seed, uint64_t offset,
                              hipStream_t s) {
-  dim3 grid((unsigned)nblocks, (unsigned)h);
+  dim3 grid((unsigned)h, (unsigned)nblocks);
   dim3 blk(256);
   const bool dropped = pdrop > 0.f;
   const long long q_ss = h * dh, k_ss = hkv * dh;
@@ -1040,7 +1043,7 @@ void flash_attn_varlen_bwd32(const void* dout, const void* q, const void* k,
   const long long ss = h * dh;
 #define VDQ(D, C, P)                                                           \
   hipLaunchKernelGGL((fa_bwd_dq32_kernel<D, C, false, P, true>),               \
-                     dim3((unsigned)nqblocks, (unsigned)h), blk, 0, s,         \
+                     dim3((unsigned)h, (unsigned)nqblocks), blk, 0, s,         \
                      (const short*)dout, (const short*)q, (const short*)k,     \
                      (const short*)v, lse, delta, (short*)dq, 1, (int)h,       \
                      (int)total_q, (int)total_k, scale, 0, dh, ss, 0, dh, ss,  \
@@ -1048,7 +1051,7 @@ void flash_attn_varlen_bwd32(const void* dout, const void* q, const void* k,
                      offset, cu_q, cu_k, qbmap)
 #define VDKV(D, C, DK, P, OUT)                                                 \
   hipLaunchKernelGGL((fa_bwd_dkv32_kernel<D, C, DK, false, P, true>),          \
-                     dim3((unsigned)nkvblocks, (unsigned)h), blk, 0, s,        \
+                     dim3((unsigned)h, (unsigned)nkvblocks), blk, 0, s,        \
                      (const short*)dout, (const short*)q, (const short*)k,     \
                      (const short*)v, lse, delta, (short*)(OUT), 1, (int)h,    \
                      (int)total_q, (int)total_k, scale, 0, dh, ss, 0, dh, ss,  \

@@ -323,8 +323,8 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
   __shared__ char dot_lds[D * QB * 2];
   __shared__ char ps_lds[NW * 16 * QB * 2];
 
-  const int kvblk = blockIdx.x;
-  const int bh = blockIdx.y;
+  const int kvblk = blockIdx.y;     // slow dim; kv0=0 (longest) first
+  const int bh = blockIdx.x;
   const int kv0 = kvblk * KVEXT;
   const int b = bh / H, h = bh % H;
   const long long qbase = (long long)b * q_sb + (long long)h * q_sh;
@@ -764,7 +764,7 @@ void flash_attn_bwd(const void* dout, const void* q, const void* k,
                        (const short*)dout, (const short*)o, delta, (int)h,
                        (int)sq, rows, dos[0], dos[1], dos[2], os[0], os[1], os[2]);
 
-  dim3 gkv((unsigned)cdiv((int)skv, 128), (unsigned)(b * h));
+  dim3 gkv((unsigned)(b * h), (unsigned)cdiv((int)skv, 128));
   dim3 gq((unsigned)cdiv((int)sq, 128), (unsigned)(b * h));
   // The 16x16 dKV kernel is the default: both 32x32 rewrites measured
   // slower (combined/transposed-output: 146 TF bwd at occupancy 1;
