@@ -74,10 +74,12 @@ std::vector<Tensor> layer_norm_bwd(const Tensor& dy, const Tensor& x,
   pa::layer_norm_bwd_dx(dy.const_data_ptr(), x.const_data_ptr(), w.const_data_ptr(),
                         mean.const_data_ptr<float>(), rstd.const_data_ptr<float>(),
                         dx.mutable_data_ptr(), n, d, dt_of(x), cur_stream());
+  auto ws = torch::empty({(int64_t)2 * pa::ln_dwdb_chunks(n, d) * d},
+                         x.options().dtype(torch::kFloat));
   pa::layer_norm_bwd_dwdb(dy.const_data_ptr(), x.const_data_ptr(),
                           mean.const_data_ptr<float>(), rstd.const_data_ptr<float>(),
                           dw.mutable_data_ptr<float>(), db.mutable_data_ptr<float>(),
-                          n, d, dt_of(x), cur_stream());
+                          n, d, dt_of(x), cur_stream(), ws.mutable_data_ptr<float>());
   return {dx, dw.to(x.scalar_type()), db.to(x.scalar_type())};
 }
 
@@ -113,9 +115,11 @@ std::vector<Tensor> rms_norm_bwd(const Tensor& dy, const Tensor& x, const Tensor
   pa::rms_norm_bwd_dx(dy.const_data_ptr(), x.const_data_ptr(), w.const_data_ptr(),
                       rstd.const_data_ptr<float>(), dx.mutable_data_ptr(), n, d,
                       dt_of(x), cur_stream());
+  auto ws = torch::empty({(int64_t)2 * pa::ln_dwdb_chunks(n, d) * d},
+                         x.options().dtype(torch::kFloat));
   pa::rms_norm_bwd_dw(dy.const_data_ptr(), x.const_data_ptr(),
                       rstd.const_data_ptr<float>(), dw.mutable_data_ptr<float>(),
-                      n, d, dt_of(x), cur_stream());
+                      n, d, dt_of(x), cur_stream(), ws.mutable_data_ptr<float>());
   return {dx, dw.to(x.scalar_type())};
 }
 

@@ -17,9 +17,11 @@ void layer_norm_fwd(const void* x, const void* w, const void* b, void* y,
 void layer_norm_bwd_dx(const void* dy, const void* x, const void* w,
                        const float* mean, const float* rstd, void* dx,
                        int64_t n, int64_t d, int dtype, hipStream_t s);
+int ln_dwdb_chunks(int64_t n, int64_t d);
 void layer_norm_bwd_dwdb(const void* dy, const void* x, const float* mean,
                          const float* rstd, float* dw, float* db, int64_t n,
-                         int64_t d, int dtype, hipStream_t s);
+                         int64_t d, int dtype, hipStream_t s,
+                         float* ws = nullptr);
 
 void rms_norm_fwd(const void* x, const void* residual, const void* w, void* y,
                   void* res_out, float* rstd, int64_t n, int64_t d, float eps,
@@ -28,7 +30,8 @@ void rms_norm_bwd_dx(const void* dy, const void* x, const void* w,
                      const float* rstd, void* dx, int64_t n, int64_t d,
                      int dtype, hipStream_t s);
 void rms_norm_bwd_dw(const void* dy, const void* x, const float* rstd,
-                     float* dw, int64_t n, int64_t d, int dtype, hipStream_t s);
+                     float* dw, int64_t n, int64_t d, int dtype, hipStream_t s,
+                     float* ws = nullptr);
 
 // ---- fused softmax cross-entropy ------------------------------------------
 // logits [n, v]; labels int64 [n]; loss/lse fp32 [n]

@@ -149,7 +149,8 @@ template <int DT, bool RMS>
 __global__ void ln_bwd_dwdb_kernel(const void* __restrict__ dy, const void* __restrict__ x,
                                    const float* __restrict__ mean, const float* __restrict__ rstd,
                                    float* __restrict__ dw, float* __restrict__ db,
-                                   int64_t n, int64_t d) {
+                                   int64_t n, int64_t d,
+                                   float* __restrict__ ws = nullptr) {
   // 8-wide column strips (16 B loads) per thread; 2-D grid tiles rows;
   // fp32 atomics once per (strip, row-chunk)
   int64_t c0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
@@ -181,12 +182,39 @@ __global__ void ln_bwd_dwdb_kernel(const void* __restrict__ dy, const void* __re
       }
     }
   }
+  if (ws) {
+    // two-stage: plain partial stores, reduced by dwdb_reduce_kernel.
+    // The atomic path serializes ~chunks adds per column at one L2 bank
+    // (measured 3x over the HBM bound at n=24k, d=4k).
+    float* wrow = ws + (int64_t)(2 * blockIdx.y) * d;
+#pragma unroll
+    for (int k = 0; k < 8; ++k)
+      if (c0 + k < d) {
+        wrow[c0 + k] = aw[k];
+        wrow[d + c0 + k] = ab[k];
+      }
+    return;
+  }
 #pragma unroll
   for (int k = 0; k < 8; ++k)
     if (c0 + k < d) {
       atomicAdd(&dw[c0 + k], aw[k]);
       if (db) atomicAdd(&db[c0 + k], ab[k]);
     }
+}
+
+// reduce [2*chunks, d] partials -> dw[d], db[d]
+__global__ void dwdb_reduce_kernel(const float* __restrict__ ws, float* __restrict__ dw,
+                                   float* __restrict__ db, int chunks, int64_t d) {
+  int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= d) return;
+  float sw = 0.f, sb = 0.f;
+  for (int ch = 0; ch < chunks; ++ch) {
+    sw += ws[(int64_t)(2 * ch) * d + c];
+    sb += ws[(int64_t)(2 * ch) * d + d + c];
+  }
+  dw[c] = sw;
+  if (db) db[c] = sb;
 }
 
 // ---------------------------------------------------------------------------
@@ -295,15 +323,23 @@ void layer_norm_bwd_dx(const void* dy, const void* x, const void* w,
                                       dim3(256), 0, s, dy, x, w, mean, rstd, dx, n, d));
 }
 
+int ln_dwdb_chunks(int64_t n, int64_t d) {
+  int xblocks = cdiv((int)d, 256 * 8);
+  return (int)hmin<int64_t>(hmax<int64_t>(1, n / 32),
+                            hmax<int64_t>(1, 2048 / xblocks));
+}
+
 void layer_norm_bwd_dwdb(const void* dy, const void* x, const float* mean,
                          const float* rstd, float* dw, float* db, int64_t n,
-                         int64_t d, int dtype, hipStream_t s) {
+                         int64_t d, int dtype, hipStream_t s, float* ws) {
   int xblocks = cdiv((int)d, 256 * 8);
-  int chunks = (int)hmin<int64_t>(hmax<int64_t>(1, n / 32),
-                                  hmax<int64_t>(1, 2048 / xblocks));
+  int chunks = ln_dwdb_chunks(n, d);
   dim3 grid((unsigned)xblocks, chunks);
   DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_kernel<DT, false>), grid,
-                                      dim3(256), 0, s, dy, x, mean, rstd, dw, db, n, d));
+                                      dim3(256), 0, s, dy, x, mean, rstd, dw, db, n, d, ws));
+  if (ws)
+    hipLaunchKernelGGL(dwdb_reduce_kernel, dim3((unsigned)cdiv((int)d, 256)),
+                       dim3(256), 0, s, ws, dw, db, chunks, d);
 }
 
 void rms_norm_fwd(const void* x, const void* residual, const void* w, void* y,
@@ -326,13 +362,16 @@ void rms_norm_bwd_dx(const void* dy, const void* x, const void* w,
 }
 
 void rms_norm_bwd_dw(const void* dy, const void* x, const float* rstd,
-                     float* dw, int64_t n, int64_t d, int dtype, hipStream_t s) {
+                     float* dw, int64_t n, int64_t d, int dtype, hipStream_t s,
+                     float* ws) {
   int xblocks = cdiv((int)d, 256 * 8);
-  int chunks = (int)hmin<int64_t>(hmax<int64_t>(1, n / 32),
-                                  hmax<int64_t>(1, 2048 / xblocks));
+  int chunks = ln_dwdb_chunks(n, d);
   dim3 grid((unsigned)xblocks, chunks);
   DT_SWITCH(dtype, hipLaunchKernelGGL((ln_bwd_dwdb_kernel<DT, true>), grid,
-                                      dim3(256), 0, s, dy, x, nullptr, rstd, dw, nullptr, n, d));
+                                      dim3(256), 0, s, dy, x, nullptr, rstd, dw, nullptr, n, d, ws));
+  if (ws)
+    hipLaunchKernelGGL(dwdb_reduce_kernel, dim3((unsigned)cdiv((int)d, 256)),
+                       dim3(256), 0, s, ws, dw, nullptr, chunks, d);
 }
 
 }  // namespace pa
