@@ -677,6 +677,20 @@ Tensor gemm_fp8_nt_batched(const Tensor& a, const Tensor& bt, double scale_ab) {
   return c;
 }
 
+Tensor gemm_bf16_nt_batched_t(const Tensor& a, const Tensor& bt) {
+  TORCH_CHECK(a.is_cuda() && a.dim() == 3 && bt.dim() == 3);
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+              bt.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(a.is_contiguous() && bt.is_contiguous());
+  int64_t e = a.size(0), m = a.size(1), k = a.size(2), n = bt.size(1);
+  TORCH_CHECK(bt.size(0) == e && bt.size(2) == k && k % 8 == 0 && n % 8 == 0);
+  auto c = torch::empty({e, m, n}, a.options());
+  pa::gemm_bf16_nt_batched(a.const_data_ptr(), bt.const_data_ptr(),
+                           c.mutable_data_ptr(), e, m, n, k, k, k, n,
+                           m * k, n * k, m * n, cur_stream());
+  return c;
+}
+
 Tensor quant_fp8(const Tensor& x, double scale) {
   CHECK_IN(x);
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.numel() % 8 == 0);
@@ -764,6 +778,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);
   m.def("gemm_bf16", &gemm_bf16);
+  m.def("gemm_bf16_nt_batched", &gemm_bf16_nt_batched_t);
   m.def("gemm_bf16_ex", &gemm_bf16_ex, py::arg("a"), py::arg("b"),
         py::arg("layout"), py::arg("epilogue") = 0,
         py::arg("bias") = c10::nullopt, py::arg("aux") = c10::nullopt,

@@ -153,6 +153,11 @@ void decode_attention(const void* q, const void* kcache, const void* vcache,
 
 // ---- hand-written bf16 MFMA GEMM (C[m][n] = op(A) x op(B)) ----------------
 // b_is_nt: B passed as Bt[n][k] row-major (fast path); else B[k][n].
+void gemm_bf16_nt_batched(const void* a, const void* b, void* c,
+                          int64_t batch, int64_t m, int64_t n, int64_t k,
+                          int64_t lda, int64_t ldb, int64_t ldc,
+                          int64_t a_bs, int64_t b_bs, int64_t c_bs,
+                          hipStream_t s);
 void gemm_bf16(const void* a, const void* b, void* c, int64_t m, int64_t n,
                int64_t k, int64_t lda, int64_t ldb, int64_t ldc, bool b_is_nt,
                hipStream_t s);

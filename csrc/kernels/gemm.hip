@@ -88,8 +88,14 @@ __global__ void gemm_bf16_kernel(const short* __restrict__ ag, const short* __re
                                  short* __restrict__ cg, const short* __restrict__ biasg,
                                  short* __restrict__ auxg, int M, int N, int K,
                                  long long lda, long long ldb, long long ldc,
-                                 int skip_interior) {
+                                 int skip_interior,
+                                 long long a_bs = 0, long long b_bs = 0,
+                                 long long c_bs = 0) {
   constexpr int BM = 256, BN = 256, BK = 64;
+  // batched (grouped-expert) mode: grid.z indexes the batch
+  ag += (long long)blockIdx.z * a_bs;
+  bg += (long long)blockIdx.z * b_bs;
+  cg += (long long)blockIdx.z * c_bs;
   constexpr bool A_DMA = (LAYOUT != LAY_TN);   // A[M][K] row-major
   constexpr bool B_DMA = (LAYOUT == LAY_NT);   // Bt[N][K] row-major
   __shared__ short a_lds[2][BM * BK];
@@ -483,6 +489,36 @@ void gemm_bf16_ex(const void* a, const void* b, void* c, const void* bias,
   else EPI_SWITCH(LAY_TN);
 #undef EPI_SWITCH
 #undef LAUNCH
+}
+
+// Batched NT (grouped experts): C[z] = A[z] @ B[z]^T, z = 0..batch-1.
+// The MoE backward is the client: dg = dy @ w2^T and dx = dz @ w1^T hit
+// the stored [E, rows, K] weight layout directly, dodging both the
+// hipBLASLt strided-transB fault (see models/moe.py) and the contiguous
+// weight-transpose workaround.
+void gemm_bf16_nt_batched(const void* a, const void* b, void* c,
+                          int64_t batch, int64_t m, int64_t n, int64_t k,
+                          int64_t lda, int64_t ldb, int64_t ldc,
+                          int64_t a_bs, int64_t b_bs, int64_t c_bs,
+                          hipStream_t s) {
+  const int mi = (int)(m / 256), ni = (int)(n / 256);
+  const int gm = (int)((m + 255) / 256), gn = (int)((n + 255) / 256);
+  const bool k_ok = (k % 64 == 0);
+  const bool has_fast = k_ok && mi > 0 && ni > 0;
+  const bool has_edge = !has_fast || mi < gm || ni < gn;
+  dim3 blk(512);
+  if (has_fast)
+    hipLaunchKernelGGL((gemm_bf16_kernel<LAY_NT, EPI_NONE, false, true>),
+                       dim3((unsigned)mi, (unsigned)ni, (unsigned)batch), blk,
+                       0, s, (const short*)a, (const short*)b, (short*)c,
+                       nullptr, nullptr, (int)m, (int)n, (int)k, lda, ldb, ldc,
+                       0, a_bs, b_bs, c_bs);
+  if (has_edge)
+    hipLaunchKernelGGL((gemm_bf16_kernel<LAY_NT, EPI_NONE, false, false>),
+                       dim3((unsigned)gm, (unsigned)gn, (unsigned)batch), blk,
+                       0, s, (const short*)a, (const short*)b, (short*)c,
+                       nullptr, nullptr, (int)m, (int)n, (int)k, lda, ldb, ldc,
+                       has_fast ? 1 : 0, a_bs, b_bs, c_bs);
 }
 
 // legacy entry (round-1 API): b_is_nt picks NT vs NN, no epilogue

@@ -76,16 +76,19 @@ class _Fp8GroupedFFN(torch.autograd.Function):
         dt = x.dtype
         dyc = dy.to(dt)
         g = hot.bias_gelu(h, None)                       # recompute gelu(h)
+        from .. import _ext
+        native = _ext.use_native(h) and w1.dtype == torch.bfloat16
         # NOTE: strided-batched bf16 GEMM with a TRANSPOSED B operand
-        # memory-faults in this ROCm/hipBLASLt build (tools/moe_prof3.py
-        # bisect); materialize the transposed weights instead (one 512 MB
-        # copy per layer, ~0.2 ms)
-        w2t = w2.transpose(1, 2).contiguous().to(dt)
-        w1t = w1.transpose(1, 2).contiguous().to(dt)
-        dg = torch.bmm(dyc, w2t)
+        # memory-faults in this ROCm/hipBLASLt build (tools/moe_prof.py
+        # bisect); use the own batched NT kernel (stored weights ARE the
+        # NT B operand) or materialize transposes on the fallback path
+        if native:
+            C = _ext.get_ext()
+            dg = C.gemm_bf16_nt_batched(dyc, w2)
+        else:
+            dg = torch.bmm(dyc, w2.transpose(1, 2).contiguous().to(dt))
         dw2 = torch.bmm(g.transpose(1, 2), dyc)
         db2 = dyc.sum(1, dtype=torch.float32).to(b2.dtype)
-        from .. import _ext
         if _ext.use_native(h):
             dz = _ext.get_ext().bias_gelu_bwd(dg.contiguous(), h, None)
         else:
@@ -93,7 +96,10 @@ class _Fp8GroupedFFN(torch.autograd.Function):
             cdf = 0.5 * (1 + torch.erf(hf * 0.7071067811865476))
             pdf = 0.3989422804014327 * torch.exp(-0.5 * hf * hf)
             dz = (dg.float() * (cdf + hf * pdf)).to(dt)
-        dx = torch.bmm(dz, w1t)
+        if native:
+            dx = C.gemm_bf16_nt_batched(dz, w1)
+        else:
+            dx = torch.bmm(dz, w1.transpose(1, 2).contiguous().to(dt))
         dw1 = torch.bmm(x.transpose(1, 2), dz)
         db1 = dz.sum(1, dtype=torch.float32).to(b1.dtype)
         return dx, dw1.to(w1.dtype), db1, dw2.to(w2.dtype), db2

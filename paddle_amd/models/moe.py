@@ -150,9 +150,15 @@ class _GroupedFFN(torch.autograd.Function):
         from .. import _ext
         x, z, g, w1, w2 = ctx.saved_tensors
         dyc = dy.contiguous()
-        w2t = w2.transpose(1, 2).contiguous()   # substrate transB-bmm bug
-        w1t = w1.transpose(1, 2).contiguous()
-        dg = torch.bmm(dyc, w2t)
+        native = _ext.use_native(z) and w1.dtype == torch.bfloat16
+        if native:
+            # own batched NT MFMA GEMM takes the stored [E, rows, K] weights
+            # directly (dg = dy @ w2^T, dx = dz @ w1^T) -- no transB bmm
+            # (faults in this hipBLASLt build) and no transpose copies
+            C = _ext.get_ext()
+            dg = C.gemm_bf16_nt_batched(dyc, w2)
+        else:
+            dg = torch.bmm(dyc, w2.transpose(1, 2).contiguous())
         dw2 = torch.bmm(g.transpose(1, 2), dyc)
         db2 = dyc.sum(1, dtype=torch.float32).to(w2.dtype)
         if _ext.use_native(z):
@@ -162,7 +168,10 @@ class _GroupedFFN(torch.autograd.Function):
             cdf = 0.5 * (1 + torch.erf(zf * 0.7071067811865476))
             pdf = 0.3989422804014327 * torch.exp(-0.5 * zf * zf)
             dz = (dg.float() * (cdf + zf * pdf)).to(z.dtype)
-        dx = torch.bmm(dz, w1t)
+        if native:
+            dx = C.gemm_bf16_nt_batched(dz, w1)
+        else:
+            dx = torch.bmm(dz, w1.transpose(1, 2).contiguous())
         dw1 = torch.bmm(x.transpose(1, 2), dz)
         db1 = dz.sum(1, dtype=torch.float32).to(w1.dtype)
         return dx, dw1, db1, dw2, db2

@@ -433,6 +433,18 @@ def test_gemm_nt_numerics(m, n, k):
     _gemm_rel_ok(out, a.float() @ bt.float().t())
 
 
+@pytest.mark.parametrize("e,m,n,k", [(4, 512, 512, 512), (8, 513, 1024, 4096),
+                                     (64, 513, 4096, 1024)])
+def test_gemm_nt_batched(e, m, n, k):
+    torch.manual_seed(2)
+    C = _ext.get_ext()
+    a = _bf(torch.randn(e, m, k, device=DEV))
+    bt = _bf(torch.randn(e, n, k, device=DEV))
+    out = C.gemm_bf16_nt_batched(a, bt)
+    ref = torch.einsum("emk,enk->emn", a.float(), bt.float())
+    _gemm_rel_ok(out, ref)
+
+
 @pytest.mark.parametrize("m,n,k", [(512, 512, 512), (2048, 1024, 2048)])
 def test_gemm_nn_numerics(m, n, k):
     torch.manual_seed(2)
