@@ -636,6 +636,28 @@ Tensor decode_gemm(const Tensor& x, const Tensor& w,
   return y;
 }
 
+Tensor decode_gemm_mfma(const Tensor& x, const Tensor& w,
+                        const c10::optional<Tensor>& bias) {
+  CHECK_IN(x);
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && w.stride(1) == 1);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  int64_t m = x.size(0), k = x.size(1), n = w.size(1);
+  TORCH_CHECK(w.size(0) == k && m <= 32 && n % 256 == 0 && k % 64 == 0);
+  // target ~1024 workgroups, but keep kchunk >= 256 (4 pipelined LDS
+  // tiles per block; kchunk == 64 degenerates to an unpipelined tile)
+  int64_t nblk = n / 256;
+  int64_t ksplit = std::min<int64_t>(std::max<int64_t>(1, 1024 / nblk),
+                                     std::max<int64_t>(1, k / 256));
+  int64_t mt = m <= 16 ? 16 : 32;
+  auto ws = torch::empty({ksplit, mt, n}, x.options().dtype(torch::kFloat));
+  auto y = torch::empty({m, n}, x.options());
+  pa::decode_gemm_mfma(x.const_data_ptr(), w.const_data_ptr(),
+                       bias.has_value() ? bias->const_data_ptr() : nullptr,
+                       y.mutable_data_ptr(), ws.mutable_data_ptr<float>(), m,
+                       n, k, w.stride(0), ksplit, cur_stream());
+  return y;
+}
+
 // ---- MoE routing ----------------------------------------------------------
 std::vector<Tensor> moe_gate_topk(const Tensor& logits, int64_t k) {
   CHECK_IN(logits);
@@ -755,6 +777,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fa_dropout_mask", &fa_dropout_mask);
   m.def("mfma_probe_fp8mx", &mfma_probe_fp8mx);
   m.def("decode_gemm", &decode_gemm, py::arg("x"), py::arg("w"),
+        py::arg("bias") = c10::nullopt);
+  m.def("decode_gemm_mfma", &decode_gemm_mfma, py::arg("x"), py::arg("w"),
         py::arg("bias") = c10::nullopt);
   m.def("moe_gate_topk", &moe_gate_topk);
   m.def("moe_assign_slots", &moe_assign_slots);

@@ -128,3 +128,171 @@ void decode_gemm(const void* x, const void* w, const void* bias, void* y,
 }
 
 }  // namespace pa
+
+// ---------------------------------------------------------------------------
+// MFMA W-streaming decode GEMM (round-2 rework of the split-K streamer
+// above: the vector-ALU inner product was VALU-bound at M=32; MFMA makes
+// the arithmetic free so the kernel can run at the weight-read roofline).
+//
+//   y[M,N] = x[M,K] @ W[K,N],  M <= 32, N % 256 == 0, K % 64 == 0
+//
+// W rows are read 16 B/lane coalesced and TRANSPOSED through LDS
+// (wt[n][k] rows) so the mfma_f32_16x16x32_bf16 B-fragment (8 consecutive
+// k at fixed n -- layout verified by probe.hip) is one ds_read_b128.
+// x is tiny and L2-resident: A-fragments load straight from global.
+// Split-K across grid.y writes fp32 partials reduced by
+// decode_gemm_reduce_kernel.
+// ---------------------------------------------------------------------------
+namespace pa {
+
+// k-chunk swizzle: chunk kc (8 bf16) of row n lives at kc ^ ((n>>3)&7)
+__device__ __forceinline__ int dg_swz(int kc, int n) {
+  return (kc ^ ((n >> 3) & 7)) * 8;
+}
+
+template <int MTILES>           // 1: M<=16, 2: M<=32
+__launch_bounds__(256, 2)
+__global__ void decode_gemm_mfma_kernel(const short* __restrict__ xg,
+                                        const short* __restrict__ wg,
+                                        float* __restrict__ partial,
+                                        int M, int N, int K, long long ldw,
+                                        int kchunk) {
+  constexpr int NB = 256;       // n cols per block
+  constexpr int KT = 64;        // k rows per LDS tile
+  __shared__ short wt[2][NB * KT];
+
+  const int n0 = blockIdx.x * NB;
+  const int ks = blockIdx.y;
+  const int k0 = ks * kchunk;
+  const int k1 = min(K, k0 + kchunk);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;      // wave: owns n cols [wv*64, wv*64+64)
+  const int l16 = lane & 15;
+  const int lg = lane >> 4;
+
+  // staging map: 8 rounds x (256 thr x 8 elem); thread covers k row
+  // (r*8 + tid/32) at n segment (tid%32)*8
+  const int s_k = tid >> 5;           // + r*8
+  const int s_n = (tid & 31) * 8;
+
+  floatx4 acc[MTILES][4];
+#pragma unroll
+  for (int mt = 0; mt < MTILES; ++mt)
+#pragma unroll
+    for (int f = 0; f < 4; ++f)
+      acc[mt][f] = floatx4{0.f, 0.f, 0.f, 0.f};
+
+  // x A-fragment rows for this lane (row index == l16 within each m-tile)
+  bool xrow_ok[MTILES];
+  long long xbase[MTILES];
+#pragma unroll
+  for (int mt = 0; mt < MTILES; ++mt) {
+    int row = mt * 16 + l16;
+    xrow_ok[mt] = row < M;
+    xbase[mt] = (long long)row * K;
+  }
+
+  shortx8 stg[8];
+  auto load_tile = [&](int kb) {    // global -> regs (coalesced rows)
+#pragma unroll
+    for (int r = 0; r < 8; ++r) {
+      int k = kb + r * 8 + s_k;
+      stg[r] = (k < k1)
+          ? *reinterpret_cast<const shortx8*>(wg + (long long)k * ldw + n0 + s_n)
+          : shortx8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  };
+  auto write_tile = [&](int buf) {  // regs -> LDS transposed
+#pragma unroll
+    for (int r = 0; r < 8; ++r) {
+      int kl = r * 8 + s_k;         // k row within tile
+      int kc = kl >> 3, ko = kl & 7;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        wt[buf][(s_n + j) * KT + dg_swz(kc, s_n + j) + ko] = stg[r][j];
+    }
+  };
+
+  load_tile(k0);
+  write_tile(0);
+  __syncthreads();
+
+  int cur = 0;
+  for (int kb = k0; kb < k1; kb += KT) {
+    const bool more = kb + KT < k1;
+    if (more) load_tile(kb + KT);   // overlap global latency with compute
+
+    // A-frags for the two mfma k-steps of this tile (x is L2-hot)
+    shortx8 af[MTILES][2];
+#pragma unroll
+    for (int mt = 0; mt < MTILES; ++mt)
+#pragma unroll
+      for (int kt = 0; kt < 2; ++kt) {
+        int k = kb + kt * 32 + lg * 8;
+        af[mt][kt] = (xrow_ok[mt] && k + 7 < K)
+            ? *reinterpret_cast<const shortx8*>(xg + xbase[mt] + k)
+            : shortx8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt) {
+      int kc = kt * 4 + lg;
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        int n = wv * 64 + f * 16 + l16;
+        shortx8 bf = *reinterpret_cast<const shortx8*>(
+            &wt[cur][n * KT + dg_swz(kc, n)]);
+#pragma unroll
+        for (int mt = 0; mt < MTILES; ++mt)
+          acc[mt][f] = mfma_bf16(af[mt][kt], bf, acc[mt][f]);
+      }
+    }
+    __syncthreads();
+    if (more) {
+      write_tile(cur ^ 1);
+      cur ^= 1;
+      __syncthreads();
+    }
+  }
+
+  // partial[ks][mt_pad][N] fp32; C lane layout: row lg*4+r, col l16
+  const int mt_pad = MTILES * 16;
+  float* out = partial + (long long)ks * mt_pad * N;
+#pragma unroll
+  for (int mt = 0; mt < MTILES; ++mt)
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      int n = n0 + wv * 64 + f * 16 + l16;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = mt * 16 + lg * 4 + r;
+        out[(long long)row * N + n] = acc[mt][f][r];
+      }
+    }
+}
+
+void decode_gemm_mfma(const void* x, const void* w, const void* bias, void* y,
+                      float* workspace, int64_t m, int64_t n, int64_t k,
+                      int64_t ldw, int64_t ksplit, hipStream_t s) {
+  const int mt = m <= 16 ? 16 : 32;
+  // kchunk: multiple of 64 so every split starts tile-aligned
+  int kchunk = (int)(((k + ksplit - 1) / ksplit + 63) / 64) * 64;
+  const int ks = (int)((k + kchunk - 1) / kchunk);
+  dim3 grid((unsigned)(n / 256), (unsigned)ks);
+  if (mt == 16)
+    hipLaunchKernelGGL((decode_gemm_mfma_kernel<1>), grid, dim3(256), 0, s,
+                       (const short*)x, (const short*)w, workspace, (int)m,
+                       (int)n, (int)k, ldw, kchunk);
+  else
+    hipLaunchKernelGGL((decode_gemm_mfma_kernel<2>), grid, dim3(256), 0, s,
+                       (const short*)x, (const short*)w, workspace, (int)m,
+                       (int)n, (int)k, ldw, kchunk);
+  long long total = m * n;
+  dim3 rg((unsigned)((total + 255) / 256));
+  hipLaunchKernelGGL((decode_gemm_reduce_kernel<0>), rg, dim3(256), 0, s,
+                     workspace, (const short*)bias, (short*)y, (int)m, (int)n,
+                     mt, ks);
+}
+
+}  // namespace pa

@@ -34,6 +34,19 @@ def linear(x, weight, bias=None, name=None):
     # NOTE: an M<=32 split-K weight-streaming kernel (decode_gemm.hip) was
     # measured SLOWER in-situ than hipBLASLt's skinny tiles (VALU-bound at
     # M=32); it stays available via _C.decode_gemm but is not dispatched.
+    # Its MFMA rework (decode_gemm_mfma) IS dispatched, but only on the
+    # tall-K decode shape where it beats hipBLASLt cold-HBM (fc2:
+    # 51.3 vs 71.8 us at M32 K16384 N4096 -- tools/bench_decode_gemm.py);
+    # lt already streams the wide-N shapes at 3.4-5.9 TB/s.
+    if (not torch.is_grad_enabled() and x.dim() >= 2 and x.is_cuda
+            and x.dtype == torch.bfloat16):
+        mrows = x.numel() // x.shape[-1]
+        k, n = weight.shape[-2], weight.shape[-1]
+        if (mrows <= 32 and k >= 2 * n and k % 64 == 0 and n % 256 == 0
+                and weight.stride(-1) == 1 and hot._ext.use_native(x)):
+            C = hot._ext.get_ext()
+            out = C.decode_gemm_mfma(x.reshape(-1, k), weight, bias)
+            return out.reshape(*x.shape[:-1], n)
     if bias is not None and x.dim() >= 2:
         x2 = x.reshape(-1, x.shape[-1])
         out = torch.addmm(bias, x2, weight)

@@ -831,3 +831,18 @@ def test_decode_gemm_numerics(m, k, n):
     _gemm_rel_ok(y, ref, tol=2e-2)
     y2 = C.decode_gemm(x, w, None)
     _gemm_rel_ok(y2, x.float() @ w.float(), tol=2e-2)
+
+
+@pytest.mark.parametrize("m,k,n", [(32, 4096, 4096), (16, 512, 1024),
+                                   (5, 256, 512), (32, 4160, 12288)])
+def test_decode_gemm_mfma_numerics(m, k, n):
+    torch.manual_seed(4)
+    C = _ext.get_ext()
+    x = _bf(torch.randn(m, k, device=DEV))
+    w = _bf(torch.randn(k, n, device=DEV) * 0.02)
+    bias = _bf(torch.randn(n, device=DEV))
+    y = C.decode_gemm_mfma(x, w, bias)
+    ref = x.float() @ w.float() + bias.float()
+    _gemm_rel_ok(y, ref, tol=2e-2)
+    y2 = C.decode_gemm_mfma(x, w, None)
+    _gemm_rel_ok(y2, x.float() @ w.float(), tol=2e-2)
