@@ -646,8 +646,10 @@ Tensor decode_gemm_mfma(const Tensor& x, const Tensor& w,
   // target ~1024 workgroups, but keep kchunk >= 256 (4 pipelined LDS
   // tiles per block; kchunk == 64 degenerates to an unpipelined tile)
   int64_t nblk = n / 256;
+  // small-N shapes need a denser grid: allow kchunk 128 to reach 2 wg/CU
+  int64_t kmin = nblk <= 24 ? 128 : 256;
   int64_t ksplit = std::min<int64_t>(std::max<int64_t>(1, 1024 / nblk),
-                                     std::max<int64_t>(1, k / 256));
+                                     std::max<int64_t>(1, k / kmin));
   int64_t mt = m <= 16 ? 16 : 32;
   auto ws = torch::empty({ksplit, mt, n}, x.options().dtype(torch::kFloat));
   auto y = torch::empty({m, n}, x.options());

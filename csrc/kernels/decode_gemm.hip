@@ -215,17 +215,7 @@ __global__ void decode_gemm_mfma_kernel(const short* __restrict__ xg,
     }
   };
 
-  load_tile(k0);
-  write_tile(0);
-  __syncthreads();
-
-  int cur = 0;
-  for (int kb = k0; kb < k1; kb += KT) {
-    const bool more = kb + KT < k1;
-    if (more) load_tile(kb + KT);   // overlap global latency with compute
-
-    // A-frags for the two mfma k-steps of this tile (x is L2-hot)
-    shortx8 af[MTILES][2];
+  auto load_a = [&](shortx8 af[][2], int kb) {
 #pragma unroll
     for (int mt = 0; mt < MTILES; ++mt)
 #pragma unroll
@@ -235,6 +225,25 @@ __global__ void decode_gemm_mfma_kernel(const short* __restrict__ xg,
             ? *reinterpret_cast<const shortx8*>(xg + xbase[mt] + k)
             : shortx8{0, 0, 0, 0, 0, 0, 0, 0};
       }
+  };
+
+  // One barrier per tile: at iter t the regs hold tile t+1 (loaded at
+  // t-1), which is written into buf (t+1)%2 BEFORE computing tile t from
+  // buf t%2.  WAR on buf t%2 (written next iter) and RAW on buf (t+1)%2
+  // (read next iter) are both cut by the single end-of-iter barrier.
+  shortx8 af[MTILES][2], afn[MTILES][2];
+  load_tile(k0);
+  write_tile(0);
+  load_tile(k0 + KT);              // regs = tile 1
+  load_a(af, k0);
+  __syncthreads();
+
+  int cur = 0;
+  for (int kb = k0; kb < k1; kb += KT) {
+    const bool more = kb + KT < k1;
+    if (more) write_tile(cur ^ 1); // regs from two iters back -> next buf
+    if (kb + 2 * KT < k1) load_tile(kb + 2 * KT);
+    if (more) load_a(afn, kb + KT);
 #pragma unroll
     for (int kt = 0; kt < 2; ++kt) {
       int kc = kt * 4 + lg;
@@ -248,12 +257,12 @@ __global__ void decode_gemm_mfma_kernel(const short* __restrict__ xg,
           acc[mt][f] = mfma_bf16(af[mt][kt], bf, acc[mt][f]);
       }
     }
+#pragma unroll
+    for (int mt = 0; mt < MTILES; ++mt)
+#pragma unroll
+      for (int kt = 0; kt < 2; ++kt) af[mt][kt] = afn[mt][kt];
+    cur ^= 1;
     __syncthreads();
-    if (more) {
-      write_tile(cur ^ 1);
-      cur ^= 1;
-      __syncthreads();
-    }
   }
 
   // partial[ks][mt_pad][N] fp32; C lane layout: row lg*4+r, col l16
