@@ -390,12 +390,16 @@ __global__ void fa_bwd_dkv_kernel(const short* __restrict__ dog, const short* __
             (__attribute__((address_space(3))) unsigned int*)(do_lds + it * NT * 16 + (tid >> 6) * 64 * 16),
             16, 0, 0);
       }
+      // transposed copies from the DMA-landed natural tiles (LDS round
+      // trip) -- re-reading Q/dO from global doubled this kernel's HBM
+      // traffic, and the kernel is Q/dO-stream-bound
+      __syncthreads();
       constexpr int elems = QB * D;
       const int rot = threadIdx.x & 7;
       for (int flat = tid * 8; flat < elems; flat += NT * 8) {
         int row = flat / D, col = flat % D;
-        shortx8 qv = *reinterpret_cast<const shortx8*>(qg + qbase + (long long)(q0 + row) * q_ss + col);
-        shortx8 dv = *reinterpret_cast<const shortx8*>(dog + dobase + (long long)(q0 + row) * do_ss + col);
+        shortx8 qv = lds_read8(q_lds, row, col, NAT_RS);
+        shortx8 dv = lds_read8(do_lds, row, col, NAT_RS);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int i = (j + rot) & 7;
