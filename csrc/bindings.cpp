@@ -689,16 +689,34 @@ std::vector<Tensor> moe_assign_slots(const Tensor& topi, int64_t num_experts,
   return {slot, counts};
 }
 
-Tensor gemm_fp8_nt_batched(const Tensor& a, const Tensor& bt, double scale_ab) {
+Tensor gemm_fp8_nt_batched(const Tensor& a, const Tensor& bt, double scale_ab,
+                           const c10::optional<Tensor>& bias) {
   TORCH_CHECK(a.is_cuda() && a.dim() == 3 && bt.dim() == 3);
   TORCH_CHECK(a.is_contiguous() && bt.is_contiguous());
   int64_t e = a.size(0), m = a.size(1), k = a.size(2), n = bt.size(1);
   TORCH_CHECK(bt.size(0) == e && bt.size(2) == k && k >= 128);
+  const void* bp = nullptr;
+  int64_t bias_bs = 0;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->is_contiguous() && bias->scalar_type() == torch::kBFloat16);
+    TORCH_CHECK(bias->dim() == 2 && bias->size(0) == e && bias->size(1) == n);
+    bp = bias->const_data_ptr();
+    bias_bs = n;
+  }
   auto c = torch::empty({e, m, n}, a.options().dtype(torch::kBFloat16));
   pa::gemm_fp8_nt_batched(a.const_data_ptr(), bt.const_data_ptr(),
-                          c.mutable_data_ptr(), nullptr, (float)scale_ab, e,
-                          m, n, k, k, k, n, m * k, n * k, m * n, cur_stream());
+                          c.mutable_data_ptr(), bp, (float)scale_ab, e,
+                          m, n, k, k, k, n, m * k, n * k, m * n, cur_stream(),
+                          bias_bs);
   return c;
+}
+
+Tensor amax_abs(const Tensor& x) {
+  CHECK_IN(x);
+  auto out = torch::zeros({}, x.options().dtype(torch::kFloat));
+  pa::amax_abs(x.const_data_ptr(), out.mutable_data_ptr<float>(), x.numel(),
+               dt_of(x), cur_stream());
+  return out;
 }
 
 Tensor gemm_bf16_nt_batched_t(const Tensor& a, const Tensor& bt) {
@@ -786,7 +804,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_assign_slots", &moe_assign_slots);
   m.def("gemm_fp8_nt", &gemm_fp8_nt, py::arg("a"), py::arg("bt"),
         py::arg("scale_ab") = 1.0, py::arg("bias") = c10::nullopt);
-  m.def("gemm_fp8_nt_batched", &gemm_fp8_nt_batched);
+  m.def("gemm_fp8_nt_batched", &gemm_fp8_nt_batched, py::arg("a"),
+        py::arg("bt"), py::arg("scale_ab"), py::arg("bias") = c10::nullopt);
+  m.def("amax_abs", &amax_abs);
   m.def("quant_fp8", &quant_fp8);
   m.def("flash_attn_varlen_fwd", &flash_attn_varlen_fwd, py::arg("q"),
         py::arg("k"), py::arg("v"), py::arg("cu_q"), py::arg("cu_k"),

@@ -65,6 +65,7 @@ void rope_fwd(const void* x, const float* cos_t, const float* sin_t, void* y,
               bool conj, int dtype, hipStream_t s);
 
 // column-sum of [n, d] into fp32 [d] (bias grads)
+void amax_abs(const void* x, float* out, int64_t n, int dtype, hipStream_t s);
 void colsum(const void* x, float* out, int64_t n, int64_t d, int dtype,
             hipStream_t s);
 
@@ -218,7 +219,8 @@ void gemm_fp8_nt_batched(const void* a, const void* bt, void* c,
                          const void* bias, float scale_ab, int64_t batch,
                          int64_t m, int64_t n, int64_t k, int64_t lda,
                          int64_t ldb, int64_t ldc, int64_t a_bs, int64_t b_bs,
-                         int64_t c_bs, hipStream_t s);
+                         int64_t c_bs, hipStream_t s,
+                         int64_t bias_bs = 0);
 // bf16 -> e4m3 cast with uniform scale (fused, no fp32 round-trip)
 void quant_fp8(const void* x, void* out, float scale, int64_t numel,
                hipStream_t s);

@@ -358,3 +358,42 @@ void dropout_add_bwd(const void* dy, const uint8_t* mask, void* dx,
 }
 
 }  // namespace pa
+
+namespace pa {
+// ---------------------------------------------------------------------------
+// fused |x| max: single read pass (x.abs().amax() materializes |x| and
+// re-reads it -- 3x the traffic; this is the fp8 quantize-scale producer)
+// ---------------------------------------------------------------------------
+template <int DT>
+__global__ void amax_abs_kernel(const void* __restrict__ x, float* __restrict__ out,
+                                int64_t n) {
+  float m = 0.f;
+  int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  for (int64_t i = i0; i + 8 <= n; i += stride) {
+    float v[8];
+    LS8<DT>::load8(x, i, v);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) m = fmaxf(m, fabsf(v[k]));
+  }
+  // scalar tail (last partial group)
+  int64_t tail = n & ~7ll;
+  if (i0 == 0)
+    for (int64_t i = tail; i < n; ++i)
+      m = fmaxf(m, fabsf(DT == kBF16
+          ? bf2f(reinterpret_cast<const short*>(x)[i])
+          : reinterpret_cast<const float*>(x)[i]));
+#pragma unroll
+  for (int off = 32; off; off >>= 1) m = fmaxf(m, __shfl_xor(m, off, 64));
+  if ((threadIdx.x & 63) == 0)
+    atomicMax(reinterpret_cast<int*>(out), __float_as_int(m));  // m >= 0
+}
+
+void amax_abs(const void* x, float* out, int64_t n, int dtype, hipStream_t s) {
+  int64_t waves = (n + 8 * 256 - 1) / (8 * 256);
+  unsigned g = (unsigned)hmin<int64_t>(hmax<int64_t>(waves, 1), 4096);
+  EDT(dtype,
+      hipLaunchKernelGGL((amax_abs_kernel<kBF16>), dim3(g), dim3(256), 0, s, x, out, n),
+      hipLaunchKernelGGL((amax_abs_kernel<kF32>), dim3(g), dim3(256), 0, s, x, out, n));
+}
+}  // namespace pa

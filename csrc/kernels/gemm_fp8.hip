@@ -40,11 +40,12 @@ __global__ void gemm_fp8_kernel(const unsigned char* __restrict__ ag,
                                 long long lda, long long ldb, long long ldc,
                                 int skip_interior, int has_bias,
                                 long long a_bs = 0, long long b_bs = 0,
-                                long long c_bs = 0) {
+                                long long c_bs = 0, long long bias_bs = 0) {
   // batched (grouped-expert) mode: blockIdx.z selects the problem
   ag += (long long)blockIdx.z * a_bs;
   bg += (long long)blockIdx.z * b_bs;
   cg += (long long)blockIdx.z * c_bs;
+  if (has_bias) biasg += (long long)blockIdx.z * bias_bs;
   constexpr int BM = 256, BN = 256, BK = 128;
   __shared__ unsigned char a_lds[2][BM * BK];
   __shared__ unsigned char b_lds[2][BN * BK];
@@ -259,7 +260,7 @@ void gemm_fp8_nt_batched(const void* a, const void* bt, void* c,
                          const void* bias, float scale_ab, int64_t batch,
                          int64_t m, int64_t n, int64_t k, int64_t lda,
                          int64_t ldb, int64_t ldc, int64_t a_bs, int64_t b_bs,
-                         int64_t c_bs, hipStream_t s) {
+                         int64_t c_bs, hipStream_t s, int64_t bias_bs) {
   const int mi = (int)(m / 256), ni = (int)(n / 256);
   const int gm = (int)((m + 255) / 256), gn = (int)((n + 255) / 256);
   const bool k_ok = (k % 128 == 0);
@@ -272,21 +273,21 @@ void gemm_fp8_nt_batched(const void* a, const void* bt, void* c,
                        blk, 0, s, (const unsigned char*)a, (const unsigned char*)bt,
                        (short*)c, (const short*)bias, scale_ab, (int)m, (int)n,
                        (int)k, lda, ldb, ldc, 0, bias != nullptr,
-                       a_bs, b_bs, c_bs);
+                       a_bs, b_bs, c_bs, bias_bs);
   if (has_edge)
     hipLaunchKernelGGL((gemm_fp8_kernel<false>),
                        dim3((unsigned)gm, (unsigned)gn, (unsigned)batch),
                        blk, 0, s, (const unsigned char*)a, (const unsigned char*)bt,
                        (short*)c, (const short*)bias, scale_ab, (int)m, (int)n,
                        (int)k, lda, ldb, ldc, has_fast ? 1 : 0, bias != nullptr,
-                       a_bs, b_bs, c_bs);
+                       a_bs, b_bs, c_bs, bias_bs);
 }
 
 void gemm_fp8_nt(const void* a, const void* bt, void* c, const void* bias,
                  float scale_ab, int64_t m, int64_t n, int64_t k, int64_t lda,
                  int64_t ldb, int64_t ldc, hipStream_t s) {
   gemm_fp8_nt_batched(a, bt, c, bias, scale_ab, 1, m, n, k, lda, ldb, ldc,
-                      0, 0, 0, s);
+                      0, 0, 0, s, 0);
 }
 
 // bf16 -> e4m3 cast with a uniform scale, packed x8 (v_cvt_pk_fp8_f32);

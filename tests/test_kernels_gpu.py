@@ -433,6 +433,31 @@ def test_gemm_nt_numerics(m, n, k):
     _gemm_rel_ok(out, a.float() @ bt.float().t())
 
 
+def test_amax_abs():
+    C = _ext.get_ext()
+    for n in (999, 4096, 1 << 20):
+        x = _bf(torch.randn(n, device=DEV) * 3)
+        ref = x.float().abs().amax()
+        out = C.amax_abs(x)
+        assert torch.allclose(out, ref), (float(out), float(ref))
+
+
+def test_gemm_fp8_batched_bias():
+    torch.manual_seed(7)
+    C = _ext.get_ext()
+    e, m, n, k = 8, 256, 512, 256
+    a = torch.randn(e, m, k, device=DEV).clamp(-2, 2)
+    bt = torch.randn(e, n, k, device=DEV).clamp(-2, 2)
+    bias = _bf(torch.randn(e, n, device=DEV))
+    qa = (a * 16).to(torch.float8_e4m3fn)
+    qb = (bt * 16).to(torch.float8_e4m3fn)
+    s = 1.0 / 256
+    out = C.gemm_fp8_nt_batched(qa, qb, s, bias)
+    ref = torch.einsum("emk,enk->emn", qa.float(), qb.float()) * s \
+        + bias.float().unsqueeze(1)
+    _gemm_rel_ok(out, ref, tol=5e-2)
+
+
 @pytest.mark.parametrize("e,m,n,k", [(4, 512, 512, 512), (8, 513, 1024, 4096),
                                      (64, 513, 4096, 1024)])
 def test_gemm_nt_batched(e, m, n, k):
