@@ -178,8 +178,40 @@ def is_float16_supported(device=None):
     return True
 
 
+class DebugMode:
+    """paddle.amp.debugging.DebugMode (reference debugging.py:56)."""
+    CHECK_NAN_INF_AND_ABORT = 0
+    CHECK_NAN_INF = 1
+    CHECK_ALL_FOR_OVERFLOW = 2
+    CHECK_ALL = 3
+
+
+class TensorCheckerConfig:
+    """Reference python/paddle/amp/debugging.py:173 -- config for the
+    module-output nan/inf checker installed by enable_tensor_checker."""
+
+    def __init__(self, enable=True, debug_mode=DebugMode.CHECK_NAN_INF_AND_ABORT,
+                 output_dir=None, checked_op_list=None, skipped_op_list=None,
+                 debug_step=None, stack_height_limit=1):
+        self.enable = enable
+        self.debug_mode = debug_mode
+        self.output_dir = output_dir
+        self.checked_op_list = checked_op_list
+        self.skipped_op_list = set(skipped_op_list or [])
+        self.debug_step = debug_step
+        self.stack_height_limit = stack_height_limit
+        self._step = 0
+
+
+_tensor_checker = {"config": None, "hooks": []}
+
+
 class debugging:
-    """paddle.amp.debugging stub surface (check_numerics etc.)."""
+    """paddle.amp.debugging (check_numerics, tensor checker, accuracy
+    compare -- reference python/paddle/amp/debugging.py)."""
+
+    DebugMode = DebugMode
+    TensorCheckerConfig = TensorCheckerConfig
 
     @staticmethod
     def enable_operator_stats_collection():
@@ -193,5 +225,101 @@ class debugging:
     def check_numerics(tensor, op_type="", var_name="", debug_mode=None):
         import torch as _t
         if not _t.isfinite(tensor.float()).all():
-            raise FloatingPointError(f"nan/inf detected in {op_type}:{var_name}")
+            bad = (~_t.isfinite(tensor.float())).sum().item()
+            msg = f"nan/inf detected in {op_type}:{var_name} ({bad} bad elements)"
+            if debug_mode in (None, DebugMode.CHECK_NAN_INF_AND_ABORT):
+                raise FloatingPointError(msg)
+            import warnings
+            warnings.warn(msg)
         return tensor
+
+    @staticmethod
+    def check_layer_numerics(layer):
+        """Wrap a layer so every forward asserts finite outputs
+        (reference debugging.py:78)."""
+        import torch as _t
+
+        def _hook(mod, inp, out):
+            for i, t in enumerate(_flat_tensors(out)):
+                debugging.check_numerics(t, type(mod).__name__, f"out{i}")
+        layer.register_forward_post_hook(_hook) if hasattr(layer, "register_forward_post_hook") \
+            else layer.register_forward_hook(_hook)
+        return layer
+
+    @staticmethod
+    def enable_tensor_checker(config):
+        """Install forward hooks on every later-constructed check via
+        torch module hooks; checks all module outputs for nan/inf."""
+        import torch as _t
+        debugging.disable_tensor_checker()
+        _tensor_checker["config"] = config
+        if not config.enable:
+            return
+
+        def _hook(mod, inp, out):
+            cfg = _tensor_checker["config"]
+            if cfg is None:
+                return
+            name = type(mod).__name__
+            if cfg.checked_op_list and name not in cfg.checked_op_list:
+                return
+            if name in cfg.skipped_op_list:
+                return
+            for i, t in enumerate(_flat_tensors(out)):
+                debugging.check_numerics(t, name, f"out{i}", cfg.debug_mode)
+        h = _t.nn.modules.module.register_module_forward_hook(_hook)
+        _tensor_checker["hooks"].append(h)
+
+    @staticmethod
+    def disable_tensor_checker():
+        for h in _tensor_checker["hooks"]:
+            h.remove()
+        _tensor_checker["hooks"].clear()
+        _tensor_checker["config"] = None
+
+    @staticmethod
+    def compare_accuracy(dump_path, another_dump_path, output_filename,
+                         loss_scale=1, dump_all_tensors=False):
+        """Compare two tensor-dump directories (torch.save'd dicts of
+        name->tensor, one file per step) and write a CSV of max abs/rel
+        divergence per tensor (reference amp/accuracy_compare.py, minus
+        the xlsx dependency)."""
+        import csv as _csv
+        import os as _os
+        import torch as _t
+        rows = []
+        for fn in sorted(_os.listdir(dump_path)):
+            p1 = _os.path.join(dump_path, fn)
+            p2 = _os.path.join(another_dump_path, fn)
+            if not _os.path.exists(p2):
+                continue
+            d1, d2 = _t.load(p1, weights_only=False), _t.load(p2, weights_only=False)
+            for k in d1:
+                if k not in d2:
+                    continue
+                a = d1[k].float() * loss_scale
+                b = d2[k].float()
+                if a.shape != b.shape:
+                    rows.append([fn, k, "shape-mismatch", "", ""])
+                    continue
+                diff = (a - b).abs()
+                rel = diff.max() / b.abs().max().clamp(min=1e-12)
+                rows.append([fn, k, float(diff.max()), float(rel),
+                             "DIVERGED" if float(rel) > 1e-2 else ""])
+        with open(output_filename, "w", newline="") as f:
+            w = _csv.writer(f)
+            w.writerow(["file", "tensor", "max_abs_diff", "max_rel_diff", "flag"])
+            w.writerows(rows)
+        return rows
+
+
+def _flat_tensors(out):
+    import torch as _t
+    if isinstance(out, _t.Tensor):
+        return [out] if out.is_floating_point() else []
+    if isinstance(out, (list, tuple)):
+        r = []
+        for o in out:
+            r.extend(_flat_tensors(o))
+        return r
+    return []

@@ -37,6 +37,9 @@ def parse_args(argv=None):
     ap.add_argument("--master_port", type=int, default=29765)
     ap.add_argument("--rank", type=int, default=0, help="node rank")
     ap.add_argument("--log_dir", type=str, default="log")
+    ap.add_argument("--gpu_watch_interval", type=float, default=0.0,
+                    help="seconds between GPU-utilization log lines "
+                         "(reference launch/controllers/watcher.py; 0 = off)")
     ap.add_argument("--job_id", type=str, default="default")
     ap.add_argument("training_script", type=str)
     ap.add_argument("training_script_args", nargs=argparse.REMAINDER)
@@ -58,6 +61,8 @@ def launch(argv=None):
     world = nproc * nnodes
 
     os.makedirs(args.log_dir, exist_ok=True)
+    watcher = _GPUWatcher(args.log_dir, args.gpu_watch_interval)
+    watcher.start()
     procs = []
     logs = []
     for local_rank in range(nproc):
@@ -115,9 +120,62 @@ def launch(argv=None):
                 p.terminate()
         code = 130
     finally:
+        watcher.stop()
         for f in logs:
             f.close()
     return code
+
+
+class _GPUWatcher:
+    """Periodic GPU utilization/VRAM logger (reference
+    launch/controllers/watcher.py) -- rocm-smi when present, else
+    torch.cuda memory counters; writes log_dir/gpulog."""
+
+    def __init__(self, log_dir, interval):
+        self.interval = interval
+        self.path = os.path.join(log_dir, "gpulog")
+        self._stop = None
+        self._thread = None
+
+    def start(self):
+        if not self.interval or self.interval <= 0:
+            return
+        import threading
+        self._stop = threading.Event()
+        self._thread = threading.Thread(target=self._run, daemon=True)
+        self._thread.start()
+
+    def _sample(self):
+        import shutil
+        if shutil.which("rocm-smi"):
+            try:
+                out = subprocess.run(
+                    ["rocm-smi", "--showuse", "--showmemuse", "--csv"],
+                    capture_output=True, text=True, timeout=10).stdout
+                return out.strip()
+            except Exception:
+                pass
+        try:
+            import torch
+            if torch.cuda.is_available():
+                return " ".join(
+                    f"gpu{i}:alloc={torch.cuda.memory_allocated(i)>>20}MiB"
+                    for i in range(torch.cuda.device_count()))
+        except Exception:
+            pass
+        return "n/a"
+
+    def _run(self):
+        with open(self.path, "a") as f:
+            while not self._stop.wait(self.interval):
+                f.write(f"{time.time():.1f} {self._sample()}\n")
+                f.flush()
+
+    def stop(self):
+        if self._stop is not None:
+            self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=5)
 
 
 if __name__ == "__main__":

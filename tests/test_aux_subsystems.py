@@ -255,3 +255,26 @@ def test_comm_desync_checker():
         comm_check.disable()
         print("rank", r, "desync checker ok")
     """, world_size=2)
+
+
+def test_tensor_checker_and_compare_accuracy(tmp_path):
+    import torch
+    import paddle_amd as paddle
+    dbg = paddle.amp.debugging
+    cfg = dbg.TensorCheckerConfig(enable=True)
+    dbg.enable_tensor_checker(cfg)
+    m = torch.nn.Linear(4, 4)
+    m(torch.randn(2, 4))
+    m.weight.data[0, 0] = float("nan")
+    import pytest as _pt
+    with _pt.raises(FloatingPointError):
+        m(torch.randn(2, 4))
+    dbg.disable_tensor_checker()
+    m(torch.randn(2, 4))  # no raise once disabled
+
+    d1 = tmp_path / "a"; d2 = tmp_path / "b"
+    d1.mkdir(); d2.mkdir()
+    torch.save({"w": torch.ones(3)}, d1 / "s0.pt")
+    torch.save({"w": torch.ones(3) * 1.5}, d2 / "s0.pt")
+    rows = dbg.compare_accuracy(str(d1), str(d2), str(tmp_path / "cmp.csv"))
+    assert rows and rows[0][4] == "DIVERGED"

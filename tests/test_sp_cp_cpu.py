@@ -133,3 +133,43 @@ def test_ring_attention_matches_single():
         assert torch.allclose(zigzag_merge(shards, w), full)
         print("rank", r, "ring ok")
     """)
+
+
+def test_sp_inner_overlap_linear_parity():
+    run_dist("""
+        import torch
+        import paddle_amd as paddle
+        from paddle_amd.distributed import fleet
+        strategy = fleet.DistributedStrategy()
+        strategy.hybrid_configs = {"dp_degree": 1, "mp_degree": 2, "pp_degree": 1,
+                                   "sharding_degree": 1}
+        fleet.init(is_collective=True, strategy=strategy)
+        from paddle_amd.distributed.fleet.sequence_parallel import (
+            ColumnSequenceParallelLinear, SPInnerOverlapLinear)
+        r = paddle.distributed.get_rank()
+        g = fleet.get_hybrid_communicate_group().get_model_parallel_group()
+        torch.manual_seed(3)
+        W = torch.randn(4, 8)
+        B = torch.randn(8)
+        x_full = torch.randn(8, 4)
+
+        col = ColumnSequenceParallelLinear(4, 8, has_bias=True, mp_group=g)
+        ovl = SPInnerOverlapLinear(4, 8, has_bias=True, mp_group=g)
+        with torch.no_grad():
+            for m in (col, ovl):
+                m.weight.copy_(W[:, r*4:(r+1)*4])
+                m.bias.copy_(B[r*4:(r+1)*4])
+
+        xs1 = x_full[r*4:(r+1)*4].clone().requires_grad_(True)
+        xs2 = x_full[r*4:(r+1)*4].clone().requires_grad_(True)
+        o1 = col(xs1)
+        o2 = ovl(xs2)
+        assert torch.allclose(o1, o2, atol=1e-5), (o1 - o2).abs().max()
+        dy = torch.randn_like(o1)
+        o1.backward(dy)
+        o2.backward(dy)
+        assert torch.allclose(xs1.grad, xs2.grad, atol=1e-5)
+        assert torch.allclose(col.weight.grad, ovl.weight.grad, atol=1e-5)
+        assert torch.allclose(col.bias.grad, ovl.bias.grad, atol=1e-5)
+        print("rank", r, "sp-overlap ok")
+    """)
