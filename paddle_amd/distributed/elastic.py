@@ -202,8 +202,26 @@ class LocalElasticAgent:
                 p.kill()
         self.procs = []
 
+    def check_failures(self):
+        """Recovery tier 2 (reference elastic/manager.py fault watch): a
+        worker that exited nonzero poisons the incarnation -- stop the
+        rest and relaunch the full set at the same world size (workers
+        resume from their own checkpoints)."""
+        if not self.procs:
+            return False
+        failed = any(p.poll() is not None and p.poll() != 0 for p in self.procs)
+        if failed:
+            world = len(self.procs)
+            self.restarts += 1
+            self._launch(world)
+        return failed
+
     def step(self, hosts=None, now=None):
         """One watch iteration; returns the plan taken."""
+        if self.check_failures():
+            return ScalePlan("restart", len(self.procs),
+                             self.scaler.last_hosts
+                             if hasattr(self.scaler, "last_hosts") else None)
         plan = self.scaler.plan(hosts, now)
         if plan.action in ("scale_out", "scale_in") or \
                 (plan.action == "none" and self.scaler.current_world is None):

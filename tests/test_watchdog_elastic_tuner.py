@@ -143,3 +143,31 @@ def test_local_elastic_agent_relaunches(tmp_path):
     w2 = [l for l in lines if l.startswith("2:")]
     w3 = [l for l in lines if l.startswith("3:")]
     assert len(w2) >= 1 and len(w3) == 3, lines
+
+
+def test_kill_rank_relaunch(tmp_path):
+    """Fault injection (SURVEY §5 test plan): rank 1 crashes in its first
+    incarnation; the agent detects the nonzero exit and relaunches the
+    full worker set, which then completes."""
+    from paddle_amd.distributed.elastic import ElasticManager, LocalElasticAgent
+    marker = tmp_path / "runs.log"
+    prog = (
+        "import os, sys\n"
+        f"open(r'{marker}', 'a').write("
+        "os.environ['PADDLE_ELASTIC_RESTART'] + ':' + os.environ['RANK'] + '\\n')\n"
+        "if os.environ['PADDLE_ELASTIC_RESTART'] == '0' and os.environ['RANK'] == '1':\n"
+        "    sys.exit(1)\n")
+    mgr = ElasticManager(job_id="t3", store_dir=str(tmp_path / "s"))
+    agent = LocalElasticAgent(mgr, "2", ["-c", prog], debounce=0.1)
+    agent.step(hosts=["a", "b"], now=1.0)          # initial launch, world 2
+    for p in agent.procs:
+        p.wait(timeout=20)
+    plan = agent.step(hosts=["a", "b"], now=2.0)   # detects rank-1 crash
+    assert plan.action == "restart"
+    for p in agent.procs:
+        p.wait(timeout=20)
+    assert all(p.poll() == 0 for p in agent.procs)
+    agent.shutdown()
+    lines = marker.read_text().strip().splitlines()
+    assert "0:1" in lines and "1:0" in lines and "1:1" in lines, lines
+    assert agent.restarts >= 1
