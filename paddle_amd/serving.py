@@ -408,6 +408,26 @@ class GPTModelRunner:
             x = x + layer.mlp(layer.ln2(x))
         return m.lm_head(m.gpt.final_norm(x[:, 0]))
 
+    def precapture(self, buckets=(1, 2, 4, 8, 16, 32)):
+        """Capture the decode hipGraphs for the given batch buckets ahead
+        of serving (otherwise the first request of each bucket pays
+        2 warmup steps + capture, which lands in its ttft)."""
+        if not self.use_graphs:
+            return
+        dev = self.dev
+        for B in buckets:
+            if B in self._graphs:
+                continue
+            toks = torch.zeros(B, 1, dtype=torch.long, device=dev)
+            pos = torch.zeros(B, 1, dtype=torch.long, device=dev)
+            rows = torch.full((B, self.max_blk), 0, dtype=torch.int32, device=dev)
+            rows[:, 0] = self.scratch_blk
+            lens = torch.zeros(B, dtype=torch.int32, device=dev)
+            wblk = torch.full((B,), self.scratch_blk, dtype=torch.long, device=dev)
+            woff = torch.zeros(B, dtype=torch.long, device=dev)
+            self._decode(B, toks, pos, rows, lens, wblk, woff)
+        torch.cuda.synchronize()
+
     def _decode(self, B, toks, pos_ids, table, lens, write_blk, write_off):
         if not self.use_graphs:
             return self._decode_forward(B, toks, pos_ids, table, lens,

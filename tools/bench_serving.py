@@ -20,6 +20,7 @@ from paddle_amd.serving import Engine, GPTModelRunner, Request  # noqa: E402
 def run_one(tag, model, n_req=64, prompt_len=128, gen_len=128, max_batch=32,
             num_blocks=4096):
     runner = GPTModelRunner(model, num_blocks=num_blocks, block_size=16)
+    runner.precapture((max_batch,))    # decode graph capture out of ttft
     eng = Engine(runner, num_blocks=num_blocks, block_size=16,
                  max_batch=max_batch)
     import random
