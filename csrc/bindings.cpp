@@ -636,6 +636,32 @@ Tensor decode_gemm(const Tensor& x, const Tensor& w,
   return y;
 }
 
+Tensor decode_gemm_int8(const Tensor& x, const Tensor& qw,
+                        const Tensor& scale,
+                        const c10::optional<Tensor>& bias) {
+  CHECK_IN(x);
+  TORCH_CHECK(x.dim() == 2 && qw.dim() == 2 && qw.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              qw.scalar_type() == torch::kChar &&
+              scale.scalar_type() == torch::kFloat);
+  int64_t m = x.size(0), k = x.size(1), n = qw.size(1);
+  TORCH_CHECK(qw.size(0) == k && m <= 32 && n % 256 == 0 && k % 64 == 0);
+  TORCH_CHECK(scale.numel() == n);
+  int64_t nblk = n / 256;
+  int64_t kmin = nblk <= 24 ? 128 : 256;
+  int64_t ksplit = std::min<int64_t>(std::max<int64_t>(1, 1024 / nblk),
+                                     std::max<int64_t>(1, k / kmin));
+  int64_t mt = m <= 16 ? 16 : 32;
+  auto ws = torch::empty({ksplit, mt, n}, x.options().dtype(torch::kFloat));
+  auto y = torch::empty({m, n}, x.options());
+  pa::decode_gemm_mfma(x.const_data_ptr(), qw.const_data_ptr(),
+                       bias.has_value() ? bias->const_data_ptr() : nullptr,
+                       y.mutable_data_ptr(), ws.mutable_data_ptr<float>(), m,
+                       n, k, n, ksplit, cur_stream(),
+                       scale.const_data_ptr<float>(), true);
+  return y;
+}
+
 Tensor decode_gemm_mfma(const Tensor& x, const Tensor& w,
                         const c10::optional<Tensor>& bias) {
   CHECK_IN(x);
@@ -800,6 +826,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bias") = c10::nullopt);
   m.def("decode_gemm_mfma", &decode_gemm_mfma, py::arg("x"), py::arg("w"),
         py::arg("bias") = c10::nullopt);
+  m.def("decode_gemm_int8", &decode_gemm_int8, py::arg("x"), py::arg("qw"),
+        py::arg("scale"), py::arg("bias") = c10::nullopt);
   m.def("moe_gate_topk", &moe_gate_topk);
   m.def("moe_assign_slots", &moe_assign_slots);
   m.def("gemm_fp8_nt", &gemm_fp8_nt, py::arg("a"), py::arg("bt"),

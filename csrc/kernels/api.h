@@ -232,7 +232,8 @@ void decode_gemm(const void* x, const void* w, const void* bias, void* y,
                  int64_t ldw, int64_t ksplit, hipStream_t s);
 void decode_gemm_mfma(const void* x, const void* w, const void* bias, void* y,
                  float* workspace, int64_t m, int64_t n, int64_t k,
-                 int64_t ldw, int64_t ksplit, hipStream_t s);
+                 int64_t ldw, int64_t ksplit, hipStream_t s,
+                 const float* chscale = nullptr, bool int8w = false);
 
 // ---- MoE routing (assign_pos/number_count/gate parity) --------------------
 void moe_gate_topk(const float* logits, float* topv, int* topi, float* me,

@@ -93,7 +93,8 @@ template <int DT>
 __global__ void decode_gemm_reduce_kernel(const float* __restrict__ partial,
                                           const short* __restrict__ bias,
                                           short* __restrict__ y, int M, int N,
-                                          int mt, int ksplit) {
+                                          int mt, int ksplit,
+                                          const float* __restrict__ chscale = nullptr) {
   long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   long long total = (long long)M * N;
   if (i >= total) return;
@@ -101,6 +102,7 @@ __global__ void decode_gemm_reduce_kernel(const float* __restrict__ partial,
   float acc = 0.f;
   for (int s = 0; s < ksplit; ++s)
     acc += partial[((long long)s * mt + m) * N + n];
+  if (chscale) acc *= chscale[n] * (1.f / 127.f);
   if (bias) acc += bf2f(bias[n]);
   y[i] = f2bf(acc);
 }
@@ -150,7 +152,13 @@ __device__ __forceinline__ int dg_swz(int kc, int n) {
   return (kc ^ ((n >> 3) & 7)) * 8;
 }
 
-template <int MTILES>           // 1: M<=16, 2: M<=32
+typedef __attribute__((ext_vector_type(8))) signed char scharx8;
+typedef __attribute__((ext_vector_type(16))) signed char scharx16;
+
+// INT8: W is [K, N] int8 (weight_quantize layout); values convert to
+// bf16 in-register before LDS staging so the MFMA path is unchanged --
+// the per-channel scale/127 applies in the reduce kernel.
+template <int MTILES, bool INT8 = false>  // MTILES 1: M<=16, 2: M<=32
 __launch_bounds__(256, 2)
 __global__ void decode_gemm_mfma_kernel(const short* __restrict__ xg,
                                         const short* __restrict__ wg,
@@ -172,10 +180,16 @@ __global__ void decode_gemm_mfma_kernel(const short* __restrict__ xg,
   const int l16 = lane & 15;
   const int lg = lane >> 4;
 
-  // staging map: 8 rounds x (256 thr x 8 elem); thread covers k row
-  // (r*8 + tid/32) at n segment (tid%32)*8
-  const int s_k = tid >> 5;           // + r*8
-  const int s_n = (tid & 31) * 8;
+  // staging map (bf16): 8 rounds x (256 thr x 8 elem); thread covers k
+  // row (r*8 + tid/32) at n segment (tid%32)*8.
+  // staging map (int8): 4 rounds x (256 thr x 16 elem) -- 16B loads keep
+  // the same bytes in flight as the bf16 path (8B loads halved the
+  // outstanding bytes and left the kernel latency-bound at ~1 TB/s).
+  // int8 map: 2 rounds x (k-pair, 16 n) per thread -- adjacent k rows
+  // pack into ONE b32 LDS write per n (b16 scatters were the bound:
+  // the int8 tile has half the bytes but the same element count)
+  const int s_k = INT8 ? (tid >> 4) * 2 : (tid >> 5);
+  const int s_n = INT8 ? (tid & 15) * 16 : (tid & 31) * 8;
 
   floatx4 acc[MTILES][4];
 #pragma unroll
@@ -194,8 +208,47 @@ __global__ void decode_gemm_mfma_kernel(const short* __restrict__ xg,
     xbase[mt] = (long long)row * K;
   }
 
+  // INT8 keeps raw 16B int8 loads in TWO reg sets (prefetch depth 2 --
+  // one tile of int8 is only half the bytes of a bf16 tile, and depth-1
+  // prefetch left the kernel in-flight-limited at ~1.2 TB/s).  The sets
+  // are NAMED arrays selected at compile time via distinct call sites
+  // (a runtime slot index into a register array spills to scratch,
+  // rule #20: measured 960 us vs 50 us).  bf16->LDS conversion happens
+  // at write time.
   shortx8 stg[8];
-  auto load_tile = [&](int kb) {    // global -> regs (coalesced rows)
+  scharx16 qstgA[4], qstgB[4];
+  const signed char* w8 = reinterpret_cast<const signed char*>(wg);
+  auto load_q = [&](scharx16(&q)[4], int kb) {
+#pragma unroll
+    for (int r = 0; r < 2; ++r)
+#pragma unroll
+      for (int p = 0; p < 2; ++p) {
+        int k = kb + r * 32 + s_k + p;
+        if (k >= k1) {
+#pragma unroll
+          for (int j = 0; j < 16; ++j) q[2 * r + p][j] = 0;
+        } else {
+          q[2 * r + p] = *reinterpret_cast<const scharx16*>(
+              w8 + (long long)k * ldw + n0 + s_n);
+        }
+      }
+  };
+  auto write_q = [&](const scharx16(&q)[4], int buf) {
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+      int kl = r * 32 + s_k;        // even; kl and kl+1 share a chunk
+      int kc = kl >> 3, ko = kl & 7;
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        int n = s_n + j;
+        unsigned lo = (unsigned short)f2bf((float)q[2 * r][j]);
+        unsigned hi = (unsigned short)f2bf((float)q[2 * r + 1][j]);
+        *reinterpret_cast<unsigned int*>(
+            &wt[buf][n * KT + dg_swz(kc, n) + ko]) = lo | (hi << 16);
+      }
+    }
+  };
+  auto load_tile = [&](int kb) {    // bf16 path: global -> regs
 #pragma unroll
     for (int r = 0; r < 8; ++r) {
       int k = kb + r * 8 + s_k;
@@ -204,7 +257,7 @@ __global__ void decode_gemm_mfma_kernel(const short* __restrict__ xg,
           : shortx8{0, 0, 0, 0, 0, 0, 0, 0};
     }
   };
-  auto write_tile = [&](int buf) {  // regs -> LDS transposed
+  auto write_tile = [&](int buf) {  // bf16 path: regs -> LDS transposed
 #pragma unroll
     for (int r = 0; r < 8; ++r) {
       int kl = r * 8 + s_k;         // k row within tile
@@ -232,17 +285,37 @@ __global__ void decode_gemm_mfma_kernel(const short* __restrict__ xg,
   // buf t%2.  WAR on buf t%2 (written next iter) and RAW on buf (t+1)%2
   // (read next iter) are both cut by the single end-of-iter barrier.
   shortx8 af[MTILES][2], afn[MTILES][2];
-  load_tile(k0);
-  write_tile(0);
-  load_tile(k0 + KT);              // regs = tile 1
+  if (INT8) {
+    load_q(qstgA, k0);
+    write_q(qstgA, 0);
+    load_q(qstgA, k0 + KT);        // A = tile 1
+    load_q(qstgB, k0 + 2 * KT);    // B = tile 2 (depth-2 prefetch)
+  } else {
+    load_tile(k0);
+    write_tile(0);
+    load_tile(k0 + KT);            // regs = tile 1
+  }
   load_a(af, k0);
   __syncthreads();
 
   int cur = 0;
+  bool useA = true;
   for (int kb = k0; kb < k1; kb += KT) {
     const bool more = kb + KT < k1;
-    if (more) write_tile(cur ^ 1); // regs from two iters back -> next buf
-    if (kb + 2 * KT < k1) load_tile(kb + 2 * KT);
+    if (INT8) {
+      if (more) {
+        if (useA) write_q(qstgA, cur ^ 1);
+        else      write_q(qstgB, cur ^ 1);
+      }
+      if (kb + 3 * KT < k1) {
+        if (useA) load_q(qstgA, kb + 3 * KT);
+        else      load_q(qstgB, kb + 3 * KT);
+      }
+      useA = !useA;
+    } else {
+      if (more) write_tile(cur ^ 1);  // regs from two iters back -> next buf
+      if (kb + 2 * KT < k1) load_tile(kb + 2 * KT);
+    }
     if (more) load_a(afn, kb + KT);
 #pragma unroll
     for (int kt = 0; kt < 2; ++kt) {
@@ -283,25 +356,25 @@ __global__ void decode_gemm_mfma_kernel(const short* __restrict__ xg,
 
 void decode_gemm_mfma(const void* x, const void* w, const void* bias, void* y,
                       float* workspace, int64_t m, int64_t n, int64_t k,
-                      int64_t ldw, int64_t ksplit, hipStream_t s) {
+                      int64_t ldw, int64_t ksplit, hipStream_t s,
+                      const float* chscale, bool int8w) {
   const int mt = m <= 16 ? 16 : 32;
   // kchunk: multiple of 64 so every split starts tile-aligned
   int kchunk = (int)(((k + ksplit - 1) / ksplit + 63) / 64) * 64;
   const int ks = (int)((k + kchunk - 1) / kchunk);
   dim3 grid((unsigned)(n / 256), (unsigned)ks);
-  if (mt == 16)
-    hipLaunchKernelGGL((decode_gemm_mfma_kernel<1>), grid, dim3(256), 0, s,
-                       (const short*)x, (const short*)w, workspace, (int)m,
-                       (int)n, (int)k, ldw, kchunk);
-  else
-    hipLaunchKernelGGL((decode_gemm_mfma_kernel<2>), grid, dim3(256), 0, s,
-                       (const short*)x, (const short*)w, workspace, (int)m,
-                       (int)n, (int)k, ldw, kchunk);
+#define DGM(MT, I8)                                                          \
+  hipLaunchKernelGGL((decode_gemm_mfma_kernel<MT, I8>), grid, dim3(256), 0,  \
+                     s, (const short*)x, (const short*)w, workspace, (int)m, \
+                     (int)n, (int)k, ldw, kchunk)
+  if (int8w) { if (mt == 16) DGM(1, true); else DGM(2, true); }
+  else       { if (mt == 16) DGM(1, false); else DGM(2, false); }
+#undef DGM
   long long total = m * n;
   dim3 rg((unsigned)((total + 255) / 256));
   hipLaunchKernelGGL((decode_gemm_reduce_kernel<0>), rg, dim3(256), 0, s,
                      workspace, (const short*)bias, (short*)y, (int)m, (int)n,
-                     mt, ks);
+                     mt, ks, chscale);
 }
 
 }  // namespace pa

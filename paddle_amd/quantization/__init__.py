@@ -62,26 +62,35 @@ class PTQ:
 # -- weight-only int8 (serving decode path) ----------------------------------
 # parity: paddle/phi/kernels/weight_quantize_kernel.cu + funcs/weight_only_gemv.cu
 def weight_quantize(w, algo="weight_only_int8"):
-    """w [K, N] (paddle Linear layout) -> (qweight [N, K] int8 row-major,
-    scale [N] fp32).  Per-output-channel absmax; w ~ qw.T * scale / 127."""
+    """w [K, N] (paddle Linear layout) -> (qweight [K, N] int8 row-major,
+    scale [N] fp32).  Per-output-channel absmax; w ~ qw * scale / 127.
+
+    The [K, N] layout feeds the MFMA W-streaming decode kernel with
+    coalesced rows (the round-1 [N, K] GEMV layout read 180-330 GB/s;
+    the streamer reads at the memory roofline on HALF the bytes)."""
     import torch
     assert algo in ("weight_only_int8",)
-    wt = w.detach().float().t().contiguous()        # [N, K]
-    scale = wt.abs().amax(dim=1).clamp(min=1e-8)    # [N]
-    q = torch.round(wt / scale.unsqueeze(1) * 127.0).clamp(-127, 127).to(torch.int8)
-    return q, scale
+    wf = w.detach().float()                          # [K, N]
+    scale = wf.abs().amax(dim=0).clamp(min=1e-8)     # [N]
+    q = torch.round(wf / scale.unsqueeze(0) * 127.0).clamp(-127, 127).to(torch.int8)
+    return q.contiguous(), scale
 
 
 def weight_only_linear(x, qweight, scale, bias=None, weight_dtype="int8"):
-    """x [..., K] @ dequant(qweight) + bias.  GPU decode path runs the
-    wave-per-channel int8 GEMV HIP kernel; fallback dequantizes."""
+    """x [..., K] @ dequant(qweight) + bias.  Decode shapes (<= 32 rows)
+    run the int8 MFMA W-streaming kernel; fallback dequantizes."""
     import torch
     from .. import _ext
+    k, n = qweight.shape
     if x.is_cuda and _ext.use_native(x):
-        C = _ext.get_ext()
-        b = bias.to(x.dtype) if bias is not None else None
-        return C.weight_only_gemv(x.contiguous(), qweight, scale, b)
-    w = (qweight.float() * scale.unsqueeze(1) / 127.0).t().to(x.dtype)  # [K,N]
+        rows = x.numel() // x.shape[-1]
+        if rows <= 32 and n % 256 == 0 and k % 64 == 0 and x.dtype == torch.bfloat16:
+            C = _ext.get_ext()
+            b = bias.to(torch.bfloat16) if bias is not None else None
+            out = C.decode_gemm_int8(x.reshape(-1, k), qweight,
+                                     scale.float(), b)
+            return out.reshape(*x.shape[:-1], n)
+    w = (qweight.float() * scale.unsqueeze(0) / 127.0).to(x.dtype)  # [K,N]
     out = x @ w
     if bias is not None:
         out = out + bias

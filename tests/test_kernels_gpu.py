@@ -321,18 +321,19 @@ def test_fused_ffn_hipblaslt_gpu():
     torch.testing.assert_close(w2.grad.float(), w2f.grad, atol=0.8, rtol=0.08)
 
 
-def test_weight_only_gemv_gpu():
-    """int8 weight-only decode GEMV vs dequantized matmul reference."""
+@pytest.mark.parametrize("m,k,n", [(4, 4096, 1024), (32, 4160, 4096),
+                                   (16, 512, 256)])
+def test_weight_only_decode_gpu(m, k, n):
+    """int8 weight-only decode (MFMA W-streamer) vs dequantized matmul."""
     from paddle_amd import quantization as Q
     torch.manual_seed(13)
-    K, N, M = 4096, 1024, 4
-    w = torch.randn(K, N, device=DEV) * 0.1
-    b = torch.randn(N, device=DEV)
+    w = torch.randn(k, n, device=DEV) * 0.1
+    b = torch.randn(n, device=DEV)
     qw, sc = Q.weight_quantize(w)
-    x = torch.randn(M, K, device=DEV, dtype=torch.bfloat16)
+    x = torch.randn(m, k, device=DEV, dtype=torch.bfloat16)
     out = Q.weight_only_linear(x, qw.to(DEV), sc.to(DEV),
                                bias=b.to(torch.bfloat16))
-    wref = (qw.to(DEV).float() * sc.to(DEV).unsqueeze(1) / 127.0).t()
+    wref = qw.to(DEV).float() * sc.to(DEV).unsqueeze(0) / 127.0
     ref = x.float() @ wref + b
     torch.testing.assert_close(out.float(), ref, atol=0.5, rtol=0.05)
 
