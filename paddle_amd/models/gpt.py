@@ -107,17 +107,18 @@ class GPTMLP(nn.Layer):
         _apply_initializer(out_init, self.fc2.weight)
 
     def forward(self, x):
-        if hot._own_linear_ok(x, self.fc1.weight):
+        w1 = getattr(self.fc1, "weight", None)   # None once weight-only-quantized
+        if w1 is not None and hot._own_linear_ok(x, w1):
             # own MFMA NT GEMM: fc1 bias+GELU in the kernel epilogue (aux
             # saved), fc2 dgrad carries dGELU in its epilogue
-            return hot.fused_ffn_own(x, self.fc1.weight, self.fc1_bias,
+            return hot.fused_ffn_own(x, w1, self.fc1_bias,
                                      self.fc2.weight, self.fc2.bias)
-        if hot._fused_ffn_available(x):
+        if w1 is not None and hot._fused_ffn_available(x):
             # fc1 bias+GELU in the hipBLASLt epilogue; backward fuses
             # dGELU + fc1 bias-grad into fc2's dgrad GEMM
-            return hot.fused_ffn(x, self.fc1.weight, self.fc1_bias,
+            return hot.fused_ffn(x, w1, self.fc1_bias,
                                  self.fc2.weight, self.fc2.bias)
-        # CPU / non-bf16 path: fused bias+gelu elementwise kernel
+        # CPU / non-bf16 / weight-only path: fused bias+gelu kernel
         return self.fc2(hot.bias_gelu(self.fc1(x), self.fc1_bias))
 
 

@@ -8,7 +8,8 @@ from ...quantization import (  # noqa: F401
 
 def weight_dequantize(qweight, scale, algo="weight_only_int8", out_dtype=None):
     import torch
-    w = (qweight.float() * scale.unsqueeze(1) / 127.0).t()
+    # weight_quantize layout: qweight [K, N], per-output-channel scale [N]
+    w = qweight.float() * scale.unsqueeze(0) / 127.0
     return w.to(out_dtype or torch.float16)
 
 

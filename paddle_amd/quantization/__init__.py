@@ -95,3 +95,57 @@ def weight_only_linear(x, qweight, scale, bias=None, weight_dtype="int8"):
     if bias is not None:
         out = out + bias
     return out
+
+
+class WeightOnlyLinear(__import__("torch").nn.Module):
+    """Drop-in replacement for a (paddle-layout) nn.Linear holding int8
+    weights; decode shapes run the MFMA W-streaming int8 kernel, larger
+    shapes dequantize on the fly.  Reference: paddle/nn/quant
+    weight-only path."""
+
+    def __init__(self, qweight, scale, bias=None):
+        super().__init__()
+        self.register_buffer("qweight", qweight)
+        self.register_buffer("scale", scale)
+        self.register_buffer("wo_bias", bias)
+        self.in_features = qweight.shape[0]
+        self.out_features = qweight.shape[1]
+
+    @classmethod
+    def from_linear(cls, linear):
+        qw, sc = weight_quantize(linear.weight.detach())
+        dev = linear.weight.device
+        bias = getattr(linear, "bias", None)
+        return cls(qw.to(dev), sc.to(dev),
+                   bias.detach().clone() if bias is not None else None)
+
+    def forward(self, x):
+        return weight_only_linear(x, self.qweight, self.scale, self.wo_bias)
+
+
+def quantize_linears_(model, min_features=1024, skip=("lm_head",)):
+    """Replace every Linear child of `model` (recursively) whose weight
+    is at least [min_features, min_features] with a WeightOnlyLinear --
+    int8 weights, ~2x less weight memory for serving.  Returns the
+    number of layers converted."""
+    import torch
+    n = 0
+
+    def walk(mod, prefix=""):
+        nonlocal n
+        for name, child in list(mod._modules.items()):
+            if child is None:
+                continue
+            full = f"{prefix}{name}"
+            if any(s in full for s in skip):
+                continue
+            w = getattr(child, "weight", None)
+            if (type(child).__name__ == "Linear" and isinstance(w, torch.Tensor)
+                    and w.dim() == 2 and min(w.shape) >= min_features):
+                mod._modules[name] = WeightOnlyLinear.from_linear(child)
+                n += 1
+            else:
+                walk(child, full + ".")
+
+    walk(model)
+    return n

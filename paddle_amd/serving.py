@@ -236,11 +236,17 @@ class GPTModelRunner:
     decode runs every active request through `paged_decode_attention`."""
 
     def __init__(self, model, num_blocks=1024, block_size=16,
-                 device=None, dtype=None, max_seq=2048, use_graphs=True):
+                 device=None, dtype=None, max_seq=2048, use_graphs=True,
+                 weight_only=False):
         import math
         from .ops import functional as hot
         self.hot = hot
         self.model = model.eval()
+        if weight_only:
+            # int8 weight-only decode: ~2x less weight memory; decode
+            # linears run the MFMA W-streaming int8 kernel
+            from .quantization import quantize_linears_
+            quantize_linears_(self.model, skip=("lm_head", "embeddings"))
         cfg = model.cfg
         self.H = cfg.num_heads
         self.D = cfg.hidden_size // self.H

@@ -116,3 +116,22 @@ def test_engine_fifo_admission():
             for _ in range(3)]          # each needs 1 block; pool of 3
     eng.run_until_done()
     assert order == rids
+
+
+def test_weight_only_quantized_model_cpu():
+    """quantize_linears_ swaps decoder Linears for int8 WeightOnlyLinear;
+    outputs stay within int8 round-off of the bf16 model."""
+    import torch
+    import paddle_amd as paddle
+    from paddle_amd import quantization as Q
+    from paddle_amd.models import build_gpt
+    torch.manual_seed(0)
+    m = build_gpt("gpt3-tiny", max_seq_len=128).eval()
+    ids = torch.randint(0, 1000, (2, 16))
+    with torch.no_grad():
+        ref = m(ids)
+        n = Q.quantize_linears_(m, min_features=8)
+        out = m(ids)
+    assert n > 0
+    rel = (out.float() - ref.float()).abs().max() / ref.float().abs().max()
+    assert rel < 0.05, float(rel)

@@ -18,8 +18,9 @@ from paddle_amd.serving import Engine, GPTModelRunner, Request  # noqa: E402
 
 
 def run_one(tag, model, n_req=64, prompt_len=128, gen_len=128, max_batch=32,
-            num_blocks=4096):
-    runner = GPTModelRunner(model, num_blocks=num_blocks, block_size=16)
+            num_blocks=4096, weight_only=False):
+    runner = GPTModelRunner(model, num_blocks=num_blocks, block_size=16,
+                            weight_only=weight_only)
     runner.precapture((max_batch,))    # decode graph capture out of ttft
     eng = Engine(runner, num_blocks=num_blocks, block_size=16,
                  max_batch=max_batch)
@@ -59,16 +60,20 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="gpt-350m",
                     choices=["gpt-350m", "gpt3-6.7b"])
+    ap.add_argument("--weight-only", action="store_true",
+                    help="int8 weight-only decode (MFMA W-streamer)")
     args = ap.parse_args()
     paddle.seed(0)
+    tag_sfx = " int8-wo" if args.weight_only else ""
     if args.model == "gpt-350m":
         cfg = GPTConfig(vocab_size=50304, hidden_size=1024, num_layers=24,
                         num_heads=16, intermediate_size=4096, max_seq_len=2048)
         m = GPTForPretraining(cfg).to("cuda", torch.bfloat16)
-        run_one("gpt-350M", m)
+        run_one("gpt-350M" + tag_sfx, m, weight_only=args.weight_only)
     else:
         m = build_gpt("gpt3-6.7b", max_seq_len=2048).to("cuda", torch.bfloat16)
-        run_one("gpt3-6.7B", m, n_req=32, max_batch=16, num_blocks=8192)
+        run_one("gpt3-6.7B" + tag_sfx, m, n_req=32, max_batch=16,
+                num_blocks=8192, weight_only=args.weight_only)
 
 
 if __name__ == "__main__":
