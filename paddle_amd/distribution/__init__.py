@@ -106,6 +106,23 @@ import torch as _t
 
 ExponentialFamily = _t.distributions.ExponentialFamily
 Independent = _t.distributions.Independent
-TransformedDistribution = _t.distributions.TransformedDistribution
+class TransformedDistribution(_t.distributions.TransformedDistribution):
+    """Accepts paddle-style Transform wrappers (unwraps to torch) or raw
+    torch transforms (reference transformed_distribution.py)."""
+
+    def __init__(self, base, transforms):
+        base_t = base._t if isinstance(base, Distribution) else base
+        if not isinstance(transforms, (list, tuple)):
+            transforms = [transforms]
+        tts = [getattr(t, "_t", t) for t in transforms]
+        super().__init__(base_t, tts)
 LKJCholesky = _t.distributions.LKJCholesky
 register_kl = _t.distributions.register_kl
+
+
+from .transform import (  # noqa: F401,E402
+    AbsTransform, AffineTransform, ChainTransform, ExpTransform,
+    IndependentTransform, PowerTransform, ReshapeTransform,
+    SigmoidTransform, SoftmaxTransform, StackTransform,
+    StickBreakingTransform, TanhTransform, Transform)
+from . import transform  # noqa: F401,E402

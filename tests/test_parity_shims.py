@@ -150,3 +150,24 @@ def test_to_static_passthrough_semantics():
     out = f(torch.ones(2), torch.ones(2))
     assert torch.allclose(out, torch.full((2,), 3.0))
     assert callable(f.dygraph_function)
+
+
+def test_distribution_transforms():
+    import torch
+    import paddle_amd as paddle
+    D = paddle.distribution
+    x = torch.tensor([0.3, 0.7])
+    for t, dom in [(D.ExpTransform(), x), (D.TanhTransform(), x),
+                   (D.SigmoidTransform(), x),
+                   (D.AffineTransform(torch.tensor(1.0), torch.tensor(3.0)), x),
+                   (D.PowerTransform(torch.tensor(2.0)), x)]:
+        y = t.forward(dom)
+        back = t.inverse(y)
+        assert torch.allclose(back, dom, atol=1e-5), type(t).__name__
+        assert t.forward_log_det_jacobian(dom).shape == dom.shape
+    ch = D.ChainTransform([D.ExpTransform(), D.PowerTransform(torch.tensor(2.0))])
+    assert torch.allclose(ch.inverse(ch.forward(x)), x, atol=1e-5)
+    td = D.TransformedDistribution(D.Normal(0.0, 1.0), [D.ExpTransform()])
+    s = td.sample((64,))
+    assert (s > 0).all()
+    assert td.log_prob(torch.tensor([1.0])).shape == (1,)
