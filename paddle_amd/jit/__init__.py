@@ -20,19 +20,57 @@ import torch
 
 
 class StaticFunction:
+    """Trace-based program capture (the MI355X stand-in for the
+    reference's AST/SOT to_static, dygraph_to_static/program_translator.py):
+    the first call with a given input signature traces the function
+    through the substrate (op wrappers in trace mode, jit/__init__.py
+    _substrate_only) into a TorchScript graph; matching calls replay the
+    captured program.  Data-dependent control flow is baked at trace
+    time -- calls whose tensor ranks/dtypes differ re-trace, and
+    non-traceable calls fall back to eager."""
+
     def __init__(self, fn, input_spec=None, full_graph=False):
         self._fn = fn
         self.input_spec = input_spec
+        self._traces = {}
+        self._trace_failed = False
+
+    @staticmethod
+    def _sig(args):
+        parts = []
+        for a in args:
+            if isinstance(a, torch.Tensor):
+                parts.append(("T", tuple(a.shape), str(a.dtype), str(a.device)))
+            else:
+                return None
+        return tuple(parts)
 
     def __call__(self, *args, **kwargs):
-        return self._fn(*args, **kwargs)
+        sig = None if (kwargs or self._trace_failed or not args) \
+            else self._sig(args)
+        if sig is None:
+            return self._fn(*args, **kwargs)
+        mod = self._traces.get(sig)
+        if mod is None:
+            try:
+                with _substrate_only():
+                    mod = torch.jit.trace(self._fn, args, check_trace=False,
+                                          strict=False)
+                self._traces[sig] = mod
+            except Exception:
+                self._trace_failed = True
+                return self._fn(*args)
+        return mod(*args)
 
     @property
     def dygraph_function(self):
         return self._fn
 
+    @property
     def concrete_program(self):
-        return None
+        """Last captured TorchScript program (None before first call)."""
+        return next(reversed(self._traces.values()), None) \
+            if self._traces else None
 
 
 def to_static(function=None, input_spec=None, full_graph=False, backend=None, **kwargs):

@@ -151,3 +151,29 @@ def test_save_load_inference_model(tmp_path):
     out = loaded.run({"x": feed_x})[0]
     torch.testing.assert_close(torch.as_tensor(out), torch.as_tensor(ref),
                                atol=1e-5, rtol=1e-5)
+
+
+def test_to_static_captures_and_replays():
+    import torch
+    import paddle_amd as paddle
+
+    calls = {"n": 0}
+
+    @paddle.jit.to_static
+    def f(x, y):
+        calls["n"] += 1
+        return paddle.nn.functional.relu(x) @ y + 1
+
+    x, y = torch.randn(4, 8), torch.randn(8, 2)
+    ref = torch.relu(x) @ y + 1
+    out1 = f(x, y)
+    assert f.concrete_program is not None
+    n_after_trace = calls["n"]
+    out2 = f(x, y)                      # replayed, no python re-execution
+    assert calls["n"] == n_after_trace
+    assert torch.allclose(out1, ref) and torch.allclose(out2, ref)
+    xg = x.clone().requires_grad_(True)
+    f(xg, y).sum().backward()           # autograd through the capture
+    assert xg.grad is not None
+    f(torch.randn(3, 8), y)             # new signature -> retrace
+    assert len(f._traces) == 2
