@@ -20,5 +20,5 @@ for m, k, n in [(16, 4096, 16384), (32, 4096, 16384), (32, 16384, 4096)]:
     ref = x.float() @ ws[0].float()
     got = Q.weight_only_linear(x, qs[0][0], qs[0][1]).float()
     err = (got - ref).abs().max().item() / ref.abs().max().item()
-    print(f"M{m} K{k} N{n}: bf16-lt {t_lt*1e6:7.1f}us ({gb/t_lt:5.2f} TB/s)  "
-          f"int8-wo {t_q*1e6:7.1f}us ({gb/2/t_q:5.2f} TB/s-int8)  relerr {err:.3f}")
+    print(f"M{m} K{k} N{n}: bf16-lt {t_lt*1e6:7.1f}us ({gb/t_lt*1e-3:5.2f} TB/s)  "
+          f"int8-wo {t_q*1e6:7.1f}us ({gb/2/t_q*1e-3:5.2f} TB/s-int8)  relerr {err:.3f}")
