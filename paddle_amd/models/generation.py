@@ -106,3 +106,74 @@ def generate_gpt(model, input_ids, max_new_tokens=16, cache_blocks=None,
         cache.advance(1)
         logits = model.lm_head(model.gpt.final_norm(x))
     return torch.cat(out_tokens, dim=1)
+
+
+def generate_llama(model, input_ids, max_new_tokens=16, cache_blocks=None,
+                   block_size=16, greedy=True):
+    """Greedy/sampled generation for LlamaForCausalLM with a paged GQA KV
+    cache: rope applied at the absolute position, K/V cached at
+    num_kv_heads, paged decode attention broadcasting HKV -> H
+    (decode_attn.hip handles the head mapping in-kernel).
+
+    Reference role: the PaddleNLP llama generation loop over the paged
+    block_multihead path; re-derived here against models/llama.py."""
+    model.eval()
+    cfg = model.cfg
+    B, S0 = input_ids.shape
+    dev = input_ids.device
+    H = cfg.num_heads
+    HKV = cfg.num_kv_heads
+    D = cfg.hidden_size // H
+    max_seq = S0 + max_new_tokens
+    nblocks = cache_blocks or (B * ((max_seq + block_size - 1) // block_size))
+    cache = PagedKVCache(cfg.num_layers, nblocks, block_size, HKV, D, B, max_seq,
+                         dev, dtype=model.lm_head.weight.dtype)
+
+    def layer_qkv(layer, x, pos0):
+        b, s, _ = x.shape
+        a = layer.self_attn
+        q = a.q_proj(x).reshape(b, s, H, D)
+        k = a.k_proj(x).reshape(b, s, HKV, D)
+        v = a.v_proj(x).reshape(b, s, HKV, D)
+        q, k = hot.fused_rotary_position_embedding(q, k, base=cfg.rope_base,
+                                                   pos_offset=pos0)
+        return q, k, v
+
+    # ---- prefill ----------------------------------------------------------
+    x = model.llama.embed_tokens(input_ids)
+    for li, layer in enumerate(model.llama.layers):
+        h = layer.input_layernorm(x)
+        q, k, v = layer_qkv(layer, h, 0)
+        cache.append(li, k, v, 0)
+        att, _ = hot.flash_attention(q, k, v, causal=True)
+        x = x + layer.self_attn.o_proj(
+            att.reshape(att.shape[0], att.shape[1], -1))
+        x = x + layer.mlp(layer.post_attention_layernorm(x))
+    cache.advance(S0)
+    logits = model.lm_head(model.llama.norm(x[:, -1:]))
+    out_tokens = []
+
+    # ---- decode loop ------------------------------------------------------
+    scale = 1.0 / math.sqrt(D)
+    for step in range(max_new_tokens):
+        if greedy:
+            nxt = logits[:, -1].argmax(-1, keepdim=True)
+        else:
+            probs = torch.softmax(logits[:, -1].float(), -1)
+            nxt = torch.multinomial(probs, 1)
+        out_tokens.append(nxt)
+        pos = S0 + step
+        x = model.llama.embed_tokens(nxt)
+        for li, layer in enumerate(model.llama.layers):
+            h = layer.input_layernorm(x)
+            q, k, v = layer_qkv(layer, h, pos)
+            cache.append(li, k, v, pos)
+            cache_lens = cache.seq_lens + 1
+            att = hot.paged_decode_attention(
+                q.reshape(B, H, D), cache.k[li], cache.v[li],
+                cache.block_table, cache_lens, scale)
+            x = x + layer.self_attn.o_proj(att.reshape(B, 1, H * D))
+            x = x + layer.mlp(layer.post_attention_layernorm(x))
+        cache.advance(1)
+        logits = model.lm_head(model.llama.norm(x))
+    return torch.cat(out_tokens, dim=1)

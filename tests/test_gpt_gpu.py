@@ -162,3 +162,23 @@ def test_serving_engine_gpu():
     eng2.add_request(r2)
     eng2.run_until_done()
     assert r2.out_ids == reqs[0].out_ids
+
+
+@pytest.mark.gpu
+def test_llama_paged_generation_gpu():
+    """GQA paged decode on the HIP kernel path (bf16): first generated
+    token matches the full-forward argmax."""
+    import torch
+    from paddle_amd.models.llama import LlamaConfig, LlamaForCausalLM
+    from paddle_amd.models.generation import generate_llama
+    torch.manual_seed(1)
+    cfg = LlamaConfig(vocab_size=1024, hidden_size=512, num_layers=3,
+                      num_heads=8, num_kv_heads=2, intermediate_size=1024,
+                      max_seq_len=256)
+    m = LlamaForCausalLM(cfg).to("cuda", torch.bfloat16).eval()
+    ids = torch.randint(0, 1024, (2, 33), device="cuda")
+    with torch.no_grad():
+        gen = generate_llama(m, ids, max_new_tokens=6)
+        ref = m(ids)[:, -1].argmax(-1)
+    assert gen.shape == (2, 6)
+    assert (gen[:, 0] == ref).all()
