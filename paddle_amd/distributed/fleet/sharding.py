@@ -285,6 +285,12 @@ class GroupShardedStage3(torch.nn.Module):
 
     # -- hooks ----------------------------------------------------------------
     def _register_hooks(self):
+        # the full-backward-pre hook intentionally keys off OUTPUT grads
+        # ("backward reached this unit"); torch warns when module inputs
+        # don't require grad (the embedding unit's int ids) -- expected
+        import warnings
+        warnings.filterwarnings(
+            "ignore", message="Full backward hook is firing")
         for u in self._units:
             u.module.register_forward_pre_hook(self._make_fwd_pre(u))
             u.module.register_forward_hook(self._make_fwd_post(u))
