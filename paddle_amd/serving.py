@@ -249,6 +249,7 @@ class GPTModelRunner:
             quantize_linears_(self.model, skip=("lm_head", "embeddings"))
         cfg = model.cfg
         self.H = cfg.num_heads
+        self.HKV = getattr(cfg, "num_kv_heads", cfg.num_heads)   # GQA-aware
         self.D = cfg.hidden_size // self.H
         self.scale = 1.0 / math.sqrt(self.D)
         self.bs = block_size
@@ -258,7 +259,7 @@ class GPTModelRunner:
         # +1 scratch block: hipGraph-padded rows write their (discarded) KV
         # into it so every captured bucket replays with fixed shapes
         self.scratch_blk = num_blocks
-        self.k = [torch.zeros(num_blocks + 1, block_size, self.H, self.D,
+        self.k = [torch.zeros(num_blocks + 1, block_size, self.HKV, self.D,
                               device=dev, dtype=dt)
                   for _ in range(cfg.num_layers)]
         self.v = [torch.zeros_like(self.k[0]) for _ in range(cfg.num_layers)]
@@ -468,3 +469,85 @@ class GPTModelRunner:
         st["woff"].copy_(write_off)
         g.replay()
         return st["logits"]
+
+
+class LlamaModelRunner(GPTModelRunner):
+    """Engine step_fn for LlamaForCausalLM: GQA paged KV (num_kv_heads
+    cache planes), rope applied at each request's OWN position via a
+    gathered cos/sin table (graph-capture-safe: the gather reads a
+    static pos tensor updated before replay), RMSNorm + swiglu layers."""
+
+    def __init__(self, model, *args, **kwargs):
+        super().__init__(model, *args, **kwargs)
+        cfg = model.cfg
+        from .ops.functional import build_rope_cache
+        cos, sin = build_rope_cache(self.max_blk * self.bs, self.D,
+                                    cfg.rope_base, self.dev)
+        self._rope_cos = cos.to(torch.float32)        # [S, D/2]
+        self._rope_sin = sin.to(torch.float32)
+
+    def _rope_rows(self, x, pos):
+        """x [B, 1, H, D] at per-row absolute positions pos [B] (long)."""
+        c = self._rope_cos[pos].view(-1, 1, 1, self.D // 2).to(x.dtype)
+        s = self._rope_sin[pos].view(-1, 1, 1, self.D // 2).to(x.dtype)
+        x1, x2 = x[..., : self.D // 2], x[..., self.D // 2:]
+        return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
+
+    def _qkv(self, layer, x):           # prefill path (positions 0..S0)
+        b, s, _ = x.shape
+        a = layer.self_attn
+        q = a.q_proj(x).reshape(b, s, self.H, self.D)
+        k = a.k_proj(x).reshape(b, s, self.HKV, self.D)
+        v = a.v_proj(x).reshape(b, s, self.HKV, self.D)
+        q, k = self.hot.fused_rotary_position_embedding(
+            q, k, base=self.model.cfg.rope_base)
+        return q, k, v
+
+    @torch.no_grad()
+    def _prefill(self, reqs, blocks):
+        m = self.model
+        N = len(reqs)
+        S0 = len(reqs[0].prompt_ids)
+        ids = torch.tensor([r.prompt_ids for r in reqs], dtype=torch.long).to(
+            self.dev, non_blocking=True)
+        pos = torch.arange(S0, device=self.dev)
+        blk_t = torch.tensor([blocks[r.rid] for r in reqs],
+                             dtype=torch.long).to(self.dev, non_blocking=True)
+        blks = blk_t[:, pos // self.bs]
+        offs = (pos % self.bs).unsqueeze(0).expand(N, S0)
+        x = m.llama.embed_tokens(ids)
+        for li, layer in enumerate(m.llama.layers):
+            h = layer.input_layernorm(x)
+            q, k, v = self._qkv(layer, h)
+            self.k[li][blks, offs] = k
+            self.v[li][blks, offs] = v
+            att, _ = self.hot.flash_attention(q, k, v, causal=True)
+            x = x + layer.self_attn.o_proj(att.reshape(N, S0, -1))
+            x = x + layer.mlp(layer.post_attention_layernorm(x))
+        logits = m.lm_head(m.llama.norm(x[:, -1]))
+        for i, r in enumerate(reqs):
+            self.seq_len[r.rid] = S0
+            self.last_token[r.rid] = sample_token(logits[i], r.temperature,
+                                                  r.top_p)
+
+    def _decode_forward(self, B, toks, pos_ids, table, lens, write_blk,
+                        write_off):
+        m = self.model
+        a_pos = lens.long()                 # absolute position of new token
+        x = m.llama.embed_tokens(toks)
+        for li, layer in enumerate(m.llama.layers):
+            h = layer.input_layernorm(x)
+            at = layer.self_attn
+            q = at.q_proj(h).reshape(B, 1, self.H, self.D)
+            k = at.k_proj(h).reshape(B, 1, self.HKV, self.D)
+            v = at.v_proj(h).reshape(B, 1, self.HKV, self.D)
+            q = self._rope_rows(q, a_pos)
+            k = self._rope_rows(k, a_pos)
+            self.k[li][write_blk, write_off] = k[:, 0]
+            self.v[li][write_blk, write_off] = v[:, 0]
+            att = self.hot.paged_decode_attention(
+                q.reshape(B, self.H, self.D), self.k[li], self.v[li],
+                table, lens + 1, self.scale)
+            x = x + layer.self_attn.o_proj(att.reshape(B, 1, -1))
+            x = x + layer.mlp(layer.post_attention_layernorm(x))
+        return m.lm_head(m.llama.norm(x[:, 0]))

@@ -135,3 +135,30 @@ def test_weight_only_quantized_model_cpu():
     assert n > 0
     rel = (out.float() - ref.float()).abs().max() / ref.float().abs().max()
     assert rel < 0.05, float(rel)
+
+
+def test_llama_runner_end_to_end_cpu():
+    """Engine + LlamaModelRunner (GQA paged cache, per-row rope in
+    decode): continuous batching matches the standalone generate_llama
+    tokens for the same prompt."""
+    import paddle_amd as paddle
+    from paddle_amd.models.llama import LlamaConfig, LlamaForCausalLM
+    from paddle_amd.models.generation import generate_llama
+    from paddle_amd.serving import Engine, LlamaModelRunner, Request
+    paddle.seed(0)
+    cfg = LlamaConfig(vocab_size=512, hidden_size=128, num_layers=2,
+                      num_heads=8, num_kv_heads=4, intermediate_size=256,
+                      max_seq_len=128)
+    m = LlamaForCausalLM(cfg).to("cpu").float().eval()
+    runner = LlamaModelRunner(m, num_blocks=64, block_size=16,
+                              device=torch.device("cpu"), max_seq=128)
+    eng = Engine(runner, num_blocks=64, block_size=16, max_batch=3)
+    reqs = [Request(prompt_ids=[3 + i, 7, 11, 20], max_new_tokens=5)
+            for i in range(3)]
+    for r in reqs:
+        eng.add_request(r)
+    eng.run_until_done()
+    assert all(r.done and len(r.out_ids) == 5 for r in reqs)
+    assert len(eng.alloc.free) == 64
+    ref = generate_llama(m, torch.tensor([[3, 7, 11, 20]]), max_new_tokens=5)
+    assert reqs[0].out_ids == ref[0].tolist()
