@@ -18,9 +18,13 @@ from paddle_amd.serving import Engine, GPTModelRunner, Request  # noqa: E402
 
 
 def run_one(tag, model, n_req=64, prompt_len=128, gen_len=128, max_batch=32,
-            num_blocks=4096, weight_only=False):
-    runner = GPTModelRunner(model, num_blocks=num_blocks, block_size=16,
-                            weight_only=weight_only)
+            num_blocks=4096, weight_only=False, runner_cls="gpt"):
+    cls = GPTModelRunner
+    if runner_cls == "llama":
+        from paddle_amd.serving import LlamaModelRunner
+        cls = LlamaModelRunner
+    runner = cls(model, num_blocks=num_blocks, block_size=16,
+                 weight_only=weight_only)
     runner.precapture((max_batch,))    # decode graph capture out of ttft
     eng = Engine(runner, num_blocks=num_blocks, block_size=16,
                  max_batch=max_batch)
@@ -59,7 +63,7 @@ def main():
     import argparse
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="gpt-350m",
-                    choices=["gpt-350m", "gpt3-6.7b"])
+                    choices=["gpt-350m", "gpt3-6.7b", "llama2-7b"])
     ap.add_argument("--weight-only", action="store_true",
                     help="int8 weight-only decode (MFMA W-streamer)")
     args = ap.parse_args()
@@ -70,10 +74,16 @@ def main():
                         num_heads=16, intermediate_size=4096, max_seq_len=2048)
         m = GPTForPretraining(cfg).to("cuda", torch.bfloat16)
         run_one("gpt-350M" + tag_sfx, m, weight_only=args.weight_only)
-    else:
+    elif args.model == "gpt3-6.7b":
         m = build_gpt("gpt3-6.7b", max_seq_len=2048).to("cuda", torch.bfloat16)
         run_one("gpt3-6.7B" + tag_sfx, m, n_req=32, max_batch=16,
                 num_blocks=8192, weight_only=args.weight_only)
+    else:
+        from paddle_amd.models.llama import build_llama
+        m = build_llama("llama2-7b", max_seq_len=2048).to("cuda", torch.bfloat16)
+        run_one("llama2-7B" + tag_sfx, m, n_req=32, max_batch=16,
+                num_blocks=8192, weight_only=args.weight_only,
+                runner_cls="llama")
 
 
 if __name__ == "__main__":
