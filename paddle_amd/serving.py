@@ -485,6 +485,27 @@ class LlamaModelRunner(GPTModelRunner):
                                     cfg.rope_base, self.dev)
         self._rope_cos = cos.to(torch.float32)        # [S, D/2]
         self._rope_sin = sin.to(torch.float32)
+        # fuse q/k/v into ONE decode GEMM per layer (3 separate skinny
+        # GEMMs read the same activations 3x and hit hipBLASLt's weak
+        # small-N tiles); bf16 weights become views into the fused matrix
+        self._qkv_fused = []
+        for layer in model.llama.layers:
+            a = layer.self_attn
+            if hasattr(a.q_proj, "qweight"):          # weight-only int8
+                qw = torch.cat([a.q_proj.qweight, a.k_proj.qweight,
+                                a.v_proj.qweight], dim=1).contiguous()
+                sc = torch.cat([a.q_proj.scale, a.k_proj.scale,
+                                a.v_proj.scale]).contiguous()
+                self._qkv_fused.append(("int8", qw, sc))
+            else:
+                w = torch.cat([a.q_proj.weight, a.k_proj.weight,
+                               a.v_proj.weight], dim=1).contiguous()
+                qo = a.q_proj.weight.shape[1]
+                ko = a.k_proj.weight.shape[1]
+                a.q_proj.weight.data = w[:, :qo]
+                a.k_proj.weight.data = w[:, qo:qo + ko]
+                a.v_proj.weight.data = w[:, qo + ko:]
+                self._qkv_fused.append(("bf16", w, None))
 
     def _rope_rows(self, x, pos):
         """x [B, 1, H, D] at per-row absolute positions pos [B] (long)."""
@@ -535,12 +556,20 @@ class LlamaModelRunner(GPTModelRunner):
         m = self.model
         a_pos = lens.long()                 # absolute position of new token
         x = m.llama.embed_tokens(toks)
+        qo = self.H * self.D
+        ko = self.HKV * self.D
         for li, layer in enumerate(m.llama.layers):
             h = layer.input_layernorm(x)
-            at = layer.self_attn
-            q = at.q_proj(h).reshape(B, 1, self.H, self.D)
-            k = at.k_proj(h).reshape(B, 1, self.HKV, self.D)
-            v = at.v_proj(h).reshape(B, 1, self.HKV, self.D)
+            kind, w, sc = self._qkv_fused[li]
+            h2 = h.reshape(B, -1)
+            if kind == "int8":
+                from .quantization import weight_only_linear
+                qkv = weight_only_linear(h2, w, sc)
+            else:
+                qkv = torch.matmul(h2, w)
+            q = qkv[:, :qo].reshape(B, 1, self.H, self.D)
+            k = qkv[:, qo:qo + ko].reshape(B, 1, self.HKV, self.D)
+            v = qkv[:, qo + ko:].reshape(B, 1, self.HKV, self.D)
             q = self._rope_rows(q, a_pos)
             k = self._rope_rows(k, a_pos)
             self.k[li][write_blk, write_off] = k[:, 0]
