@@ -39,7 +39,7 @@ def linear(x, weight, bias=None, name=None):
     # 51.3 vs 71.8 us at M32 K16384 N4096 -- tools/bench_decode_gemm.py);
     # lt already streams the wide-N shapes at 3.4-5.9 TB/s.
     if (not torch.is_grad_enabled() and x.dim() >= 2 and x.is_cuda
-            and x.dtype == torch.bfloat16):
+            and x.dtype == torch.bfloat16 and not hot._tracing()):
         mrows = x.numel() // x.shape[-1]
         k, n = weight.shape[-2], weight.shape[-1]
         if (mrows <= 32 and k >= 2 * n and k % 64 == 0 and n % 256 == 0
