@@ -745,6 +745,17 @@ Tensor amax_abs(const Tensor& x) {
   return out;
 }
 
+Tensor gemm_bf16_8p_t(const Tensor& a, const Tensor& bt) {
+  TORCH_CHECK(a.is_cuda() && a.dim() == 2 && bt.dim() == 2);
+  TORCH_CHECK(a.is_contiguous() && bt.is_contiguous());
+  int64_t m = a.size(0), k = a.size(1), n = bt.size(0);
+  TORCH_CHECK(bt.size(1) == k && m % 256 == 0 && n % 256 == 0 && k % 32 == 0);
+  auto c = torch::empty({m, n}, a.options());
+  pa::gemm_bf16_8p(a.const_data_ptr(), bt.const_data_ptr(),
+                   c.mutable_data_ptr(), m, n, k, k, k, n, cur_stream());
+  return c;
+}
+
 Tensor gemm_bf16_nt_batched_t(const Tensor& a, const Tensor& bt) {
   TORCH_CHECK(a.is_cuda() && a.dim() == 3 && bt.dim() == 3);
   TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
@@ -853,6 +864,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_bwd", &embedding_bwd);
   m.def("gemm_bf16", &gemm_bf16);
   m.def("gemm_bf16_nt_batched", &gemm_bf16_nt_batched_t);
+  m.def("gemm_bf16_8p", &gemm_bf16_8p_t);
   m.def("gemm_bf16_ex", &gemm_bf16_ex, py::arg("a"), py::arg("b"),
         py::arg("layout"), py::arg("epilogue") = 0,
         py::arg("bias") = c10::nullopt, py::arg("aux") = c10::nullopt,

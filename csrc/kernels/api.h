@@ -154,6 +154,9 @@ void decode_attention(const void* q, const void* kcache, const void* vcache,
 
 // ---- hand-written bf16 MFMA GEMM (C[m][n] = op(A) x op(B)) ----------------
 // b_is_nt: B passed as Bt[n][k] row-major (fast path); else B[k][n].
+void gemm_bf16_8p(const void* a, const void* b, void* c, int64_t m, int64_t n,
+                  int64_t k, int64_t lda, int64_t ldb, int64_t ldc,
+                  hipStream_t s);
 void gemm_bf16_nt_batched(const void* a, const void* b, void* c,
                           int64_t batch, int64_t m, int64_t n, int64_t k,
                           int64_t lda, int64_t ldb, int64_t ldc,
