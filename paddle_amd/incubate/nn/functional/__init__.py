@@ -477,3 +477,14 @@ def variable_length_memory_efficient_attention(query, key, value, seq_lens=None,
         s0, s1 = int(cu_q[i]), int(cu_q[i + 1])
         out[i, :, :lens[i]] = op[s0:s1].transpose(0, 1)
     return out
+
+
+def blha_get_max_len(seq_lens_encoder, seq_lens_decoder, batch_size):
+    """Max encoder/decoder sequence lengths for the current step, used
+    ahead of block_multihead_attention (reference:
+    incubate/nn/functional/blha_get_max_len.py:26 -> fused
+    BlhaGetMaxLen kernel; a two-max reduction here)."""
+    import torch
+    max_enc = seq_lens_encoder.max().reshape([1])
+    max_dec = seq_lens_decoder.max().reshape([1])
+    return max_enc, max_dec

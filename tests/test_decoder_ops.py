@@ -98,3 +98,12 @@ def test_variable_length_memory_efficient_attention_cpu():
                            v[i:i+1, :, :L].float(), 1 / math.sqrt(D), False)
         torch.testing.assert_close(out[i, :, :L], ref.squeeze(0), atol=1e-4,
                                    rtol=1e-4)
+
+
+def test_blha_get_max_len():
+    import torch
+    from paddle_amd.incubate.nn import functional as F
+    enc = torch.tensor([3, 9, 0], dtype=torch.int32)
+    dec = torch.tensor([5, 2, 7], dtype=torch.int32)
+    me, md = F.blha_get_max_len(enc, dec, torch.tensor(3))
+    assert int(me) == 9 and int(md) == 7 and me.shape == (1,)
