@@ -88,5 +88,28 @@ def gather(tensor, gather_list=None, dst=0, group=None, sync_op=True):
     return C.gather(tensor, gather_list, dst=dst, group=group)
 
 
+from . import checkpoint as io  # noqa: E402,F401  (distributed.io save/load tier)
+from .auto_parallel import DistAttr, Placement, Strategy  # noqa: E402,F401
+from .checkpoint import load_state_dict, save_state_dict  # noqa: E402,F401
+from .extras import (  # noqa: E402,F401
+    DistModel,
+    ParallelMode,
+    ReduceType,
+    ShardDataloader,
+    ShardingStage1,
+    ShardingStage2,
+    ShardingStage3,
+    get_backend,
+    gloo_barrier,
+    gloo_init_parallel_env,
+    gloo_release,
+    is_available,
+    scatter_object_list,
+    shard_dataloader,
+    shard_scaler,
+    split,
+    to_static,
+    unshard_dtensor,
+)
 from . import launch  # noqa: E402,F401
 from . import passes  # noqa: E402,F401

@@ -388,3 +388,37 @@ def test_merge_sharded_checkpoint(tmp_path):
     total = sum(v.numel() for v in masters)
     assert total >= 16 * 16 + 16          # weight + bias (padded)
     assert full["step"] == 1
+
+
+def test_distributed_longtail_api(tmp_path):
+    """split/ParallelMode/DistModel/to_static/unshard surface (reference
+    distributed/__init__.py exports)."""
+    import torch
+    import paddle_amd as paddle
+    d = paddle.distributed
+    assert d.ParallelMode.TENSOR_PARALLEL == 1
+    assert d.ReduceType.kRedSum == 0
+    # split at world 1 == plain ops
+    x = torch.randn(3, 8)
+    out = d.split(x, (8, 6), operation="linear", axis=1, num_partitions=1)
+    assert out.shape == (3, 6)
+    emb = d.split(torch.tensor([[1, 2]]), (16, 8), operation="embedding")
+    assert emb.shape == (1, 2, 8)
+    # DistModel one train step
+    net = torch.nn.Linear(4, 2)
+    opt = paddle.optimizer.SGD(learning_rate=0.1, parameters=net.parameters())
+    dm = d.to_static(net, loss=torch.nn.functional.mse_loss, optimizer=opt)
+    w0 = net.weight.detach().clone()
+    loss = dm(torch.randn(5, 4), torch.randn(5, 2))
+    assert loss.dim() == 0 and not torch.allclose(net.weight, w0)
+    dm.eval()
+    _ = dm(torch.randn(5, 4), torch.randn(5, 2))
+    # unshard on a plain tensor is identity
+    t = torch.randn(4)
+    assert d.unshard_dtensor(t) is t
+    # shard_dataloader wraps and iterates
+    dl = [[torch.ones(2, 3), torch.zeros(2)]]
+    from paddle_amd.distributed.auto_parallel import ProcessMesh
+    sd = d.shard_dataloader(dl, ProcessMesh([[0]] if False else [0]))
+    batches = list(sd)
+    assert len(batches) == 1 and batches[0][0].shape == (2, 3)
