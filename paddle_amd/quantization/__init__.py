@@ -39,24 +39,191 @@ def fp8_dequant(q, scale):
     return q.to(torch.float32) / scale
 
 
+class BaseObserver(__import__("torch").nn.Module):
+    """Activation-range observer (reference paddle/quantization/observer.py
+    family): tracks running abs-max during calibration forwards."""
+
+    def __init__(self, quant_bits=8):
+        super().__init__()
+        import torch
+        self.quant_bits = quant_bits
+        self.register_buffer("absmax", torch.zeros(()))
+
+    def forward(self, x):
+        import torch
+        with torch.no_grad():
+            m = x.detach().abs().amax().float()
+            if m > self.absmax:
+                self.absmax.fill_(m)
+        return x
+
+    def scale(self):
+        qmax = 2 ** (self.quant_bits - 1) - 1
+        return (self.absmax.clamp(min=1e-8) / qmax)
+
+
+class BaseQuanter(__import__("torch").nn.Module):
+    """Fake-quantizer (reference paddle/quantization/quanters/abs_max.py):
+    quantize-dequantize with a straight-through estimator so QAT
+    backprops through the rounding."""
+
+    def __init__(self, quant_bits=8):
+        super().__init__()
+        import torch
+        self.quant_bits = quant_bits
+        self.register_buffer("absmax", torch.zeros(()))
+
+    def forward(self, x):
+        import torch
+        qmax = 2 ** (self.quant_bits - 1) - 1
+        with torch.no_grad():
+            m = x.detach().abs().amax().float()
+            if self.training and m > self.absmax:
+                self.absmax.fill_(m)
+        s = self.absmax.clamp(min=1e-8) / qmax
+        q = (x / s).round().clamp(-qmax, qmax) * s
+        # straight-through: forward quantized, backward identity
+        return x + (q - x.to(q.dtype)).detach()
+
+
+def quanter(name):
+    """Reference paddle.quantization.quanter class decorator: registers a
+    quanter factory under `name`."""
+    def deco(cls):
+        _QUANTER_REGISTRY[name] = cls
+        return cls
+    return deco
+
+
+_QUANTER_REGISTRY = {"FakeQuanterWithAbsMax": BaseQuanter}
+
+
 class QuantConfig:
+    """reference paddle/quantization/config.py: which layers get which
+    activation/weight quanters."""
+
     def __init__(self, activation=None, weight=None):
         self.activation = activation
         self.weight = weight
+        self._types = None
 
-    def add_layer_config(self, layer, activation=None, weight=None):
-        pass
+    def add_type_config(self, layer_types, activation=None, weight=None):
+        self._types = tuple(layer_types) if isinstance(layer_types, (list, tuple)) \
+            else (layer_types,)
+        if activation is not None:
+            self.activation = activation
+        if weight is not None:
+            self.weight = weight
+
+    def _match(self, layer):
+        import torch
+        if self._types is not None:
+            return isinstance(layer, self._types)
+        w = getattr(layer, "weight", None)
+        return (type(layer).__name__ in ("Linear", "Conv2D", "Conv2d")
+                and isinstance(w, torch.Tensor))
+
+
+class _QuantedLayer(__import__("torch").nn.Module):
+    """Wrapper inserted by PTQ/QAT: observer/fake-quant on the input and
+    (for QAT) on the weight."""
+
+    def __init__(self, layer, act_q, weight_q=None):
+        super().__init__()
+        self.layer = layer
+        self.act_q = act_q
+        self.weight_q = weight_q
+
+    def forward(self, *args, **kwargs):
+        args = (self.act_q(args[0]),) + args[1:]
+        if self.weight_q is not None:
+            w = self.layer.weight
+            orig = w.data
+            self.layer.weight.data = self.weight_q(orig)
+            try:
+                return self.layer(*args, **kwargs)
+            finally:
+                self.layer.weight.data = orig
+        return self.layer(*args, **kwargs)
+
+
+def _wrap_layers(model, config, make_act, make_w):
+    n = 0
+    for mod in list(model.modules()):
+        for name, child in list(mod._modules.items()):
+            if child is None or isinstance(child, _QuantedLayer):
+                continue
+            if config._match(child):
+                mod._modules[name] = _QuantedLayer(child, make_act(),
+                                                   make_w() if make_w else None)
+                n += 1
+    return n
 
 
 class PTQ:
+    """Post-training quantization (reference paddle/quantization/ptq.py):
+    quantize() inserts observers, run calibration forwards, convert()
+    bakes int8 weights + the observed activation fake-quant."""
+
     def __init__(self, config: QuantConfig = None):
         self.config = config or QuantConfig()
 
     def quantize(self, model, inplace=False):
+        import copy
+        m = model if inplace else copy.deepcopy(model)
+        _wrap_layers(m, self.config, lambda: BaseObserver(), None)
+        return m
+
+    def convert(self, model, inplace=True):
+        import torch
+        qmax = 127
+        for mod in model.modules():
+            if isinstance(mod, _QuantedLayer):
+                if isinstance(mod.act_q, BaseObserver):
+                    fq = BaseQuanter()
+                    fq.absmax.copy_(mod.act_q.absmax)
+                    fq.eval()
+                    mod.act_q = fq
+                w = getattr(mod.layer, "weight", None)
+                if isinstance(w, torch.Tensor):
+                    with torch.no_grad():
+                        s = w.abs().amax().clamp(min=1e-8) / qmax
+                        w.copy_(((w / s).round().clamp(-qmax, qmax)) * s)
         return model
 
-    def convert(self, model, inplace=False):
+
+class QAT:
+    """Quantization-aware training (reference paddle/quantization/qat.py):
+    quantize() inserts straight-through fake-quanters on activations and
+    weights so training sees int8 rounding."""
+
+    def __init__(self, config: QuantConfig = None):
+        self.config = config or QuantConfig()
+
+    def quantize(self, model, inplace=False):
+        import copy
+        m = model if inplace else copy.deepcopy(model)
+        act_cls = self.config.activation or BaseQuanter
+        w_cls = self.config.weight or BaseQuanter
+        act_f = act_cls if callable(act_cls) else BaseQuanter
+        _wrap_layers(m, self.config, lambda: _mk(act_f), lambda: _mk(w_cls))
+        return m
+
+    def convert(self, model, inplace=True):
+        for mod in model.modules():
+            if isinstance(mod, _QuantedLayer):
+                mod.eval()
         return model
+
+
+def _mk(cls_or_inst):
+    import copy
+    if isinstance(cls_or_inst, type):
+        return cls_or_inst()
+    try:
+        return copy.deepcopy(cls_or_inst)
+    except Exception:
+        return BaseQuanter()
 
 
 # -- weight-only int8 (serving decode path) ----------------------------------
