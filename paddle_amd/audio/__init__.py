@@ -5,3 +5,5 @@ from . import backends  # noqa: F401
 from . import datasets  # noqa: F401
 from . import features  # noqa: F401
 from . import functional  # noqa: F401
+
+from .backends import info, load, save  # noqa: F401,E402

@@ -32,3 +32,39 @@ def load(filepath, frame_offset=0, num_frames=-1, normalize=True):
     if normalize:
         t = t / 32768.0
     return t, sr
+
+
+def info(filepath):
+    """Audio metadata (reference audio/backends wave backend)."""
+    import wave
+    from dataclasses import dataclass
+
+    @dataclass
+    class AudioInfo:
+        sample_rate: int
+        num_samples: int
+        num_channels: int
+        bits_per_sample: int
+        encoding: str = "PCM_S"
+
+    with wave.open(filepath, "rb") as w:
+        return AudioInfo(w.getframerate(), w.getnframes(), w.getnchannels(),
+                         w.getsampwidth() * 8)
+
+
+def save(filepath, src, sample_rate, channels_first=True,
+         encoding="PCM_S", bits_per_sample=16):
+    """Write a [C, N] (or [N, C]) float tensor as 16-bit PCM WAV."""
+    import wave
+
+    import numpy as np
+    import torch
+    t = src.detach().cpu()
+    if not channels_first:
+        t = t.t()
+    data = (t.clamp(-1, 1).numpy().T * 32767.0).astype(np.int16)
+    with wave.open(filepath, "wb") as w:
+        w.setnchannels(data.shape[1] if data.ndim > 1 else 1)
+        w.setsampwidth(2)
+        w.setframerate(int(sample_rate))
+        w.writeframes(data.tobytes())

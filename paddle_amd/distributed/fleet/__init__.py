@@ -195,3 +195,91 @@ from .recompute import recompute  # noqa: E402,F401
 
 from . import meta_parallel  # noqa: E402,F401
 from . import metrics  # noqa: E402,F401
+
+
+# -- role makers + Fleet class (reference fleet/base/role_maker.py:40, 548
+#    and fleet/fleet.py; the PS-specific SERVER/HETER roles are plumbing
+#    only -- this build is collective-mode) --------------------------------
+class Role:
+    WORKER = 1
+    SERVER = 2
+    HETER_WORKER = 3
+    ALL = 4
+    COORDINATOR = 5
+
+
+class RoleMakerBase:
+    def __init__(self, is_collective=True, **kwargs):
+        import os
+        self._is_collective = is_collective
+        self._rank = int(os.environ.get("RANK",
+                         os.environ.get("PADDLE_TRAINER_ID", "0")))
+        self._world = int(os.environ.get("WORLD_SIZE",
+                          os.environ.get("PADDLE_TRAINERS_NUM", "1")))
+
+    def _role(self):
+        return Role.WORKER
+
+    def worker_index(self):
+        return self._rank
+
+    def worker_num(self):
+        return self._world
+
+    def is_worker(self):
+        return True
+
+    def is_first_worker(self):
+        return self._rank == 0
+
+    def is_server(self):
+        return False
+
+
+class PaddleCloudRoleMaker(RoleMakerBase):
+    """Env-var driven role discovery (reference role_maker.py:548);
+    collective mode reads RANK/WORLD_SIZE (or PADDLE_TRAINER_*)."""
+
+
+class UserDefinedRoleMaker(RoleMakerBase):
+    """Explicit role assignment (reference role_maker.py UserDefined)."""
+
+    def __init__(self, is_collective=True, current_id=0, role=Role.WORKER,
+                 worker_num=1, **kwargs):
+        super().__init__(is_collective, **kwargs)
+        self._rank = current_id
+        self._world = worker_num
+        self._role_v = role
+
+    def _role(self):
+        return self._role_v
+
+
+class Fleet:
+    """The class behind the module-level fleet facade (reference
+    fleet/fleet.py Fleet); init() delegates to the module functions so
+    `fleet.Fleet().init(...)` and `fleet.init(...)` behave alike."""
+
+    def __init__(self):
+        self._role_maker = None
+
+    def init(self, role_maker=None, is_collective=True, strategy=None, log_level="INFO"):
+        self._role_maker = role_maker or PaddleCloudRoleMaker(is_collective)
+        init(is_collective=is_collective, strategy=strategy)
+        return self
+
+    def worker_index(self):
+        return (self._role_maker or PaddleCloudRoleMaker()).worker_index()
+
+    def worker_num(self):
+        return (self._role_maker or PaddleCloudRoleMaker()).worker_num()
+
+    def is_first_worker(self):
+        return self.worker_index() == 0
+
+    def __getattr__(self, name):
+        import sys
+        mod = sys.modules[__name__]
+        if hasattr(mod, name):
+            return getattr(mod, name)
+        raise AttributeError(name)

@@ -127,3 +127,16 @@ def load_profiler_result(path):
     import json
     with open(path) as f:
         return json.load(f)
+
+
+class SummaryView:
+    """reference profiler/profiler.py:55 SummaryView enum."""
+    DeviceView = 0
+    OverView = 1
+    ModelView = 2
+    DistributedView = 3
+    KernelView = 4
+    OperatorView = 5
+    MemoryView = 6
+    MemoryManipulationView = 7
+    UDFView = 8

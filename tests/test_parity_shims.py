@@ -203,3 +203,16 @@ def test_ptq_qat_quantization():
         opt.zero_grad(); loss.backward(); opt.step()
         losses.append(float(loss.detach()))
     assert losses[-1] < losses[0] * 0.8  # trains through the STE
+
+
+def test_audio_io_roundtrip(tmp_path):
+    import torch
+    import paddle_amd as paddle
+    x = torch.sin(torch.linspace(0, 100, 8000)).unsqueeze(0)
+    p = str(tmp_path / "t.wav")
+    paddle.audio.save(p, x, 8000)
+    meta = paddle.audio.info(p)
+    y, sr = paddle.audio.load(p)
+    assert sr == 8000 and meta.num_channels == 1
+    assert (y - x).abs().max() < 1e-3
+    assert paddle.profiler.SummaryView.KernelView == 4
