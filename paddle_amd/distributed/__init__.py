@@ -92,7 +92,12 @@ from . import checkpoint as io  # noqa: E402,F401  (distributed.io save/load tie
 from .auto_parallel import DistAttr, Placement, Strategy  # noqa: E402,F401
 from .checkpoint import load_state_dict, save_state_dict  # noqa: E402,F401
 from .extras import (  # noqa: E402,F401
+    CountFilterEntry,
     DistModel,
+    InMemoryDataset,
+    ProbabilityEntry,
+    QueueDataset,
+    ShowClickEntry,
     ParallelMode,
     ReduceType,
     ShardDataloader,

@@ -217,3 +217,27 @@ def to_static(layer, loader=None, loss=None, optimizer=None, strategy=None):
     """reference dist.to_static: returns a DistModel driving one
     train/eval/predict step per call."""
     return DistModel(layer, loader, loss, optimizer, strategy)
+
+
+# -- parameter-server data pipeline (reference entry_attr.py, fleet
+#    InMemoryDataset/QueueDataset): PS mode is out of scope per SURVEY
+#    §2.3 -- the names exist and raise with an explicit reason ----------
+def _ps_gate(name):
+    class _Gated:
+        def __init__(self, *a, **k):
+            raise NotImplementedError(
+                f"paddle.distributed.{name} belongs to parameter-server "
+                "mode, which is out of scope for this MI355X collective-"
+                "mode build (SURVEY.md §2.3); use collective data "
+                "loading (paddle.io.DataLoader + DistributedBatchSampler).")
+    _Gated.__name__ = name
+    return _Gated
+
+
+QueueDataset = _ps_gate("QueueDataset")
+InMemoryDataset = _ps_gate("InMemoryDataset")
+CountFilterEntry = _ps_gate("CountFilterEntry")
+ShowClickEntry = _ps_gate("ShowClickEntry")
+ProbabilityEntry = _ps_gate("ProbabilityEntry")
+MultiSlotDataGenerator = _ps_gate("MultiSlotDataGenerator")
+MultiSlotStringDataGenerator = _ps_gate("MultiSlotStringDataGenerator")

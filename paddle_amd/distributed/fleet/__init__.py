@@ -283,3 +283,9 @@ class Fleet:
         if hasattr(mod, name):
             return getattr(mod, name)
         raise AttributeError(name)
+
+
+from ..extras import (  # noqa: E402,F401  (PS-mode gates)
+    MultiSlotDataGenerator,
+    MultiSlotStringDataGenerator,
+)
