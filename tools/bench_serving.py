@@ -76,12 +76,12 @@ def main():
         run_one("gpt-350M" + tag_sfx, m, weight_only=args.weight_only)
     elif args.model == "gpt3-6.7b":
         m = build_gpt("gpt3-6.7b", max_seq_len=2048).to("cuda", torch.bfloat16)
-        run_one("gpt3-6.7B" + tag_sfx, m, n_req=32, max_batch=16,
+        run_one("gpt3-6.7B" + tag_sfx, m, n_req=64, max_batch=32,
                 num_blocks=8192, weight_only=args.weight_only)
     else:
         from paddle_amd.models.llama import build_llama
         m = build_llama("llama2-7b", max_seq_len=2048).to("cuda", torch.bfloat16)
-        run_one("llama2-7B" + tag_sfx, m, n_req=32, max_batch=16,
+        run_one("llama2-7B" + tag_sfx, m, n_req=64, max_batch=32,
                 num_blocks=8192, weight_only=args.weight_only,
                 runner_cls="llama")
 
