@@ -648,7 +648,9 @@ Tensor decode_gemm_int8(const Tensor& x, const Tensor& qw,
   TORCH_CHECK(qw.size(0) == k && m <= 32 && n % 256 == 0 && k % 64 == 0);
   TORCH_CHECK(scale.numel() == n);
   int64_t nblk = n / 256;
-  int64_t kmin = nblk <= 24 ? 128 : 256;
+  // fp32 partials cost ksplit*M*N*8 B of traffic -- cap harder (see the
+  // bf16 binding's note; the reduce was 13% of serving kernel time)
+  int64_t kmin = nblk <= 24 ? 256 : 512;
   int64_t ksplit = std::min<int64_t>(std::max<int64_t>(1, 1024 / nblk),
                                      std::max<int64_t>(1, k / kmin));
   int64_t mt = m <= 16 ? 16 : 32;
@@ -672,8 +674,12 @@ Tensor decode_gemm_mfma(const Tensor& x, const Tensor& w,
   // target ~1024 workgroups, but keep kchunk >= 256 (4 pipelined LDS
   // tiles per block; kchunk == 64 degenerates to an unpipelined tile)
   int64_t nblk = n / 256;
-  // small-N shapes need a denser grid: allow kchunk 128 to reach 2 wg/CU
-  int64_t kmin = nblk <= 24 ? 128 : 256;
+  // small-N shapes need a denser grid: allow kchunk 128 to reach 2 wg/CU.
+  // Cap ksplit harder than the grid target alone suggests: the fp32
+  // partial buffer costs ksplit*M*N*8 B of write+read, which at int8
+  // weights rivals the weight stream itself (the reduce kernel measured
+  // 13% of serving kernel time at the 1024-wg target).
+  int64_t kmin = nblk <= 24 ? 256 : 512;
   int64_t ksplit = std::min<int64_t>(std::max<int64_t>(1, 1024 / nblk),
                                      std::max<int64_t>(1, k / kmin));
   int64_t mt = m <= 16 ? 16 : 32;
