@@ -49,30 +49,46 @@ __global__ void decode_attn_kernel(const short* __restrict__ qg,
 #pragma unroll
   for (int i = 0; i < E; ++i) oacc[i] = 0.f;
 
-  // each wave processes 4 positions per step; 4 waves -> 16 pos/block/step
-  for (int pos0 = wid * 4; pos0 < S; pos0 += 16) {
+  // each wave processes 4 positions per step; 4 waves -> 16 pos/block/step.
+  // Software-pipelined: K/V for step t+1 load while step t computes --
+  // the serial position walk was latency-bound (back-to-back dependent
+  // 16 B loads per step).
+  auto load_kv = [&](int pos0, bool& ok, shortx8& kv8, shortx8& vv8) {
     int pos = pos0 + lp;
-    bool ok = pos < S;
+    ok = pos < S;
     int phys = 0, off = 0;
     if (ok) {
       phys = block_table[(long long)b * max_blocks + pos / bs];
       off = pos % bs;
     }
-    const short* kp = kcache + (long long)phys * blk_str + (long long)off * pos_str
-                     + (long long)hkv * head_str;
-    const short* vp = vcache + (long long)phys * blk_str + (long long)off * pos_str
-                     + (long long)hkv * head_str;
-    float sc = 0.f;
+    const long long base = (long long)phys * blk_str +
+                           (long long)off * pos_str +
+                           (long long)hkv * head_str + l16 * E;
     if (ok) {
       if (E == 8) {
-        shortx8 kv8 = *reinterpret_cast<const shortx8*>(kp + l16 * E);
-#pragma unroll
-        for (int i = 0; i < E; ++i) sc += qv[i] * bf2f(kv8[i]);
+        kv8 = *reinterpret_cast<const shortx8*>(kcache + base);
+        vv8 = *reinterpret_cast<const shortx8*>(vcache + base);
       } else {
-        shortx4 kv4 = *reinterpret_cast<const shortx4*>(kp + l16 * E);
+        shortx4 k4 = *reinterpret_cast<const shortx4*>(kcache + base);
+        shortx4 v4 = *reinterpret_cast<const shortx4*>(vcache + base);
 #pragma unroll
-        for (int i = 0; i < E; ++i) sc += qv[i] * bf2f(kv4[i & 3]);
+        for (int i = 0; i < 4; ++i) { kv8[i] = k4[i]; vv8[i] = v4[i]; }
       }
+    }
+  };
+
+  bool okc = false, okn = false;
+  shortx8 kc8, vc8, kn8, vn8;
+  const int p_start = wid * 4;
+  load_kv(p_start, okc, kc8, vc8);
+  for (int pos0 = p_start; pos0 < S; pos0 += 16) {
+    if (pos0 + 16 < S) load_kv(pos0 + 16, okn, kn8, vn8);
+    else okn = false;
+    bool ok = okc;
+    float sc = 0.f;
+    if (ok) {
+#pragma unroll
+      for (int i = 0; i < E; ++i) sc += qv[i] * bf2f(kc8[i]);
     }
     // reduce over the 16-lane group
     sc = group16_reduce(sc, SumOp());
@@ -83,22 +99,12 @@ __global__ void decode_attn_kernel(const short* __restrict__ qg,
     float p = ok ? __expf(sc - m_new) : 0.f;
     l = l * corr + p;
     m = m_new;
-    float vvf[E];
 #pragma unroll
-    for (int i = 0; i < E; ++i) vvf[i] = 0.f;
-    if (ok) {
-      if (E == 8) {
-        shortx8 vv8 = *reinterpret_cast<const shortx8*>(vp + l16 * E);
-#pragma unroll
-        for (int i = 0; i < E; ++i) vvf[i] = bf2f(vv8[i]);
-      } else {
-        shortx4 vv4 = *reinterpret_cast<const shortx4*>(vp + l16 * E);
-#pragma unroll
-        for (int i = 0; i < E; ++i) vvf[i] = bf2f(vv4[i & 3]);
-      }
+    for (int i = 0; i < E; ++i) {
+      float vvf = ok ? bf2f(vc8[i]) : 0.f;
+      oacc[i] = oacc[i] * corr + p * vvf;
     }
-#pragma unroll
-    for (int i = 0; i < E; ++i) oacc[i] = oacc[i] * corr + p * vvf[i];
+    okc = okn; kc8 = kn8; vc8 = vn8;
   }
 
   // ---- merge the 16 position-slot partials (4 waves x 4 slots) ----------

@@ -216,8 +216,14 @@ class GroupShardedStage3(torch.nn.Module):
         self.group = group
         self.world = get_world_size(group)
         self.rank = get_rank(group) if group is None else group.rank
-        dev = device or (torch.device("cuda", torch.cuda.current_device())
-                         if torch.cuda.is_available() else torch.device("cpu"))
+        if device is None:
+            # follow the MODEL, not CUDA availability: a CPU-built layer
+            # on a GPU box must shard on CPU (gloo tests run everywhere)
+            p = next(iter(layer.parameters()), None)   # paddle .parameters() is a list
+            device = p.device if p is not None else (
+                torch.device("cuda", torch.cuda.current_device())
+                if torch.cuda.is_available() else torch.device("cpu"))
+        dev = torch.device(device)
         self.device = dev
         self.sync_comm = sync_comm
         self._units: List[_Unit] = []

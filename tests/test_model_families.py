@@ -143,7 +143,7 @@ def test_llama_paged_generation_cpu():
     cfg = LlamaConfig(vocab_size=512, hidden_size=128, num_layers=2,
                       num_heads=8, num_kv_heads=4, intermediate_size=256,
                       max_seq_len=128)
-    m = LlamaForCausalLM(cfg).eval()
+    m = LlamaForCausalLM(cfg).to("cpu").float().eval()
     ids = torch.randint(0, 512, (2, 12))
     with torch.no_grad():
         gen = generate_llama(m, ids, max_new_tokens=4)

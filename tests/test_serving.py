@@ -126,7 +126,7 @@ def test_weight_only_quantized_model_cpu():
     from paddle_amd import quantization as Q
     from paddle_amd.models import build_gpt
     torch.manual_seed(0)
-    m = build_gpt("gpt3-tiny", max_seq_len=128).eval()
+    m = build_gpt("gpt3-tiny", max_seq_len=128).to("cpu").float().eval()
     ids = torch.randint(0, 1000, (2, 16))
     with torch.no_grad():
         ref = m(ids)
