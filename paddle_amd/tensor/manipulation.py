@@ -8,12 +8,25 @@ from __future__ import annotations
 import torch
 
 
+def _resolve_shape(x, shape):
+    # paddle reshape semantics (manipulation.py reshape docs): 0 copies
+    # the corresponding INPUT dim, -1 infers; torch has no 0 rule
+    out = []
+    for i, s in enumerate(shape):
+        s = int(s.item()) if isinstance(s, torch.Tensor) else int(s)
+        out.append(x.shape[i] if s == 0 else s)
+    return out
+
+
 def reshape(x, shape, name=None):
-    shape = [int(s) if not isinstance(s, torch.Tensor) else int(s.item()) for s in shape]
-    return torch.reshape(x, shape)
+    """paddle.reshape: 0 copies the input dim, -1 infers.  (The
+    Tensor.reshape METHOD keeps torch semantics -- patching it globally
+    would break torch-internal empty-tensor reshapes.)"""
+    return torch.reshape(x, _resolve_shape(x, shape))
 
 
 def reshape_(x, shape, name=None):
+    shape = _resolve_shape(x, shape)
     return x.reshape_(*shape) if hasattr(x, "reshape_") else x.view(shape)
 
 

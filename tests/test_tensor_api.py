@@ -95,3 +95,13 @@ def test_cast_astype():
     x = paddle.ones([2], dtype="float32")
     assert x.astype("int64").dtype == torch.int64
     assert paddle.cast(x, "float16").dtype == torch.float16
+
+
+def test_reshape_zero_copies_dim():
+    """paddle reshape semantics: 0 copies the corresponding input dim
+    (reference tensor/manipulation.py reshape); -1 still infers."""
+    import paddle_amd as paddle
+    x = paddle.ones([2, 3, 4])
+    assert tuple(paddle.reshape(x, [0, 3, 4]).shape) == (2, 3, 4)
+    assert tuple(paddle.reshape(x, [0, -1]).shape) == (2, 12)
+    assert tuple(paddle.reshape(x, [0, 0, -1]).shape) == (2, 3, 4)
