@@ -26,17 +26,21 @@ for _ in range(2):
     step()
 torch.cuda.synchronize()
 from torch.profiler import profile, ProfilerActivity
+import torch._C._profiler as _p
 with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
-             with_stack=True) as prof:
+             with_stack=True,
+             experimental_config=_p._ExperimentalConfig(verbose=True)) as prof:
     step()
     torch.cuda.synchronize()
 evs = prof.key_averages(group_by_stack_n=6)
 rows = [e for e in evs
         if e.device_time_total > 0 and
-        ("copy" in e.key.lower() or "Memcpy" in e.key or "to_" in e.key.lower()
+        ("copy" in e.key.lower() or "Memcpy" in e.key or "to" == e.key[-2:].lower()
+         or "_to_copy" in e.key or "cast" in e.key.lower()
          or "add" in e.key.lower() or "fill" in e.key.lower())]
 rows.sort(key=lambda e: -e.device_time_total)
 for e in rows[:12]:
     print(f"{e.device_time_total/1e3:8.2f}ms {e.count:5d}x  {e.key[:60]}")
-    for ln in (e.stack or [])[:4]:
-        print("     ", ln[-110:])
+    for ln in (e.stack or [])[:6]:
+        if "paddle_amd" in ln or "bench" in ln:
+            print("     ", ln[-110:])
