@@ -47,8 +47,14 @@ class ElasticManager:
         self._threads.append(t)
 
     def _heartbeat(self):
-        with open(self._my_key(), "w") as f:
+        # atomic lease refresh: open("w") truncates in place, so a
+        # concurrent hosts() read can see a partial file and silently
+        # drop the host (observed as membership flaps under load)
+        key = self._my_key()
+        tmp = key + f".tmp{os.getpid()}"
+        with open(tmp, "w") as f:
             json.dump({"host": self.host, "ts": time.time()}, f)
+        os.replace(tmp, key)
 
     def _hb_loop(self):
         while not self._stop.wait(self.hb_interval):

@@ -43,10 +43,13 @@ def test_watchdog_completed_task_no_fire():
 def test_elastic_membership(tmp_path):
     # generous margins: a loaded CI box can stall a heartbeat thread for
     # hundreds of ms, which must not look like a lost lease
+    # ttl must survive multi-second scheduler stalls on a saturated box
+    # (exit() removes the lease explicitly, so detection never waits on
+    # the ttl -- it only guards against a's own lease lapsing)
     m1 = ElasticManager(job_id="t", host="a:1", store_dir=str(tmp_path),
-                        heartbeat_interval=0.2, lease_ttl=3.0)
+                        heartbeat_interval=0.2, lease_ttl=15.0)
     m2 = ElasticManager(job_id="t", host="b:2", store_dir=str(tmp_path),
-                        heartbeat_interval=0.2, lease_ttl=3.0)
+                        heartbeat_interval=0.2, lease_ttl=15.0)
     m1.register()
     m2.register()
     time.sleep(0.6)
