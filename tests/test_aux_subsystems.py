@@ -79,8 +79,9 @@ def test_launch_cli_single_proc(tmp_path):
     env = dict(os.environ)
     env["CUDA_VISIBLE_DEVICES"] = ""
     env["PYTHONPATH"] = REPO
+    from dist_util import free_port
     r = subprocess.run([sys.executable, "-m", "paddle_amd.distributed.launch",
-                        "--nproc_per_node", "2", "--master_port", "29777",
+                        "--nproc_per_node", "2", "--master_port", str(free_port()),
                         "--log_dir", str(tmp_path / "logs"), str(script)],
                        env=env, capture_output=True, text=True, timeout=180,
                        cwd=str(tmp_path))
