@@ -84,3 +84,62 @@ def test_transpose_perm(data):
     x = paddle.randn(shape)
     np.testing.assert_array_equal(paddle.transpose(x, perm).numpy(),
                                   np.transpose(x.numpy(), perm))
+
+
+@settings(max_examples=25, deadline=None)
+@given(data=st.data())
+def test_matmul_broadcast_semantics(data):
+    b = data.draw(st.integers(1, 3))
+    m = data.draw(st.integers(1, 5))
+    k = data.draw(st.integers(1, 5))
+    n = data.draw(st.integers(1, 5))
+    x = paddle.randn([b, m, k])
+    y = paddle.randn([k, n])
+    out = paddle.matmul(x, y)
+    np.testing.assert_allclose(out.numpy(), x.numpy() @ y.numpy(),
+                               rtol=1e-4, atol=1e-5)
+    xt = paddle.randn([b, k, m])
+    out2 = paddle.matmul(xt, y, transpose_x=True)
+    np.testing.assert_allclose(out2.numpy(),
+                               np.swapaxes(xt.numpy(), -1, -2) @ y.numpy(),
+                               rtol=1e-4, atol=1e-5)
+
+
+@settings(max_examples=25, deadline=None)
+@given(data=st.data())
+def test_einsum_vs_numpy(data):
+    i = data.draw(st.integers(1, 4))
+    j = data.draw(st.integers(1, 4))
+    k = data.draw(st.integers(1, 4))
+    a = paddle.randn([i, j])
+    b = paddle.randn([j, k])
+    np.testing.assert_allclose(paddle.einsum("ij,jk->ik", a, b).numpy(),
+                               np.einsum("ij,jk->ik", a.numpy(), b.numpy()),
+                               rtol=1e-4, atol=1e-5)
+    np.testing.assert_allclose(paddle.einsum("ij->j", a).numpy(),
+                               a.numpy().sum(0), rtol=1e-4, atol=1e-5)
+
+
+@settings(max_examples=25, deadline=None)
+@given(shape=st.lists(st.integers(1, 5), min_size=2, max_size=2))
+def test_linalg_matches_numpy(shape):
+    n = max(shape)
+    x = paddle.randn([n, n]) + paddle.eye(n) * n   # well-conditioned
+    np.testing.assert_allclose(
+        paddle.linalg.inv(x).numpy() @ x.numpy(), np.eye(n),
+        rtol=1e-3, atol=1e-3)
+    s = paddle.linalg.svd(x)[1]
+    assert (s.numpy() >= -1e-6).all()
+
+
+@settings(max_examples=20, deadline=None)
+@given(data=st.data())
+def test_indexing_semantics(data):
+    n = data.draw(st.integers(2, 8))
+    x = paddle.arange(n * 3, dtype="float32").reshape([n, 3])
+    idx = data.draw(st.lists(st.integers(0, n - 1), min_size=1, max_size=4))
+    it = paddle.to_tensor(idx, dtype="int64")
+    np.testing.assert_array_equal(paddle.gather(x, it, axis=0).numpy(),
+                                  x.numpy()[idx])
+    np.testing.assert_array_equal(paddle.index_select(x, it, axis=0).numpy(),
+                                  x.numpy()[idx])
