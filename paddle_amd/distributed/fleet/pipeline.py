@@ -123,6 +123,8 @@ class _P2P:
         self.sent_meta = False
         self.recv_shape = None
         self.recv_dtype = None
+        self._sent_meta_to = set()   # peer-aware path (ZB-V placement)
+        self._recv_meta_from = {}
 
     def _send_meta(self, t, dst):
         hdr = torch.tensor([t.dim(), *t.shape, _DTYPE_IDS.index(t.dtype)],
@@ -142,27 +144,47 @@ class _P2P:
         dtype = _DTYPE_IDS[vals[1 + ndim]]
         return shape, dtype
 
-    def send_forward(self, t):
-        dst = self.hcg.get_p2p_next_rank()
-        if not self.sent_meta:
+    def send_forward(self, t, dst=None):
+        explicit = dst is not None
+        if dst is None:
+            dst = self.hcg.get_p2p_next_rank()
+        if explicit:
+            # ZB-V placement: forward peers vary by stage (up-leg -> next,
+            # down-leg -> prev); meta handshake cached per peer
+            if dst not in self._sent_meta_to:
+                self._send_meta(t, dst)
+                self._sent_meta_to.add(dst)
+        elif not self.sent_meta:
             self._send_meta(t, dst)
             self.sent_meta = True
         C.send(t.contiguous(), dst=dst, group=self.group)
 
-    def recv_forward(self, device):
-        src = self.hcg.get_p2p_prev_rank()
-        if self.recv_shape is None:
-            self.recv_shape, self.recv_dtype = self._recv_meta(src, device)
-        t = torch.zeros(self.recv_shape, dtype=self.recv_dtype, device=device)
+    def recv_forward(self, device, src=None):
+        explicit = src is not None
+        if src is None:
+            src = self.hcg.get_p2p_prev_rank()
+        if explicit:
+            if src not in self._recv_meta_from:
+                self._recv_meta_from[src] = self._recv_meta(src, device)
+            shape, dtype = self._recv_meta_from[src]
+        else:
+            if self.recv_shape is None:
+                self.recv_shape, self.recv_dtype = self._recv_meta(src, device)
+            shape, dtype = self.recv_shape, self.recv_dtype
+        t = torch.zeros(shape, dtype=dtype, device=device)
         C.recv(t, src=src, group=self.group)
         return t
 
-    def send_backward(self, g):
-        C.send(g.contiguous(), dst=self.hcg.get_p2p_prev_rank(), group=self.group)
+    def send_backward(self, g, dst=None):
+        if dst is None:
+            dst = self.hcg.get_p2p_prev_rank()
+        C.send(g.contiguous(), dst=dst, group=self.group)
 
-    def recv_backward(self, like):
+    def recv_backward(self, like, src=None):
+        if src is None:
+            src = self.hcg.get_p2p_next_rank()
         g = torch.zeros_like(like)
-        C.recv(g, src=self.hcg.get_p2p_next_rank(), group=self.group)
+        C.recv(g, src=src, group=self.group)
         return g
 
     # combined pairs (deadlock-free: post isend+irecv, then wait both --
@@ -344,7 +366,8 @@ class VirtualPipelineLayer(torch.nn.Module):
     """
 
     def __init__(self, layers, topology=None, loss_fn=None,
-                 num_virtual_pipeline_stages=2, recompute_interval=0, hcg=None):
+                 num_virtual_pipeline_stages=2, recompute_interval=0, hcg=None,
+                 placement="interleaved"):
         super().__init__()
         from . import get_hybrid_communicate_group
         self._hcg = hcg or get_hybrid_communicate_group()
@@ -357,7 +380,18 @@ class VirtualPipelineLayer(torch.nn.Module):
         n = len(self.descs)
         per = [n // total + (1 if i < n % total else 0) for i in range(total)]
         starts = [sum(per[:i]) for i in range(total + 1)]
-        self.my_stages = [g for g in range(total) if g % self.pp == self.rank]
+        # stage->rank placement: "interleaved" (g % pp) or "zbv" (ZB-V
+        # paper V-shape, v == 2: rank r owns chunks r and 2*pp-1-r, so
+        # the last stage's B starts on the rank that also holds stage 0
+        # -- the geometry that removes the H1 schedule's tail bubble)
+        if placement == "zbv":
+            assert self.v == 2, "ZB-V placement is defined for v == 2"
+            self.owner = [g if g < self.pp else 2 * self.pp - 1 - g
+                          for g in range(total)]
+        else:
+            self.owner = [g % self.pp for g in range(total)]
+        self.placement = placement
+        self.my_stages = [g for g in range(total) if self.owner[g] == self.rank]
         self.total_stages = total
         self.chunks = torch.nn.ModuleList()
         self._chunk_funcs = {}
@@ -414,6 +448,12 @@ class InterleavedPipelineParallel(torch.nn.Module):
         ys = y.chunk(n) if y is not None else [None] * n
         return list(zip(xs, ys))
 
+    def _hcg_pp_ranks(self):
+        ranks = getattr(self._hcg, "_pp_group", None)
+        if ranks is None:
+            ranks = list(range(self.pp))
+        return ranks
+
     def train_batch(self, data, optimizer, lr_scheduler=None, scaler=None):
         micro = self._split_micro(data)
         n_micro = len(micro)
@@ -425,32 +465,51 @@ class InterleavedPipelineParallel(torch.nn.Module):
         # saved[(mb, g)] = (input, output) for the backward relay
         saved = {}
 
+        owner = getattr(L, "owner", [g % self.pp for g in range(L.total_stages)])
+        zbv = getattr(L, "placement", "interleaved") != "interleaved"
+        pp_ranks = (self._hcg_pp_ranks() if zbv else None)
+
+        def peer(stage):
+            return pp_ranks[owner[stage]] if zbv else None
+
+        local_fwd = {}
         for i in range(n_micro):
             for g in L.my_stages:
                 if g == 0:
                     inp = micro[i][0]
+                elif owner[g - 1] == self.rank:
+                    inp = local_fwd.pop((i, g)).detach()   # V-turn: same rank
+                    inp.requires_grad_(True)
                 else:
-                    inp = self.p2p.recv_forward(dev)
+                    inp = self.p2p.recv_forward(dev, src=peer(g - 1))
                     inp.requires_grad_(True)
                 out = L.run_chunk(g, inp)
                 if g == last_g:
                     loss = L.loss_fn(out, micro[i][1]) / n_micro
                     losses.append(loss.detach())
                     out = loss
+                elif owner[g + 1] == self.rank:
+                    local_fwd[(i, g + 1)] = out
                 else:
-                    self.p2p.send_forward(out)
+                    self.p2p.send_forward(out, dst=peer(g + 1))
                 saved[(i, g)] = (inp, out)
 
+        local_bwd = {}
         for i in reversed(range(n_micro)):
             for g in reversed(L.my_stages):
                 inp, out = saved.pop((i, g))
                 if g == last_g:
                     out.backward()
+                elif owner[g + 1] == self.rank:
+                    out.backward(gradient=local_bwd.pop((i, g)))
                 else:
-                    grad_out = self.p2p.recv_backward(out)
+                    grad_out = self.p2p.recv_backward(out, src=peer(g + 1))
                     out.backward(gradient=grad_out)
                 if g != 0:
-                    self.p2p.send_backward(inp.grad)
+                    if owner[g - 1] == self.rank:
+                        local_bwd[(i, g - 1)] = inp.grad
+                    else:
+                        self.p2p.send_backward(inp.grad, dst=peer(g - 1))
 
         if optimizer is not None:
             if scaler is not None:
@@ -544,9 +603,9 @@ class _ZBP2P:
         self._store = store
         self._per_recv = per_recv
 
-    def recv_backward(self, like):
+    def recv_backward(self, like, src=None):
         _flush_w(self._store, limit=self._per_recv)
-        return self._inner.recv_backward(like)
+        return self._inner.recv_backward(like, src=src)
 
     def send_backward_recv_forward(self, grad_in, device):
         _flush_w(self._store, limit=self._per_recv)
