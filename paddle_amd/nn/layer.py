@@ -27,9 +27,18 @@ class Layer(torch.nn.Module):
                          default_initializer=None):
         dtype = framework.convert_dtype(dtype) if dtype is not None else self._dtype
         t = torch.empty(list(shape), dtype=dtype, device=framework.get_default_device())
-        init = default_initializer
-        if init is None and attr is not None and getattr(attr, "initializer", None) is not None:
+        init = None
+        if attr is not None and getattr(attr, "initializer", None) is not None:
             init = attr.initializer
+        if init is None:
+            # set_global_initializer overrides layer defaults but not an
+            # explicit ParamAttr initializer (reference initializer.py
+            # set_global_initializer semantics)
+            from . import initializer as _I
+            init = (_I._GLOBAL_BIAS_INIT if is_bias
+                    else _I._GLOBAL_WEIGHT_INIT)
+        if init is None:
+            init = default_initializer
         if init is None:
             init = Constant(0.0) if is_bias else XavierNormal()
         _apply_initializer(init, t)

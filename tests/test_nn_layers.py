@@ -105,3 +105,24 @@ def test_clip_grad_by_global_norm():
     opt._clip_grads()
     total = sum(float(p.grad.square().sum()) for p in m.parameters())
     assert total ** 0.5 <= 0.11
+
+
+def test_set_global_initializer():
+    """reference nn/initializer set_global_initializer: overrides layer
+    defaults, loses to an explicit ParamAttr initializer, cleared by
+    passing None."""
+    import paddle_amd as paddle
+    from paddle_amd.nn import initializer as I
+    from paddle_amd.param_attr import ParamAttr
+    I.set_global_initializer(I.Constant(0.5), I.Constant(0.25))
+    try:
+        l = paddle.nn.Linear(3, 3)
+        assert abs(l.weight[0, 0].item() - 0.5) < 1e-6
+        assert abs(l.bias[0].item() - 0.25) < 1e-6
+        l2 = paddle.nn.Linear(3, 3,
+                              weight_attr=ParamAttr(initializer=I.Constant(2.0)))
+        assert abs(l2.weight[0, 0].item() - 2.0) < 1e-6
+    finally:
+        I.set_global_initializer(None)
+    l3 = paddle.nn.Linear(3, 3)
+    assert l3.weight.std() > 1e-4      # back to the default distribution
