@@ -94,12 +94,12 @@ def _apply_initializer(init, t: torch.Tensor):
         fi, _ = _fans(t)
         fi = init.fan_in if init.fan_in is not None else fi
         gain = math.sqrt(2.0 / (1 + init.negative_slope ** 2))
-        if isinstance(init, KaimingNormal):
-            std = gain / math.sqrt(fi)
-            t.copy_(torch.randn_like(t, dtype=torch.float32).mul_(std).to(t.dtype))
-        else:
+        if isinstance(init, KaimingUniform):   # subclass: must test FIRST
             limit = gain * math.sqrt(3.0 / fi)
             t.copy_((torch.rand_like(t, dtype=torch.float32) * 2 * limit - limit).to(t.dtype))
+        else:
+            std = gain / math.sqrt(fi)
+            t.copy_(torch.randn_like(t, dtype=torch.float32).mul_(std).to(t.dtype))
     elif isinstance(init, Assign):
         v = init.value
         if not isinstance(v, torch.Tensor):

@@ -126,3 +126,18 @@ def test_set_global_initializer():
         I.set_global_initializer(None)
     l3 = paddle.nn.Linear(3, 3)
     assert l3.weight.std() > 1e-4      # back to the default distribution
+
+
+def test_kaiming_uniform_is_uniform():
+    """KaimingUniform subclasses KaimingNormal; the dispatch must not
+    send it down the normal branch (bounded support distinguishes)."""
+    import math
+    import torch
+    from paddle_amd.nn.initializer import (KaimingNormal, KaimingUniform,
+                                           _apply_initializer)
+    t = torch.empty(256, 256)
+    _apply_initializer(KaimingUniform(), t)
+    limit = math.sqrt(2.0) * math.sqrt(3.0 / 256)
+    assert t.abs().max() <= limit + 1e-6
+    _apply_initializer(KaimingNormal(), t)
+    assert t.abs().max() > limit
