@@ -220,13 +220,42 @@ def kl_div(input, label, reduction="mean", name=None):
 
 
 # -- conv / pool ------------------------------------------------------------
+def _conv_padding(padding, x, weight, stride, dilation, nd):
+    """paddle accepts padding "SAME"/"VALID" (any case, stride>1 ok);
+    torch's "same" string rejects stride>1, so SAME computes explicit
+    asymmetric padding (pads the high side like TF/paddle)."""
+    if not isinstance(padding, str):
+        return x, padding
+    p = padding.lower()
+    if p == "valid":
+        return x, 0
+    assert p == "same", padding
+    strides = [stride] * nd if isinstance(stride, int) else list(stride)
+    dils = [dilation] * nd if isinstance(dilation, int) else list(dilation)
+    pads = []          # [lo, hi] per spatial dim, torch F.pad order reversed
+    for i in range(nd):
+        size = x.shape[2 + i]
+        k = (weight.shape[2 + i] - 1) * dils[i] + 1
+        out = -(-size // strides[i])
+        total = max(0, (out - 1) * strides[i] + k - size)
+        pads.append((total // 2, total - total // 2))
+    if all(lo == hi for lo, hi in pads):
+        return x, [lo for lo, _ in pads]
+    flat = []
+    for lo, hi in reversed(pads):
+        flat += [lo, hi]
+    return TF.pad(x, flat), 0
+
+
 def conv2d(x, weight, bias=None, stride=1, padding=0, dilation=1, groups=1,
            data_format="NCHW", name=None):
+    x, padding = _conv_padding(padding, x, weight, stride, dilation, 2)
     return TF.conv2d(x, weight, bias, stride, padding, dilation, groups)
 
 
 def conv1d(x, weight, bias=None, stride=1, padding=0, dilation=1, groups=1,
            data_format="NCL", name=None):
+    x, padding = _conv_padding(padding, x, weight, stride, dilation, 1)
     return TF.conv1d(x, weight, bias, stride, padding, dilation, groups)
 
 

@@ -141,3 +141,20 @@ def test_kaiming_uniform_is_uniform():
     assert t.abs().max() <= limit + 1e-6
     _apply_initializer(KaimingNormal(), t)
     assert t.abs().max() > limit
+
+
+def test_conv_string_padding():
+    """paddle padding="SAME"/"VALID" (uppercase, stride>1 allowed --
+    torch's string form rejects both)."""
+    import torch
+    import torch.nn.functional as TF
+    import paddle_amd as paddle
+    from paddle_amd import nn
+    x = torch.randn(2, 3, 16, 16)
+    c = nn.Conv2D(3, 8, 3, padding="SAME")
+    assert tuple(c(x).shape) == (2, 8, 16, 16)
+    assert tuple(nn.Conv2D(3, 8, 3, padding="VALID")(x).shape) == (2, 8, 14, 14)
+    assert tuple(nn.Conv2D(3, 8, 3, stride=2, padding="SAME")(x).shape) == (2, 8, 8, 8)
+    ref = TF.conv2d(x, c.weight, c.bias, 1, "same")
+    out = paddle.nn.functional.conv2d(x, c.weight, c.bias, 1, "SAME")
+    assert torch.allclose(out, ref, atol=1e-5)
