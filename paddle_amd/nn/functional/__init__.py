@@ -264,8 +264,36 @@ def conv2d_transpose(x, weight, bias=None, stride=1, padding=0, output_padding=0
     return TF.conv_transpose2d(x, weight, bias, stride, padding, output_padding, groups, dilation)
 
 
+def _pool_padding(padding, x, kernel_size, stride, nd, pad_value=0.0):
+    """Pools also accept paddle's "SAME"/"VALID" strings (pad_value
+    -inf for max pooling so padded cells never win)."""
+    if not isinstance(padding, str):
+        return x, padding
+    p = padding.lower()
+    if p == "valid":
+        return x, 0
+    assert p == "same", padding
+    ks = [kernel_size] * nd if isinstance(kernel_size, int) else list(kernel_size)
+    stride = stride or kernel_size
+    strides = [stride] * nd if isinstance(stride, int) else list(stride)
+    flat = []
+    sym = []
+    for i in range(nd):
+        size = x.shape[2 + i]
+        out = -(-size // strides[i])
+        total = max(0, (out - 1) * strides[i] + ks[i] - size)
+        sym.append((total // 2, total - total // 2))
+    if all(lo == hi for lo, hi in sym):
+        return x, [lo for lo, _ in sym]
+    for lo, hi in reversed(sym):
+        flat += [lo, hi]
+    return TF.pad(x, flat, value=pad_value), 0
+
+
 def max_pool2d(x, kernel_size, stride=None, padding=0, return_mask=False,
                ceil_mode=False, data_format="NCHW", name=None):
+    x, padding = _pool_padding(padding, x, kernel_size, stride, 2,
+                               pad_value=float("-inf"))
     out = TF.max_pool2d(x, kernel_size, stride, padding, ceil_mode=ceil_mode,
                         return_indices=return_mask)
     return out
@@ -273,6 +301,7 @@ def max_pool2d(x, kernel_size, stride=None, padding=0, return_mask=False,
 
 def avg_pool2d(x, kernel_size, stride=None, padding=0, ceil_mode=False,
                exclusive=True, divisor_override=None, data_format="NCHW", name=None):
+    x, padding = _pool_padding(padding, x, kernel_size, stride, 2)
     return TF.avg_pool2d(x, kernel_size, stride, padding, ceil_mode=ceil_mode,
                          count_include_pad=not exclusive, divisor_override=divisor_override)
 
