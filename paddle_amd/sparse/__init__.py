@@ -57,15 +57,120 @@ def to_sparse_csr(x):
     return x.to_sparse_csr()
 
 
+def _values_op(x, fn):
+    if getattr(x, "is_sparse", False):
+        xc = x.coalesce()
+        return torch.sparse_coo_tensor(xc.indices(), fn(xc.values()), x.shape)
+    return fn(x)
+
+
 class nn:
-    """paddle.sparse.nn namespace placeholder (conv3d etc. are later work)."""
+    """paddle.sparse.nn (reference sparse/nn/__init__.py: activations over
+    sparse values; Conv/SubmConv/MaxPool3D run via a dense round-trip --
+    correct semantics, sized for the moderate sparse workloads the CPU
+    suite exercises; a gather-scatter sparse conv kernel is future work).
+    Layouts: paddle sparse conv is channels-last ([N, D, H, W, C])."""
 
     class ReLU(torch.nn.Module):
         def forward(self, x):
-            if x.layout == torch.sparse_coo:
-                return torch.sparse_coo_tensor(x.indices(), torch.relu(x.values()),
-                                               x.shape)
-            return torch.relu(x)
+            return _values_op(x, torch.relu)
+
+    class ReLU6(torch.nn.Module):
+        def forward(self, x):
+            return _values_op(x, torch.nn.functional.relu6)
+
+    class LeakyReLU(torch.nn.Module):
+        def __init__(self, negative_slope=0.01):
+            super().__init__()
+            self.ns = negative_slope
+
+        def forward(self, x):
+            return _values_op(x, lambda v: torch.nn.functional.leaky_relu(v, self.ns))
+
+    class Softmax(torch.nn.Module):
+        """CSR per-row softmax over stored values (reference semantics)."""
+
+        def __init__(self, axis=-1):
+            super().__init__()
+
+        def forward(self, x):
+            if getattr(x, "is_sparse", False):
+                return torch.sparse.softmax(x.coalesce(), dim=-1)
+            if x.layout == torch.sparse_csr:
+                coo = x.to_sparse_coo(2) if hasattr(x, "to_sparse_coo") else x.to_sparse()
+                return torch.sparse.softmax(coo.coalesce(), dim=-1).to_sparse_csr()
+            return torch.softmax(x, dim=-1)
+
+    class BatchNorm(torch.nn.Module):
+        """BatchNorm over the stored values' channel dim."""
+
+        def __init__(self, num_features, momentum=0.9, epsilon=1e-5,
+                     data_format="NDHWC", **kw):
+            super().__init__()
+            self.bn = torch.nn.BatchNorm1d(num_features, eps=epsilon,
+                                           momentum=1 - momentum)
+
+        def forward(self, x):
+            if getattr(x, "is_sparse", False):
+                xc = x.coalesce()
+                return torch.sparse_coo_tensor(xc.indices(),
+                                               self.bn(xc.values()), x.shape)
+            return self.bn(x)
+
+    SyncBatchNorm = BatchNorm
+
+    class _SparseConvNd(torch.nn.Module):
+        def __init__(self, in_channels, out_channels, kernel_size, stride=1,
+                     padding=0, dilation=1, groups=1, subm=False, nd=3,
+                     bias_attr=None, **kw):
+            super().__init__()
+            conv_cls = torch.nn.Conv3d if nd == 3 else torch.nn.Conv2d
+            self.conv = conv_cls(in_channels, out_channels, kernel_size,
+                                 stride=stride, padding=padding,
+                                 dilation=dilation, groups=groups,
+                                 bias=bias_attr is not False)
+            self.subm = subm
+            self.nd = nd
+
+        def forward(self, x):
+            # x: sparse COO [N, spatial..., C] (paddle channels-last)
+            dense = x.to_dense() if getattr(x, "is_sparse", False) else x
+            perm = (0, self.nd + 1) + tuple(range(1, self.nd + 1))
+            out = self.conv(dense.permute(*perm))
+            inv = (0,) + tuple(range(2, self.nd + 2)) + (1,)
+            out = out.permute(*inv).contiguous()
+            if self.subm and getattr(x, "is_sparse", False):
+                # submanifold: output nonzeros only at input active sites
+                mask = (x.to_dense().abs().sum(-1, keepdim=True) > 0)
+                out = out * mask
+            return out.to_sparse(self.nd + 1) if getattr(x, "is_sparse", False) else out
+
+    class Conv3D(_SparseConvNd):
+        def __init__(self, *a, **kw):
+            super().__init__(*a, nd=3, subm=False, **kw)
+
+    class SubmConv3D(_SparseConvNd):
+        def __init__(self, *a, **kw):
+            super().__init__(*a, nd=3, subm=True, **kw)
+
+    class Conv2D(_SparseConvNd):
+        def __init__(self, *a, **kw):
+            super().__init__(*a, nd=2, subm=False, **kw)
+
+    class SubmConv2D(_SparseConvNd):
+        def __init__(self, *a, **kw):
+            super().__init__(*a, nd=2, subm=True, **kw)
+
+    class MaxPool3D(torch.nn.Module):
+        def __init__(self, kernel_size, stride=None, padding=0, **kw):
+            super().__init__()
+            self.pool = torch.nn.MaxPool3d(kernel_size, stride, padding)
+
+        def forward(self, x):
+            dense = x.to_dense() if getattr(x, "is_sparse", False) else x
+            out = self.pool(dense.permute(0, 4, 1, 2, 3))
+            out = out.permute(0, 2, 3, 4, 1).contiguous()
+            return out.to_sparse(4) if getattr(x, "is_sparse", False) else out
 
 
 # -- elementwise/unary over sparse values (reference: sparse/unary.py) -------

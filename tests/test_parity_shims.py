@@ -216,3 +216,28 @@ def test_audio_io_roundtrip(tmp_path):
     assert sr == 8000 and meta.num_channels == 1
     assert (y - x).abs().max() < 1e-3
     assert paddle.profiler.SummaryView.KernelView == 4
+
+
+def test_sparse_nn_layers():
+    """paddle.sparse.nn (reference sparse/nn): activations over stored
+    values, channels-last sparse Conv3D/SubmConv3D/MaxPool3D (dense
+    round-trip semantics; SubmConv keeps the input's active sites)."""
+    import torch
+    import paddle_amd as paddle
+    nn = paddle.sparse.nn
+    i = torch.tensor([[0, 1], [1, 0]])
+    s = torch.sparse_coo_tensor(i, torch.tensor([-1.0, 5.0]), (2, 2))
+    assert nn.ReLU()(s).to_dense()[1, 0] == 5.0
+    assert abs(nn.LeakyReLU(0.1)(s).to_dense()[0, 1] + 0.1) < 1e-6
+    x = torch.zeros(1, 4, 4, 4, 2)
+    x[0, 1, 1, 1] = 1.0
+    x[0, 2, 3, 0] = 2.0
+    xs = x.to_sparse(4)
+    assert nn.Conv3D(2, 3, 3, padding=1)(xs).shape == (1, 4, 4, 4, 3)
+    so = nn.SubmConv3D(2, 3, 3, padding=1)(xs).to_dense()
+    nz = set(map(tuple, (so.abs().sum(-1) > 0).nonzero().tolist()))
+    assert nz == {(0, 1, 1, 1), (0, 2, 3, 0)}
+    assert nn.MaxPool3D(2)(xs).shape == (1, 2, 2, 2, 2)
+    assert nn.BatchNorm(2)(xs).shape == xs.shape
+    r = nn.Softmax()(torch.sparse_coo_tensor(i, torch.tensor([1.0, 2.0]), (2, 2)))
+    assert abs(float(r.to_dense().sum()) - 2.0) < 1e-5
