@@ -241,3 +241,36 @@ def test_sparse_nn_layers():
     assert nn.BatchNorm(2)(xs).shape == xs.shape
     r = nn.Softmax()(torch.sparse_coo_tensor(i, torch.tensor([1.0, 2.0]), (2, 2)))
     assert abs(float(r.to_dense().sum()) - 2.0) < 1e-5
+
+
+def test_semantics_spot_checks():
+    """Round-2 probe batch, kept as regressions: paddle-specific behaviors
+    verified against the reference's documented semantics."""
+    import numpy as np
+    import torch
+    import paddle_amd as paddle
+    x = paddle.to_tensor([[3., 1., 2.], [6., 5., 4.]])
+    # reductions return plain tensors (no namedtuples)
+    assert isinstance(paddle.max(x, axis=1), torch.Tensor)
+    assert isinstance(paddle.sort(x, axis=1), torch.Tensor)
+    # one_hot returns float
+    assert paddle.nn.functional.one_hot(paddle.to_tensor([1]), 4).dtype == torch.float32
+    # scatter overwrite vs add
+    base = paddle.zeros([4, 2])
+    idx = paddle.to_tensor([1, 1], dtype="int64")
+    upd = paddle.ones([2, 2])
+    assert paddle.scatter(base, idx, upd, overwrite=True)[1].tolist() == [1.0, 1.0]
+    assert paddle.scatter(base, idx, upd, overwrite=False)[1].tolist() == [2.0, 2.0]
+    # cumsum with no axis flattens
+    assert tuple(paddle.cumsum(x).shape) == (6,)
+    # multi-axis unsqueeze / squeeze
+    assert tuple(paddle.unsqueeze(x, [0, 2]).shape) == (1, 2, 1, 3)
+    assert tuple(paddle.squeeze(paddle.ones([1, 2, 1, 3]), [0, 2]).shape) == (2, 3)
+    # split with -1 section
+    assert [tuple(t.shape) for t in paddle.split(x, [1, -1], axis=1)] == [(2, 1), (2, 2)]
+    # summary/flops run
+    net = paddle.nn.Sequential(paddle.nn.Linear(4, 8), paddle.nn.ReLU(),
+                               paddle.nn.Linear(8, 2))
+    s = paddle.summary(net, (1, 4))
+    assert s["total_params"] == 58
+    assert paddle.flops(net, [1, 4], print_detail=False) > 0
