@@ -491,21 +491,23 @@ class LlamaModelRunner(GPTModelRunner):
         self._qkv_fused = []
         for layer in model.llama.layers:
             a = layer.self_attn
-            if hasattr(a.q_proj, "qweight"):          # weight-only int8
-                qw = torch.cat([a.q_proj.qweight, a.k_proj.qweight,
-                                a.v_proj.qweight], dim=1).contiguous()
-                sc = torch.cat([a.q_proj.scale, a.k_proj.scale,
-                                a.v_proj.scale]).contiguous()
+            projs = (a.q_proj, a.k_proj, a.v_proj)
+            if all(hasattr(p, "qweight") for p in projs):   # weight-only int8
+                qw = torch.cat([p.qweight for p in projs], dim=1).contiguous()
+                sc = torch.cat([p.scale for p in projs]).contiguous()
                 self._qkv_fused.append(("int8", qw, sc))
-            else:
-                w = torch.cat([a.q_proj.weight, a.k_proj.weight,
-                               a.v_proj.weight], dim=1).contiguous()
+            elif all(getattr(p, "weight", None) is not None for p in projs):
+                w = torch.cat([p.weight for p in projs], dim=1).contiguous()
                 qo = a.q_proj.weight.shape[1]
                 ko = a.k_proj.weight.shape[1]
                 a.q_proj.weight.data = w[:, :qo]
                 a.k_proj.weight.data = w[:, qo:qo + ko]
                 a.v_proj.weight.data = w[:, qo + ko:]
                 self._qkv_fused.append(("bf16", w, None))
+            else:
+                # mixed (e.g. GQA k/v under quantize_linears_' min_features
+                # cutoff while q was converted): run the module calls
+                self._qkv_fused.append(("none", None, None))
 
     def _rope_rows(self, x, pos):
         """x [B, 1, H, D] at per-row absolute positions pos [B] (long)."""
@@ -562,11 +564,15 @@ class LlamaModelRunner(GPTModelRunner):
             h = layer.input_layernorm(x)
             kind, w, sc = self._qkv_fused[li]
             h2 = h.reshape(B, -1)
+            at = layer.self_attn
             if kind == "int8":
                 from .quantization import weight_only_linear
                 qkv = weight_only_linear(h2, w, sc)
-            else:
+            elif kind == "bf16":
                 qkv = torch.matmul(h2, w)
+            else:
+                qkv = torch.cat([at.q_proj(h2), at.k_proj(h2), at.v_proj(h2)],
+                                dim=-1)
             q = qkv[:, :qo].reshape(B, 1, self.H, self.D)
             k = qkv[:, qo:qo + ko].reshape(B, 1, self.HKV, self.D)
             v = qkv[:, qo + ko:].reshape(B, 1, self.HKV, self.D)

@@ -162,3 +162,28 @@ def test_llama_runner_end_to_end_cpu():
     assert len(eng.alloc.free) == 64
     ref = generate_llama(m, torch.tensor([[3, 7, 11, 20]]), max_new_tokens=5)
     assert reqs[0].out_ids == ref[0].tolist()
+
+
+def test_llama_runner_weight_only_mixed_cpu():
+    """weight_only llama with GQA k/v below quantize_linears_'
+    min_features: q is int8, k/v stay bf16 -- the fused-qkv init must
+    fall back to module calls instead of crashing."""
+    import paddle_amd as paddle
+    from paddle_amd.models.llama import LlamaConfig, LlamaForCausalLM
+    from paddle_amd.serving import Engine, LlamaModelRunner, Request
+    paddle.seed(0)
+    cfg = LlamaConfig(vocab_size=512, hidden_size=1024, num_layers=2,
+                      num_heads=8, num_kv_heads=4, intermediate_size=1024,
+                      max_seq_len=128)
+    m = LlamaForCausalLM(cfg).to("cpu").float().eval()
+    runner = LlamaModelRunner(m, num_blocks=64, block_size=16,
+                              device=torch.device("cpu"), max_seq=128,
+                              weight_only=True)
+    assert any(k[0] == "none" for k in runner._qkv_fused)
+    eng = Engine(runner, num_blocks=64, block_size=16, max_batch=2)
+    reqs = [Request(prompt_ids=[3, 7, 11, 20], max_new_tokens=4)
+            for _ in range(2)]
+    for r in reqs:
+        eng.add_request(r)
+    eng.run_until_done()
+    assert all(r.done and len(r.out_ids) == 4 for r in reqs)
