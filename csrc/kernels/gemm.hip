@@ -164,7 +164,7 @@ __global__ void gemm_bf16_kernel(const short* __restrict__ ag, const short* __re
         (__attribute__((address_space(3))) unsigned int*)&a_lds[buf][it * 4096 + tid * 8],
         16, 0, 0);
   };
-  auto stage_b_round = [&](int buf, int kt, int it) {
+  [[maybe_unused]] auto stage_b_round = [&](int buf, int kt, int it) {
     const long long k0 = (long long)kt * BK;
     const short* src = bg + (long long)(col0 + s_r[it]) * ldb + k0 + s_cp[it];
     __builtin_amdgcn_global_load_lds(
